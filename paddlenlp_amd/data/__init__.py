@@ -1,0 +1,7 @@
+from .data_collator import (  # noqa: F401
+    DataCollatorForLanguageModeling,
+    DataCollatorForSeq2Seq,
+    DataCollatorWithPadding,
+    default_data_collator,
+)
+from .sampler import DistributedBatchSampler  # noqa: F401

@@ -1,0 +1,254 @@
+// Python bindings for the paddlenlp_amd gfx950 kernel pack.
+// Built by setup.py via torch.utils.cpp_extension (hipcc, PYTORCH_ROCM_ARCH=gfx950).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <vector>
+
+// ---- launcher decls (implemented in the .hip files) ----
+void launch_rms_norm_fwd(const void*, const void*, void*, float*, long long, int, float, hipStream_t);
+void launch_rms_norm_bwd(const void*, const void*, const void*, const float*, void*, float*, void*, bool, long long, int, int, hipStream_t);
+void launch_rope(const void*, void*, const float*, const float*, long long, int, int, int, bool, hipStream_t);
+void launch_swiglu_fwd(const void*, void*, long long, int, hipStream_t);
+void launch_swiglu_bwd(const void*, const void*, void*, long long, int, hipStream_t);
+void launch_ce_fwd(const void*, const long long*, float*, float*, long long, int, long long, hipStream_t);
+void launch_ce_bwd(const float*, const void*, const long long*, const float*, void*, long long, int, long long, hipStream_t);
+void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
+void launch_flash_fwd(const void*, const void*, const void*, void*, float*, int, int, int, int, int, int, float, bool, hipStream_t);
+void launch_flash_bwd(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, int, int, int, int, int, int, float, bool, hipStream_t);
+
+struct AdamWChunk {
+    void* param;
+    const void* grad;
+    float* m;
+    float* v;
+    float* master;
+    long long offset;
+    long long n;
+    int is_bf16;
+};
+void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, float, float, hipStream_t);
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+#define CHECK_BF16(x) TORCH_CHECK(x.scalar_type() == torch::kBFloat16, #x " must be bf16")
+
+static hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> rms_norm_fwd(torch::Tensor x, torch::Tensor w, double eps) {
+    CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x); CHECK_BF16(w);
+    int H = x.size(-1);
+    TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+    long long rows = x.numel() / H;
+    auto y = torch::empty_like(x);
+    auto invrms = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+    launch_rms_norm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                        invrms.data_ptr<float>(), rows, H, (float)eps, cur_stream());
+    return {y, invrms};
+}
+
+std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor w,
+                                        torch::Tensor invrms) {
+    CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+    int H = x.size(-1);
+    long long rows = x.numel() / H;
+    int P = (int)std::min<long long>(rows, 2048);
+    auto dx = torch::empty_like(x);
+    auto dw_partial = torch::zeros({P, H}, x.options().dtype(torch::kFloat32));
+    auto dw = torch::empty_like(w);
+    launch_rms_norm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                        invrms.data_ptr<float>(), dx.data_ptr(),
+                        dw_partial.data_ptr<float>(), dw.data_ptr(),
+                        w.scalar_type() == torch::kBFloat16, rows, H, P, cur_stream());
+    return {dx, dw};
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> rope_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor cos_t, torch::Tensor sin_t,
+                                    bool backward) {
+    CHECK_GPU(q); CHECK_CONTIG(q); CHECK_BF16(q); CHECK_CONTIG(k); CHECK_BF16(k);
+    TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "q/k must be [B,S,H,D]");
+    int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Hk = k.size(2);
+    TORCH_CHECK(D % 8 == 0, "head dim must be a multiple of 8");
+    auto cf = cos_t.to(torch::kFloat32).contiguous();
+    auto sf = sin_t.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(cf.size(0) == S && cf.size(-1) == D, "cos table must be [S, D]");
+    auto q_out = torch::empty_like(q);
+    auto k_out = torch::empty_like(k);
+    launch_rope(q.data_ptr(), q_out.data_ptr(), cf.data_ptr<float>(), sf.data_ptr<float>(),
+                (long long)B * S * Hq, Hq, D, S, backward, cur_stream());
+    launch_rope(k.data_ptr(), k_out.data_ptr(), cf.data_ptr<float>(), sf.data_ptr<float>(),
+                (long long)B * S * Hk, Hk, D, S, backward, cur_stream());
+    return {q_out, k_out};
+}
+
+// ---------------------------------------------------------------------------
+torch::Tensor swiglu_fwd(torch::Tensor x) {
+    CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+    int twoI = x.size(-1);
+    TORCH_CHECK(twoI % 16 == 0, "last dim must be a multiple of 16");
+    int I = twoI / 2;
+    long long N = x.numel() / twoI;
+    auto sizes = x.sizes().vec();
+    sizes.back() = I;
+    auto y = torch::empty(sizes, x.options());
+    launch_swiglu_fwd(x.data_ptr(), y.data_ptr(), N, I, cur_stream());
+    return y;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor x) {
+    CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+    int twoI = x.size(-1);
+    int I = twoI / 2;
+    long long N = x.numel() / twoI;
+    auto dx = torch::empty_like(x);
+    launch_swiglu_bwd(dy.data_ptr(), x.data_ptr(), dx.data_ptr(), N, I, cur_stream());
+    return dx;
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits, torch::Tensor labels,
+                                             int64_t ignore_index) {
+    CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+    TORCH_CHECK(logits.dim() == 2, "logits must be [N, V]");
+    long long N = logits.size(0);
+    int V = logits.size(1);
+    TORCH_CHECK(V % 8 == 0, "vocab must be a multiple of 8");
+    auto labels64 = labels.to(torch::kInt64).contiguous();
+    auto loss = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+    auto maxlse = torch::empty({N, 2}, logits.options().dtype(torch::kFloat32));
+    launch_ce_fwd(logits.data_ptr(), labels64.data_ptr<long long>(),
+                  loss.data_ptr<float>(), maxlse.data_ptr<float>(), N, V,
+                  ignore_index, cur_stream());
+    return {loss, maxlse};
+}
+
+torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logits,
+                                torch::Tensor labels, torch::Tensor maxlse,
+                                int64_t ignore_index) {
+    CHECK_GPU(dloss);
+    long long N = logits.size(0);
+    int V = logits.size(1);
+    auto labels64 = labels.to(torch::kInt64).contiguous();
+    auto dl = dloss.to(torch::kFloat32).contiguous();
+    auto dlogits = torch::empty_like(logits);
+    launch_ce_bwd(dl.data_ptr<float>(), logits.data_ptr(),
+                  labels64.data_ptr<long long>(), maxlse.data_ptr<float>(),
+                  dlogits.data_ptr(), N, V, ignore_index, cur_stream());
+    return dlogits;
+}
+
+// ---------------------------------------------------------------------------
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+    CHECK_GPU(A); CHECK_BF16(A);
+    auto C = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+    launch_mfma_probe(A.contiguous().data_ptr(), B.contiguous().data_ptr(),
+                      C.data_ptr<float>(), cur_stream());
+    return C;
+}
+
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                          bool causal) {
+    CHECK_GPU(q); CHECK_CONTIG(q); CHECK_BF16(q);
+    CHECK_CONTIG(k); CHECK_CONTIG(v);
+    TORCH_CHECK(q.dim() == 4, "q must be [B,S,H,D]");
+    int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Skv = k.size(1), Hk = k.size(2);
+    TORCH_CHECK(D == 128 || D == 64 || D == 32, "head dim must be 32/64/128");
+    TORCH_CHECK(Hq % Hk == 0, "GQA requires Hq % Hk == 0");
+    auto o = torch::empty_like(q);
+    auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     lse.data_ptr<float>(), B, Sq, Skv, Hq, Hk, D, scale, causal,
+                     cur_stream());
+    return {o, lse};
+}
+
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                          torch::Tensor k, torch::Tensor v,
+                                          torch::Tensor o, torch::Tensor lse,
+                                          bool causal) {
+    CHECK_GPU(dout); CHECK_CONTIG(dout); CHECK_BF16(dout);
+    int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Skv = k.size(1), Hk = k.size(2);
+    auto dq = torch::empty_like(q);
+    // per-q-head buffers; summed over GQA groups below
+    auto dk_h = torch::empty({B, Skv, Hq, D}, k.options());
+    auto dv_h = torch::empty({B, Skv, Hq, D}, v.options());
+    auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_flash_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                     o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(),
+                     B, Sq, Skv, Hq, Hk, D, scale, causal, cur_stream());
+    torch::Tensor dk, dv;
+    if (Hq == Hk) {
+        dk = dk_h;
+        dv = dv_h;
+    } else {
+        int G = Hq / Hk;
+        dk = dk_h.view({B, Skv, Hk, G, D}).sum(3);
+        dv = dv_h.view({B, Skv, Hk, G, D}).sum(3);
+    }
+    return {dq, dk, dv};
+}
+
+// ---------------------------------------------------------------------------
+void fused_adamw(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> exp_avgs, std::vector<torch::Tensor> exp_avg_sqs,
+                 std::vector<torch::Tensor> masters,
+                 double lr, double beta1, double beta2, double eps, double wd,
+                 int64_t step) {
+    TORCH_CHECK(!params.empty());
+    bool has_master = !masters.empty();
+    const long long CHUNK = 1 << 22;  // 4M elements per chunk
+    std::vector<AdamWChunk> chunks;
+    for (size_t i = 0; i < params.size(); i++) {
+        auto& p = params[i];
+        CHECK_GPU(p); CHECK_CONTIG(p);
+        bool is_bf16 = p.scalar_type() == torch::kBFloat16;
+        TORCH_CHECK(!is_bf16 || has_master, "bf16 params need master weights");
+        long long n = p.numel();
+        torch::Tensor g = grads[i];
+        if (g.scalar_type() != p.scalar_type()) g = g.to(p.scalar_type());
+        for (long long off = 0; off < n; off += CHUNK) {
+            AdamWChunk c;
+            c.param = p.data_ptr();
+            c.grad = g.data_ptr();
+            c.m = exp_avgs[i].data_ptr<float>();
+            c.v = exp_avg_sqs[i].data_ptr<float>();
+            c.master = has_master ? masters[i].data_ptr<float>() : nullptr;
+            c.offset = off;
+            c.n = std::min(CHUNK, n - off);
+            c.is_bf16 = is_bf16 ? 1 : 0;
+            chunks.push_back(c);
+        }
+    }
+    auto meta = torch::from_blob(chunks.data(), {(long long)(chunks.size() * sizeof(AdamWChunk))},
+                                 torch::kUInt8).clone();
+    auto dev_meta = meta.to(params[0].device(), /*non_blocking=*/true);
+    float bias1 = 1.0f - std::pow((float)beta1, (float)step);
+    float bias2 = 1.0f - std::pow((float)beta2, (float)step);
+    launch_adamw(reinterpret_cast<const AdamWChunk*>(dev_meta.data_ptr()),
+                 (int)chunks.size(), (float)lr, (float)beta1, (float)beta2,
+                 (float)eps, (float)wd, bias1, bias2, cur_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rms_norm_fwd", &rms_norm_fwd);
+    m.def("rms_norm_bwd", &rms_norm_bwd);
+    m.def("rope_fwd", &rope_fwd);
+    m.def("swiglu_fwd", &swiglu_fwd);
+    m.def("swiglu_bwd", &swiglu_bwd);
+    m.def("cross_entropy_fwd", &cross_entropy_fwd);
+    m.def("cross_entropy_bwd", &cross_entropy_bwd);
+    m.def("mfma_probe", &mfma_probe);
+    m.def("flash_attn_fwd", &flash_attn_fwd);
+    m.def("flash_attn_bwd", &flash_attn_bwd);
+    m.def("fused_adamw", &fused_adamw);
+}

@@ -1,0 +1,694 @@
+// FlashAttention-2 forward + backward — hand-written gfx950 (CDNA4) MFMA.
+//
+// MI355X-native replacement for the reference's training attention
+// (paddle `flash_attention` fused op; the csrc append_attn
+// family is the inference-side sibling).  Not a CUDA port: tiles are sized
+// for 64-wide waves and mfma_f32_16x16x32_bf16, K/V staged through LDS with
+// +8-element padded rows (≤2-way bank aliasing, free on CDNA4 per guide G4),
+// fp32 online-softmax state in registers.
+//
+// v1 structure (correctness-first; the 8-wave 32x32 swizzled "ladder"
+// structure is the planned optimization):
+//   block = 4 waves (256 thr); Q tile 64 rows (16/wave); KV tile 64.
+//
+// MFMA fragment layouts for mfma_f32_16x16x32_bf16 (HW-validated by the
+// mfma_layout_probe kernel + tests/test_ops_gpu.py asymmetric check):
+//   A (16x32): lane l holds A[l%16][(l/16)*8 + j], j=0..7   (one short8)
+//   B (32x16): lane l holds B[(l/16)*8 + j][l%16]
+//   C (16x16): lane l, reg r holds C[(l/16)*4 + r][l%16]
+//
+// Operand-layout cheat sheet (X row-major [row][col] in LDS):
+//   "B-frag contraction over col of X" (e.g. S=QK^T contracts d):
+//       read x_lds[n*16 + l16][kk*32 + lk8 .. +8]        (contiguous ✓)
+//   "B-frag contraction over row of X" (e.g. O=PV contracts kv):
+//       needs X transposed in LDS: xt_lds[col][row]
+#include "common.h"
+
+#define FA_WAVES 4
+#define FA_BLOCK (FA_WAVES * 64)
+#define BLK_M 64
+#define BLK_N 64
+#define LDS_PAD 8
+
+typedef short8v frag_ab;
+typedef f32x4 frag_c;
+
+__device__ __forceinline__ frag_c mfma16(frag_ab a, frag_ab b, frag_c c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// ---------------------------------------------------------------------------
+// layout probe: one wave computes C = A @ B for 16x32 @ 32x16.  Used by the
+// GPU layout-validation test with asymmetric random inputs (transpose-
+// detecting, guide G9).
+// ---------------------------------------------------------------------------
+__global__ void mfma_layout_probe(const ushort_t* A, const ushort_t* B, float* C) {
+    int l = threadIdx.x;
+    frag_ab a, b;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+        a[j] = (short)A[(l % 16) * 32 + (l / 16) * 8 + j];
+        b[j] = (short)B[((l / 16) * 8 + j) * 16 + (l % 16)];
+    }
+    frag_c c = {0.f, 0.f, 0.f, 0.f};
+    c = mfma16(a, b, c);
+#pragma unroll
+    for (int r = 0; r < 4; r++) C[((l / 16) * 4 + r) * 16 + (l % 16)] = c[r];
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
+    const ushort_t* __restrict__ q,   // [B, Sq, Hq, D]
+    const ushort_t* __restrict__ k,   // [B, Skv, Hk, D]
+    const ushort_t* __restrict__ v,   // [B, Skv, Hk, D]
+    ushort_t* __restrict__ o,         // [B, Sq, Hq, D]
+    float* __restrict__ lse,          // [B, Hq, Sq]
+    int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
+    constexpr int KD = D / 32;   // MFMA K-steps over the head dim
+    constexpr int ND = D / 16;   // d-frags of the O accumulator
+    constexpr int NN = BLK_N / 16;
+    constexpr int LDK = D + LDS_PAD;
+    constexpr int LDT = BLK_N + LDS_PAD;
+
+    __shared__ ushort_t k_lds[BLK_N][LDK];   // K row-major [kv][d]
+    __shared__ ushort_t vt_lds[D][LDT];      // V transposed [d][kv]
+    __shared__ ushort_t p_lds[BLK_M][LDT];   // P row-major [q][kv]
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int lk8 = (lane >> 4) * 8;
+
+    const int qt = blockIdx.x;
+    const int bh = blockIdx.y;
+    const int b = bh / Hq, hq = bh % Hq;
+    const int hk = hq / (Hq / Hk);
+    const int q_base = qt * BLK_M;
+    const int causal_off = Skv - Sq;  // kv visible iff kv <= q + off
+
+    const long long q_row_stride = (long long)Hq * D;
+    const long long kv_row_stride = (long long)Hk * D;
+    const ushort_t* q_ptr = q + ((long long)b * Sq * Hq + hq) * D;
+    const ushort_t* k_ptr = k + ((long long)b * Skv * Hk + hk) * D;
+    const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
+
+    // Q A-frags in registers (16 rows x D per wave)
+    frag_ab aq[KD];
+    {
+        int qrow = q_base + wave * 16 + l16;
+        if (qrow < Sq) {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++)
+                aq[kk] = *reinterpret_cast<const frag_ab*>(
+                    q_ptr + (long long)qrow * q_row_stride + kk * 32 + lk8);
+        } else {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) aq[kk] = frag_ab{0};
+        }
+    }
+
+    float m_run[4], l_run[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+    frag_c acc_o[ND];
+#pragma unroll
+    for (int n = 0; n < ND; n++) acc_o[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+
+    int n_kv_tiles = (Skv + BLK_N - 1) / BLK_N;
+    if (causal) {
+        int max_kv = q_base + BLK_M - 1 + causal_off;
+        int lim = (max_kv + BLK_N) / BLK_N;
+        n_kv_tiles = min(n_kv_tiles, max(lim, 0));
+    }
+
+    for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
+        const int kv_base = kvt * BLK_N;
+        for (int idx = tid * 8; idx < BLK_N * D; idx += FA_BLOCK * 8) {
+            int row = idx / D, col = idx % D;
+            int kvg = kv_base + row;
+            short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
+            short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (kvg < Skv) {
+                k8 = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg * kv_row_stride + col);
+                v8 = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg * kv_row_stride + col);
+            }
+            *reinterpret_cast<short8v*>(&k_lds[row][col]) = k8;
+#pragma unroll
+            for (int j = 0; j < 8; j++) vt_lds[col + j][row] = (ushort_t)v8[j];
+        }
+        __syncthreads();
+
+        // S = Q K^T  (contract d; B-frag from K row-major)
+        frag_c acc_s[NN];
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) {
+                frag_ab bk = *reinterpret_cast<const frag_ab*>(
+                    &k_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_s[n] = mfma16(aq[kk], bk, acc_s[n]);
+            }
+        }
+
+        // scale + mask (C layout: row q = qrow0 + r, col kv)
+        const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            int kvg = kv_base + n * 16 + l16;
+#pragma unroll
+            for (int r = 0; r < 4; r++) {
+                int qg = qrow0 + r;
+                bool vis = (kvg < Skv) && (qg < Sq);
+                if (causal) vis = vis && (kvg <= qg + causal_off);
+                acc_s[n][r] = vis ? acc_s[n][r] * scale : -INFINITY;
+            }
+        }
+
+        // online softmax update
+        float p_tile[NN][4];
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            float row_max = -INFINITY;
+#pragma unroll
+            for (int n = 0; n < NN; n++) row_max = fmaxf(row_max, acc_s[n][r]);
+#pragma unroll
+            for (int off = 8; off > 0; off >>= 1)
+                row_max = fmaxf(row_max, __shfl_xor(row_max, off, 64));
+            float m_new = fmaxf(m_run[r], row_max);
+            float alpha;
+            if (m_new == -INFINITY) {            // nothing visible yet
+                alpha = 1.f;
+            } else if (m_run[r] == -INFINITY) {  // first visible tile
+                alpha = 0.f;
+            } else {
+                alpha = __expf(m_run[r] - m_new);
+            }
+            float row_sum = 0.f;
+#pragma unroll
+            for (int n = 0; n < NN; n++) {
+                float p = (m_new == -INFINITY || acc_s[n][r] == -INFINITY)
+                              ? 0.f : __expf(acc_s[n][r] - m_new);
+                p_tile[n][r] = p;
+                row_sum += p;
+            }
+#pragma unroll
+            for (int off = 8; off > 0; off >>= 1) row_sum += __shfl_xor(row_sum, off, 64);
+            l_run[r] = l_run[r] * alpha + row_sum;
+            m_run[r] = m_new;
+#pragma unroll
+            for (int n = 0; n < ND; n++) acc_o[n][r] *= alpha;
+        }
+
+        // P -> LDS (C layout out, A layout back in)
+#pragma unroll
+        for (int n = 0; n < NN; n++)
+#pragma unroll
+            for (int r = 0; r < 4; r++)
+                p_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                    f32_to_bf16(p_tile[n][r]);
+        __syncthreads();
+
+        // O += P V  (contract kv; A from p_lds, B from vt_lds)
+#pragma unroll
+        for (int kk = 0; kk < BLK_N / 32; kk++) {
+            frag_ab ap = *reinterpret_cast<const frag_ab*>(
+                &p_lds[wave * 16 + l16][kk * 32 + lk8]);
+#pragma unroll
+            for (int n = 0; n < ND; n++) {
+                frag_ab bv = *reinterpret_cast<const frag_ab*>(
+                    &vt_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_o[n] = mfma16(ap, bv, acc_o[n]);
+            }
+        }
+        __syncthreads();
+    }
+
+    // epilogue: O/l and LSE
+    const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+        int qg = qrow0 + r;
+        if (qg >= Sq) continue;
+        float inv_l = (l_run[r] > 0.f) ? 1.0f / l_run[r] : 0.f;
+        ushort_t* orow = o + ((long long)b * Sq + qg) * q_row_stride + (long long)hq * D;
+#pragma unroll
+        for (int n = 0; n < ND; n++)
+            orow[n * 16 + l16] = f32_to_bf16(acc_o[n][r] * inv_l);
+        if (l16 == 0) {
+            float lv = (l_run[r] > 0.f) ? (m_run[r] + __logf(l_run[r])) : -INFINITY;
+            lse[((long long)b * Hq + hq) * Sq + qg] = lv;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// backward preprocess: delta[b,h,q] = rowsum(dO * O)
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ void flash_bwd_delta_kernel(
+    const ushort_t* __restrict__ dout, const ushort_t* __restrict__ o,
+    float* __restrict__ delta, int B, int Sq, int Hq) {
+    long long row = (long long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    long long total = (long long)B * Sq * Hq;
+    if (row >= total) return;
+    int lane = threadIdx.x & 63;
+    const ushort_t* dor = dout + row * D;
+    const ushort_t* orow = o + row * D;
+    float acc = 0.f;
+    for (int i = lane * 2; i < D; i += 128) {
+        acc += bf16_to_f32(dor[i]) * bf16_to_f32(orow[i]);
+        acc += bf16_to_f32(dor[i + 1]) * bf16_to_f32(orow[i + 1]);
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) {
+        long long h = row % Hq;
+        long long bq = row / Hq;
+        long long bb = bq / Sq, qi = bq % Sq;
+        delta[(bb * Hq + h) * Sq + qi] = acc;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// backward dQ: fixed q tile, loop kv tiles.
+//   S = scale*QK^T; P = exp(S - lse); dP = dO V^T;
+//   dS = P*(dP - delta)*scale; dQ += dS K
+// LDS: K row-major (S B-frag), V row-major (dP B-frag),
+//      K^T (dQ B-frag), dS staging.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
+    const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    ushort_t* __restrict__ dq,
+    int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
+    constexpr int KD = D / 32;
+    constexpr int ND = D / 16;
+    constexpr int NN = BLK_N / 16;
+    constexpr int LDK = D + LDS_PAD;
+    constexpr int LDT = BLK_N + LDS_PAD;
+
+    __shared__ ushort_t k_lds[BLK_N][LDK];
+    __shared__ ushort_t v_lds[BLK_N][LDK];
+    __shared__ ushort_t kt_lds[D][LDT];
+    __shared__ ushort_t ds_lds[BLK_M][LDT];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int lk8 = (lane >> 4) * 8;
+
+    const int qt = blockIdx.x;
+    const int bh = blockIdx.y;
+    const int b = bh / Hq, hq = bh % Hq;
+    const int hk = hq / (Hq / Hk);
+    const int q_base = qt * BLK_M;
+    const int causal_off = Skv - Sq;
+
+    const long long q_row_stride = (long long)Hq * D;
+    const long long kv_row_stride = (long long)Hk * D;
+    const ushort_t* q_ptr = q + ((long long)b * Sq * Hq + hq) * D;
+    const ushort_t* k_ptr = k + ((long long)b * Skv * Hk + hk) * D;
+    const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
+    const ushort_t* do_ptr = dout + ((long long)b * Sq * Hq + hq) * D;
+    const float* lse_row = lse + ((long long)b * Hq + hq) * Sq;
+    const float* dl_row = delta + ((long long)b * Hq + hq) * Sq;
+
+    frag_ab aq[KD], ado[KD];
+    {
+        int qrow = q_base + wave * 16 + l16;
+        if (qrow < Sq) {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) {
+                aq[kk] = *reinterpret_cast<const frag_ab*>(
+                    q_ptr + (long long)qrow * q_row_stride + kk * 32 + lk8);
+                ado[kk] = *reinterpret_cast<const frag_ab*>(
+                    do_ptr + (long long)qrow * q_row_stride + kk * 32 + lk8);
+            }
+        } else {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) { aq[kk] = frag_ab{0}; ado[kk] = frag_ab{0}; }
+        }
+    }
+    const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
+    float lse_r[4], dl_r[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+        int qg = qrow0 + r;
+        lse_r[r] = (qg < Sq) ? lse_row[qg] : INFINITY;
+        dl_r[r] = (qg < Sq) ? dl_row[qg] : 0.f;
+    }
+
+    frag_c acc_dq[ND];
+#pragma unroll
+    for (int n = 0; n < ND; n++) acc_dq[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+
+    int n_kv_tiles = (Skv + BLK_N - 1) / BLK_N;
+    if (causal) {
+        int max_kv = q_base + BLK_M - 1 + causal_off;
+        int lim = (max_kv + BLK_N) / BLK_N;
+        n_kv_tiles = min(n_kv_tiles, max(lim, 0));
+    }
+
+    for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
+        const int kv_base = kvt * BLK_N;
+        for (int idx = tid * 8; idx < BLK_N * D; idx += FA_BLOCK * 8) {
+            int row = idx / D, col = idx % D;
+            int kvg = kv_base + row;
+            short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
+            short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (kvg < Skv) {
+                k8 = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg * kv_row_stride + col);
+                v8 = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg * kv_row_stride + col);
+            }
+            *reinterpret_cast<short8v*>(&k_lds[row][col]) = k8;
+            *reinterpret_cast<short8v*>(&v_lds[row][col]) = v8;
+#pragma unroll
+            for (int j = 0; j < 8; j++) kt_lds[col + j][row] = (ushort_t)k8[j];
+        }
+        __syncthreads();
+
+        // S and dP
+        frag_c acc_s[NN], acc_dp[NN];
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+            acc_dp[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) {
+                frag_ab bk = *reinterpret_cast<const frag_ab*>(
+                    &k_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bv = *reinterpret_cast<const frag_ab*>(
+                    &v_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_s[n] = mfma16(aq[kk], bk, acc_s[n]);
+                acc_dp[n] = mfma16(ado[kk], bv, acc_dp[n]);
+            }
+        }
+
+        // dS = P * (dP - delta) * scale   (C layout)
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            int kvg = kv_base + n * 16 + l16;
+#pragma unroll
+            for (int r = 0; r < 4; r++) {
+                int qg = qrow0 + r;
+                bool vis = (kvg < Skv) && (qg < Sq);
+                if (causal) vis = vis && (kvg <= qg + causal_off);
+                float p = vis ? __expf(acc_s[n][r] * scale - lse_r[r]) : 0.f;
+                float ds = p * (acc_dp[n][r] - dl_r[r]) * scale;
+                ds_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] = f32_to_bf16(ds);
+            }
+        }
+        __syncthreads();
+
+        // dQ += dS K  (contract kv; B from kt_lds)
+#pragma unroll
+        for (int kk = 0; kk < BLK_N / 32; kk++) {
+            frag_ab ads = *reinterpret_cast<const frag_ab*>(
+                &ds_lds[wave * 16 + l16][kk * 32 + lk8]);
+#pragma unroll
+            for (int n = 0; n < ND; n++) {
+                frag_ab bkt = *reinterpret_cast<const frag_ab*>(
+                    &kt_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_dq[n] = mfma16(ads, bkt, acc_dq[n]);
+            }
+        }
+        __syncthreads();
+    }
+
+    // store dQ
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+        int qg = qrow0 + r;
+        if (qg >= Sq) continue;
+        ushort_t* dqr = dq + ((long long)b * Sq + qg) * q_row_stride + (long long)hq * D;
+#pragma unroll
+        for (int n = 0; n < ND; n++)
+            dqr[n * 16 + l16] = f32_to_bf16(acc_dq[n][r]);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// backward dK/dV: fixed kv tile, loop q tiles.  Wave owns 16 kv rows.
+//   S^T = scale*K Q^T;  P^T = exp(S^T - lse[q]);
+//   dV += P^T dO;  dP^T = V dO^T;  dS^T = P^T*(dP^T - delta[q])*scale;
+//   dK += dS^T Q
+// GQA: outputs are PER Q-HEAD ([B, Skv, Hq, D]); python sums head groups.
+// LDS: Q row-major (S^T B), Q^T (dK B), dO row-major (dP^T B),
+//      dO^T (dV B), P^T/dS^T staging (one reused buffer).
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
+    const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    ushort_t* __restrict__ dk,   // [B, Skv, Hq, D] (per q-head)
+    ushort_t* __restrict__ dv,   // [B, Skv, Hq, D]
+    int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
+    constexpr int KD = D / 32;
+    constexpr int ND = D / 16;
+    constexpr int NN = BLK_M / 16;  // q-frags per tile
+    constexpr int LDK = D + LDS_PAD;
+    constexpr int LDT = BLK_M + LDS_PAD;
+
+    __shared__ ushort_t q_lds[BLK_M][LDK];
+    __shared__ ushort_t do_lds[BLK_M][LDK];
+    __shared__ ushort_t qt_lds[D][LDT];
+    __shared__ ushort_t dot_lds[D][LDT];
+    __shared__ ushort_t pt_lds[BLK_N][LDT];  // P^T then dS^T (reused)
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int lk8 = (lane >> 4) * 8;
+
+    const int kvt = blockIdx.x;
+    const int bh = blockIdx.y;
+    const int b = bh / Hq, hq = bh % Hq;
+    const int hk = hq / (Hq / Hk);
+    const int kv_base = kvt * BLK_N;
+    const int causal_off = Skv - Sq;
+
+    const long long q_row_stride = (long long)Hq * D;
+    const long long kv_row_stride = (long long)Hk * D;
+    const ushort_t* q_ptr = q + ((long long)b * Sq * Hq + hq) * D;
+    const ushort_t* k_ptr = k + ((long long)b * Skv * Hk + hk) * D;
+    const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
+    const ushort_t* do_ptr = dout + ((long long)b * Sq * Hq + hq) * D;
+    const float* lse_row = lse + ((long long)b * Hq + hq) * Sq;
+    const float* dl_row = delta + ((long long)b * Hq + hq) * Sq;
+
+    // K and V A-frags in registers (wave's 16 kv rows, fixed all kernel)
+    frag_ab ak[KD], av[KD];
+    {
+        int kvrow = kv_base + wave * 16 + l16;
+        if (kvrow < Skv) {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) {
+                ak[kk] = *reinterpret_cast<const frag_ab*>(
+                    k_ptr + (long long)kvrow * kv_row_stride + kk * 32 + lk8);
+                av[kk] = *reinterpret_cast<const frag_ab*>(
+                    v_ptr + (long long)kvrow * kv_row_stride + kk * 32 + lk8);
+            }
+        } else {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) { ak[kk] = frag_ab{0}; av[kk] = frag_ab{0}; }
+        }
+    }
+
+    frag_c acc_dk[ND], acc_dv[ND];
+#pragma unroll
+    for (int n = 0; n < ND; n++) {
+        acc_dk[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+        acc_dv[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+    }
+
+    int qt_start = 0;
+    if (causal) {
+        // first q that can see kv_base: q >= kv_base - off
+        int first_q = kv_base - causal_off;
+        if (first_q > 0) qt_start = first_q / BLK_M;
+    }
+    int n_q_tiles = (Sq + BLK_M - 1) / BLK_M;
+
+    for (int qt = qt_start; qt < n_q_tiles; qt++) {
+        const int q_base = qt * BLK_M;
+        for (int idx = tid * 8; idx < BLK_M * D; idx += FA_BLOCK * 8) {
+            int row = idx / D, col = idx % D;
+            int qg = q_base + row;
+            short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
+            short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (qg < Sq) {
+                q8 = *reinterpret_cast<const short8v*>(q_ptr + (long long)qg * q_row_stride + col);
+                d8 = *reinterpret_cast<const short8v*>(do_ptr + (long long)qg * q_row_stride + col);
+            }
+            *reinterpret_cast<short8v*>(&q_lds[row][col]) = q8;
+            *reinterpret_cast<short8v*>(&do_lds[row][col]) = d8;
+#pragma unroll
+            for (int j = 0; j < 8; j++) {
+                qt_lds[col + j][row] = (ushort_t)q8[j];
+                dot_lds[col + j][row] = (ushort_t)d8[j];
+            }
+        }
+        __syncthreads();
+
+        // S^T = K Q^T and dP^T = V dO^T   (M = kv, N = q, contract d)
+        frag_c acc_s[NN], acc_dp[NN];
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+            acc_dp[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) {
+                frag_ab bq = *reinterpret_cast<const frag_ab*>(
+                    &q_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bdo = *reinterpret_cast<const frag_ab*>(
+                    &do_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_s[n] = mfma16(ak[kk], bq, acc_s[n]);
+                acc_dp[n] = mfma16(av[kk], bdo, acc_dp[n]);
+            }
+        }
+
+        // P^T (C layout: row = kv, col = q); lse/delta gathered per col
+        const int kvrow0 = kv_base + wave * 16 + (lane >> 4) * 4;
+        float pt_vals[NN][4], dst_vals[NN][4];
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            int qg = q_base + n * 16 + l16;
+            float lse_q = (qg < Sq) ? lse_row[qg] : INFINITY;
+            float dl_q = (qg < Sq) ? dl_row[qg] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 4; r++) {
+                int kvg = kvrow0 + r;
+                bool vis = (kvg < Skv) && (qg < Sq);
+                if (causal) vis = vis && (kvg <= qg + causal_off);
+                float p = vis ? __expf(acc_s[n][r] * scale - lse_q) : 0.f;
+                pt_vals[n][r] = p;
+                dst_vals[n][r] = p * (acc_dp[n][r] - dl_q) * scale;
+            }
+        }
+
+        // stage P^T; dV += P^T dO (contract q; B from dot_lds)
+#pragma unroll
+        for (int n = 0; n < NN; n++)
+#pragma unroll
+            for (int r = 0; r < 4; r++)
+                pt_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                    f32_to_bf16(pt_vals[n][r]);
+        __syncthreads();
+#pragma unroll
+        for (int kk = 0; kk < BLK_M / 32; kk++) {
+            frag_ab apt = *reinterpret_cast<const frag_ab*>(
+                &pt_lds[wave * 16 + l16][kk * 32 + lk8]);
+#pragma unroll
+            for (int n = 0; n < ND; n++) {
+                frag_ab bdot = *reinterpret_cast<const frag_ab*>(
+                    &dot_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_dv[n] = mfma16(apt, bdot, acc_dv[n]);
+            }
+        }
+        __syncthreads();
+
+        // stage dS^T; dK += dS^T Q (contract q; B from qt_lds)
+#pragma unroll
+        for (int n = 0; n < NN; n++)
+#pragma unroll
+            for (int r = 0; r < 4; r++)
+                pt_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                    f32_to_bf16(dst_vals[n][r]);
+        __syncthreads();
+#pragma unroll
+        for (int kk = 0; kk < BLK_M / 32; kk++) {
+            frag_ab adst = *reinterpret_cast<const frag_ab*>(
+                &pt_lds[wave * 16 + l16][kk * 32 + lk8]);
+#pragma unroll
+            for (int n = 0; n < ND; n++) {
+                frag_ab bqt = *reinterpret_cast<const frag_ab*>(
+                    &qt_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_dk[n] = mfma16(adst, bqt, acc_dk[n]);
+            }
+        }
+        __syncthreads();
+    }
+
+    // store per-q-head dK/dV: [B, Skv, Hq, D]
+    const int kvrow0 = kv_base + wave * 16 + (lane >> 4) * 4;
+    const long long out_row_stride = (long long)Hq * D;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+        int kvg = kvrow0 + r;
+        if (kvg >= Skv) continue;
+        long long base = ((long long)b * Skv + kvg) * out_row_stride + (long long)hq * D;
+#pragma unroll
+        for (int n = 0; n < ND; n++) {
+            dk[base + n * 16 + l16] = f32_to_bf16(acc_dk[n][r]);
+            dv[base + n * 16 + l16] = f32_to_bf16(acc_dv[n][r]);
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+void launch_mfma_probe(const void* A, const void* B, float* C, hipStream_t stream) {
+    hipLaunchKernelGGL(mfma_layout_probe, dim3(1), dim3(64), 0, stream,
+                       (const ushort_t*)A, (const ushort_t*)B, C);
+}
+
+template <int D>
+static void flash_fwd_t(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int B, int Sq, int Skv, int Hq, int Hk,
+                        float scale, bool causal, hipStream_t stream) {
+    dim3 grid((Sq + BLK_M - 1) / BLK_M, B * Hq);
+    hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(FA_BLOCK), 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                       (ushort_t*)o, lse, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+}
+
+void launch_flash_fwd(const void* q, const void* k, const void* v, void* o,
+                      float* lse, int B, int Sq, int Skv, int Hq, int Hk, int D,
+                      float scale, bool causal, hipStream_t stream) {
+    if (D == 128) flash_fwd_t<128>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    else if (D == 64) flash_fwd_t<64>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    else if (D == 32) flash_fwd_t<32>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+}
+
+template <int D>
+static void flash_bwd_t(const void* dout, const void* q, const void* k, const void* v,
+                        const void* o, const float* lse, float* delta,
+                        void* dq, void* dk, void* dv,
+                        int B, int Sq, int Skv, int Hq, int Hk,
+                        float scale, bool causal, hipStream_t stream) {
+    long long rows = (long long)B * Sq * Hq;
+    int waves_per_block = FA_BLOCK / 64;
+    int dgrid = (int)((rows + waves_per_block - 1) / waves_per_block);
+    hipLaunchKernelGGL(flash_bwd_delta_kernel<D>, dim3(dgrid), dim3(FA_BLOCK), 0, stream,
+                       (const ushort_t*)dout, (const ushort_t*)o, delta, B, Sq, Hq);
+    dim3 gq((Sq + BLK_M - 1) / BLK_M, B * Hq);
+    hipLaunchKernelGGL(flash_bwd_dq_kernel<D>, gq, dim3(FA_BLOCK), 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dout, lse, delta, (ushort_t*)dq,
+                       B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+    dim3 gkv((Skv + BLK_N - 1) / BLK_N, B * Hq);
+    hipLaunchKernelGGL(flash_bwd_dkv_kernel<D>, gkv, dim3(FA_BLOCK), 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dout, lse, delta, (ushort_t*)dk, (ushort_t*)dv,
+                       B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+}
+
+void launch_flash_bwd(const void* dout, const void* q, const void* k, const void* v,
+                      const void* o, const float* lse, float* delta,
+                      void* dq, void* dk, void* dv,
+                      int B, int Sq, int Skv, int Hq, int Hk, int D,
+                      float scale, bool causal, hipStream_t stream) {
+    if (D == 128) flash_bwd_t<128>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    else if (D == 64) flash_bwd_t<64>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    else if (D == 32) flash_bwd_t<32>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+}

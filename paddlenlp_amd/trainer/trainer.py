@@ -1,0 +1,623 @@
+"""Trainer: HF-style training loop with 4D-parallel wiring for MI355X.
+
+Reference behavior: paddlenlp/trainer/trainer.py — Trainer.__init__ :273,
+train :687, _inner_training_loop :855, training_step :2211, _wrap_model
+:1895, _maybe_log_save_evaluate :1388, _save_checkpoint :2363,
+evaluation_loop :2911.  The distributed runtime that the reference gets from
+paddle fleet is implemented natively here: DP gradient sync via
+parallel.data_parallel (bucketed RCCL all-reduce at the accumulation
+boundary), ZeRO sharding via parallel.zero, bf16 "O2" via bf16 params +
+fp32 master weights in FusedAdamW.
+"""
+from __future__ import annotations
+
+import contextlib
+import math
+import os
+import shutil
+import time
+from typing import Any, Callable, Dict, List, Optional, Tuple, Union
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+from torch.utils.data import DataLoader, Dataset
+
+from ..data import default_data_collator
+from ..data.sampler import DistributedBatchSampler
+from ..parallel.data_parallel import broadcast_parameters, fused_allreduce_gradients
+from ..parallel.topology import get_topology
+from ..transformers.model_utils import PretrainedModel, unwrap_model
+from ..utils.env import (
+    OPTIMIZER_STATE_NAME,
+    PREFIX_CHECKPOINT_DIR,
+    SCHEDULER_NAME,
+    TRAINER_STATE_NAME,
+)
+from ..utils.log import logger
+from .optimizer import FusedAdamW
+from .trainer_callback import (
+    CallbackHandler,
+    DefaultFlowCallback,
+    ProgressCallback,
+    TrainerControl,
+    TrainerState,
+)
+from .trainer_utils import (
+    TrainOutput,
+    caculate_llm_flops,
+    get_last_checkpoint,
+    get_scheduler,
+    set_hybrid_seed,
+    speed_metrics,
+)
+from .training_args import TrainingArguments
+
+DEFAULT_CALLBACKS = [DefaultFlowCallback, ProgressCallback]
+
+
+class Trainer:
+    def __init__(
+        self,
+        model: nn.Module = None,
+        args: TrainingArguments = None,
+        data_collator: Optional[Callable] = None,
+        train_dataset: Optional[Dataset] = None,
+        eval_dataset: Optional[Dataset] = None,
+        tokenizer=None,
+        compute_metrics: Optional[Callable] = None,
+        callbacks: Optional[List] = None,
+        optimizers: Tuple = (None, None),
+        criterion: Optional[nn.Module] = None,
+    ):
+        if args is None:
+            args = TrainingArguments(output_dir="output")
+        self.args = args
+        self.model = model
+        self.criterion = criterion
+        self.data_collator = data_collator or default_data_collator
+        self.train_dataset = train_dataset
+        self.eval_dataset = eval_dataset
+        self.tokenizer = tokenizer
+        self.compute_metrics = compute_metrics
+        self.optimizer, self.lr_scheduler = optimizers
+        self.topology = args.topology
+
+        set_hybrid_seed(args.seed, self.topology)
+
+        self.state = TrainerState()
+        self.state.is_world_process_zero = args.process_index == 0
+        self.control = TrainerControl()
+        callbacks = list(DEFAULT_CALLBACKS) + (callbacks or [])
+        self.callback_handler = CallbackHandler(
+            callbacks, self.model, self.tokenizer, self.optimizer, self.lr_scheduler
+        )
+        self._model_wrapped = None
+        self._zero = None  # ZeRO sharded-optimizer engine when sharding>0
+        self.control = self.callback_handler.call_event(
+            "on_init_end", args, self.state, self.control
+        )
+
+    # ------------------------------------------------------------------
+    # dataloaders
+    # ------------------------------------------------------------------
+    def get_train_dataloader(self) -> DataLoader:
+        sampler = DistributedBatchSampler(
+            self.train_dataset,
+            batch_size=self.args.per_device_train_batch_size,
+            num_replicas=self.args.dataset_world_size,
+            rank=self.args.dataset_rank,
+            shuffle=False,
+            drop_last=self.args.dataloader_drop_last,
+            seed=self.args.seed,
+            consumed_samples=self.state.consumed_samples,
+        )
+        return DataLoader(
+            self.train_dataset,
+            batch_sampler=sampler,
+            collate_fn=self.data_collator,
+            num_workers=self.args.dataloader_num_workers,
+            pin_memory=torch.cuda.is_available(),
+        )
+
+    def get_eval_dataloader(self, eval_dataset=None) -> DataLoader:
+        eval_dataset = eval_dataset or self.eval_dataset
+        sampler = DistributedBatchSampler(
+            eval_dataset,
+            batch_size=self.args.per_device_eval_batch_size,
+            num_replicas=self.args.dataset_world_size,
+            rank=self.args.dataset_rank,
+            shuffle=False,
+            drop_last=False,
+        )
+        return DataLoader(
+            eval_dataset,
+            batch_sampler=sampler,
+            collate_fn=self.data_collator,
+            num_workers=self.args.dataloader_num_workers,
+            pin_memory=torch.cuda.is_available(),
+        )
+
+    # ------------------------------------------------------------------
+    # optimizer / scheduler
+    # ------------------------------------------------------------------
+    @staticmethod
+    def _no_decay(name: str) -> bool:
+        return name.endswith("bias") or "norm" in name.lower()
+
+    def create_optimizer_and_scheduler(self, num_training_steps: int):
+        if self.optimizer is None:
+            model = unwrap_model(self.model)
+            decay_params, no_decay_params = [], []
+            for name, p in model.named_parameters():
+                if not p.requires_grad:
+                    continue
+                p.param_name = name
+                (no_decay_params if self._no_decay(name) else decay_params).append(p)
+            groups = [
+                {"params": decay_params, "weight_decay": self.args.weight_decay},
+                {"params": no_decay_params, "weight_decay": 0.0},
+            ]
+            self.optimizer = FusedAdamW(
+                groups,
+                lr=self.args.learning_rate,
+                betas=(self.args.adam_beta1, self.args.adam_beta2),
+                eps=self.args.adam_epsilon,
+                master_weights=self.args.bf16 or self.args.fp16,
+            )
+        if self.lr_scheduler is None:
+            self.lr_scheduler = get_scheduler(
+                self.args.lr_scheduler_type,
+                self.optimizer,
+                num_warmup_steps=self.args.warmup_steps,
+                num_training_steps=num_training_steps,
+                min_lr_ratio=self.args.min_lr_ratio,
+            )
+        self.callback_handler.optimizer = self.optimizer
+        self.callback_handler.lr_scheduler = self.lr_scheduler
+
+    # ------------------------------------------------------------------
+    # model wrapping (reference _wrap_model :1895)
+    # ------------------------------------------------------------------
+    def _wrap_model(self, model: nn.Module) -> nn.Module:
+        args = self.args
+        device = args.device
+        # bf16 "O2": cast parameters to bf16; FusedAdamW holds fp32 masters
+        if args.bf16:
+            model = model.to(dtype=torch.bfloat16)
+        elif args.fp16:
+            model = model.to(dtype=torch.float16)
+        model = model.to(device)
+
+        topo = self.topology
+        if topo.dp_degree > 1 and topo.data_parallel_group is not None:
+            broadcast_parameters(model, topo.data_parallel_group)
+        if topo.sharding_degree > 1 and topo.sharding_parallel_group is not None:
+            stage = args.sharding_stage()
+            if stage in (1, 2):
+                from ..parallel.zero import ZeroShardedEngine
+
+                # engine is attached after optimizer creation in train()
+                self._zero_stage = stage
+            else:
+                raise NotImplementedError("sharding stage3 lands in a later milestone")
+            broadcast_parameters(model, topo.sharding_parallel_group)
+        return model
+
+    # ------------------------------------------------------------------
+    # train
+    # ------------------------------------------------------------------
+    def train(self, resume_from_checkpoint: Optional[Union[str, bool]] = None):
+        args = self.args
+        self.is_in_train = True
+
+        train_dataloader = self.get_train_dataloader()
+        steps_per_epoch = max(1, len(train_dataloader) // args.gradient_accumulation_steps)
+        if args.max_steps > 0:
+            max_steps = args.max_steps
+            num_train_epochs = math.ceil(max_steps / steps_per_epoch)
+        else:
+            max_steps = int(steps_per_epoch * args.num_train_epochs)
+            num_train_epochs = math.ceil(args.num_train_epochs)
+
+        if resume_from_checkpoint is True or (
+            resume_from_checkpoint is None and args.resume_from_checkpoint
+        ):
+            resume_from_checkpoint = args.resume_from_checkpoint or get_last_checkpoint(args.output_dir)
+
+        model = self._wrap_model(self.model)
+        self._model_wrapped = model
+        self.callback_handler.model = model
+
+        self.create_optimizer_and_scheduler(max_steps)
+
+        # ZeRO engine wraps the optimizer after both exist
+        if getattr(self, "_zero_stage", 0):
+            from ..parallel.zero import ZeroShardedEngine
+
+            self._zero = ZeroShardedEngine(
+                model, self.optimizer,
+                stage=self._zero_stage,
+                group=self.topology.sharding_parallel_group,
+                bucket_mb=args.sharding_comm_buffer_size_MB,
+            )
+
+        if resume_from_checkpoint and isinstance(resume_from_checkpoint, str):
+            self._load_from_checkpoint(resume_from_checkpoint)
+            train_dataloader = self.get_train_dataloader()  # re-skip consumed samples
+
+        model.train()
+
+        total_batch_size = args.global_train_batch_size
+        if args.should_log:
+            logger.info("***** Running training *****")
+            logger.info(f"  Num examples = {len(self.train_dataset):,}")
+            logger.info(f"  Num epochs = {num_train_epochs}")
+            logger.info(f"  Per-device batch size = {args.per_device_train_batch_size}")
+            logger.info(f"  Global batch size (w. accumulation) = {total_batch_size}")
+            logger.info(f"  Gradient accumulation steps = {args.gradient_accumulation_steps}")
+            logger.info(f"  Total optimization steps = {max_steps:,}")
+
+        self.state.max_steps = max_steps
+        self.state.num_train_epochs = num_train_epochs
+        self.callback_handler.train_dataloader = train_dataloader
+        self.control = self.callback_handler.on_train_begin(args, self.state, self.control)
+
+        tr_loss = torch.tensor(0.0, device=args.device)
+        self._total_loss_scalar = 0.0
+        self._globalstep_last_logged = self.state.global_step
+        self._tokens_since_last_log = 0
+        start_time = time.time()
+        self._last_log_time = start_time
+
+        epoch = self.state.epoch
+        accum_count = 0
+        done = False
+        while not done:
+            for step, inputs in enumerate(train_dataloader):
+                if accum_count == 0:
+                    self.control = self.callback_handler.on_step_begin(args, self.state, self.control)
+
+                loss = self.training_step(model, inputs)
+                tr_loss += loss.detach()
+                if "input_ids" in inputs:
+                    self._tokens_since_last_log += inputs["input_ids"].numel()
+                accum_count += 1
+                self.state.consumed_samples += (
+                    args.per_device_train_batch_size * args.dataset_world_size
+                )
+
+                if accum_count < args.gradient_accumulation_steps:
+                    self.control = self.callback_handler.on_substep_end(args, self.state, self.control)
+                    continue
+                accum_count = 0
+
+                self.optimizer_step(model)
+                self.state.global_step += 1
+                self.state.epoch = epoch + (step + 1) / max(1, len(train_dataloader))
+                self.control = self.callback_handler.on_step_end(args, self.state, self.control)
+                self._maybe_log_save_evaluate(tr_loss, model, start_time)
+
+                if self.control.should_training_stop or self.state.global_step >= max_steps:
+                    done = True
+                    break
+            epoch += 1
+            self.control = self.callback_handler.on_epoch_end(args, self.state, self.control)
+            if epoch >= num_train_epochs:
+                done = True
+
+        self.control = self.callback_handler.on_train_end(args, self.state, self.control)
+        metrics = speed_metrics(
+            "train", start_time,
+            num_samples=self.state.consumed_samples,
+            num_steps=self.state.global_step,
+        )
+        train_loss = self._total_loss_scalar / max(1, self.state.global_step)
+        self.is_in_train = False
+        return TrainOutput(self.state.global_step, train_loss, metrics)
+
+    # ------------------------------------------------------------------
+    def _prepare_inputs(self, inputs: Dict[str, Any]) -> Dict[str, Any]:
+        device = self.args.device
+        return {
+            k: v.to(device, non_blocking=True) if isinstance(v, torch.Tensor) else v
+            for k, v in inputs.items()
+        }
+
+    def compute_loss(self, model, inputs, return_outputs=False):
+        """Reference trainer.py:2157."""
+        labels = None
+        if self.criterion is not None and "labels" in inputs:
+            labels = inputs.pop("labels")
+        outputs = model(**inputs)
+        if self.criterion is not None and labels is not None:
+            logits = outputs[0] if isinstance(outputs, tuple) else outputs
+            loss = self.criterion(logits, labels)
+            outputs = (loss, logits)
+        elif isinstance(outputs, tuple):
+            loss = outputs[0]
+        elif isinstance(outputs, dict):
+            loss = outputs["loss"]
+        else:
+            loss = outputs
+        return (loss, outputs) if return_outputs else loss
+
+    def training_step(self, model: nn.Module, inputs: Dict[str, Any]) -> torch.Tensor:
+        """Forward + backward for one micro-batch (reference :2211)."""
+        model.train()
+        inputs = self._prepare_inputs(inputs)
+        loss = self.compute_loss(model, inputs)
+        if self.args.gradient_accumulation_steps > 1:
+            loss = loss / self.args.gradient_accumulation_steps
+        loss.backward()
+        return loss.detach()
+
+    def optimizer_step(self, model: nn.Module):
+        args = self.args
+        topo = self.topology
+        # DP gradient sync (reference fused_allreduce_gradients :1079-1110)
+        if self._zero is not None:
+            self._zero.reduce_gradients_and_step_pre()
+        if topo.dp_degree > 1:
+            fused_allreduce_gradients(model.parameters(), topo.data_parallel_group)
+
+        if args.max_grad_norm and args.max_grad_norm > 0:
+            self._clip_grad_norm(model)
+
+        self.optimizer.step()
+        if self._zero is not None:
+            self._zero.step_post()
+        self.lr_scheduler.step()
+        self.optimizer.zero_grad(set_to_none=False)
+
+    def _clip_grad_norm(self, model):
+        """Global grad-norm clip, TP/PP-aware: local sum-of-squares, reduced
+        over mp (sharded params) before the norm."""
+        args = self.args
+        topo = self.topology
+        params = [p for p in model.parameters() if p.grad is not None]
+        if not params:
+            return
+        device = params[0].grad.device
+        local_sq = torch.zeros((), dtype=torch.float32, device=device)
+        replicated_sq = torch.zeros((), dtype=torch.float32, device=device)
+        for p in params:
+            sq = p.grad.float().pow(2).sum()
+            if getattr(p, "is_column_parallel", False) or getattr(p, "is_row_parallel", False):
+                local_sq += sq
+            else:
+                replicated_sq += sq
+        total = local_sq.clone()
+        if topo.mp_degree > 1 and topo.model_parallel_group is not None:
+            dist.all_reduce(total, group=topo.model_parallel_group)
+        total += replicated_sq
+        if topo.pp_degree > 1 and topo.pipe_parallel_group is not None:
+            dist.all_reduce(total, group=topo.pipe_parallel_group)
+        if self._zero is not None and self._zero.stage >= 2:
+            dist.all_reduce(total, group=topo.sharding_parallel_group)
+        norm = total.sqrt()
+        clip = (args.max_grad_norm / (norm + 1e-6)).clamp(max=1.0)
+        for p in params:
+            p.grad.mul_(clip.to(p.grad.dtype))
+
+    # ------------------------------------------------------------------
+    # logging / save / evaluate
+    # ------------------------------------------------------------------
+    def _model_flops_per_step(self) -> Optional[float]:
+        model = unwrap_model(self.model)
+        cfg = getattr(model, "config", None)
+        if cfg is None or not hasattr(cfg, "num_hidden_layers"):
+            return None
+        seq_len = getattr(cfg, "max_position_embeddings", 4096)
+        seq_len = getattr(self, "_observed_seq_len", seq_len)
+        tokens = self._tokens_since_last_log
+        if tokens == 0:
+            return None
+        batch_equiv = tokens / seq_len
+        return caculate_llm_flops(
+            cfg.hidden_size, cfg.intermediate_size, cfg.num_hidden_layers,
+            cfg.vocab_size, seq_len, batch_size=batch_equiv,
+            recompute=getattr(cfg, "recompute", False),
+        )
+
+    def _maybe_log_save_evaluate(self, tr_loss, model, start_time):
+        args, state, control = self.args, self.state, self.control
+        if control.should_log:
+            steps = state.global_step - self._globalstep_last_logged
+            loss_val = tr_loss.item() / max(1, steps) / args.gradient_accumulation_steps * args.gradient_accumulation_steps
+            loss_scalar = (tr_loss / max(1, steps)).item()
+            tr_loss.zero_()
+            self._total_loss_scalar += loss_scalar * steps
+            interval = time.time() - self._last_log_time
+            logs = {
+                "loss": round(loss_scalar, 6),
+                "learning_rate": self.lr_scheduler.get_last_lr()[0],
+                "global_step": state.global_step,
+            }
+            if interval > 0 and self._tokens_since_last_log:
+                logs["interval_tokens_per_second_per_device"] = round(
+                    self._tokens_since_last_log / interval, 1
+                )
+                flops = self._model_flops_per_step()
+                if flops:
+                    logs["interval_hardware_tflops_per_device"] = round(flops / interval / 1e12, 1)
+            if torch.cuda.is_available():
+                logs["gpu_mem_max_allocated_gb"] = round(
+                    torch.cuda.max_memory_allocated() / 1e9, 2
+                )
+            self._tokens_since_last_log = 0
+            self._last_log_time = time.time()
+            self._globalstep_last_logged = state.global_step
+            self.log(logs)
+        if control.should_evaluate and self.eval_dataset is not None:
+            metrics = self.evaluate()
+            self.control = self.callback_handler.on_evaluate(args, state, control, metrics=metrics)
+        if control.should_save:
+            self._save_checkpoint(model)
+            self.control = self.callback_handler.on_save(args, state, self.control)
+
+    def log(self, logs: Dict[str, float]):
+        logs["epoch"] = round(self.state.epoch, 4)
+        self.state.log_history.append(dict(logs))
+        self.control = self.callback_handler.on_log(self.args, self.state, self.control, logs=logs)
+
+    # ------------------------------------------------------------------
+    # checkpointing (reference _save_checkpoint :2363)
+    # ------------------------------------------------------------------
+    def _checkpoint_dir(self) -> str:
+        return os.path.join(self.args.output_dir, f"{PREFIX_CHECKPOINT_DIR}-{self.state.global_step}")
+
+    def _save_checkpoint(self, model):
+        args = self.args
+        ckpt_dir = self._checkpoint_dir()
+        os.makedirs(ckpt_dir, exist_ok=True)
+        self.save_model(ckpt_dir)
+
+        if args.unified_checkpoint:
+            from .unified_checkpoint import save_unified_optimizer
+
+            save_unified_optimizer(self.optimizer, unwrap_model(model), ckpt_dir, self.topology, zero=self._zero)
+        elif args.process_index == 0 or (self._zero is not None):
+            torch.save(self.optimizer.state_dict(), os.path.join(ckpt_dir, OPTIMIZER_STATE_NAME))
+
+        if args.process_index == 0:
+            torch.save(self.lr_scheduler.state_dict(), os.path.join(ckpt_dir, SCHEDULER_NAME))
+            self.state.save_to_json(os.path.join(ckpt_dir, TRAINER_STATE_NAME))
+            self._save_rng_state(ckpt_dir)
+            self._rotate_checkpoints()
+        if dist.is_initialized():
+            dist.barrier()
+
+    def _save_rng_state(self, ckpt_dir):
+        rng = {
+            "python": __import__("random").getstate(),
+            "numpy": __import__("numpy").random.get_state(),
+            "torch": torch.get_rng_state(),
+        }
+        if torch.cuda.is_available():
+            rng["cuda"] = torch.cuda.get_rng_state_all()
+        torch.save(rng, os.path.join(ckpt_dir, f"rng_state_{self.topology.world_size}.pth"))
+
+    def _rotate_checkpoints(self):
+        limit = self.args.save_total_limit
+        if not limit:
+            return
+        import re
+
+        pat = re.compile(rf"^{PREFIX_CHECKPOINT_DIR}-(\d+)$")
+        ckpts = sorted(
+            (
+                d for d in os.listdir(self.args.output_dir)
+                if pat.match(d) and os.path.isdir(os.path.join(self.args.output_dir, d))
+            ),
+            key=lambda d: int(pat.match(d).group(1)),
+        )
+        for d in ckpts[:-limit]:
+            shutil.rmtree(os.path.join(self.args.output_dir, d), ignore_errors=True)
+
+    def save_model(self, output_dir: Optional[str] = None):
+        output_dir = output_dir or self.args.output_dir
+        model = unwrap_model(self.model)
+        if self.args.unified_checkpoint:
+            from .unified_checkpoint import save_unified_model
+
+            save_unified_model(model, output_dir, self.topology)
+        elif self.args.process_index == 0:
+            if isinstance(model, PretrainedModel):
+                model.save_pretrained(output_dir)
+            else:
+                torch.save(model.state_dict(), os.path.join(output_dir, "pytorch_model.bin"))
+        if self.tokenizer is not None and self.args.process_index == 0:
+            self.tokenizer.save_pretrained(output_dir)
+
+    def _load_from_checkpoint(self, ckpt_dir: str):
+        logger.info(f"Resuming from checkpoint {ckpt_dir}")
+        model = unwrap_model(self.model)
+        if self.args.unified_checkpoint:
+            from .unified_checkpoint import load_unified_checkpoint
+
+            load_unified_checkpoint(model, self.optimizer, ckpt_dir, self.topology, zero=self._zero)
+        else:
+            from safetensors.torch import load_file
+
+            state = load_file(os.path.join(ckpt_dir, "model.safetensors"))
+            model.load_state_dict(state, strict=False)
+            opt_path = os.path.join(ckpt_dir, OPTIMIZER_STATE_NAME)
+            if os.path.isfile(opt_path):
+                self.optimizer.load_state_dict(torch.load(opt_path, weights_only=False))
+        sched_path = os.path.join(ckpt_dir, SCHEDULER_NAME)
+        if os.path.isfile(sched_path):
+            self.lr_scheduler.load_state_dict(torch.load(sched_path, weights_only=False))
+            # re-apply the restored LR now: the optimizer keeps its
+            # construction-time LR until the next scheduler.step() otherwise
+            for group, lr in zip(self.optimizer.param_groups, self.lr_scheduler.get_last_lr()):
+                group["lr"] = lr
+        state_path = os.path.join(ckpt_dir, TRAINER_STATE_NAME)
+        if os.path.isfile(state_path):
+            self.state = TrainerState.load_from_json(state_path)
+            self.state.is_world_process_zero = self.args.process_index == 0
+        rng_path = os.path.join(ckpt_dir, f"rng_state_{self.topology.world_size}.pth")
+        if os.path.isfile(rng_path):
+            rng = torch.load(rng_path, weights_only=False)
+            __import__("random").setstate(rng["python"])
+            __import__("numpy").random.set_state(rng["numpy"])
+            torch.set_rng_state(rng["torch"])
+            if torch.cuda.is_available() and "cuda" in rng:
+                try:
+                    torch.cuda.set_rng_state_all(rng["cuda"])
+                except RuntimeError:
+                    logger.warning("CUDA rng restore skipped (device count changed)")
+
+    # ------------------------------------------------------------------
+    # evaluation
+    # ------------------------------------------------------------------
+    def evaluate(self, eval_dataset=None, metric_key_prefix: str = "eval") -> Dict[str, float]:
+        model = self._model_wrapped or self.model
+        dataloader = self.get_eval_dataloader(eval_dataset)
+        model.eval()
+        losses = []
+        start = time.time()
+        n_samples = 0
+        with torch.no_grad():
+            for i, inputs in enumerate(dataloader):
+                if 0 < self.args.max_evaluate_steps <= i:
+                    break
+                inputs = self._prepare_inputs(inputs)
+                loss = self.compute_loss(model, inputs)
+                losses.append(loss.float())
+                n_samples += next(iter(inputs.values())).shape[0]
+        model.train()
+        mean_loss = torch.stack(losses).mean() if losses else torch.tensor(float("nan"))
+        if dist.is_initialized() and self.topology.data_parallel_group is not None:
+            dist.all_reduce(mean_loss, group=self.topology.data_parallel_group)
+            mean_loss /= dist.get_world_size(self.topology.data_parallel_group)
+        metrics = {f"{metric_key_prefix}_loss": mean_loss.item()}
+        try:
+            metrics[f"{metric_key_prefix}_ppl"] = math.exp(mean_loss.item())
+        except OverflowError:
+            metrics[f"{metric_key_prefix}_ppl"] = float("inf")
+        metrics.update(speed_metrics(metric_key_prefix, start, num_samples=n_samples))
+        self.log(metrics)
+        return metrics
+
+    def predict(self, test_dataset, metric_key_prefix: str = "test"):
+        model = self._model_wrapped or self.model
+        dataloader = self.get_eval_dataloader(test_dataset)
+        model.eval()
+        all_logits, all_labels = [], []
+        with torch.no_grad():
+            for inputs in dataloader:
+                inputs = self._prepare_inputs(inputs)
+                labels = inputs.get("labels")
+                loss, outputs = self.compute_loss(model, inputs, return_outputs=True)
+                logits = outputs[1] if isinstance(outputs, tuple) and len(outputs) > 1 else outputs
+                all_logits.append(logits.float().cpu())
+                if labels is not None:
+                    all_labels.append(labels.cpu())
+        model.train()
+        logits = torch.cat(all_logits) if all_logits else None
+        labels = torch.cat(all_labels) if all_labels else None
+        metrics = {}
+        if self.compute_metrics is not None and logits is not None:
+            metrics = self.compute_metrics((logits, labels))
+        return logits, labels, metrics

@@ -1,0 +1,182 @@
+"""Trainer utilities (reference: paddlenlp/trainer/trainer_utils.py).
+
+set_seed, speed_metrics (:351 — tokens/s/device + hardware TFLOPS for MFU),
+LR scheduler factory get_scheduler, ShardingOption (:315), checkpoint
+discovery get_last_checkpoint, and the LLM FLOPs formula
+(reference: paddlenlp/transformers/utils.py:963 caculate_llm_flops,
+Megatron eq. 3).
+"""
+from __future__ import annotations
+
+import math
+import os
+import random
+import re
+import time
+from enum import Enum
+from typing import Dict, Optional
+
+import numpy as np
+import torch
+
+from ..utils.env import PREFIX_CHECKPOINT_DIR
+
+# MI355X dense bf16 peak (MFU denominator): 2.5 PFLOP/s
+# (AMD's 5 PF marketing figure includes 2:1 sparsity.)
+MI355X_BF16_PEAK_FLOPS = 2.5e15
+MI355X_FP8_PEAK_FLOPS = 5.0e15
+
+
+class ShardingOption(Enum):
+    """Reference: trainer_utils.py ShardingOption :315."""
+
+    SHARD_OP = "stage1"
+    SHARD_GRAD_OP = "stage2"
+    FULL_SHARD = "stage3"
+    OFFLOAD = "offload"
+
+
+class IntervalStrategy(Enum):
+    NO = "no"
+    STEPS = "steps"
+    EPOCH = "epoch"
+
+
+def set_seed(seed: int):
+    random.seed(seed)
+    np.random.seed(seed)
+    torch.manual_seed(seed)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(seed)
+
+
+def set_hybrid_seed(seed: int, topology=None):
+    """Per-rank seed offsets: dp/sharding ranks must differ for dropout,
+    mp/pp ranks must match for parameter init on replicated params."""
+    from ..parallel.topology import get_topology
+
+    topo = topology or get_topology()
+    # global seed for replicated-state ops; offset by data-parallel coordinate
+    set_seed(seed + topo.dataset_rank * 1024)
+
+
+def caculate_llm_flops(
+    hidden_size: int,
+    intermediate_size: int,
+    layer_num: int,
+    vocab_size: int,
+    seq_length: int,
+    batch_size: int = 1,
+    recompute: bool = False,
+) -> int:
+    """Megatron eq.3 FLOPs per optimization step for causal LM training.
+
+    Reference: paddlenlp/transformers/utils.py:963-1003 (name kept with the
+    reference's spelling for parity).
+    """
+    flops_per_token = (
+        # attention qkvo + mlp
+        2 * (4 * hidden_size * hidden_size + 3 * hidden_size * intermediate_size)
+        # attention scores + context
+        + 4 * hidden_size * seq_length
+    )
+    flops_fwd = batch_size * seq_length * layer_num * flops_per_token
+    # lm head
+    flops_fwd += 2 * batch_size * seq_length * hidden_size * vocab_size
+    # bwd = 2x fwd; full recompute adds another fwd
+    mult = 4 if recompute else 3
+    return flops_fwd * mult
+
+
+def speed_metrics(
+    split: str,
+    start_time: float,
+    num_samples: Optional[int] = None,
+    num_steps: Optional[int] = None,
+    num_tokens: Optional[int] = None,
+    model_flops: Optional[float] = None,
+) -> Dict[str, float]:
+    """Reference: trainer_utils.py:351-379 — reports
+    *_tokens_per_second_per_device and *_hardware_tflops_per_device."""
+    runtime = time.time() - start_time
+    result = {f"{split}_runtime": round(runtime, 4)}
+    if runtime == 0:
+        return result
+    if num_samples is not None:
+        result[f"{split}_samples_per_second"] = round(num_samples / runtime, 3)
+    if num_steps is not None:
+        result[f"{split}_steps_per_second"] = round(num_steps / runtime, 3)
+    if num_tokens is not None:
+        result[f"{split}_tokens_per_second_per_device"] = round(num_tokens / runtime, 2)
+    if model_flops is not None:
+        tflops = model_flops / runtime / 1e12
+        result[f"{split}_hardware_tflops_per_device"] = round(tflops, 2)
+        result[f"{split}_mfu_percent"] = round(
+            100.0 * model_flops / runtime / MI355X_BF16_PEAK_FLOPS, 2
+        )
+    return result
+
+
+# ---------------------------------------------------------------------------
+# LR schedulers (reference: paddlenlp/transformers/optimization.py)
+# ---------------------------------------------------------------------------
+def get_scheduler(
+    name: str,
+    optimizer: torch.optim.Optimizer,
+    num_warmup_steps: int,
+    num_training_steps: int,
+    min_lr_ratio: float = 0.0,
+    last_epoch: int = -1,
+):
+    name = name.lower()
+
+    def linear(step):
+        if step < num_warmup_steps:
+            return step / max(1, num_warmup_steps)
+        return max(
+            min_lr_ratio,
+            (num_training_steps - step) / max(1, num_training_steps - num_warmup_steps),
+        )
+
+    def cosine(step):
+        if step < num_warmup_steps:
+            return step / max(1, num_warmup_steps)
+        progress = (step - num_warmup_steps) / max(1, num_training_steps - num_warmup_steps)
+        progress = min(progress, 1.0)
+        return min_lr_ratio + (1 - min_lr_ratio) * 0.5 * (1 + math.cos(math.pi * progress))
+
+    def constant(step):
+        if step < num_warmup_steps:
+            return step / max(1, num_warmup_steps)
+        return 1.0
+
+    fns = {
+        "linear": linear,
+        "cosine": cosine,
+        "constant": constant,
+        "constant_with_warmup": constant,
+    }
+    if name not in fns:
+        raise ValueError(f"Unknown scheduler {name}; choices: {sorted(fns)}")
+    return torch.optim.lr_scheduler.LambdaLR(optimizer, fns[name], last_epoch=last_epoch)
+
+
+def get_last_checkpoint(folder: str) -> Optional[str]:
+    """Reference: trainer_utils.py get_last_checkpoint."""
+    if not os.path.isdir(folder):
+        return None
+    pat = re.compile(rf"^{PREFIX_CHECKPOINT_DIR}-(\d+)$")
+    ckpts = [
+        d for d in os.listdir(folder)
+        if pat.match(d) and os.path.isdir(os.path.join(folder, d))
+    ]
+    if not ckpts:
+        return None
+    return os.path.join(folder, max(ckpts, key=lambda d: int(pat.match(d).group(1))))
+
+
+class TrainOutput:
+    def __init__(self, global_step: int, training_loss: float, metrics: Dict[str, float]):
+        self.global_step = global_step
+        self.training_loss = training_loss
+        self.metrics = metrics

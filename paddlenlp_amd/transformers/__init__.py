@@ -1,0 +1,10 @@
+from .configuration_utils import LlmMetaConfig, PretrainedConfig  # noqa: F401
+from .model_utils import PretrainedModel, unwrap_model  # noqa: F401
+from .tokenizer_utils import PretrainedTokenizer  # noqa: F401
+from .auto import AutoConfig, AutoModel, AutoModelForCausalLM, AutoTokenizer  # noqa: F401
+from .llama import (  # noqa: F401
+    LlamaConfig,
+    LlamaForCausalLM,
+    LlamaModel,
+    LlamaPretrainingCriterion,
+)

@@ -1,0 +1,25 @@
+"""AutoModel / AutoModelForCausalLM (reference: auto/modeling.py:405)."""
+from .registry import get_class, resolve_model_type
+
+
+class _AutoBase:
+    _kind = "base"
+
+    @classmethod
+    def from_pretrained(cls, path: str, **kwargs):
+        model_type = resolve_model_type(path)
+        model_cls = get_class(model_type, cls._kind)
+        return model_cls.from_pretrained(path, **kwargs)
+
+    @classmethod
+    def from_config(cls, config, **kwargs):
+        model_cls = get_class(config.model_type, cls._kind)
+        return model_cls.from_config(config, **kwargs)
+
+
+class AutoModel(_AutoBase):
+    _kind = "base"
+
+
+class AutoModelForCausalLM(_AutoBase):
+    _kind = "causal_lm"
