@@ -121,7 +121,7 @@ std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits, torch::Tensor
     auto labels64 = labels.to(torch::kInt64).contiguous();
     auto loss = torch::empty({N}, logits.options().dtype(torch::kFloat32));
     auto maxlse = torch::empty({N, 2}, logits.options().dtype(torch::kFloat32));
-    launch_ce_fwd(logits.data_ptr(), labels64.data_ptr<long long>(),
+    launch_ce_fwd(logits.data_ptr(), reinterpret_cast<const long long*>(labels64.data_ptr<int64_t>()),
                   loss.data_ptr<float>(), maxlse.data_ptr<float>(), N, V,
                   ignore_index, cur_stream());
     return {loss, maxlse};
@@ -137,7 +137,7 @@ torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logits,
     auto dl = dloss.to(torch::kFloat32).contiguous();
     auto dlogits = torch::empty_like(logits);
     launch_ce_bwd(dl.data_ptr<float>(), logits.data_ptr(),
-                  labels64.data_ptr<long long>(), maxlse.data_ptr<float>(),
+                  reinterpret_cast<const long long*>(labels64.data_ptr<int64_t>()), maxlse.data_ptr<float>(),
                   dlogits.data_ptr(), N, V, ignore_index, cur_stream());
     return dlogits;
 }
