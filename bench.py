@@ -138,7 +138,7 @@ def main():
         })
 
     def one_step(step_idx: int):
-        optimizer.zero_grad(set_to_none=False)
+        optimizer.zero_grad(set_to_none=True)
         for a in range(args.accum):
             batch = batches[(step_idx * args.accum + a) % len(batches)]
             loss, _ = model(**batch)

@@ -6,7 +6,7 @@
 
 // ---- launcher decls (implemented in the .hip files) ----
 void launch_rms_norm_fwd(const void*, const void*, void*, float*, long long, int, float, hipStream_t);
-void launch_rms_norm_bwd(const void*, const void*, const void*, const float*, void*, float*, void*, bool, long long, int, int, hipStream_t);
+void launch_rms_norm_bwd(const void*, const void*, const void*, const float*, void*, float*, float*, void*, bool, long long, int, int, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, long long, int, int, int, bool, hipStream_t);
 void launch_swiglu_fwd(const void*, void*, long long, int, hipStream_t);
 void launch_swiglu_bwd(const void*, const void*, void*, long long, int, hipStream_t);
@@ -57,11 +57,13 @@ std::vector<torch::Tensor> rms_norm_bwd(torch::Tensor dy, torch::Tensor x, torch
     int P = (int)std::min<long long>(rows, 2048);
     auto dx = torch::empty_like(x);
     auto dw_partial = torch::zeros({P, H}, x.options().dtype(torch::kFloat32));
-    auto dw = torch::empty_like(w);
+    bool w_bf16 = w.scalar_type() == torch::kBFloat16;
+    auto dw32 = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+    auto dw = w_bf16 ? torch::empty_like(w) : dw32;
     launch_rms_norm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                         invrms.data_ptr<float>(), dx.data_ptr(),
-                        dw_partial.data_ptr<float>(), dw.data_ptr(),
-                        w.scalar_type() == torch::kBFloat16, rows, H, P, cur_stream());
+                        dw_partial.data_ptr<float>(), dw32.data_ptr<float>(),
+                        dw.data_ptr(), w_bf16, rows, H, P, cur_stream());
     return {dx, dw};
 }
 

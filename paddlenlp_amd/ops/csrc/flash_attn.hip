@@ -57,8 +57,12 @@ __global__ void mfma_layout_probe(const ushort_t* A, const ushort_t* B, float* C
 }
 
 // ---------------------------------------------------------------------------
-// forward
+// forward.  Each wave owns MF x 16 q rows (MF M-frags), so one K/V staging
+// round feeds 2x the MFMA work (BLK_M = FA_WAVES * MF * 16 = 128).
 // ---------------------------------------------------------------------------
+#define FA_MF 2
+#define FWD_BLK_M (FA_WAVES * FA_MF * 16)
+
 template <int D>
 __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const ushort_t* __restrict__ q,   // [B, Sq, Hq, D]
@@ -73,9 +77,9 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     constexpr int LDK = D + LDS_PAD;
     constexpr int LDT = BLK_N + LDS_PAD;
 
-    __shared__ ushort_t k_lds[BLK_N][LDK];   // K row-major [kv][d]
-    __shared__ ushort_t vt_lds[D][LDT];      // V transposed [d][kv]
-    __shared__ ushort_t p_lds[BLK_M][LDT];   // P row-major [q][kv]
+    __shared__ ushort_t k_lds[BLK_N][LDK];       // K row-major [kv][d]
+    __shared__ ushort_t vt_lds[D][LDT];          // V transposed [d][kv]
+    __shared__ ushort_t p_lds[FWD_BLK_M][LDT];   // P row-major [q][kv]
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -87,7 +91,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const int bh = blockIdx.y;
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
-    const int q_base = qt * BLK_M;
+    const int q_base = qt * FWD_BLK_M;
     const int causal_off = Skv - Sq;  // kv visible iff kv <= q + off
 
     const long long q_row_stride = (long long)Hq * D;
@@ -96,152 +100,172 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const ushort_t* k_ptr = k + ((long long)b * Skv * Hk + hk) * D;
     const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
 
-    // Q A-frags in registers (16 rows x D per wave)
-    frag_ab aq[KD];
-    {
-        int qrow = q_base + wave * 16 + l16;
+    // Q A-frags in registers (MF x 16 rows x D per wave)
+    frag_ab aq[FA_MF][KD];
+#pragma unroll
+    for (int mf = 0; mf < FA_MF; mf++) {
+        int qrow = q_base + (wave * FA_MF + mf) * 16 + l16;
         if (qrow < Sq) {
 #pragma unroll
             for (int kk = 0; kk < KD; kk++)
-                aq[kk] = *reinterpret_cast<const frag_ab*>(
+                aq[mf][kk] = *reinterpret_cast<const frag_ab*>(
                     q_ptr + (long long)qrow * q_row_stride + kk * 32 + lk8);
         } else {
 #pragma unroll
-            for (int kk = 0; kk < KD; kk++) aq[kk] = frag_ab{0};
+            for (int kk = 0; kk < KD; kk++) aq[mf][kk] = frag_ab{0};
         }
     }
 
-    float m_run[4], l_run[4];
+    float m_run[FA_MF][4], l_run[FA_MF][4];
 #pragma unroll
-    for (int r = 0; r < 4; r++) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
-    frag_c acc_o[ND];
+    for (int mf = 0; mf < FA_MF; mf++)
 #pragma unroll
-    for (int n = 0; n < ND; n++) acc_o[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+        for (int r = 0; r < 4; r++) { m_run[mf][r] = -INFINITY; l_run[mf][r] = 0.f; }
+    frag_c acc_o[FA_MF][ND];
+#pragma unroll
+    for (int mf = 0; mf < FA_MF; mf++)
+#pragma unroll
+        for (int n = 0; n < ND; n++) acc_o[mf][n] = frag_c{0.f, 0.f, 0.f, 0.f};
 
     int n_kv_tiles = (Skv + BLK_N - 1) / BLK_N;
     if (causal) {
-        int max_kv = q_base + BLK_M - 1 + causal_off;
+        int max_kv = q_base + FWD_BLK_M - 1 + causal_off;
         int lim = (max_kv + BLK_N) / BLK_N;
         n_kv_tiles = min(n_kv_tiles, max(lim, 0));
     }
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
         const int kv_base = kvt * BLK_N;
-        for (int idx = tid * 8; idx < BLK_N * D; idx += FA_BLOCK * 8) {
-            int row = idx / D, col = idx % D;
-            int kvg = kv_base + row;
-            short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-            short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (kvg < Skv) {
-                k8 = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg * kv_row_stride + col);
-                v8 = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg * kv_row_stride + col);
+        // stage K row-major (vector) + V transposed (b32-packed: two kv rows
+        // per thread so the LDS transpose writes are 4-byte, halving ds ops)
+        for (int idx = tid * 8; idx < (BLK_N / 2) * D; idx += FA_BLOCK * 8) {
+            int rr = idx / D, col = idx % D;
+            int row0 = rr * 2;
+            int kvg0 = kv_base + row0;
+            short8v k8a = {0,0,0,0,0,0,0,0}, k8b = {0,0,0,0,0,0,0,0};
+            short8v v8a = {0,0,0,0,0,0,0,0}, v8b = {0,0,0,0,0,0,0,0};
+            if (kvg0 < Skv) {
+                k8a = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg0 * kv_row_stride + col);
+                v8a = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg0 * kv_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(&k_lds[row][col]) = k8;
+            if (kvg0 + 1 < Skv) {
+                k8b = *reinterpret_cast<const short8v*>(k_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+                v8b = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+            }
+            *reinterpret_cast<short8v*>(&k_lds[row0][col]) = k8a;
+            *reinterpret_cast<short8v*>(&k_lds[row0 + 1][col]) = k8b;
 #pragma unroll
-            for (int j = 0; j < 8; j++) vt_lds[col + j][row] = (ushort_t)v8[j];
+            for (int j = 0; j < 8; j++) {
+                unsigned int packed = ((unsigned int)(unsigned short)v8a[j]) |
+                                      (((unsigned int)(unsigned short)v8b[j]) << 16);
+                *reinterpret_cast<unsigned int*>(&vt_lds[col + j][row0]) = packed;
+            }
         }
         __syncthreads();
 
-        // S = Q K^T  (contract d; B-frag from K row-major)
-        frag_c acc_s[NN];
+        // S = Q K^T (contract d; B-frag from K row-major), then softmax with
+        // P written in place into the S accumulator, then P -> LDS
 #pragma unroll
-        for (int n = 0; n < NN; n++) {
-            acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-            for (int kk = 0; kk < KD; kk++) {
-                frag_ab bk = *reinterpret_cast<const frag_ab*>(
-                    &k_lds[n * 16 + l16][kk * 32 + lk8]);
-                acc_s[n] = mfma16(aq[kk], bk, acc_s[n]);
-            }
-        }
-
-        // scale + mask (C layout: row q = qrow0 + r, col kv)
-        const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
-#pragma unroll
-        for (int n = 0; n < NN; n++) {
-            int kvg = kv_base + n * 16 + l16;
-#pragma unroll
-            for (int r = 0; r < 4; r++) {
-                int qg = qrow0 + r;
-                bool vis = (kvg < Skv) && (qg < Sq);
-                if (causal) vis = vis && (kvg <= qg + causal_off);
-                acc_s[n][r] = vis ? acc_s[n][r] * scale : -INFINITY;
-            }
-        }
-
-        // online softmax update
-        float p_tile[NN][4];
-#pragma unroll
-        for (int r = 0; r < 4; r++) {
-            float row_max = -INFINITY;
-#pragma unroll
-            for (int n = 0; n < NN; n++) row_max = fmaxf(row_max, acc_s[n][r]);
-#pragma unroll
-            for (int off = 8; off > 0; off >>= 1)
-                row_max = fmaxf(row_max, __shfl_xor(row_max, off, 64));
-            float m_new = fmaxf(m_run[r], row_max);
-            float alpha;
-            if (m_new == -INFINITY) {            // nothing visible yet
-                alpha = 1.f;
-            } else if (m_run[r] == -INFINITY) {  // first visible tile
-                alpha = 0.f;
-            } else {
-                alpha = __expf(m_run[r] - m_new);
-            }
-            float row_sum = 0.f;
+        for (int mf = 0; mf < FA_MF; mf++) {
+            frag_c acc_s[NN];
 #pragma unroll
             for (int n = 0; n < NN; n++) {
-                float p = (m_new == -INFINITY || acc_s[n][r] == -INFINITY)
-                              ? 0.f : __expf(acc_s[n][r] - m_new);
-                p_tile[n][r] = p;
-                row_sum += p;
+                acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int kk = 0; kk < KD; kk++) {
+                    frag_ab bk = *reinterpret_cast<const frag_ab*>(
+                        &k_lds[n * 16 + l16][kk * 32 + lk8]);
+                    acc_s[n] = mfma16(aq[mf][kk], bk, acc_s[n]);
+                }
+            }
+            const int qrow0 = q_base + (wave * FA_MF + mf) * 16 + (lane >> 4) * 4;
+#pragma unroll
+            for (int n = 0; n < NN; n++) {
+                int kvg = kv_base + n * 16 + l16;
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int qg = qrow0 + r;
+                    bool vis = (kvg < Skv) && (qg < Sq);
+                    if (causal) vis = vis && (kvg <= qg + causal_off);
+                    acc_s[n][r] = vis ? acc_s[n][r] * scale : -INFINITY;
+                }
             }
 #pragma unroll
-            for (int off = 8; off > 0; off >>= 1) row_sum += __shfl_xor(row_sum, off, 64);
-            l_run[r] = l_run[r] * alpha + row_sum;
-            m_run[r] = m_new;
+            for (int r = 0; r < 4; r++) {
+                float row_max = -INFINITY;
 #pragma unroll
-            for (int n = 0; n < ND; n++) acc_o[n][r] *= alpha;
+                for (int n = 0; n < NN; n++) row_max = fmaxf(row_max, acc_s[n][r]);
+#pragma unroll
+                for (int off = 8; off > 0; off >>= 1)
+                    row_max = fmaxf(row_max, __shfl_xor(row_max, off, 64));
+                float m_new = fmaxf(m_run[mf][r], row_max);
+                float alpha;
+                if (m_new == -INFINITY) {
+                    alpha = 1.f;
+                } else if (m_run[mf][r] == -INFINITY) {
+                    alpha = 0.f;
+                } else {
+                    alpha = __expf(m_run[mf][r] - m_new);
+                }
+                float row_sum = 0.f;
+#pragma unroll
+                for (int n = 0; n < NN; n++) {
+                    float p = (m_new == -INFINITY || acc_s[n][r] == -INFINITY)
+                                  ? 0.f : __expf(acc_s[n][r] - m_new);
+                    acc_s[n][r] = p;  // reuse the S accumulator for P
+                    row_sum += p;
+                }
+#pragma unroll
+                for (int off = 8; off > 0; off >>= 1) row_sum += __shfl_xor(row_sum, off, 64);
+                l_run[mf][r] = l_run[mf][r] * alpha + row_sum;
+                m_run[mf][r] = m_new;
+#pragma unroll
+                for (int n = 0; n < ND; n++) acc_o[mf][n][r] *= alpha;
+            }
+#pragma unroll
+            for (int n = 0; n < NN; n++)
+#pragma unroll
+                for (int r = 0; r < 4; r++)
+                    p_lds[(wave * FA_MF + mf) * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                        f32_to_bf16(acc_s[n][r]);
         }
-
-        // P -> LDS (C layout out, A layout back in)
-#pragma unroll
-        for (int n = 0; n < NN; n++)
-#pragma unroll
-            for (int r = 0; r < 4; r++)
-                p_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
-                    f32_to_bf16(p_tile[n][r]);
         __syncthreads();
 
         // O += P V  (contract kv; A from p_lds, B from vt_lds)
 #pragma unroll
         for (int kk = 0; kk < BLK_N / 32; kk++) {
-            frag_ab ap = *reinterpret_cast<const frag_ab*>(
-                &p_lds[wave * 16 + l16][kk * 32 + lk8]);
 #pragma unroll
-            for (int n = 0; n < ND; n++) {
-                frag_ab bv = *reinterpret_cast<const frag_ab*>(
-                    &vt_lds[n * 16 + l16][kk * 32 + lk8]);
-                acc_o[n] = mfma16(ap, bv, acc_o[n]);
+            for (int mf = 0; mf < FA_MF; mf++) {
+                frag_ab ap = *reinterpret_cast<const frag_ab*>(
+                    &p_lds[(wave * FA_MF + mf) * 16 + l16][kk * 32 + lk8]);
+#pragma unroll
+                for (int n = 0; n < ND; n++) {
+                    frag_ab bv = *reinterpret_cast<const frag_ab*>(
+                        &vt_lds[n * 16 + l16][kk * 32 + lk8]);
+                    acc_o[mf][n] = mfma16(ap, bv, acc_o[mf][n]);
+                }
             }
         }
         __syncthreads();
     }
 
     // epilogue: O/l and LSE
-    const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int r = 0; r < 4; r++) {
-        int qg = qrow0 + r;
-        if (qg >= Sq) continue;
-        float inv_l = (l_run[r] > 0.f) ? 1.0f / l_run[r] : 0.f;
-        ushort_t* orow = o + ((long long)b * Sq + qg) * q_row_stride + (long long)hq * D;
+    for (int mf = 0; mf < FA_MF; mf++) {
+        const int qrow0 = q_base + (wave * FA_MF + mf) * 16 + (lane >> 4) * 4;
 #pragma unroll
-        for (int n = 0; n < ND; n++)
-            orow[n * 16 + l16] = f32_to_bf16(acc_o[n][r] * inv_l);
-        if (l16 == 0) {
-            float lv = (l_run[r] > 0.f) ? (m_run[r] + __logf(l_run[r])) : -INFINITY;
-            lse[((long long)b * Hq + hq) * Sq + qg] = lv;
+        for (int r = 0; r < 4; r++) {
+            int qg = qrow0 + r;
+            if (qg >= Sq) continue;
+            float inv_l = (l_run[mf][r] > 0.f) ? 1.0f / l_run[mf][r] : 0.f;
+            ushort_t* orow = o + ((long long)b * Sq + qg) * q_row_stride + (long long)hq * D;
+#pragma unroll
+            for (int n = 0; n < ND; n++)
+                orow[n * 16 + l16] = f32_to_bf16(acc_o[mf][n][r] * inv_l);
+            if (l16 == 0) {
+                float lv = (l_run[mf][r] > 0.f) ? (m_run[mf][r] + __logf(l_run[mf][r])) : -INFINITY;
+                lse[((long long)b * Hq + hq) * Sq + qg] = lv;
+            }
         }
     }
 }
@@ -358,19 +382,30 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
         const int kv_base = kvt * BLK_N;
-        for (int idx = tid * 8; idx < BLK_N * D; idx += FA_BLOCK * 8) {
-            int row = idx / D, col = idx % D;
-            int kvg = kv_base + row;
-            short8v k8 = {0, 0, 0, 0, 0, 0, 0, 0};
-            short8v v8 = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (kvg < Skv) {
-                k8 = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg * kv_row_stride + col);
-                v8 = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg * kv_row_stride + col);
+        for (int idx = tid * 8; idx < (BLK_N / 2) * D; idx += FA_BLOCK * 8) {
+            int rr = idx / D, col = idx % D;
+            int row0 = rr * 2;
+            int kvg0 = kv_base + row0;
+            short8v k8a = {0,0,0,0,0,0,0,0}, k8b = {0,0,0,0,0,0,0,0};
+            short8v v8a = {0,0,0,0,0,0,0,0}, v8b = {0,0,0,0,0,0,0,0};
+            if (kvg0 < Skv) {
+                k8a = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg0 * kv_row_stride + col);
+                v8a = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg0 * kv_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(&k_lds[row][col]) = k8;
-            *reinterpret_cast<short8v*>(&v_lds[row][col]) = v8;
+            if (kvg0 + 1 < Skv) {
+                k8b = *reinterpret_cast<const short8v*>(k_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+                v8b = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+            }
+            *reinterpret_cast<short8v*>(&k_lds[row0][col]) = k8a;
+            *reinterpret_cast<short8v*>(&k_lds[row0 + 1][col]) = k8b;
+            *reinterpret_cast<short8v*>(&v_lds[row0][col]) = v8a;
+            *reinterpret_cast<short8v*>(&v_lds[row0 + 1][col]) = v8b;
 #pragma unroll
-            for (int j = 0; j < 8; j++) kt_lds[col + j][row] = (ushort_t)k8[j];
+            for (int j = 0; j < 8; j++) {
+                unsigned int packed = ((unsigned int)(unsigned short)k8a[j]) |
+                                      (((unsigned int)(unsigned short)k8b[j]) << 16);
+                *reinterpret_cast<unsigned int*>(&kt_lds[col + j][row0]) = packed;
+            }
         }
         __syncthreads();
 
@@ -520,21 +555,32 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
 
     for (int qt = qt_start; qt < n_q_tiles; qt++) {
         const int q_base = qt * BLK_M;
-        for (int idx = tid * 8; idx < BLK_M * D; idx += FA_BLOCK * 8) {
-            int row = idx / D, col = idx % D;
-            int qg = q_base + row;
-            short8v q8 = {0, 0, 0, 0, 0, 0, 0, 0};
-            short8v d8 = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (qg < Sq) {
-                q8 = *reinterpret_cast<const short8v*>(q_ptr + (long long)qg * q_row_stride + col);
-                d8 = *reinterpret_cast<const short8v*>(do_ptr + (long long)qg * q_row_stride + col);
+        for (int idx = tid * 8; idx < (BLK_M / 2) * D; idx += FA_BLOCK * 8) {
+            int rr = idx / D, col = idx % D;
+            int row0 = rr * 2;
+            int qg0 = q_base + row0;
+            short8v q8a = {0,0,0,0,0,0,0,0}, q8b = {0,0,0,0,0,0,0,0};
+            short8v d8a = {0,0,0,0,0,0,0,0}, d8b = {0,0,0,0,0,0,0,0};
+            if (qg0 < Sq) {
+                q8a = *reinterpret_cast<const short8v*>(q_ptr + (long long)qg0 * q_row_stride + col);
+                d8a = *reinterpret_cast<const short8v*>(do_ptr + (long long)qg0 * q_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(&q_lds[row][col]) = q8;
-            *reinterpret_cast<short8v*>(&do_lds[row][col]) = d8;
+            if (qg0 + 1 < Sq) {
+                q8b = *reinterpret_cast<const short8v*>(q_ptr + (long long)(qg0 + 1) * q_row_stride + col);
+                d8b = *reinterpret_cast<const short8v*>(do_ptr + (long long)(qg0 + 1) * q_row_stride + col);
+            }
+            *reinterpret_cast<short8v*>(&q_lds[row0][col]) = q8a;
+            *reinterpret_cast<short8v*>(&q_lds[row0 + 1][col]) = q8b;
+            *reinterpret_cast<short8v*>(&do_lds[row0][col]) = d8a;
+            *reinterpret_cast<short8v*>(&do_lds[row0 + 1][col]) = d8b;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
-                qt_lds[col + j][row] = (ushort_t)q8[j];
-                dot_lds[col + j][row] = (ushort_t)d8[j];
+                unsigned int pq = ((unsigned int)(unsigned short)q8a[j]) |
+                                  (((unsigned int)(unsigned short)q8b[j]) << 16);
+                unsigned int pd = ((unsigned int)(unsigned short)d8a[j]) |
+                                  (((unsigned int)(unsigned short)d8b[j]) << 16);
+                *reinterpret_cast<unsigned int*>(&qt_lds[col + j][row0]) = pq;
+                *reinterpret_cast<unsigned int*>(&dot_lds[col + j][row0]) = pd;
             }
         }
         __syncthreads();
@@ -646,7 +692,7 @@ template <int D>
 static void flash_fwd_t(const void* q, const void* k, const void* v, void* o,
                         float* lse, int B, int Sq, int Skv, int Hq, int Hk,
                         float scale, bool causal, hipStream_t stream) {
-    dim3 grid((Sq + BLK_M - 1) / BLK_M, B * Hq);
+    dim3 grid((Sq + FWD_BLK_M - 1) / FWD_BLK_M, B * Hq);
     hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (ushort_t*)o, lse, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);

@@ -95,26 +95,24 @@ __global__ void rms_norm_bwd_dx_kernel(
     }
 }
 
-// reduce dw partials [P, H] -> dw [H] (bf16 out to match weight dtype grads)
+// reduce dw partials [P, H] -> dw32 [H] fp32.  Parallelized over BOTH axes:
+// grid.x covers columns, grid.y slices the P partials; one atomicAdd per
+// (column, slice) — ~grid.y adders per address, negligible contention (G12).
 __global__ void rms_norm_bwd_dw_reduce_kernel(
-    const float* __restrict__ dw_partial, ushort_t* __restrict__ dw,
+    const float* __restrict__ dw_partial, float* __restrict__ dw32,
     int H, int P) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= H) return;
+    int p0 = blockIdx.y;
     float acc = 0.f;
-    for (int p = 0; p < P; p++) acc += dw_partial[(long long)p * H + i];
-    dw[i] = f32_to_bf16(acc);
+    for (int p = p0; p < P; p += gridDim.y) acc += dw_partial[(long long)p * H + i];
+    atomicAdd(&dw32[i], acc);
 }
 
-// fp32-weight variant of the dw reduce (when weights are fp32)
-__global__ void rms_norm_bwd_dw_reduce_f32_kernel(
-    const float* __restrict__ dw_partial, float* __restrict__ dw,
-    int H, int P) {
+__global__ void rms_norm_bwd_dw_cast_kernel(
+    const float* __restrict__ dw32, ushort_t* __restrict__ dw, int H) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= H) return;
-    float acc = 0.f;
-    for (int p = 0; p < P; p++) acc += dw_partial[(long long)p * H + i];
-    dw[i] = acc;
+    if (i < H) dw[i] = f32_to_bf16(dw32[i]);
 }
 
 // ---- launchers (called from bindings.cpp) ----
@@ -127,17 +125,17 @@ void launch_rms_norm_fwd(const void* x, const void* w, void* y, float* invrms,
 
 void launch_rms_norm_bwd(const void* dy, const void* x, const void* w,
                          const float* invrms, void* dx, float* dw_partial,
-                         void* dw, bool dw_is_bf16, long long rows, int H, int P,
-                         hipStream_t stream) {
+                         float* dw32, void* dw, bool dw_is_bf16,
+                         long long rows, int H, int P, hipStream_t stream) {
     hipLaunchKernelGGL(rms_norm_bwd_dx_kernel, dim3(P), dim3(RMS_BLOCK), 0, stream,
                        (const ushort_t*)dy, (const ushort_t*)x, (const ushort_t*)w,
                        invrms, (ushort_t*)dx, dw_partial, H, rows);
     int rgrid = (H + 255) / 256;
+    int pslices = P > 64 ? 64 : P;
+    hipLaunchKernelGGL(rms_norm_bwd_dw_reduce_kernel, dim3(rgrid, pslices), dim3(256), 0, stream,
+                       dw_partial, dw32, H, P);
     if (dw_is_bf16) {
-        hipLaunchKernelGGL(rms_norm_bwd_dw_reduce_kernel, dim3(rgrid), dim3(256), 0, stream,
-                           dw_partial, (ushort_t*)dw, H, P);
-    } else {
-        hipLaunchKernelGGL(rms_norm_bwd_dw_reduce_f32_kernel, dim3(rgrid), dim3(256), 0, stream,
-                           dw_partial, (float*)dw, H, P);
+        hipLaunchKernelGGL(rms_norm_bwd_dw_cast_kernel, dim3(rgrid), dim3(256), 0, stream,
+                           dw32, (ushort_t*)dw, H);
     }
 }

@@ -368,7 +368,7 @@ class Trainer:
         if self._zero is not None:
             self._zero.step_post()
         self.lr_scheduler.step()
-        self.optimizer.zero_grad(set_to_none=False)
+        self.optimizer.zero_grad(set_to_none=True)
 
     def _clip_grad_norm(self, model):
         """Global grad-norm clip, TP/PP-aware: local sum-of-squares, reduced
