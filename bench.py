@@ -58,7 +58,7 @@ def main():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="llama3-8b")
-    p.add_argument("--micro-batch", type=int, default=2)
+    p.add_argument("--micro-batch", type=int, default=8)
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--accum", type=int, default=1)
     p.add_argument("--recompute", action="store_true")
