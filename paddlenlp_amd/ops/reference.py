@@ -87,7 +87,8 @@ def flash_attention(
     causal: bool = True,
     attn_mask: Optional[torch.Tensor] = None,
     startend_row_indices: Optional[torch.Tensor] = None,
-) -> torch.Tensor:
+    return_lse: bool = False,
+):
     """Reference attention in fp32 (materializes scores — small inputs only).
 
     GQA: k/v heads are repeated to match q heads.
@@ -124,7 +125,11 @@ def flash_attention(
     # fully-masked rows produce NaN via softmax(-inf row); zero them
     probs = torch.nan_to_num(probs, nan=0.0)
     out = probs @ v32  # [B, H, S, D]
-    return out.permute(0, 2, 1, 3).to(q.dtype)
+    out = out.permute(0, 2, 1, 3).to(q.dtype)
+    if return_lse:
+        lse = torch.logsumexp(scores, dim=-1)  # [B, H, S]; -inf for masked rows
+        return out, lse
+    return out
 
 
 def cross_entropy(

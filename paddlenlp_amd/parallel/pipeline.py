@@ -1,0 +1,298 @@
+"""Pipeline parallelism: LayerDesc model builder + 1F1B schedule over p2p.
+
+Reference behavior: paddle's PipelineLayer/PipelineParallel as used by
+LlamaForCausalLMPipe (paddlenlp/transformers/llama/modeling_pp.py:296 —
+add_sequential_layer LayerDesc list, SharedLayerDesc for tied embeddings
+:359-385, seg_method "layer:LlamaDecoderLayer" :391-393) and the trainer's
+training_pipeline_step (trainer.py:2246-2290, forward_backward_pipeline).
+
+MI355X design: the schedule is implemented directly on torch.distributed
+point-to-point ops (isend/irecv over RCCL on xGMI; gloo in CPU tests).
+Activations crossing stage boundaries are fixed-shape hidden-state tensors
+[B, S, H] in the compute dtype, so no shape negotiation is needed per
+micro-batch.  Non-interleaved 1F1B; virtual stages are a planned extension.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..utils.log import logger
+from .topology import get_topology
+
+
+class LayerDesc:
+    """Lazy layer constructor so only the owning stage materializes weights."""
+
+    def __init__(self, layer_cls, *args, name: Optional[str] = None, **kwargs):
+        self.layer_cls = layer_cls
+        self.args = args
+        self.kwargs = kwargs
+        self.name = name or layer_cls.__name__
+
+    def build(self) -> nn.Module:
+        return self.layer_cls(*self.args, **self.kwargs)
+
+
+class SharedLayerDesc(LayerDesc):
+    """Layer whose weights are tied across stages (e.g. embedding <-> head).
+
+    Reference: modeling_pp.py SharedLayerDesc :359-385.  All stages in
+    `shared_group` build the layer; after backward the tied weights' grads
+    are all-reduced so the copies stay identical.
+    """
+
+    def __init__(self, key: str, layer_cls, *args, forward_fn=None, **kwargs):
+        super().__init__(layer_cls, *args, **kwargs)
+        self.key = key
+        self.forward_fn = forward_fn
+
+
+def _segment_uniform(n_items: int, n_parts: int) -> List[int]:
+    """Boundaries (len n_parts+1) distributing n_items as evenly as possible."""
+    base = n_items // n_parts
+    extra = n_items % n_parts
+    bounds = [0]
+    for i in range(n_parts):
+        bounds.append(bounds[-1] + base + (1 if i < extra else 0))
+    return bounds
+
+
+class PipelineModule(nn.Module):
+    """Holds this stage's slice of a LayerDesc list.
+
+    seg_method:
+      - "uniform": split all descs evenly
+      - "layer:ClassName": balance by counting only that class; leading
+        descs attach to stage 0, trailing to the last stage
+        (reference seg_method llama modeling_pp.py:391-393)
+    """
+
+    def __init__(
+        self,
+        layer_descs: List[LayerDesc],
+        loss_fn: Optional[nn.Module] = None,
+        seg_method: str = "uniform",
+        topology=None,
+    ):
+        super().__init__()
+        self.topo = topology or get_topology()
+        self.pp_rank = self.topo.get_rank_in("pp")
+        self.pp_degree = self.topo.pp_degree
+        self.pp_group = self.topo.pipe_parallel_group
+        self.loss_fn = loss_fn
+        self.descs = layer_descs
+
+        start, end = self._segment(layer_descs, seg_method)
+        self.local_start, self.local_end = start, end
+        self.local_layers = nn.ModuleList()
+        self.shared_layers: Dict[str, nn.Module] = {}
+        self._layer_descs_local = []
+        for i in range(start, end):
+            desc = layer_descs[i]
+            layer = desc.build()
+            self.local_layers.append(layer)
+            self._layer_descs_local.append(desc)
+            if isinstance(desc, SharedLayerDesc):
+                self.shared_layers[desc.key] = layer
+        logger.info(
+            f"PP stage {self.pp_rank}/{self.pp_degree}: layers [{start}, {end}) of {len(layer_descs)}"
+        )
+
+        # shared-weight groups: ranks sharing a key all-reduce tied grads
+        self._shared_comm = self._build_shared_comm(layer_descs)
+
+    def _segment(self, descs, seg_method) -> Tuple[int, int]:
+        n = len(descs)
+        if self.pp_degree == 1:
+            return 0, n
+        if seg_method.startswith("layer:"):
+            cls_name = seg_method.split(":", 1)[1]
+            marks = [i for i, d in enumerate(descs) if d.layer_cls.__name__ == cls_name]
+            if not marks:
+                seg_method = "uniform"
+            else:
+                bounds = _segment_uniform(len(marks), self.pp_degree)
+                # stage boundaries in desc indices
+                starts = []
+                for p in range(self.pp_degree):
+                    starts.append(marks[bounds[p]] if bounds[p] < len(marks) else n)
+                starts.append(n)
+                start = starts[self.pp_rank] if self.pp_rank > 0 else 0
+                end = starts[self.pp_rank + 1] if self.pp_rank < self.pp_degree - 1 else n
+                return start, end
+        bounds = _segment_uniform(n, self.pp_degree)
+        return bounds[self.pp_rank], bounds[self.pp_rank + 1]
+
+    def _build_shared_comm(self, descs):
+        """For each SharedLayerDesc key, create a group of pp ranks holding it."""
+        comm = {}
+        if self.pp_degree == 1 or not dist.is_initialized():
+            return comm
+        # gather key -> owning stages via all_gather_object, then make one
+        # group per key (every pp rank participates in every new_group call)
+        my_keys = sorted(self.shared_layers.keys())
+        all_keys = [None] * dist.get_world_size(self.pp_group)
+        dist.all_gather_object(all_keys, my_keys, group=self.pp_group)
+        key_stages: Dict[str, List[int]] = {}
+        for stage, keys in enumerate(all_keys):
+            for k in keys or []:
+                key_stages.setdefault(k, []).append(stage)
+        pp_ranks = dist.get_process_group_ranks(self.pp_group)
+        for key, stages in key_stages.items():
+            if len(stages) > 1:
+                ranks = [pp_ranks[s] for s in stages]
+                # every rank must call new_group
+                group = dist.new_group(ranks=ranks)
+                if self.pp_rank in stages:
+                    comm[key] = (group, len(stages))
+        return comm
+
+    def allreduce_shared_weight_gradients(self):
+        for key, (group, _) in self._shared_comm.items():
+            layer = self.shared_layers.get(key)
+            if layer is None:
+                continue
+            for p in layer.parameters():
+                if p.grad is not None:
+                    dist.all_reduce(p.grad, group=group)
+
+    @property
+    def is_first_stage(self):
+        return self.pp_rank == 0
+
+    @property
+    def is_last_stage(self):
+        return self.pp_rank == self.pp_degree - 1
+
+    def stage_forward(self, x):
+        for layer, desc in zip(self.local_layers, self._layer_descs_local):
+            if isinstance(desc, SharedLayerDesc) and desc.forward_fn is not None:
+                x = desc.forward_fn(layer, x)
+            else:
+                x = layer(x)
+        return x
+
+    def forward(self, x):
+        if self.pp_degree != 1:
+            raise RuntimeError("use PipelineEngine.forward_backward for pp > 1")
+        return self.stage_forward(x)
+
+
+class PipelineEngine:
+    """Non-interleaved 1F1B over torch.distributed p2p.
+
+    The hidden state crossing stages is a single tensor [B, S, H].  The
+    first stage consumes `input_fn(micro_batch)`; the last stage computes
+    `loss_fn(output, micro_batch)`.
+    """
+
+    def __init__(self, module: PipelineModule, hidden_shape_fn: Callable,
+                 dtype: torch.dtype, device: torch.device):
+        self.module = module
+        self.topo = module.topo
+        self.pp_group = module.pp_group
+        self.pp_rank = module.pp_rank
+        self.pp_degree = module.pp_degree
+        self.hidden_shape_fn = hidden_shape_fn  # micro_batch -> (B, S, H)
+        self.dtype = dtype
+        self.device = device
+        ranks = (dist.get_process_group_ranks(self.pp_group)
+                 if self.pp_group is not None else [0])
+        self.prev_rank = ranks[self.pp_rank - 1] if self.pp_rank > 0 else None
+        self.next_rank = ranks[self.pp_rank + 1] if self.pp_rank < self.pp_degree - 1 else None
+
+    # -- p2p helpers.  Sends are non-blocking (isend) so the steady-state
+    # 1F1B pattern (rank r sending fwd to r+1 while r+1 sends grad to r)
+    # cannot rendezvous-deadlock; buffers are kept alive until waited.
+    def _send(self, tensor, dst):
+        buf = tensor.contiguous()
+        work = dist.isend(buf, dst=dst, group=self.pp_group)
+        self._pending.append((work, buf))
+
+    def _recv(self, shape, src):
+        buf = torch.empty(shape, dtype=self.dtype, device=self.device)
+        dist.recv(buf, src=src, group=self.pp_group)
+        return buf
+
+    def _drain_sends(self):
+        for work, _ in self._pending:
+            work.wait()
+        self._pending.clear()
+
+    def forward_backward(self, micro_batches: List[Dict[str, torch.Tensor]],
+                         input_fn: Callable, scale_loss: bool = True):
+        """Run 1F1B over the micro-batches; returns mean loss (last stage)."""
+        M = len(micro_batches)
+        P, r = self.pp_degree, self.pp_rank
+        num_warmup = min(P - r - 1, M)
+        num_steady = M - num_warmup
+        self._pending: List = []
+
+        fwd_inputs: List[Optional[torch.Tensor]] = []
+        fwd_outputs: List[Optional[torch.Tensor]] = []
+        losses = []
+        fwd_idx = 0
+        bwd_idx = 0
+
+        def run_forward(i):
+            mb = micro_batches[i]
+            if self.module.is_first_stage:
+                x = input_fn(mb)
+                x_in = None
+            else:
+                shape = self.hidden_shape_fn(mb)
+                x_in = self._recv(shape, self.prev_rank)
+                x_in.requires_grad_()
+                x = x_in
+            out = self.module.stage_forward(x)
+            if self.module.is_last_stage:
+                loss = self.module.loss_fn(out, mb)
+                if scale_loss:
+                    loss = loss / M
+                losses.append(loss)
+                fwd_inputs.append(x_in)
+                fwd_outputs.append(loss)
+            else:
+                self._send(out.detach(), self.next_rank)
+                fwd_inputs.append(x_in)
+                fwd_outputs.append(out)
+
+        def run_backward(i):
+            out = fwd_outputs[i]
+            x_in = fwd_inputs[i]
+            if self.module.is_last_stage:
+                out.backward()
+            else:
+                grad = self._recv(out.shape, self.next_rank)
+                out.backward(gradient=grad)
+            if not self.module.is_first_stage:
+                self._send(x_in.grad, self.prev_rank)
+            fwd_outputs[i] = None
+            fwd_inputs[i] = None
+
+        # warmup forwards
+        for _ in range(num_warmup):
+            run_forward(fwd_idx)
+            fwd_idx += 1
+        # steady 1F1B
+        for _ in range(num_steady):
+            run_forward(fwd_idx)
+            fwd_idx += 1
+            run_backward(bwd_idx)
+            bwd_idx += 1
+        # cooldown backwards
+        while bwd_idx < M:
+            run_backward(bwd_idx)
+            bwd_idx += 1
+
+        self._drain_sends()
+        self.module.allreduce_shared_weight_gradients()
+
+        if self.module.is_last_stage and losses:
+            return torch.stack([l.detach() for l in losses]).sum()
+        return torch.zeros((), device=self.device)

@@ -190,6 +190,23 @@ class Trainer:
         model = model.to(device)
 
         topo = self.topology
+        if topo.pp_degree > 1:
+            from ..parallel.pipeline import PipelineEngine, PipelineModule
+
+            if not isinstance(unwrap_model(model), PipelineModule):
+                raise ValueError(
+                    "pipeline_parallel_degree > 1 requires a PipelineModule "
+                    "(e.g. LlamaForCausalLMPipe)"
+                )
+            pipe = unwrap_model(model)
+            hidden = pipe.config.hidden_size
+
+            self._pipe_engine = PipelineEngine(
+                pipe,
+                hidden_shape_fn=lambda mb: (*mb["input_ids"].shape, hidden),
+                dtype=args.compute_dtype,
+                device=device,
+            )
         if topo.dp_degree > 1 and topo.data_parallel_group is not None:
             broadcast_parameters(model, topo.data_parallel_group)
         if topo.sharding_degree > 1 and topo.sharding_parallel_group is not None:
@@ -272,14 +289,22 @@ class Trainer:
 
         epoch = self.state.epoch
         accum_count = 0
+        pipe_buffer = []
+        is_pipeline = self.topology.pp_degree > 1
         done = False
         while not done:
             for step, inputs in enumerate(train_dataloader):
                 if accum_count == 0:
                     self.control = self.callback_handler.on_step_begin(args, self.state, self.control)
 
-                loss = self.training_step(model, inputs)
-                tr_loss += loss.detach()
+                if is_pipeline:
+                    # buffer micro-batches; the 1F1B engine consumes them as
+                    # one optimizer step (reference training_pipeline_step
+                    # trainer.py:2246-2290)
+                    pipe_buffer.append(self._prepare_inputs(inputs))
+                else:
+                    loss = self.training_step(model, inputs)
+                    tr_loss += loss.detach()
                 if "input_ids" in inputs:
                     self._tokens_since_last_log += inputs["input_ids"].numel()
                 accum_count += 1
@@ -291,6 +316,11 @@ class Trainer:
                     self.control = self.callback_handler.on_substep_end(args, self.state, self.control)
                     continue
                 accum_count = 0
+
+                if is_pipeline:
+                    loss = self.training_pipeline_step(pipe_buffer)
+                    pipe_buffer = []
+                    tr_loss += loss.detach()
 
                 self.optimizer_step(model)
                 self.state.global_step += 1
@@ -342,10 +372,23 @@ class Trainer:
             loss = outputs
         return (loss, outputs) if return_outputs else loss
 
+    def training_pipeline_step(self, micro_batches) -> torch.Tensor:
+        """One 1F1B pipeline pass over the buffered micro-batches."""
+        loss = self._pipe_engine.forward_backward(
+            micro_batches, input_fn=lambda mb: mb["input_ids"]
+        )
+        return loss
+
     def training_step(self, model: nn.Module, inputs: Dict[str, Any]) -> torch.Tensor:
         """Forward + backward for one micro-batch (reference :2211)."""
         model.train()
         inputs = self._prepare_inputs(inputs)
+        if self.topology.sep_degree > 1:
+            # sep/cp: every sep rank sees the same batch, sliced on seq
+            # (reference split_inputs_sequence_dim, trainer.py:972-975)
+            from ..parallel.segment_parallel import split_inputs_sequence_dim
+
+            inputs = split_inputs_sequence_dim(inputs, self.topology.sep_parallel_group)
         loss = self.compute_loss(model, inputs)
         if self.args.gradient_accumulation_steps > 1:
             loss = loss / self.args.gradient_accumulation_steps
@@ -360,6 +403,15 @@ class Trainer:
             self._zero.reduce_gradients_and_step_pre()
         if topo.dp_degree > 1:
             fused_allreduce_gradients(model.parameters(), topo.data_parallel_group)
+        if topo.sep_degree > 1:
+            # sep/cp ranks each backprop their seq slice of the same batch
+            fused_allreduce_gradients(model.parameters(), topo.sep_parallel_group)
+        if topo.mp_degree > 1 and getattr(self.model, "config", None) is not None and \
+                getattr(unwrap_model(self.model).config, "sequence_parallel", False):
+            # norm/bias params under sequence parallel see sharded activations
+            for p in model.parameters():
+                if getattr(p, "sequence_parallel", False) and p.grad is not None:
+                    dist.all_reduce(p.grad, group=topo.model_parallel_group)
 
         if args.max_grad_norm and args.max_grad_norm > 0:
             self._clip_grad_norm(model)

@@ -48,6 +48,10 @@ class LlamaRMSNorm(nn.Module):
         hidden_size = hidden_size or config.hidden_size
         self.weight = nn.Parameter(torch.ones(hidden_size))
         self.eps = config.rms_norm_eps
+        if config.sequence_parallel and config.tensor_parallel_degree > 1:
+            # sees seq-sharded activations: grad needs an mp all-reduce
+            # (reference register_sequence_parallel_allreduce_hooks)
+            self.weight.sequence_parallel = True
 
     def forward(self, x):
         return ops.rms_norm(x, self.weight, self.eps)
@@ -131,8 +135,7 @@ class LlamaAttention(nn.Module):
 
         q_out = self.num_heads * self.head_dim
         kv_out = self.num_kv_heads * self.head_dim
-        Column = ColumnParallelLinear if tp > 1 else _Linear
-        Row = RowParallelLinear if tp > 1 else _Linear
+        Column, Row = _linear_classes(config)
 
         if config.fuse_attention_qkv:
             self.qkv_proj = Column(self.hidden_size, q_out + 2 * kv_out, bias=False)
@@ -142,6 +145,16 @@ class LlamaAttention(nn.Module):
             self.v_proj = Column(self.hidden_size, kv_out, bias=False)
         self.o_proj = Row(q_out, self.hidden_size, bias=False)
         self.rotary_emb = LlamaRotaryEmbedding(config)
+        # Ulysses reshard around the attention core (reference
+        # segment_parallel_utils.py ReshardLayer, llama/modeling.py:811-815)
+        self.sep_degree = max(config.sep_parallel_degree, 1)
+        self.cp_degree = max(config.context_parallel_degree, 1)
+        if self.sep_degree > 1:
+            from ...parallel.segment_parallel import ReshardLayer
+
+            assert self.num_heads_local % self.sep_degree == 0
+            assert self.num_kv_heads_local % self.sep_degree == 0
+            self.reshard = ReshardLayer()
 
     def forward(
         self,
@@ -152,7 +165,7 @@ class LlamaAttention(nn.Module):
         use_cache: bool = False,
         position_offset: int = 0,
     ):
-        B, S, _ = hidden_states.shape
+        B = hidden_states.shape[0]
         hl, kl, d = self.num_heads_local, self.num_kv_heads_local, self.head_dim
 
         if self.config.fuse_attention_qkv:
@@ -162,9 +175,17 @@ class LlamaAttention(nn.Module):
             q = self.q_proj(hidden_states)
             k = self.k_proj(hidden_states)
             v = self.v_proj(hidden_states)
+        # with sequence parallel the column linear all-gathered the seq dim,
+        # so infer S from the projection output, not the input
+        S = q.shape[1]
         q = q.view(B, S, hl, d)
         k = k.view(B, S, kl, d)
         v = v.view(B, S, kl, d)
+
+        # sep/cp: this rank holds a seq chunk; rope positions are offset
+        if self.sep_degree > 1 or self.cp_degree > 1:
+            sep_rank = get_topology().get_rank_in("sep")
+            position_offset = position_offset + sep_rank * S
 
         cos, sin = self.rotary_emb(S, hidden_states.device, position_offset)
         if self.config.use_fused_rope:
@@ -177,11 +198,23 @@ class LlamaAttention(nn.Module):
             v = torch.cat([past_key_value[1], v], dim=1)
         present = (k, v) if use_cache else None
 
-        attn_out = ops.flash_attention(
-            q, k, v, causal=True, attn_mask=attn_mask,
-            startend_row_indices=startend_row_indices,
-        )
-        attn_out = attn_out.reshape(B, S, hl * d)
+        if self.cp_degree > 1:
+            from ...parallel.ring_attention import ring_flash_attention
+
+            attn_out = ring_flash_attention(q, k, v, causal=True)
+        elif self.sep_degree > 1:
+            # Ulysses: [B, S/sep, H, D] -> [B, S, H/sep, D] around the core
+            q = self.reshard.seq_to_head(q)
+            k = self.reshard.seq_to_head(k)
+            v = self.reshard.seq_to_head(v)
+            attn_out = ops.flash_attention(q, k, v, causal=True)
+            attn_out = self.reshard.head_to_seq(attn_out)
+        else:
+            attn_out = ops.flash_attention(
+                q, k, v, causal=True, attn_mask=attn_mask,
+                startend_row_indices=startend_row_indices,
+            )
+        attn_out = attn_out.reshape(B, attn_out.shape[1], hl * d)
         out = self.o_proj(attn_out)
         if use_cache:
             return out, present
@@ -195,12 +228,26 @@ class _Linear(nn.Linear):
         super().__init__(in_features, out_features, bias=bias)
 
 
+def _linear_classes(config):
+    """(Column, Row) linear classes for this parallel config.
+
+    Reference: paddlenlp/transformers/linear_utils.py:33-38 aliasing."""
+    if config.tensor_parallel_degree > 1:
+        if config.sequence_parallel:
+            from ...parallel.sequence_parallel import (
+                ColumnSequenceParallelLinear,
+                RowSequenceParallelLinear,
+            )
+
+            return ColumnSequenceParallelLinear, RowSequenceParallelLinear
+        return ColumnParallelLinear, RowParallelLinear
+    return _Linear, _Linear
+
+
 class LlamaMLP(nn.Module):
     def __init__(self, config: LlamaConfig):
         super().__init__()
-        tp = config.tensor_parallel_degree
-        Column = ColumnParallelLinear if tp > 1 else _Linear
-        Row = RowParallelLinear if tp > 1 else _Linear
+        Column, Row = _linear_classes(config)
         self.config = config
         h, i = config.hidden_size, config.intermediate_size
         if config.fuse_attention_ffn:
@@ -376,6 +423,12 @@ class LlamaModel(LlamaPretrainedModel):
         if inputs_embeds is None:
             inputs_embeds = self.embed_tokens(input_ids)
         hidden_states = inputs_embeds
+        if self.config.sequence_parallel and self.config.tensor_parallel_degree > 1:
+            # shard activations on the seq dim over the mp group
+            # (reference llama/modeling.py ScatterOp after embedding)
+            from ...parallel.sequence_parallel import ScatterOp
+
+            hidden_states = ScatterOp(hidden_states)
 
         position_offset = 0
         if past_key_values is not None and past_key_values[0] is not None:
@@ -479,6 +532,12 @@ class LlamaForCausalLM(LlamaPretrainedModel):
             hidden_states, presents = out
         else:
             hidden_states, presents = out, None
+        if self.config.sequence_parallel and self.config.tensor_parallel_degree > 1:
+            # gather the full sequence before the LM head
+            # (reference llama/modeling.py:1896 GatherOp)
+            from ...parallel.sequence_parallel import GatherOp
+
+            hidden_states = GatherOp(hidden_states)
         logits = self.lm_head(hidden_states)
         if labels is not None:
             loss = self.criterion(logits, labels)

@@ -1,0 +1,128 @@
+"""Pipeline-parallel Llama: flattened LayerDesc stack.
+
+Reference behavior: paddlenlp/transformers/llama/modeling_pp.py:296
+(LlamaForCausalLMPipe — add_sequential_layer list Embedding -> N x
+DecoderLayer -> RMSNorm -> LMHead :360-387, SharedLayerDesc for tied
+embeddings :359-385, seg_method "layer:LlamaDecoderLayer" :391-393).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ...parallel.pipeline import LayerDesc, PipelineModule, SharedLayerDesc
+from ...parallel.topology import get_topology
+from .configuration import LlamaConfig
+from .modeling import (
+    LlamaDecoderLayer,
+    LlamaPretrainedModel,
+    LlamaPretrainingCriterion,
+    LlamaRMSNorm,
+    _Linear,
+)
+from ...parallel.tensor_parallel import ColumnParallelLinear, VocabParallelEmbedding
+
+__all__ = ["LlamaForCausalLMPipe"]
+
+
+class EmbeddingPipe(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.config = config
+        if config.tensor_parallel_degree > 1:
+            self.embed_tokens = VocabParallelEmbedding(config.vocab_size, config.hidden_size)
+        else:
+            self.embed_tokens = nn.Embedding(config.vocab_size, config.hidden_size)
+
+    @property
+    def embedding_weight(self):
+        return self.embed_tokens.weight
+
+    def forward(self, input_ids):
+        return self.embed_tokens(input_ids)
+
+
+class RMSNormPipe(LlamaRMSNorm):
+    pass
+
+
+class LMHeadPipe(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.config = config
+        if config.tensor_parallel_degree > 1:
+            self.lm_head = ColumnParallelLinear(
+                config.hidden_size, config.vocab_size, bias=False,
+                gather_output=not config.tensor_parallel_output,
+            )
+        else:
+            self.lm_head = _Linear(config.hidden_size, config.vocab_size, bias=False)
+
+    def forward(self, hidden):
+        return self.lm_head(hidden)
+
+
+class LlamaForCausalLMPipe(PipelineModule):
+    """Build with `LlamaForCausalLMPipe(config)`; run through PipelineEngine.
+
+    Weight names intentionally match LlamaForCausalLM once remapped by
+    `pp_param_name_map` (PipelinePretrainedModel behavior in the reference
+    model_utils.py: maps pipe-layer param names back to base names).
+    """
+
+    config_class = LlamaConfig
+
+    def __init__(self, config: LlamaConfig):
+        criterion = LlamaPretrainingCriterion(config)
+        descs = [LayerDesc(EmbeddingPipe, config, name="embedding")]
+        for i in range(config.num_hidden_layers):
+            descs.append(LayerDesc(LlamaDecoderLayer, config, i, name=f"layer_{i}"))
+        descs.append(LayerDesc(RMSNormPipe, config, name="final_norm"))
+        descs.append(LayerDesc(LMHeadPipe, config, name="lm_head"))
+
+        def loss_fn(logits, micro_batch):
+            return criterion(logits, micro_batch["labels"])
+
+        super().__init__(
+            descs,
+            loss_fn=loss_fn,
+            seg_method="layer:LlamaDecoderLayer",
+            topology=get_topology(),
+        )
+        self.config = config
+        self._local_names = [d.name for d in self._layer_descs_local]
+
+    def pp_param_name_map(self):
+        """pipe-local param name -> LlamaForCausalLM name (for checkpoints)."""
+        mapping = {}
+        for idx, (layer, desc) in enumerate(zip(self.local_layers, self._layer_descs_local)):
+            for pname, _ in layer.named_parameters():
+                local = f"local_layers.{idx}.{pname}"
+                if desc.name == "embedding":
+                    base = pname.replace("embed_tokens", "llama.embed_tokens")
+                elif desc.name.startswith("layer_"):
+                    li = desc.name.split("_")[1]
+                    base = f"llama.layers.{li}.{pname}"
+                elif desc.name == "final_norm":
+                    base = f"llama.norm.{pname}"
+                else:  # lm_head
+                    base = pname if pname.startswith("lm_head") else f"lm_head.{pname}"
+                mapping[local] = base
+        return mapping
+
+    def state_dict_with_base_names(self):
+        mapping = self.pp_param_name_map()
+        sd = self.state_dict()
+        return {mapping.get(k, k): v for k, v in sd.items()}
+
+    def load_base_state_dict(self, base_sd, strict: bool = False):
+        mapping = {v: k for k, v in self.pp_param_name_map().items()}
+        local_sd = {}
+        for base_name, tensor in base_sd.items():
+            if base_name in mapping:
+                local_sd[mapping[base_name]] = tensor
+        missing, unexpected = self.load_state_dict(local_sd, strict=False)
+        real_missing = [m for m in missing if not m.startswith("local_layers")] if strict else []
+        return missing, unexpected

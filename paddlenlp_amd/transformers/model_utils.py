@@ -282,6 +282,10 @@ class PretrainedModel(nn.Module):
         return model
 
 
+# parallelism metadata attached to parameters that must survive reload
+_PARAM_FLAGS = ("sequence_parallel", "is_column_parallel", "is_row_parallel", "no_sync")
+
+
 def _assign_param(model: nn.Module, name: str, tensor: torch.Tensor):
     module = model
     parts = name.split(".")
@@ -290,7 +294,11 @@ def _assign_param(model: nn.Module, name: str, tensor: torch.Tensor):
     leaf = parts[-1]
     old = getattr(module, leaf)
     if isinstance(old, nn.Parameter):
-        setattr(module, leaf, nn.Parameter(tensor, requires_grad=old.requires_grad))
+        new = nn.Parameter(tensor, requires_grad=old.requires_grad)
+        for flag in _PARAM_FLAGS:
+            if hasattr(old, flag):
+                setattr(new, flag, getattr(old, flag))
+        setattr(module, leaf, new)
     else:
         # buffer
         module.register_buffer(leaf, tensor, persistent=leaf in dict(module.named_buffers()))

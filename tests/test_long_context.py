@@ -1,0 +1,257 @@
+"""SP / SEP (Ulysses) / ring-attention tests on gloo, world 2."""
+import torch
+import torch.distributed as dist
+
+from tests.test_distributed import _run_workers
+
+
+def _w_sp_ops(rank, world):
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.sequence_parallel import (
+        AllGatherOp, GatherOp, ReduceScatterOp, ScatterOp,
+        ColumnSequenceParallelLinear, RowSequenceParallelLinear,
+    )
+
+    topo = init_parallel_env(mp_degree=world, backend="gloo")
+    g = topo.model_parallel_group
+    torch.manual_seed(0)
+    full = torch.randn(2, 8, 4)
+
+    # scatter -> gather round trip
+    local = ScatterOp(full, g)
+    assert local.shape == (2, 4, 4)
+    back = GatherOp(local, g)
+    assert torch.equal(back, full)
+
+    # AllGather fwd == full; bwd reduce-scatters
+    x = ScatterOp(full, g).requires_grad_()
+    y = AllGatherOp(x, g)
+    assert torch.allclose(y, full)
+    y.sum().backward()
+    assert torch.allclose(x.grad, torch.full_like(x, float(world)))
+
+    # Column->Row SP pair == plain 2-layer on full seq
+    torch.manual_seed(3)
+    w1 = torch.randn(12, 4)
+    w2 = torch.randn(4, 12)
+    col = ColumnSequenceParallelLinear(4, 12, group=g)
+    row = RowSequenceParallelLinear(12, 4, group=g)
+    with torch.no_grad():
+        col.weight.copy_(w1.chunk(world, 0)[rank])
+        row.weight.copy_(w2.chunk(world, 1)[rank])
+    xs = ScatterOp(full, g)
+    out_local = row(col(xs))
+    ref = (full @ w1.t()) @ w2.t()
+    ref_local = ref.chunk(world, dim=1)[rank]
+    assert torch.allclose(out_local, ref_local, atol=1e-5)
+
+
+def _w_ulysses(rank, world):
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.segment_parallel import ReshardLayer, split_inputs_sequence_dim
+
+    topo = init_parallel_env(sep_degree=world, backend="gloo")
+    g = topo.sep_parallel_group
+    torch.manual_seed(1)
+    B, S, H, D = 2, 8, 4, 6
+    full = torch.randn(B, S, H, D)
+    local = full.chunk(world, dim=1)[rank].clone().requires_grad_()
+
+    rs = ReshardLayer(g)
+    x = rs.seq_to_head(local)   # [B, S, H/w, D]
+    assert x.shape == (B, S, H // world, D)
+    # must equal the full tensor's head chunk
+    expect = full[:, :, rank * (H // world):(rank + 1) * (H // world)]
+    assert torch.allclose(x, expect, atol=1e-6), (x - expect).abs().max()
+    back = rs.head_to_seq(x)
+    assert torch.allclose(back, local, atol=1e-6)
+    # gradient flows through both all-to-alls
+    back.sum().backward()
+    assert torch.allclose(local.grad, torch.ones_like(local))
+
+    inputs = {"input_ids": torch.arange(16).reshape(2, 8)}
+    sliced = split_inputs_sequence_dim(inputs, g)
+    assert sliced["input_ids"].shape == (2, 4)
+
+
+def _w_ring_attention(rank, world):
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.ring_attention import ring_flash_attention
+    from paddlenlp_amd.ops import reference
+
+    topo = init_parallel_env(sep_degree=world, backend="gloo")
+    g = topo.sep_parallel_group
+    torch.manual_seed(2)
+    B, S, Hq, Hk, D = 1, 16, 4, 2, 8
+    q = torch.randn(B, S, Hq, D)
+    k = torch.randn(B, S, Hk, D)
+    v = torch.randn(B, S, Hk, D)
+
+    Sl = S // world
+    ql = q[:, rank * Sl:(rank + 1) * Sl].clone().requires_grad_()
+    kl = k[:, rank * Sl:(rank + 1) * Sl].clone().requires_grad_()
+    vl = v[:, rank * Sl:(rank + 1) * Sl].clone().requires_grad_()
+
+    out = ring_flash_attention(ql, kl, vl, group=g, causal=True)
+
+    qr = q.clone().requires_grad_()
+    kr = k.clone().requires_grad_()
+    vr = v.clone().requires_grad_()
+    ref = reference.flash_attention(qr, kr, vr, causal=True)
+    ref_local = ref[:, rank * Sl:(rank + 1) * Sl]
+    assert torch.allclose(out, ref_local, atol=1e-4), (out - ref_local).abs().max()
+
+    # backward parity: seed with a deterministic full-seq grad
+    gfull = torch.arange(ref.numel(), dtype=torch.float32).reshape(ref.shape) / ref.numel()
+    out.backward(gfull[:, rank * Sl:(rank + 1) * Sl])
+    ref.backward(gfull)
+    assert torch.allclose(ql.grad, qr.grad[:, rank * Sl:(rank + 1) * Sl], atol=1e-4)
+    assert torch.allclose(kl.grad, kr.grad[:, rank * Sl:(rank + 1) * Sl], atol=1e-4), \
+        (kl.grad - kr.grad[:, rank * Sl:(rank + 1) * Sl]).abs().max()
+    assert torch.allclose(vl.grad, vr.grad[:, rank * Sl:(rank + 1) * Sl], atol=1e-4)
+
+
+def _w_ring_noncausal(rank, world):
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.ring_attention import ring_flash_attention
+    from paddlenlp_amd.ops import reference
+
+    topo = init_parallel_env(sep_degree=world, backend="gloo")
+    g = topo.sep_parallel_group
+    torch.manual_seed(4)
+    B, S, H, D = 1, 8, 2, 8
+    q = torch.randn(B, S, H, D)
+    k = torch.randn(B, S, H, D)
+    v = torch.randn(B, S, H, D)
+    Sl = S // world
+    out = ring_flash_attention(
+        q[:, rank * Sl:(rank + 1) * Sl].contiguous(),
+        k[:, rank * Sl:(rank + 1) * Sl].contiguous(),
+        v[:, rank * Sl:(rank + 1) * Sl].contiguous(),
+        group=g, causal=False,
+    )
+    ref = reference.flash_attention(q, k, v, causal=False)
+    assert torch.allclose(out, ref[:, rank * Sl:(rank + 1) * Sl], atol=1e-4)
+
+
+def test_sequence_parallel_ops():
+    _run_workers(_w_sp_ops)
+
+
+def test_ulysses_reshard():
+    _run_workers(_w_ulysses)
+
+
+def test_ring_attention_causal():
+    _run_workers(_w_ring_attention)
+
+
+def test_ring_attention_noncausal():
+    _run_workers(_w_ring_noncausal)
+
+
+def _w_sp_llama(rank, world):
+    """TP2 + sequence-parallel Llama forward == single-process forward."""
+    import os, tempfile, shutil
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    topo = init_parallel_env(mp_degree=world, backend="gloo")
+    torch.manual_seed(21)
+    base_cfg = dict(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, dtype="float32",
+    )
+    full = LlamaForCausalLM.from_config(LlamaConfig(**base_cfg))
+    tmp = "/tmp/pdnlp_sp_test"
+    if rank == 0:
+        os.makedirs(tmp, exist_ok=True)
+        full.save_pretrained(tmp)
+    dist.barrier()
+    cfg = LlamaConfig(**{**base_cfg, "tensor_parallel_degree": world,
+                         "tensor_parallel_rank": rank, "sequence_parallel": True,
+                         "tensor_parallel_output": False})
+    sp_model = LlamaForCausalLM.from_pretrained(tmp, config=cfg)
+    ids = torch.randint(0, 128, (2, 16), generator=torch.Generator().manual_seed(22))
+    labels = torch.randint(0, 128, (2, 16), generator=torch.Generator().manual_seed(23))
+    ref_loss, ref_logits = full(input_ids=ids, labels=labels)
+    loss, logits = sp_model(input_ids=ids, labels=labels)
+    assert torch.allclose(logits, ref_logits, atol=1e-4), (logits - ref_logits).abs().max()
+    assert torch.allclose(loss, ref_loss, atol=1e-5)
+    # backward: norm-weight grads need the mp allreduce to match
+    loss.backward()
+    ref_loss.backward()
+    import torch.distributed as _d
+    for n, p in sp_model.named_parameters():
+        if getattr(p, "sequence_parallel", False) and p.grad is not None:
+            _d.all_reduce(p.grad, group=topo.model_parallel_group)
+    ref_params = dict(full.named_parameters())
+    g1 = sp_model.llama.norm.weight.grad
+    g2 = ref_params["llama.norm.weight"].grad
+    assert torch.allclose(g1, g2, atol=1e-4), (g1 - g2).abs().max()
+    dist.barrier()
+    if rank == 0:
+        shutil.rmtree(tmp, ignore_errors=True)
+
+
+def _w_sep_llama(rank, world):
+    """sep(Ulysses)-parallel Llama forward == sliced single-process forward."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    topo = init_parallel_env(sep_degree=world, backend="gloo")
+    base_cfg = dict(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, dtype="float32",
+    )
+    torch.manual_seed(31)
+    full = LlamaForCausalLM.from_config(LlamaConfig(**base_cfg))
+    cfg = LlamaConfig(**{**base_cfg, "sep_parallel_degree": world})
+    torch.manual_seed(31)
+    sep_model = LlamaForCausalLM.from_config(cfg)
+    ids = torch.randint(0, 128, (2, 16), generator=torch.Generator().manual_seed(32))
+    with torch.no_grad():
+        ref = full(input_ids=ids)
+        S = ids.shape[1] // world
+        local = sep_model(input_ids=ids[:, rank * S:(rank + 1) * S])
+    assert torch.allclose(local, ref[:, rank * S:(rank + 1) * S], atol=1e-4), \
+        (local - ref[:, rank * S:(rank + 1) * S]).abs().max()
+
+
+def _w_cp_llama(rank, world):
+    """context-parallel (ring) Llama forward == sliced single-process."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    topo = init_parallel_env(sep_degree=world, backend="gloo")
+    base_cfg = dict(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, dtype="float32",
+    )
+    torch.manual_seed(41)
+    full = LlamaForCausalLM.from_config(LlamaConfig(**base_cfg))
+    cfg = LlamaConfig(**{**base_cfg, "context_parallel_degree": world})
+    torch.manual_seed(41)
+    cp_model = LlamaForCausalLM.from_config(cfg)
+    ids = torch.randint(0, 128, (2, 16), generator=torch.Generator().manual_seed(42))
+    with torch.no_grad():
+        ref = full(input_ids=ids)
+    S = ids.shape[1] // world
+    local = cp_model(input_ids=ids[:, rank * S:(rank + 1) * S])
+    assert torch.allclose(local, ref[:, rank * S:(rank + 1) * S], atol=1e-4), \
+        (local - ref[:, rank * S:(rank + 1) * S]).abs().max()
+
+
+def test_sp_llama_parity():
+    _run_workers(_w_sp_llama)
+
+
+def test_sep_llama_parity():
+    _run_workers(_w_sep_llama)
+
+
+def test_cp_llama_parity():
+    _run_workers(_w_cp_llama)
