@@ -8,3 +8,4 @@ from .llama import (  # noqa: F401
     LlamaModel,
     LlamaPretrainingCriterion,
 )
+from .gpt import GPTConfig, GPTForCausalLM, GPTModel  # noqa: F401
