@@ -1,0 +1,9 @@
+from .configuration_utils import GenerationConfig  # noqa: F401
+from .logits_process import (  # noqa: F401
+    LogitsProcessorList,
+    RepetitionPenaltyLogitsProcessor,
+    TemperatureLogitsWarper,
+    TopKLogitsWarper,
+    TopPLogitsWarper,
+)
+from .utils import GenerationMixin  # noqa: F401

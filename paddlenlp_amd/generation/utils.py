@@ -1,0 +1,205 @@
+"""GenerationMixin: training-free decoding loops.
+
+Reference behavior: paddlenlp/generation/utils.py — GenerationMixin :319,
+generate :609 dispatching to greedy_search :1036, sample :1137, beam_search
+:1496; KV-cached decode via the model's use_cache/past_key_values protocol.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+
+from .configuration_utils import GenerationConfig
+from .logits_process import (
+    LogitsProcessorList,
+    MinNewTokensLengthLogitsProcessor,
+    RepetitionPenaltyLogitsProcessor,
+    TemperatureLogitsWarper,
+    TopKLogitsWarper,
+    TopPLogitsWarper,
+)
+
+
+class GenerationMixin:
+    """Mixed into CausalLM models.  The model forward must support
+    (input_ids, use_cache=True, past_key_values=...) -> (logits, presents)."""
+
+    generation_config: Optional[GenerationConfig] = None
+
+    def prepare_inputs_for_generation(self, input_ids, past_key_values=None, **kwargs):
+        if past_key_values is not None:
+            input_ids = input_ids[:, -1:]
+        return {"input_ids": input_ids, "past_key_values": past_key_values}
+
+    def _get_logits_processors(self, gen_config: GenerationConfig, prompt_len: int):
+        procs = LogitsProcessorList()
+        if gen_config.repetition_penalty != 1.0:
+            procs.append(RepetitionPenaltyLogitsProcessor(gen_config.repetition_penalty))
+        if gen_config.min_new_tokens > 0 and gen_config.eos_ids():
+            procs.append(MinNewTokensLengthLogitsProcessor(
+                prompt_len, gen_config.min_new_tokens, gen_config.eos_ids()))
+        return procs
+
+    def _get_logits_warpers(self, gen_config: GenerationConfig):
+        warpers = LogitsProcessorList()
+        if gen_config.temperature != 1.0:
+            warpers.append(TemperatureLogitsWarper(gen_config.temperature))
+        if gen_config.top_k > 0:
+            warpers.append(TopKLogitsWarper(gen_config.top_k))
+        if gen_config.top_p < 1.0:
+            warpers.append(TopPLogitsWarper(gen_config.top_p))
+        return warpers
+
+    @torch.no_grad()
+    def generate(
+        self,
+        input_ids: torch.Tensor,
+        generation_config: Optional[GenerationConfig] = None,
+        **kwargs,
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Returns (generated_ids [B, new], scores placeholder)."""
+        gen = generation_config or self.generation_config or GenerationConfig.from_model_config(self.config)
+        for k, v in kwargs.items():
+            if hasattr(gen, k):
+                setattr(gen, k, v)
+        if gen.max_length:
+            gen.max_new_tokens = max(1, gen.max_length - input_ids.shape[1])
+
+        if gen.num_beams > 1:
+            return self.beam_search(input_ids, gen)
+        if gen.do_sample:
+            return self.sample(input_ids, gen)
+        return self.greedy_search(input_ids, gen)
+
+    def _decode_loop(self, input_ids, gen: GenerationConfig, select_fn):
+        B = input_ids.shape[0]
+        device = input_ids.device
+        eos_ids = gen.eos_ids()
+        procs = self._get_logits_processors(gen, input_ids.shape[1])
+        unfinished = torch.ones(B, dtype=torch.bool, device=device)
+        pad_id = gen.pad_token_id if gen.pad_token_id is not None else (eos_ids[0] if eos_ids else 0)
+
+        past = None
+        all_ids = input_ids
+        new_tokens = []
+        cur = input_ids
+        for step in range(gen.max_new_tokens):
+            if gen.use_cache:
+                out = self(input_ids=cur, use_cache=True, past_key_values=past)
+                logits, past = out
+            else:
+                logits = self(input_ids=all_ids)
+            next_logits = logits[:, -1].float()
+            next_logits = procs(all_ids, next_logits)
+            token = select_fn(all_ids, next_logits)
+            token = torch.where(unfinished, token, torch.full_like(token, pad_id))
+            new_tokens.append(token)
+            all_ids = torch.cat([all_ids, token[:, None]], dim=1)
+            cur = token[:, None]
+            if eos_ids:
+                for eos in eos_ids:
+                    unfinished = unfinished & (token != eos)
+                if not unfinished.any():
+                    break
+        gen_ids = torch.stack(new_tokens, dim=1) if new_tokens else input_ids.new_zeros(B, 0)
+        return gen_ids, None
+
+    def greedy_search(self, input_ids, gen: GenerationConfig):
+        return self._decode_loop(input_ids, gen, lambda ids, lg: lg.argmax(-1))
+
+    def sample(self, input_ids, gen: GenerationConfig):
+        warpers = self._get_logits_warpers(gen)
+
+        def select(ids, lg):
+            lg = warpers(ids, lg)
+            probs = lg.softmax(-1)
+            return torch.multinomial(probs, 1).squeeze(-1)
+
+        return self._decode_loop(input_ids, gen, select)
+
+    def beam_search(self, input_ids, gen: GenerationConfig):
+        """Standard length-penalized beam search (no cache reordering
+        subtleties: the KV cache is reindexed per step)."""
+        B, prompt_len = input_ids.shape
+        K = gen.num_beams
+        device = input_ids.device
+        eos_ids = gen.eos_ids()
+        procs = self._get_logits_processors(gen, prompt_len)
+
+        # expand to beams
+        ids = input_ids.repeat_interleave(K, dim=0)  # [B*K, L]
+        beam_scores = torch.full((B, K), float("-inf"), device=device)
+        beam_scores[:, 0] = 0.0
+        beam_scores = beam_scores.view(-1)
+        past = None
+        cur = ids
+        finished = [[] for _ in range(B)]  # (score, seq)
+
+        for step in range(gen.max_new_tokens):
+            out = self(input_ids=cur, use_cache=True, past_key_values=past)
+            logits, past = out
+            logp = logits[:, -1].float().log_softmax(-1)
+            logp = procs(ids, logp)
+            vocab = logp.shape[-1]
+            scores = beam_scores[:, None] + logp  # [B*K, V]
+            scores = scores.view(B, K * vocab)
+            top_scores, top_idx = scores.topk(2 * K, dim=-1)
+            beam_idx = top_idx // vocab           # [B, 2K]
+            token_idx = top_idx % vocab
+
+            new_ids, new_scores, new_beam_src = [], [], []
+            for b in range(B):
+                row_ids, row_scores, row_src = [], [], []
+                for j in range(2 * K):
+                    tok = token_idx[b, j].item()
+                    src = b * K + beam_idx[b, j].item()
+                    seq = torch.cat([ids[src], token_idx[b, j:j + 1]])
+                    if eos_ids and tok in eos_ids:
+                        lp = (seq.shape[0] - prompt_len) ** gen.length_penalty
+                        finished[b].append((top_scores[b, j].item() / lp, seq))
+                    elif len(row_ids) < K:
+                        row_ids.append(seq)
+                        row_scores.append(top_scores[b, j])
+                        row_src.append(src)
+                while len(row_ids) < K:  # degenerate: pad with best
+                    row_ids.append(row_ids[-1])
+                    row_scores.append(row_scores[-1])
+                    row_src.append(row_src[-1])
+                new_ids.extend(row_ids)
+                new_scores.extend(row_scores)
+                new_beam_src.extend(row_src)
+
+            ids = torch.stack(new_ids)
+            beam_scores = torch.stack(new_scores)
+            src = torch.tensor(new_beam_src, device=device)
+            past = _reorder_cache(past, src)
+            cur = ids[:, -1:]
+            if all(len(f) >= K for f in finished):
+                break
+
+        results = []
+        for b in range(B):
+            cands = finished[b]
+            if not cands:
+                # no EOS reached: use live beams
+                for j in range(K):
+                    lp = (ids.shape[1] - prompt_len) ** gen.length_penalty
+                    cands.append((beam_scores[b * K + j].item() / lp, ids[b * K + j]))
+            cands.sort(key=lambda x: -x[0])
+            results.append(cands[0][1][prompt_len:])
+        maxlen = max(r.shape[0] for r in results)
+        pad_id = gen.pad_token_id or (eos_ids[0] if eos_ids else 0)
+        out = torch.full((B, maxlen), pad_id, dtype=torch.long, device=device)
+        for b, r in enumerate(results):
+            out[b, : r.shape[0]] = r
+        return out, None
+
+
+def _reorder_cache(past, beam_idx):
+    if past is None:
+        return None
+    return [
+        (k.index_select(0, beam_idx), v.index_select(0, beam_idx))
+        for k, v in past
+    ]

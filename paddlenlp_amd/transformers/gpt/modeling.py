@@ -13,6 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ... import ops
+from ...generation import GenerationConfig, GenerationMixin
 from ..model_utils import PretrainedModel
 from .configuration import GPTConfig
 
@@ -139,7 +140,7 @@ class GPTPretrainingCriterion(nn.Module):
         )
 
 
-class GPTForCausalLM(GPTPretrainedModel):
+class GPTForCausalLM(GPTPretrainedModel, GenerationMixin):
     _tied_weights_keys = ["lm_head.weight"]
 
     def __init__(self, config: GPTConfig):
@@ -147,6 +148,7 @@ class GPTForCausalLM(GPTPretrainedModel):
         self.gpt = GPTModel(config)
         self.lm_head = nn.Linear(config.hidden_size, config.vocab_size, bias=False)
         self.criterion = GPTPretrainingCriterion(config)
+        self.generation_config = GenerationConfig.from_model_config(config)
 
     def get_output_embeddings(self):
         return self.lm_head

@@ -28,6 +28,7 @@ from ...parallel.tensor_parallel import (
     parallel_matmul,
 )
 from ...parallel.topology import get_topology
+from ...generation import GenerationConfig, GenerationMixin
 from ..model_utils import PretrainedModel
 from .configuration import LlamaConfig
 
@@ -485,7 +486,7 @@ class LlamaPretrainingCriterion(nn.Module):
         return ops.cross_entropy(logits, labels, self.ignore_index, reduction="mean")
 
 
-class LlamaForCausalLM(LlamaPretrainedModel):
+class LlamaForCausalLM(LlamaPretrainedModel, GenerationMixin):
     _tied_weights_keys = []  # lm_head untied by default for llama
 
     def __init__(self, config: LlamaConfig):
@@ -500,6 +501,7 @@ class LlamaForCausalLM(LlamaPretrainedModel):
         else:
             self.lm_head = _Linear(config.hidden_size, config.vocab_size, bias=False)
         self.criterion = LlamaPretrainingCriterion(config)
+        self.generation_config = GenerationConfig.from_model_config(config)
 
     def get_output_embeddings(self):
         return self.lm_head
