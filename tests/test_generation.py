@@ -80,12 +80,18 @@ def test_generation_config_roundtrip(tmp_path):
 
 
 def test_eos_stopping(model):
-    """Force EOS as the argmax token by biasing the lm_head."""
+    """Zero the EOS row and add a huge bias via the weight so EOS wins
+    regardless of the hidden state's sign."""
     with torch.no_grad():
-        model.lm_head.weight[2] += 100.0
+        saved = model.lm_head.weight[2].clone()
+        model.lm_head.weight[2] = 0.0
+    # monkeypatch forward-time bias through a hook on lm_head
+    handle = model.lm_head.register_forward_hook(
+        lambda mod, inp, out: out.index_fill(-1, torch.tensor(2), 1000.0))
     ids = torch.randint(3, 128, (2, 4))
     out, _ = model.generate(ids, max_new_tokens=20)
+    handle.remove()
     with torch.no_grad():
-        model.lm_head.weight[2] -= 100.0
-    assert out.shape[1] < 20  # stopped early
+        model.lm_head.weight[2] = saved
+    assert out.shape[1] < 20  # stopped early at EOS
     assert (out[:, 0] == 2).all()
