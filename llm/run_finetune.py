@@ -1,0 +1,129 @@
+"""SFT / LoRA / prefix fine-tuning entry point.
+
+Reference behavior: llm/run_finetune.py main :77 — same trainer skeleton as
+pretrain with: load_dataset + chat-template tokenization (llm/utils/data.py),
+ZeroPadding packing (:385-412), FlashMask via attn_mask_startend_row_indices,
+PEFT wrapping (LoRA/prefix) before the Trainer.
+"""
+from __future__ import annotations
+
+import os
+import sys
+from dataclasses import dataclass, field
+from functools import partial
+from typing import Optional
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from paddlenlp_amd.data import DataCollatorWithPadding
+from paddlenlp_amd.datasets import ZeroPaddingMapDataset, load_dataset
+from paddlenlp_amd.peft import LoRAConfig, LoRAModel, PrefixConfig, PrefixModelForCausalLM
+from paddlenlp_amd.trainer import PdArgumentParser, Trainer, TrainingArguments
+from paddlenlp_amd.transformers import (
+    AutoConfig,
+    AutoModelForCausalLM,
+    AutoTokenizer,
+    LlmMetaConfig,
+)
+from paddlenlp_amd.utils.log import logger
+from utils.data import convert_example
+
+
+@dataclass
+class FinetuneArguments(TrainingArguments):
+    pass
+
+
+@dataclass
+class ModelArgument:
+    model_name_or_path: str = field(default=None)
+    lora: bool = field(default=False)
+    lora_rank: int = field(default=8)
+    lora_alpha: float = field(default=16.0)
+    lora_dropout: float = field(default=0.0)
+    rslora: bool = field(default=False)
+    prefix_tuning: bool = field(default=False)
+    num_prefix_tokens: int = field(default=16)
+    flash_mask: bool = field(default=False)
+
+
+@dataclass
+class DataArgument:
+    dataset_name_or_path: str = field(default=None)
+    max_length: int = field(default=2048)
+    zero_padding: bool = field(default=False)
+
+
+def main():
+    parser = PdArgumentParser((ModelArgument, DataArgument, FinetuneArguments))
+    model_args, data_args, training_args = parser.parse_json_file_and_cmd_lines()
+
+    config = AutoConfig.from_pretrained(model_args.model_name_or_path)
+    LlmMetaConfig.set_llm_config(config, training_args)
+    config.tensor_parallel_rank = training_args.topology.get_rank_in("mp")
+    config.dtype = "bfloat16" if training_args.bf16 else "float32"
+
+    tokenizer = AutoTokenizer.from_pretrained(model_args.model_name_or_path)
+    model = AutoModelForCausalLM.from_pretrained(model_args.model_name_or_path, config=config)
+    if training_args.bf16:
+        model = model.to(torch.bfloat16)
+
+    if model_args.flash_mask and not data_args.zero_padding:
+        raise ValueError("flash_mask requires zero_padding (reference run_finetune.py:182-184)")
+
+    if model_args.lora:
+        lora_config = LoRAConfig(
+            r=model_args.lora_rank, lora_alpha=model_args.lora_alpha,
+            lora_dropout=model_args.lora_dropout, rslora=model_args.rslora,
+        )
+        model = LoRAModel(model, lora_config)
+    elif model_args.prefix_tuning:
+        model = PrefixModelForCausalLM(
+            model, PrefixConfig(num_prefix_tokens=model_args.num_prefix_tokens))
+
+    train_ds = load_dataset(data_args.dataset_name_or_path, splits="train")
+    dev_ds = None
+    try:
+        dev_ds = load_dataset(data_args.dataset_name_or_path, splits="dev")
+    except (FileNotFoundError, TypeError):
+        pass
+
+    trans_fn = partial(convert_example, tokenizer=tokenizer, max_length=data_args.max_length)
+    train_ds = train_ds.map(trans_fn)
+    if dev_ds is not None:
+        dev_ds = dev_ds.map(trans_fn)
+
+    if data_args.zero_padding:
+        train_ds = ZeroPaddingMapDataset(train_ds, tokenizer=tokenizer,
+                                         max_length=data_args.max_length)
+        if dev_ds is not None:
+            dev_ds = ZeroPaddingMapDataset(dev_ds, tokenizer=tokenizer,
+                                           max_length=data_args.max_length)
+        collator = None  # already fixed-shape
+    else:
+        collator = DataCollatorWithPadding(tokenizer=tokenizer, max_length=data_args.max_length)
+
+    trainer = Trainer(
+        model=model,
+        args=training_args,
+        train_dataset=train_ds,
+        eval_dataset=dev_ds,
+        data_collator=collator,
+        tokenizer=tokenizer,
+    )
+    if training_args.do_train:
+        trainer.train(resume_from_checkpoint=training_args.resume_from_checkpoint)
+        if model_args.lora or model_args.prefix_tuning:
+            if training_args.process_index == 0:
+                model.save_pretrained(training_args.output_dir)
+                tokenizer.save_pretrained(training_args.output_dir)
+        else:
+            trainer.save_model()
+    if training_args.do_eval and dev_ds is not None:
+        trainer.evaluate()
+
+
+if __name__ == "__main__":
+    main()

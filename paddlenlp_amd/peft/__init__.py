@@ -1,0 +1,2 @@
+from .lora import LoRAConfig, LoRAModel  # noqa: F401
+from .prefix import PrefixConfig, PrefixModelForCausalLM  # noqa: F401

@@ -516,12 +516,15 @@ class LlamaForCausalLM(LlamaPretrainedModel, GenerationMixin):
         attn_mask=None,
         attention_mask=None,  # accepted and ignored when causal-only (parity arg)
         startend_row_indices=None,
+        attn_mask_startend_row_indices=None,  # reference kw name (FlashMask)
         past_key_values=None,
         use_cache=False,
         inputs_embeds=None,
         position_ids=None,  # parity arg; rope uses position_offset from cache
         **kwargs,
     ):
+        if startend_row_indices is None:
+            startend_row_indices = attn_mask_startend_row_indices
         out = self.llama(
             input_ids=input_ids,
             attn_mask=attn_mask,
