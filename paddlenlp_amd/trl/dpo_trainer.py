@@ -1,0 +1,108 @@
+"""DPO trainer: concatenated chosen/rejected forward with a frozen reference.
+
+Reference behavior: paddlenlp/trl/dpo_trainer.py (565 LoC) — concatenated
+forward over packed sequences, per-sequence logprob extraction, frozen
+reference model (or reference-free loss types).
+"""
+from __future__ import annotations
+
+import copy
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+
+from ..trainer.trainer import Trainer
+from ..utils.log import logger
+from .dpo_criterion import DPOCriterion
+
+REFERENCE_FREE_LOSSES = {"simpo", "orpo"}
+
+
+def sequence_logprob(logits: torch.Tensor, labels: torch.Tensor,
+                     average: bool = False) -> torch.Tensor:
+    """Sum (or mean) log p(label) over non-masked positions.  [B, S, V], [B, S]."""
+    mask = labels != -100
+    safe = labels.clamp(min=0)
+    logps = logits.float().log_softmax(-1)
+    token_lp = logps.gather(-1, safe.unsqueeze(-1)).squeeze(-1)
+    token_lp = token_lp * mask
+    total = token_lp.sum(-1)
+    if average:
+        return total / mask.sum(-1).clamp(min=1)
+    return total
+
+
+class DPOTrainer(Trainer):
+    def __init__(self, *args, beta: float = 0.1, loss_type: str = "sigmoid",
+                 label_smoothing: float = 0.0, reference_model=None, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.dpo_criterion = DPOCriterion(
+            beta=beta, label_smoothing=label_smoothing, loss_type=loss_type)
+        self.loss_type = loss_type
+        self.average_logps = loss_type in ("ipo", "simpo")
+        self.reference_model = None
+        if loss_type not in REFERENCE_FREE_LOSSES:
+            if reference_model is None:
+                logger.info("DPO: cloning the policy as the frozen reference model")
+                reference_model = copy.deepcopy(self.model)
+            for p in reference_model.parameters():
+                p.requires_grad_(False)
+            reference_model.eval()
+            self.reference_model = reference_model
+
+    def _wrap_model(self, model):
+        model = super()._wrap_model(model)
+        if self.reference_model is not None:
+            self.reference_model = self.reference_model.to(self.args.device)
+            if self.args.bf16:
+                self.reference_model = self.reference_model.to(torch.bfloat16)
+        return model
+
+    def compute_loss(self, model, inputs, return_outputs=False):
+        """inputs: chosen_input_ids/chosen_labels/rejected_input_ids/rejected_labels
+        — concatenated into one forward (reference concatenated_forward)."""
+        B = inputs["chosen_input_ids"].shape[0]
+        maxlen = max(inputs["chosen_input_ids"].shape[1], inputs["rejected_input_ids"].shape[1])
+
+        def pad(t, fill):
+            if t.shape[1] == maxlen:
+                return t
+            pad_t = t.new_full((t.shape[0], maxlen - t.shape[1]), fill)
+            return torch.cat([t, pad_t], dim=1)
+
+        ids = torch.cat([
+            pad(inputs["chosen_input_ids"], 0), pad(inputs["rejected_input_ids"], 0)
+        ], dim=0)
+        labels = torch.cat([
+            pad(inputs["chosen_labels"], -100), pad(inputs["rejected_labels"], -100)
+        ], dim=0)
+
+        logits = model(input_ids=ids)
+        if isinstance(logits, tuple):
+            logits = logits[0]
+        logps = sequence_logprob(logits, labels, average=self.average_logps)
+        policy_chosen, policy_rejected = logps[:B], logps[B:]
+
+        ref_chosen = ref_rejected = None
+        if self.reference_model is not None:
+            with torch.no_grad():
+                ref_logits = self.reference_model(input_ids=ids)
+                if isinstance(ref_logits, tuple):
+                    ref_logits = ref_logits[0]
+                ref_logps = sequence_logprob(ref_logits, labels, average=self.average_logps)
+                ref_chosen, ref_rejected = ref_logps[:B], ref_logps[B:]
+
+        loss, chosen_r, rejected_r = self.dpo_criterion(
+            policy_chosen, policy_rejected, ref_chosen, ref_rejected)
+        if self.loss_type == "orpo":
+            # orpo adds the chosen NLL
+            nll = -sequence_logprob(logits[:B], labels[:B], average=True).mean()
+            loss = loss + nll
+        if return_outputs:
+            return loss, {
+                "rewards/chosen": chosen_r.mean(),
+                "rewards/rejected": rejected_r.mean(),
+                "rewards/accuracy": (chosen_r > rejected_r).float().mean(),
+            }
+        return loss
