@@ -14,6 +14,7 @@ void launch_ce_fwd(const void*, const long long*, float*, float*, long long, int
 void launch_ce_bwd(const float*, const void*, const long long*, const float*, void*, long long, int, long long, hipStream_t);
 void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_flash_fwd(const void*, const void*, const void*, void*, float*, int, int, int, int, int, int, float, bool, hipStream_t);
+void launch_flash_fwd_variant(const void*, const void*, const void*, void*, float*, int, int, int, int, int, int, float, bool, int, hipStream_t);
 void launch_flash_bwd(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, int, int, int, int, int, int, float, bool, hipStream_t);
 
 struct AdamWChunk {
@@ -153,6 +154,19 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
     return C;
 }
 
+std::vector<torch::Tensor> flash_attn_fwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                             bool causal, int64_t variant) {
+    int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Skv = k.size(1), Hk = k.size(2);
+    auto o = torch::empty_like(q);
+    auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_flash_fwd_variant(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                             lse.data_ptr<float>(), B, Sq, Skv, Hq, Hk, D, scale,
+                             causal, (int)variant, cur_stream());
+    return {o, lse};
+}
+
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                           bool causal) {
     CHECK_GPU(q); CHECK_CONTIG(q); CHECK_BF16(q);
@@ -251,6 +265,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("cross_entropy_bwd", &cross_entropy_bwd);
     m.def("mfma_probe", &mfma_probe);
     m.def("flash_attn_fwd", &flash_attn_fwd);
+    m.def("flash_attn_fwd_ex", &flash_attn_fwd_ex);
     m.def("flash_attn_bwd", &flash_attn_bwd);
     m.def("fused_adamw", &fused_adamw);
 }

@@ -38,6 +38,28 @@ __device__ __forceinline__ frag_c mfma16(frag_ab a, frag_ab b, frag_c c) {
 }
 
 // ---------------------------------------------------------------------------
+// LDS XOR swizzle (guide §6 G4 / T2): a row-major bf16 tile read as B-frags
+// has 16 lanes x 4 row-groups hitting (4r + 4g) mod 32 banks = 8-way
+// conflicts (measured 1.8e8 conflicts/dispatch on the unswizzled kernels).
+// XOR-ing the byte offset with (row & 7) << 4 spreads each column across 8
+// distinct 16-byte slots.  Writes and reads must apply the SAME swizzle;
+// the XOR is a multiple of 16 so 4/8/16-byte access alignment is preserved.
+// ---------------------------------------------------------------------------
+template <int LD>
+__device__ __forceinline__ char* swz(ushort_t (*tile)[LD], int row, int col_elem) {
+    return reinterpret_cast<char*>(&tile[row][0]) + ((col_elem * 2) ^ ((row & 7) << 4));
+}
+
+// compile-time on/off variant for within-probe A/B (guide rule #13)
+template <bool SWZ, int LD>
+__device__ __forceinline__ char* swzb(ushort_t (*tile)[LD], int row, int col_elem) {
+    if constexpr (SWZ)
+        return reinterpret_cast<char*>(&tile[row][0]) + ((col_elem * 2) ^ ((row & 7) << 4));
+    else
+        return reinterpret_cast<char*>(&tile[row][0]) + col_elem * 2;
+}
+
+// ---------------------------------------------------------------------------
 // layout probe: one wave computes C = A @ B for 16x32 @ 32x16.  Used by the
 // GPU layout-validation test with asymmetric random inputs (transpose-
 // detecting, guide G9).
@@ -60,10 +82,7 @@ __global__ void mfma_layout_probe(const ushort_t* A, const ushort_t* B, float* C
 // forward.  Each wave owns MF x 16 q rows (MF M-frags), so one K/V staging
 // round feeds 2x the MFMA work (BLK_M = FA_WAVES * MF * 16 = 128).
 // ---------------------------------------------------------------------------
-#define FA_MF 2
-#define FWD_BLK_M (FA_WAVES * FA_MF * 16)
-
-template <int D>
+template <int D, int MF, bool SWZ>
 __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const ushort_t* __restrict__ q,   // [B, Sq, Hq, D]
     const ushort_t* __restrict__ k,   // [B, Skv, Hk, D]
@@ -71,6 +90,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     ushort_t* __restrict__ o,         // [B, Sq, Hq, D]
     float* __restrict__ lse,          // [B, Hq, Sq]
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
+    constexpr int FWD_BLKM = FA_WAVES * MF * 16;
     constexpr int KD = D / 32;   // MFMA K-steps over the head dim
     constexpr int ND = D / 16;   // d-frags of the O accumulator
     constexpr int NN = BLK_N / 16;
@@ -79,7 +99,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
 
     __shared__ ushort_t k_lds[BLK_N][LDK];       // K row-major [kv][d]
     __shared__ ushort_t vt_lds[D][LDT];          // V transposed [d][kv]
-    __shared__ ushort_t p_lds[FWD_BLK_M][LDT];   // P row-major [q][kv]
+    __shared__ ushort_t p_lds[FWD_BLKM][LDT];   // P row-major [q][kv]
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -91,7 +111,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const int bh = blockIdx.y;
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
-    const int q_base = qt * FWD_BLK_M;
+    const int q_base = qt * FWD_BLKM;
     const int causal_off = Skv - Sq;  // kv visible iff kv <= q + off
 
     const long long q_row_stride = (long long)Hq * D;
@@ -101,10 +121,10 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
 
     // Q A-frags in registers (MF x 16 rows x D per wave)
-    frag_ab aq[FA_MF][KD];
+    frag_ab aq[MF][KD];
 #pragma unroll
-    for (int mf = 0; mf < FA_MF; mf++) {
-        int qrow = q_base + (wave * FA_MF + mf) * 16 + l16;
+    for (int mf = 0; mf < MF; mf++) {
+        int qrow = q_base + (wave * MF + mf) * 16 + l16;
         if (qrow < Sq) {
 #pragma unroll
             for (int kk = 0; kk < KD; kk++)
@@ -116,20 +136,20 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
         }
     }
 
-    float m_run[FA_MF][4], l_run[FA_MF][4];
+    float m_run[MF][4], l_run[MF][4];
 #pragma unroll
-    for (int mf = 0; mf < FA_MF; mf++)
+    for (int mf = 0; mf < MF; mf++)
 #pragma unroll
         for (int r = 0; r < 4; r++) { m_run[mf][r] = -INFINITY; l_run[mf][r] = 0.f; }
-    frag_c acc_o[FA_MF][ND];
+    frag_c acc_o[MF][ND];
 #pragma unroll
-    for (int mf = 0; mf < FA_MF; mf++)
+    for (int mf = 0; mf < MF; mf++)
 #pragma unroll
         for (int n = 0; n < ND; n++) acc_o[mf][n] = frag_c{0.f, 0.f, 0.f, 0.f};
 
     int n_kv_tiles = (Skv + BLK_N - 1) / BLK_N;
     if (causal) {
-        int max_kv = q_base + FWD_BLK_M - 1 + causal_off;
+        int max_kv = q_base + FWD_BLKM - 1 + causal_off;
         int lim = (max_kv + BLK_N) / BLK_N;
         n_kv_tiles = min(n_kv_tiles, max(lim, 0));
     }
@@ -152,13 +172,13 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
                 k8b = *reinterpret_cast<const short8v*>(k_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
                 v8b = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(&k_lds[row0][col]) = k8a;
-            *reinterpret_cast<short8v*>(&k_lds[row0 + 1][col]) = k8b;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(k_lds, row0, col)) = k8a;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(k_lds, row0 + 1, col)) = k8b;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
                 unsigned int packed = ((unsigned int)(unsigned short)v8a[j]) |
                                       (((unsigned int)(unsigned short)v8b[j]) << 16);
-                *reinterpret_cast<unsigned int*>(&vt_lds[col + j][row0]) = packed;
+                *reinterpret_cast<unsigned int*>(swzb<SWZ>(vt_lds, col + j, row0)) = packed;
             }
         }
         __syncthreads();
@@ -166,19 +186,18 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
         // S = Q K^T (contract d; B-frag from K row-major), then softmax with
         // P written in place into the S accumulator, then P -> LDS
 #pragma unroll
-        for (int mf = 0; mf < FA_MF; mf++) {
+        for (int mf = 0; mf < MF; mf++) {
             frag_c acc_s[NN];
 #pragma unroll
             for (int n = 0; n < NN; n++) {
                 acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                 for (int kk = 0; kk < KD; kk++) {
-                    frag_ab bk = *reinterpret_cast<const frag_ab*>(
-                        &k_lds[n * 16 + l16][kk * 32 + lk8]);
+                    frag_ab bk = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(k_lds, n * 16 + l16, kk * 32 + lk8));
                     acc_s[n] = mfma16(aq[mf][kk], bk, acc_s[n]);
                 }
             }
-            const int qrow0 = q_base + (wave * FA_MF + mf) * 16 + (lane >> 4) * 4;
+            const int qrow0 = q_base + (wave * MF + mf) * 16 + (lane >> 4) * 4;
 #pragma unroll
             for (int n = 0; n < NN; n++) {
                 int kvg = kv_base + n * 16 + l16;
@@ -226,7 +245,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
             for (int n = 0; n < NN; n++)
 #pragma unroll
                 for (int r = 0; r < 4; r++)
-                    p_lds[(wave * FA_MF + mf) * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                    *reinterpret_cast<ushort_t*>(swzb<SWZ>(p_lds, (wave * MF + mf) * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
                         f32_to_bf16(acc_s[n][r]);
         }
         __syncthreads();
@@ -235,13 +254,11 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
 #pragma unroll
         for (int kk = 0; kk < BLK_N / 32; kk++) {
 #pragma unroll
-            for (int mf = 0; mf < FA_MF; mf++) {
-                frag_ab ap = *reinterpret_cast<const frag_ab*>(
-                    &p_lds[(wave * FA_MF + mf) * 16 + l16][kk * 32 + lk8]);
+            for (int mf = 0; mf < MF; mf++) {
+                frag_ab ap = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(p_lds, (wave * MF + mf) * 16 + l16, kk * 32 + lk8));
 #pragma unroll
                 for (int n = 0; n < ND; n++) {
-                    frag_ab bv = *reinterpret_cast<const frag_ab*>(
-                        &vt_lds[n * 16 + l16][kk * 32 + lk8]);
+                    frag_ab bv = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(vt_lds, n * 16 + l16, kk * 32 + lk8));
                     acc_o[mf][n] = mfma16(ap, bv, acc_o[mf][n]);
                 }
             }
@@ -251,8 +268,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
 
     // epilogue: O/l and LSE
 #pragma unroll
-    for (int mf = 0; mf < FA_MF; mf++) {
-        const int qrow0 = q_base + (wave * FA_MF + mf) * 16 + (lane >> 4) * 4;
+    for (int mf = 0; mf < MF; mf++) {
+        const int qrow0 = q_base + (wave * MF + mf) * 16 + (lane >> 4) * 4;
 #pragma unroll
         for (int r = 0; r < 4; r++) {
             int qg = qrow0 + r;
@@ -396,15 +413,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
                 k8b = *reinterpret_cast<const short8v*>(k_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
                 v8b = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(&k_lds[row0][col]) = k8a;
-            *reinterpret_cast<short8v*>(&k_lds[row0 + 1][col]) = k8b;
-            *reinterpret_cast<short8v*>(&v_lds[row0][col]) = v8a;
-            *reinterpret_cast<short8v*>(&v_lds[row0 + 1][col]) = v8b;
+            *reinterpret_cast<short8v*>(swz(k_lds, row0, col)) = k8a;
+            *reinterpret_cast<short8v*>(swz(k_lds, row0 + 1, col)) = k8b;
+            *reinterpret_cast<short8v*>(swz(v_lds, row0, col)) = v8a;
+            *reinterpret_cast<short8v*>(swz(v_lds, row0 + 1, col)) = v8b;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
                 unsigned int packed = ((unsigned int)(unsigned short)k8a[j]) |
                                       (((unsigned int)(unsigned short)k8b[j]) << 16);
-                *reinterpret_cast<unsigned int*>(&kt_lds[col + j][row0]) = packed;
+                *reinterpret_cast<unsigned int*>(swz(kt_lds, col + j, row0)) = packed;
             }
         }
         __syncthreads();
@@ -417,10 +434,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
             acc_dp[n] = frag_c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kk = 0; kk < KD; kk++) {
-                frag_ab bk = *reinterpret_cast<const frag_ab*>(
-                    &k_lds[n * 16 + l16][kk * 32 + lk8]);
-                frag_ab bv = *reinterpret_cast<const frag_ab*>(
-                    &v_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bk = *reinterpret_cast<const frag_ab*>(swz(k_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bv = *reinterpret_cast<const frag_ab*>(swz(v_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_s[n] = mfma16(aq[kk], bk, acc_s[n]);
                 acc_dp[n] = mfma16(ado[kk], bv, acc_dp[n]);
             }
@@ -437,7 +452,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
                 if (causal) vis = vis && (kvg <= qg + causal_off);
                 float p = vis ? __expf(acc_s[n][r] * scale - lse_r[r]) : 0.f;
                 float ds = p * (acc_dp[n][r] - dl_r[r]) * scale;
-                ds_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] = f32_to_bf16(ds);
+                *reinterpret_cast<ushort_t*>(swz(ds_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) = f32_to_bf16(ds);
             }
         }
         __syncthreads();
@@ -445,12 +460,10 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
         // dQ += dS K  (contract kv; B from kt_lds)
 #pragma unroll
         for (int kk = 0; kk < BLK_N / 32; kk++) {
-            frag_ab ads = *reinterpret_cast<const frag_ab*>(
-                &ds_lds[wave * 16 + l16][kk * 32 + lk8]);
+            frag_ab ads = *reinterpret_cast<const frag_ab*>(swz(ds_lds, wave * 16 + l16, kk * 32 + lk8));
 #pragma unroll
             for (int n = 0; n < ND; n++) {
-                frag_ab bkt = *reinterpret_cast<const frag_ab*>(
-                    &kt_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bkt = *reinterpret_cast<const frag_ab*>(swz(kt_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_dq[n] = mfma16(ads, bkt, acc_dq[n]);
             }
         }
@@ -569,18 +582,18 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
                 q8b = *reinterpret_cast<const short8v*>(q_ptr + (long long)(qg0 + 1) * q_row_stride + col);
                 d8b = *reinterpret_cast<const short8v*>(do_ptr + (long long)(qg0 + 1) * q_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(&q_lds[row0][col]) = q8a;
-            *reinterpret_cast<short8v*>(&q_lds[row0 + 1][col]) = q8b;
-            *reinterpret_cast<short8v*>(&do_lds[row0][col]) = d8a;
-            *reinterpret_cast<short8v*>(&do_lds[row0 + 1][col]) = d8b;
+            *reinterpret_cast<short8v*>(swz(q_lds, row0, col)) = q8a;
+            *reinterpret_cast<short8v*>(swz(q_lds, row0 + 1, col)) = q8b;
+            *reinterpret_cast<short8v*>(swz(do_lds, row0, col)) = d8a;
+            *reinterpret_cast<short8v*>(swz(do_lds, row0 + 1, col)) = d8b;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
                 unsigned int pq = ((unsigned int)(unsigned short)q8a[j]) |
                                   (((unsigned int)(unsigned short)q8b[j]) << 16);
                 unsigned int pd = ((unsigned int)(unsigned short)d8a[j]) |
                                   (((unsigned int)(unsigned short)d8b[j]) << 16);
-                *reinterpret_cast<unsigned int*>(&qt_lds[col + j][row0]) = pq;
-                *reinterpret_cast<unsigned int*>(&dot_lds[col + j][row0]) = pd;
+                *reinterpret_cast<unsigned int*>(swz(qt_lds, col + j, row0)) = pq;
+                *reinterpret_cast<unsigned int*>(swz(dot_lds, col + j, row0)) = pd;
             }
         }
         __syncthreads();
@@ -593,10 +606,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
             acc_dp[n] = frag_c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kk = 0; kk < KD; kk++) {
-                frag_ab bq = *reinterpret_cast<const frag_ab*>(
-                    &q_lds[n * 16 + l16][kk * 32 + lk8]);
-                frag_ab bdo = *reinterpret_cast<const frag_ab*>(
-                    &do_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bq = *reinterpret_cast<const frag_ab*>(swz(q_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bdo = *reinterpret_cast<const frag_ab*>(swz(do_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_s[n] = mfma16(ak[kk], bq, acc_s[n]);
                 acc_dp[n] = mfma16(av[kk], bdo, acc_dp[n]);
             }
@@ -626,17 +637,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
         for (int n = 0; n < NN; n++)
 #pragma unroll
             for (int r = 0; r < 4; r++)
-                pt_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                *reinterpret_cast<ushort_t*>(swz(pt_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
                     f32_to_bf16(pt_vals[n][r]);
         __syncthreads();
 #pragma unroll
         for (int kk = 0; kk < BLK_M / 32; kk++) {
-            frag_ab apt = *reinterpret_cast<const frag_ab*>(
-                &pt_lds[wave * 16 + l16][kk * 32 + lk8]);
+            frag_ab apt = *reinterpret_cast<const frag_ab*>(swz(pt_lds, wave * 16 + l16, kk * 32 + lk8));
 #pragma unroll
             for (int n = 0; n < ND; n++) {
-                frag_ab bdot = *reinterpret_cast<const frag_ab*>(
-                    &dot_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bdot = *reinterpret_cast<const frag_ab*>(swz(dot_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_dv[n] = mfma16(apt, bdot, acc_dv[n]);
             }
         }
@@ -647,17 +656,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
         for (int n = 0; n < NN; n++)
 #pragma unroll
             for (int r = 0; r < 4; r++)
-                pt_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                *reinterpret_cast<ushort_t*>(swz(pt_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
                     f32_to_bf16(dst_vals[n][r]);
         __syncthreads();
 #pragma unroll
         for (int kk = 0; kk < BLK_M / 32; kk++) {
-            frag_ab adst = *reinterpret_cast<const frag_ab*>(
-                &pt_lds[wave * 16 + l16][kk * 32 + lk8]);
+            frag_ab adst = *reinterpret_cast<const frag_ab*>(swz(pt_lds, wave * 16 + l16, kk * 32 + lk8));
 #pragma unroll
             for (int n = 0; n < ND; n++) {
-                frag_ab bqt = *reinterpret_cast<const frag_ab*>(
-                    &qt_lds[n * 16 + l16][kk * 32 + lk8]);
+                frag_ab bqt = *reinterpret_cast<const frag_ab*>(swz(qt_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_dk[n] = mfma16(adst, bqt, acc_dk[n]);
             }
         }
@@ -688,22 +695,39 @@ void launch_mfma_probe(const void* A, const void* B, float* C, hipStream_t strea
                        (const ushort_t*)A, (const ushort_t*)B, C);
 }
 
-template <int D>
+template <int D, int MF, bool SWZ>
 static void flash_fwd_t(const void* q, const void* k, const void* v, void* o,
                         float* lse, int B, int Sq, int Skv, int Hq, int Hk,
                         float scale, bool causal, hipStream_t stream) {
-    dim3 grid((Sq + FWD_BLK_M - 1) / FWD_BLK_M, B * Hq);
-    hipLaunchKernelGGL(flash_fwd_kernel<D>, grid, dim3(FA_BLOCK), 0, stream,
+    constexpr int BLKM = FA_WAVES * MF * 16;
+    dim3 grid((Sq + BLKM - 1) / BLKM, B * Hq);
+    hipLaunchKernelGGL((flash_fwd_kernel<D, MF, SWZ>), grid, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (ushort_t*)o, lse, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+}
+
+// variant dispatch for A/B benchmarking (variant: 0=MF1+swz [default],
+// 1=MF2+swz, 2=MF1 linear, 3=MF2 linear)
+void launch_flash_fwd_variant(const void* q, const void* k, const void* v, void* o,
+                              float* lse, int B, int Sq, int Skv, int Hq, int Hk,
+                              int D, float scale, bool causal, int variant,
+                              hipStream_t stream) {
+    if (D != 128) variant = 0;
+    switch (variant) {
+        case 1: flash_fwd_t<128, 2, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
+        case 2: flash_fwd_t<128, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
+        case 3: flash_fwd_t<128, 2, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
+        default:
+            if (D == 128) flash_fwd_t<128, 1, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+            else if (D == 64) flash_fwd_t<64, 1, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+            else if (D == 32) flash_fwd_t<32, 1, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    }
 }
 
 void launch_flash_fwd(const void* q, const void* k, const void* v, void* o,
                       float* lse, int B, int Sq, int Skv, int Hq, int Hk, int D,
                       float scale, bool causal, hipStream_t stream) {
-    if (D == 128) flash_fwd_t<128>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
-    else if (D == 64) flash_fwd_t<64>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
-    else if (D == 32) flash_fwd_t<32>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    launch_flash_fwd_variant(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, D, scale, causal, 0, stream);
 }
 
 template <int D>

@@ -1,0 +1,79 @@
+"""Within-probe A/B microbenchmark for the flash-attention kernels.
+
+Guide rule #13: cross-run/cross-box deltas under 10% are noise — variants
+must be timed interleaved in ONE process on ONE box.
+"""
+import argparse
+import sys
+import os
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from paddlenlp_amd.ops.functional import _load_extension
+
+VARIANTS = {0: "MF1+swz", 1: "MF2+swz", 2: "MF1+linear", 3: "MF2+linear"}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--B", type=int, default=8)
+    p.add_argument("--S", type=int, default=4096)
+    p.add_argument("--Hq", type=int, default=32)
+    p.add_argument("--Hk", type=int, default=8)
+    p.add_argument("--D", type=int, default=128)
+    p.add_argument("--reps", type=int, default=10)
+    args = p.parse_args()
+
+    C = _load_extension()
+    torch.manual_seed(0)
+    dev = "cuda:0"
+    q = torch.randn(args.B, args.S, args.Hq, args.D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(args.B, args.S, args.Hk, args.D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(args.B, args.S, args.Hk, args.D, device=dev, dtype=torch.bfloat16)
+
+    # causal attention matmul FLOPs (0.5 visibility): 2 matmuls x 2 flops
+    flops = 0.5 * 2 * 2 * args.B * args.Hq * args.S * args.S * args.D
+
+    # correctness cross-check between variants first
+    ref_o, ref_lse = C.flash_attn_fwd_ex(q, k, v, True, 0)
+    for var in (1, 2, 3):
+        o, lse = C.flash_attn_fwd_ex(q, k, v, True, var)
+        err = (o.float() - ref_o.float()).abs().max().item()
+        assert err < 1e-2, (var, err)
+
+    results = {v: [] for v in VARIANTS}
+    for rep in range(args.reps):
+        for var in VARIANTS:  # interleaved
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            C.flash_attn_fwd_ex(q, k, v, True, var)
+            torch.cuda.synchronize()
+            results[var].append(time.perf_counter() - t0)
+
+    print(f"shape B{args.B} S{args.S} Hq{args.Hq} Hk{args.Hk} D{args.D}, {args.reps} reps:")
+    for var, name in VARIANTS.items():
+        ts = sorted(results[var])[1:-1] or results[var]  # trim outliers
+        mean = sum(ts) / len(ts)
+        print(f"  fwd variant {var} ({name:<11}): {mean*1e3:8.2f} ms  {flops/mean/1e12:7.1f} TF/s")
+
+    # backward timing (current build's single variant)
+    o, lse = C.flash_attn_fwd_ex(q, k, v, True, 0)
+    do = torch.randn_like(o)
+    torch.cuda.synchronize()
+    ts = []
+    for _ in range(args.reps):
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        C.flash_attn_bwd(do, q, k, v, o, lse, True)
+        torch.cuda.synchronize()
+        ts.append(time.perf_counter() - t0)
+    mean = sum(sorted(ts)[1:-1] or ts) / max(1, len(ts) - 2)
+    bwd_flops = flops * 2.5  # dq(3 matmuls) + dkv(4 matmuls) vs fwd's 2 -> 3.5x total/1.4... report raw
+    print(f"  bwd (dq+dkv+delta)        : {mean*1e3:8.2f} ms  {bwd_flops/mean/1e12:7.1f} TF/s")
+
+
+if __name__ == "__main__":
+    main()
