@@ -1,0 +1,256 @@
+"""Inference predictor: dynamic-batching paged-KV serving loop.
+
+Reference behavior: llm/predict/predictor.py — PredictorArgument :54,
+DygraphPredictor :232 (vanilla model.generate), BlockInferencePredictorMixin
+:727 (block tables, free list, dynamic insert), create_predictor :1163,
+decode loop `while not_need_stop: _infer` :1003.
+
+MI355X design: the block predictor drives the FusedMultiTransformer engine
+(gfx950 paged decode kernels); sampling/penalties run as torch ops on-device;
+the block manager is host-side.
+"""
+from __future__ import annotations
+
+import os
+import sys
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)))))
+
+import torch
+
+from paddlenlp_amd.experimental.block_manager import BlockManager
+from paddlenlp_amd.experimental.fused_transformer import FusedMultiTransformer
+from paddlenlp_amd.generation import GenerationConfig
+from paddlenlp_amd.trainer import PdArgumentParser
+from paddlenlp_amd.transformers import AutoConfig, AutoModelForCausalLM, AutoTokenizer
+from paddlenlp_amd.utils.log import logger
+
+
+@dataclass
+class PredictorArgument:
+    model_name_or_path: str = field(default=None)
+    src_length: int = field(default=1024)
+    max_length: int = field(default=1024)
+    total_max_length: int = field(default=4096)
+    batch_size: int = field(default=8)
+    block_size: int = field(default=64)
+    block_attn: bool = field(default=True)
+    inference_model: bool = field(default=True)
+    dtype: str = field(default="bfloat16")
+    top_p: float = field(default=0.7)
+    temperature: float = field(default=0.95)
+    repetition_penalty: float = field(default=1.0)
+    decode_strategy: str = field(default="sampling")  # or "greedy"
+    mode: str = field(default="dynamic")
+    benchmark: bool = field(default=False)
+
+
+class BasePredictor:
+    def __init__(self, config: PredictorArgument, tokenizer=None):
+        self.config = config
+        self.tokenizer = tokenizer
+
+    def _preprocess(self, texts: List[str]):
+        out = self.tokenizer(list(texts), padding=True, return_tensors="pt",
+                             truncation=True, max_length=self.config.src_length)
+        return out
+
+
+class DygraphPredictor(BasePredictor):
+    """Vanilla model.generate path (reference DygraphPredictor :232)."""
+
+    def __init__(self, config, model=None, tokenizer=None):
+        super().__init__(config, tokenizer)
+        self.model = model
+        self.model.eval()
+
+    @torch.no_grad()
+    def predict(self, texts: List[str]) -> List[str]:
+        inputs = self._preprocess(texts)
+        device = next(self.model.parameters()).device
+        ids = inputs["input_ids"].to(device)
+        gen = GenerationConfig(
+            max_new_tokens=self.config.max_length,
+            do_sample=self.config.decode_strategy == "sampling",
+            top_p=self.config.top_p,
+            temperature=self.config.temperature,
+            repetition_penalty=self.config.repetition_penalty,
+            eos_token_id=self.tokenizer.eos_token_id,
+            pad_token_id=self.tokenizer.pad_token_id or 0,
+        )
+        out, _ = self.model.generate(ids, gen)
+        return self.tokenizer.batch_decode(out, skip_special_tokens=True)
+
+
+class BlockInferencePredictor(BasePredictor):
+    """Paged-KV dynamic-batching predictor over the fused engine."""
+
+    def __init__(self, config: PredictorArgument, engine: FusedMultiTransformer,
+                 tokenizer=None, device=None):
+        super().__init__(config, tokenizer)
+        self.engine = engine
+        self.device = device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+        c = engine.config
+        max_blocks_per_seq = (config.total_max_length + c.block_size - 1) // c.block_size
+        # enough blocks for a full batch at total_max_length (288 GB HBM: be generous)
+        num_blocks = max_blocks_per_seq * config.batch_size + 8
+        self.manager = BlockManager(
+            num_blocks, c.block_size, max_blocks_per_seq, config.batch_size)
+        engine.allocate_caches(num_blocks, self.device)
+        self.engine = engine.to(self.device)
+
+    def _sample(self, logits, prev_ids=None):
+        cfg = self.config
+        if cfg.repetition_penalty != 1.0 and prev_ids is not None and prev_ids.numel():
+            score = torch.gather(logits, 1, prev_ids)
+            score = torch.where(score < 0, score * cfg.repetition_penalty,
+                                score / cfg.repetition_penalty)
+            logits.scatter_(1, prev_ids, score)
+        if cfg.decode_strategy == "greedy":
+            return logits.argmax(-1)
+        logits = logits / max(cfg.temperature, 1e-6)
+        if cfg.top_p < 1.0:
+            sorted_logits, sorted_idx = torch.sort(logits, descending=True)
+            probs = sorted_logits.softmax(-1)
+            cum = probs.cumsum(-1)
+            remove = cum - probs > cfg.top_p
+            mask = remove.scatter(1, sorted_idx, remove)
+            logits = logits.masked_fill(mask, float("-inf"))
+        return torch.multinomial(logits.softmax(-1), 1).squeeze(-1)
+
+    @torch.no_grad()
+    def predict(self, texts: List[str]) -> List[str]:
+        """Dynamic batching: insert prompts while capacity allows, decode the
+        active batch, refill slots as sequences finish."""
+        cfg = self.config
+        tok = self.tokenizer
+        eos = tok.eos_token_id
+        pending = list(enumerate(texts))
+        results = {i: [] for i in range(len(texts))}
+        slot_to_req = {}
+        generated = {}
+        mgr = self.manager
+        prev_tokens = {}
+
+        def try_insert():
+            inserted = []
+            while pending:
+                req_id, text = pending[0]
+                ids = tok.encode(text)[-cfg.src_length:]
+                slot = mgr.allocate_slot(len(ids))
+                if slot is None:
+                    break
+                pending.pop(0)
+                slot_to_req[slot] = req_id
+                generated[slot] = 0
+                prev_tokens[slot] = list(ids)
+                inserted.append((slot, ids))
+            if inserted:
+                T = max(len(ids) for _, ids in inserted)
+                batch = torch.zeros(len(inserted), T, dtype=torch.long, device=self.device)
+                lens = torch.zeros(len(inserted), dtype=torch.int32, device=self.device)
+                bt = torch.empty(len(inserted), mgr.max_blocks_per_seq, dtype=torch.int32,
+                                 device=self.device)
+                for i, (slot, ids) in enumerate(inserted):
+                    batch[i, :len(ids)] = torch.tensor(ids, device=self.device)
+                    lens[i] = len(ids)
+                    bt[i] = mgr.block_table[slot].to(self.device)
+                logits = self.engine.prefill(batch, bt, lens)
+                prev = torch.tensor([prev_tokens[s] + [0] * (T - len(prev_tokens[s]))
+                                     for s, _ in inserted], device=self.device, dtype=torch.long)
+                tokens = self._sample(logits, prev)
+                for i, (slot, ids) in enumerate(inserted):
+                    t = int(tokens[i])
+                    self._commit_token(slot, t, results, slot_to_req, generated, prev_tokens, eos)
+
+        def _active_slots():
+            return sorted(slot_to_req.keys())
+
+        try_insert()
+        while slot_to_req or pending:
+            slots = _active_slots()
+            if not slots:
+                try_insert()
+                continue
+            B = len(slots)
+            input_ids = torch.tensor(
+                [[prev_tokens[s][-1]] for s in slots], dtype=torch.long, device=self.device)
+            lens_before = torch.tensor(
+                [int(mgr.seq_lens[s]) for s in slots], dtype=torch.int32, device=self.device)
+            # extend block tables for the incoming token
+            for s in slots:
+                if not mgr.extend(s, 1):
+                    mgr.preempt_longest()  # (v1: preempted request is dropped back to pending)
+            bt = torch.stack([mgr.block_table[s] for s in slots]).to(self.device, torch.int32)
+            logits = self.engine.decode_step(input_ids, bt, lens_before)
+            maxlen = max(len(prev_tokens[s]) for s in slots)
+            prev = torch.zeros(B, maxlen, dtype=torch.long, device=self.device)
+            for i, s in enumerate(slots):
+                prev[i, :len(prev_tokens[s])] = torch.tensor(
+                    prev_tokens[s], device=self.device)
+            tokens = self._sample(logits, prev)
+            finished_any = False
+            for i, s in enumerate(slots):
+                t = int(tokens[i])
+                done = self._commit_token(s, t, results, slot_to_req, generated, prev_tokens, eos)
+                finished_any = finished_any or done
+            if finished_any:
+                try_insert()
+
+        return [tok.decode(results[i], skip_special_tokens=True) for i in range(len(texts))]
+
+    def _commit_token(self, slot, token, results, slot_to_req, generated, prev_tokens, eos):
+        req = slot_to_req[slot]
+        generated[slot] += 1
+        is_eos = (eos is not None and token == eos)
+        if not is_eos:
+            results[req].append(token)
+            prev_tokens[slot].append(token)
+        if is_eos or generated[slot] >= self.config.max_length:
+            self.manager.release(slot)
+            del slot_to_req[slot]
+            del generated[slot]
+            del prev_tokens[slot]
+            return True
+        return False
+
+
+def create_predictor(predictor_args: PredictorArgument, model=None, tokenizer=None):
+    """Factory (reference create_predictor :1163)."""
+    if tokenizer is None:
+        tokenizer = AutoTokenizer.from_pretrained(predictor_args.model_name_or_path)
+    if model is None:
+        model = AutoModelForCausalLM.from_pretrained(
+            predictor_args.model_name_or_path,
+            dtype=predictor_args.dtype,
+        )
+        if torch.cuda.is_available():
+            model = model.to("cuda:0")
+    if predictor_args.inference_model and predictor_args.block_attn:
+        engine = FusedMultiTransformer.from_llama(
+            model, block_size=predictor_args.block_size,
+            max_seq_len=predictor_args.total_max_length)
+        device = next(model.parameters()).device
+        del model
+        return BlockInferencePredictor(predictor_args, engine, tokenizer, device=device)
+    return DygraphPredictor(predictor_args, model, tokenizer)
+
+
+def predict():
+    parser = PdArgumentParser((PredictorArgument,))
+    (args,) = parser.parse_json_file_and_cmd_lines()
+    predictor = create_predictor(args)
+    import json as _json
+
+    for line in sys.stdin:
+        line = line.strip()
+        if not line:
+            continue
+        outs = predictor.predict([line])
+        print(_json.dumps({"src": line, "output": outs[0]}, ensure_ascii=False))
+
+
+if __name__ == "__main__":
+    predict()

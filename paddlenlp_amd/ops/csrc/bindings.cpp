@@ -28,6 +28,8 @@ struct AdamWChunk {
     int is_bf16;
 };
 void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, float, float, hipStream_t);
+void launch_paged_decode_attn(const void*, const void*, const void*, const int*, const int*, void*, int, int, int, int, int, int, float, hipStream_t);
+void launch_rope_cache_append(const void*, void*, void*, void*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -255,6 +257,50 @@ void fused_adamw(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
                  (float)eps, (float)wd, bias1, bias2, cur_stream());
 }
 
+// ---------------------------------------------------------------------------
+// paged-KV inference ops
+// ---------------------------------------------------------------------------
+torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache,
+                                torch::Tensor block_table, torch::Tensor seq_lens) {
+    CHECK_GPU(q); CHECK_CONTIG(q); CHECK_BF16(q);
+    TORCH_CHECK(q.dim() == 3, "q must be [B, Hq, D] (single decode token)");
+    int B = q.size(0), Hq = q.size(1), D = q.size(2);
+    int block_size = k_cache.size(1), Hk = k_cache.size(2);
+    int max_blocks = block_table.size(1);
+    auto out = torch::empty_like(q);
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_paged_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                             block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                             out.data_ptr(), B, Hq, Hk, D, block_size, max_blocks,
+                             scale, cur_stream());
+    return out;
+}
+
+torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch::Tensor v_cache,
+                                torch::Tensor block_table, torch::Tensor seq_lens_before,
+                                torch::Tensor cos_t, torch::Tensor sin_t,
+                                int64_t Hq, int64_t Hk,
+                                c10::optional<torch::Tensor> token_counts) {
+    CHECK_GPU(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
+    TORCH_CHECK(qkv.dim() == 3, "qkv must be [B, T, (Hq+2Hk)*D]");
+    int B = qkv.size(0), T = qkv.size(1);
+    int D = k_cache.size(3);
+    int block_size = k_cache.size(1);
+    int max_blocks = block_table.size(1);
+    auto q_out = torch::empty({B, T, Hq, (long)D}, qkv.options());
+    auto cf = cos_t.to(torch::kFloat32).contiguous();
+    auto sf = sin_t.to(torch::kFloat32).contiguous();
+    const int* tc = nullptr;
+    if (token_counts.has_value()) tc = token_counts->data_ptr<int>();
+    launch_rope_cache_append(qkv.data_ptr(), q_out.data_ptr(),
+                             k_cache.data_ptr(), v_cache.data_ptr(),
+                             block_table.data_ptr<int>(), seq_lens_before.data_ptr<int>(),
+                             cf.data_ptr<float>(), sf.data_ptr<float>(),
+                             B, T, (int)Hq, (int)Hk, D, block_size, max_blocks, tc,
+                             cur_stream());
+    return q_out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rms_norm_fwd", &rms_norm_fwd);
     m.def("rms_norm_bwd", &rms_norm_bwd);
@@ -268,4 +314,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flash_attn_fwd_ex", &flash_attn_fwd_ex);
     m.def("flash_attn_bwd", &flash_attn_bwd);
     m.def("fused_adamw", &fused_adamw);
+    m.def("paged_decode_attn", &paged_decode_attn);
+    m.def("rope_cache_append", &rope_cache_append);
 }

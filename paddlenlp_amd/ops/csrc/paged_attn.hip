@@ -1,0 +1,258 @@
+// Paged-KV decode attention + fused RoPE cache writers — gfx950.
+//
+// MI355X-native replacement for the reference's append_attn family
+// (csrc/gpu/append_attention.cu + encoder/decoder_write_cache_with_rope,
+// SURVEY §2.9) re-designed for CDNA4 instead of ported: decode attention is
+// memory-bound, so the kernel is organized around GQA kv-reuse — one
+// workgroup per (batch, kv-head) computes ALL q-heads of the group while
+// streaming each K/V block from HBM exactly once, with vectorized short8
+// loads and fp32 online-softmax state in registers.
+//
+// Cache layout: [num_blocks, block_size, Hk, D] bf16 per layer.
+#include "common.h"
+
+#define PA_WAVES 4
+#define PA_BLOCK (PA_WAVES * 64)
+#define MAX_GQA 16  // max q-heads per kv head
+
+// ---------------------------------------------------------------------------
+// decode attention: q [B, Hq, D] (current token, post-RoPE),
+// block_table [B, max_blocks] int32, seq_lens [B] int32 (kv length incl.
+// the current token), out [B, Hq, D].
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
+    const ushort_t* __restrict__ q,
+    const ushort_t* __restrict__ k_cache,  // [nblocks, bs, Hk, D]
+    const ushort_t* __restrict__ v_cache,
+    const int* __restrict__ block_table,   // [B, max_blocks]
+    const int* __restrict__ seq_lens,      // [B]
+    ushort_t* __restrict__ out,
+    int B, int Hq, int Hk, int block_size, int max_blocks, float scale) {
+    const int b = blockIdx.x;
+    const int hk = blockIdx.y;
+    const int G = Hq / Hk;
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int seq_len = seq_lens[b];
+    if (seq_len <= 0) return;
+
+    constexpr int EPL = D / 64;  // elements per lane (2 for D=128)
+
+    // q for the G heads of this group -> registers (fp32), pre-scaled
+    float qreg[MAX_GQA][EPL];
+    const int hq0 = hk * G;
+#pragma unroll
+    for (int g = 0; g < MAX_GQA; g++) {
+        if (g >= G) break;
+        const ushort_t* qp = q + ((long long)b * Hq + hq0 + g) * D;
+#pragma unroll
+        for (int e = 0; e < EPL; e++)
+            qreg[g][e] = bf16_to_f32(qp[lane * EPL + e]) * scale;
+    }
+
+    float m[MAX_GQA], l[MAX_GQA], acc[MAX_GQA][EPL];
+#pragma unroll
+    for (int g = 0; g < MAX_GQA; g++) {
+        m[g] = -INFINITY;
+        l[g] = 0.f;
+#pragma unroll
+        for (int e = 0; e < EPL; e++) acc[g][e] = 0.f;
+    }
+
+    // waves split the kv range into contiguous chunks
+    const int per_wave = (seq_len + PA_WAVES - 1) / PA_WAVES;
+    const int j0 = wave * per_wave;
+    const int j1 = min(seq_len, j0 + per_wave);
+    const int* bt = block_table + (long long)b * max_blocks;
+
+    for (int j = j0; j < j1; j++) {
+        const int blk = bt[j / block_size];
+        const long long base =
+            (((long long)blk * block_size + (j % block_size)) * Hk + hk) * D;
+        // lane loads its EPL elements of K and V (coalesced short2/short4)
+        float kv_k[EPL], kv_v[EPL];
+        if (EPL == 2) {
+            short2v k2 = *reinterpret_cast<const short2v*>(k_cache + base + lane * 2);
+            short2v v2 = *reinterpret_cast<const short2v*>(v_cache + base + lane * 2);
+            kv_k[0] = bf16_to_f32((ushort_t)k2[0]); kv_k[1] = bf16_to_f32((ushort_t)k2[1]);
+            kv_v[0] = bf16_to_f32((ushort_t)v2[0]); kv_v[1] = bf16_to_f32((ushort_t)v2[1]);
+        } else {
+#pragma unroll
+            for (int e = 0; e < EPL; e++) {
+                kv_k[e] = bf16_to_f32(k_cache[base + lane * EPL + e]);
+                kv_v[e] = bf16_to_f32(v_cache[base + lane * EPL + e]);
+            }
+        }
+#pragma unroll
+        for (int g = 0; g < MAX_GQA; g++) {
+            if (g >= G) break;
+            float s = 0.f;
+#pragma unroll
+            for (int e = 0; e < EPL; e++) s += qreg[g][e] * kv_k[e];
+            s = wave_reduce_sum(s);  // full dot product, replicated in wave
+            float m_new = fmaxf(m[g], s);
+            float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
+            float p = __expf(s - m_new);
+            l[g] = l[g] * alpha + p;
+#pragma unroll
+            for (int e = 0; e < EPL; e++) acc[g][e] = acc[g][e] * alpha + p * kv_v[e];
+            m[g] = m_new;
+        }
+    }
+
+    // merge the 4 waves' partial (m, l, acc) through LDS
+    __shared__ float s_m[PA_WAVES][MAX_GQA];
+    __shared__ float s_l[PA_WAVES][MAX_GQA];
+    __shared__ float s_acc[PA_WAVES][MAX_GQA][64 * EPL];
+#pragma unroll
+    for (int g = 0; g < MAX_GQA; g++) {
+        if (g >= G) break;
+        if (lane == 0) { s_m[wave][g] = m[g]; s_l[wave][g] = l[g]; }
+#pragma unroll
+        for (int e = 0; e < EPL; e++) s_acc[wave][g][lane * EPL + e] = acc[g][e];
+    }
+    __syncthreads();
+    if (wave == 0) {
+#pragma unroll
+        for (int g = 0; g < MAX_GQA; g++) {
+            if (g >= G) break;
+            float gm = -INFINITY;
+            for (int w = 0; w < PA_WAVES; w++) gm = fmaxf(gm, s_m[w][g]);
+            float gl = 0.f;
+            float oacc[EPL] = {0.f};
+            for (int w = 0; w < PA_WAVES; w++) {
+                float mw = s_m[w][g];
+                float a = (mw == -INFINITY) ? 0.f : __expf(mw - gm);
+                gl += s_l[w][g] * a;
+#pragma unroll
+                for (int e = 0; e < EPL; e++)
+                    oacc[e] += s_acc[w][g][lane * EPL + e] * a;
+            }
+            float inv = (gl > 0.f) ? 1.0f / gl : 0.f;
+            ushort_t* op = out + ((long long)b * Hq + hq0 + g) * D;
+#pragma unroll
+            for (int e = 0; e < EPL; e++)
+                op[lane * EPL + e] = f32_to_bf16(oacc[e] * inv);
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// fused RoPE + paged cache append for a batch of new tokens.
+// qkv: [B, T, (Hq + 2*Hk) * D] (fused projection output; T tokens per seq
+// this step — 1 for decode, prompt length for prefill).
+// positions p = seq_lens_before[b] + t; writes roped q to q_out [B, T, Hq, D]
+// and roped k / raw v into the paged cache.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ void rope_cache_append_kernel(
+    const ushort_t* __restrict__ qkv,
+    ushort_t* __restrict__ q_out,
+    ushort_t* __restrict__ k_cache, ushort_t* __restrict__ v_cache,
+    const int* __restrict__ block_table, const int* __restrict__ seq_lens_before,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,  // [max_pos, D]
+    int B, int T, int Hq, int Hk, int block_size, int max_blocks,
+    const int* __restrict__ token_counts  // [B] tokens this step (<= T), or null
+) {
+    // grid: (B*T, Hq + 2*Hk); one wave (64 lanes) per head
+    const int bt_idx = blockIdx.x;
+    const int head = blockIdx.y;
+    const int b = bt_idx / T;
+    const int t = bt_idx % T;
+    const int n_tok = token_counts ? token_counts[b] : T;
+    if (t >= n_tok) return;
+    const int lane = threadIdx.x & 63;
+    const int H_all = Hq + 2 * Hk;
+    const int pos = seq_lens_before[b] + t;
+
+    const ushort_t* src = qkv + (((long long)b * T + t) * H_all + head) * D;
+    constexpr int half = D / 2;
+    constexpr int EPL = half / 64;  // rotation pairs per lane
+
+    bool is_q = head < Hq;
+    bool is_k = head >= Hq && head < Hq + Hk;
+
+    if (is_q || is_k) {
+        const float* cr = cos_t + (long long)pos * D;
+        const float* sr = sin_t + (long long)pos * D;
+        float o1[EPL > 0 ? EPL : 1], o2[EPL > 0 ? EPL : 1];
+#pragma unroll
+        for (int e = 0; e < EPL; e++) {
+            int i = lane * EPL + e;
+            float a = bf16_to_f32(src[i]);
+            float bb = bf16_to_f32(src[i + half]);
+            o1[e] = a * cr[i] - bb * sr[i];
+            o2[e] = bb * cr[i + half] + a * sr[i + half];
+        }
+        if (is_q) {
+            ushort_t* dst = q_out + (((long long)b * T + t) * Hq + head) * D;
+#pragma unroll
+            for (int e = 0; e < EPL; e++) {
+                int i = lane * EPL + e;
+                dst[i] = f32_to_bf16(o1[e]);
+                dst[i + half] = f32_to_bf16(o2[e]);
+            }
+        } else {
+            int hk = head - Hq;
+            int blk = block_table[(long long)b * max_blocks + pos / block_size];
+            ushort_t* dst = k_cache +
+                (((long long)blk * block_size + pos % block_size) * Hk + hk) * D;
+#pragma unroll
+            for (int e = 0; e < EPL; e++) {
+                int i = lane * EPL + e;
+                dst[i] = f32_to_bf16(o1[e]);
+                dst[i + half] = f32_to_bf16(o2[e]);
+            }
+        }
+    } else {
+        // V: raw copy into the cache
+        int hv = head - Hq - Hk;
+        int blk = block_table[(long long)b * max_blocks + pos / block_size];
+        ushort_t* dst = v_cache +
+            (((long long)blk * block_size + pos % block_size) * Hk + hv) * D;
+        constexpr int VE = D / 64;
+#pragma unroll
+        for (int e = 0; e < VE; e++) dst[lane * VE + e] = src[lane * VE + e];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+void launch_paged_decode_attn(const void* q, const void* k_cache, const void* v_cache,
+                              const int* block_table, const int* seq_lens, void* out,
+                              int B, int Hq, int Hk, int D, int block_size,
+                              int max_blocks, float scale, hipStream_t stream) {
+    dim3 grid(B, Hk);
+    if (D == 128)
+        hipLaunchKernelGGL(paged_decode_attn_kernel<128>, grid, dim3(PA_BLOCK), 0, stream,
+                           (const ushort_t*)q, (const ushort_t*)k_cache, (const ushort_t*)v_cache,
+                           block_table, seq_lens, (ushort_t*)out,
+                           B, Hq, Hk, block_size, max_blocks, scale);
+    else if (D == 64)
+        hipLaunchKernelGGL(paged_decode_attn_kernel<64>, grid, dim3(PA_BLOCK), 0, stream,
+                           (const ushort_t*)q, (const ushort_t*)k_cache, (const ushort_t*)v_cache,
+                           block_table, seq_lens, (ushort_t*)out,
+                           B, Hq, Hk, block_size, max_blocks, scale);
+}
+
+void launch_rope_cache_append(const void* qkv, void* q_out, void* k_cache, void* v_cache,
+                              const int* block_table, const int* seq_lens_before,
+                              const float* cos_t, const float* sin_t,
+                              int B, int T, int Hq, int Hk, int D, int block_size,
+                              int max_blocks, const int* token_counts, hipStream_t stream) {
+    dim3 grid(B * T, Hq + 2 * Hk);
+    if (D == 128)
+        hipLaunchKernelGGL(rope_cache_append_kernel<128>, grid, dim3(64), 0, stream,
+                           (const ushort_t*)qkv, (ushort_t*)q_out,
+                           (ushort_t*)k_cache, (ushort_t*)v_cache,
+                           block_table, seq_lens_before, cos_t, sin_t,
+                           B, T, Hq, Hk, block_size, max_blocks, token_counts);
+    else if (D == 64)
+        hipLaunchKernelGGL(rope_cache_append_kernel<64>, grid, dim3(64), 0, stream,
+                           (const ushort_t*)qkv, (ushort_t*)q_out,
+                           (ushort_t*)k_cache, (ushort_t*)v_cache,
+                           block_table, seq_lens_before, cos_t, sin_t,
+                           B, T, Hq, Hk, block_size, max_blocks, token_counts);
+}
