@@ -16,6 +16,7 @@ void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_flash_fwd(const void*, const void*, const void*, void*, float*, int, int, int, int, int, int, float, bool, hipStream_t);
 void launch_flash_fwd_variant(const void*, const void*, const void*, void*, float*, int, int, int, int, int, int, float, bool, int, hipStream_t);
 void launch_flash_bwd(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, int, int, int, int, int, int, float, bool, hipStream_t);
+void launch_flash_bwd_variant(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, int, int, int, int, int, int, float, bool, int, hipStream_t);
 
 struct AdamWChunk {
     void* param;
@@ -187,6 +188,30 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torc
     return {o, lse};
 }
 
+std::vector<torch::Tensor> flash_attn_bwd_ex(torch::Tensor dout, torch::Tensor q,
+                                             torch::Tensor k, torch::Tensor v,
+                                             torch::Tensor o, torch::Tensor lse,
+                                             bool causal, int64_t variant) {
+    int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Skv = k.size(1), Hk = k.size(2);
+    auto dq = torch::empty_like(q);
+    auto dk_h = torch::empty({B, Skv, Hq, D}, k.options());
+    auto dv_h = torch::empty({B, Skv, Hq, D}, v.options());
+    auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_flash_bwd_variant(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                             o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
+                             dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(),
+                             B, Sq, Skv, Hq, Hk, D, scale, causal, (int)variant, cur_stream());
+    torch::Tensor dk = dk_h, dv = dv_h;
+    if (Hq != Hk) {
+        int G = Hq / Hk;
+        dk = dk_h.view({B, Skv, Hk, G, D}).sum(3);
+        dv = dv_h.view({B, Skv, Hk, G, D}).sum(3);
+    }
+    return {dq, dk, dv};
+}
+
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor k, torch::Tensor v,
                                           torch::Tensor o, torch::Tensor lse,
@@ -313,6 +338,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flash_attn_fwd", &flash_attn_fwd);
     m.def("flash_attn_fwd_ex", &flash_attn_fwd_ex);
     m.def("flash_attn_bwd", &flash_attn_bwd);
+    m.def("flash_attn_bwd_ex", &flash_attn_bwd_ex);
     m.def("fused_adamw", &fused_adamw);
     m.def("paged_decode_attn", &paged_decode_attn);
     m.def("rope_cache_append", &rope_cache_append);

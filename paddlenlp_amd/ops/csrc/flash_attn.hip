@@ -248,7 +248,9 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
                     *reinterpret_cast<ushort_t*>(swzb<SWZ>(p_lds, (wave * MF + mf) * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
                         f32_to_bf16(acc_s[n][r]);
         }
-        __syncthreads();
+        // each wave reads back only its own p_lds rows: a wave-local LDS
+        // drain is enough (no cross-wave barrier needed here)
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
         // O += P V  (contract kv; A from p_lds, B from vt_lds)
 #pragma unroll
@@ -321,7 +323,7 @@ __global__ void flash_bwd_delta_kernel(
 // LDS: K row-major (S B-frag), V row-major (dP B-frag),
 //      K^T (dQ B-frag), dS staging.
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool SWZ>
 __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
@@ -413,15 +415,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
                 k8b = *reinterpret_cast<const short8v*>(k_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
                 v8b = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(swz(k_lds, row0, col)) = k8a;
-            *reinterpret_cast<short8v*>(swz(k_lds, row0 + 1, col)) = k8b;
-            *reinterpret_cast<short8v*>(swz(v_lds, row0, col)) = v8a;
-            *reinterpret_cast<short8v*>(swz(v_lds, row0 + 1, col)) = v8b;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(k_lds, row0, col)) = k8a;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(k_lds, row0 + 1, col)) = k8b;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(v_lds, row0, col)) = v8a;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(v_lds, row0 + 1, col)) = v8b;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
                 unsigned int packed = ((unsigned int)(unsigned short)k8a[j]) |
                                       (((unsigned int)(unsigned short)k8b[j]) << 16);
-                *reinterpret_cast<unsigned int*>(swz(kt_lds, col + j, row0)) = packed;
+                *reinterpret_cast<unsigned int*>(swzb<SWZ>(kt_lds, col + j, row0)) = packed;
             }
         }
         __syncthreads();
@@ -434,8 +436,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
             acc_dp[n] = frag_c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kk = 0; kk < KD; kk++) {
-                frag_ab bk = *reinterpret_cast<const frag_ab*>(swz(k_lds, n * 16 + l16, kk * 32 + lk8));
-                frag_ab bv = *reinterpret_cast<const frag_ab*>(swz(v_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bk = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(k_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bv = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(v_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_s[n] = mfma16(aq[kk], bk, acc_s[n]);
                 acc_dp[n] = mfma16(ado[kk], bv, acc_dp[n]);
             }
@@ -452,7 +454,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
                 if (causal) vis = vis && (kvg <= qg + causal_off);
                 float p = vis ? __expf(acc_s[n][r] * scale - lse_r[r]) : 0.f;
                 float ds = p * (acc_dp[n][r] - dl_r[r]) * scale;
-                *reinterpret_cast<ushort_t*>(swz(ds_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) = f32_to_bf16(ds);
+                *reinterpret_cast<ushort_t*>(swzb<SWZ>(ds_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) = f32_to_bf16(ds);
             }
         }
         __syncthreads();
@@ -460,10 +462,10 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
         // dQ += dS K  (contract kv; B from kt_lds)
 #pragma unroll
         for (int kk = 0; kk < BLK_N / 32; kk++) {
-            frag_ab ads = *reinterpret_cast<const frag_ab*>(swz(ds_lds, wave * 16 + l16, kk * 32 + lk8));
+            frag_ab ads = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(ds_lds, wave * 16 + l16, kk * 32 + lk8));
 #pragma unroll
             for (int n = 0; n < ND; n++) {
-                frag_ab bkt = *reinterpret_cast<const frag_ab*>(swz(kt_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bkt = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(kt_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_dq[n] = mfma16(ads, bkt, acc_dq[n]);
             }
         }
@@ -491,7 +493,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
 // LDS: Q row-major (S^T B), Q^T (dK B), dO row-major (dP^T B),
 //      dO^T (dV B), P^T/dS^T staging (one reused buffer).
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool SWZ>
 __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
@@ -582,18 +584,18 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
                 q8b = *reinterpret_cast<const short8v*>(q_ptr + (long long)(qg0 + 1) * q_row_stride + col);
                 d8b = *reinterpret_cast<const short8v*>(do_ptr + (long long)(qg0 + 1) * q_row_stride + col);
             }
-            *reinterpret_cast<short8v*>(swz(q_lds, row0, col)) = q8a;
-            *reinterpret_cast<short8v*>(swz(q_lds, row0 + 1, col)) = q8b;
-            *reinterpret_cast<short8v*>(swz(do_lds, row0, col)) = d8a;
-            *reinterpret_cast<short8v*>(swz(do_lds, row0 + 1, col)) = d8b;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(q_lds, row0, col)) = q8a;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(q_lds, row0 + 1, col)) = q8b;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(do_lds, row0, col)) = d8a;
+            *reinterpret_cast<short8v*>(swzb<SWZ>(do_lds, row0 + 1, col)) = d8b;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
                 unsigned int pq = ((unsigned int)(unsigned short)q8a[j]) |
                                   (((unsigned int)(unsigned short)q8b[j]) << 16);
                 unsigned int pd = ((unsigned int)(unsigned short)d8a[j]) |
                                   (((unsigned int)(unsigned short)d8b[j]) << 16);
-                *reinterpret_cast<unsigned int*>(swz(qt_lds, col + j, row0)) = pq;
-                *reinterpret_cast<unsigned int*>(swz(dot_lds, col + j, row0)) = pd;
+                *reinterpret_cast<unsigned int*>(swzb<SWZ>(qt_lds, col + j, row0)) = pq;
+                *reinterpret_cast<unsigned int*>(swzb<SWZ>(dot_lds, col + j, row0)) = pd;
             }
         }
         __syncthreads();
@@ -606,8 +608,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
             acc_dp[n] = frag_c{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kk = 0; kk < KD; kk++) {
-                frag_ab bq = *reinterpret_cast<const frag_ab*>(swz(q_lds, n * 16 + l16, kk * 32 + lk8));
-                frag_ab bdo = *reinterpret_cast<const frag_ab*>(swz(do_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bq = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(q_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bdo = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(do_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_s[n] = mfma16(ak[kk], bq, acc_s[n]);
                 acc_dp[n] = mfma16(av[kk], bdo, acc_dp[n]);
             }
@@ -637,15 +639,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
         for (int n = 0; n < NN; n++)
 #pragma unroll
             for (int r = 0; r < 4; r++)
-                *reinterpret_cast<ushort_t*>(swz(pt_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
+                *reinterpret_cast<ushort_t*>(swzb<SWZ>(pt_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
                     f32_to_bf16(pt_vals[n][r]);
         __syncthreads();
 #pragma unroll
         for (int kk = 0; kk < BLK_M / 32; kk++) {
-            frag_ab apt = *reinterpret_cast<const frag_ab*>(swz(pt_lds, wave * 16 + l16, kk * 32 + lk8));
+            frag_ab apt = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(pt_lds, wave * 16 + l16, kk * 32 + lk8));
 #pragma unroll
             for (int n = 0; n < ND; n++) {
-                frag_ab bdot = *reinterpret_cast<const frag_ab*>(swz(dot_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bdot = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(dot_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_dv[n] = mfma16(apt, bdot, acc_dv[n]);
             }
         }
@@ -656,15 +658,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
         for (int n = 0; n < NN; n++)
 #pragma unroll
             for (int r = 0; r < 4; r++)
-                *reinterpret_cast<ushort_t*>(swz(pt_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
+                *reinterpret_cast<ushort_t*>(swzb<SWZ>(pt_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) =
                     f32_to_bf16(dst_vals[n][r]);
         __syncthreads();
 #pragma unroll
         for (int kk = 0; kk < BLK_M / 32; kk++) {
-            frag_ab adst = *reinterpret_cast<const frag_ab*>(swz(pt_lds, wave * 16 + l16, kk * 32 + lk8));
+            frag_ab adst = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(pt_lds, wave * 16 + l16, kk * 32 + lk8));
 #pragma unroll
             for (int n = 0; n < ND; n++) {
-                frag_ab bqt = *reinterpret_cast<const frag_ab*>(swz(qt_lds, n * 16 + l16, kk * 32 + lk8));
+                frag_ab bqt = *reinterpret_cast<const frag_ab*>(swzb<SWZ>(qt_lds, n * 16 + l16, kk * 32 + lk8));
                 acc_dk[n] = mfma16(adst, bqt, acc_dk[n]);
             }
         }
@@ -718,6 +720,9 @@ void launch_flash_fwd_variant(const void* q, const void* k, const void* v, void*
         case 2: flash_fwd_t<128, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
         case 3: flash_fwd_t<128, 2, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
         default:
+            // within-probe A/B (tools/bench_fa.py, after the wave-local
+            // lgkmcnt replaced the P barrier): MF1+swz 174 TF > MF1+linear
+            // 170 > MF2 variants (MF2 = 298 VGPR = 1 wave/SIMD)
             if (D == 128) flash_fwd_t<128, 1, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
             else if (D == 64) flash_fwd_t<64, 1, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
             else if (D == 32) flash_fwd_t<32, 1, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream);
@@ -730,7 +735,7 @@ void launch_flash_fwd(const void* q, const void* k, const void* v, void* o,
     launch_flash_fwd_variant(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, D, scale, causal, 0, stream);
 }
 
-template <int D>
+template <int D, bool SWZ>
 static void flash_bwd_t(const void* dout, const void* q, const void* k, const void* v,
                         const void* o, const float* lse, float* delta,
                         void* dq, void* dk, void* dv,
@@ -742,15 +747,31 @@ static void flash_bwd_t(const void* dout, const void* q, const void* k, const vo
     hipLaunchKernelGGL(flash_bwd_delta_kernel<D>, dim3(dgrid), dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)dout, (const ushort_t*)o, delta, B, Sq, Hq);
     dim3 gq((Sq + BLK_M - 1) / BLK_M, B * Hq);
-    hipLaunchKernelGGL(flash_bwd_dq_kernel<D>, gq, dim3(FA_BLOCK), 0, stream,
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<D, SWZ>), gq, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dq,
                        B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
     dim3 gkv((Skv + BLK_N - 1) / BLK_N, B * Hq);
-    hipLaunchKernelGGL(flash_bwd_dkv_kernel<D>, gkv, dim3(FA_BLOCK), 0, stream,
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, SWZ>), gkv, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dk, (ushort_t*)dv,
                        B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+}
+
+void launch_flash_bwd_variant(const void* dout, const void* q, const void* k, const void* v,
+                              const void* o, const float* lse, float* delta,
+                              void* dq, void* dk, void* dv,
+                              int B, int Sq, int Skv, int Hq, int Hk, int D,
+                              float scale, bool causal, int variant, hipStream_t stream) {
+    // within-probe A/B: linear 202 TF > swizzled 172 for the bwd pair
+    // (their extra LDS tiles make the swizzle's address VALU the bottleneck)
+    if (variant == 0 && D == 128) {
+        flash_bwd_t<128, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+        return;
+    }
+    if (D == 128) flash_bwd_t<128, true>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    else if (D == 64) flash_bwd_t<64, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    else if (D == 32) flash_bwd_t<32, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
 }
 
 void launch_flash_bwd(const void* dout, const void* q, const void* k, const void* v,
@@ -758,7 +779,6 @@ void launch_flash_bwd(const void* dout, const void* q, const void* k, const void
                       void* dq, void* dk, void* dv,
                       int B, int Sq, int Skv, int Hq, int Hk, int D,
                       float scale, bool causal, hipStream_t stream) {
-    if (D == 128) flash_bwd_t<128>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
-    else if (D == 64) flash_bwd_t<64>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
-    else if (D == 32) flash_bwd_t<32>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, causal, stream);
+    launch_flash_bwd_variant(dout, q, k, v, o, lse, delta, dq, dk, dv,
+                             B, Sq, Skv, Hq, Hk, D, scale, causal, 0, stream);
 }

@@ -59,20 +59,22 @@ def main():
         mean = sum(ts) / len(ts)
         print(f"  fwd variant {var} ({name:<11}): {mean*1e3:8.2f} ms  {flops/mean/1e12:7.1f} TF/s")
 
-    # backward timing (current build's single variant)
+    # backward variants A/B (0=swizzled, 1=linear)
     o, lse = C.flash_attn_fwd_ex(q, k, v, True, 0)
     do = torch.randn_like(o)
-    torch.cuda.synchronize()
-    ts = []
-    for _ in range(args.reps):
-        torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        C.flash_attn_bwd(do, q, k, v, o, lse, True)
-        torch.cuda.synchronize()
-        ts.append(time.perf_counter() - t0)
-    mean = sum(sorted(ts)[1:-1] or ts) / max(1, len(ts) - 2)
-    bwd_flops = flops * 2.5  # dq(3 matmuls) + dkv(4 matmuls) vs fwd's 2 -> 3.5x total/1.4... report raw
-    print(f"  bwd (dq+dkv+delta)        : {mean*1e3:8.2f} ms  {bwd_flops/mean/1e12:7.1f} TF/s")
+    bres = {0: [], 1: []}
+    for rep in range(args.reps):
+        for var in (0, 1):
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            C.flash_attn_bwd_ex(do, q, k, v, o, lse, True, var)
+            torch.cuda.synchronize()
+            bres[var].append(time.perf_counter() - t0)
+    bwd_flops = flops * 3.5  # 7 matmuls vs fwd's 2
+    for var, name in ((0, "swz"), (1, "linear")):
+        ts = sorted(bres[var])[1:-1] or bres[var]
+        mean = sum(ts) / len(ts)
+        print(f"  bwd variant {var} ({name:<6}): {mean*1e3:8.2f} ms  {bwd_flops/mean/1e12:7.1f} TF/s")
 
 
 if __name__ == "__main__":
