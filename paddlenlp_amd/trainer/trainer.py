@@ -332,6 +332,12 @@ class Trainer:
                     done = True
                     break
             epoch += 1
+            # the consumed-samples resume skip applies only to the first
+            # (resumed) epoch; later epochs start from sample 0
+            sampler = getattr(train_dataloader, "batch_sampler", None)
+            if sampler is not None and hasattr(sampler, "consumed_samples"):
+                sampler.consumed_samples = 0
+                sampler.set_epoch(int(epoch))
             self.control = self.callback_handler.on_epoch_end(args, self.state, self.control)
             if epoch >= num_train_epochs:
                 done = True
