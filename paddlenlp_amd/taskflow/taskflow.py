@@ -1,0 +1,88 @@
+"""Taskflow: task-name -> pipeline dispatcher.
+
+Reference behavior: paddlenlp/taskflow/taskflow.py:758 (Taskflow(task, model))
+with the TASKS registry :48.  Round-1 scope: generation-backed tasks
+(text_generation / question_answering / dialogue / poetry_generation /
+code_generation) run on any local CausalLM; the classic NLU pipelines are
+registered with informative errors until their task models are ported.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from ..generation import GenerationConfig
+from ..transformers import AutoModelForCausalLM, AutoTokenizer
+from ..utils.log import logger
+
+
+class _GenerationTask:
+    TEMPLATES = {
+        "text_generation": "{text}",
+        "question_answering": "问题：{text}\n答案：",
+        "dialogue": "{text}",
+        "poetry_generation": "以「{text}」为题作诗：",
+        "code_generation": "{text}",
+    }
+
+    def __init__(self, task: str, model_path: str, max_new_tokens: int = 64,
+                 do_sample: bool = False, **kwargs):
+        self.task = task
+        self.tokenizer = AutoTokenizer.from_pretrained(model_path)
+        self.model = AutoModelForCausalLM.from_pretrained(model_path)
+        if torch.cuda.is_available():
+            self.model = self.model.to("cuda:0")
+        self.model.eval()
+        self.gen = GenerationConfig(
+            max_new_tokens=max_new_tokens, do_sample=do_sample,
+            eos_token_id=self.tokenizer.eos_token_id,
+            pad_token_id=self.tokenizer.pad_token_id or 0,
+        )
+
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        template = self.TEMPLATES.get(self.task, "{text}")
+        prompts = [template.format(text=t) for t in texts]
+        enc = self.tokenizer(prompts, padding=True, return_tensors="pt")
+        device = next(self.model.parameters()).device
+        out, _ = self.model.generate(enc["input_ids"].to(device), self.gen)
+        decoded = self.tokenizer.batch_decode(out, skip_special_tokens=True)
+        results = [{"text": t, "answer": d} for t, d in zip(texts, decoded)]
+        return results[0] if single else results
+
+
+GENERATION_TASKS = set(_GenerationTask.TEMPLATES)
+
+# API-surface parity with the reference registry (taskflow/taskflow.py:48);
+# pipelines beyond generation are pending their task models.
+PENDING_TASKS = {
+    "dependency_parsing", "document_intelligence", "fill_mask",
+    "information_extraction", "knowledge_mining", "lexical_analysis", "ner",
+    "pos_tagging", "sentiment_analysis", "text_classification",
+    "text_correction", "text_similarity", "feature_extraction",
+    "zero_shot_text_classification", "word_segmentation",
+}
+
+TASKS = sorted(GENERATION_TASKS | PENDING_TASKS)
+
+
+class Taskflow:
+    def __init__(self, task: str, model: Optional[str] = None, **kwargs):
+        if task in GENERATION_TASKS:
+            if model is None:
+                raise ValueError(
+                    f"Taskflow('{task}') needs a local model path via model= "
+                    "(no network access in this environment)")
+            self.task_instance = _GenerationTask(task, model, **kwargs)
+        elif task in PENDING_TASKS:
+            raise NotImplementedError(
+                f"Task '{task}' is registered but its task model is not ported "
+                f"yet. Available now: {sorted(GENERATION_TASKS)}")
+        else:
+            raise ValueError(f"Unknown task '{task}'. Registered: {TASKS}")
+        self.task = task
+
+    def __call__(self, inputs):
+        return self.task_instance(inputs)

@@ -43,6 +43,8 @@ from .trainer_callback import (
     TrainerControl,
     TrainerState,
 )
+from .integrations import get_reporting_integration_callbacks
+from .plugins.timer import get_timers
 from .trainer_utils import (
     TrainOutput,
     caculate_llm_flops,
@@ -89,6 +91,7 @@ class Trainer:
         self.state.is_world_process_zero = args.process_index == 0
         self.control = TrainerControl()
         callbacks = list(DEFAULT_CALLBACKS) + (callbacks or [])
+        callbacks += get_reporting_integration_callbacks(args.report_to)
         self.callback_handler = CallbackHandler(
             callbacks, self.model, self.tokenizer, self.optimizer, self.lr_scheduler
         )
@@ -297,13 +300,16 @@ class Trainer:
                 if accum_count == 0:
                     self.control = self.callback_handler.on_step_begin(args, self.state, self.control)
 
+                timers = get_timers()
                 if is_pipeline:
                     # buffer micro-batches; the 1F1B engine consumes them as
                     # one optimizer step (reference training_pipeline_step
                     # trainer.py:2246-2290)
                     pipe_buffer.append(self._prepare_inputs(inputs))
                 else:
+                    timers("forward-backward").start()
                     loss = self.training_step(model, inputs)
+                    timers("forward-backward").stop()
                     tr_loss += loss.detach()
                 if "input_ids" in inputs:
                     self._tokens_since_last_log += inputs["input_ids"].numel()
@@ -404,6 +410,8 @@ class Trainer:
     def optimizer_step(self, model: nn.Module):
         args = self.args
         topo = self.topology
+        timers = get_timers()
+        timers("all-reduce").start()
         # DP gradient sync (reference fused_allreduce_gradients :1079-1110)
         if self._zero is not None:
             self._zero.reduce_gradients_and_step_pre()
@@ -419,12 +427,15 @@ class Trainer:
                 if getattr(p, "sequence_parallel", False) and p.grad is not None:
                     dist.all_reduce(p.grad, group=topo.model_parallel_group)
 
+        timers("all-reduce").stop()
         if args.max_grad_norm and args.max_grad_norm > 0:
             self._clip_grad_norm(model)
 
+        timers("optimizer-step").start()
         self.optimizer.step()
         if self._zero is not None:
             self._zero.step_post()
+        timers("optimizer-step").stop()
         self.lr_scheduler.step()
         self.optimizer.zero_grad(set_to_none=True)
 
@@ -506,6 +517,10 @@ class Trainer:
             self._tokens_since_last_log = 0
             self._last_log_time = time.time()
             self._globalstep_last_logged = state.global_step
+            if not args.skip_profile_timer:
+                timer_msg = get_timers().log()
+                if timer_msg:
+                    logs["timers"] = timer_msg
             self.log(logs)
         if control.should_evaluate and self.eval_dataset is not None:
             metrics = self.evaluate()
