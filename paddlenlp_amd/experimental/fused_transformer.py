@@ -82,7 +82,7 @@ def rope_cache_append_ref(qkv, k_cache, v_cache, block_table, seq_lens_before,
     q = qkv[:, :, :Hq]
     k = qkv[:, :, Hq:Hq + Hk]
     v = qkv[:, :, Hq + Hk:]
-    q_out = torch.empty_like(q)
+    q_out = torch.zeros_like(q)
     for b in range(B):
         n_tok = int(token_counts[b]) if token_counts is not None else T
         start = int(seq_lens_before[b])

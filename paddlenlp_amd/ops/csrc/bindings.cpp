@@ -312,7 +312,9 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     int D = k_cache.size(3);
     int block_size = k_cache.size(1);
     int max_blocks = block_table.size(1);
-    auto q_out = torch::empty({B, T, Hq, (long)D}, qkv.options());
+    // zeros (not empty): tokens beyond token_counts[b] are skipped by the
+    // kernel and must compare deterministically
+    auto q_out = torch::zeros({B, T, Hq, (long)D}, qkv.options());
     auto cf = cos_t.to(torch::kFloat32).contiguous();
     auto sf = sin_t.to(torch::kFloat32).contiguous();
     const int* tc = nullptr;

@@ -59,7 +59,7 @@ def main():
         mean = sum(ts) / len(ts)
         print(f"  fwd variant {var} ({name:<11}): {mean*1e3:8.2f} ms  {flops/mean/1e12:7.1f} TF/s")
 
-    # backward variants A/B (0=swizzled, 1=linear)
+    # backward variants A/B (0=linear [default], 1=swizzled)
     o, lse = C.flash_attn_fwd_ex(q, k, v, True, 0)
     do = torch.randn_like(o)
     bres = {0: [], 1: []}
@@ -71,7 +71,7 @@ def main():
             torch.cuda.synchronize()
             bres[var].append(time.perf_counter() - t0)
     bwd_flops = flops * 3.5  # 7 matmuls vs fwd's 2
-    for var, name in ((0, "swz"), (1, "linear")):
+    for var, name in ((0, "linear"), (1, "swz")):
         ts = sorted(bres[var])[1:-1] or bres[var]
         mean = sum(ts) / len(ts)
         print(f"  bwd variant {var} ({name:<6}): {mean*1e3:8.2f} ms  {bwd_flops/mean/1e12:7.1f} TF/s")
