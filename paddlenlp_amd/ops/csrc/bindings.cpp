@@ -17,6 +17,8 @@ void launch_flash_fwd(const void*, const void*, const void*, void*, float*, int,
 void launch_flash_fwd_variant(const void*, const void*, const void*, void*, float*, int, int, int, int, int, int, float, bool, int, hipStream_t);
 void launch_flash_bwd(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, int, int, int, int, int, int, float, bool, hipStream_t);
 void launch_flash_bwd_variant(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, int, int, int, int, int, int, float, bool, int, hipStream_t);
+void launch_flash_fwd_mask(const void*, const void*, const void*, void*, float*, const int*, int, int, int, int, int, int, float, hipStream_t);
+void launch_flash_bwd_mask(const void*, const void*, const void*, const void*, const void*, const float*, float*, void*, void*, void*, const int*, int, int, int, int, int, int, float, hipStream_t);
 
 struct AdamWChunk {
     void* param;
@@ -282,6 +284,47 @@ void fused_adamw(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
                  (float)eps, (float)wd, bias1, bias2, cur_stream());
 }
 
+std::vector<torch::Tensor> flashmask_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                              torch::Tensor startend) {
+    CHECK_GPU(q); CHECK_CONTIG(q); CHECK_BF16(q);
+    int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Skv = k.size(1), Hk = k.size(2);
+    TORCH_CHECK(startend.numel() == (long long)B * Skv, "startend must be [B, Skv]");
+    auto se = startend.to(torch::kInt32).contiguous();
+    auto o = torch::empty_like(q);
+    auto lse = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_flash_fwd_mask(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                          lse.data_ptr<float>(), se.data_ptr<int>(),
+                          B, Sq, Skv, Hq, Hk, D, scale, cur_stream());
+    return {o, lse};
+}
+
+std::vector<torch::Tensor> flashmask_attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                              torch::Tensor k, torch::Tensor v,
+                                              torch::Tensor o, torch::Tensor lse,
+                                              torch::Tensor startend) {
+    int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+    int Skv = k.size(1), Hk = k.size(2);
+    auto se = startend.to(torch::kInt32).contiguous();
+    auto dq = torch::empty_like(q);
+    auto dk_h = torch::empty({B, Skv, Hq, D}, k.options());
+    auto dv_h = torch::empty({B, Skv, Hq, D}, v.options());
+    auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
+    float scale = 1.0f / std::sqrt((float)D);
+    launch_flash_bwd_mask(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
+                          dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(),
+                          se.data_ptr<int>(), B, Sq, Skv, Hq, Hk, D, scale, cur_stream());
+    torch::Tensor dk = dk_h, dv = dv_h;
+    if (Hq != Hk) {
+        int G = Hq / Hk;
+        dk = dk_h.view({B, Skv, Hk, G, D}).sum(3);
+        dv = dv_h.view({B, Skv, Hk, G, D}).sum(3);
+    }
+    return {dq, dk, dv};
+}
+
 // ---------------------------------------------------------------------------
 // paged-KV inference ops
 // ---------------------------------------------------------------------------
@@ -341,6 +384,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flash_attn_fwd_ex", &flash_attn_fwd_ex);
     m.def("flash_attn_bwd", &flash_attn_bwd);
     m.def("flash_attn_bwd_ex", &flash_attn_bwd_ex);
+    m.def("flashmask_attn_fwd", &flashmask_attn_fwd);
+    m.def("flashmask_attn_bwd", &flashmask_attn_bwd);
     m.def("fused_adamw", &fused_adamw);
     m.def("paged_decode_attn", &paged_decode_attn);
     m.def("rope_cache_append", &rope_cache_append);

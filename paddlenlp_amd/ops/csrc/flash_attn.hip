@@ -89,6 +89,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const ushort_t* __restrict__ v,   // [B, Skv, Hk, D]
     ushort_t* __restrict__ o,         // [B, Sq, Hq, D]
     float* __restrict__ lse,          // [B, Hq, Sq]
+    const int* __restrict__ startend, // FlashMask [B, Skv] or nullptr:
+                                      // key j visible to queries j<=i<start[j]
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int FWD_BLKM = FA_WAVES * MF * 16;
     constexpr int KD = D / 32;   // MFMA K-steps over the head dim
@@ -201,11 +203,14 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
 #pragma unroll
             for (int n = 0; n < NN; n++) {
                 int kvg = kv_base + n * 16 + l16;
+                int kv_end = (startend && kvg < Skv)
+                                 ? startend[(long long)b * Skv + kvg] : Sq + causal_off;
 #pragma unroll
                 for (int r = 0; r < 4; r++) {
                     int qg = qrow0 + r;
                     bool vis = (kvg < Skv) && (qg < Sq);
                     if (causal) vis = vis && (kvg <= qg + causal_off);
+                    if (startend) vis = vis && (qg < kv_end);
                     acc_s[n][r] = vis ? acc_s[n][r] * scale : -INFINITY;
                 }
             }
@@ -328,7 +333,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    ushort_t* __restrict__ dq,
+    ushort_t* __restrict__ dq, const int* __restrict__ startend,
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int KD = D / 32;
     constexpr int ND = D / 16;
@@ -447,11 +452,14 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
 #pragma unroll
         for (int n = 0; n < NN; n++) {
             int kvg = kv_base + n * 16 + l16;
+            int kv_end = (startend && kvg < Skv)
+                             ? startend[(long long)b * Skv + kvg] : Sq + causal_off;
 #pragma unroll
             for (int r = 0; r < 4; r++) {
                 int qg = qrow0 + r;
                 bool vis = (kvg < Skv) && (qg < Sq);
                 if (causal) vis = vis && (kvg <= qg + causal_off);
+                if (startend) vis = vis && (qg < kv_end);
                 float p = vis ? __expf(acc_s[n][r] * scale - lse_r[r]) : 0.f;
                 float ds = p * (acc_dp[n][r] - dl_r[r]) * scale;
                 *reinterpret_cast<ushort_t*>(swzb<SWZ>(ds_lds, wave * 16 + (lane >> 4) * 4 + r, n * 16 + l16)) = f32_to_bf16(ds);
@@ -501,6 +509,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     ushort_t* __restrict__ dk,   // [B, Skv, Hq, D] (per q-head)
     ushort_t* __restrict__ dv,   // [B, Skv, Hq, D]
+    const int* __restrict__ startend,
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int KD = D / 32;
     constexpr int ND = D / 16;
@@ -629,6 +638,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
                 int kvg = kvrow0 + r;
                 bool vis = (kvg < Skv) && (qg < Sq);
                 if (causal) vis = vis && (kvg <= qg + causal_off);
+                if (startend) vis = vis && (kvg < Skv) &&
+                    (qg < startend[(long long)b * Skv + kvg]);
                 float p = vis ? __expf(acc_s[n][r] * scale - lse_q) : 0.f;
                 pt_vals[n][r] = p;
                 dst_vals[n][r] = p * (acc_dp[n][r] - dl_q) * scale;
@@ -703,12 +714,23 @@ void launch_mfma_probe(const void* A, const void* B, float* C, hipStream_t strea
 template <int D, int MF, bool SWZ>
 static void flash_fwd_t(const void* q, const void* k, const void* v, void* o,
                         float* lse, int B, int Sq, int Skv, int Hq, int Hk,
-                        float scale, bool causal, hipStream_t stream) {
+                        float scale, bool causal, hipStream_t stream,
+                        const int* startend = nullptr) {
     constexpr int BLKM = FA_WAVES * MF * 16;
     dim3 grid((Sq + BLKM - 1) / BLKM, B * Hq);
     hipLaunchKernelGGL((flash_fwd_kernel<D, MF, SWZ>), grid, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
-                       (ushort_t*)o, lse, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+                       (ushort_t*)o, lse, startend, B, Sq, Skv, Hq, Hk, scale,
+                       causal ? 1 : 0);
+}
+
+void launch_flash_fwd_mask(const void* q, const void* k, const void* v, void* o,
+                           float* lse, const int* startend,
+                           int B, int Sq, int Skv, int Hq, int Hk, int D,
+                           float scale, hipStream_t stream) {
+    if (D == 128) flash_fwd_t<128, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
+    else if (D == 64) flash_fwd_t<64, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
+    else if (D == 32) flash_fwd_t<32, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
 }
 
 // variant dispatch for A/B benchmarking (variant: 0=MF1+swz [default],
@@ -744,7 +766,8 @@ static void flash_bwd_t(const void* dout, const void* q, const void* k, const vo
                         const void* o, const float* lse, float* delta,
                         void* dq, void* dk, void* dv,
                         int B, int Sq, int Skv, int Hq, int Hk,
-                        float scale, bool causal, hipStream_t stream) {
+                        float scale, bool causal, hipStream_t stream,
+                        const int* startend = nullptr) {
     long long rows = (long long)B * Sq * Hq;
     int waves_per_block = FA_BLOCK / 64;
     int dgrid = (int)((rows + waves_per_block - 1) / waves_per_block);
@@ -753,13 +776,23 @@ static void flash_bwd_t(const void* dout, const void* q, const void* k, const vo
     dim3 gq((Sq + BLK_M - 1) / BLK_M, B * Hq);
     hipLaunchKernelGGL((flash_bwd_dq_kernel<D, SWZ>), gq, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
-                       (const ushort_t*)dout, lse, delta, (ushort_t*)dq,
+                       (const ushort_t*)dout, lse, delta, (ushort_t*)dq, startend,
                        B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
     dim3 gkv((Skv + BLK_N - 1) / BLK_N, B * Hq);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, SWZ>), gkv, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dk, (ushort_t*)dv,
-                       B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+                       startend, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+}
+
+void launch_flash_bwd_mask(const void* dout, const void* q, const void* k, const void* v,
+                           const void* o, const float* lse, float* delta,
+                           void* dq, void* dk, void* dv, const int* startend,
+                           int B, int Sq, int Skv, int Hq, int Hk, int D,
+                           float scale, hipStream_t stream) {
+    if (D == 128) flash_bwd_t<128, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
+    else if (D == 64) flash_bwd_t<64, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
+    else if (D == 32) flash_bwd_t<32, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
 }
 
 void launch_flash_bwd_variant(const void* dout, const void* q, const void* k, const void* v,

@@ -148,6 +148,26 @@ class _FlashAttnFunction(torch.autograd.Function):
         return dq, dk, dv, None
 
 
+class _FlashMaskFunction(torch.autograd.Function):
+    """FlashMask sparse-causal attention (reference flashmask_attention,
+    llama/fusion_ops.py:218-238): key j visible to queries j <= i < start[j]."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, startend):
+        C = _load_extension()
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = C.flashmask_attn_fwd(q, k, v, startend)
+        ctx.save_for_backward(q, k, v, o, lse, startend)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        C = _load_extension()
+        q, k, v, o, lse, startend = ctx.saved_tensors
+        dq, dk, dv = C.flashmask_attn_bwd(do.contiguous(), q, k, v, o, lse, startend)
+        return dq, dk, dv, None
+
+
 def flash_attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -157,8 +177,16 @@ def flash_attention(
     startend_row_indices: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """q: [B,S,Hq,D]; k,v: [B,S,Hk,D].  Returns [B,S,Hq,D]."""
-    if q.is_cuda and attn_mask is None and startend_row_indices is None:
-        return _FlashAttnFunction.apply(q, k, v, causal)
+    if q.is_cuda and attn_mask is None:
+        if startend_row_indices is None:
+            return _FlashAttnFunction.apply(q, k, v, causal)
+        if causal:
+            # [B, h(=1), Skv, 1] -> [B, Skv] (per-head masks pending)
+            se = startend_row_indices
+            if se.dim() == 4:
+                assert se.shape[1] == 1, "per-kv-head FlashMask pending"
+                se = se[:, 0, :, 0]
+            return _FlashMaskFunction.apply(q, k, v, se.contiguous())
     return reference.flash_attention(
         q, k, v, causal=causal, attn_mask=attn_mask,
         startend_row_indices=startend_row_indices,
