@@ -543,6 +543,16 @@ class LlamaForCausalLM(LlamaPretrainedModel, GenerationMixin):
             from ...parallel.sequence_parallel import GatherOp
 
             hidden_states = GatherOp(hidden_states)
+        if (labels is not None and not use_cache
+                and self.config.use_fused_linear_cross_entropy
+                and self.config.tensor_parallel_degree <= 1):
+            # chunked head+CE: never materializes [tokens, vocab] logits
+            # (reference fused_head_and_loss_fn, tensor_parallel_utils.py:112)
+            from ..tensor_parallel_utils import fused_head_and_loss_fn
+
+            loss = fused_head_and_loss_fn(
+                hidden_states, self.lm_head.weight, labels)
+            return loss, None
         logits = self.lm_head(hidden_states)
         if labels is not None:
             loss = self.criterion(logits, labels)

@@ -1,0 +1,93 @@
+"""Fused LM-head + cross-entropy (chunked).
+
+Reference behavior: paddlenlp/transformers/tensor_parallel_utils.py:112
+fused_head_and_loss_fn / FusedHeadAndCrossEntropy :145 — computes the loss in
+token chunks without ever materializing the full [tokens, vocab] logits
+(memory-critical for 128k vocabularies: Llama-3-8B at batch 8 x 4096 would
+need 8.4 GB for bf16 logits + the same again for grads).  The backward
+recomputes each chunk's logits from the saved hidden states.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..ops import reference
+
+
+class FusedHeadAndCrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, hidden, weight, labels, ignore_index, chunk_tokens):
+        """hidden [N, H] (flattened tokens), weight [V, H], labels [N]."""
+        N = hidden.shape[0]
+        total = hidden.new_zeros((), dtype=torch.float32)
+        count = hidden.new_zeros((), dtype=torch.float32)
+        maxlses = []
+        for s in range(0, N, chunk_tokens):
+            e = min(N, s + chunk_tokens)
+            logits = hidden[s:e] @ weight.t()
+            if hidden.is_cuda:
+                from ..ops.functional import _load_extension
+
+                C = _load_extension()
+                loss, maxlse = C.cross_entropy_fwd(logits.contiguous(), labels[s:e], ignore_index)
+            else:
+                l32 = logits.float()
+                m = l32.max(-1, keepdim=True).values
+                lse = (l32 - m).exp().sum(-1).log()
+                safe = labels[s:e].clamp(min=0)
+                tgt = (l32 - m).gather(-1, safe[:, None]).squeeze(-1)
+                loss = torch.where(labels[s:e] != ignore_index, lse - tgt,
+                                   torch.zeros_like(lse))
+                maxlse = torch.stack([m.squeeze(-1), lse], dim=-1)
+            total = total + loss.sum()
+            count = count + (labels[s:e] != ignore_index).sum()
+            maxlses.append(maxlse)
+        ctx.save_for_backward(hidden, weight, labels, torch.cat(maxlses))
+        ctx.ignore_index = ignore_index
+        ctx.chunk_tokens = chunk_tokens
+        count = count.clamp(min=1)
+        ctx.count = count
+        return total / count
+
+    @staticmethod
+    def backward(ctx, dloss):
+        hidden, weight, labels, maxlse = ctx.saved_tensors
+        ignore_index = ctx.ignore_index
+        chunk = ctx.chunk_tokens
+        N = hidden.shape[0]
+        dhidden = torch.empty_like(hidden)
+        dweight = torch.zeros_like(weight, dtype=torch.float32)
+        scale = (dloss / ctx.count).float()
+        for s in range(0, N, chunk):
+            e = min(N, s + chunk)
+            h = hidden[s:e]
+            logits = h @ weight.t()
+            lab = labels[s:e]
+            if hidden.is_cuda:
+                from ..ops.functional import _load_extension
+
+                C = _load_extension()
+                dl = scale.expand(e - s).contiguous()
+                dlogits = C.cross_entropy_bwd(dl, logits.contiguous(), lab,
+                                              maxlse[s:e].contiguous(), ignore_index)
+            else:
+                l32 = logits.float()
+                p = (l32 - maxlse[s:e, 0:1]).exp() / maxlse[s:e, 1:2].exp()
+                onehot = F.one_hot(lab.clamp(min=0), weight.shape[0]).float()
+                valid = (lab != ignore_index).float()[:, None]
+                dlogits = ((p - onehot) * valid * scale).to(logits.dtype)
+            dhidden[s:e] = dlogits @ weight
+            dweight += (dlogits.t().float() @ h.float())
+        return dhidden, dweight.to(weight.dtype), None, None, None
+
+
+def fused_head_and_loss_fn(hidden, head_weight, labels, ignore_index: int = -100,
+                           chunk_tokens: int = 2048):
+    """hidden [B, S, H] or [N, H]; labels matching leading dims; mean loss."""
+    H = hidden.shape[-1]
+    hidden = hidden.reshape(-1, H)
+    labels = labels.reshape(-1)
+    return FusedHeadAndCrossEntropy.apply(hidden, head_weight, labels,
+                                          ignore_index, chunk_tokens)

@@ -102,3 +102,24 @@ def test_flashmask_startend_row_indices():
     out = ops.flash_attention(q, k, v, causal=True, startend_row_indices=idx)
     ref = ops.flash_attention(q, k, v, causal=True)
     assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_fused_head_and_loss_fn():
+    """Chunked head+CE == plain lm_head + cross_entropy (values and grads)."""
+    from paddlenlp_amd.transformers.tensor_parallel_utils import fused_head_and_loss_fn
+
+    torch.manual_seed(0)
+    N, H, V = 20, 16, 64
+    hidden = torch.randn(N, H, requires_grad=True)
+    weight = torch.randn(V, H, requires_grad=True)
+    labels = torch.randint(0, V, (N,))
+    labels[3] = -100
+    loss = fused_head_and_loss_fn(hidden, weight, labels, chunk_tokens=7)
+    hr = hidden.detach().clone().requires_grad_()
+    wr = weight.detach().clone().requires_grad_()
+    ref = F.cross_entropy(hr @ wr.t(), labels, ignore_index=-100)
+    assert torch.allclose(loss, ref, atol=1e-5), (loss, ref)
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(hidden.grad, hr.grad, atol=1e-5), (hidden.grad - hr.grad).abs().max()
+    assert torch.allclose(weight.grad, wr.grad, atol=1e-4), (weight.grad - wr.grad).abs().max()

@@ -230,3 +230,24 @@ def test_flashmask_attention_gpu(C):
     assert torch.allclose(q.grad.float(), qr.grad, atol=5e-2, rtol=5e-2)
     assert torch.allclose(k.grad.float(), kr.grad, atol=5e-2, rtol=5e-2)
     assert torch.allclose(v.grad.float(), vr.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_fused_head_and_loss_fn_gpu(C):
+    from paddlenlp_amd.transformers.tensor_parallel_utils import fused_head_and_loss_fn
+
+    torch.manual_seed(0)
+    N, H, V = 256, 128, 1024
+    hidden = _bf16(torch.randn(N, H, device="cuda")).requires_grad_()
+    weight = _bf16(torch.randn(V, H, device="cuda")).requires_grad_()
+    labels = torch.randint(0, V, (N,), device="cuda")
+    labels[5] = -100
+    loss = fused_head_and_loss_fn(hidden, weight, labels, chunk_tokens=100)
+    hr = hidden.detach().float().requires_grad_()
+    wr = weight.detach().float().requires_grad_()
+    import torch.nn.functional as F
+    ref = F.cross_entropy(hr @ wr.t(), labels, ignore_index=-100)
+    assert torch.allclose(loss.float(), ref, atol=2e-2, rtol=1e-2), (loss, ref)
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(hidden.grad.float(), hr.grad, atol=2e-2, rtol=5e-2)
+    assert torch.allclose(weight.grad.float(), wr.grad, atol=2e-2, rtol=5e-2)
