@@ -41,12 +41,19 @@ def _param_names(model) -> Dict[int, str]:
 def save_unified_model(model, output_dir: str, topology) -> None:
     """Write model weights.  tp=1: dp/sharding rank 0 writes everything.
     tp>1: every mp rank gathers-and-merges its TP-split params to mp rank 0
-    (full merge; per-rank slice writing is a later optimization)."""
+    (full merge; per-rank slice writing is a later optimization).
+    pp>1: each stage writes a model-XXXXX-of-N.safetensors shard with base
+    names + a gathered index (reference unified_checkpoint_into_shards :928)."""
     os.makedirs(output_dir, exist_ok=True)
     from ..transformers.model_utils import PretrainedModel
+    from ..parallel.pipeline import PipelineModule
 
     tp = topology.mp_degree
     is_dp0 = topology.coords.get("dp", 0) == 0 and topology.coords.get("sharding", 0) == 0
+
+    if isinstance(model, PipelineModule):
+        _save_unified_pipe_model(model, output_dir, topology, is_dp0)
+        return
 
     if tp == 1:
         if topology.rank == 0:
@@ -86,6 +93,54 @@ def save_unified_model(model, output_dir: str, topology) -> None:
                 os.path.join(output_dir, "model.safetensors"),
                 metadata={"format": "pt"},
             )
+
+
+def _save_unified_pipe_model(model, output_dir, topology, is_dp0):
+    """Per-pp-stage shards with base param names; TP merged within the stage."""
+    from safetensors.torch import save_file
+    from ..utils.env import SAFE_WEIGHTS_INDEX_NAME, SAFE_WEIGHTS_NAME
+
+    tp = topology.mp_degree
+    mp_rank = topology.get_rank_in("mp")
+    pp_rank = topology.get_rank_in("pp")
+    pp_deg = topology.pp_degree
+    sd = model.state_dict_with_base_names()
+
+    actions = {}
+    if tp > 1 and hasattr(type(model), "_get_tensor_parallel_mappings"):
+        actions = type(model)._get_tensor_parallel_mappings(model.config, is_split=False)
+    merged = {}
+    mp_group = topology.model_parallel_group
+    for name, t in sd.items():
+        if name in actions and tp > 1:
+            shards = [torch.empty_like(t) for _ in range(tp)]
+            dist.all_gather(shards, t.contiguous(), group=mp_group)
+            if mp_rank == 0 and is_dp0:
+                merged[name] = actions[name]([x.cpu() for x in shards]).contiguous()
+        elif mp_rank == 0 and is_dp0:
+            merged[name] = t.detach().cpu().contiguous()
+
+    fname = SAFE_WEIGHTS_NAME.replace(
+        ".safetensors", f"-{pp_rank + 1:05d}-of-{pp_deg:05d}.safetensors")
+    local_map = {}
+    if mp_rank == 0 and is_dp0:
+        save_file(merged, os.path.join(output_dir, fname), metadata={"format": "pt"})
+        local_map = {k: fname for k in merged}
+        if hasattr(model, "config"):
+            model.config.save_pretrained(output_dir)
+
+    if dist.is_initialized():
+        maps = [None] * topology.world_size
+        dist.all_gather_object(maps, local_map)
+    else:
+        maps = [local_map]
+    if topology.rank == 0:
+        weight_map = {}
+        total = 0
+        for m in maps:
+            weight_map.update(m or {})
+        with open(os.path.join(output_dir, SAFE_WEIGHTS_INDEX_NAME), "w") as f:
+            json.dump({"metadata": {"total_size": total}, "weight_map": weight_map}, f, indent=2)
 
 
 def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=None) -> None:

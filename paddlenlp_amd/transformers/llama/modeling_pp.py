@@ -74,6 +74,12 @@ class LlamaForCausalLMPipe(PipelineModule):
 
     config_class = LlamaConfig
 
+    @classmethod
+    def _get_tensor_parallel_mappings(cls, config, is_split=True):
+        from .modeling import LlamaForCausalLM
+
+        return LlamaForCausalLM._get_tensor_parallel_mappings(config, is_split)
+
     def __init__(self, config: LlamaConfig):
         criterion = LlamaPretrainingCriterion(config)
         descs = [LayerDesc(EmbeddingPipe, config, name="embedding")]
