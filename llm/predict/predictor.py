@@ -45,6 +45,7 @@ class PredictorArgument:
     decode_strategy: str = field(default="sampling")  # or "greedy"
     mode: str = field(default="dynamic")
     benchmark: bool = field(default=False)
+    quant_type: str = field(default="")  # "" | "fp8" | "weight_only_int8"
 
 
 class BasePredictor:
@@ -232,6 +233,8 @@ def create_predictor(predictor_args: PredictorArgument, model=None, tokenizer=No
         engine = FusedMultiTransformer.from_llama(
             model, block_size=predictor_args.block_size,
             max_seq_len=predictor_args.total_max_length)
+        if predictor_args.quant_type:
+            engine.quantize(predictor_args.quant_type)
         device = next(model.parameters()).device
         del model
         return BlockInferencePredictor(predictor_args, engine, tokenizer, device=device)

@@ -125,6 +125,8 @@ class FusedMultiTransformer(nn.Module):
         cos, sin = ops.build_rope_cache(c.max_seq_len, hd, base=c.rope_theta)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+        self.quant_algo = None  # None | "fp8" | "weight_only_int8"
+        self._qw = {}
 
     # ------------------------------------------------------------------
     @classmethod
@@ -163,6 +165,35 @@ class FusedMultiTransformer(nn.Module):
         eng.final_norm.copy_(base.norm.weight)
         eng.lm_head.copy_(model.lm_head.weight)
         return eng
+
+    @torch.no_grad()
+    def quantize(self, algo: str = "fp8"):
+        """Weight-only quantization of the projection weights (fp8 e4m3fn via
+        the gfx950 fp8 MFMA path, or per-channel int8).  Norm scales,
+        embeddings and the LM head stay in the compute dtype."""
+        from ..quantization import quantize_fp8, quantize_int8
+
+        qfn = quantize_fp8 if algo == "fp8" else quantize_int8
+        self._qw = {}
+        for name in ("qkv_weights", "out_proj_weights", "gate_up_weights", "down_weights"):
+            plist = getattr(self, name)
+            qs = []
+            for w in plist:
+                q, sc = qfn(w.data)
+                qs.append((q, sc))
+                w.data = w.data.new_zeros(1)  # free the bf16 copy
+            self._qw[name] = qs
+        self.quant_algo = algo
+        return self
+
+    def _mm(self, x, name: str, i: int):
+        if self.quant_algo is None:
+            return x @ getattr(self, name)[i].t()
+        from ..quantization import weight_only_linear
+
+        q, sc = self._qw[name][i]
+        algo = self.quant_algo if self.quant_algo == "fp8" else "weight_only_int8"
+        return weight_only_linear(x, q, sc, None, algo)
 
     def allocate_caches(self, num_blocks: int, device):
         c = self.config
@@ -216,14 +247,14 @@ class FusedMultiTransformer(nn.Module):
         seq_lens_after = seq_lens_before + 1
         for i in range(c.num_layers):
             h = self._rms(x, self.ln_scales[i])
-            qkv = h @ self.qkv_weights[i].t()          # [B, 1, qkv_out]
+            qkv = self._mm(h, "qkv_weights", i)          # [B, 1, qkv_out]
             q = self._rope_append(i, qkv, block_table, seq_lens_before)
             attn = self._paged_attn(i, q[:, 0], block_table, seq_lens_after)
-            x = x + (attn.reshape(B, 1, -1) @ self.out_proj_weights[i].t())
+            x = x + self._mm(attn.reshape(B, 1, -1), "out_proj_weights", i)
             h = self._rms(x, self.ffn_ln_scales[i])
-            gu = h @ self.gate_up_weights[i].t()
+            gu = self._mm(h, "gate_up_weights", i)
             act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
-            x = x + act @ self.down_weights[i].t()
+            x = x + self._mm(act, "down_weights", i)
         x = self._rms(x, self.final_norm)
         logits = x[:, 0] @ self.lm_head.t()
         return logits.float()
@@ -241,7 +272,7 @@ class FusedMultiTransformer(nn.Module):
         zeros = torch.zeros_like(prompt_lens)
         for i in range(c.num_layers):
             h = self._rms(x, self.ln_scales[i])
-            qkv = h @ self.qkv_weights[i].t()          # [B, T, *]
+            qkv = self._mm(h, "qkv_weights", i)          # [B, T, *]
             q = self._rope_append(i, qkv, block_table, zeros, token_counts=prompt_lens)
             # contiguous roped K + raw V for the flash kernel
             qkv_v = qkv.view(B, T, c.num_heads + 2 * c.num_kv_heads, c.head_dim)
@@ -255,11 +286,11 @@ class FusedMultiTransformer(nn.Module):
                 attn = ops.flash_attention(q, k_roped, v, causal=True)
             else:
                 attn = reference.flash_attention(q, k_roped, v, causal=True)
-            x = x + attn.reshape(B, T, -1) @ self.out_proj_weights[i].t()
+            x = x + self._mm(attn.reshape(B, T, -1), "out_proj_weights", i)
             h = self._rms(x, self.ffn_ln_scales[i])
-            gu = h @ self.gate_up_weights[i].t()
+            gu = self._mm(h, "gate_up_weights", i)
             act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
-            x = x + act @ self.down_weights[i].t()
+            x = x + self._mm(act, "down_weights", i)
         x = self._rms(x, self.final_norm)
         idx = (prompt_lens.long() - 1).clamp(min=0)
         last = x[torch.arange(B, device=x.device), idx]

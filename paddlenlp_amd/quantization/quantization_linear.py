@@ -1,0 +1,101 @@
+"""Weight-only quantized linear layers (int8 / fp8).
+
+Reference behavior: paddlenlp/quantization/quantization_linear.py:42
+(QuantizationLinear + Column/RowParallel variants :165/:277 — weight-only
+int4/8 runtime) and the fp8 cutlass path (fused_transformer_layers.py
+FusedMultiTransformerFP8 :2491).
+
+MI355X design: fp8 uses OCP e4m3fn (the gfx950-native format — NOT the
+MI300X fnuz variant) through torch._scaled_mm, which lowers to hipBLASLt's
+fp8 MFMA path on gfx950 (~2x the bf16 rate).  int8 weight-only dequantizes
+per-channel into the activation dtype and uses the bf16 GEMM; a fused
+dequant-GEMM epilogue kernel is the planned upgrade.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+FP8_E4M3_MAX = 448.0
+
+
+def quantize_int8(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Per-output-channel symmetric int8: w [out, in] -> (int8 w, fp32 scale[out])."""
+    scale = w.abs().amax(dim=1).clamp(min=1e-8).float() / 127.0
+    q = torch.clamp(torch.round(w.float() / scale[:, None]), -127, 127).to(torch.int8)
+    return q, scale
+
+
+def quantize_fp8(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Per-tensor fp8 e4m3fn: w -> (fp8 w, fp32 scale scalar)."""
+    scale = (w.abs().amax().clamp(min=1e-8).float() / FP8_E4M3_MAX)
+    q = (w.float() / scale).clamp(-FP8_E4M3_MAX, FP8_E4M3_MAX).to(torch.float8_e4m3fn)
+    return q, scale
+
+
+def weight_only_linear(x: torch.Tensor, qweight: torch.Tensor, scale: torch.Tensor,
+                       bias: Optional[torch.Tensor] = None,
+                       algo: str = "weight_only_int8") -> torch.Tensor:
+    """y = x @ dequant(qweight)^T (+ bias)."""
+    if algo == "fp8":
+        orig_shape = x.shape
+        x2 = x.reshape(-1, orig_shape[-1])
+        if x2.is_cuda:
+            # dynamic per-tensor activation scale -> fp8 x fp8 scaled GEMM
+            x_scale = (x2.abs().amax().clamp(min=1e-8).float() / FP8_E4M3_MAX)
+            x8 = (x2.float() / x_scale).clamp(-FP8_E4M3_MAX, FP8_E4M3_MAX).to(torch.float8_e4m3fn)
+            y = torch._scaled_mm(
+                x8, qweight.t(), scale_a=x_scale, scale_b=scale,
+                bias=None, out_dtype=x.dtype,
+            )
+        else:
+            y = x2 @ (qweight.float() * scale).t().to(x.dtype)
+        y = y.reshape(*orig_shape[:-1], -1)
+    else:
+        w = (qweight.float() * scale[:, None]).to(x.dtype)
+        y = x @ w.t()
+    if bias is not None:
+        y = y + bias
+    return y
+
+
+class QuantizationLinear(nn.Module):
+    """Inference-only quantized linear (reference QuantizationLinear :42)."""
+
+    def __init__(self, in_features: int, out_features: int,
+                 quant_algo: str = "weight_only_int8", bias: bool = False,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        self.quant_algo = quant_algo
+        if quant_algo == "fp8":
+            self.register_buffer("quant_weight",
+                                 torch.zeros(out_features, in_features, dtype=torch.float8_e4m3fn))
+            self.register_buffer("quant_scale", torch.ones((), dtype=torch.float32))
+        else:
+            self.register_buffer("quant_weight",
+                                 torch.zeros(out_features, in_features, dtype=torch.int8))
+            self.register_buffer("quant_scale", torch.ones(out_features, dtype=torch.float32))
+        self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
+
+    @classmethod
+    def from_linear(cls, linear: nn.Linear, quant_algo: str = "weight_only_int8"):
+        m = cls(linear.in_features, linear.out_features, quant_algo,
+                bias=linear.bias is not None, dtype=linear.weight.dtype)
+        if quant_algo == "fp8":
+            q, s = quantize_fp8(linear.weight.data)
+        else:
+            q, s = quantize_int8(linear.weight.data)
+        m.quant_weight.copy_(q)
+        m.quant_scale.copy_(s)
+        if linear.bias is not None:
+            m.bias.data.copy_(linear.bias.data)
+        m = m.to(linear.weight.device)
+        return m
+
+    def forward(self, x):
+        return weight_only_linear(x, self.quant_weight, self.quant_scale,
+                                  self.bias, self.quant_algo)

@@ -146,3 +146,50 @@ def test_block_predictor_end_to_end():
     outs_dy = dy.predict(["the quick brown fox"])
     outs_blk = pred.predict(["the quick brown fox"])
     assert outs_blk[0] == outs_dy[0], (outs_blk, outs_dy)
+
+
+def test_quantization_linear_int8():
+    from paddlenlp_amd.quantization import QuantizationLinear
+
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(64, 32, bias=True)
+    q = QuantizationLinear.from_linear(lin, "weight_only_int8")
+    x = torch.randn(4, 64)
+    ref = lin(x)
+    out = q(x)
+    # int8 weight-only: relative error bounded by quantization step
+    assert (out - ref).abs().max() / ref.abs().max() < 0.05
+
+
+def test_quantization_linear_fp8_cpu_path():
+    from paddlenlp_amd.quantization import QuantizationLinear
+
+    torch.manual_seed(1)
+    lin = torch.nn.Linear(64, 32)
+    q = QuantizationLinear.from_linear(lin, "fp8")
+    x = torch.randn(4, 64)
+    ref = lin(x)
+    out = q(x)
+    assert (out - ref).abs().max() / ref.abs().max() < 0.12
+
+
+def test_engine_quantized_decode():
+    """int8 weight-only engine still decodes near the bf16 engine (CPU)."""
+    model = tiny_llama(seed=9)
+    eng = FusedMultiTransformer.from_llama(model, block_size=4, max_seq_len=64)
+    eng.allocate_caches(num_blocks=64, device="cpu")
+    eng_q = FusedMultiTransformer.from_llama(model, block_size=4, max_seq_len=64)
+    eng_q.allocate_caches(num_blocks=64, device="cpu")
+    eng_q.quantize("weight_only_int8")
+
+    B, T = 1, 8
+    ids = torch.randint(3, 128, (B, T), generator=torch.Generator().manual_seed(3))
+    lens = torch.tensor([T], dtype=torch.int32)
+    mgr1, mgr2 = BlockManager(64, 4, 16, B), BlockManager(64, 4, 16, B)
+    s1, s2 = mgr1.allocate_slot(T), mgr2.allocate_slot(T)
+    bt1 = mgr1.block_table[s1][None].to(torch.int32)
+    bt2 = mgr2.block_table[s2][None].to(torch.int32)
+    l1 = eng.prefill(ids, bt1, lens)
+    l2 = eng_q.prefill(ids, bt2, lens)
+    # quantized logits stay close enough to keep the same top-1 most of the time
+    assert (l1 - l2).abs().max() / l1.abs().max() < 0.2

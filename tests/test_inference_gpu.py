@@ -112,3 +112,46 @@ def test_engine_decode_gpu_matches_dygraph(C):
         chosen = ref.float().gather(-1, logits.argmax(-1, keepdim=True)).squeeze(-1)
         assert (ref.float().max(-1).values - chosen < 0.25).all(), \
             (ref.float().max(-1).values - chosen).max()
+
+
+def test_fp8_scaled_mm_gpu(C):
+    """fp8 e4m3fn weight-only GEMM via the gfx950 fp8 MFMA path."""
+    from paddlenlp_amd.quantization import QuantizationLinear
+
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(256, 512, bias=False).to("cuda", torch.bfloat16)
+    q = QuantizationLinear.from_linear(lin, "fp8")
+    x = torch.randn(64, 256, device="cuda", dtype=torch.bfloat16)
+    ref = lin(x)
+    out = q(x)
+    rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max()
+    assert rel < 0.15, rel
+
+
+def test_engine_fp8_decode_gpu(C):
+    from paddlenlp_amd.experimental import BlockManager, FusedMultiTransformer
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(3)
+    cfg = LlamaConfig(
+        vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128,
+    )
+    model = LlamaForCausalLM.from_config(cfg, dtype=torch.bfloat16, device="cuda")
+    model.eval()
+    eng = FusedMultiTransformer.from_llama(model, block_size=16, max_seq_len=128).to("cuda")
+    eng.allocate_caches(num_blocks=32, device="cuda")
+    eng.quantize("fp8")
+    B, T = 2, 16
+    ids = torch.randint(3, 512, (B, T), device="cuda")
+    lens = torch.tensor([T, T], dtype=torch.int32, device="cuda")
+    mgr = BlockManager(32, 16, 8, B)
+    slots = [mgr.allocate_slot(T) for _ in range(B)]
+    bt = torch.stack([mgr.block_table[s] for s in slots]).to("cuda", torch.int32)
+    logits = eng.prefill(ids, bt, lens)
+    with torch.no_grad():
+        ref = model(input_ids=ids)[:, -1]
+    rel = (logits - ref.float()).abs().max() / ref.float().abs().max()
+    assert torch.isfinite(logits).all()
+    assert rel < 0.3, rel
