@@ -104,3 +104,34 @@ def test_reward_model_step(tmp_path):
     trainer = RewardTrainer(model=rm, args=args, train_dataset=RMDS())
     out = trainer.train()
     assert out.global_step == 3
+
+
+def test_ppo_step():
+    """Two PPO steps on a tiny model: finite losses, actor weights move."""
+    import copy
+
+    from paddlenlp_amd.trl import PPOConfig, PPOTrainer, ValueHeadModel
+
+    torch.manual_seed(0)
+    actor = tiny_llama(seed=1)
+    reference = copy.deepcopy(actor)
+    critic = ValueHeadModel(copy.deepcopy(actor), hidden_size=64)
+
+    def reward_fn(prompt, response):
+        # toy reward: prefer longer non-pad responses with token diversity
+        return float(len(set(response.tolist()))) / max(1, len(response))
+
+    trainer = PPOTrainer(actor, critic, reference, reward_fn,
+                         config=PPOConfig(max_new_tokens=8, minibatch_size=2,
+                                          ppo_epochs=1))
+    before = actor.lm_head.weight.detach().clone()
+    prompts = torch.randint(3, 128, (4, 6))
+    stats1 = trainer.step(prompts)
+    stats2 = trainer.step(prompts)
+    for st in (stats1, stats2):
+        for k, v in st.items():
+            assert abs(v) < 1e6 and v == v, (k, v)  # finite
+    assert not torch.equal(before, actor.lm_head.weight)
+    # reference stayed frozen
+    for p in reference.parameters():
+        assert not p.requires_grad
