@@ -7,9 +7,17 @@
 // +8-element padded rows (≤2-way bank aliasing, free on CDNA4 per guide G4),
 // fp32 online-softmax state in registers.
 //
-// v1 structure (correctness-first; the 8-wave 32x32 swizzled "ladder"
-// structure is the planned optimization):
-//   block = 4 waves (256 thr); Q tile 64 rows (16/wave); KV tile 64.
+// v1 structure: block = 4 waves (256 thr); Q tile 64 rows (16/wave);
+// KV tile 64.  Within-probe A/B results at the bench shape
+// (B8 S4096 Hq32 Hk8 D128, tools/bench_fa.py):
+//   fwd: MF1+padded-linear ~167 TF (default).  Tried and rejected:
+//     MF2 (298 VGPR -> 1 wave/SIMD, 80 TF), XOR swizzle (codegen-unstable
+//     149-174 TF; conflicts are hidden behind the staging barrier in this
+//     2-phase structure, guide T2 gate), global_load_lds K staging
+//     (128 TF: unpadded tile re-exposes 16-way read conflicts).
+//   bwd: linear ~200 TF (default); swizzle -15%.
+// The planned next structure is the guide's 256^2-style multi-phase
+// pipeline with counted vmcnt (T3+T4), which is what unlocks T2/T5.
 //
 // MFMA fragment layouts for mfma_f32_16x16x32_bf16 (HW-validated by the
 // mfma_layout_probe kernel + tests/test_ops_gpu.py asymmetric check):
@@ -82,7 +90,7 @@ __global__ void mfma_layout_probe(const ushort_t* A, const ushort_t* B, float* C
 // forward.  Each wave owns MF x 16 q rows (MF M-frags), so one K/V staging
 // round feeds 2x the MFMA work (BLK_M = FA_WAVES * MF * 16 = 128).
 // ---------------------------------------------------------------------------
-template <int D, int MF, bool SWZ>
+template <int D, int MF, bool SWZ, bool GLL = false>
 __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const ushort_t* __restrict__ q,   // [B, Sq, Hq, D]
     const ushort_t* __restrict__ k,   // [B, Skv, Hk, D]
@@ -96,7 +104,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     constexpr int KD = D / 32;   // MFMA K-steps over the head dim
     constexpr int ND = D / 16;   // d-frags of the O accumulator
     constexpr int NN = BLK_N / 16;
-    constexpr int LDK = D + LDS_PAD;
+    // global_load_lds requires a linear (unpadded, unswizzled) K tile
+    constexpr int LDK = GLL ? D : (D + LDS_PAD);
     constexpr int LDT = BLK_N + LDS_PAD;
 
     __shared__ ushort_t k_lds[BLK_N][LDK];       // K row-major [kv][d]
@@ -158,6 +167,39 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
         const int kv_base = kvt * BLK_N;
+        if constexpr (GLL) {
+            // direct HBM -> LDS for the K tile (guide §5: the compiler never
+            // auto-emits global_load_lds; width 16 removes the VGPR staging
+            // round-trip).  Dest is wave-uniform base + lane*16.
+            for (int chunk = wave; chunk < BLK_N * D * 2 / 1024; chunk += FA_WAVES) {
+                int byte_off = chunk * 1024 + lane * 16;
+                int row = byte_off / (D * 2);
+                int col = (byte_off % (D * 2)) / 2;
+                int kvg = kv_base + row;
+                if (kvg >= Skv) kvg = Skv - 1;  // clamp: masked later, avoids NaN garbage
+                const ushort_t* src = k_ptr + (long long)kvg * kv_row_stride + col;
+                __builtin_amdgcn_global_load_lds(
+                    reinterpret_cast<const unsigned int*>(src),
+                    reinterpret_cast<unsigned int*>(&k_lds[0][0]) + byte_off / 4,
+                    16, 0, 0);
+            }
+            for (int idx = tid * 8; idx < (BLK_N / 2) * D; idx += FA_BLOCK * 8) {
+                int rr = idx / D, col = idx % D;
+                int row0 = rr * 2;
+                int kvg0 = kv_base + row0;
+                short8v v8a = {0,0,0,0,0,0,0,0}, v8b = {0,0,0,0,0,0,0,0};
+                if (kvg0 < Skv)
+                    v8a = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg0 * kv_row_stride + col);
+                if (kvg0 + 1 < Skv)
+                    v8b = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+#pragma unroll
+                for (int j = 0; j < 8; j++) {
+                    unsigned int packed = ((unsigned int)(unsigned short)v8a[j]) |
+                                          (((unsigned int)(unsigned short)v8b[j]) << 16);
+                    *reinterpret_cast<unsigned int*>(swzb<SWZ>(vt_lds, col + j, row0)) = packed;
+                }
+            }
+        } else {
         // stage K row-major (vector) + V transposed (b32-packed: two kv rows
         // per thread so the LDS transpose writes are 4-byte, halving ds ops)
         for (int idx = tid * 8; idx < (BLK_N / 2) * D; idx += FA_BLOCK * 8) {
@@ -182,6 +224,7 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
                                       (((unsigned int)(unsigned short)v8b[j]) << 16);
                 *reinterpret_cast<unsigned int*>(swzb<SWZ>(vt_lds, col + j, row0)) = packed;
             }
+        }
         }
         __syncthreads();
 
@@ -741,6 +784,14 @@ void launch_flash_fwd_variant(const void* q, const void* k, const void* v, void*
                               hipStream_t stream) {
     if (D != 128) variant = 0;
     switch (variant) {
+        case 4: {
+            constexpr int BLKM = FA_WAVES * 1 * 16;
+            dim3 grid((Sq + BLKM - 1) / BLKM, B * Hq);
+            hipLaunchKernelGGL((flash_fwd_kernel<128, 1, false, true>), grid, dim3(FA_BLOCK), 0, stream,
+                               (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                               (ushort_t*)o, lse, nullptr, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+            break;
+        }
         case 1: flash_fwd_t<128, 2, true>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
         case 2: flash_fwd_t<128, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
         case 3: flash_fwd_t<128, 2, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, causal, stream); break;
