@@ -197,21 +197,15 @@ std::vector<torch::Tensor> flash_attn_bwd_ex(torch::Tensor dout, torch::Tensor q
     int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
     int Skv = k.size(1), Hk = k.size(2);
     auto dq = torch::empty_like(q);
-    auto dk_h = torch::empty({B, Skv, Hq, D}, k.options());
-    auto dv_h = torch::empty({B, Skv, Hq, D}, v.options());
+    auto dk_h = torch::empty({B, Skv, Hk, D}, k.options());
+    auto dv_h = torch::empty({B, Skv, Hk, D}, v.options());
     auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
     float scale = 1.0f / std::sqrt((float)D);
     launch_flash_bwd_variant(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                              o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
                              dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(),
                              B, Sq, Skv, Hq, Hk, D, scale, causal, (int)variant, cur_stream());
-    torch::Tensor dk = dk_h, dv = dv_h;
-    if (Hq != Hk) {
-        int G = Hq / Hk;
-        dk = dk_h.view({B, Skv, Hk, G, D}).sum(3);
-        dv = dv_h.view({B, Skv, Hk, G, D}).sum(3);
-    }
-    return {dq, dk, dv};
+    return {dq, dk_h, dv_h};
 }
 
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
@@ -222,25 +216,15 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
     int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
     int Skv = k.size(1), Hk = k.size(2);
     auto dq = torch::empty_like(q);
-    // per-q-head buffers; summed over GQA groups below
-    auto dk_h = torch::empty({B, Skv, Hq, D}, k.options());
-    auto dv_h = torch::empty({B, Skv, Hq, D}, v.options());
+    auto dk_h = torch::empty({B, Skv, Hk, D}, k.options());
+    auto dv_h = torch::empty({B, Skv, Hk, D}, v.options());
     auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
     float scale = 1.0f / std::sqrt((float)D);
     launch_flash_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
                      dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(),
                      B, Sq, Skv, Hq, Hk, D, scale, causal, cur_stream());
-    torch::Tensor dk, dv;
-    if (Hq == Hk) {
-        dk = dk_h;
-        dv = dv_h;
-    } else {
-        int G = Hq / Hk;
-        dk = dk_h.view({B, Skv, Hk, G, D}).sum(3);
-        dv = dv_h.view({B, Skv, Hk, G, D}).sum(3);
-    }
-    return {dq, dk, dv};
+    return {dq, dk_h, dv_h};
 }
 
 // ---------------------------------------------------------------------------
@@ -308,21 +292,15 @@ std::vector<torch::Tensor> flashmask_attn_bwd(torch::Tensor dout, torch::Tensor 
     int Skv = k.size(1), Hk = k.size(2);
     auto se = startend.to(torch::kInt32).contiguous();
     auto dq = torch::empty_like(q);
-    auto dk_h = torch::empty({B, Skv, Hq, D}, k.options());
-    auto dv_h = torch::empty({B, Skv, Hq, D}, v.options());
+    auto dk_h = torch::empty({B, Skv, Hk, D}, k.options());
+    auto dv_h = torch::empty({B, Skv, Hk, D}, v.options());
     auto delta = torch::empty({B, Hq, Sq}, q.options().dtype(torch::kFloat32));
     float scale = 1.0f / std::sqrt((float)D);
     launch_flash_bwd_mask(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                           o.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
                           dq.data_ptr(), dk_h.data_ptr(), dv_h.data_ptr(),
                           se.data_ptr<int>(), B, Sq, Skv, Hq, Hk, D, scale, cur_stream());
-    torch::Tensor dk = dk_h, dv = dv_h;
-    if (Hq != Hk) {
-        int G = Hq / Hk;
-        dk = dk_h.view({B, Skv, Hk, G, D}).sum(3);
-        dv = dv_h.view({B, Skv, Hk, G, D}).sum(3);
-    }
-    return {dq, dk, dv};
+    return {dq, dk_h, dv_h};
 }
 
 // ---------------------------------------------------------------------------

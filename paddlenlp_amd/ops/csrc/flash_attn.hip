@@ -550,8 +550,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    ushort_t* __restrict__ dk,   // [B, Skv, Hq, D] (per q-head)
-    ushort_t* __restrict__ dv,   // [B, Skv, Hq, D]
+    ushort_t* __restrict__ dk,   // [B, Skv, Hk, D] (GQA group summed in-kernel)
+    ushort_t* __restrict__ dv,   // [B, Skv, Hk, D]
     const int* __restrict__ startend,
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int KD = D / 32;
@@ -573,20 +573,16 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
     const int lk8 = (lane >> 4) * 8;
 
     const int kvt = blockIdx.x;
-    const int bh = blockIdx.y;
-    const int b = bh / Hq, hq = bh % Hq;
-    const int hk = hq / (Hq / Hk);
+    const int bh = blockIdx.y;           // b * Hk + hk (one block per kv head)
+    const int b = bh / Hk, hk = bh % Hk;
+    const int G = Hq / Hk;               // q heads accumulated in-kernel
     const int kv_base = kvt * BLK_N;
     const int causal_off = Skv - Sq;
 
     const long long q_row_stride = (long long)Hq * D;
     const long long kv_row_stride = (long long)Hk * D;
-    const ushort_t* q_ptr = q + ((long long)b * Sq * Hq + hq) * D;
     const ushort_t* k_ptr = k + ((long long)b * Skv * Hk + hk) * D;
     const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
-    const ushort_t* do_ptr = dout + ((long long)b * Sq * Hq + hq) * D;
-    const float* lse_row = lse + ((long long)b * Hq + hq) * Sq;
-    const float* dl_row = delta + ((long long)b * Hq + hq) * Sq;
 
     // K and V A-frags in registers (wave's 16 kv rows, fixed all kernel)
     frag_ab ak[KD], av[KD];
@@ -621,6 +617,15 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
     }
     int n_q_tiles = (Sq + BLK_M - 1) / BLK_M;
 
+    // accumulate the whole GQA group into acc_dk/acc_dv: K/V stream from HBM
+    // once per kv head, and the outputs are written [B, Skv, Hk, D] directly
+    // (no per-q-head buffers + reduction pass)
+    for (int g = 0; g < G; g++) {
+    const int hq = hk * G + g;
+    const ushort_t* q_ptr = q + ((long long)b * Sq * Hq + hq) * D;
+    const ushort_t* do_ptr = dout + ((long long)b * Sq * Hq + hq) * D;
+    const float* lse_row = lse + ((long long)b * Hq + hq) * Sq;
+    const float* dl_row = delta + ((long long)b * Hq + hq) * Sq;
     for (int qt = qt_start; qt < n_q_tiles; qt++) {
         const int q_base = qt * BLK_M;
         for (int idx = tid * 8; idx < (BLK_M / 2) * D; idx += FA_BLOCK * 8) {
@@ -729,15 +734,16 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
         }
         __syncthreads();
     }
+    }  // g loop over the GQA group
 
-    // store per-q-head dK/dV: [B, Skv, Hq, D]
+    // store dK/dV directly in [B, Skv, Hk, D]
     const int kvrow0 = kv_base + wave * 16 + (lane >> 4) * 4;
-    const long long out_row_stride = (long long)Hq * D;
+    const long long out_row_stride = (long long)Hk * D;
 #pragma unroll
     for (int r = 0; r < 4; r++) {
         int kvg = kvrow0 + r;
         if (kvg >= Skv) continue;
-        long long base = ((long long)b * Skv + kvg) * out_row_stride + (long long)hq * D;
+        long long base = ((long long)b * Skv + kvg) * out_row_stride + (long long)hk * D;
 #pragma unroll
         for (int n = 0; n < ND; n++) {
             dk[base + n * 16 + l16] = f32_to_bf16(acc_dk[n][r]);
@@ -829,7 +835,7 @@ static void flash_bwd_t(const void* dout, const void* q, const void* k, const vo
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dq, startend,
                        B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
-    dim3 gkv((Skv + BLK_N - 1) / BLK_N, B * Hq);
+    dim3 gkv((Skv + BLK_N - 1) / BLK_N, B * Hk);
     hipLaunchKernelGGL((flash_bwd_dkv_kernel<D, SWZ>), gkv, dim3(FA_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dk, (ushort_t*)dv,
