@@ -214,13 +214,8 @@ class Trainer:
             broadcast_parameters(model, topo.data_parallel_group)
         if topo.sharding_degree > 1 and topo.sharding_parallel_group is not None:
             stage = args.sharding_stage()
-            if stage in (1, 2):
-                from ..parallel.zero import ZeroShardedEngine
-
-                # engine is attached after optimizer creation in train()
-                self._zero_stage = stage
-            else:
-                raise NotImplementedError("sharding stage3 lands in a later milestone")
+            # engine is attached after optimizer creation in train()
+            self._zero_stage = stage
             broadcast_parameters(model, topo.sharding_parallel_group)
         return model
 
@@ -252,7 +247,7 @@ class Trainer:
         self.create_optimizer_and_scheduler(max_steps)
 
         # ZeRO engine wraps the optimizer after both exist
-        if getattr(self, "_zero_stage", 0):
+        if getattr(self, "_zero_stage", 0) in (1, 2):
             from ..parallel.zero import ZeroShardedEngine
 
             self._zero = ZeroShardedEngine(
@@ -261,6 +256,11 @@ class Trainer:
                 group=self.topology.sharding_parallel_group,
                 bucket_mb=args.sharding_comm_buffer_size_MB,
             )
+        elif getattr(self, "_zero_stage", 0) == 3:
+            from ..parallel.zero3 import Zero3Engine
+
+            self._zero = Zero3Engine(
+                unwrap_model(model), group=self.topology.sharding_parallel_group)
 
         if resume_from_checkpoint and isinstance(resume_from_checkpoint, str):
             self._load_from_checkpoint(resume_from_checkpoint)
@@ -429,7 +429,10 @@ class Trainer:
 
         timers("all-reduce").stop()
         if args.max_grad_norm and args.max_grad_norm > 0:
-            self._clip_grad_norm(model)
+            if self._zero is not None and self._zero.stage == 3:
+                self._zero.clip_grads(args.max_grad_norm)
+            else:
+                self._clip_grad_norm(model)
 
         timers("optimizer-step").start()
         self.optimizer.step()
