@@ -31,7 +31,8 @@ struct AdamWChunk {
     int is_bf16;
 };
 void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, float, float, hipStream_t);
-void launch_paged_decode_attn(const void*, const void*, const void*, const int*, const int*, void*, int, int, int, int, int, int, float, hipStream_t);
+void launch_paged_decode_attn(const void*, const void*, const void*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, hipStream_t);
+int paged_decode_nsplit(int, int);
 void launch_rope_cache_append(const void*, void*, void*, void*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -315,10 +316,19 @@ torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor k_cache, torch::T
     int max_blocks = block_table.size(1);
     auto out = torch::empty_like(q);
     float scale = 1.0f / std::sqrt((float)D);
+    int G = Hq / Hk;
+    int nsplit = paged_decode_nsplit(B, Hk);
+    torch::Tensor partials;
+    float* pptr = nullptr;
+    if (nsplit > 1) {
+        partials = torch::empty({(long)B, (long)Hk, (long)nsplit, (long)G * (2 + D)},
+                                q.options().dtype(torch::kFloat32));
+        pptr = partials.data_ptr<float>();
+    }
     launch_paged_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
                              block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                             out.data_ptr(), B, Hq, Hk, D, block_size, max_blocks,
-                             scale, cur_stream());
+                             out.data_ptr(), pptr, nsplit, B, Hq, Hk, D, block_size,
+                             max_blocks, scale, cur_stream());
     return out;
 }
 

@@ -20,6 +20,9 @@
 // block_table [B, max_blocks] int32, seq_lens [B] int32 (kv length incl.
 // the current token), out [B, Hq, D].
 // ---------------------------------------------------------------------------
+// With NSPLIT > 1 the kv range is chunked across blockIdx.z so small-batch
+// decode fills the chip; each split writes fp32 partials
+// partial[b, hk, split] = {m[G], l[G], acc[G][D]} merged by the kernel below.
 template <int D>
 __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
     const ushort_t* __restrict__ q,
@@ -28,9 +31,12 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
     const int* __restrict__ block_table,   // [B, max_blocks]
     const int* __restrict__ seq_lens,      // [B]
     ushort_t* __restrict__ out,
-    int B, int Hq, int Hk, int block_size, int max_blocks, float scale) {
+    float* __restrict__ partials,          // [B, Hk, NSPLIT, G*(2+D)] or null
+    int B, int Hq, int Hk, int block_size, int max_blocks, float scale,
+    int nsplit) {
     const int b = blockIdx.x;
     const int hk = blockIdx.y;
+    const int split = blockIdx.z;
     const int G = Hq / Hk;
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
@@ -60,10 +66,13 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
         for (int e = 0; e < EPL; e++) acc[g][e] = 0.f;
     }
 
-    // waves split the kv range into contiguous chunks
-    const int per_wave = (seq_len + PA_WAVES - 1) / PA_WAVES;
-    const int j0 = wave * per_wave;
-    const int j1 = min(seq_len, j0 + per_wave);
+    // split chunk, then waves split the chunk into contiguous sub-ranges
+    const int chunk = (seq_len + nsplit - 1) / nsplit;
+    const int c0 = split * chunk;
+    const int c1 = min(seq_len, c0 + chunk);
+    const int per_wave = (c1 - c0 + PA_WAVES - 1) / PA_WAVES;
+    const int j0 = c0 + wave * per_wave;
+    const int j1 = min(c1, j0 + per_wave);
     const int* bt = block_table + (long long)b * max_blocks;
 
     for (int j = j0; j < j1; j++) {
@@ -129,13 +138,63 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
                 for (int e = 0; e < EPL; e++)
                     oacc[e] += s_acc[w][g][lane * EPL + e] * a;
             }
-            float inv = (gl > 0.f) ? 1.0f / gl : 0.f;
-            ushort_t* op = out + ((long long)b * Hq + hq0 + g) * D;
+            if (nsplit == 1) {
+                float inv = (gl > 0.f) ? 1.0f / gl : 0.f;
+                ushort_t* op = out + ((long long)b * Hq + hq0 + g) * D;
 #pragma unroll
-            for (int e = 0; e < EPL; e++)
-                op[lane * EPL + e] = f32_to_bf16(oacc[e] * inv);
+                for (int e = 0; e < EPL; e++)
+                    op[lane * EPL + e] = f32_to_bf16(oacc[e] * inv);
+            } else {
+                float* pp = partials +
+                    (((long long)b * Hk + hk) * nsplit + split) * (long long)G * (2 + D)
+                    + (long long)g * (2 + D);
+                if (lane == 0) { pp[0] = gm; pp[1] = gl; }
+#pragma unroll
+                for (int e = 0; e < EPL; e++)
+                    pp[2 + lane * EPL + e] = oacc[e];
+            }
         }
     }
+}
+
+// merge the per-split partials: one wave per (b, hk, g)
+template <int D>
+__global__ void paged_decode_merge_kernel(
+    const float* __restrict__ partials, ushort_t* __restrict__ out,
+    const int* __restrict__ seq_lens,
+    int B, int Hq, int Hk, int nsplit) {
+    const int b = blockIdx.x;
+    const int hk = blockIdx.y;
+    const int g = blockIdx.z;
+    const int G = Hq / Hk;
+    const int lane = threadIdx.x & 63;
+    constexpr int EPL = D / 64;
+    const int seq_len = seq_lens[b];
+    const int chunk = (seq_len + nsplit - 1) / nsplit;
+    const float* base = partials +
+        (((long long)b * Hk + hk) * nsplit) * (long long)G * (2 + D)
+        + (long long)g * (2 + D);
+    float gm = -INFINITY;
+    for (int sp = 0; sp < nsplit; sp++) {
+        if (sp * chunk >= seq_len) break;
+        gm = fmaxf(gm, base[(long long)sp * G * (2 + D)]);
+    }
+    float gl = 0.f;
+    float oacc[EPL] = {0.f};
+    for (int sp = 0; sp < nsplit; sp++) {
+        if (sp * chunk >= seq_len) break;
+        const float* pp = base + (long long)sp * G * (2 + D);
+        float a = (pp[0] == -INFINITY) ? 0.f : __expf(pp[0] - gm);
+        gl += pp[1] * a;
+#pragma unroll
+        for (int e = 0; e < EPL; e++)
+            oacc[e] += pp[2 + lane * EPL + e] * a;
+    }
+    float inv = (gl > 0.f) ? 1.0f / gl : 0.f;
+    ushort_t* op = out + ((long long)b * Hq + hk * G + g) * D;
+#pragma unroll
+    for (int e = 0; e < EPL; e++)
+        op[lane * EPL + e] = f32_to_bf16(oacc[e] * inv);
 }
 
 // ---------------------------------------------------------------------------
@@ -220,21 +279,39 @@ __global__ void rope_cache_append_kernel(
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
+int paged_decode_nsplit(int B, int Hk) {
+    // fill ~2x the 256 CUs; cap so chunks stay useful
+    int target = 512;
+    int base = B * Hk;
+    int nsplit = (target + base - 1) / base;
+    if (nsplit < 1) nsplit = 1;
+    if (nsplit > 64) nsplit = 64;
+    return nsplit;
+}
+
 void launch_paged_decode_attn(const void* q, const void* k_cache, const void* v_cache,
                               const int* block_table, const int* seq_lens, void* out,
+                              float* partials, int nsplit,
                               int B, int Hq, int Hk, int D, int block_size,
                               int max_blocks, float scale, hipStream_t stream) {
-    dim3 grid(B, Hk);
-    if (D == 128)
+    dim3 grid(B, Hk, nsplit);
+    if (D == 128) {
         hipLaunchKernelGGL(paged_decode_attn_kernel<128>, grid, dim3(PA_BLOCK), 0, stream,
                            (const ushort_t*)q, (const ushort_t*)k_cache, (const ushort_t*)v_cache,
-                           block_table, seq_lens, (ushort_t*)out,
-                           B, Hq, Hk, block_size, max_blocks, scale);
-    else if (D == 64)
+                           block_table, seq_lens, (ushort_t*)out, partials,
+                           B, Hq, Hk, block_size, max_blocks, scale, nsplit);
+        if (nsplit > 1)
+            hipLaunchKernelGGL(paged_decode_merge_kernel<128>, dim3(B, Hk, Hq / Hk), dim3(64),
+                               0, stream, partials, (ushort_t*)out, seq_lens, B, Hq, Hk, nsplit);
+    } else if (D == 64) {
         hipLaunchKernelGGL(paged_decode_attn_kernel<64>, grid, dim3(PA_BLOCK), 0, stream,
                            (const ushort_t*)q, (const ushort_t*)k_cache, (const ushort_t*)v_cache,
-                           block_table, seq_lens, (ushort_t*)out,
-                           B, Hq, Hk, block_size, max_blocks, scale);
+                           block_table, seq_lens, (ushort_t*)out, partials,
+                           B, Hq, Hk, block_size, max_blocks, scale, nsplit);
+        if (nsplit > 1)
+            hipLaunchKernelGGL(paged_decode_merge_kernel<64>, dim3(B, Hk, Hq / Hk), dim3(64),
+                               0, stream, partials, (ushort_t*)out, seq_lens, B, Hq, Hk, nsplit);
+    }
 }
 
 void launch_rope_cache_append(const void* qkv, void* q_out, void* k_cache, void* v_cache,
