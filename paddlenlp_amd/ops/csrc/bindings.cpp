@@ -33,6 +33,7 @@ struct AdamWChunk {
 void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, float, float, hipStream_t);
 void launch_paged_decode_attn(const void*, const void*, const void*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, hipStream_t);
 int paged_decode_nsplit(int, int);
+void launch_wint8_gemv(const void*, const void*, const float*, void*, int, int, int, hipStream_t);
 void launch_rope_cache_append(const void*, void*, void*, void*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -359,6 +360,20 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     return q_out;
 }
 
+torch::Tensor wint8_gemv(torch::Tensor x, torch::Tensor wq, torch::Tensor scale) {
+    CHECK_GPU(x); CHECK_BF16(x);
+    auto x2 = x.contiguous().view({-1, x.size(-1)});
+    int M = x2.size(0), K = x2.size(1), N = wq.size(0);
+    TORCH_CHECK(M <= 16, "wint8_gemv is for decode shapes (M <= 16)");
+    TORCH_CHECK(K % 128 == 0, "K must be a multiple of 128");
+    auto sizes = x.sizes().vec();
+    sizes.back() = N;
+    auto y = torch::empty(sizes, x.options());
+    launch_wint8_gemv(x2.data_ptr(), wq.data_ptr(), scale.data_ptr<float>(),
+                      y.data_ptr(), M, N, K, cur_stream());
+    return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rms_norm_fwd", &rms_norm_fwd);
     m.def("rms_norm_bwd", &rms_norm_bwd);
@@ -377,4 +392,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_adamw", &fused_adamw);
     m.def("paged_decode_attn", &paged_decode_attn);
     m.def("rope_cache_append", &rope_cache_append);
+    m.def("wint8_gemv", &wint8_gemv);
 }

@@ -155,3 +155,17 @@ def test_engine_fp8_decode_gpu(C):
     rel = (logits - ref.float()).abs().max() / ref.float().abs().max()
     assert torch.isfinite(logits).all()
     assert rel < 0.3, rel
+
+
+def test_wint8_gemv_kernel(C):
+    from paddlenlp_amd.quantization import quantize_int8
+
+    torch.manual_seed(0)
+    for M in (1, 3, 8, 16):
+        x = torch.randn(M, 512, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(384, 512, device="cuda", dtype=torch.bfloat16)
+        q, sc = quantize_int8(w)
+        y = C.wint8_gemv(x, q, sc)
+        ref = x.float() @ (q.float() * sc[:, None]).t()
+        rel = (y.float() - ref).abs().max() / ref.abs().max()
+        assert rel < 0.02, (M, rel)
