@@ -53,21 +53,17 @@ def weight_only_linear(x: torch.Tensor, qweight: torch.Tensor, scale: torch.Tens
         else:
             y = x2 @ (qweight.float() * scale).t().to(x.dtype)
         y = y.reshape(*orig_shape[:-1], -1)
-    else:
-        use_kernel = (
-            x.is_cuda and x.dtype == torch.bfloat16
-            and x.numel() // x.shape[-1] <= 16 and x.shape[-1] % 128 == 0
-        )
-        if use_kernel:
-            # fused dequant GEMV: streams int8 weights once (half the bf16
-            # traffic) instead of materializing a dequantized copy
-            from ..ops.functional import _load_extension
+    elif algo == "weight_only_int8_gemv":
+        # experimental fused dequant GEMV (ops/csrc/wint8_gemv.hip): correct
+        # but currently VALU-bound (per-FMA LDS reads in the M loop) and
+        # slower than dequant+hipBLASLt — kept for the register-tiled rework
+        from ..ops.functional import _load_extension
 
-            C = _load_extension()
-            y = C.wint8_gemv(x, qweight, scale.float())
-        else:
-            w = (qweight.float() * scale[:, None]).to(x.dtype)
-            y = x @ w.t()
+        C = _load_extension()
+        y = C.wint8_gemv(x, qweight, scale.float())
+    else:
+        w = (qweight.float() * scale[:, None]).to(x.dtype)
+        y = x @ w.t()
     if bias is not None:
         y = y + bias
     return y
