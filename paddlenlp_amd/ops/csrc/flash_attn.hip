@@ -337,6 +337,222 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// forward, software-pipelined (guide T14 async-stage split): the next KV
+// tile's global loads are ISSUED before this tile's PV MFMAs so the ~500-
+// cycle HBM latency hides under compute; the register-staged data is written
+// to LDS after the barrier.  Same barrier count per tile as the base kernel.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(FA_BLOCK) void flash_fwd_pipe_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
+    const ushort_t* __restrict__ v, ushort_t* __restrict__ o,
+    float* __restrict__ lse,
+    int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
+    constexpr int BLKM = FA_WAVES * 16;
+    constexpr int KD = D / 32;
+    constexpr int ND = D / 16;
+    constexpr int NN = BLK_N / 16;
+    constexpr int LDK = D + LDS_PAD;
+    constexpr int LDT = BLK_N + LDS_PAD;
+    constexpr int ITERS = (BLK_N / 2) * D / (FA_BLOCK * 8);  // reg-staged chunks/thread
+
+    __shared__ ushort_t k_lds[BLK_N][LDK];
+    __shared__ ushort_t vt_lds[D][LDT];
+    __shared__ ushort_t p_lds[BLKM][LDT];
+
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int l16 = lane & 15;
+    const int lk8 = (lane >> 4) * 8;
+
+    const int qt = blockIdx.x;
+    const int bh = blockIdx.y;
+    const int b = bh / Hq, hq = bh % Hq;
+    const int hk = hq / (Hq / Hk);
+    const int q_base = qt * BLKM;
+    const int causal_off = Skv - Sq;
+
+    const long long q_row_stride = (long long)Hq * D;
+    const long long kv_row_stride = (long long)Hk * D;
+    const ushort_t* q_ptr = q + ((long long)b * Sq * Hq + hq) * D;
+    const ushort_t* k_ptr = k + ((long long)b * Skv * Hk + hk) * D;
+    const ushort_t* v_ptr = v + ((long long)b * Skv * Hk + hk) * D;
+
+    frag_ab aq[KD];
+    {
+        int qrow = q_base + wave * 16 + l16;
+        if (qrow < Sq) {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++)
+                aq[kk] = *reinterpret_cast<const frag_ab*>(
+                    q_ptr + (long long)qrow * q_row_stride + kk * 32 + lk8);
+        } else {
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) aq[kk] = frag_ab{0};
+        }
+    }
+
+    float m_run[4], l_run[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+    frag_c acc_o[ND];
+#pragma unroll
+    for (int n = 0; n < ND; n++) acc_o[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+
+    int n_kv_tiles = (Skv + BLK_N - 1) / BLK_N;
+    if (causal) {
+        int max_kv = q_base + BLKM - 1 + causal_off;
+        int lim = (max_kv + BLK_N) / BLK_N;
+        n_kv_tiles = min(n_kv_tiles, max(lim, 0));
+    }
+
+    // register staging buffers: this thread's chunks of the NEXT tile
+    short8v sk[ITERS][2], sv[ITERS][2];
+
+    auto load_tile = [&](int kv_base) {
+#pragma unroll
+        for (int it = 0; it < ITERS; it++) {
+            int idx = (tid + it * FA_BLOCK) * 8;
+            int rr = idx / D, col = idx % D;
+            int row0 = rr * 2;
+            int kvg0 = kv_base + row0;
+            sk[it][0] = short8v{0,0,0,0,0,0,0,0}; sk[it][1] = short8v{0,0,0,0,0,0,0,0};
+            sv[it][0] = short8v{0,0,0,0,0,0,0,0}; sv[it][1] = short8v{0,0,0,0,0,0,0,0};
+            if (kvg0 < Skv) {
+                sk[it][0] = *reinterpret_cast<const short8v*>(k_ptr + (long long)kvg0 * kv_row_stride + col);
+                sv[it][0] = *reinterpret_cast<const short8v*>(v_ptr + (long long)kvg0 * kv_row_stride + col);
+            }
+            if (kvg0 + 1 < Skv) {
+                sk[it][1] = *reinterpret_cast<const short8v*>(k_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+                sv[it][1] = *reinterpret_cast<const short8v*>(v_ptr + (long long)(kvg0 + 1) * kv_row_stride + col);
+            }
+        }
+    };
+
+    auto write_tile = [&]() {
+#pragma unroll
+        for (int it = 0; it < ITERS; it++) {
+            int idx = (tid + it * FA_BLOCK) * 8;
+            int rr = idx / D, col = idx % D;
+            int row0 = rr * 2;
+            *reinterpret_cast<short8v*>(&k_lds[row0][col]) = sk[it][0];
+            *reinterpret_cast<short8v*>(&k_lds[row0 + 1][col]) = sk[it][1];
+#pragma unroll
+            for (int j = 0; j < 8; j++) {
+                unsigned int packed = ((unsigned int)(unsigned short)sv[it][0][j]) |
+                                      (((unsigned int)(unsigned short)sv[it][1][j]) << 16);
+                *reinterpret_cast<unsigned int*>(&vt_lds[col + j][row0]) = packed;
+            }
+        }
+    };
+
+    // prologue: tile 0 into LDS
+    load_tile(0);
+    write_tile();
+    __syncthreads();
+
+    for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
+        const int kv_base = kvt * BLK_N;
+
+        frag_c acc_s[NN];
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            acc_s[n] = frag_c{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kk = 0; kk < KD; kk++) {
+                frag_ab bk = *reinterpret_cast<const frag_ab*>(&k_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_s[n] = mfma16(aq[kk], bk, acc_s[n]);
+            }
+        }
+
+        // issue the NEXT tile's global loads now: their latency hides under
+        // the softmax + PV work below (guide T14)
+        if (kvt + 1 < n_kv_tiles) load_tile(kv_base + BLK_N);
+
+        const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
+#pragma unroll
+        for (int n = 0; n < NN; n++) {
+            int kvg = kv_base + n * 16 + l16;
+#pragma unroll
+            for (int r = 0; r < 4; r++) {
+                int qg = qrow0 + r;
+                bool vis = (kvg < Skv) && (qg < Sq);
+                if (causal) vis = vis && (kvg <= qg + causal_off);
+                acc_s[n][r] = vis ? acc_s[n][r] * scale : -INFINITY;
+            }
+        }
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            float row_max = -INFINITY;
+#pragma unroll
+            for (int n = 0; n < NN; n++) row_max = fmaxf(row_max, acc_s[n][r]);
+#pragma unroll
+            for (int off = 8; off > 0; off >>= 1)
+                row_max = fmaxf(row_max, __shfl_xor(row_max, off, 64));
+            float m_new = fmaxf(m_run[r], row_max);
+            float alpha;
+            if (m_new == -INFINITY) alpha = 1.f;
+            else if (m_run[r] == -INFINITY) alpha = 0.f;
+            else alpha = __expf(m_run[r] - m_new);
+            float row_sum = 0.f;
+#pragma unroll
+            for (int n = 0; n < NN; n++) {
+                float pp = (m_new == -INFINITY || acc_s[n][r] == -INFINITY)
+                               ? 0.f : __expf(acc_s[n][r] - m_new);
+                acc_s[n][r] = pp;
+                row_sum += pp;
+            }
+#pragma unroll
+            for (int off = 8; off > 0; off >>= 1) row_sum += __shfl_xor(row_sum, off, 64);
+            l_run[r] = l_run[r] * alpha + row_sum;
+            m_run[r] = m_new;
+#pragma unroll
+            for (int n = 0; n < ND; n++) acc_o[n][r] *= alpha;
+        }
+#pragma unroll
+        for (int n = 0; n < NN; n++)
+#pragma unroll
+            for (int r = 0; r < 4; r++)
+                p_lds[wave * 16 + (lane >> 4) * 4 + r][n * 16 + l16] =
+                    f32_to_bf16(acc_s[n][r]);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+#pragma unroll
+        for (int kk = 0; kk < BLK_N / 32; kk++) {
+            frag_ab ap = *reinterpret_cast<const frag_ab*>(&p_lds[wave * 16 + l16][kk * 32 + lk8]);
+#pragma unroll
+            for (int n = 0; n < ND; n++) {
+                frag_ab bv = *reinterpret_cast<const frag_ab*>(&vt_lds[n * 16 + l16][kk * 32 + lk8]);
+                acc_o[n] = mfma16(ap, bv, acc_o[n]);
+            }
+        }
+        __syncthreads();      // everyone done reading k_lds / vt_lds
+        if (kvt + 1 < n_kv_tiles) {
+            write_tile();      // staged regs -> LDS for the next iteration
+            __syncthreads();
+        }
+    }
+
+    const int qrow0 = q_base + wave * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+        int qg = qrow0 + r;
+        if (qg >= Sq) continue;
+        float inv_l = (l_run[r] > 0.f) ? 1.0f / l_run[r] : 0.f;
+        ushort_t* orow = o + ((long long)b * Sq + qg) * q_row_stride + (long long)hq * D;
+#pragma unroll
+        for (int n = 0; n < ND; n++)
+            orow[n * 16 + l16] = f32_to_bf16(acc_o[n][r] * inv_l);
+        if (l16 == 0) {
+            float lv = (l_run[r] > 0.f) ? (m_run[r] + __logf(l_run[r])) : -INFINITY;
+            lse[((long long)b * Hq + hq) * Sq + qg] = lv;
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // backward preprocess: delta[b,h,q] = rowsum(dO * O)
 // ---------------------------------------------------------------------------
@@ -790,6 +1006,14 @@ void launch_flash_fwd_variant(const void* q, const void* k, const void* v, void*
                               hipStream_t stream) {
     if (D != 128) variant = 0;
     switch (variant) {
+        case 5: {
+            constexpr int BLKM = FA_WAVES * 16;
+            dim3 grid((Sq + BLKM - 1) / BLKM, B * Hq);
+            hipLaunchKernelGGL((flash_fwd_pipe_kernel<128>), grid, dim3(FA_BLOCK), 0, stream,
+                               (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                               (ushort_t*)o, lse, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+            break;
+        }
         case 4: {
             constexpr int BLKM = FA_WAVES * 1 * 16;
             dim3 grid((Sq + BLKM - 1) / BLKM, B * Hq);

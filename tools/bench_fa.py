@@ -14,7 +14,7 @@ import torch
 
 from paddlenlp_amd.ops.functional import _load_extension
 
-VARIANTS = {0: "default", 1: "MF2+swz", 2: "MF1+linear", 3: "MF2+linear", 4: "MF1+gll"}
+VARIANTS = {0: "default", 2: "MF1+linear", 4: "MF1+gll", 5: "MF1+pipe"}
 
 
 def main():
