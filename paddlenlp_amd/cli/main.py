@@ -6,8 +6,6 @@ no network in this environment).
 """
 from __future__ import annotations
 
-import json
-import os
 import sys
 
 try:
@@ -16,7 +14,6 @@ except ImportError:  # pragma: no cover
     typer = None
 
 from ..transformers.auto.registry import MODEL_REGISTRY
-from ..utils.log import logger
 
 
 def _build_app():

@@ -16,7 +16,7 @@ import numpy as np
 import torch
 
 from ..utils.log import logger
-from .indexed_dataset import MMapIndexedDataset, make_indexed_dataset
+from .indexed_dataset import make_indexed_dataset
 
 
 def get_train_valid_test_split_(splits_string: str, size: int) -> List[int]:

@@ -8,7 +8,7 @@ receive the batch too (first stage uses input_ids, last uses labels).
 """
 from __future__ import annotations
 
-from typing import Iterator, Optional
+from typing import Iterator
 
 import torch
 import torch.distributed as dist

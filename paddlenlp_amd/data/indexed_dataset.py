@@ -12,10 +12,7 @@ load directly:
 """
 from __future__ import annotations
 
-import os
-import shutil
 import struct
-from functools import lru_cache
 from typing import List, Optional
 
 import numpy as np

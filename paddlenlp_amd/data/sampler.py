@@ -6,7 +6,7 @@ Reference: paddlenlp/data/sampler.py + paddlenlp/utils/batch_sampler.py
 from __future__ import annotations
 
 import math
-from typing import Iterator, List, Optional
+from typing import Iterator, List
 
 import numpy as np
 import torch

@@ -7,7 +7,7 @@ from __future__ import annotations
 
 import json
 import os
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, List
 
 import torch
 

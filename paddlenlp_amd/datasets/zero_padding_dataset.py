@@ -7,7 +7,7 @@ between packed samples is expressed as `attn_mask_startend_row_indices`
 """
 from __future__ import annotations
 
-from typing import Dict, Iterable, List
+from typing import List
 
 import numpy as np
 import torch

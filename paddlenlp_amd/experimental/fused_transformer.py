@@ -17,7 +17,6 @@ from __future__ import annotations
 
 import math
 from dataclasses import dataclass
-from typing import List, Optional, Tuple
 
 import torch
 import torch.nn as nn
@@ -25,7 +24,6 @@ import torch.nn.functional as F
 
 from .. import ops
 from ..ops import reference
-from ..utils.log import logger
 
 
 @dataclass

@@ -6,7 +6,7 @@ generate :609 dispatching to greedy_search :1036, sample :1137, beam_search
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

@@ -8,7 +8,7 @@ tests run without a GPU.
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

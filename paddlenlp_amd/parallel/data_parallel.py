@@ -11,7 +11,7 @@ big enough to hit link peak while bounding the staging memory.
 """
 from __future__ import annotations
 
-from typing import Iterable, List, Optional
+from typing import Iterable, List
 
 import torch
 import torch.distributed as dist

@@ -13,12 +13,11 @@ fabric — each of the 7 links carries exactly one peer's shard.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List
 
 import torch
 import torch.distributed as dist
 
-from .topology import get_topology
 
 
 class _AllToAllVar(torch.autograd.Function):

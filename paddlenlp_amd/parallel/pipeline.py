@@ -14,8 +14,7 @@ micro-batch.  Non-interleaved 1F1B; virtual stages are a planned extension.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

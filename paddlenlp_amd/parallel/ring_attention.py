@@ -20,7 +20,6 @@ balanced chunking (chunks i and 2cp-1-i) is a planned refinement.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
 
 import torch
 import torch.distributed as dist

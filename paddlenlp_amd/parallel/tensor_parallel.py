@@ -14,8 +14,6 @@ direct links.
 """
 from __future__ import annotations
 
-import math
-from typing import Optional
 
 import torch
 import torch.distributed as dist

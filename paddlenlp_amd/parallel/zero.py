@@ -20,7 +20,7 @@ stage-2 memory savings for free.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict
 
 import torch
 import torch.distributed as dist

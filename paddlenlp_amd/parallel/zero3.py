@@ -20,7 +20,6 @@ refinement.
 """
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional
 
 import torch

@@ -7,7 +7,6 @@ ColumnParallelLoRALinear/RowParallelLoRALinear with explicit collectives
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.nn as nn

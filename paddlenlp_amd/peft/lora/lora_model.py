@@ -7,13 +7,11 @@ rslora, lora_plus_scale).
 from __future__ import annotations
 
 import json
-import math
 import os
 import re
 from dataclasses import asdict, dataclass, field
-from typing import List, Optional
+from typing import List
 
-import torch
 import torch.nn as nn
 
 from ...parallel.tensor_parallel import ColumnParallelLinear, RowParallelLinear

@@ -4,7 +4,7 @@ Reference behavior: paddlenlp/server (SimpleServer FastAPI wrapper).
 """
 from __future__ import annotations
 
-from typing import Callable, Optional
+from typing import Callable
 
 from ..utils.log import logger
 

@@ -8,13 +8,12 @@ registered with informative errors until their task models are ported.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch
 
 from ..generation import GenerationConfig
 from ..transformers import AutoModelForCausalLM, AutoTokenizer
-from ..utils.log import logger
 
 
 class _GenerationTask:

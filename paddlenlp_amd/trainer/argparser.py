@@ -11,7 +11,7 @@ import sys
 from argparse import ArgumentParser
 from enum import Enum
 from pathlib import Path
-from typing import Any, Iterable, List, Optional, Tuple, Union, get_args, get_origin
+from typing import List, Tuple, Union, get_args, get_origin
 
 
 def _str2bool(v):

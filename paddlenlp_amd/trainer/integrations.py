@@ -4,7 +4,6 @@ from __future__ import annotations
 
 import json
 import os
-from typing import Optional
 
 from ..utils.log import logger
 from .trainer_callback import TrainerCallback
@@ -12,7 +11,6 @@ from .trainer_callback import TrainerCallback
 
 def is_tensorboard_available() -> bool:
     try:
-        import torch.utils.tensorboard  # noqa: F401
 
         return True
     except ImportError:

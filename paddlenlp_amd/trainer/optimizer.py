@@ -10,7 +10,6 @@ Gradients arrive in the params' dtype (bf16); the kernel reads them as fp32.
 """
 from __future__ import annotations
 
-from typing import List, Optional
 
 import torch
 from torch.optim import Optimizer

@@ -11,7 +11,6 @@ fp32 master weights in FusedAdamW.
 """
 from __future__ import annotations
 
-import contextlib
 import math
 import os
 import shutil
@@ -26,7 +25,6 @@ from torch.utils.data import DataLoader, Dataset
 from ..data import default_data_collator
 from ..data.sampler import DistributedBatchSampler
 from ..parallel.data_parallel import broadcast_parameters, fused_allreduce_gradients
-from ..parallel.topology import get_topology
 from ..transformers.model_utils import PretrainedModel, unwrap_model
 from ..utils.env import (
     OPTIMIZER_STATE_NAME,

@@ -17,7 +17,7 @@ import torch
 
 from ..parallel.topology import Topology, get_topology, init_parallel_env
 from ..utils.log import logger
-from .trainer_utils import IntervalStrategy, ShardingOption
+from .trainer_utils import ShardingOption
 
 
 @dataclass

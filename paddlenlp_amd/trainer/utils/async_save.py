@@ -8,7 +8,6 @@ step) and a worker thread writes the files, so the training loop never waits
 on filesystem IO."""
 from __future__ import annotations
 
-import os
 import queue
 import threading
 from typing import Dict, Optional

@@ -2,7 +2,6 @@
 distributed_concat, broadcast_dp_optimizer :233, nested_* utilities)."""
 from __future__ import annotations
 
-from typing import Any, List, Optional
 
 import torch
 import torch.distributed as dist

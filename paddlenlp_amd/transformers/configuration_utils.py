@@ -12,8 +12,8 @@ from __future__ import annotations
 import copy
 import json
 import os
-from dataclasses import dataclass, fields
-from typing import Any, Dict, Optional
+from dataclasses import dataclass
+from typing import Any, Dict
 
 from ..utils.env import CONFIG_NAME
 from ..utils.log import logger

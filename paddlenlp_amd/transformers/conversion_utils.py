@@ -11,10 +11,8 @@ HuggingFace Llama/Qwen2/Mistral checkpoints load directly.
 """
 from __future__ import annotations
 
-import json
 import os
-import re
-from typing import Dict, List
+from typing import Dict
 
 import torch
 

@@ -6,7 +6,6 @@ flash-attention seam as Llama ([B, S, H, D] layout).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
 
 import torch
 import torch.nn as nn

@@ -16,17 +16,10 @@ from typing import Optional, Tuple
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 from torch.utils.checkpoint import checkpoint
 
 from ... import ops
-from ...parallel.tensor_parallel import (
-    ColumnParallelLinear,
-    ParallelCrossEntropy,
-    RowParallelLinear,
-    VocabParallelEmbedding,
-    parallel_matmul,
-)
+from ...parallel.tensor_parallel import ColumnParallelLinear, ParallelCrossEntropy, RowParallelLinear, VocabParallelEmbedding
 from ...parallel.topology import get_topology
 from ...generation import GenerationConfig, GenerationMixin
 from ..model_utils import PretrainedModel

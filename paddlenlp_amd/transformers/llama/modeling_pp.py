@@ -7,21 +7,13 @@ embeddings :359-385, seg_method "layer:LlamaDecoderLayer" :391-393).
 """
 from __future__ import annotations
 
-from typing import Optional
 
-import torch
 import torch.nn as nn
 
-from ...parallel.pipeline import LayerDesc, PipelineModule, SharedLayerDesc
+from ...parallel.pipeline import LayerDesc, PipelineModule
 from ...parallel.topology import get_topology
 from .configuration import LlamaConfig
-from .modeling import (
-    LlamaDecoderLayer,
-    LlamaPretrainedModel,
-    LlamaPretrainingCriterion,
-    LlamaRMSNorm,
-    _Linear,
-)
+from .modeling import LlamaDecoderLayer, LlamaPretrainingCriterion, LlamaRMSNorm, _Linear
 from ...parallel.tensor_parallel import ColumnParallelLinear, VocabParallelEmbedding
 
 __all__ = ["LlamaForCausalLMPipe"]

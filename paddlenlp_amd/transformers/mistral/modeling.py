@@ -7,7 +7,6 @@ sequence lengths).
 """
 from __future__ import annotations
 
-import torch.nn as nn
 
 from ...generation import GenerationConfig, GenerationMixin
 from ..llama.modeling import (

@@ -10,7 +10,6 @@ gradient all-reduce skips them (reference trainer.py:1079-1085).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
 
 import torch
 import torch.nn as nn

@@ -12,7 +12,6 @@ from __future__ import annotations
 
 import gc
 import json
-import math
 import os
 import re
 from typing import Dict, Optional
@@ -20,11 +19,7 @@ from typing import Dict, Optional
 import torch
 import torch.nn as nn
 
-from ..utils.env import (
-    CONFIG_NAME,
-    SAFE_WEIGHTS_INDEX_NAME,
-    SAFE_WEIGHTS_NAME,
-)
+from ..utils.env import SAFE_WEIGHTS_INDEX_NAME, SAFE_WEIGHTS_NAME
 from ..utils.log import logger
 from .configuration_utils import PretrainedConfig
 

@@ -6,7 +6,6 @@ differ, handled by a config-driven subclass.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 
 from ...generation import GenerationConfig, GenerationMixin
@@ -19,7 +18,6 @@ from ..llama.modeling import (
     _Linear,
     _linear_classes,
 )
-from ..model_utils import PretrainedModel
 from .configuration import Qwen2Config
 
 __all__ = ["Qwen2Model", "Qwen2ForCausalLM"]

@@ -12,8 +12,6 @@ from __future__ import annotations
 import torch
 import torch.nn.functional as F
 
-from .. import ops
-from ..ops import reference
 
 
 class FusedHeadAndCrossEntropy(torch.autograd.Function):

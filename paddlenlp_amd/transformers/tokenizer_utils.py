@@ -14,7 +14,6 @@ import os
 from typing import Dict, List, Optional, Union
 
 from ..utils.env import TOKENIZER_CONFIG_NAME
-from ..utils.log import logger
 
 __all__ = ["PretrainedTokenizer"]
 

@@ -7,10 +7,8 @@ reference model (or reference-free loss types).
 from __future__ import annotations
 
 import copy
-from typing import Dict, Optional
 
 import torch
-import torch.nn as nn
 
 from ..trainer.trainer import Trainer
 from ..utils.log import logger

@@ -10,15 +10,12 @@ the update steps use the framework's FusedAdamW.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Callable, List, Optional
+from typing import Callable, Optional
 
 import torch
-import torch.nn.functional as F
 
 from ..generation import GenerationConfig
 from ..trainer.optimizer import FusedAdamW
-from ..utils.log import logger
-from .dpo_trainer import sequence_logprob
 
 
 @dataclass
