@@ -74,6 +74,21 @@ def main():
     torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
 
+    # Pre-tuned hipBLASLt/rocBLAS GEMM algo selections for the bench shapes
+    # (tools/tune_gemms.py).  Loaded read-only: no runtime tuning cost.
+    tuned_csv = os.environ.get(
+        "PNLP_TUNABLEOP_CSV",
+        os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                     "paddlenlp_amd", "ops", "tunableop_gfx950.csv"))
+    if os.path.exists(tuned_csv) and os.environ.get("PNLP_TUNABLEOP", "1") == "1":
+        import torch.cuda.tunable as tunable
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.read_file(tuned_csv)
+        if local_rank == 0:
+            print(f"[bench] loaded {len(tunable.get_results())} tuned GEMM "
+                  f"algos from {tuned_csv}", file=sys.stderr)
+
     # data-parallel + ZeRO stage2 over the sharding axis (reference recipe)
     use_sharding = world_size > 1
     topo = init_parallel_env(
