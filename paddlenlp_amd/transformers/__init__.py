@@ -9,3 +9,40 @@ from .llama import (  # noqa: F401
     LlamaPretrainingCriterion,
 )
 from .gpt import GPTConfig, GPTForCausalLM, GPTModel  # noqa: F401
+from .bert import (  # noqa: F401
+    BertConfig,
+    BertForMaskedLM,
+    BertForMultipleChoice,
+    BertForPretraining,
+    BertForQuestionAnswering,
+    BertForSequenceClassification,
+    BertForTokenClassification,
+    BertModel,
+)
+from .ernie import (  # noqa: F401
+    UIE,
+    ErnieConfig,
+    ErnieForMaskedLM,
+    ErnieForPretraining,
+    ErnieForQuestionAnswering,
+    ErnieForSequenceClassification,
+    ErnieForTokenClassification,
+    ErnieModel,
+)
+from .roberta import (  # noqa: F401
+    RobertaConfig,
+    RobertaForMaskedLM,
+    RobertaForQuestionAnswering,
+    RobertaForSequenceClassification,
+    RobertaForTokenClassification,
+    RobertaModel,
+)
+from .electra import (  # noqa: F401
+    ElectraConfig,
+    ElectraDiscriminator,
+    ElectraForSequenceClassification,
+    ElectraForTokenClassification,
+    ElectraForTotalPretraining,
+    ElectraGenerator,
+    ElectraModel,
+)

@@ -1,3 +1,10 @@
 from .configuration import AutoConfig  # noqa: F401
-from .modeling import AutoModel, AutoModelForCausalLM  # noqa: F401
+from .modeling import (  # noqa: F401
+    AutoModel,
+    AutoModelForCausalLM,
+    AutoModelForMaskedLM,
+    AutoModelForQuestionAnswering,
+    AutoModelForSequenceClassification,
+    AutoModelForTokenClassification,
+)
 from .tokenizer import AutoTokenizer  # noqa: F401

@@ -23,3 +23,19 @@ class AutoModel(_AutoBase):
 
 class AutoModelForCausalLM(_AutoBase):
     _kind = "causal_lm"
+
+
+class AutoModelForSequenceClassification(_AutoBase):
+    _kind = "sequence_classification"
+
+
+class AutoModelForTokenClassification(_AutoBase):
+    _kind = "token_classification"
+
+
+class AutoModelForQuestionAnswering(_AutoBase):
+    _kind = "question_answering"
+
+
+class AutoModelForMaskedLM(_AutoBase):
+    _kind = "masked_lm"

@@ -1,4 +1,4 @@
-"""model_type -> (module, class names) registry shared by the Auto* classes.
+"""model_type -> class-name registry shared by the Auto* classes.
 
 Reference: paddlenlp/transformers/auto/{configuration,modeling,tokenizer}.py
 name->class registries resolved via importlib.
@@ -7,16 +7,38 @@ import importlib
 import json
 import os
 
-# model_type: (module path under paddlenlp_amd.transformers, config class,
-#              causal-lm class, base-model class)
+# model_type -> {kind: class name}; module defaults to the model_type key.
 MODEL_REGISTRY = {
-    "llama": ("llama", "LlamaConfig", "LlamaForCausalLM", "LlamaModel"),
-    "gpt2": ("gpt", "GPTConfig", "GPTForCausalLM", "GPTModel"),
-    "gpt": ("gpt", "GPTConfig", "GPTForCausalLM", "GPTModel"),
-    "qwen2": ("qwen2", "Qwen2Config", "Qwen2ForCausalLM", "Qwen2Model"),
-    "mixtral": ("mixtral", "MixtralConfig", "MixtralForCausalLM", "MixtralModel"),
-    "qwen2_moe": ("qwen2_moe", "Qwen2MoeConfig", "Qwen2MoeForCausalLM", "Qwen2MoeModel"),
-    "mistral": ("mistral", "MistralConfig", "MistralForCausalLM", "MistralModel"),
+    "llama": {"module": "llama", "config": "LlamaConfig",
+              "causal_lm": "LlamaForCausalLM", "base": "LlamaModel"},
+    "gpt2": {"module": "gpt", "config": "GPTConfig",
+             "causal_lm": "GPTForCausalLM", "base": "GPTModel"},
+    "gpt": {"module": "gpt", "config": "GPTConfig",
+            "causal_lm": "GPTForCausalLM", "base": "GPTModel"},
+    "qwen2": {"module": "qwen2", "config": "Qwen2Config",
+              "causal_lm": "Qwen2ForCausalLM", "base": "Qwen2Model"},
+    "mixtral": {"module": "mixtral", "config": "MixtralConfig",
+                "causal_lm": "MixtralForCausalLM", "base": "MixtralModel"},
+    "mistral": {"module": "mistral", "config": "MistralConfig",
+                "causal_lm": "MistralForCausalLM", "base": "MistralModel"},
+    "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
+             "sequence_classification": "BertForSequenceClassification",
+             "token_classification": "BertForTokenClassification",
+             "question_answering": "BertForQuestionAnswering",
+             "masked_lm": "BertForMaskedLM"},
+    "ernie": {"module": "ernie", "config": "ErnieConfig", "base": "ErnieModel",
+              "sequence_classification": "ErnieForSequenceClassification",
+              "token_classification": "ErnieForTokenClassification",
+              "question_answering": "ErnieForQuestionAnswering",
+              "masked_lm": "ErnieForMaskedLM"},
+    "roberta": {"module": "roberta", "config": "RobertaConfig", "base": "RobertaModel",
+                "sequence_classification": "RobertaForSequenceClassification",
+                "token_classification": "RobertaForTokenClassification",
+                "question_answering": "RobertaForQuestionAnswering",
+                "masked_lm": "RobertaForMaskedLM"},
+    "electra": {"module": "electra", "config": "ElectraConfig", "base": "ElectraModel",
+                "sequence_classification": "ElectraForSequenceClassification",
+                "token_classification": "ElectraForTokenClassification"},
 }
 
 
@@ -28,20 +50,23 @@ def resolve_model_type(path: str) -> str:
     if model_type is None:
         archs = cfg.get("architectures") or []
         for arch in archs:
-            for mt, (_, _, lm_cls, base_cls) in MODEL_REGISTRY.items():
-                if arch in (lm_cls, base_cls):
+            for mt, entry in MODEL_REGISTRY.items():
+                if arch in entry.values():
                     return mt
         raise ValueError(f"Cannot infer model_type from {config_file}")
     return model_type
 
 
 def get_class(model_type: str, kind: str):
-    """kind in {config, causal_lm, base}."""
+    """kind in {config, base, causal_lm, sequence_classification,
+    token_classification, question_answering, masked_lm}."""
     if model_type not in MODEL_REGISTRY:
         raise ValueError(
             f"Unknown model_type '{model_type}'. Registered: {sorted(MODEL_REGISTRY)}"
         )
-    module_name, cfg_cls, lm_cls, base_cls = MODEL_REGISTRY[model_type]
-    module = importlib.import_module(f"paddlenlp_amd.transformers.{module_name}")
-    name = {"config": cfg_cls, "causal_lm": lm_cls, "base": base_cls}[kind]
-    return getattr(module, name)
+    entry = MODEL_REGISTRY[model_type]
+    if kind not in entry:
+        raise ValueError(f"model_type '{model_type}' has no {kind} head")
+    module = importlib.import_module(
+        f"paddlenlp_amd.transformers.{entry['module']}")
+    return getattr(module, entry[kind])

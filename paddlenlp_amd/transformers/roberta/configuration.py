@@ -1,0 +1,56 @@
+"""RoBERTa config (reference: paddlenlp/transformers/roberta/configuration.py)."""
+from ..configuration_utils import PretrainedConfig
+
+__all__ = ["RobertaConfig"]
+
+
+class RobertaConfig(PretrainedConfig):
+    model_type = "roberta"
+
+    attribute_map = {
+        "num_classes": "num_labels",
+    }
+
+    def __init__(
+        self,
+        vocab_size=50265,
+        hidden_size=768,
+        num_hidden_layers=12,
+        num_attention_heads=12,
+        intermediate_size=3072,
+        hidden_act="gelu",
+        hidden_dropout_prob=0.1,
+        attention_probs_dropout_prob=0.1,
+        max_position_embeddings=514,
+        type_vocab_size=1,
+        initializer_range=0.02,
+        layer_norm_eps=1e-5,
+        pad_token_id=1,
+        bos_token_id=0,
+        eos_token_id=2,
+        classifier_dropout=None,
+        num_labels=2,
+        **kwargs,
+    ):
+        super().__init__(**kwargs)
+        self.vocab_size = vocab_size
+        self.hidden_size = hidden_size
+        self.num_hidden_layers = num_hidden_layers
+        self.num_attention_heads = num_attention_heads
+        self.intermediate_size = intermediate_size
+        self.hidden_act = hidden_act
+        self.hidden_dropout_prob = hidden_dropout_prob
+        self.attention_probs_dropout_prob = attention_probs_dropout_prob
+        self.max_position_embeddings = max_position_embeddings
+        self.type_vocab_size = type_vocab_size
+        self.initializer_range = initializer_range
+        self.layer_norm_eps = layer_norm_eps
+        self.pad_token_id = pad_token_id
+        self.bos_token_id = bos_token_id
+        self.eos_token_id = eos_token_id
+        self.classifier_dropout = classifier_dropout
+        self.num_labels = num_labels
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.num_attention_heads
