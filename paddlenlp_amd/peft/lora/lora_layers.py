@@ -21,9 +21,12 @@ from ...parallel.tensor_parallel import (
 
 
 class LoRALinear(nn.Module):
-    """y = x W^T + scale * (dropout(x) A^T) B^T; only A/B train."""
+    """y = x W^T + scale * (dropout(x) A^T) B^T; only A/B train.
 
-    def __init__(self, base: nn.Linear, r: int, lora_alpha: float = 1.0,
+    `base` may be an nn.Linear or a QuantizationLinear (QLoRA: frozen 4-bit
+    base + trainable adapters, reference peft/lora over quantization_linear)."""
+
+    def __init__(self, base: nn.Module, r: int, lora_alpha: float = 1.0,
                  lora_dropout: float = 0.0, rslora: bool = False,
                  lora_plus_scale: float = 1.0):
         super().__init__()
@@ -33,21 +36,26 @@ class LoRALinear(nn.Module):
         self.lora_dropout = nn.Dropout(lora_dropout) if lora_dropout > 0 else nn.Identity()
         in_f = base.in_features
         out_f = base.out_features
-        dtype = base.weight.dtype
-        device = base.weight.device
+        base_w = getattr(base, "weight", None)
+        if base_w is not None:
+            dtype, device = base_w.dtype, base_w.device
+        else:  # quantized base: packed buffers, compute dtype recorded on it
+            dtype = getattr(base, "compute_dtype", torch.float32)
+            device = base.quant_weight.device
         self.lora_A = nn.Parameter(torch.zeros(r, in_f, dtype=dtype, device=device))
         self.lora_B = nn.Parameter(torch.zeros(out_f, r, dtype=dtype, device=device))
         # lora+ : B gets a higher LR via a param attribute the optimizer reads
         self.lora_B.lr_scale = lora_plus_scale
         nn.init.kaiming_uniform_(self.lora_A, a=math.sqrt(5))
-        base.weight.requires_grad_(False)
-        if base.bias is not None:
+        if base_w is not None:
+            base_w.requires_grad_(False)
+        if getattr(base, "bias", None) is not None:
             base.bias.requires_grad_(False)
         self.merged = False
 
     @property
     def weight(self):
-        return self.base.weight
+        return getattr(self.base, "weight", None)
 
     def forward(self, x):
         y = self.base(x)
@@ -57,9 +65,12 @@ class LoRALinear(nn.Module):
 
     @torch.no_grad()
     def merge(self):
-        if not self.merged:
-            self.base.weight += self.scaling * (self.lora_B @ self.lora_A).to(self.base.weight.dtype)
-            self.merged = True
+        if self.merged:
+            return
+        if getattr(self.base, "weight", None) is None:
+            raise RuntimeError("cannot merge LoRA into a quantized (packed) base")
+        self.base.weight += self.scaling * (self.lora_B @ self.lora_A).to(self.base.weight.dtype)
+        self.merged = True
 
     @torch.no_grad()
     def unmerge(self):

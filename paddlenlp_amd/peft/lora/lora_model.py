@@ -19,6 +19,7 @@ from ...parallel.sequence_parallel import (
     ColumnSequenceParallelLinear,
     RowSequenceParallelLinear,
 )
+from ...quantization.quantization_linear import QuantizationLinear
 from ...utils.log import logger
 from .lora_layers import ColumnParallelLoRALinear, LoRALinear, RowParallelLoRALinear
 
@@ -76,7 +77,7 @@ class LoRAModel(nn.Module):
                 new = ColumnParallelLoRALinear(module, **kwargs)
             elif isinstance(module, (RowParallelLinear, RowSequenceParallelLinear)):
                 new = RowParallelLoRALinear(module, **kwargs)
-            elif isinstance(module, nn.Linear):
+            elif isinstance(module, (nn.Linear, QuantizationLinear)):
                 new = LoRALinear(module, **kwargs)
             else:
                 continue

@@ -28,6 +28,25 @@ def quantize_int8(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     return q, scale
 
 
+def quantize_int4(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Per-output-channel symmetric int4: w [out, in] -> (packed uint8
+    [out, in//2] with two nibbles per byte, fp32 scale[out])."""
+    assert w.shape[1] % 2 == 0, w.shape
+    scale = w.abs().amax(dim=1).clamp(min=1e-8).float() / 7.0
+    q = torch.clamp(torch.round(w.float() / scale[:, None]), -7, 7).to(torch.int8)
+    u = (q + 8).to(torch.uint8)  # offset-binary nibbles
+    packed = (u[:, 0::2] << 4) | u[:, 1::2]
+    return packed, scale
+
+
+def dequantize_int4(packed: torch.Tensor, scale: torch.Tensor,
+                    dtype: torch.dtype) -> torch.Tensor:
+    hi = (packed >> 4).to(torch.int8) - 8
+    lo = (packed & 0xF).to(torch.int8) - 8
+    q = torch.stack([hi, lo], dim=2).reshape(packed.shape[0], -1)
+    return (q.float() * scale[:, None]).to(dtype)
+
+
 def quantize_fp8(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     """Per-tensor fp8 e4m3fn: w -> (fp8 w, fp32 scale scalar)."""
     scale = (w.abs().amax().clamp(min=1e-8).float() / FP8_E4M3_MAX)
@@ -53,6 +72,9 @@ def weight_only_linear(x: torch.Tensor, qweight: torch.Tensor, scale: torch.Tens
         else:
             y = x2 @ (qweight.float() * scale).t().to(x.dtype)
         y = y.reshape(*orig_shape[:-1], -1)
+    elif algo == "weight_only_int4":
+        w = dequantize_int4(qweight, scale, x.dtype)
+        y = x @ w.t()
     elif algo == "weight_only_int8_gemv":
         # experimental fused dequant GEMV (ops/csrc/wint8_gemv.hip): correct
         # but currently VALU-bound (per-FMA LDS reads in the M loop) and
@@ -74,15 +96,26 @@ class QuantizationLinear(nn.Module):
 
     def __init__(self, in_features: int, out_features: int,
                  quant_algo: str = "weight_only_int8", bias: bool = False,
-                 dtype: torch.dtype = torch.bfloat16):
+                 dtype: torch.dtype = torch.bfloat16, block_size: int = 64):
         super().__init__()
         self.in_features = in_features
         self.out_features = out_features
         self.quant_algo = quant_algo
+        self.block_size = block_size
+        self.compute_dtype = dtype
         if quant_algo == "fp8":
             self.register_buffer("quant_weight",
                                  torch.zeros(out_features, in_features, dtype=torch.float8_e4m3fn))
             self.register_buffer("quant_scale", torch.ones((), dtype=torch.float32))
+        elif quant_algo == "weight_only_int4":
+            self.register_buffer("quant_weight",
+                                 torch.zeros(out_features, in_features // 2, dtype=torch.uint8))
+            self.register_buffer("quant_scale", torch.ones(out_features, dtype=torch.float32))
+        elif quant_algo in ("nf4", "fp4"):
+            n = out_features * in_features
+            assert n % block_size == 0, (out_features, in_features, block_size)
+            self.register_buffer("quant_weight", torch.zeros(n // 2, dtype=torch.uint8))
+            self.register_buffer("quant_scale", torch.ones(n // block_size, dtype=torch.float32))
         else:
             self.register_buffer("quant_weight",
                                  torch.zeros(out_features, in_features, dtype=torch.int8))
@@ -90,11 +123,18 @@ class QuantizationLinear(nn.Module):
         self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
 
     @classmethod
-    def from_linear(cls, linear: nn.Linear, quant_algo: str = "weight_only_int8"):
+    def from_linear(cls, linear: nn.Linear, quant_algo: str = "weight_only_int8",
+                    block_size: int = 64):
         m = cls(linear.in_features, linear.out_features, quant_algo,
-                bias=linear.bias is not None, dtype=linear.weight.dtype)
+                bias=linear.bias is not None, dtype=linear.weight.dtype,
+                block_size=block_size)
         if quant_algo == "fp8":
             q, s = quantize_fp8(linear.weight.data)
+        elif quant_algo == "weight_only_int4":
+            q, s = quantize_int4(linear.weight.data)
+        elif quant_algo in ("nf4", "fp4"):
+            from .qlora import quant_blockwise
+            q, s = quant_blockwise(linear.weight.data, quant_algo, block_size)
         else:
             q, s = quantize_int8(linear.weight.data)
         m.quant_weight.copy_(q)
@@ -105,5 +145,14 @@ class QuantizationLinear(nn.Module):
         return m
 
     def forward(self, x):
+        if self.quant_algo in ("nf4", "fp4"):
+            from .qlora import dequant_blockwise
+            w = dequant_blockwise(self.quant_weight, self.quant_scale,
+                                  self.quant_algo, self.block_size,
+                                  x.dtype).reshape(self.out_features, self.in_features)
+            y = x @ w.t()
+            if self.bias is not None:
+                y = y + self.bias
+            return y
         return weight_only_linear(x, self.quant_weight, self.quant_scale,
                                   self.bias, self.quant_algo)
