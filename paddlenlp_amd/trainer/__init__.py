@@ -19,3 +19,5 @@ from .trainer_utils import (  # noqa: F401
     speed_metrics,
 )
 from .training_args import TrainingArguments  # noqa: F401
+from .compression_args import CompressionArguments  # noqa: F401
+from .trainer_compress import compress  # noqa: F401

@@ -103,7 +103,7 @@ class EncoderSelfAttention(nn.Module):
                 attn_mask=add_mask,
                 dropout_p=self.dropout_p if self.training else 0.0,
             ).transpose(1, 2)
-        return self.out_proj(out.reshape(B, S, H))
+        return self.out_proj(out.reshape(B, S, self.num_heads * self.head_dim))
 
 
 class EncoderLayer(nn.Module):
