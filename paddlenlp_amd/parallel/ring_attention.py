@@ -12,10 +12,17 @@ point-to-point hop.  GPU blocks run the gfx950 flash kernels via ops._C
 (which return/consume the LSE); the CPU path uses fp32 torch math (the same
 formulas) so gloo tests validate the ring logic end-to-end.
 
-v1 shards the sequence contiguously (rank r owns chunk r).  Causal masking
-per ring step: kv from an earlier chunk is fully visible (causal=False), the
-own chunk is causal, later chunks are skipped.  The reference's load-
-balanced chunking (chunks i and 2cp-1-i) is a planned refinement.
+Two sharding modes:
+- contiguous: rank r owns chunk r.  Per ring step, kv from an earlier chunk
+  is fully visible (causal=False), the own chunk is causal, later chunks are
+  skipped — simple, but causal work is unbalanced (rank 0 does 1 block,
+  rank w-1 does w blocks).
+- balanced (zigzag, `balanced=True` + `zigzag_split`): the sequence is cut
+  into 2w chunks and rank r owns chunks (r, 2w-1-r), so every rank does the
+  same causal work per step (reference's load-balanced chunking).  Per step
+  against the kv of source rank s: s == r is plain causal on the local
+  concat; s < r attends all local q to the kv's FIRST chunk (full
+  visibility); s > r attends the local SECOND chunk of q to all kv.
 """
 from __future__ import annotations
 
@@ -171,11 +178,114 @@ class RingFlashAttention(torch.autograd.Function):
         return dq_acc, dk_cur, dv_cur, None, None
 
 
-def ring_flash_attention(q, k, v, group=None, causal: bool = True):
-    """q/k/v: this rank's sequence chunk [B, S/cp, H, D]."""
+def zigzag_split(t: torch.Tensor, world: int, rank: int, dim: int = 1) -> torch.Tensor:
+    """Load-balanced context-parallel shard: cut `dim` into 2*world chunks and
+    return cat(chunk[rank], chunk[2*world-1-rank]).  Inverse: zigzag_gather."""
+    chunks = t.chunk(2 * world, dim=dim)
+    return torch.cat([chunks[rank], chunks[2 * world - 1 - rank]], dim=dim).contiguous()
+
+
+def zigzag_gather(locals_per_rank, dim: int = 1) -> torch.Tensor:
+    """Reassemble the full sequence from every rank's zigzag shard (test /
+    logging helper; locals_per_rank[r] = that rank's [.., S/w, ..] shard)."""
+    world = len(locals_per_rank)
+    slots = [None] * (2 * world)
+    for r, t in enumerate(locals_per_rank):
+        first, second = t.chunk(2, dim=dim)
+        slots[r] = first
+        slots[2 * world - 1 - r] = second
+    return torch.cat(slots, dim=dim)
+
+
+class ZigzagRingFlashAttention(torch.autograd.Function):
+    """Balanced causal ring attention over zigzag shards (see module doc)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, group):
+        rank = dist.get_rank(group)
+        world = dist.get_world_size(group)
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        B, S2, Hq, D = q.shape  # S2 = 2 chunks
+        C = S2 // 2
+
+        out = torch.zeros(B, S2, Hq, D, dtype=torch.float32, device=q.device)
+        lse = torch.full((B, Hq, S2), float("-inf"),
+                         dtype=torch.float32, device=q.device)
+        k_cur, v_cur = k, v
+        for step in range(world):
+            src = (rank - step) % world
+            if step < world - 1:
+                nxt = _ring_send_recv([k_cur, v_cur], group)
+            if src == rank:
+                o_s, lse_s = _block_fwd(q, k_cur, v_cur, True)
+                out, lse = _merge(out, lse, o_s, lse_s)
+            elif src < rank:
+                # kv chunks (src, 2w-1-src) straddle ours: only the FIRST is
+                # earlier than all local q -> full visibility, second skipped
+                o_s, lse_s = _block_fwd(q, k_cur[:, :C].contiguous(), v_cur[:, :C].contiguous(), False)
+                out, lse = _merge(out, lse, o_s, lse_s)
+            else:
+                # only our SECOND q chunk (2w-1-rank) is after both kv chunks
+                o_s, lse_s = _block_fwd(q[:, C:].contiguous(), k_cur, v_cur, False)
+                o_m, lse_m = _merge(out[:, C:], lse[:, :, C:], o_s, lse_s)
+                out = torch.cat([out[:, :C], o_m], dim=1)
+                lse = torch.cat([lse[:, :, :C], lse_m], dim=2)
+            if step < world - 1:
+                k_cur, v_cur = nxt
+        out = out.to(q.dtype)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.group = group
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        group = ctx.group
+        rank = dist.get_rank(group)
+        world = dist.get_world_size(group)
+        dout = dout.contiguous()
+        C = q.shape[1] // 2
+
+        dq_acc = torch.zeros_like(q)
+        k_cur, v_cur = k, v
+        dk_cur = torch.zeros_like(k)
+        dv_cur = torch.zeros_like(v)
+        for step in range(world):
+            src = (rank - step) % world
+            if src == rank:
+                dq_s, dk_s, dv_s = _block_bwd(dout, q, k_cur, v_cur, out, lse, True)
+                dq_acc += dq_s
+                dk_cur = dk_cur + dk_s
+                dv_cur = dv_cur + dv_s
+            elif src < rank:
+                dq_s, dk_s, dv_s = _block_bwd(
+                    dout, q, k_cur[:, :C].contiguous(), v_cur[:, :C].contiguous(), out, lse, False)
+                dq_acc += dq_s
+                dk_cur = torch.cat([dk_cur[:, :C] + dk_s, dk_cur[:, C:]], dim=1)
+                dv_cur = torch.cat([dv_cur[:, :C] + dv_s, dv_cur[:, C:]], dim=1)
+            else:
+                dq_s, dk_s, dv_s = _block_bwd(
+                    dout[:, C:].contiguous(), q[:, C:].contiguous(), k_cur, v_cur,
+                    out[:, C:].contiguous(), lse[:, :, C:].contiguous(), False)
+                dq_acc[:, C:] += dq_s
+                dk_cur = dk_cur + dk_s
+                dv_cur = dv_cur + dv_s
+            if world > 1:
+                k_cur, v_cur, dk_cur, dv_cur = _ring_send_recv(
+                    [k_cur, v_cur, dk_cur, dv_cur], group)
+        return dq_acc, dk_cur, dv_cur, None
+
+
+def ring_flash_attention(q, k, v, group=None, causal: bool = True,
+                         balanced: bool = False):
+    """q/k/v: this rank's sequence chunk [B, S/cp, H, D].  With
+    balanced=True the chunk must be a `zigzag_split` shard (causal only)."""
     group = group if group is not None else get_topology().sep_parallel_group
     if group is None:
         from .. import ops
 
         return ops.flash_attention(q, k, v, causal=causal)
+    if balanced:
+        assert causal, "balanced (zigzag) sharding only applies to causal attention"
+        return ZigzagRingFlashAttention.apply(q, k, v, group)
     return RingFlashAttention.apply(q, k, v, group, causal)

@@ -255,3 +255,60 @@ def test_sep_llama_parity():
 
 def test_cp_llama_parity():
     _run_workers(_w_cp_llama)
+
+
+def _w_ring_balanced(rank, world):
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.ring_attention import (
+        ring_flash_attention, zigzag_split)
+    from paddlenlp_amd.ops import reference
+
+    topo = init_parallel_env(sep_degree=world, backend="gloo")
+    g = topo.sep_parallel_group
+    torch.manual_seed(7)
+    B, S, Hq, Hk, D = 1, 16, 4, 2, 8
+    q = torch.randn(B, S, Hq, D)
+    k = torch.randn(B, S, Hk, D)
+    v = torch.randn(B, S, Hk, D)
+
+    ql = zigzag_split(q, world, rank).requires_grad_()
+    kl = zigzag_split(k, world, rank).requires_grad_()
+    vl = zigzag_split(v, world, rank).requires_grad_()
+
+    out = ring_flash_attention(ql, kl, vl, group=g, causal=True, balanced=True)
+
+    qr = q.clone().requires_grad_()
+    kr = k.clone().requires_grad_()
+    vr = v.clone().requires_grad_()
+    ref = reference.flash_attention(qr, kr, vr, causal=True)
+    ref_local = zigzag_split(ref, world, rank)
+    assert torch.allclose(out, ref_local, atol=1e-4), (out - ref_local).abs().max()
+
+    gfull = torch.arange(ref.numel(), dtype=torch.float32).reshape(ref.shape) / ref.numel()
+    out.backward(zigzag_split(gfull, world, rank))
+    ref.backward(gfull)
+    for loc, full in ((ql, qr), (kl, kr), (vl, vr)):
+        expect = zigzag_split(full.grad, world, rank)
+        assert torch.allclose(loc.grad, expect, atol=1e-4), \
+            (loc.grad - expect).abs().max()
+
+
+def test_ring_attention_balanced_zigzag():
+    _run_workers(_w_ring_balanced)
+
+
+def test_zigzag_split_gather_roundtrip():
+    from paddlenlp_amd.parallel.ring_attention import zigzag_gather, zigzag_split
+
+    x = torch.arange(32.0).reshape(1, 32, 1)
+    world = 4
+    shards = [zigzag_split(x, world, r) for r in range(world)]
+    assert all(s.shape[1] == 8 for s in shards)
+    torch.testing.assert_close(zigzag_gather(shards), x)
+    # rank 0 owns the first and the LAST chunk (the balance property)
+    torch.testing.assert_close(shards[0][0, :4, 0], torch.arange(4.0))
+    torch.testing.assert_close(shards[0][0, 4:, 0], torch.arange(28.0, 32.0))
+
+
+def test_ring_attention_balanced_zigzag_world4():
+    _run_workers(_w_ring_balanced, world_size=4)
