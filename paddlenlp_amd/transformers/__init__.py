@@ -46,3 +46,9 @@ from .electra import (  # noqa: F401
     ElectraGenerator,
     ElectraModel,
 )
+from .t5 import (  # noqa: F401
+    T5Config,
+    T5EncoderModel,
+    T5ForConditionalGeneration,
+    T5Model,
+)

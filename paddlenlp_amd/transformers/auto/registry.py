@@ -21,6 +21,8 @@ MODEL_REGISTRY = {
                 "causal_lm": "MixtralForCausalLM", "base": "MixtralModel"},
     "mistral": {"module": "mistral", "config": "MistralConfig",
                 "causal_lm": "MistralForCausalLM", "base": "MistralModel"},
+    "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
+           "seq2seq_lm": "T5ForConditionalGeneration"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",
