@@ -52,3 +52,8 @@ from .t5 import (  # noqa: F401
     T5ForConditionalGeneration,
     T5Model,
 )
+from .qwen2_moe import (  # noqa: F401
+    Qwen2MoeConfig,
+    Qwen2MoeForCausalLM,
+    Qwen2MoeModel,
+)

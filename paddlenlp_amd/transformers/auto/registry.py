@@ -19,6 +19,8 @@ MODEL_REGISTRY = {
               "causal_lm": "Qwen2ForCausalLM", "base": "Qwen2Model"},
     "mixtral": {"module": "mixtral", "config": "MixtralConfig",
                 "causal_lm": "MixtralForCausalLM", "base": "MixtralModel"},
+    "qwen2_moe": {"module": "qwen2_moe", "config": "Qwen2MoeConfig",
+                  "causal_lm": "Qwen2MoeForCausalLM", "base": "Qwen2MoeModel"},
     "mistral": {"module": "mistral", "config": "MistralConfig",
                 "causal_lm": "MistralForCausalLM", "base": "MistralModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
