@@ -214,8 +214,11 @@ class _CrossEntropyFunction(torch.autograd.Function):
 
 
 def cross_entropy(logits, labels, ignore_index: int = -100, reduction: str = "none"):
-    """Per-token loss [N] from logits [N, V]; reduction applied on top."""
-    if logits.is_cuda:
+    """Per-token loss [N] from logits [N, V]; reduction applied on top.
+
+    The fused gfx950 kernel handles bf16 logits with V % 8 == 0 (the LLM
+    vocab shapes); anything else computes eager on the same device."""
+    if logits.is_cuda and logits.dtype == torch.bfloat16 and logits.shape[-1] % 8 == 0:
         loss = _CrossEntropyFunction.apply(logits, labels, ignore_index)
     else:
         loss = reference.cross_entropy(logits, labels, ignore_index, reduction="none")

@@ -310,7 +310,7 @@ class T5ForConditionalGeneration(T5PretrainedModel):
         logits = self.lm_head(hidden)
         if labels is not None:
             loss = ops.cross_entropy(
-                logits.reshape(-1, logits.shape[-1]).float(),
+                logits.reshape(-1, logits.shape[-1]),
                 labels.reshape(-1), -100, reduction="mean")
             return (loss, logits) if not use_cache else (loss, logits, presents, enc)
         return logits if not use_cache else (logits, presents, enc)
