@@ -10,7 +10,9 @@ MI355X design: the schedule is implemented directly on torch.distributed
 point-to-point ops (isend/irecv over RCCL on xGMI; gloo in CPU tests).
 Activations crossing stage boundaries are fixed-shape hidden-state tensors
 [B, S, H] in the compute dtype, so no shape negotiation is needed per
-micro-batch.  Non-interleaved 1F1B; virtual stages are a planned extension.
+micro-batch.  PipelineEngine runs non-interleaved 1F1B;
+InterleavedPipelineEngine runs the virtual-stage (VPP) schedule with
+per-(chunk, direction) communication channels.
 """
 from __future__ import annotations
 
@@ -77,6 +79,7 @@ class PipelineModule(nn.Module):
         loss_fn: Optional[nn.Module] = None,
         seg_method: str = "uniform",
         topology=None,
+        num_virtual_stages: int = 1,
     ):
         super().__init__()
         self.topo = topology or get_topology()
@@ -85,47 +88,56 @@ class PipelineModule(nn.Module):
         self.pp_group = self.topo.pipe_parallel_group
         self.loss_fn = loss_fn
         self.descs = layer_descs
+        self.num_virtual_stages = num_virtual_stages
 
-        start, end = self._segment(layer_descs, seg_method)
-        self.local_start, self.local_end = start, end
+        # rank r owns virtual stages {c*P + r}: ranges[c] = (start, end)
+        ranges = self._segment(layer_descs, seg_method)
+        self.local_start, self.local_end = ranges[0]
         self.local_layers = nn.ModuleList()
         self.shared_layers: Dict[str, nn.Module] = {}
         self._layer_descs_local = []
-        for i in range(start, end):
-            desc = layer_descs[i]
-            layer = desc.build()
-            self.local_layers.append(layer)
-            self._layer_descs_local.append(desc)
-            if isinstance(desc, SharedLayerDesc):
-                self.shared_layers[desc.key] = layer
+        self.chunk_bounds: List[Tuple[int, int]] = []  # into local_layers
+        for start, end in ranges:
+            lo = len(self.local_layers)
+            for i in range(start, end):
+                desc = layer_descs[i]
+                layer = desc.build()
+                self.local_layers.append(layer)
+                self._layer_descs_local.append(desc)
+                if isinstance(desc, SharedLayerDesc):
+                    self.shared_layers[desc.key] = layer
+            self.chunk_bounds.append((lo, len(self.local_layers)))
         logger.info(
-            f"PP stage {self.pp_rank}/{self.pp_degree}: layers [{start}, {end}) of {len(layer_descs)}"
+            f"PP stage {self.pp_rank}/{self.pp_degree} x{num_virtual_stages}v: "
+            f"desc ranges {ranges} of {len(layer_descs)}"
         )
 
         # shared-weight groups: ranks sharing a key all-reduce tied grads
         self._shared_comm = self._build_shared_comm(layer_descs)
 
-    def _segment(self, descs, seg_method) -> Tuple[int, int]:
+    def _segment(self, descs, seg_method) -> List[Tuple[int, int]]:
+        """Desc ranges for this rank's chunks (one per virtual stage)."""
         n = len(descs)
-        if self.pp_degree == 1:
-            return 0, n
+        v = self.num_virtual_stages
+        P = self.pp_degree
+        if P == 1 and v == 1:
+            return [(0, n)]
+        n_parts = P * v
+        starts = None
         if seg_method.startswith("layer:"):
             cls_name = seg_method.split(":", 1)[1]
             marks = [i for i, d in enumerate(descs) if d.layer_cls.__name__ == cls_name]
-            if not marks:
-                seg_method = "uniform"
-            else:
-                bounds = _segment_uniform(len(marks), self.pp_degree)
-                # stage boundaries in desc indices
-                starts = []
-                for p in range(self.pp_degree):
+            if marks:
+                bounds = _segment_uniform(len(marks), n_parts)
+                starts = [0]
+                for p in range(1, n_parts):
                     starts.append(marks[bounds[p]] if bounds[p] < len(marks) else n)
                 starts.append(n)
-                start = starts[self.pp_rank] if self.pp_rank > 0 else 0
-                end = starts[self.pp_rank + 1] if self.pp_rank < self.pp_degree - 1 else n
-                return start, end
-        bounds = _segment_uniform(n, self.pp_degree)
-        return bounds[self.pp_rank], bounds[self.pp_rank + 1]
+        if starts is None:
+            bounds = _segment_uniform(n, n_parts)
+            starts = list(bounds)
+        return [(starts[c * P + self.pp_rank], starts[c * P + self.pp_rank + 1])
+                for c in range(v)]
 
     def _build_shared_comm(self, descs):
         """For each SharedLayerDesc key, create a group of pp ranks holding it."""
@@ -168,8 +180,10 @@ class PipelineModule(nn.Module):
     def is_last_stage(self):
         return self.pp_rank == self.pp_degree - 1
 
-    def stage_forward(self, x):
-        for layer, desc in zip(self.local_layers, self._layer_descs_local):
+    def stage_forward(self, x, chunk: int = 0):
+        lo, hi = self.chunk_bounds[chunk]
+        for i in range(lo, hi):
+            layer, desc = self.local_layers[i], self._layer_descs_local[i]
             if isinstance(desc, SharedLayerDesc) and desc.forward_fn is not None:
                 x = desc.forward_fn(layer, x)
             else:
@@ -293,5 +307,133 @@ class PipelineEngine:
         self.module.allreduce_shared_weight_gradients()
 
         if self.module.is_last_stage and losses:
+            return torch.stack([l.detach() for l in losses]).sum()
+        return torch.zeros((), device=self.device)
+
+
+class InterleavedPipelineEngine:
+    """Interleaved 1F1B over virtual stages (VPP, Megatron-style schedule).
+
+    Each physical rank r holds v model chunks; virtual stage s = c*P + r runs
+    chunk c on rank r, shrinking the pipeline bubble from (P-1)/M to
+    (P-1)/(v*M).  Per-(direction, chunk) process groups give every message
+    stream its own FIFO channel, so eager isend + blocking recv follow the
+    dataflow DAG and cannot deadlock or cross-match.  Requires M % P == 0
+    (the schedule interleaves microbatches in groups of P).
+    """
+
+    def __init__(self, module: PipelineModule, hidden_shape_fn: Callable,
+                 dtype: torch.dtype, device: torch.device):
+        self.module = module
+        self.topo = module.topo
+        self.pp_group = module.pp_group
+        self.pp_rank = module.pp_rank
+        self.pp_degree = module.pp_degree
+        self.v = module.num_virtual_stages
+        self.hidden_shape_fn = hidden_shape_fn
+        self.dtype = dtype
+        self.device = device
+        ranks = (dist.get_process_group_ranks(self.pp_group)
+                 if self.pp_group is not None else [0])
+        self.prev_rank = ranks[(self.pp_rank - 1) % self.pp_degree]
+        self.next_rank = ranks[(self.pp_rank + 1) % self.pp_degree]
+        # one channel per (chunk, direction): independent FIFO orderings
+        self.fwd_ch = [dist.new_group(ranks=ranks) for _ in range(self.v)]
+        self.bwd_ch = [dist.new_group(ranks=ranks) for _ in range(self.v)]
+
+    def _send(self, tensor, dst, group):
+        buf = tensor.contiguous()
+        work = dist.isend(buf, dst=dst, group=group)
+        self._pending.append((work, buf))
+
+    def _recv(self, shape, src, group):
+        buf = torch.empty(shape, dtype=self.dtype, device=self.device)
+        dist.recv(buf, src=src, group=group)
+        return buf
+
+    def forward_backward(self, micro_batches: List[Dict[str, torch.Tensor]],
+                         input_fn: Callable, scale_loss: bool = True):
+        M = len(micro_batches)
+        P, r, v = self.pp_degree, self.pp_rank, self.v
+        V = P * v
+        assert M % P == 0, f"interleaved schedule needs M % P == 0 (M={M}, P={P})"
+        total = M * v
+        warmup = min((P - r - 1) * 2 + (v - 1) * P, total)
+        self._pending: List = []
+
+        saved: Dict[Tuple[int, int], Tuple] = {}
+        losses = []
+
+        def fwd_unit_of(k):
+            c = (k % (P * v)) // P
+            mb = (k // (P * v)) * P + (k % P)
+            return c, mb
+
+        def bwd_unit_of(k):
+            c = v - 1 - (k % (P * v)) // P
+            mb = (k // (P * v)) * P + (k % P)
+            return c, mb
+
+        def run_forward(k):
+            c, mb = fwd_unit_of(k)
+            s = c * P + r
+            batch = micro_batches[mb]
+            if s == 0:
+                x_in = None
+                x = input_fn(batch)
+            else:
+                # sender chunk: same c from rank r-1, or c-1 wrapping from
+                # the last rank into our chunk c
+                src_chunk = c if r > 0 else c - 1
+                x_in = self._recv(self.hidden_shape_fn(batch), self.prev_rank,
+                                  self.fwd_ch[src_chunk])
+                x_in.requires_grad_()
+                x = x_in
+            out = self.module.stage_forward(x, chunk=c)
+            if s == V - 1:
+                loss = self.module.loss_fn(out, batch)
+                if scale_loss:
+                    loss = loss / M
+                losses.append(loss)
+                saved[(c, mb)] = (x_in, loss)
+            else:
+                self._send(out.detach(), self.next_rank, self.fwd_ch[c])
+                saved[(c, mb)] = (x_in, out)
+
+        def run_backward(k):
+            c, mb = bwd_unit_of(k)
+            s = c * P + r
+            x_in, out = saved.pop((c, mb))
+            if s == V - 1:
+                out.backward()
+            else:
+                # grad comes from the next virtual stage: rank r+1 chunk c,
+                # or (wrapping) rank 0 chunk c+1
+                src_chunk = c if r < P - 1 else c + 1
+                grad = self._recv(out.shape, self.next_rank, self.bwd_ch[src_chunk])
+                out.backward(gradient=grad)
+            if s > 0:
+                self._send(x_in.grad, self.prev_rank, self.bwd_ch[c])
+
+        fk = bk = 0
+        for _ in range(warmup):
+            run_forward(fk)
+            fk += 1
+        for _ in range(total - warmup):
+            run_forward(fk)
+            fk += 1
+            run_backward(bk)
+            bk += 1
+        while bk < total:
+            run_backward(bk)
+            bk += 1
+
+        for work, _ in self._pending:
+            work.wait()
+        self._pending.clear()
+        self.module.allreduce_shared_weight_gradients()
+
+        last_stage_rank = (V - 1) % P
+        if r == last_stage_rank and losses:
             return torch.stack([l.detach() for l in losses]).sum()
         return torch.zeros((), device=self.device)

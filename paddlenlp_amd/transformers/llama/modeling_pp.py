@@ -72,7 +72,7 @@ class LlamaForCausalLMPipe(PipelineModule):
 
         return LlamaForCausalLM._get_tensor_parallel_mappings(config, is_split)
 
-    def __init__(self, config: LlamaConfig):
+    def __init__(self, config: LlamaConfig, num_virtual_stages: int = 1):
         criterion = LlamaPretrainingCriterion(config)
         descs = [LayerDesc(EmbeddingPipe, config, name="embedding")]
         for i in range(config.num_hidden_layers):
@@ -88,6 +88,7 @@ class LlamaForCausalLMPipe(PipelineModule):
             loss_fn=loss_fn,
             seg_method="layer:LlamaDecoderLayer",
             topology=get_topology(),
+            num_virtual_stages=num_virtual_stages,
         )
         self.config = config
         self._local_names = [d.name for d in self._layer_descs_local]
