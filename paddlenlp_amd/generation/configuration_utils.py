@@ -16,6 +16,8 @@ class GenerationConfig:
     max_length: Optional[int] = None
     do_sample: bool = False
     num_beams: int = 1
+    num_beam_groups: int = 1
+    diversity_penalty: float = 0.0
     temperature: float = 1.0
     top_k: int = 50
     top_p: float = 1.0

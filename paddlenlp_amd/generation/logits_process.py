@@ -71,3 +71,20 @@ class TopPLogitsWarper:
         remove = cum - probs > self.top_p  # keep first token exceeding p
         mask = remove.scatter(1, sorted_idx, remove)
         return logits.masked_fill(mask, float("-inf"))
+
+
+class HammingDiversityLogitsProcessor:
+    """Subtract `diversity_penalty` for tokens earlier beam groups already
+    chose at this step (reference logits_process group beam search)."""
+
+    def __init__(self, diversity_penalty: float, num_beams: int, num_beam_groups: int):
+        self.diversity_penalty = diversity_penalty
+        self.num_sub_beams = num_beams // num_beam_groups
+
+    def __call__(self, scores, used_token_counts):
+        """scores: [B*Kg, V]; used_token_counts: [B, V] picks by earlier
+        groups this step."""
+        if self.diversity_penalty == 0.0:
+            return scores
+        return scores - self.diversity_penalty * used_token_counts.repeat_interleave(
+            self.num_sub_beams, dim=0)

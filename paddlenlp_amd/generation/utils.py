@@ -66,6 +66,8 @@ class GenerationMixin:
         if gen.max_length:
             gen.max_new_tokens = max(1, gen.max_length - input_ids.shape[1])
 
+        if gen.num_beams > 1 and gen.num_beam_groups > 1:
+            return self.group_beam_search(input_ids, gen)
         if gen.num_beams > 1:
             return self.beam_search(input_ids, gen)
         if gen.do_sample:
@@ -186,6 +188,99 @@ class GenerationMixin:
                 for j in range(K):
                     lp = (ids.shape[1] - prompt_len) ** gen.length_penalty
                     cands.append((beam_scores[b * K + j].item() / lp, ids[b * K + j]))
+            cands.sort(key=lambda x: -x[0])
+            results.append(cands[0][1][prompt_len:])
+        maxlen = max(r.shape[0] for r in results)
+        pad_id = gen.pad_token_id or (eos_ids[0] if eos_ids else 0)
+        out = torch.full((B, maxlen), pad_id, dtype=torch.long, device=device)
+        for b, r in enumerate(results):
+            out[b, : r.shape[0]] = r
+        return out, None
+
+
+    def group_beam_search(self, input_ids, gen: GenerationConfig):
+        """Diverse (group) beam search: beams split into num_beam_groups
+        groups; later groups pay `diversity_penalty` per use of a token an
+        earlier group already picked at the same step (reference
+        group_beam_search / HammingDiversityLogitsProcessor)."""
+        B, prompt_len = input_ids.shape
+        K = gen.num_beams
+        G = gen.num_beam_groups
+        assert K % G == 0, f"num_beams {K} not divisible by num_beam_groups {G}"
+        Kg = K // G
+        device = input_ids.device
+        eos_ids = gen.eos_ids()
+        procs = self._get_logits_processors(gen, prompt_len)
+
+        ids = input_ids.repeat_interleave(K, dim=0)   # beam layout: group-major
+        beam_scores = torch.full((B, K), float("-inf"), device=device)
+        beam_scores[:, [g * Kg for g in range(G)]] = 0.0
+        beam_scores = beam_scores.view(-1)
+        past = None
+        cur = ids
+        finished = [[[] for _ in range(G)] for _ in range(B)]
+
+        for step in range(gen.max_new_tokens):
+            out = self(input_ids=cur, use_cache=True, past_key_values=past)
+            logits, past = out
+            logp_all = logits[:, -1].float().log_softmax(-1)   # [B*K, V]
+            logp_all = procs(ids, logp_all)
+            vocab = logp_all.shape[-1]
+
+            new_ids = [None] * (B * K)
+            new_scores = torch.empty(B * K, device=device)
+            new_src = [0] * (B * K)
+            used = torch.zeros(B, vocab, device=device)  # earlier groups' picks
+            for g in range(G):
+                rows = torch.cat([
+                    torch.arange(b * K + g * Kg, b * K + (g + 1) * Kg)
+                    for b in range(B)]).to(device)
+                logp = logp_all[rows].view(B, Kg, vocab)
+                if g > 0 and gen.diversity_penalty > 0:
+                    logp = logp - gen.diversity_penalty * used[:, None, :]
+                scores = beam_scores[rows].view(B, Kg, 1) + logp
+                scores = scores.view(B, Kg * vocab)
+                top_scores, top_idx = scores.topk(2 * Kg, dim=-1)
+                beam_idx = top_idx // vocab
+                token_idx = top_idx % vocab
+                for b in range(B):
+                    n_live = 0
+                    for j in range(2 * Kg):
+                        tok = int(token_idx[b, j])
+                        src = b * K + g * Kg + int(beam_idx[b, j])
+                        seq = torch.cat([ids[src], token_idx[b, j:j + 1]])
+                        if eos_ids and tok in eos_ids:
+                            lp = (seq.shape[0] - prompt_len) ** gen.length_penalty
+                            finished[b][g].append((float(top_scores[b, j]) / lp, seq))
+                            used[b, tok] += 1.0
+                        elif n_live < Kg:
+                            slot = b * K + g * Kg + n_live
+                            new_ids[slot] = seq
+                            new_scores[slot] = top_scores[b, j]
+                            new_src[slot] = src
+                            used[b, tok] += 1.0
+                            n_live += 1
+                    while n_live < Kg:  # degenerate fill
+                        slot = b * K + g * Kg + n_live
+                        new_ids[slot] = new_ids[slot - 1]
+                        new_scores[slot] = new_scores[slot - 1]
+                        new_src[slot] = new_src[slot - 1]
+                        n_live += 1
+
+            ids = torch.stack(new_ids)
+            beam_scores = new_scores
+            past = _reorder_cache(past, torch.tensor(new_src, device=device))
+            cur = ids[:, -1:]
+            if all(len(fg) >= Kg for f in finished for fg in f):
+                break
+
+        results = []
+        for b in range(B):
+            cands = [c for fg in finished[b] for c in fg]
+            if not cands:
+                for j in range(K):
+                    lp = (ids.shape[1] - prompt_len) ** gen.length_penalty
+                    cands.append((float(beam_scores[b * K + j]) / lp, ids[b * K + j]))
             cands.sort(key=lambda x: -x[0])
             results.append(cands[0][1][prompt_len:])
         maxlen = max(r.shape[0] for r in results)

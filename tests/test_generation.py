@@ -95,3 +95,41 @@ def test_eos_stopping(model):
         model.lm_head.weight[2] = saved
     assert out.shape[1] < 20  # stopped early at EOS
     assert (out[:, 0] == 2).all()
+
+
+def test_group_beam_search_diversity(model):
+    """Group beam search runs, and with a strong diversity penalty the two
+    groups pick different first tokens."""
+    ids = torch.randint(3, 128, (2, 6))
+    gen = GenerationConfig(max_new_tokens=6, num_beams=4, num_beam_groups=2,
+                           diversity_penalty=0.0, eos_token_id=2, pad_token_id=0)
+    out, _ = model.generate(ids, gen)
+    assert out.shape[0] == 2 and out.shape[1] <= 6
+
+    # plain beam search and group beam search with zero penalty agree on the
+    # best sequence (same search, grouped)
+    plain_gen = GenerationConfig(max_new_tokens=6, num_beams=2, eos_token_id=2,
+                                 pad_token_id=0)
+    plain, _ = model.generate(ids, plain_gen)
+    gen0 = GenerationConfig(max_new_tokens=6, num_beams=4, num_beam_groups=2,
+                            diversity_penalty=0.0, eos_token_id=2, pad_token_id=0)
+    grouped, _ = model.generate(ids, gen0)
+    L = min(plain.shape[1], grouped.shape[1])
+    assert torch.equal(plain[:, :L], grouped[:, :L])
+
+
+def test_hamming_diversity_processor():
+    from paddlenlp_amd.generation.logits_process import (
+        HammingDiversityLogitsProcessor,
+    )
+
+    proc = HammingDiversityLogitsProcessor(
+        diversity_penalty=1.5, num_beams=4, num_beam_groups=2)
+    scores = torch.zeros(4, 10)  # B=2 x Kg=2
+    used = torch.zeros(2, 10)
+    used[0, 3] = 2.0
+    used[1, 7] = 1.0
+    out = proc(scores, used)
+    assert out[0, 3] == -3.0 and out[1, 3] == -3.0   # batch 0 rows
+    assert out[2, 7] == -1.5 and out[3, 7] == -1.5   # batch 1 rows
+    assert out[0, 0] == 0.0
