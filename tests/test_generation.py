@@ -133,3 +133,43 @@ def test_hamming_diversity_processor():
     assert out[0, 3] == -3.0 and out[1, 3] == -3.0   # batch 0 rows
     assert out[2, 7] == -1.5 and out[3, 7] == -1.5   # batch 1 rows
     assert out[0, 0] == 0.0
+
+
+def test_speculative_greedy_matches_target_greedy(model):
+    """Greedy speculative decode must be IDENTICAL to target-only greedy."""
+    from paddlenlp_amd.generation.speculative import speculative_generate
+
+    torch.manual_seed(1)
+    draft_cfg = LlamaConfig(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=2,
+        max_position_embeddings=256, eos_token_id=2, pad_token_id=0,
+    )
+    draft = LlamaForCausalLM(draft_cfg).eval()
+
+    for seed in (0, 3):
+        g = torch.Generator().manual_seed(seed)
+        ids = torch.randint(3, 128, (1, 8), generator=g)
+        gen = GenerationConfig(max_new_tokens=12, do_sample=False,
+                               eos_token_id=2, pad_token_id=0)
+        ref, _ = model.generate(ids, gen)
+        spec, stats = speculative_generate(model, draft, ids, gen, gamma=3)
+        L = min(ref.shape[1], spec.shape[1])
+        assert torch.equal(ref[0, :L], spec[0, :L]), (ref, spec)
+        assert stats["drafted"] > 0
+
+    # a perfect draft (the target itself) accepts everything
+    _, stats = speculative_generate(model, model, ids, gen, gamma=3)
+    assert stats["acceptance_rate"] == 1.0
+
+
+def test_speculative_sampling_runs(model):
+    from paddlenlp_amd.generation.speculative import speculative_generate
+
+    torch.manual_seed(2)
+    ids = torch.randint(3, 128, (1, 6))
+    gen = GenerationConfig(max_new_tokens=8, do_sample=True, temperature=1.0,
+                           eos_token_id=2, pad_token_id=0)
+    out, stats = speculative_generate(model, model, ids, gen, gamma=2)
+    assert out.shape[1] <= 8
+    assert 0.0 <= stats["acceptance_rate"] <= 1.0
