@@ -54,31 +54,55 @@ class _GenerationTask:
 
 GENERATION_TASKS = set(_GenerationTask.TEMPLATES)
 
-# API-surface parity with the reference registry (taskflow/taskflow.py:48);
-# pipelines beyond generation are pending their task models.
-PENDING_TASKS = {
-    "dependency_parsing", "document_intelligence", "fill_mask",
-    "information_extraction", "knowledge_mining", "lexical_analysis", "ner",
-    "pos_tagging", "sentiment_analysis", "text_classification",
-    "text_correction", "text_similarity", "feature_extraction",
-    "zero_shot_text_classification", "word_segmentation",
+# encoder-backed pipelines (taskflow/tasks.py): task name -> class
+from .tasks import (  # noqa: E402
+    FeatureExtractionTask,
+    FillMaskTask,
+    InformationExtractionTask,
+    TextClassificationTask,
+    TextSimilarityTask,
+    TokenClassificationTask,
+)
+
+ENCODER_TASKS = {
+    "text_classification": TextClassificationTask,
+    "sentiment_analysis": TextClassificationTask,
+    "ner": TokenClassificationTask,
+    "pos_tagging": TokenClassificationTask,
+    "lexical_analysis": TokenClassificationTask,
+    "fill_mask": FillMaskTask,
+    "feature_extraction": FeatureExtractionTask,
+    "text_similarity": TextSimilarityTask,
+    "information_extraction": InformationExtractionTask,
 }
 
-TASKS = sorted(GENERATION_TASKS | PENDING_TASKS)
+# API-surface parity with the reference registry (taskflow/taskflow.py:48);
+# remaining pipelines need task models that don't exist offline.
+PENDING_TASKS = {
+    "dependency_parsing", "document_intelligence", "knowledge_mining",
+    "text_correction", "zero_shot_text_classification", "word_segmentation",
+}
+
+TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)
 
 
 class Taskflow:
     def __init__(self, task: str, model: Optional[str] = None, **kwargs):
-        if task in GENERATION_TASKS:
+        if task in GENERATION_TASKS or task in ENCODER_TASKS:
             if model is None:
                 raise ValueError(
                     f"Taskflow('{task}') needs a local model path via model= "
                     "(no network access in this environment)")
-            self.task_instance = _GenerationTask(task, model, **kwargs)
+            cls = ENCODER_TASKS.get(task, _GenerationTask)
+            if cls is _GenerationTask:
+                self.task_instance = _GenerationTask(task, model, **kwargs)
+            else:
+                self.task_instance = cls(model, **kwargs)
         elif task in PENDING_TASKS:
             raise NotImplementedError(
                 f"Task '{task}' is registered but its task model is not ported "
-                f"yet. Available now: {sorted(GENERATION_TASKS)}")
+                f"yet. Available now: "
+                f"{sorted(GENERATION_TASKS | set(ENCODER_TASKS))}")
         else:
             raise ValueError(f"Unknown task '{task}'. Registered: {TASKS}")
         self.task = task

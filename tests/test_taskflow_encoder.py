@@ -1,0 +1,127 @@
+"""Encoder-backed Taskflow pipelines over tiny local models.
+
+Reference behavior: paddlenlp/taskflow pipelines (text_classification, ner,
+fill_mask, text_similarity, information_extraction).
+"""
+import pytest
+import torch
+
+from paddlenlp_amd.taskflow import Taskflow
+from paddlenlp_amd.transformers import (
+    UIE,
+    BertConfig,
+    BertForMaskedLM,
+    BertForSequenceClassification,
+    BertForTokenClassification,
+    ErnieConfig,
+)
+
+VOCAB = ["[PAD]", "[UNK]", "[MASK]", "the", "cat", "sat", "on", "mat",
+         "paris", "visited", "alice", "bob"]
+
+
+def _save_tokenizer(path):
+    from tokenizers import Tokenizer, models, pre_tokenizers
+
+    tok = Tokenizer(models.WordLevel({w: i for i, w in enumerate(VOCAB)},
+                                     unk_token="[UNK]"))
+    tok.pre_tokenizer = pre_tokenizers.WhitespaceSplit()
+    tok.save(str(path / "tokenizer.json"))
+
+
+def _tiny_cfg(**kw):
+    return BertConfig(vocab_size=len(VOCAB), hidden_size=32,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      intermediate_size=64, max_position_embeddings=64,
+                      hidden_dropout_prob=0.0,
+                      attention_probs_dropout_prob=0.0, **kw)
+
+
+def test_text_classification_task(tmp_path):
+    torch.manual_seed(0)
+    m = BertForSequenceClassification(_tiny_cfg(num_labels=2))
+    m.config.id2label = {0: "negative", 1: "positive"}
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("text_classification", model=str(tmp_path))
+    out = flow("the cat sat")
+    assert set(out) == {"text", "label", "score"}
+    assert out["label"] in ("negative", "positive")
+    outs = flow(["the cat", "the mat"])
+    assert len(outs) == 2
+
+
+def test_ner_task_bio_spans(tmp_path):
+    torch.manual_seed(0)
+    m = BertForTokenClassification(_tiny_cfg(num_labels=3))
+    m.config.id2label = {0: "O", 1: "B-PER", 2: "I-PER"}
+    # force deterministic predictions: bias the classifier to B-PER for all
+    with torch.no_grad():
+        m.classifier.weight.zero_()
+        m.classifier.bias.copy_(torch.tensor([0.0, 1.0, -1.0]))
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("ner", model=str(tmp_path))
+    out = flow("alice visited paris")
+    # every token B-PER -> three single-token entities
+    assert [e["text"] for e in out["entities"]] == ["alice", "visited", "paris"]
+    assert all(e["entity"] == "PER" for e in out["entities"])
+
+
+def test_fill_mask_task(tmp_path):
+    torch.manual_seed(0)
+    m = BertForMaskedLM(_tiny_cfg())
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("fill_mask", model=str(tmp_path), top_k=3)
+    out = flow("the [MASK] sat")
+    assert len(out["predictions"]) == 1
+    assert len(out["predictions"][0]) == 3
+    assert all("token" in p and "score" in p for p in out["predictions"][0])
+
+
+def test_text_similarity_task(tmp_path):
+    torch.manual_seed(0)
+    m = BertForSequenceClassification(_tiny_cfg(num_labels=2))
+    m.bert.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("text_similarity", model=str(tmp_path))
+    out = flow(("the cat sat", "the cat sat"))
+    assert abs(out["similarity"] - 1.0) < 1e-4
+    diff = flow(("the cat sat", "paris visited alice"))
+    assert diff["similarity"] <= 1.0
+
+
+def test_information_extraction_task(tmp_path):
+    torch.manual_seed(0)
+    cfg = ErnieConfig(vocab_size=len(VOCAB), hidden_size=32,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      intermediate_size=64, max_position_embeddings=64,
+                      hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
+    m = UIE(cfg)
+    # bias the pointers so every position fires: spans decode deterministically
+    with torch.no_grad():
+        m.linear_start.weight.zero_()
+        m.linear_start.bias.fill_(5.0)
+        m.linear_end.weight.zero_()
+        m.linear_end.bias.fill_(5.0)
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("information_extraction", model=str(tmp_path),
+                    schema=["person"])
+    out = flow("alice visited paris")
+    assert "person" in out
+    assert out["person"][0]["text"] == "alice"
+    assert all(0 <= s["probability"] <= 1 for s in out["person"])
+
+
+def test_unknown_and_pending_tasks():
+    with pytest.raises(ValueError):
+        Taskflow("bogus_task")
+    with pytest.raises(NotImplementedError):
+        Taskflow("dependency_parsing")
