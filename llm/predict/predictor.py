@@ -46,6 +46,7 @@ class PredictorArgument:
     mode: str = field(default="dynamic")
     benchmark: bool = field(default=False)
     quant_type: str = field(default="")  # "" | "fp8" | "weight_only_int8"
+    use_hipgraph: bool = field(default=False)  # capture decode in a hipGraph
 
 
 class BasePredictor:
@@ -101,6 +102,12 @@ class BlockInferencePredictor(BasePredictor):
             num_blocks, c.block_size, max_blocks_per_seq, config.batch_size)
         engine.allocate_caches(num_blocks, self.device)
         self.engine = engine.to(self.device)
+        if config.use_hipgraph and torch.cuda.is_available():
+            from paddlenlp_amd.experimental.fused_transformer import GraphDecodeRunner
+
+            self.decode_fn = GraphDecodeRunner(self.engine)
+        else:
+            self.decode_fn = self.engine.decode_step
 
     def _sample(self, logits, prev_ids=None):
         cfg = self.config
@@ -185,7 +192,7 @@ class BlockInferencePredictor(BasePredictor):
                 if not mgr.extend(s, 1):
                     mgr.preempt_longest()  # (v1: preempted request is dropped back to pending)
             bt = torch.stack([mgr.block_table[s] for s in slots]).to(self.device, torch.int32)
-            logits = self.engine.decode_step(input_ids, bt, lens_before)
+            logits = self.decode_fn(input_ids, bt, lens_before)
             maxlen = max(len(prev_tokens[s]) for s in slots)
             prev = torch.zeros(B, maxlen, dtype=torch.long, device=self.device)
             for i, s in enumerate(slots):
