@@ -48,7 +48,13 @@ class LlamaRMSNorm(nn.Module):
             self.weight.sequence_parallel = True
 
     def forward(self, x):
-        return ops.rms_norm(x, self.weight, self.eps)
+        out = ops.rms_norm(x, self.weight, self.eps)
+        # outlier-suppression shift (llm/utils/quant.py apply_shift): the
+        # following linears see shifted activations and compensate in bias
+        shift = getattr(self, "shift_bias", None)
+        if shift is not None:
+            out = out + shift
+        return out
 
 
 class LlamaRotaryEmbedding(nn.Module):
