@@ -57,3 +57,8 @@ from .qwen2_moe import (  # noqa: F401
     Qwen2MoeForCausalLM,
     Qwen2MoeModel,
 )
+from .deepseek_v2 import (  # noqa: F401
+    DeepseekV2Config,
+    DeepseekV2ForCausalLM,
+    DeepseekV2Model,
+)

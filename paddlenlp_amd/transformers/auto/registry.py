@@ -21,6 +21,8 @@ MODEL_REGISTRY = {
                 "causal_lm": "MixtralForCausalLM", "base": "MixtralModel"},
     "qwen2_moe": {"module": "qwen2_moe", "config": "Qwen2MoeConfig",
                   "causal_lm": "Qwen2MoeForCausalLM", "base": "Qwen2MoeModel"},
+    "deepseek_v2": {"module": "deepseek_v2", "config": "DeepseekV2Config",
+                    "causal_lm": "DeepseekV2ForCausalLM", "base": "DeepseekV2Model"},
     "mistral": {"module": "mistral", "config": "MistralConfig",
                 "causal_lm": "MistralForCausalLM", "base": "MistralModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
