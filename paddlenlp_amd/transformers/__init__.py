@@ -62,3 +62,8 @@ from .deepseek_v2 import (  # noqa: F401
     DeepseekV2ForCausalLM,
     DeepseekV2Model,
 )
+from .bart import (  # noqa: F401
+    BartConfig,
+    BartForConditionalGeneration,
+    BartModel,
+)

@@ -27,6 +27,8 @@ MODEL_REGISTRY = {
                 "causal_lm": "MistralForCausalLM", "base": "MistralModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
            "seq2seq_lm": "T5ForConditionalGeneration"},
+    "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",
+             "seq2seq_lm": "BartForConditionalGeneration"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",

@@ -1,0 +1,77 @@
+"""BART: post-LN encoder-decoder, offset positions, cached decode parity.
+
+Reference behavior: paddlenlp/transformers/bart/modeling.py.
+"""
+import torch
+
+from paddlenlp_amd.transformers import (
+    BartConfig,
+    BartForConditionalGeneration,
+    BartModel,
+)
+
+torch.manual_seed(0)
+
+
+def tiny_bart(**kw):
+    return BartConfig(vocab_size=120, d_model=32, encoder_layers=2,
+                      decoder_layers=2, encoder_attention_heads=4,
+                      decoder_attention_heads=4, encoder_ffn_dim=64,
+                      decoder_ffn_dim=64, max_position_embeddings=64,
+                      dropout=0.0, **kw)
+
+
+def test_bart_position_offset():
+    from paddlenlp_amd.transformers.bart.modeling import (
+        BartLearnedPositionalEmbedding,
+    )
+
+    emb = BartLearnedPositionalEmbedding(64, 32)
+    assert emb.num_embeddings == 66  # +2 reserved rows
+    out = emb(3)
+    torch.testing.assert_close(out[0], emb.weight[2])  # position 0 -> row 2
+
+
+def test_bart_training_and_shift_right():
+    m = BartForConditionalGeneration(tiny_bart())
+    src = torch.randint(3, 120, (2, 10))
+    labels = torch.randint(3, 120, (2, 6))
+    loss, logits = m(input_ids=src, labels=labels)
+    assert logits.shape == (2, 6, 120)
+    loss.backward()
+    assert m.bart.shared.weight.grad is not None
+    shifted = m._shift_right(labels)
+    assert (shifted[:, 0] == m.config.decoder_start_token_id).all()
+    assert (shifted[:, 1:] == labels[:, :-1]).all()
+
+
+def test_bart_cached_decode_parity():
+    m = BartForConditionalGeneration(tiny_bart()).eval()
+    src = torch.randint(3, 120, (2, 10))
+    dec = torch.randint(3, 120, (2, 5))
+    with torch.no_grad():
+        full = m(input_ids=src, decoder_input_ids=dec)
+        enc = m.bart.encoder(src)
+        past, outs = None, []
+        for t in range(dec.shape[1]):
+            lg, past, _ = m(decoder_input_ids=dec[:, t:t + 1],
+                            encoder_output=enc, past_key_values=past,
+                            use_cache=True)
+            outs.append(lg[:, 0])
+    torch.testing.assert_close(torch.stack(outs, 1), full, rtol=1e-4, atol=1e-4)
+
+
+def test_bart_generate_and_save_load(tmp_path):
+    m = BartForConditionalGeneration(tiny_bart()).eval()
+    src = torch.randint(3, 120, (2, 8))
+    out, _ = m.generate(src, max_new_tokens=5, do_sample=False)
+    assert out.shape[0] == 2 and out.shape[1] <= 5
+
+    m.save_pretrained(str(tmp_path))
+    m2 = BartForConditionalGeneration.from_pretrained(str(tmp_path)).eval()
+    assert m2.lm_head.weight.data_ptr() == m2.bart.shared.weight.data_ptr()
+    labels = torch.randint(3, 120, (2, 4))
+    with torch.no_grad():
+        a = m(input_ids=src, labels=labels)[1]
+        b = m2(input_ids=src, labels=labels)[1]
+    torch.testing.assert_close(a, b)
