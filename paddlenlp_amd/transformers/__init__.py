@@ -67,3 +67,6 @@ from .bart import (  # noqa: F401
     BartForConditionalGeneration,
     BartModel,
 )
+from .gemma import GemmaConfig, GemmaForCausalLM, GemmaModel  # noqa: F401
+from .opt import OPTConfig, OPTForCausalLM, OPTModel  # noqa: F401
+from .bloom import BloomConfig, BloomForCausalLM, BloomModel  # noqa: F401

@@ -25,6 +25,12 @@ MODEL_REGISTRY = {
                     "causal_lm": "DeepseekV2ForCausalLM", "base": "DeepseekV2Model"},
     "mistral": {"module": "mistral", "config": "MistralConfig",
                 "causal_lm": "MistralForCausalLM", "base": "MistralModel"},
+    "gemma": {"module": "gemma", "config": "GemmaConfig",
+              "causal_lm": "GemmaForCausalLM", "base": "GemmaModel"},
+    "opt": {"module": "opt", "config": "OPTConfig",
+            "causal_lm": "OPTForCausalLM", "base": "OPTModel"},
+    "bloom": {"module": "bloom", "config": "BloomConfig",
+              "causal_lm": "BloomForCausalLM", "base": "BloomModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
            "seq2seq_lm": "T5ForConditionalGeneration"},
     "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",
