@@ -70,3 +70,4 @@ from .bart import (  # noqa: F401
 from .gemma import GemmaConfig, GemmaForCausalLM, GemmaModel  # noqa: F401
 from .opt import OPTConfig, OPTForCausalLM, OPTModel  # noqa: F401
 from .bloom import BloomConfig, BloomForCausalLM, BloomModel  # noqa: F401
+from .falcon import FalconConfig, FalconForCausalLM, FalconModel  # noqa: F401

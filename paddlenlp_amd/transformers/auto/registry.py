@@ -31,6 +31,8 @@ MODEL_REGISTRY = {
             "causal_lm": "OPTForCausalLM", "base": "OPTModel"},
     "bloom": {"module": "bloom", "config": "BloomConfig",
               "causal_lm": "BloomForCausalLM", "base": "BloomModel"},
+    "falcon": {"module": "falcon", "config": "FalconConfig",
+               "causal_lm": "FalconForCausalLM", "base": "FalconModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
            "seq2seq_lm": "T5ForConditionalGeneration"},
     "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",

@@ -107,3 +107,24 @@ def test_bloom_alibi_monotone():
     bias = m.bloom._alibi(8, torch.device("cpu"))
     assert bias.shape == (1, 4, 1, 8)
     assert (bias[0, :, 0, 1:] <= bias[0, :, 0, :-1]).all()
+
+
+def test_falcon_parallel_residual():
+    from paddlenlp_amd.transformers import FalconConfig, FalconForCausalLM
+
+    cfg = FalconConfig(vocab_size=128, hidden_size=64, num_hidden_layers=2,
+                       num_attention_heads=4, num_key_value_heads=1,
+                       intermediate_size=128, max_position_embeddings=64)
+    m = FalconForCausalLM(cfg)
+    # parallel form: single LN per layer (no post_attention_layernorm)
+    assert not hasattr(m.falcon.h[0], "post_attention_layernorm")
+    _check_family(m)
+
+    # sequential form has both norms
+    cfg2 = FalconConfig(vocab_size=128, hidden_size=64, num_hidden_layers=1,
+                        num_attention_heads=4, num_key_value_heads=2,
+                        intermediate_size=128, max_position_embeddings=64,
+                        parallel_attn=False)
+    m2 = FalconForCausalLM(cfg2)
+    assert hasattr(m2.falcon.h[0], "post_attention_layernorm")
+    _check_family(m2)
