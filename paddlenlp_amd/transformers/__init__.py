@@ -71,3 +71,8 @@ from .gemma import GemmaConfig, GemmaForCausalLM, GemmaModel  # noqa: F401
 from .opt import OPTConfig, OPTForCausalLM, OPTModel  # noqa: F401
 from .bloom import BloomConfig, BloomForCausalLM, BloomModel  # noqa: F401
 from .falcon import FalconConfig, FalconForCausalLM, FalconModel  # noqa: F401
+from .chatglm_v2 import (  # noqa: F401
+    ChatGLMv2Config,
+    ChatGLMv2ForCausalLM,
+    ChatGLMv2Model,
+)

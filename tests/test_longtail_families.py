@@ -128,3 +128,25 @@ def test_falcon_parallel_residual():
     m2 = FalconForCausalLM(cfg2)
     assert hasattr(m2.falcon.h[0], "post_attention_layernorm")
     _check_family(m2)
+
+
+def test_chatglm_v2_partial_rope():
+    from paddlenlp_amd.transformers import ChatGLMv2Config, ChatGLMv2ForCausalLM
+    from paddlenlp_amd.transformers.chatglm_v2.modeling import _glm_rope
+
+    cfg = ChatGLMv2Config(vocab_size=128, hidden_size=64, num_hidden_layers=2,
+                          num_attention_heads=4, multi_query_group_num=2,
+                          ffn_hidden_size=128, kv_channels=16,
+                          max_position_embeddings=64)
+    m = ChatGLMv2ForCausalLM(cfg)
+    # rotary touches only the first half of the head dim
+    x = torch.randn(1, 4, 2, 16)
+    n = 16 // 4
+    inv = 1.0 / (10000.0 ** (torch.arange(n).float() / n))
+    freqs = torch.outer(torch.arange(4).float(), inv)
+    out = _glm_rope(x, freqs.cos(), freqs.sin())
+    torch.testing.assert_close(out[..., 8:], x[..., 8:])       # pass-through
+    assert not torch.allclose(out[..., :8], x[..., :8])        # rotated
+    # position 0 is identity everywhere
+    torch.testing.assert_close(out[:, 0], x[:, 0])
+    _check_family(m)
