@@ -43,12 +43,15 @@ from .trainer_callback import (
 )
 from .integrations import get_reporting_integration_callbacks
 from .plugins.timer import get_timers
+from ..utils.profiler import add_profiler_step
 from .trainer_utils import (
+    TrainerMemoryTracker,
     TrainOutput,
     caculate_llm_flops,
     get_last_checkpoint,
     get_scheduler,
     set_hybrid_seed,
+    should_skip_data,
     speed_metrics,
 )
 from .training_args import TrainingArguments
@@ -223,6 +226,8 @@ class Trainer:
     def train(self, resume_from_checkpoint: Optional[Union[str, bool]] = None):
         args = self.args
         self.is_in_train = True
+        mem_tracker = TrainerMemoryTracker(skip=args.skip_memory_metrics)
+        mem_tracker.start("train")
 
         train_dataloader = self.get_train_dataloader()
         steps_per_epoch = max(1, len(train_dataloader) // args.gradient_accumulation_steps)
@@ -298,6 +303,21 @@ class Trainer:
                 if accum_count == 0:
                     self.control = self.callback_handler.on_step_begin(args, self.state, self.control)
 
+                # corrupted-data replay jump: consume the batch, train nothing
+                if should_skip_data(self.state.global_step + 1,
+                                    args.skip_data_intervals):
+                    accum_count += 1
+                    self.state.consumed_samples += (
+                        args.per_device_train_batch_size * args.dataset_world_size)
+                    if accum_count >= args.gradient_accumulation_steps:
+                        accum_count = 0
+                        self.state.global_step += 1
+                        if self.state.global_step >= max_steps:
+                            done = True
+                            break
+                    continue
+
+                add_profiler_step()
                 timers = get_timers()
                 if is_pipeline:
                     # buffer micro-batches; the 1F1B engine consumes them as
@@ -352,6 +372,7 @@ class Trainer:
             num_samples=self.state.consumed_samples,
             num_steps=self.state.global_step,
         )
+        mem_tracker.stop_and_update_metrics(metrics)
         train_loss = self._total_loss_scalar / max(1, self.state.global_step)
         self.is_in_train = False
         return TrainOutput(self.state.global_step, train_loss, metrics)

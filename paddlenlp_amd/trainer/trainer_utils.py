@@ -180,3 +180,79 @@ class TrainOutput:
         self.global_step = global_step
         self.training_loss = training_loss
         self.metrics = metrics
+
+
+def should_skip_data(global_step: int, skip_data_intervals) -> bool:
+    """True when global_step (1-based, the step ABOUT to run) falls inside any
+    [start, end] interval (reference trainer.py:1007 should_skip_data) —
+    used to jump over corrupted data ranges on resume."""
+    if not skip_data_intervals:
+        return False
+    for interval in skip_data_intervals:
+        if len(interval) != 2 or interval[0] > interval[1]:
+            raise ValueError(f"invalid skip interval {interval}")
+        if interval[0] <= global_step <= interval[1]:
+            return True
+    return False
+
+
+class TrainerMemoryTracker:
+    """CPU RSS + device peak memory deltas per stage (reference
+    trainer_utils.py:726).  Usage: start() ... stop_and_update_metrics(m)."""
+
+    def __init__(self, skip: bool = False):
+        self.skip = skip
+        self._stage = None
+        try:
+            import psutil
+
+            self._proc = psutil.Process()
+        except ImportError:  # psutil is present in this image; belt+braces
+            self._proc = None
+            self.skip = True
+
+    def _cpu_mem(self) -> int:
+        return self._proc.memory_info().rss if self._proc else 0
+
+    def start(self, stage: str = "train"):
+        if self.skip:
+            return
+        import torch
+
+        self._stage = stage
+        self._cpu_begin = self._cpu_mem()
+        if torch.cuda.is_available():
+            torch.cuda.reset_peak_memory_stats()
+            self._gpu_begin = torch.cuda.memory_allocated()
+        else:
+            self._gpu_begin = None
+
+    def stop_and_update_metrics(self, metrics: dict):
+        if self.skip or self._stage is None:
+            return
+        import torch
+
+        stage = self._stage
+        metrics[f"{stage}_mem_cpu_rss"] = self._cpu_mem()
+        metrics[f"{stage}_mem_cpu_delta"] = self._cpu_mem() - self._cpu_begin
+        if self._gpu_begin is not None:
+            peak = torch.cuda.max_memory_allocated()
+            metrics[f"{stage}_mem_gpu_alloc_delta"] = (
+                torch.cuda.memory_allocated() - self._gpu_begin)
+            metrics[f"{stage}_mem_gpu_peaked_delta"] = peak - self._gpu_begin
+        self._stage = None
+
+
+def enable_determinism(seed: int = 42):
+    """Bitwise-reproducible training mode for CI (reference determinism
+    knobs FLAGS_cudnn_deterministic etc.): seeds everything, turns on
+    torch deterministic algorithms, MIOpen determinism, and routes custom
+    kernels with atomic reductions (rms_norm backward) to their
+    deterministic fallbacks via PNLP_DETERMINISTIC=1."""
+    import os
+
+    set_seed(seed)
+    os.environ["PNLP_DETERMINISTIC"] = "1"
+    os.environ.setdefault("MIOPEN_FIND_MODE", "1")
+    os.environ.setdefault("CUBLAS_WORKSPACE_CONFIG", ":4096:8")
+    torch.use_deterministic_algorithms(True, warn_only=True)

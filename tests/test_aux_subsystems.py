@@ -1,0 +1,108 @@
+"""Auxiliary subsystems: data-skip replay, memory tracker,
+profiler hook, determinism knob.
+"""
+import json
+import os
+
+import pytest
+import torch
+
+from paddlenlp_amd.trainer.trainer_utils import (
+    TrainerMemoryTracker,
+    enable_determinism,
+    should_skip_data,
+)
+
+
+def test_should_skip_data():
+    assert not should_skip_data(5, None)
+    assert not should_skip_data(5, [])
+    assert should_skip_data(5, [[3, 7]])
+    assert should_skip_data(3, [[3, 7]]) and should_skip_data(7, [[3, 7]])
+    assert not should_skip_data(8, [[3, 7]])
+    assert should_skip_data(20, [[3, 7], [20, 20]])
+    with pytest.raises(ValueError):
+        should_skip_data(1, [[9, 3]])
+
+
+def test_trainer_skips_intervals(tmp_path):
+    from paddlenlp_amd.trainer import Trainer
+    from paddlenlp_amd.trainer.training_args import TrainingArguments
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=2,
+                      num_key_value_heads=2, max_position_embeddings=32,
+                      dtype="float32")
+    model = LlamaForCausalLM.from_config(cfg)
+    data = [{"input_ids": torch.randint(0, 64, (8,)),
+             "labels": torch.randint(0, 64, (8,))} for _ in range(12)]
+    stepped = []
+    orig = Trainer.training_step
+
+    def spy(self, m, inputs):
+        stepped.append(self.state.global_step + 1)
+        return orig(self, m, inputs)
+
+    Trainer.training_step = spy
+    try:
+        args = TrainingArguments(
+            output_dir=str(tmp_path), do_train=True, max_steps=8,
+            per_device_train_batch_size=1, logging_steps=1000,
+            save_steps=1 << 30, skip_data_intervals=[[2, 3]], report_to=[])
+        trainer = Trainer(model=model, args=args, train_dataset=data)
+        trainer.train()
+    finally:
+        Trainer.training_step = orig
+    # steps 2 and 3 consumed data but ran no forward/backward
+    assert 2 not in stepped and 3 not in stepped
+    assert 1 in stepped and 4 in stepped
+    assert trainer.state.global_step == 8
+
+
+def test_memory_tracker():
+    m = {}
+    t = TrainerMemoryTracker(skip=False)
+    t.start("train")
+    _ = [bytearray(1 << 20) for _ in range(8)]
+    t.stop_and_update_metrics(m)
+    assert "train_mem_cpu_rss" in m and m["train_mem_cpu_rss"] > 0
+    # skip mode writes nothing
+    m2 = {}
+    t2 = TrainerMemoryTracker(skip=True)
+    t2.start("train")
+    t2.stop_and_update_metrics(m2)
+    assert m2 == {}
+
+
+def test_profiler_hook(tmp_path, monkeypatch):
+    import paddlenlp_amd.utils.profiler as prof
+
+    trace = tmp_path / "trace.json"
+    monkeypatch.setattr(prof, "_profiler", None)
+    monkeypatch.setattr(prof, "_batch_range", None)
+    monkeypatch.setattr(prof, "_step", 0)
+    opts = f"batch_range=[2, 3]; profile_path={trace}"
+    x = torch.randn(4, 4)
+    for _ in range(5):
+        prof.add_profiler_step(opts)
+        _ = x @ x
+    assert trace.exists()
+    data = json.loads(trace.read_text())
+    assert "traceEvents" in data
+
+
+def test_enable_determinism_routes_rms_norm(monkeypatch):
+    from paddlenlp_amd import ops
+
+    enable_determinism(7)
+    assert os.environ["PNLP_DETERMINISTIC"] == "1"
+    a = torch.randn(3, 8)
+    b = torch.randn(3, 8)
+    torch.manual_seed(7)
+    r1 = ops.rms_norm(a, torch.ones(8))
+    r2 = ops.rms_norm(a, torch.ones(8))
+    torch.testing.assert_close(r1, r2)
+    monkeypatch.delenv("PNLP_DETERMINISTIC", raising=False)
+    torch.use_deterministic_algorithms(False)
