@@ -76,3 +76,4 @@ from .chatglm_v2 import (  # noqa: F401
     ChatGLMv2ForCausalLM,
     ChatGLMv2Model,
 )
+from .mamba import MambaConfig, MambaForCausalLM, MambaModel  # noqa: F401

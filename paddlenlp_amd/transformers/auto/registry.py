@@ -35,6 +35,8 @@ MODEL_REGISTRY = {
                "causal_lm": "FalconForCausalLM", "base": "FalconModel"},
     "chatglm_v2": {"module": "chatglm_v2", "config": "ChatGLMv2Config",
                    "causal_lm": "ChatGLMv2ForCausalLM", "base": "ChatGLMv2Model"},
+    "mamba": {"module": "mamba", "config": "MambaConfig",
+              "causal_lm": "MambaForCausalLM", "base": "MambaModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
            "seq2seq_lm": "T5ForConditionalGeneration"},
     "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",
