@@ -62,6 +62,7 @@ from .tasks import (  # noqa: E402
     TextClassificationTask,
     TextSimilarityTask,
     TokenClassificationTask,
+    ZeroShotTextClassificationTask,
 )
 
 ENCODER_TASKS = {
@@ -74,13 +75,14 @@ ENCODER_TASKS = {
     "feature_extraction": FeatureExtractionTask,
     "text_similarity": TextSimilarityTask,
     "information_extraction": InformationExtractionTask,
+    "zero_shot_text_classification": ZeroShotTextClassificationTask,
 }
 
 # API-surface parity with the reference registry (taskflow/taskflow.py:48);
 # remaining pipelines need task models that don't exist offline.
 PENDING_TASKS = {
     "dependency_parsing", "document_intelligence", "knowledge_mining",
-    "text_correction", "zero_shot_text_classification", "word_segmentation",
+    "text_correction", "word_segmentation",
 }
 
 TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)

@@ -227,3 +227,63 @@ class InformationExtractionTask:
                 "probability": float(start_p[s] * end_p[e]),
             })
         return spans
+
+
+class ZeroShotTextClassificationTask:
+    """UTC zero-shot classification (reference taskflow
+    zero_shot_text_classification): prompt = [CLS] [O-MASK] option_1 ...
+    [O-MASK] option_n [SEP] text; the UTC head scores each [O-MASK]."""
+
+    def __init__(self, model_path: str, schema: Optional[List[str]] = None,
+                 omask_token: str = "[O-MASK]", single_label: bool = True,
+                 pred_threshold: float = 0.5, **kwargs):
+        from ..transformers import UTC
+
+        self.tokenizer = AutoTokenizer.from_pretrained(model_path)
+        self.model = UTC.from_pretrained(model_path)
+        if torch.cuda.is_available():
+            self.model = self.model.to("cuda:0")
+        self.model.eval()
+        self.device = next(self.model.parameters()).device
+        self.schema = schema or []
+        self.single_label = single_label
+        self.pred_threshold = pred_threshold
+        tok = self.tokenizer._tokenizer
+        self.omask_id = tok.token_to_id(omask_token)
+        self.cls_id = tok.token_to_id("[CLS]")
+        self.sep_id = tok.token_to_id("[SEP]")
+        assert self.omask_id is not None, f"{omask_token} not in vocab"
+
+    def set_schema(self, schema: List[str]):
+        self.schema = list(schema)
+
+    @torch.no_grad()
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        tok = self.tokenizer._tokenizer
+        results = []
+        for text in texts:
+            ids = [self.cls_id] if self.cls_id is not None else []
+            omask_positions = []
+            for option in self.schema:
+                omask_positions.append(len(ids))
+                ids.append(self.omask_id)
+                ids.extend(tok.encode(option).ids)
+            if self.sep_id is not None:
+                ids.append(self.sep_id)
+            ids.extend(tok.encode(text).ids)
+            input_ids = torch.tensor([ids], device=self.device)
+            om = torch.tensor([omask_positions], device=self.device)
+            logits = self.model(input_ids, omask_positions=om)[0]
+            if self.single_label:
+                probs = logits.float().softmax(-1)
+                idx = int(probs.argmax())
+                preds = [{"label": self.schema[idx], "score": float(probs[idx])}]
+            else:
+                probs = torch.sigmoid(logits.float())
+                preds = [{"label": o, "score": float(p)}
+                         for o, p in zip(self.schema, probs)
+                         if float(p) >= self.pred_threshold]
+            results.append({"text_a": text, "predictions": preds})
+        return results[0] if single else results

@@ -21,6 +21,7 @@ from .bert import (  # noqa: F401
 )
 from .ernie import (  # noqa: F401
     UIE,
+    UTC,
     ErnieConfig,
     ErnieForMaskedLM,
     ErnieForPretraining,

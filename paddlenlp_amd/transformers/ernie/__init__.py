@@ -1,6 +1,7 @@
 from .configuration import ErnieConfig
 from .modeling import (
     UIE,
+    UTC,
     ErnieForMaskedLM,
     ErnieForPretraining,
     ErnieForQuestionAnswering,

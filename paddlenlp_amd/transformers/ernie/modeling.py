@@ -30,6 +30,7 @@ __all__ = [
     "ErnieForPretraining",
     "ErniePretrainingCriterion",
     "UIE",
+    "UTC",
 ]
 
 
@@ -185,6 +186,37 @@ class ErniePretrainingCriterion(nn.Module):
             masked_lm_labels.view(-1), ignore_index=-100)
         nsp = F.cross_entropy(seq_relationship_logits, next_sentence_label.view(-1))
         return mlm + nsp
+
+
+class UTC(ErniePretrainedModel):
+    """Unified Tag Classification (reference :1286-1384): a query projection
+    of the [CLS] state dotted against key projections of each option's
+    [O-MASK] state gives one logit per candidate label — the backbone of
+    zero-shot text classification."""
+
+    def __init__(self, config: ErnieConfig):
+        super().__init__(config)
+        self.ernie = ErnieModel(config)
+        self.predict_size = 64
+        self.linear_q = nn.Linear(config.hidden_size, self.predict_size)
+        self.linear_k = nn.Linear(config.hidden_size, self.predict_size)
+
+    def forward(self, input_ids, token_type_ids=None, position_ids=None,
+                attention_mask=None, omask_positions=None, cls_positions=None):
+        """omask_positions [B, max_options] (0-padded); cls_positions [B]."""
+        sequence_output, _ = self.ernie(
+            input_ids, token_type_ids, position_ids, attention_mask)
+        B = sequence_output.shape[0]
+        batch_idx = torch.arange(B, device=sequence_output.device)
+        if cls_positions is None:
+            cls_positions = torch.zeros(B, dtype=torch.long,
+                                        device=sequence_output.device)
+        q = self.linear_q(sequence_output[batch_idx, cls_positions])  # [B, P]
+        option_states = sequence_output[
+            batch_idx[:, None], omask_positions]                       # [B, O, H]
+        k = self.linear_k(option_states)                               # [B, O, P]
+        logits = torch.einsum("bp,bop->bo", q, k) / self.predict_size ** 0.5
+        return logits
 
 
 class UIE(ErniePretrainedModel):

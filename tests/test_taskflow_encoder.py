@@ -125,3 +125,43 @@ def test_unknown_and_pending_tasks():
         Taskflow("bogus_task")
     with pytest.raises(NotImplementedError):
         Taskflow("dependency_parsing")
+
+
+def test_zero_shot_text_classification(tmp_path):
+    from paddlenlp_amd.transformers import UTC
+
+    torch.manual_seed(0)
+    cfg = ErnieConfig(vocab_size=len(VOCAB) + 2, hidden_size=32,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      intermediate_size=64, max_position_embeddings=64,
+                      hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
+    m = UTC(cfg)
+    m.save_pretrained(str(tmp_path))
+    from tokenizers import Tokenizer, models, pre_tokenizers
+
+    vocab = VOCAB + ["[O-MASK]", "[SEP]", "[CLS]"]
+    # keep ids in range: rebuild with the right size
+    cfg2 = ErnieConfig(vocab_size=len(vocab), hidden_size=32,
+                       num_hidden_layers=2, num_attention_heads=4,
+                       intermediate_size=64, max_position_embeddings=64,
+                       hidden_dropout_prob=0.0,
+                       attention_probs_dropout_prob=0.0)
+    m2 = UTC(cfg2)
+    m2.save_pretrained(str(tmp_path))
+    tok = Tokenizer(models.WordLevel({w: i for i, w in enumerate(vocab)},
+                                     unk_token="[UNK]"))
+    tok.pre_tokenizer = pre_tokenizers.WhitespaceSplit()
+    tok.save(str(tmp_path / "tokenizer.json"))
+
+    flow = Taskflow("zero_shot_text_classification", model=str(tmp_path),
+                    schema=["cat", "mat"])
+    out = flow("the cat sat")
+    assert len(out["predictions"]) == 1
+    assert out["predictions"][0]["label"] in ("cat", "mat")
+    assert 0 <= out["predictions"][0]["score"] <= 1
+
+    multi = Taskflow("zero_shot_text_classification", model=str(tmp_path),
+                     schema=["cat", "mat"], single_label=False,
+                     pred_threshold=0.0)
+    out2 = multi("the cat sat")
+    assert len(out2["predictions"]) == 2
