@@ -60,6 +60,7 @@ from .tasks import (  # noqa: E402
     FillMaskTask,
     InformationExtractionTask,
     TextClassificationTask,
+    TextCorrectionTask,
     TextSimilarityTask,
     TokenClassificationTask,
     ZeroShotTextClassificationTask,
@@ -76,13 +77,14 @@ ENCODER_TASKS = {
     "text_similarity": TextSimilarityTask,
     "information_extraction": InformationExtractionTask,
     "zero_shot_text_classification": ZeroShotTextClassificationTask,
+    "text_correction": TextCorrectionTask,
 }
 
 # API-surface parity with the reference registry (taskflow/taskflow.py:48);
 # remaining pipelines need task models that don't exist offline.
 PENDING_TASKS = {
     "dependency_parsing", "document_intelligence", "knowledge_mining",
-    "text_correction", "word_segmentation",
+    "word_segmentation",
 }
 
 TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)

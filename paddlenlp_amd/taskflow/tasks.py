@@ -287,3 +287,42 @@ class ZeroShotTextClassificationTask:
                          if float(p) >= self.pred_threshold]
             results.append({"text_a": text, "predictions": preds})
         return results[0] if single else results
+
+
+class TextCorrectionTask(_EncoderTaskBase):
+    """Spelling/word correction via masked-LM rescoring (reference taskflow
+    text_correction CSC pipeline): a token is flagged when the MLM strongly
+    prefers a different token at its position."""
+
+    auto_cls = AutoModelForMaskedLM
+
+    def __init__(self, model_path: str, threshold: float = 0.9, **kwargs):
+        super().__init__(model_path)
+        self.threshold = threshold
+
+    @torch.no_grad()
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        tok = self.tokenizer._tokenizer
+        results = []
+        for text in texts:
+            encoding = tok.encode(text)
+            ids = torch.tensor([encoding.ids], device=self.device)
+            logits = self.model(ids)
+            probs = logits[0].float().softmax(-1)
+            corrections = []
+            out_tokens = list(encoding.tokens)
+            for i, tid in enumerate(encoding.ids):
+                p_best, best = probs[i].max(-1)
+                if int(best) != tid and float(p_best) >= self.threshold:
+                    corrections.append({
+                        "position": i,
+                        "source": encoding.tokens[i],
+                        "target": tok.id_to_token(int(best)),
+                        "score": float(p_best),
+                    })
+                    out_tokens[i] = tok.id_to_token(int(best))
+            results.append({"source": text, "target": " ".join(out_tokens),
+                            "errors": corrections})
+        return results[0] if single else results

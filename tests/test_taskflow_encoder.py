@@ -165,3 +165,20 @@ def test_zero_shot_text_classification(tmp_path):
                      pred_threshold=0.0)
     out2 = multi("the cat sat")
     assert len(out2["predictions"]) == 2
+
+
+def test_text_correction(tmp_path):
+    torch.manual_seed(0)
+    m = BertForMaskedLM(_tiny_cfg())
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("text_correction", model=str(tmp_path), threshold=0.0)
+    out = flow("the cat sat")
+    assert set(out) == {"source", "target", "errors"}
+    for err in out["errors"]:
+        assert err["source"] != err["target"]
+    # a high threshold on a random model yields few/no corrections
+    strict = Taskflow("text_correction", model=str(tmp_path), threshold=0.9999)
+    out2 = strict("the cat sat")
+    assert isinstance(out2["errors"], list)
