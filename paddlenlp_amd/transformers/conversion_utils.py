@@ -46,9 +46,79 @@ def _fuse_llama_family(sd: Dict[str, torch.Tensor], prefix: str, config) -> Dict
     return out
 
 
+def _convert_bert(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
+    """HF BERT naming -> shared-encoder-core naming + fused qkv."""
+    import re
+
+    out = {}
+    qkv: Dict[str, dict] = {}
+    rules = [
+        (r"\.embeddings\.LayerNorm\.", ".embeddings.layer_norm."),
+        (r"\.encoder\.layer\.(\d+)\.attention\.output\.dense\.",
+         r".encoder.layers.\1.self_attn.out_proj."),
+        (r"\.encoder\.layer\.(\d+)\.attention\.output\.LayerNorm\.",
+         r".encoder.layers.\1.attn_norm."),
+        (r"\.encoder\.layer\.(\d+)\.intermediate\.dense\.",
+         r".encoder.layers.\1.fc_in."),
+        (r"\.encoder\.layer\.(\d+)\.output\.dense\.",
+         r".encoder.layers.\1.fc_out."),
+        (r"\.encoder\.layer\.(\d+)\.output\.LayerNorm\.",
+         r".encoder.layers.\1.mlp_norm."),
+        (r"cls\.predictions\.transform\.dense\.", "cls.dense."),
+        (r"cls\.predictions\.transform\.LayerNorm\.", "cls.layer_norm."),
+        (r"cls\.predictions\.decoder\.", "cls.decoder."),
+        (r"cls\.predictions\.bias$", "cls.decoder.bias"),
+    ]
+    for k, v in sd.items():
+        m = re.match(r"(.*\.encoder\.layer\.(\d+))\.attention\.self\.(query|key|value)\.(weight|bias)$", k)
+        if m:
+            layer_key = m.group(1).replace(".layer.", ".layers.") \
+                + ".self_attn.qkv_proj." + m.group(4)
+            qkv.setdefault(layer_key, {})[m.group(3)] = v
+            continue
+        nk = k
+        for pat, rep in rules:
+            nk = re.sub(pat, rep, nk)
+        out[nk] = v
+    for key, parts in qkv.items():
+        out[key] = torch.cat([parts["query"], parts["key"], parts["value"]], dim=0)
+    return out
+
+
+def _convert_t5(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
+    """HF T5 naming -> this framework's stack naming."""
+    import re
+
+    rules = [
+        (r"\.block\.(\d+)\.layer\.0\.SelfAttention\.", r".blocks.\1.self_attn."),
+        (r"\.block\.(\d+)\.layer\.0\.layer_norm\.", r".blocks.\1.self_norm."),
+        (r"\.block\.(\d+)\.layer\.1\.EncDecAttention\.", r".blocks.\1.cross_attn."),
+        (r"decoder\.block\.(\d+)\.layer\.1\.layer_norm\.", r"decoder.blocks.\1.cross_norm."),
+        (r"decoder\.block\.(\d+)\.layer\.2\.DenseReluDense\.", r"decoder.blocks.\1.ff."),
+        (r"decoder\.block\.(\d+)\.layer\.2\.layer_norm\.", r"decoder.blocks.\1.ff_norm."),
+        (r"encoder\.block\.(\d+)\.layer\.1\.DenseReluDense\.", r"encoder.blocks.\1.ff."),
+        (r"encoder\.block\.(\d+)\.layer\.1\.layer_norm\.", r"encoder.blocks.\1.ff_norm."),
+        (r"\.final_layer_norm\.", ".final_norm."),
+    ]
+    out = {}
+    for k, v in sd.items():
+        nk = k
+        for pat, rep in rules:
+            nk = re.sub(pat, rep, nk)
+        out[nk] = v
+    return out
+
+
 def convert_hf_state_dict(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
     """HF names (model.layers.N...) -> framework names + fused projections."""
     model_type = config.model_type
+    if model_type == "bert":
+        return _convert_bert(sd, config)
+    if model_type == "t5":
+        converted = _convert_t5(sd, config)
+        # T5ForConditionalGeneration nests the stacks under "t5."
+        return {("t5." + k if k.startswith(("encoder.", "decoder.", "shared."))
+                 else k): v for k, v in converted.items()}
     base_prefix = {"llama": "llama", "qwen2": "qwen2", "mistral": "mistral",
                    "mixtral": "mixtral"}.get(model_type, model_type)
     renamed = {}

@@ -204,3 +204,54 @@ def test_electra_tied_generator_head():
     gen = ElectraGenerator(cfg)
     assert gen.generator_lm_head.weight.data_ptr() == \
         gen.electra.embeddings.word_embeddings.weight.data_ptr()
+
+
+def test_hf_bert_name_conversion():
+    """Round-trip: export a model's weights under HF BERT naming, convert
+    back, reload — outputs must match exactly."""
+    import re
+
+    from paddlenlp_amd.transformers.conversion_utils import convert_hf_state_dict
+
+    m = BertForMaskedLM(tiny_bert()).eval()
+    sd = {k: v for k, v in m.state_dict().items()}
+
+    def to_hf(name, tensor):
+        # inverse mapping of _convert_bert for the names this model uses
+        n = name.replace(".embeddings.layer_norm.", ".embeddings.LayerNorm.")
+        n = re.sub(r"\.encoder\.layers\.(\d+)\.self_attn\.out_proj\.",
+                   r".encoder.layer.\1.attention.output.dense.", n)
+        n = re.sub(r"\.encoder\.layers\.(\d+)\.attn_norm\.",
+                   r".encoder.layer.\1.attention.output.LayerNorm.", n)
+        n = re.sub(r"\.encoder\.layers\.(\d+)\.fc_in\.",
+                   r".encoder.layer.\1.intermediate.dense.", n)
+        n = re.sub(r"\.encoder\.layers\.(\d+)\.fc_out\.",
+                   r".encoder.layer.\1.output.dense.", n)
+        n = re.sub(r"\.encoder\.layers\.(\d+)\.mlp_norm\.",
+                   r".encoder.layer.\1.output.LayerNorm.", n)
+        n = n.replace("cls.dense.", "cls.predictions.transform.dense.")
+        n = n.replace("cls.layer_norm.", "cls.predictions.transform.LayerNorm.")
+        n = n.replace("cls.decoder.", "cls.predictions.decoder.")
+        return n
+
+    hf_sd = {}
+    for k, v in sd.items():
+        mt = re.match(r"(.*\.encoder\.layers\.(\d+))\.self_attn\.qkv_proj\.(weight|bias)$", k)
+        if mt:
+            base = mt.group(1).replace(".layers.", ".layer.") + ".attention.self."
+            H = v.shape[0] // 3
+            hf_sd[base + "query." + mt.group(3)] = v[:H]
+            hf_sd[base + "key." + mt.group(3)] = v[H:2 * H]
+            hf_sd[base + "value." + mt.group(3)] = v[2 * H:]
+        else:
+            hf_sd[to_hf(k, v)] = v
+
+    converted = convert_hf_state_dict(hf_sd, m.config)
+    m2 = BertForMaskedLM(tiny_bert())
+    missing, unexpected = m2.load_state_dict(converted, strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    m2.eval()
+    ids = torch.randint(0, 120, (2, 10))
+    with torch.no_grad():
+        torch.testing.assert_close(m(ids), m2(ids))

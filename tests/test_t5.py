@@ -109,3 +109,50 @@ def test_t5_for_conditional_generation_save_load(tmp_path):
         a = m(input_ids=src, labels=labels)[1]
         b = m2(input_ids=src, labels=labels)[1]
     torch.testing.assert_close(a, b)
+
+
+def test_hf_t5_name_conversion():
+    import re
+
+    from paddlenlp_amd.transformers.conversion_utils import convert_hf_state_dict
+
+    m = T5ForConditionalGeneration(tiny_t5()).eval()
+
+    def to_hf(n):
+        n = re.sub(r"^t5\.", "", n)
+        n = re.sub(r"encoder\.blocks\.(\d+)\.self_attn\.",
+                   r"encoder.block.\1.layer.0.SelfAttention.", n)
+        n = re.sub(r"encoder\.blocks\.(\d+)\.self_norm\.",
+                   r"encoder.block.\1.layer.0.layer_norm.", n)
+        n = re.sub(r"encoder\.blocks\.(\d+)\.ff\.",
+                   r"encoder.block.\1.layer.1.DenseReluDense.", n)
+        n = re.sub(r"encoder\.blocks\.(\d+)\.ff_norm\.",
+                   r"encoder.block.\1.layer.1.layer_norm.", n)
+        n = re.sub(r"decoder\.blocks\.(\d+)\.self_attn\.",
+                   r"decoder.block.\1.layer.0.SelfAttention.", n)
+        n = re.sub(r"decoder\.blocks\.(\d+)\.self_norm\.",
+                   r"decoder.block.\1.layer.0.layer_norm.", n)
+        n = re.sub(r"decoder\.blocks\.(\d+)\.cross_attn\.",
+                   r"decoder.block.\1.layer.1.EncDecAttention.", n)
+        n = re.sub(r"decoder\.blocks\.(\d+)\.cross_norm\.",
+                   r"decoder.block.\1.layer.1.layer_norm.", n)
+        n = re.sub(r"decoder\.blocks\.(\d+)\.ff\.",
+                   r"decoder.block.\1.layer.2.DenseReluDense.", n)
+        n = re.sub(r"decoder\.blocks\.(\d+)\.ff_norm\.",
+                   r"decoder.block.\1.layer.2.layer_norm.", n)
+        n = n.replace(".final_norm.", ".final_layer_norm.")
+        return n
+
+    hf_sd = {to_hf(k): v for k, v in m.state_dict().items()}
+    converted = convert_hf_state_dict(hf_sd, m.config)
+    m2 = T5ForConditionalGeneration(tiny_t5())
+    missing, unexpected = m2.load_state_dict(converted, strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    m2.tie_weights()
+    m2.eval()
+    src = torch.randint(0, 100, (2, 8))
+    dec = torch.randint(0, 100, (2, 4))
+    with torch.no_grad():
+        torch.testing.assert_close(m(src, decoder_input_ids=dec),
+                                   m2(src, decoder_input_ids=dec))
