@@ -7,7 +7,6 @@ masked-LM, MLM+NSP pretraining with BertPretrainingCriterion).
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

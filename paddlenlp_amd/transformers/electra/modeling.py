@@ -14,7 +14,6 @@ import torch.nn.functional as F
 from ..encoder import (
     ACT2FN,
     EncoderEmbeddings,
-    LMPredictionHead,
     TransformerEncoder,
     init_encoder_weights,
 )

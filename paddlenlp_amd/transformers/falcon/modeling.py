@@ -8,7 +8,6 @@ MQA/GQA rotary attention reuses the Llama GQA path.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
