@@ -54,10 +54,99 @@ class ZeroShardedEngine:
             f"{sum(p.numel() for p in self.owned_params):,} / {sum(sizes):,} params"
         )
 
+        # ---- optional comm/compute overlap (enable_overlap_comm) ----
+        # Buckets in REVERSE parameter order (the backward pass produces
+        # grads last-layer-first) so each bucket's reduce launches as soon
+        # as its grads exist, overlapping with the rest of backward.
+        self._overlap = False
+        self.overlap_active = False
+        self._param_list = params
+        self._buckets = []          # dicts: params, owner_rank, fired
+        self._param_bucket = {}
+        self._pending = []          # (work, flat, grads, owner_rank)
+        self._hooks = []
+
+    # ------------------------------------------------------------------
+    # comm overlap
+    # ------------------------------------------------------------------
+    def enable_overlap_comm(self):
+        """Register per-param hooks that reduce each owner-bucket
+        asynchronously the moment its last gradient accumulates (only while
+        `overlap_active`, i.e. during the final micro-batch's backward)."""
+        if self.group is None or self._overlap:
+            return
+        self._overlap = True
+        per_owner = {r: [] for r in range(self.world)}
+        for p in reversed(self._param_list):
+            per_owner[self.owner[p]].append(p)
+        for owner_rank, plist in per_owner.items():
+            bucket, size = [], 0
+            for p in plist:
+                bucket.append(p)
+                size += p.numel() * p.element_size()
+                if size >= self.bucket_bytes:
+                    self._buckets.append(
+                        {"params": bucket, "owner_rank": owner_rank, "fired": 0})
+                    bucket, size = [], 0
+            if bucket:
+                self._buckets.append(
+                    {"params": bucket, "owner_rank": owner_rank, "fired": 0})
+        for bi, b in enumerate(self._buckets):
+            for p in b["params"]:
+                self._param_bucket[p] = bi
+                self._hooks.append(p.register_post_accumulate_grad_hook(
+                    self._on_grad_ready))
+        logger.info(f"ZeRO: overlap enabled over {len(self._buckets)} buckets")
+
+    def _on_grad_ready(self, p):
+        if not self.overlap_active:
+            return
+        b = self._buckets[self._param_bucket[p]]
+        b["fired"] += 1
+        if b["fired"] == len(b["params"]):
+            self._launch_bucket(b)
+
+    def _launch_bucket(self, b):
+        grads = [p.grad for p in b["params"] if p.grad is not None]
+        if not grads:
+            b["fired"] = -1  # mark done
+            return
+        flat = torch._utils._flatten_dense_tensors(grads)
+        if self.stage == 1:
+            work = dist.all_reduce(flat, group=self.group, async_op=True)
+        else:
+            work = dist.reduce(flat, dst=self.group_ranks[b["owner_rank"]],
+                               group=self.group, async_op=True)
+        self._pending.append((work, flat, grads, b["owner_rank"]))
+        b["fired"] = -1
+
+    def _finish_overlap(self):
+        # launch any bucket that never completed (params without grads)
+        for b in self._buckets:
+            if b["fired"] >= 0:
+                self._launch_bucket(b)
+        for work, flat, grads, owner_rank in self._pending:
+            work.wait()
+            if self.stage == 1 or owner_rank == self.rank:
+                flat.div_(self.world)
+                for g, out in zip(grads,
+                                  torch._utils._unflatten_dense_tensors(flat, grads)):
+                    g.copy_(out)
+        self._pending.clear()
+        for b in self._buckets:
+            b["fired"] = 0
+        self.overlap_active = False
+
     # ------------------------------------------------------------------
     def reduce_gradients_and_step_pre(self):
         """Reduce grads over the sharding group, then drop non-owned grads."""
         if self.group is None:
+            return
+        if self._overlap and (self.overlap_active or self._pending):
+            self._finish_overlap()
+            for p in self.owner:
+                if self.owner[p] != self.rank:
+                    p.grad = None
             return
         params = [p for p in self.owner if p.grad is not None]
         if self.stage == 1:

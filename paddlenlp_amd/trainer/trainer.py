@@ -259,6 +259,8 @@ class Trainer:
                 group=self.topology.sharding_parallel_group,
                 bucket_mb=args.sharding_comm_buffer_size_MB,
             )
+            if args.sharding_overlap_comm and self.topology.pp_degree == 1:
+                self._zero.enable_overlap_comm()
         elif getattr(self, "_zero_stage", 0) == 3:
             from ..parallel.zero3 import Zero3Engine
 
@@ -326,6 +328,11 @@ class Trainer:
                     pipe_buffer.append(self._prepare_inputs(inputs))
                 else:
                     timers("forward-backward").start()
+                    if (self._zero is not None
+                            and getattr(self._zero, "_overlap", False)
+                            and accum_count == args.gradient_accumulation_steps - 1):
+                        # final micro-batch: buckets reduce during backward
+                        self._zero.overlap_active = True
                     loss = self.training_step(model, inputs)
                     timers("forward-backward").stop()
                     tr_loss += loss.detach()

@@ -70,6 +70,8 @@ class TrainingArguments:
     context_parallel_degree: int = 1
     sharding: str = ""  # "stage1" | "stage2" | "stage3" | "" (space-separated options)
     sharding_comm_buffer_size_MB: int = 256
+    # overlap the sharding-group gradient reduction with backward compute
+    sharding_overlap_comm: bool = True
     # [[start, end], ...] 1-based global-step intervals whose data is skipped
     # (corrupted-range replay jump, reference trainer.py should_skip_data)
     skip_data_intervals: Optional[List[List[int]]] = None

@@ -371,3 +371,68 @@ def test_zero3_parity():
 
 def test_zero3_llama_grad_accum_parity():
     _run_workers(_w_zero3_llama_accum)
+
+
+def _zero_overlap_parity_body(rank, world, stage):
+    """Overlapped (hook-launched async) reduction == the synchronous path."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.zero import ZeroShardedEngine
+    from paddlenlp_amd.parallel.data_parallel import broadcast_parameters
+    from paddlenlp_amd.trainer.optimizer import FusedAdamW
+
+    topo = init_parallel_env(sharding_degree=world, backend="gloo")
+
+    def build():
+        torch.manual_seed(321)
+        return torch.nn.Sequential(
+            torch.nn.Linear(16, 32), torch.nn.Tanh(), torch.nn.Linear(32, 4)
+        )
+
+    torch.manual_seed(2000)
+    xs = [torch.randn(world * 4, 16) for _ in range(4)]
+    ys = [torch.randn(world * 4, 4) for _ in range(4)]
+
+    def run(overlap):
+        model = build()
+        broadcast_parameters(model, topo.sharding_parallel_group)
+        opt = FusedAdamW(model.parameters(), lr=1e-2, master_weights=False)
+        # tiny buckets force several per-owner buckets in both paths
+        zero = ZeroShardedEngine(model, opt, stage=stage,
+                                 group=topo.sharding_parallel_group,
+                                 bucket_mb=0)
+        if overlap:
+            zero.enable_overlap_comm()
+        for x, y in zip(xs, ys):
+            xl = x[rank * 4:(rank + 1) * 4]
+            yl = y[rank * 4:(rank + 1) * 4]
+            opt.zero_grad(set_to_none=True)
+            if overlap:
+                zero.overlap_active = True
+            loss = ((model(xl) - yl) ** 2).mean()
+            loss.backward()
+            zero.reduce_gradients_and_step_pre()
+            opt.step()
+            zero.step_post()
+        return {n: p.detach().clone() for n, p in model.named_parameters()}
+
+    sync_params = run(False)
+    overlap_params = run(True)
+    for n in sync_params:
+        assert torch.allclose(sync_params[n], overlap_params[n], atol=1e-6), \
+            (n, (sync_params[n] - overlap_params[n]).abs().max())
+
+
+def _w_zero2_overlap(rank, world):
+    _zero_overlap_parity_body(rank, world, 2)
+
+
+def _w_zero1_overlap(rank, world):
+    _zero_overlap_parity_body(rank, world, 1)
+
+
+def test_zero2_overlap_comm_parity():
+    _run_workers(_w_zero2_overlap)
+
+
+def test_zero1_overlap_comm_parity():
+    _run_workers(_w_zero1_overlap)
