@@ -47,6 +47,9 @@ class ModelArgument:
     prefix_tuning: bool = field(default=False)
     num_prefix_tokens: int = field(default=16)
     flash_mask: bool = field(default=False)
+    # QLoRA: quantize the frozen base to 4-bit before wrapping with LoRA
+    # (reference run_finetune weight_quantize_algo; "" disables)
+    weight_quantize_algo: str = field(default="")
 
 
 @dataclass
@@ -72,6 +75,15 @@ def main():
 
     if model_args.flash_mask and not data_args.zero_padding:
         raise ValueError("flash_mask requires zero_padding (reference run_finetune.py:182-184)")
+
+    if model_args.weight_quantize_algo:
+        from paddlenlp_amd.quantization import (
+            QuantizationConfig, replace_with_quantization_linear)
+
+        replace_with_quantization_linear(
+            model, QuantizationConfig(
+                weight_quantize_algo=model_args.weight_quantize_algo))
+        logger.info(f"base model quantized with {model_args.weight_quantize_algo}")
 
     if model_args.lora:
         lora_config = LoRAConfig(

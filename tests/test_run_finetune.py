@@ -82,3 +82,9 @@ def test_run_finetune_lora(sft_setup):
 def test_run_finetune_zero_padding_flashmask(sft_setup):
     out = _run_finetune(sft_setup, ("--zero_padding", "true", "--flash_mask", "true"))
     assert (out / "model.safetensors").is_file()
+
+
+def test_run_finetune_qlora(sft_setup):
+    out = _run_finetune(sft_setup, ("--lora", "true", "--lora_rank", "4",
+                                    "--weight_quantize_algo", "nf4"))
+    assert (out / "lora_model_state.safetensors").is_file()
