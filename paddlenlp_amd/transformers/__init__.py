@@ -78,3 +78,4 @@ from .chatglm_v2 import (  # noqa: F401
     ChatGLMv2Model,
 )
 from .mamba import MambaConfig, MambaForCausalLM, MambaModel  # noqa: F401
+from .gptj import GPTJConfig, GPTJForCausalLM, GPTJModel  # noqa: F401

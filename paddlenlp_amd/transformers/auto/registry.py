@@ -37,6 +37,8 @@ MODEL_REGISTRY = {
                    "causal_lm": "ChatGLMv2ForCausalLM", "base": "ChatGLMv2Model"},
     "mamba": {"module": "mamba", "config": "MambaConfig",
               "causal_lm": "MambaForCausalLM", "base": "MambaModel"},
+    "gptj": {"module": "gptj", "config": "GPTJConfig",
+             "causal_lm": "GPTJForCausalLM", "base": "GPTJModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
            "seq2seq_lm": "T5ForConditionalGeneration"},
     "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",

@@ -150,3 +150,25 @@ def test_chatglm_v2_partial_rope():
     # position 0 is identity everywhere
     torch.testing.assert_close(out[:, 0], x[:, 0])
     _check_family(m)
+
+
+def test_gptj_partial_rope_parallel_residual():
+    from paddlenlp_amd.transformers import GPTJConfig, GPTJForCausalLM
+    from paddlenlp_amd.transformers.gptj.modeling import _gptj_rope
+
+    cfg = GPTJConfig(vocab_size=128, hidden_size=64, num_hidden_layers=2,
+                     num_attention_heads=4, rotary_dim=8,
+                     intermediate_size=128, max_position_embeddings=64)
+    m = GPTJForCausalLM(cfg)
+    # untied head with bias
+    assert m.lm_head.bias is not None
+    assert m.lm_head.weight.data_ptr() != m.gptj.wte.weight.data_ptr()
+    # rope leaves dims >= rotary_dim untouched; position 0 is identity
+    x = torch.randn(1, 4, 2, 16)
+    n = 8 // 2
+    inv = 1.0 / (10000.0 ** (torch.arange(n).float() * 2 / 8))
+    freqs = torch.outer(torch.arange(4).float(), inv)
+    out = _gptj_rope(x, freqs.cos(), freqs.sin(), 8)
+    torch.testing.assert_close(out[..., 8:], x[..., 8:])
+    torch.testing.assert_close(out[:, 0], x[:, 0])
+    _check_family(m)
