@@ -56,6 +56,7 @@ GENERATION_TASKS = set(_GenerationTask.TEMPLATES)
 
 # encoder-backed pipelines (taskflow/tasks.py): task name -> class
 from .tasks import (  # noqa: E402
+    DependencyParsingTask,
     FeatureExtractionTask,
     FillMaskTask,
     InformationExtractionTask,
@@ -63,6 +64,7 @@ from .tasks import (  # noqa: E402
     TextCorrectionTask,
     TextSimilarityTask,
     TokenClassificationTask,
+    WordSegmentationTask,
     ZeroShotTextClassificationTask,
 )
 
@@ -78,13 +80,14 @@ ENCODER_TASKS = {
     "information_extraction": InformationExtractionTask,
     "zero_shot_text_classification": ZeroShotTextClassificationTask,
     "text_correction": TextCorrectionTask,
+    "word_segmentation": WordSegmentationTask,
+    "dependency_parsing": DependencyParsingTask,
 }
 
 # API-surface parity with the reference registry (taskflow/taskflow.py:48);
 # remaining pipelines need task models that don't exist offline.
 PENDING_TASKS = {
-    "dependency_parsing", "document_intelligence", "knowledge_mining",
-    "word_segmentation",
+    "document_intelligence", "knowledge_mining",
 }
 
 TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)

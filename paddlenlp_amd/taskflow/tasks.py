@@ -326,3 +326,71 @@ class TextCorrectionTask(_EncoderTaskBase):
             results.append({"source": text, "target": " ".join(out_tokens),
                             "errors": corrections})
         return results[0] if single else results
+
+
+class WordSegmentationTask(TokenClassificationTask):
+    """Sequence-labeling word segmentation (reference taskflow
+    word_segmentation): B/I labels aggregate characters/subtokens into
+    words; returns the word list."""
+
+    @torch.no_grad()
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        results = []
+        for text in texts:
+            encoding = self.tokenizer._tokenizer.encode(text)
+            ids = torch.tensor([encoding.ids], device=self.device)
+            logits = self.model(ids)
+            pred = logits[0].argmax(-1).tolist()
+            labels = [self.label_map.get(p, str(p)) for p in pred]
+            words = []
+            cur = None
+            for (s, e), lab in zip(encoding.offsets, labels):
+                if lab.startswith("B") or cur is None:
+                    if cur is not None:
+                        words.append(text[cur[0]:cur[1]])
+                    cur = [s, e]
+                else:  # I: extend the current word
+                    cur[1] = e
+            if cur is not None:
+                words.append(text[cur[0]:cur[1]])
+            results.append(words)
+        return results[0] if single else results
+
+
+class DependencyParsingTask:
+    """DDParser-style biaffine dependency parsing (reference taskflow
+    dependency_parsing.py): greedy head + relation decoding per token."""
+
+    DEFAULT_RELS = ["ATT", "SBV", "VOB", "ADV", "CMP", "COO", "POB", "MT",
+                    "HED", "IC", "DE", "DI", "DOB", "F", "DBL", "VV"]
+
+    def __init__(self, model_path: str, rel_labels: Optional[List[str]] = None,
+                 **kwargs):
+        from .models import BiAffineParser
+
+        self.tokenizer = AutoTokenizer.from_pretrained(model_path)
+        self.rel_labels = rel_labels or self.DEFAULT_RELS
+        self.model = BiAffineParser.from_pretrained(
+            model_path, n_rels=len(self.rel_labels))
+        if torch.cuda.is_available():
+            self.model = self.model.to("cuda:0")
+        self.model.eval()
+        self.device = next(self.model.parameters()).device
+
+    @torch.no_grad()
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        results = []
+        for text in texts:
+            encoding = self.tokenizer._tokenizer.encode(text)
+            ids = torch.tensor([encoding.ids], device=self.device)
+            heads, rels = self.model.decode(ids)
+            results.append({
+                "word": list(encoding.tokens),
+                "head": [int(h) for h in heads[0]],
+                "deprel": [self.rel_labels[int(r)] for r in rels[0]],
+            })
+        return results[0] if single else results

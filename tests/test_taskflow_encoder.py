@@ -124,7 +124,7 @@ def test_unknown_and_pending_tasks():
     with pytest.raises(ValueError):
         Taskflow("bogus_task")
     with pytest.raises(NotImplementedError):
-        Taskflow("dependency_parsing")
+        Taskflow("document_intelligence")
 
 
 def test_zero_shot_text_classification(tmp_path):
@@ -182,3 +182,44 @@ def test_text_correction(tmp_path):
     strict = Taskflow("text_correction", model=str(tmp_path), threshold=0.9999)
     out2 = strict("the cat sat")
     assert isinstance(out2["errors"], list)
+
+
+def test_word_segmentation(tmp_path):
+    torch.manual_seed(0)
+    m = BertForTokenClassification(_tiny_cfg(num_labels=2))
+    m.config.id2label = {0: "B", 1: "I"}
+    with torch.no_grad():
+        m.classifier.weight.zero_()
+        m.classifier.bias.copy_(torch.tensor([1.0, -1.0]))  # all B
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+    flow = Taskflow("word_segmentation", model=str(tmp_path))
+    out = flow("the cat sat")
+    assert out == ["the", "cat", "sat"]
+
+
+def test_dependency_parsing(tmp_path):
+    from paddlenlp_amd.taskflow.models import BiAffineParser
+
+    torch.manual_seed(0)
+    cfg = ErnieConfig(vocab_size=len(VOCAB), hidden_size=32,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      intermediate_size=64, max_position_embeddings=64,
+                      hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
+    parser = BiAffineParser(cfg, n_rels=16)
+
+    # training step: arc + rel loss backward
+    ids = torch.randint(0, len(VOCAB), (2, 6))
+    arc_labels = torch.randint(0, 6, (2, 6))
+    rel_labels = torch.randint(0, 16, (2, 6))
+    loss, arc, rel = parser(ids, arc_labels=arc_labels, rel_labels=rel_labels)
+    assert arc.shape == (2, 6, 6) and rel.shape == (2, 16, 6, 6)
+    loss.backward()
+
+    parser.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+    flow = Taskflow("dependency_parsing", model=str(tmp_path))
+    out = flow("the cat sat")
+    assert set(out) == {"word", "head", "deprel"}
+    assert len(out["word"]) == len(out["head"]) == len(out["deprel"]) == 3
+    assert all(0 <= h < 3 for h in out["head"])
