@@ -5,3 +5,11 @@ from .data_collator import (  # noqa: F401
     default_data_collator,
 )
 from .sampler import DistributedBatchSampler  # noqa: F401
+from .collate import (  # noqa: F401
+    Dict,
+    JiebaLikeTokenizer,
+    Pad,
+    Stack,
+    Tuple,
+    Vocab,
+)

@@ -118,3 +118,49 @@ def test_flashmask_semantics_match_block_diagonal():
     ref2 = reference.flash_attention(q[:, s1:], k[:, s1:], v[:, s1:], causal=True)
     assert torch.allclose(out[:, :s1], ref1, atol=1e-5)
     assert torch.allclose(out[:, s1:], ref2, atol=1e-5)
+
+
+def test_collate_stack_pad_tuple_dict():
+    import numpy as np
+
+    from paddlenlp_amd.data import Dict, Pad, Stack, Tuple
+
+    stack = Stack(dtype="int64")
+    out = stack([[1, 2], [3, 4]])
+    assert out.shape == (2, 2) and out.dtype == np.int64
+
+    pad = Pad(pad_val=9, ret_length=True)
+    batch, lens = pad([[1, 2, 3], [4]])
+    assert batch.tolist() == [[1, 2, 3], [4, 9, 9]]
+    assert lens.tolist() == [3, 1]
+    left = Pad(pad_val=0, pad_right=False)
+    assert left([[1, 2], [3]]).tolist() == [[1, 2], [0, 3]]
+
+    tup = Tuple(Stack(), Pad(pad_val=0))
+    a, b = tup([([1, 2], [5]), ([3, 4], [6, 7])])
+    assert a.tolist() == [[1, 2], [3, 4]]
+    assert b.tolist() == [[5, 0], [6, 7]]
+
+    dic = Dict({"x": Stack(), "y": Pad(pad_val=-1)})
+    a, b = dic([{"x": [1], "y": [2, 3]}, {"x": [4], "y": [5]}])
+    assert a.tolist() == [[1], [4]]
+    assert b.tolist() == [[2, 3], [5, -1]]
+
+
+def test_vocab_and_greedy_tokenizer():
+    from paddlenlp_amd.data import JiebaLikeTokenizer, Vocab
+
+    vocab = Vocab.build_vocab(
+        [["hello", "world"], ["hello", "there"]],
+        unk_token="[UNK]", pad_token="[PAD]")
+    assert vocab.to_indices("hello") != vocab.to_indices("[UNK]")
+    assert vocab.to_indices("missing") == vocab.to_indices("[UNK]")
+    assert vocab.to_tokens(vocab.to_indices(["hello", "world"])) == ["hello", "world"]
+    assert "hello" in vocab and len(vocab) >= 4
+
+    cn_vocab = Vocab(token_to_idx={"深度": 0, "学习": 1, "深": 2, "度": 3,
+                                   "[UNK]": 4}, unk_token="[UNK]")
+    tok = JiebaLikeTokenizer(cn_vocab)
+    assert tok.cut("深度学习") == ["深度", "学习"]  # longest match wins
+    assert tok.cut("深度x") == ["深度", "x"]
+    assert tok.encode("深度学习") == [0, 1]
