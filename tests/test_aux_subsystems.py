@@ -106,3 +106,52 @@ def test_enable_determinism_routes_rms_norm(monkeypatch):
     torch.testing.assert_close(r1, r2)
     monkeypatch.delenv("PNLP_DETERMINISTIC", raising=False)
     torch.use_deterministic_algorithms(False)
+
+
+def test_llm_utils_rotary_and_read_res():
+    import queue
+
+    from paddlenlp_amd.utils.llm_utils import (
+        get_rotary_position_embedding,
+        read_res,
+    )
+
+    pos = torch.arange(8)[None]
+    rot = get_rotary_position_embedding(pos, head_dim=16)
+    assert rot.shape == (2, 1, 8, 1, 8)
+    torch.testing.assert_close(rot[0, 0, 0, 0], torch.ones(8))  # cos(0)
+    # linear scaling halves the effective position
+    lin = get_rotary_position_embedding(pos, 16, rope_scaling={"type": "linear", "factor": 2.0})
+    torch.testing.assert_close(lin[:, :, 2], rot[:, :, 1])
+
+    class Tok:
+        eos_token_id = 99
+
+        def decode(self, ids, skip_special_tokens=True):
+            return " ".join(str(i) for i in ids)
+
+    rq, oq = queue.Queue(), queue.Queue()
+    rq.put([(0, 5), (1, 7)])
+    rq.put([(0, 6), (1, 99)])   # seq 1 ends
+    rq.put(None)
+    read_res(Tok(), rq, oq)
+    results = dict(oq.get_nowait() for _ in range(2))
+    assert results[1] == "7"
+    assert results[0] == "5 6"   # flushed at stream end
+
+
+def test_import_utils_and_tools():
+    from paddlenlp_amd.utils.import_utils import (
+        import_module,
+        is_package_available,
+    )
+    from paddlenlp_amd.utils.tools import compare_version, get_env_device
+
+    assert is_package_available("torch")
+    assert not is_package_available("definitely_not_a_module_xyz")
+    assert import_module("math") is not None
+    assert import_module("not_a_module_abc") is None
+    assert get_env_device() in ("cpu", "gpu")
+    assert compare_version("2.10.0", "2.9.1") == 1
+    assert compare_version("1.0", "1.0.0") == 0
+    assert compare_version("1.2", "1.10") == -1
