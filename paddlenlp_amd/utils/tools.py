@@ -28,7 +28,10 @@ def is_gfx950() -> bool:
 
 
 def compare_version(version: str, target: str) -> int:
-    """-1 / 0 / 1 comparison of dotted version strings."""
+    """-1 / 0 / 1 comparison of dotted version strings ('1.0' == '1.0.0')."""
     a = [int(x) for x in version.split(".")[:3] if x.isdigit()]
     b = [int(x) for x in target.split(".")[:3] if x.isdigit()]
+    n = max(len(a), len(b))
+    a += [0] * (n - len(a))
+    b += [0] * (n - len(b))
     return (a > b) - (a < b)
