@@ -48,8 +48,10 @@ class FusedMultiTransformerConfig:
 # ---------------------------------------------------------------------------
 # CPU reference versions of the paged kernels (test oracle)
 # ---------------------------------------------------------------------------
-def paged_decode_attn_ref(q, k_cache, v_cache, block_table, seq_lens):
-    """q [B, Hq, D] -> out [B, Hq, D] (fp32 math)."""
+def paged_decode_attn_ref(q, k_cache, v_cache, block_table, seq_lens,
+                          k_scale=None, v_scale=None):
+    """q [B, Hq, D] -> out [B, Hq, D] (fp32 math; int8 caches dequant with
+    the per-(token, head) scales)."""
     B, Hq, D = q.shape
     bs = k_cache.shape[1]
     Hk = k_cache.shape[2]
@@ -60,6 +62,11 @@ def paged_decode_attn_ref(q, k_cache, v_cache, block_table, seq_lens):
         blocks = block_table[b, :nb].long()
         k = k_cache[blocks].reshape(-1, Hk, D)[:L].float()  # [L, Hk, D]
         v = v_cache[blocks].reshape(-1, Hk, D)[:L].float()
+        if k_scale is not None:
+            ks = k_scale[blocks].reshape(-1, Hk)[:L].float()
+            vs = v_scale[blocks].reshape(-1, Hk)[:L].float()
+            k = k * ks[:, :, None]
+            v = v * vs[:, :, None]
         rep = Hq // Hk
         k = k.repeat_interleave(rep, dim=1)
         v = v.repeat_interleave(rep, dim=1)
@@ -71,7 +78,8 @@ def paged_decode_attn_ref(q, k_cache, v_cache, block_table, seq_lens):
 
 
 def rope_cache_append_ref(qkv, k_cache, v_cache, block_table, seq_lens_before,
-                          cos_t, sin_t, Hq, Hk, token_counts=None):
+                          cos_t, sin_t, Hq, Hk, token_counts=None,
+                          k_scale=None, v_scale=None):
     """CPU mirror of the fused kernel; returns roped q [B, T, Hq, D]."""
     B, T, _ = qkv.shape
     D = k_cache.shape[3]
@@ -93,13 +101,28 @@ def rope_cache_append_ref(qkv, k_cache, v_cache, block_table, seq_lens_before,
                 cos[None], sin[None])
             q_out[b, t] = qr[0, 0].to(q.dtype)
             blk = int(block_table[b, pos // bs])
-            k_cache[blk, pos % bs] = kr[0, 0].to(k_cache.dtype)
-            v_cache[blk, pos % bs] = v[b, t].to(v_cache.dtype)
+            if k_scale is not None:  # int8 cache: per-(token, head) absmax
+                kv32 = kr[0, 0].float()
+                vv32 = v[b, t].float()
+                ks = kv32.abs().amax(-1).clamp(min=1e-8) / 127.0
+                vs = vv32.abs().amax(-1).clamp(min=1e-8) / 127.0
+                k_cache[blk, pos % bs] = torch.round(
+                    kv32 / ks[:, None]).clamp(-127, 127).to(torch.int8)
+                v_cache[blk, pos % bs] = torch.round(
+                    vv32 / vs[:, None]).clamp(-127, 127).to(torch.int8)
+                k_scale[blk, pos % bs] = ks
+                v_scale[blk, pos % bs] = vs
+            else:
+                k_cache[blk, pos % bs] = kr[0, 0].to(k_cache.dtype)
+                v_cache[blk, pos % bs] = v[b, t].to(v_cache.dtype)
     return q_out
 
 
 class FusedMultiTransformer(nn.Module):
     """All decoder layers in one module; weights as per-layer lists."""
+
+    k_scales = None  # set by allocate_caches(cachekv_dtype="int8")
+    v_scales = None
 
     def __init__(self, config: FusedMultiTransformerConfig):
         super().__init__()
@@ -193,13 +216,30 @@ class FusedMultiTransformer(nn.Module):
         algo = self.quant_algo if self.quant_algo == "fp8" else "weight_only_int8"
         return weight_only_linear(x, q, sc, None, algo)
 
-    def allocate_caches(self, num_blocks: int, device):
+    def allocate_caches(self, num_blocks: int, device, cachekv_dtype: str = "bf16"):
+        """cachekv_dtype="int8" halves KV memory: caches store int8 with one
+        fp32 absmax scale per cached (token, kv-head) vector; the decode and
+        append kernels (de)quantize in-register (reference cachekv int8,
+        write_int8_cache_kv / append_attention_c8)."""
         c = self.config
         shape = (num_blocks, c.block_size, c.num_kv_heads, c.head_dim)
-        self.k_caches = [torch.zeros(shape, dtype=c.dtype, device=device)
-                         for _ in range(c.num_layers)]
-        self.v_caches = [torch.zeros(shape, dtype=c.dtype, device=device)
-                         for _ in range(c.num_layers)]
+        self.cachekv_dtype = cachekv_dtype
+        if cachekv_dtype == "int8":
+            self.k_caches = [torch.zeros(shape, dtype=torch.int8, device=device)
+                             for _ in range(c.num_layers)]
+            self.v_caches = [torch.zeros(shape, dtype=torch.int8, device=device)
+                             for _ in range(c.num_layers)]
+            sshape = (num_blocks, c.block_size, c.num_kv_heads)
+            self.k_scales = [torch.zeros(sshape, dtype=torch.float32, device=device)
+                             for _ in range(c.num_layers)]
+            self.v_scales = [torch.zeros(sshape, dtype=torch.float32, device=device)
+                             for _ in range(c.num_layers)]
+        else:
+            self.k_caches = [torch.zeros(shape, dtype=c.dtype, device=device)
+                             for _ in range(c.num_layers)]
+            self.v_caches = [torch.zeros(shape, dtype=c.dtype, device=device)
+                             for _ in range(c.num_layers)]
+            self.k_scales = self.v_scales = None
         return self.k_caches, self.v_caches
 
     # ------------------------------------------------------------------
@@ -217,20 +257,28 @@ class FusedMultiTransformer(nn.Module):
             return C.rope_cache_append(
                 qkv, self.k_caches[i], self.v_caches[i], block_table,
                 lens_before, self.rope_cos, self.rope_sin,
-                c.num_heads, c.num_kv_heads, token_counts)
+                c.num_heads, c.num_kv_heads, token_counts,
+                self.k_scales[i] if self.k_scales is not None else None,
+                self.v_scales[i] if self.v_scales is not None else None)
         return rope_cache_append_ref(
             qkv, self.k_caches[i], self.v_caches[i], block_table, lens_before,
-            self.rope_cos, self.rope_sin, c.num_heads, c.num_kv_heads, token_counts)
+            self.rope_cos, self.rope_sin, c.num_heads, c.num_kv_heads, token_counts,
+            self.k_scales[i] if self.k_scales is not None else None,
+            self.v_scales[i] if self.v_scales is not None else None)
 
     def _paged_attn(self, i, q, block_table, seq_lens):
         if q.is_cuda:
             from ..ops.functional import _load_extension
 
             C = _load_extension()
-            return C.paged_decode_attn(q, self.k_caches[i], self.v_caches[i],
-                                       block_table, seq_lens)
-        return paged_decode_attn_ref(q, self.k_caches[i], self.v_caches[i],
-                                     block_table, seq_lens)
+            return C.paged_decode_attn(
+                q, self.k_caches[i], self.v_caches[i], block_table, seq_lens,
+                self.k_scales[i] if self.k_scales is not None else None,
+                self.v_scales[i] if self.v_scales is not None else None)
+        return paged_decode_attn_ref(
+            q, self.k_caches[i], self.v_caches[i], block_table, seq_lens,
+            self.k_scales[i] if self.k_scales is not None else None,
+            self.v_scales[i] if self.v_scales is not None else None)
 
     # ------------------------------------------------------------------
     @torch.no_grad()

@@ -31,10 +31,10 @@ struct AdamWChunk {
     int is_bf16;
 };
 void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, float, float, hipStream_t);
-void launch_paged_decode_attn(const void*, const void*, const void*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, hipStream_t);
+void launch_paged_decode_attn(const void*, const void*, const void*, const float*, const float*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, hipStream_t);
 int paged_decode_nsplit(int, int);
 void launch_wint8_gemv(const void*, const void*, const float*, void*, int, int, int, hipStream_t);
-void launch_rope_cache_append(const void*, void*, void*, void*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, hipStream_t);
+void launch_rope_cache_append(const void*, void*, void*, void*, float*, float*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -309,12 +309,17 @@ std::vector<torch::Tensor> flashmask_attn_bwd(torch::Tensor dout, torch::Tensor 
 // paged-KV inference ops
 // ---------------------------------------------------------------------------
 torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor k_cache, torch::Tensor v_cache,
-                                torch::Tensor block_table, torch::Tensor seq_lens) {
+                                torch::Tensor block_table, torch::Tensor seq_lens,
+                                c10::optional<torch::Tensor> k_scale,
+                                c10::optional<torch::Tensor> v_scale) {
     CHECK_GPU(q); CHECK_CONTIG(q); CHECK_BF16(q);
     TORCH_CHECK(q.dim() == 3, "q must be [B, Hq, D] (single decode token)");
     int B = q.size(0), Hq = q.size(1), D = q.size(2);
     int block_size = k_cache.size(1), Hk = k_cache.size(2);
     int max_blocks = block_table.size(1);
+    const bool c8 = (k_cache.scalar_type() == torch::kChar);
+    TORCH_CHECK(!c8 || (k_scale.has_value() && v_scale.has_value()),
+                "int8 KV cache needs k_scale/v_scale");
     auto out = torch::empty_like(q);
     float scale = 1.0f / std::sqrt((float)D);
     int G = Hq / Hk;
@@ -327,6 +332,8 @@ torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor k_cache, torch::T
         pptr = partials.data_ptr<float>();
     }
     launch_paged_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                             c8 ? k_scale->data_ptr<float>() : nullptr,
+                             c8 ? v_scale->data_ptr<float>() : nullptr,
                              block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                              out.data_ptr(), pptr, nsplit, B, Hq, Hk, D, block_size,
                              max_blocks, scale, cur_stream());
@@ -337,7 +344,9 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
                                 torch::Tensor block_table, torch::Tensor seq_lens_before,
                                 torch::Tensor cos_t, torch::Tensor sin_t,
                                 int64_t Hq, int64_t Hk,
-                                c10::optional<torch::Tensor> token_counts) {
+                                c10::optional<torch::Tensor> token_counts,
+                                c10::optional<torch::Tensor> k_scale,
+                                c10::optional<torch::Tensor> v_scale) {
     CHECK_GPU(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
     TORCH_CHECK(qkv.dim() == 3, "qkv must be [B, T, (Hq+2Hk)*D]");
     int B = qkv.size(0), T = qkv.size(1);
@@ -351,8 +360,13 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     auto sf = sin_t.to(torch::kFloat32).contiguous();
     const int* tc = nullptr;
     if (token_counts.has_value()) tc = token_counts->data_ptr<int>();
+    const bool c8 = (k_cache.scalar_type() == torch::kChar);
+    TORCH_CHECK(!c8 || (k_scale.has_value() && v_scale.has_value()),
+                "int8 KV cache needs k_scale/v_scale");
     launch_rope_cache_append(qkv.data_ptr(), q_out.data_ptr(),
                              k_cache.data_ptr(), v_cache.data_ptr(),
+                             c8 ? k_scale->data_ptr<float>() : nullptr,
+                             c8 ? v_scale->data_ptr<float>() : nullptr,
                              block_table.data_ptr<int>(), seq_lens_before.data_ptr<int>(),
                              cf.data_ptr<float>(), sf.data_ptr<float>(),
                              B, T, (int)Hq, (int)Hk, D, block_size, max_blocks, tc,
@@ -390,7 +404,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flashmask_attn_fwd", &flashmask_attn_fwd);
     m.def("flashmask_attn_bwd", &flashmask_attn_bwd);
     m.def("fused_adamw", &fused_adamw);
-    m.def("paged_decode_attn", &paged_decode_attn);
-    m.def("rope_cache_append", &rope_cache_append);
+    m.def("paged_decode_attn", &paged_decode_attn,
+          py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+          py::arg("block_table"), py::arg("seq_lens"),
+          py::arg("k_scale") = c10::nullopt, py::arg("v_scale") = c10::nullopt);
+    m.def("rope_cache_append", &rope_cache_append,
+          py::arg("qkv"), py::arg("k_cache"), py::arg("v_cache"),
+          py::arg("block_table"), py::arg("seq_lens_before"),
+          py::arg("cos_t"), py::arg("sin_t"), py::arg("Hq"), py::arg("Hk"),
+          py::arg("token_counts") = c10::nullopt,
+          py::arg("k_scale") = c10::nullopt, py::arg("v_scale") = c10::nullopt);
     m.def("wint8_gemv", &wint8_gemv);
 }

@@ -23,17 +23,23 @@
 // With NSPLIT > 1 the kv range is chunked across blockIdx.z so small-batch
 // decode fills the chip; each split writes fp32 partials
 // partial[b, hk, split] = {m[G], l[G], acc[G][D]} merged by the kernel below.
-template <int D>
+template <int D, bool C8>
 __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
     const ushort_t* __restrict__ q,
-    const ushort_t* __restrict__ k_cache,  // [nblocks, bs, Hk, D]
-    const ushort_t* __restrict__ v_cache,
+    const void* __restrict__ k_cache,      // [nblocks, bs, Hk, D] bf16|int8
+    const void* __restrict__ v_cache,
+    const float* __restrict__ k_scale,     // [nblocks*bs*Hk] (C8 only)
+    const float* __restrict__ v_scale,
     const int* __restrict__ block_table,   // [B, max_blocks]
     const int* __restrict__ seq_lens,      // [B]
     ushort_t* __restrict__ out,
     float* __restrict__ partials,          // [B, Hk, NSPLIT, G*(2+D)] or null
     int B, int Hq, int Hk, int block_size, int max_blocks, float scale,
     int nsplit) {
+    const ushort_t* k16 = (const ushort_t*)k_cache;
+    const ushort_t* v16 = (const ushort_t*)v_cache;
+    const signed char* k8 = (const signed char*)k_cache;
+    const signed char* v8 = (const signed char*)v_cache;
     const int b = blockIdx.x;
     const int hk = blockIdx.y;
     const int split = blockIdx.z;
@@ -77,20 +83,28 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
 
     for (int j = j0; j < j1; j++) {
         const int blk = bt[j / block_size];
-        const long long base =
-            (((long long)blk * block_size + (j % block_size)) * Hk + hk) * D;
+        const long long tok = ((long long)blk * block_size + (j % block_size)) * Hk + hk;
+        const long long base = tok * D;
         // lane loads its EPL elements of K and V (coalesced short2/short4)
         float kv_k[EPL], kv_v[EPL];
-        if (EPL == 2) {
-            short2v k2 = *reinterpret_cast<const short2v*>(k_cache + base + lane * 2);
-            short2v v2 = *reinterpret_cast<const short2v*>(v_cache + base + lane * 2);
+        if (C8) {
+            const float ks = k_scale[tok];
+            const float vs = v_scale[tok];
+#pragma unroll
+            for (int e = 0; e < EPL; e++) {
+                kv_k[e] = (float)k8[base + lane * EPL + e] * ks;
+                kv_v[e] = (float)v8[base + lane * EPL + e] * vs;
+            }
+        } else if (EPL == 2) {
+            short2v k2 = *reinterpret_cast<const short2v*>(k16 + base + lane * 2);
+            short2v v2 = *reinterpret_cast<const short2v*>(v16 + base + lane * 2);
             kv_k[0] = bf16_to_f32((ushort_t)k2[0]); kv_k[1] = bf16_to_f32((ushort_t)k2[1]);
             kv_v[0] = bf16_to_f32((ushort_t)v2[0]); kv_v[1] = bf16_to_f32((ushort_t)v2[1]);
         } else {
 #pragma unroll
             for (int e = 0; e < EPL; e++) {
-                kv_k[e] = bf16_to_f32(k_cache[base + lane * EPL + e]);
-                kv_v[e] = bf16_to_f32(v_cache[base + lane * EPL + e]);
+                kv_k[e] = bf16_to_f32(k16[base + lane * EPL + e]);
+                kv_v[e] = bf16_to_f32(v16[base + lane * EPL + e]);
             }
         }
 #pragma unroll
@@ -204,11 +218,12 @@ __global__ void paged_decode_merge_kernel(
 // positions p = seq_lens_before[b] + t; writes roped q to q_out [B, T, Hq, D]
 // and roped k / raw v into the paged cache.
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool C8>
 __global__ void rope_cache_append_kernel(
     const ushort_t* __restrict__ qkv,
     ushort_t* __restrict__ q_out,
-    ushort_t* __restrict__ k_cache, ushort_t* __restrict__ v_cache,
+    void* __restrict__ k_cache, void* __restrict__ v_cache,
+    float* __restrict__ k_scale, float* __restrict__ v_scale,
     const int* __restrict__ block_table, const int* __restrict__ seq_lens_before,
     const float* __restrict__ cos_t, const float* __restrict__ sin_t,  // [max_pos, D]
     int B, int T, int Hq, int Hk, int block_size, int max_blocks,
@@ -255,24 +270,61 @@ __global__ void rope_cache_append_kernel(
         } else {
             int hk = head - Hq;
             int blk = block_table[(long long)b * max_blocks + pos / block_size];
-            ushort_t* dst = k_cache +
-                (((long long)blk * block_size + pos % block_size) * Hk + hk) * D;
+            const long long tok =
+                ((long long)blk * block_size + pos % block_size) * Hk + hk;
+            if (C8) {
+                float amax = 0.f;
 #pragma unroll
-            for (int e = 0; e < EPL; e++) {
-                int i = lane * EPL + e;
-                dst[i] = f32_to_bf16(o1[e]);
-                dst[i + half] = f32_to_bf16(o2[e]);
+                for (int e = 0; e < EPL; e++)
+                    amax = fmaxf(amax, fmaxf(fabsf(o1[e]), fabsf(o2[e])));
+                amax = wave_reduce_max(amax);
+                const float sc = fmaxf(amax, 1e-8f) / 127.f;
+                signed char* dst = (signed char*)k_cache + tok * D;
+#pragma unroll
+                for (int e = 0; e < EPL; e++) {
+                    int i = lane * EPL + e;
+                    dst[i] = (signed char)lrintf(fminf(fmaxf(o1[e] / sc, -127.f), 127.f));
+                    dst[i + half] = (signed char)lrintf(fminf(fmaxf(o2[e] / sc, -127.f), 127.f));
+                }
+                if (lane == 0) k_scale[tok] = sc;
+            } else {
+                ushort_t* dst = (ushort_t*)k_cache + tok * D;
+#pragma unroll
+                for (int e = 0; e < EPL; e++) {
+                    int i = lane * EPL + e;
+                    dst[i] = f32_to_bf16(o1[e]);
+                    dst[i + half] = f32_to_bf16(o2[e]);
+                }
             }
         }
     } else {
-        // V: raw copy into the cache
+        // V: copy (or quantize) into the cache
         int hv = head - Hq - Hk;
         int blk = block_table[(long long)b * max_blocks + pos / block_size];
-        ushort_t* dst = v_cache +
-            (((long long)blk * block_size + pos % block_size) * Hk + hv) * D;
+        const long long tok =
+            ((long long)blk * block_size + pos % block_size) * Hk + hv;
         constexpr int VE = D / 64;
+        if (C8) {
+            float vals[VE];
+            float amax = 0.f;
 #pragma unroll
-        for (int e = 0; e < VE; e++) dst[lane * VE + e] = src[lane * VE + e];
+            for (int e = 0; e < VE; e++) {
+                vals[e] = bf16_to_f32(src[lane * VE + e]);
+                amax = fmaxf(amax, fabsf(vals[e]));
+            }
+            amax = wave_reduce_max(amax);
+            const float sc = fmaxf(amax, 1e-8f) / 127.f;
+            signed char* dst = (signed char*)v_cache + tok * D;
+#pragma unroll
+            for (int e = 0; e < VE; e++)
+                dst[lane * VE + e] =
+                    (signed char)lrintf(fminf(fmaxf(vals[e] / sc, -127.f), 127.f));
+            if (lane == 0) v_scale[tok] = sc;
+        } else {
+            ushort_t* dst = (ushort_t*)v_cache + tok * D;
+#pragma unroll
+            for (int e = 0; e < VE; e++) dst[lane * VE + e] = src[lane * VE + e];
+        }
     }
 }
 
@@ -290,46 +342,50 @@ int paged_decode_nsplit(int B, int Hk) {
 }
 
 void launch_paged_decode_attn(const void* q, const void* k_cache, const void* v_cache,
+                              const float* k_scale, const float* v_scale,
                               const int* block_table, const int* seq_lens, void* out,
                               float* partials, int nsplit,
                               int B, int Hq, int Hk, int D, int block_size,
                               int max_blocks, float scale, hipStream_t stream) {
     dim3 grid(B, Hk, nsplit);
+    const bool c8 = (k_scale != nullptr);
+#define PA_LAUNCH(DD, CC)                                                          \
+    hipLaunchKernelGGL((paged_decode_attn_kernel<DD, CC>), grid, dim3(PA_BLOCK),   \
+                       0, stream, (const ushort_t*)q, k_cache, v_cache,            \
+                       k_scale, v_scale, block_table, seq_lens, (ushort_t*)out,    \
+                       partials, B, Hq, Hk, block_size, max_blocks, scale, nsplit)
     if (D == 128) {
-        hipLaunchKernelGGL(paged_decode_attn_kernel<128>, grid, dim3(PA_BLOCK), 0, stream,
-                           (const ushort_t*)q, (const ushort_t*)k_cache, (const ushort_t*)v_cache,
-                           block_table, seq_lens, (ushort_t*)out, partials,
-                           B, Hq, Hk, block_size, max_blocks, scale, nsplit);
+        if (c8) PA_LAUNCH(128, true); else PA_LAUNCH(128, false);
         if (nsplit > 1)
             hipLaunchKernelGGL(paged_decode_merge_kernel<128>, dim3(B, Hk, Hq / Hk), dim3(64),
                                0, stream, partials, (ushort_t*)out, seq_lens, B, Hq, Hk, nsplit);
     } else if (D == 64) {
-        hipLaunchKernelGGL(paged_decode_attn_kernel<64>, grid, dim3(PA_BLOCK), 0, stream,
-                           (const ushort_t*)q, (const ushort_t*)k_cache, (const ushort_t*)v_cache,
-                           block_table, seq_lens, (ushort_t*)out, partials,
-                           B, Hq, Hk, block_size, max_blocks, scale, nsplit);
+        if (c8) PA_LAUNCH(64, true); else PA_LAUNCH(64, false);
         if (nsplit > 1)
             hipLaunchKernelGGL(paged_decode_merge_kernel<64>, dim3(B, Hk, Hq / Hk), dim3(64),
                                0, stream, partials, (ushort_t*)out, seq_lens, B, Hq, Hk, nsplit);
     }
+#undef PA_LAUNCH
 }
 
 void launch_rope_cache_append(const void* qkv, void* q_out, void* k_cache, void* v_cache,
+                              float* k_scale, float* v_scale,
                               const int* block_table, const int* seq_lens_before,
                               const float* cos_t, const float* sin_t,
                               int B, int T, int Hq, int Hk, int D, int block_size,
                               int max_blocks, const int* token_counts, hipStream_t stream) {
     dim3 grid(B * T, Hq + 2 * Hk);
-    if (D == 128)
-        hipLaunchKernelGGL(rope_cache_append_kernel<128>, grid, dim3(64), 0, stream,
-                           (const ushort_t*)qkv, (ushort_t*)q_out,
-                           (ushort_t*)k_cache, (ushort_t*)v_cache,
-                           block_table, seq_lens_before, cos_t, sin_t,
-                           B, T, Hq, Hk, block_size, max_blocks, token_counts);
-    else if (D == 64)
-        hipLaunchKernelGGL(rope_cache_append_kernel<64>, grid, dim3(64), 0, stream,
-                           (const ushort_t*)qkv, (ushort_t*)q_out,
-                           (ushort_t*)k_cache, (ushort_t*)v_cache,
-                           block_table, seq_lens_before, cos_t, sin_t,
-                           B, T, Hq, Hk, block_size, max_blocks, token_counts);
+    const bool c8 = (k_scale != nullptr);
+#define RC_LAUNCH(DD, CC)                                                         \
+    hipLaunchKernelGGL((rope_cache_append_kernel<DD, CC>), grid, dim3(64), 0,     \
+                       stream, (const ushort_t*)qkv, (ushort_t*)q_out,            \
+                       k_cache, v_cache, k_scale, v_scale,                        \
+                       block_table, seq_lens_before, cos_t, sin_t,                \
+                       B, T, Hq, Hk, block_size, max_blocks, token_counts)
+    if (D == 128) {
+        if (c8) RC_LAUNCH(128, true); else RC_LAUNCH(128, false);
+    } else if (D == 64) {
+        if (c8) RC_LAUNCH(64, true); else RC_LAUNCH(64, false);
+    }
+#undef RC_LAUNCH
 }

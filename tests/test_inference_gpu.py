@@ -221,3 +221,56 @@ def test_graph_decode_runner_matches_eager(C):
         # free for the next batch size
         for s in slots:
             mgr.release(s)
+
+
+def test_int8_kv_cache_kernels(C):
+    """int8 cache kernels (quantized append + dequantizing decode) match the
+    CPU int8 reference and stay close to the bf16 kernels."""
+    import torch
+
+    from paddlenlp_amd.experimental.fused_transformer import (
+        paged_decode_attn_ref, rope_cache_append_ref)
+    from paddlenlp_amd.ops import reference
+
+    torch.manual_seed(0)
+    device = torch.device("cuda:0")
+    B, T, Hq, Hk, D = 2, 6, 4, 2, 128
+    nblocks, bs, max_blocks = 8, 8, 4
+    qkv = torch.randn(B, T, (Hq + 2 * Hk) * D, device=device,
+                      dtype=torch.bfloat16)
+    bt = torch.arange(B * max_blocks, device=device,
+                      dtype=torch.int32).reshape(B, max_blocks)
+    lens0 = torch.zeros(B, dtype=torch.int32, device=device)
+    cos, sin = reference.build_rope_cache(64, D, 10000.0)
+    cos = cos.to(device)
+    sin = sin.to(device)
+
+    kc = torch.zeros(nblocks, bs, Hk, D, dtype=torch.int8, device=device)
+    vc = torch.zeros_like(kc)
+    ks = torch.zeros(nblocks, bs, Hk, dtype=torch.float32, device=device)
+    vs = torch.zeros_like(ks)
+    q_out = C.rope_cache_append(qkv, kc, vc, bt, lens0, cos, sin, Hq, Hk,
+                                None, ks, vs)
+
+    kc_ref = torch.zeros_like(kc, device="cpu")
+    vc_ref = torch.zeros_like(vc, device="cpu")
+    ks_ref = torch.zeros_like(ks, device="cpu")
+    vs_ref = torch.zeros_like(vs, device="cpu")
+    q_ref = rope_cache_append_ref(
+        qkv.cpu(), kc_ref, vc_ref, bt.cpu(), lens0.cpu(), cos.cpu(), sin.cpu(),
+        Hq, Hk, None, ks_ref, vs_ref)
+    torch.testing.assert_close(q_out.cpu().float(), q_ref.float(),
+                               rtol=2e-2, atol=2e-2)
+    # quantized cache contents match (int8 rounding is deterministic)
+    assert (kc.cpu() - kc_ref).abs().max() <= 1
+    torch.testing.assert_close(ks.cpu(), ks_ref, rtol=1e-3, atol=1e-5)
+
+    # decode against the int8 cache
+    q = torch.randn(B, Hq, D, device=device, dtype=torch.bfloat16)
+    seq_lens = torch.full((B,), T, dtype=torch.int32, device=device)
+    out = C.paged_decode_attn(q, kc, vc, bt, seq_lens, ks, vs)
+    out_ref = paged_decode_attn_ref(q.cpu(), kc_ref, vc_ref, bt.cpu(),
+                                    seq_lens.cpu(), ks_ref, vs_ref)
+    rel = (out.cpu().float() - out_ref.float()).abs().max() \
+        / out_ref.float().abs().max()
+    assert rel < 0.03, rel
