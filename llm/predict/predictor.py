@@ -47,6 +47,7 @@ class PredictorArgument:
     benchmark: bool = field(default=False)
     quant_type: str = field(default="")  # "" | "fp8" | "weight_only_int8"
     use_hipgraph: bool = field(default=False)  # capture decode in a hipGraph
+    cachekv_int8: bool = field(default=False)  # int8 paged KV (2x capacity)
 
 
 class BasePredictor:
@@ -100,7 +101,9 @@ class BlockInferencePredictor(BasePredictor):
         num_blocks = max_blocks_per_seq * config.batch_size + 8
         self.manager = BlockManager(
             num_blocks, c.block_size, max_blocks_per_seq, config.batch_size)
-        engine.allocate_caches(num_blocks, self.device)
+        engine.allocate_caches(
+            num_blocks, self.device,
+            cachekv_dtype="int8" if config.cachekv_int8 else "bf16")
         self.engine = engine.to(self.device)
         if config.use_hipgraph and torch.cuda.is_available():
             from paddlenlp_amd.experimental.fused_transformer import GraphDecodeRunner

@@ -235,3 +235,32 @@ def test_int8_kv_cache_cpu():
     for a, b in zip(outs["bf16"], outs["int8"]):
         rel = (a - b).abs().max() / a.abs().max().clamp(min=1e-6)
         assert rel < 0.05, rel  # int8 KV keeps logits close
+
+
+def test_predictor_int8_cachekv():
+    import sys
+
+    sys.path.insert(0, "llm")
+    import importlib
+
+    import predict.predictor as predictor_mod
+
+    importlib.reload(predictor_mod)
+    import torch
+
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=128)
+    model = LlamaForCausalLM.from_config(cfg).eval()
+    tok = _make_tiny_tokenizer()
+    args = predictor_mod.PredictorArgument(
+        src_length=32, max_length=8, total_max_length=64, batch_size=2,
+        block_size=8, decode_strategy="greedy", dtype="float32",
+        cachekv_int8=True)
+    pred = predictor_mod.create_predictor(args, model=model, tokenizer=tok)
+    out = pred.predict(["hello world", "the cat"])
+    assert len(out) == 2 and all(isinstance(o, str) for o in out)
+    assert pred.engine.k_caches[0].dtype == torch.int8
