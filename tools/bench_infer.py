@@ -96,6 +96,7 @@ def main():
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--quant", default="")          # "" | fp8 | weight_only_int8
     p.add_argument("--graph", action="store_true")
+    p.add_argument("--cachekv", default="bf16", choices=["bf16", "int8"])
     args = p.parse_args()
 
     device = "cuda:0"
@@ -112,7 +113,7 @@ def main():
 
     max_b = max(int(b) for b in args.batches.split(","))
     blocks_per_seq = (args.prompt_len + args.steps + args.warmup) // 64 + 2
-    eng.allocate_caches(max_b * blocks_per_seq + 8, device)
+    eng.allocate_caches(max_b * blocks_per_seq + 8, device, cachekv_dtype=args.cachekv)
 
     for b in args.batches.split(","):
         B = int(b)
@@ -123,6 +124,7 @@ def main():
             "prompt_len": args.prompt_len, "value": round(tps, 1),
             "ms_per_step": round(ms, 2),
             "quant": args.quant or "bf16", "hipgraph": bool(args.graph),
+            "cachekv": args.cachekv,
         }))
 
 
