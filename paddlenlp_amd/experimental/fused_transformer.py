@@ -39,10 +39,12 @@ class FusedMultiTransformerConfig:
     block_size: int = 64
     max_seq_len: int = 8192
     dtype: torch.dtype = torch.bfloat16
+    # explicit head_dim for TP shards (local num_heads != hidden/head_dim)
+    head_dim_override: int = 0
 
     @property
     def head_dim(self):
-        return self.hidden_size // self.num_heads
+        return self.head_dim_override or self.hidden_size // self.num_heads
 
 
 # ---------------------------------------------------------------------------
@@ -123,6 +125,7 @@ class FusedMultiTransformer(nn.Module):
 
     k_scales = None  # set by allocate_caches(cachekv_dtype="int8")
     v_scales = None
+    tp_group = None  # set by from_llama(tp_degree > 1): row-parallel reduce
 
     def __init__(self, config: FusedMultiTransformerConfig):
         super().__init__()
@@ -152,40 +155,73 @@ class FusedMultiTransformer(nn.Module):
     # ------------------------------------------------------------------
     @classmethod
     @torch.no_grad()
-    def from_llama(cls, model, block_size: int = 64, max_seq_len: int = 8192):
-        """Import weights from a LlamaForCausalLM (fused qkv / gate_up)."""
+    def from_llama(cls, model, block_size: int = 64, max_seq_len: int = 8192,
+                   tp_degree: int = 1, tp_rank: int = 0, tp_group=None):
+        """Import weights from a LlamaForCausalLM (fused qkv / gate_up).
+
+        With tp_degree > 1 this builds the tensor-parallel SHARD for
+        tp_rank: qkv/gate_up are head/channel column-sharded, o/down are
+        row-sharded and their partial outputs all-reduce over tp_group
+        (Megatron inference split; embeddings/head replicated)."""
         mc = model.config
+        assert mc.num_attention_heads % tp_degree == 0
+        assert mc.num_key_value_heads % tp_degree == 0
+        assert mc.intermediate_size % tp_degree == 0
+        Hq, Hk = mc.num_attention_heads, mc.num_key_value_heads
+        I = mc.intermediate_size
+        hd = mc.hidden_size // Hq
         cfg = FusedMultiTransformerConfig(
-            hidden_size=mc.hidden_size, num_heads=mc.num_attention_heads,
-            num_kv_heads=mc.num_key_value_heads, intermediate_size=mc.intermediate_size,
+            hidden_size=mc.hidden_size, num_heads=Hq // tp_degree,
+            num_kv_heads=Hk // tp_degree, intermediate_size=I // tp_degree,
             num_layers=mc.num_hidden_layers, vocab_size=mc.vocab_size,
             rms_norm_eps=mc.rms_norm_eps, rope_theta=mc.rope_theta,
             block_size=block_size, max_seq_len=max_seq_len,
-            dtype=next(model.parameters()).dtype,
+            dtype=next(model.parameters()).dtype, head_dim_override=hd,
         )
         eng = cls(cfg)
+        eng.tp_group = tp_group if tp_degree > 1 else None
+        r = tp_rank
+
+        def qsl(w):   # rows of this rank's q heads
+            return w[r * (Hq // tp_degree) * hd:(r + 1) * (Hq // tp_degree) * hd]
+
+        def kvsl(w):  # rows of this rank's kv heads
+            return w[r * (Hk // tp_degree) * hd:(r + 1) * (Hk // tp_degree) * hd]
+
         base = model.llama
         for i, layer in enumerate(base.layers):
             eng.ln_scales[i].copy_(layer.input_layernorm.weight)
             a = layer.self_attn
             if mc.fuse_attention_qkv:
-                eng.qkv_weights[i].copy_(a.qkv_proj.weight)
+                w = a.qkv_proj.weight
+                qw, kw, vw = w[:Hq * hd], w[Hq * hd:(Hq + Hk) * hd], w[(Hq + Hk) * hd:]
             else:
-                eng.qkv_weights[i].copy_(torch.cat(
-                    [a.q_proj.weight, a.k_proj.weight, a.v_proj.weight], dim=0))
-            eng.out_proj_weights[i].copy_(a.o_proj.weight)
+                qw, kw, vw = a.q_proj.weight, a.k_proj.weight, a.v_proj.weight
+            eng.qkv_weights[i].copy_(torch.cat([qsl(qw), kvsl(kw), kvsl(vw)], dim=0))
+            eng.out_proj_weights[i].copy_(
+                a.o_proj.weight[:, r * (Hq // tp_degree) * hd:(r + 1) * (Hq // tp_degree) * hd])
             eng.ffn_ln_scales[i].copy_(layer.post_attention_layernorm.weight)
             m = layer.mlp
             if mc.fuse_attention_ffn:
-                eng.gate_up_weights[i].copy_(m.gate_up_fused_proj.weight)
+                w = m.gate_up_fused_proj.weight
+                gw, uw = w[:I], w[I:]
             else:
-                eng.gate_up_weights[i].copy_(torch.cat(
-                    [m.gate_proj.weight, m.up_proj.weight], dim=0))
-            eng.down_weights[i].copy_(m.down_proj.weight)
+                gw, uw = m.gate_proj.weight, m.up_proj.weight
+            Il = I // tp_degree
+            eng.gate_up_weights[i].copy_(torch.cat(
+                [gw[r * Il:(r + 1) * Il], uw[r * Il:(r + 1) * Il]], dim=0))
+            eng.down_weights[i].copy_(m.down_proj.weight[:, r * Il:(r + 1) * Il])
         eng.embed_tokens.copy_(base.embed_tokens.weight)
         eng.final_norm.copy_(base.norm.weight)
         eng.lm_head.copy_(model.lm_head.weight)
         return eng
+
+    def _tp_reduce(self, t):
+        if self.tp_group is not None:
+            import torch.distributed as dist
+
+            dist.all_reduce(t, group=self.tp_group)
+        return t
 
     @torch.no_grad()
     def quantize(self, algo: str = "fp8"):
@@ -296,11 +332,11 @@ class FusedMultiTransformer(nn.Module):
             qkv = self._mm(h, "qkv_weights", i)          # [B, 1, qkv_out]
             q = self._rope_append(i, qkv, block_table, seq_lens_before)
             attn = self._paged_attn(i, q[:, 0], block_table, seq_lens_after)
-            x = x + self._mm(attn.reshape(B, 1, -1), "out_proj_weights", i)
+            x = x + self._tp_reduce(self._mm(attn.reshape(B, 1, -1), "out_proj_weights", i))
             h = self._rms(x, self.ffn_ln_scales[i])
             gu = self._mm(h, "gate_up_weights", i)
             act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
-            x = x + self._mm(act, "down_weights", i)
+            x = x + self._tp_reduce(self._mm(act, "down_weights", i))
         x = self._rms(x, self.final_norm)
         logits = x[:, 0] @ self.lm_head.t()
         return logits.float()
@@ -332,11 +368,11 @@ class FusedMultiTransformer(nn.Module):
                 attn = ops.flash_attention(q, k_roped, v, causal=True)
             else:
                 attn = reference.flash_attention(q, k_roped, v, causal=True)
-            x = x + self._mm(attn.reshape(B, T, -1), "out_proj_weights", i)
+            x = x + self._tp_reduce(self._mm(attn.reshape(B, T, -1), "out_proj_weights", i))
             h = self._rms(x, self.ffn_ln_scales[i])
             gu = self._mm(h, "gate_up_weights", i)
             act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
-            x = x + self._mm(act, "down_weights", i)
+            x = x + self._tp_reduce(self._mm(act, "down_weights", i))
         x = self._rms(x, self.final_norm)
         idx = (prompt_lens.long() - 1).clamp(min=0)
         last = x[torch.arange(B, device=x.device), idx]

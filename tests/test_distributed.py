@@ -436,3 +436,57 @@ def test_zero2_overlap_comm_parity():
 
 def test_zero1_overlap_comm_parity():
     _run_workers(_w_zero1_overlap)
+
+
+def _w_tp_inference_engine(rank, world):
+    """TP-sharded fused inference engine == single-process engine logits."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.experimental import BlockManager, FusedMultiTransformer
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    topo = init_parallel_env(mp_degree=world, backend="gloo")
+    torch.manual_seed(11)
+    cfg = LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, dtype="float32",
+    )
+    model = LlamaForCausalLM.from_config(cfg).eval()
+    # identical weights on every rank
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+
+    def run(eng):
+        eng.allocate_caches(16, torch.device("cpu"))
+        mgr = BlockManager(16, 8, 8, 2)
+        g = torch.Generator().manual_seed(3)
+        ids = torch.randint(3, 128, (2, 10), generator=g)
+        lens = torch.tensor([10, 10], dtype=torch.int32)
+        slots = [mgr.allocate_slot(10) for _ in range(2)]
+        bt = torch.stack([mgr.block_table[s] for s in slots]).to(torch.int32)
+        logits = eng.prefill(ids, bt, lens)
+        tok = logits.argmax(-1, keepdim=True)
+        lens_before = torch.tensor([10, 10], dtype=torch.int32)
+        for s in slots:
+            assert mgr.extend(s, 1)
+        bt = torch.stack([mgr.block_table[s] for s in slots]).to(torch.int32)
+        step = eng.decode_step(tok, bt, lens_before)
+        return logits, step
+
+    tp_eng = FusedMultiTransformer.from_llama(
+        model, block_size=8, max_seq_len=64, tp_degree=world, tp_rank=rank,
+        tp_group=topo.model_parallel_group)
+    assert tp_eng.config.num_heads == 4 // world
+    tp_logits, tp_step = run(tp_eng)
+
+    full_eng = FusedMultiTransformer.from_llama(model, block_size=8, max_seq_len=64)
+    full_logits, full_step = run(full_eng)
+
+    assert torch.allclose(tp_logits, full_logits, atol=1e-4), \
+        (tp_logits - full_logits).abs().max()
+    assert torch.allclose(tp_step, full_step, atol=1e-4), \
+        (tp_step - full_step).abs().max()
+
+
+def test_tp_inference_engine_parity():
+    _run_workers(_w_tp_inference_engine)
