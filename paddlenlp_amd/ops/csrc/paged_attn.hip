@@ -81,11 +81,83 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
     const int j1 = min(c1, j0 + per_wave);
     const int* bt = block_table + (long long)b * max_blocks;
 
-    for (int j = j0; j < j1; j++) {
+    // The per-token online-softmax update is a serial dependency chain
+    // (wave_reduce -> exp -> acc scale).  Unrolling U tokens per iteration
+    // pipelines the K/V loads and the U wave-reductions (independent), and
+    // folds U softmax updates into ONE rescale (single max over the tile) —
+    // long-context decode is latency-bound on this chain, not on HBM bytes.
+    constexpr int U = 8;
+    int j = j0;
+    for (; j + U <= j1; j += U) {
+        float kk[U][EPL], vv[U][EPL];
+#pragma unroll
+        for (int u = 0; u < U; u++) {
+            const int jj = j + u;
+            const int blk = bt[jj / block_size];
+            const long long tok =
+                ((long long)blk * block_size + (jj % block_size)) * Hk + hk;
+            const long long base = tok * D;
+            if (C8) {
+                const float ks = k_scale[tok];
+                const float vs = v_scale[tok];
+#pragma unroll
+                for (int e = 0; e < EPL; e++) {
+                    kk[u][e] = (float)k8[base + lane * EPL + e] * ks;
+                    vv[u][e] = (float)v8[base + lane * EPL + e] * vs;
+                }
+            } else if (EPL == 2) {
+                short2v k2 = *reinterpret_cast<const short2v*>(k16 + base + lane * 2);
+                short2v v2 = *reinterpret_cast<const short2v*>(v16 + base + lane * 2);
+                kk[u][0] = bf16_to_f32((ushort_t)k2[0]);
+                kk[u][1] = bf16_to_f32((ushort_t)k2[1]);
+                vv[u][0] = bf16_to_f32((ushort_t)v2[0]);
+                vv[u][1] = bf16_to_f32((ushort_t)v2[1]);
+            } else {
+#pragma unroll
+                for (int e = 0; e < EPL; e++) {
+                    kk[u][e] = bf16_to_f32(k16[base + lane * EPL + e]);
+                    vv[u][e] = bf16_to_f32(v16[base + lane * EPL + e]);
+                }
+            }
+        }
+#pragma unroll
+        for (int g = 0; g < MAX_GQA; g++) {
+            if (g >= G) break;
+            float sc[U];
+#pragma unroll
+            for (int u = 0; u < U; u++) {
+                float s = 0.f;
+#pragma unroll
+                for (int e = 0; e < EPL; e++) s += qreg[g][e] * kk[u][e];
+                sc[u] = s;
+            }
+            // U independent reductions pipeline their shuffle latency
+#pragma unroll
+            for (int u = 0; u < U; u++) sc[u] = wave_reduce_sum(sc[u]);
+            float tile_max = sc[0];
+#pragma unroll
+            for (int u = 1; u < U; u++) tile_max = fmaxf(tile_max, sc[u]);
+            const float m_new = fmaxf(m[g], tile_max);
+            const float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
+            float p[U];
+            float psum = 0.f;
+#pragma unroll
+            for (int u = 0; u < U; u++) { p[u] = __expf(sc[u] - m_new); psum += p[u]; }
+            l[g] = l[g] * alpha + psum;
+#pragma unroll
+            for (int e = 0; e < EPL; e++) {
+                float a = acc[g][e] * alpha;
+#pragma unroll
+                for (int u = 0; u < U; u++) a += p[u] * vv[u][e];
+                acc[g][e] = a;
+            }
+            m[g] = m_new;
+        }
+    }
+    for (; j < j1; j++) {
         const int blk = bt[j / block_size];
         const long long tok = ((long long)blk * block_size + (j % block_size)) * Hk + hk;
         const long long base = tok * D;
-        // lane loads its EPL elements of K and V (coalesced short2/short4)
         float kv_k[EPL], kv_v[EPL];
         if (C8) {
             const float ks = k_scale[tok];
@@ -95,11 +167,6 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
                 kv_k[e] = (float)k8[base + lane * EPL + e] * ks;
                 kv_v[e] = (float)v8[base + lane * EPL + e] * vs;
             }
-        } else if (EPL == 2) {
-            short2v k2 = *reinterpret_cast<const short2v*>(k16 + base + lane * 2);
-            short2v v2 = *reinterpret_cast<const short2v*>(v16 + base + lane * 2);
-            kv_k[0] = bf16_to_f32((ushort_t)k2[0]); kv_k[1] = bf16_to_f32((ushort_t)k2[1]);
-            kv_v[0] = bf16_to_f32((ushort_t)v2[0]); kv_v[1] = bf16_to_f32((ushort_t)v2[1]);
         } else {
 #pragma unroll
             for (int e = 0; e < EPL; e++) {
@@ -113,7 +180,7 @@ __global__ __launch_bounds__(PA_BLOCK) void paged_decode_attn_kernel(
             float s = 0.f;
 #pragma unroll
             for (int e = 0; e < EPL; e++) s += qreg[g][e] * kv_k[e];
-            s = wave_reduce_sum(s);  // full dot product, replicated in wave
+            s = wave_reduce_sum(s);
             float m_new = fmaxf(m[g], s);
             float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
             float p = __expf(s - m_new);
