@@ -22,10 +22,12 @@ def _build_app():
     @app.command()
     def search(query: str = typer.Argument("", help="filter model types")):
         """List registered model families."""
-        for mt, (mod, cfg, lm, base) in sorted(MODEL_REGISTRY.items()):
+        for mt, entry in sorted(MODEL_REGISTRY.items()):
             if query and query not in mt:
                 continue
-            typer.echo(f"{mt:<12} {lm} ({mod})")
+            main_cls = entry.get("causal_lm") or entry.get("seq2seq_lm") \
+                or entry.get("base")
+            typer.echo(f"{mt:<14} {main_cls} ({entry['module']})")
 
     @app.command()
     def server(model: str = typer.Option(..., help="local model dir"),

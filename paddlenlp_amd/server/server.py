@@ -25,12 +25,17 @@ class SimpleServer:
 
     def register(self, path: str, handler: Callable, methods=("POST",)):
         """Register a prediction handler: handler(dict) -> dict."""
-        from fastapi import Request
+        # NOTE: `from __future__ import annotations` turns the Request
+        # annotation into a string FastAPI resolves against module globals —
+        # a request-object endpoint signature would silently degrade to a
+        # query param.  Read the body via the raw starlette scope instead.
+        from starlette.requests import Request as _Req
 
-        async def endpoint(request: Request):
+        async def endpoint(request):
             data = await request.json()
             return handler(data)
 
+        endpoint.__annotations__ = {"request": _Req}
         self.app.add_api_route(path, endpoint, methods=list(methods))
 
     def register_taskflow(self, path: str, taskflow):
@@ -54,3 +59,21 @@ class SimpleServer:
 
         logger.info(f"SimpleServer listening on {host}:{port}")
         uvicorn.run(self.app, host=host, port=port, workers=workers)
+
+    def register_openai_compat(self, predictor, model_name: str = "default"):
+        """Minimal OpenAI-style /v1/completions endpoint over a predictor."""
+
+        def handler(data):
+            prompt = data.get("prompt", "")
+            prompts = [prompt] if isinstance(prompt, str) else list(prompt)
+            outs = predictor.predict(prompts)
+            return {
+                "object": "text_completion",
+                "model": data.get("model", model_name),
+                "choices": [
+                    {"index": i, "text": o, "finish_reason": "stop"}
+                    for i, o in enumerate(outs)
+                ],
+            }
+
+        self.register("/v1/completions", handler)
