@@ -7,3 +7,11 @@ from .logits_process import (  # noqa: F401
     TopPLogitsWarper,
 )
 from .utils import GenerationMixin  # noqa: F401
+from .stopping_criteria import (  # noqa: F401
+    MaxLengthCriteria,
+    MaxNewTokensCriteria,
+    MaxTimeCriteria,
+    StoppingCriteria,
+    StoppingCriteriaList,
+    StopStringsCriteria,
+)

@@ -74,7 +74,8 @@ class GenerationMixin:
             return self.sample(input_ids, gen)
         return self.greedy_search(input_ids, gen)
 
-    def _decode_loop(self, input_ids, gen: GenerationConfig, select_fn):
+    def _decode_loop(self, input_ids, gen: GenerationConfig, select_fn,
+                     stopping_criteria=None):
         B = input_ids.shape[0]
         device = input_ids.device
         eos_ids = gen.eos_ids()
@@ -104,11 +105,15 @@ class GenerationMixin:
                     unfinished = unfinished & (token != eos)
                 if not unfinished.any():
                     break
+            if stopping_criteria is not None and stopping_criteria(all_ids, None):
+                break
         gen_ids = torch.stack(new_tokens, dim=1) if new_tokens else input_ids.new_zeros(B, 0)
         return gen_ids, None
 
-    def greedy_search(self, input_ids, gen: GenerationConfig):
-        return self._decode_loop(input_ids, gen, lambda ids, lg: lg.argmax(-1))
+    def greedy_search(self, input_ids, gen: GenerationConfig,
+                      stopping_criteria=None):
+        return self._decode_loop(input_ids, gen, lambda ids, lg: lg.argmax(-1),
+                                 stopping_criteria)
 
     def sample(self, input_ids, gen: GenerationConfig):
         warpers = self._get_logits_warpers(gen)

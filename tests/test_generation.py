@@ -173,3 +173,28 @@ def test_speculative_sampling_runs(model):
     out, stats = speculative_generate(model, model, ids, gen, gamma=2)
     assert out.shape[1] <= 8
     assert 0.0 <= stats["acceptance_rate"] <= 1.0
+
+
+def test_stopping_criteria(model):
+    import time
+
+    from paddlenlp_amd.generation import (
+        MaxLengthCriteria,
+        MaxNewTokensCriteria,
+        MaxTimeCriteria,
+        StoppingCriteriaList,
+    )
+
+    ids = torch.randint(3, 128, (1, 6))
+    crit = StoppingCriteriaList([MaxLengthCriteria(10)])
+    gen = GenerationConfig(max_new_tokens=20, do_sample=False,
+                           eos_token_id=None, pad_token_id=0)
+    out, _ = model.greedy_search(ids, gen, stopping_criteria=crit)
+    assert out.shape[1] == 4  # stopped at total length 10
+
+    assert MaxNewTokensCriteria(6, 3)(torch.zeros(1, 9), None)
+    assert not MaxNewTokensCriteria(6, 3)(torch.zeros(1, 8), None)
+    t = MaxTimeCriteria(max_time=1000.0, initial_timestamp=time.time())
+    assert not t(torch.zeros(1, 1), None)
+    t2 = MaxTimeCriteria(max_time=0.0, initial_timestamp=time.time() - 1)
+    assert t2(torch.zeros(1, 1), None)
