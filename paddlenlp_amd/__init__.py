@@ -6,14 +6,23 @@ PyTorch-ROCm with hand-written CDNA4 (gfx950) HIP kernels for the hot
 path and RCCL collectives over xGMI for 4D parallelism.
 
 Public surface mirrors the reference layer map:
-  - paddlenlp_amd.transformers: model zoo + Auto* registries + PretrainedModel
-  - paddlenlp_amd.trainer:      Trainer / TrainingArguments / callbacks
-  - paddlenlp_amd.data:         collators, Megatron-style causal datasets
-  - paddlenlp_amd.parallel:     topology + DP/ZeRO/TP/SP/PP/CP engines (RCCL)
-  - paddlenlp_amd.ops:          CDNA4 HIP fused ops (flash attention, RMSNorm,
-                                RoPE, SwiGLU, fused AdamW, fused cross-entropy)
-  - paddlenlp_amd.generation:   decoding loops (greedy/sample/beam)
-  - paddlenlp_amd.peft:         LoRA / prefix tuning
+  - paddlenlp_amd.transformers: model zoo (21 families) + Auto* registries
+  - paddlenlp_amd.trainer:      Trainer / TrainingArguments / callbacks /
+                                unified checkpoint / compression
+  - paddlenlp_amd.data:         collators, Megatron-style causal datasets,
+                                Stack/Pad/Tuple/Dict batchify, Vocab
+  - paddlenlp_amd.parallel:     topology + DP/ZeRO-1/2/3/TP/SP/PP(+VPP)/
+                                SEP/CP(zigzag)/EP engines (RCCL over xGMI)
+  - paddlenlp_amd.ops:          CDNA4 HIP fused ops (flash attention +
+                                FlashMask, RMSNorm, RoPE, SwiGLU, fused
+                                AdamW, fused CE, paged decode, int8 KV)
+  - paddlenlp_amd.generation:   greedy/sample/beam/group-beam, speculative
+                                decoding, stopping criteria
+  - paddlenlp_amd.peft:         LoRA / prefix / VeRA / QLoRA
+  - paddlenlp_amd.experimental: fused inference engine (paged KV, TP,
+                                hipGraph), AutoNLP
+  - paddlenlp_amd.quantization / taskflow / prompt / seq2vec / embeddings /
+    dataaug / metrics / losses / server / cli
 """
 
 __version__ = "0.1.0"
