@@ -79,3 +79,9 @@ from .chatglm_v2 import (  # noqa: F401
 )
 from .mamba import MambaConfig, MambaForCausalLM, MambaModel  # noqa: F401
 from .gptj import GPTJConfig, GPTJForCausalLM, GPTJModel  # noqa: F401
+from .albert import (  # noqa: F401
+    AlbertConfig,
+    AlbertForMaskedLM,
+    AlbertForSequenceClassification,
+    AlbertModel,
+)

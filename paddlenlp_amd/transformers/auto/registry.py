@@ -58,6 +58,9 @@ MODEL_REGISTRY = {
                 "token_classification": "RobertaForTokenClassification",
                 "question_answering": "RobertaForQuestionAnswering",
                 "masked_lm": "RobertaForMaskedLM"},
+    "albert": {"module": "albert", "config": "AlbertConfig", "base": "AlbertModel",
+               "sequence_classification": "AlbertForSequenceClassification",
+               "masked_lm": "AlbertForMaskedLM"},
     "electra": {"module": "electra", "config": "ElectraConfig", "base": "ElectraModel",
                 "sequence_classification": "ElectraForSequenceClassification",
                 "token_classification": "ElectraForTokenClassification"},

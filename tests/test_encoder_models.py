@@ -255,3 +255,38 @@ def test_hf_bert_name_conversion():
     ids = torch.randint(0, 120, (2, 10))
     with torch.no_grad():
         torch.testing.assert_close(m(ids), m2(ids))
+
+
+# -------------------------------------------------------------------- albert
+def test_albert_parameter_sharing():
+    from paddlenlp_amd.transformers import AlbertConfig, AlbertForMaskedLM, AlbertModel
+
+    cfg = AlbertConfig(vocab_size=120, embedding_size=16, hidden_size=32,
+                       num_hidden_layers=4, num_hidden_groups=1,
+                       num_attention_heads=4, intermediate_size=64,
+                       max_position_embeddings=64)
+    m = AlbertModel(cfg)
+    # 4 layer applications share ONE parameter group
+    assert len(m.groups) == 1
+    # factorized embeddings: table at embedding_size, projected to hidden
+    assert m.embeddings.word_embeddings.weight.shape == (120, 16)
+    assert m.embedding_hidden_mapping.out_features == 32
+    ids = torch.randint(0, 120, (2, 10))
+    seq, pooled = m(ids)
+    assert seq.shape == (2, 10, 32) and pooled.shape == (2, 32)
+
+    # param count stays flat as layers grow (the ALBERT property)
+    cfg8 = AlbertConfig(vocab_size=120, embedding_size=16, hidden_size=32,
+                        num_hidden_layers=8, num_hidden_groups=1,
+                        num_attention_heads=4, intermediate_size=64,
+                        max_position_embeddings=64)
+    n4 = sum(p.numel() for p in AlbertModel(cfg).parameters())
+    n8 = sum(p.numel() for p in AlbertModel(cfg8).parameters())
+    assert n4 == n8
+
+    mlm = AlbertForMaskedLM(cfg)
+    labels = ids.clone()
+    labels[:, :5] = -100
+    loss, logits = mlm(ids, labels=labels)
+    loss.backward()
+    assert logits.shape == (2, 10, 120)
