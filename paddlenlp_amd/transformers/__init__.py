@@ -85,3 +85,9 @@ from .albert import (  # noqa: F401
     AlbertForSequenceClassification,
     AlbertModel,
 )
+from .clip import (  # noqa: F401
+    CLIPConfig,
+    CLIPModel,
+    CLIPTextModel,
+    CLIPVisionModel,
+)

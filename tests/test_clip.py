@@ -1,0 +1,62 @@
+"""CLIP dual-tower: patch embedding, EOS pooling, contrastive logits.
+
+Reference behavior: paddlenlp/transformers/clip/modeling.py.
+"""
+import torch
+
+from paddlenlp_amd.transformers import CLIPConfig, CLIPModel
+
+torch.manual_seed(0)
+
+
+def tiny_clip():
+    return CLIPConfig(
+        text_config=dict(vocab_size=100, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64,
+                         max_position_embeddings=16, eos_token_id=99),
+        vision_config=dict(hidden_size=32, num_hidden_layers=2,
+                           num_attention_heads=4, intermediate_size=64,
+                           image_size=32, patch_size=8),
+        projection_dim=24)
+
+
+def test_clip_towers_and_logits():
+    m = CLIPModel(tiny_clip()).eval()
+    ids = torch.randint(0, 98, (3, 10))
+    ids[:, -1] = 99  # eos
+    pix = torch.randn(3, 3, 32, 32)
+
+    tf = m.get_text_features(ids)
+    vf = m.get_image_features(pix)
+    assert tf.shape == (3, 24) and vf.shape == (3, 24)
+
+    li, lt = m(ids, pix)
+    assert li.shape == (3, 3)
+    torch.testing.assert_close(li, lt.t())
+
+    loss, _, _ = m(ids, pix, return_loss=True)
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert m.logit_scale.grad is not None
+
+
+def test_clip_eos_pooling_position():
+    m = CLIPModel(tiny_clip()).eval()
+    ids = torch.randint(0, 98, (1, 10))
+    ids[0, 4] = 99  # eos mid-sequence
+    with torch.no_grad():
+        x, pooled = m.text_model(ids)
+    torch.testing.assert_close(pooled[0], x[0, 4])
+    # causal text tower: tokens after EOS cannot change the pooled feature
+    ids2 = ids.clone()
+    ids2[0, 7] = (ids2[0, 7] + 1) % 98
+    with torch.no_grad():
+        _, pooled2 = m.text_model(ids2)
+    torch.testing.assert_close(pooled, pooled2)
+
+
+def test_clip_vision_patches():
+    m = CLIPModel(tiny_clip())
+    x, pooled = m.vision_model(torch.randn(2, 3, 32, 32))
+    assert x.shape == (2, 1 + 16, 32)  # class token + (32/8)^2 patches
+    assert pooled.shape == (2, 32)
