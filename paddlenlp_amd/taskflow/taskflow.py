@@ -57,6 +57,7 @@ GENERATION_TASKS = set(_GenerationTask.TEMPLATES)
 # encoder-backed pipelines (taskflow/tasks.py): task name -> class
 from .tasks import (  # noqa: E402
     DependencyParsingTask,
+    DocumentIntelligenceTask,
     FeatureExtractionTask,
     FillMaskTask,
     InformationExtractionTask,
@@ -82,12 +83,13 @@ ENCODER_TASKS = {
     "text_correction": TextCorrectionTask,
     "word_segmentation": WordSegmentationTask,
     "dependency_parsing": DependencyParsingTask,
+    "document_intelligence": DocumentIntelligenceTask,
 }
 
 # API-surface parity with the reference registry (taskflow/taskflow.py:48);
 # remaining pipelines need task models that don't exist offline.
 PENDING_TASKS = {
-    "document_intelligence", "knowledge_mining",
+    "knowledge_mining",  # needs the wordtag TermTree KB (not available offline)
 }
 
 TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)

@@ -394,3 +394,56 @@ class DependencyParsingTask:
                 "deprel": [self.rel_labels[int(r)] for r in rels[0]],
             })
         return results[0] if single else results
+
+
+class DocumentIntelligenceTask:
+    """ERNIE-Layout extractive doc QA (reference taskflow
+    document_intelligence.py): input is pre-OCR'd {"doc_tokens", "doc_boxes",
+    "prompt"} (no OCR engine offline); answer = argmax start/end span over
+    the document segment."""
+
+    def __init__(self, model_path: str, **kwargs):
+        from ..transformers import ErnieLayoutForQuestionAnswering
+
+        self.tokenizer = AutoTokenizer.from_pretrained(model_path)
+        self.model = ErnieLayoutForQuestionAnswering.from_pretrained(model_path)
+        if torch.cuda.is_available():
+            self.model = self.model.to("cuda:0")
+        self.model.eval()
+        self.device = next(self.model.parameters()).device
+
+    @torch.no_grad()
+    def __call__(self, inputs):
+        single = isinstance(inputs, dict)
+        batch = [inputs] if single else list(inputs)
+        tok = self.tokenizer._tokenizer
+        results = []
+        for ex in batch:
+            prompt_ids = tok.encode(ex["prompt"]).ids
+            doc_tokens = list(ex["doc_tokens"])
+            doc_boxes = list(ex["doc_boxes"])
+            doc_ids = []
+            tok_spans = []  # subtoken -> doc token index
+            for ti, word in enumerate(doc_tokens):
+                sub = tok.encode(word).ids
+                doc_ids.extend(sub)
+                tok_spans.extend([ti] * len(sub))
+            ids = torch.tensor([prompt_ids + doc_ids], device=self.device)
+            zero_box = [0, 0, 0, 0]
+            boxes = [zero_box] * len(prompt_ids) + \
+                [list(doc_boxes[t]) for t in tok_spans]
+            bbox = torch.tensor([boxes], device=self.device)
+            start_logits, end_logits = self.model(ids, bbox)
+            off = len(prompt_ids)
+            s = int(start_logits[0, off:].argmax())
+            e_cands = end_logits[0, off + s:]
+            e = s + int(e_cands.argmax())
+            ans_tokens = sorted(set(tok_spans[s:e + 1]))
+            results.append({
+                "prompt": ex["prompt"],
+                "result": [{
+                    "value": " ".join(doc_tokens[t] for t in ans_tokens),
+                    "start": s, "end": e,
+                }],
+            })
+        return results[0] if single else results

@@ -91,3 +91,9 @@ from .clip import (  # noqa: F401
     CLIPTextModel,
     CLIPVisionModel,
 )
+from .ernie_layout import (  # noqa: F401
+    ErnieLayoutConfig,
+    ErnieLayoutForQuestionAnswering,
+    ErnieLayoutForTokenClassification,
+    ErnieLayoutModel,
+)

@@ -124,7 +124,7 @@ def test_unknown_and_pending_tasks():
     with pytest.raises(ValueError):
         Taskflow("bogus_task")
     with pytest.raises(NotImplementedError):
-        Taskflow("document_intelligence")
+        Taskflow("knowledge_mining")
 
 
 def test_zero_shot_text_classification(tmp_path):
@@ -223,3 +223,32 @@ def test_dependency_parsing(tmp_path):
     assert set(out) == {"word", "head", "deprel"}
     assert len(out["word"]) == len(out["head"]) == len(out["deprel"]) == 3
     assert all(0 <= h < 3 for h in out["head"])
+
+
+def test_document_intelligence(tmp_path):
+    from paddlenlp_amd.transformers import (
+        ErnieLayoutConfig,
+        ErnieLayoutForQuestionAnswering,
+    )
+
+    torch.manual_seed(0)
+    cfg = ErnieLayoutConfig(vocab_size=len(VOCAB), hidden_size=32,
+                            num_hidden_layers=2, num_attention_heads=4,
+                            intermediate_size=64, max_position_embeddings=64,
+                            max_2d_position_embeddings=100,
+                            hidden_dropout_prob=0.0,
+                            attention_probs_dropout_prob=0.0)
+    m = ErnieLayoutForQuestionAnswering(cfg)
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("document_intelligence", model=str(tmp_path))
+    out = flow({
+        "prompt": "the cat",
+        "doc_tokens": ["alice", "visited", "paris"],
+        "doc_boxes": [[0, 0, 10, 10], [12, 0, 30, 10], [32, 0, 50, 10]],
+    })
+    assert out["prompt"] == "the cat"
+    ans = out["result"][0]
+    assert ans["value"] and ans["start"] <= ans["end"]
+    assert all(w in ("alice", "visited", "paris") for w in ans["value"].split())
