@@ -61,6 +61,7 @@ from .tasks import (  # noqa: E402
     FeatureExtractionTask,
     FillMaskTask,
     InformationExtractionTask,
+    KnowledgeMiningTask,
     TextClassificationTask,
     TextCorrectionTask,
     TextSimilarityTask,
@@ -84,13 +85,12 @@ ENCODER_TASKS = {
     "word_segmentation": WordSegmentationTask,
     "dependency_parsing": DependencyParsingTask,
     "document_intelligence": DocumentIntelligenceTask,
+    "knowledge_mining": KnowledgeMiningTask,
 }
 
 # API-surface parity with the reference registry (taskflow/taskflow.py:48);
 # remaining pipelines need task models that don't exist offline.
-PENDING_TASKS = {
-    "knowledge_mining",  # needs the wordtag TermTree KB (not available offline)
-}
+PENDING_TASKS = set()  # every registered pipeline is live
 
 TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)
 

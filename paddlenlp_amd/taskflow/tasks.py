@@ -447,3 +447,34 @@ class DocumentIntelligenceTask:
                 }],
             })
         return results[0] if single else results
+
+
+class KnowledgeMiningTask(TokenClassificationTask):
+    """WordTag-style knowledge mining (reference taskflow
+    knowledge_mining.py): token-classification tags per word plus optional
+    term linking against a USER-SUPPLIED term dictionary
+    ({surface_form: termid}) — the reference's TermTree KB is a downloadable
+    artifact that does not exist offline, so linking is bring-your-own."""
+
+    def __init__(self, model_path: str, label_map=None, term_dict=None,
+                 **kwargs):
+        super().__init__(model_path, label_map=label_map)
+        self.term_dict = term_dict or {}
+
+    @torch.no_grad()
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        results = []
+        for text in texts:
+            base = super().__call__(text)
+            items = []
+            for ent in base["entities"]:
+                item = {"item": ent["text"], "offset": ent["start"],
+                        "wordtag_label": ent["entity"]}
+                termid = self.term_dict.get(ent["text"])
+                if termid is not None:
+                    item["termid"] = termid
+                items.append(item)
+            results.append({"text": text, "items": items})
+        return results[0] if single else results

@@ -123,8 +123,6 @@ def test_information_extraction_task(tmp_path):
 def test_unknown_and_pending_tasks():
     with pytest.raises(ValueError):
         Taskflow("bogus_task")
-    with pytest.raises(NotImplementedError):
-        Taskflow("knowledge_mining")
 
 
 def test_zero_shot_text_classification(tmp_path):
@@ -252,3 +250,23 @@ def test_document_intelligence(tmp_path):
     ans = out["result"][0]
     assert ans["value"] and ans["start"] <= ans["end"]
     assert all(w in ("alice", "visited", "paris") for w in ans["value"].split())
+
+
+def test_knowledge_mining(tmp_path):
+    torch.manual_seed(0)
+    m = BertForTokenClassification(_tiny_cfg(num_labels=3))
+    m.config.id2label = {0: "O", 1: "B-人物类_实体", 2: "I-人物类_实体"}
+    with torch.no_grad():
+        m.classifier.weight.zero_()
+        m.classifier.bias.copy_(torch.tensor([0.0, 1.0, -1.0]))
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+
+    flow = Taskflow("knowledge_mining", model=str(tmp_path),
+                    term_dict={"alice": "person_alice_001"})
+    out = flow("alice visited paris")
+    assert out["text"] == "alice visited paris"
+    items = {i["item"]: i for i in out["items"]}
+    assert items["alice"]["termid"] == "person_alice_001"
+    assert "termid" not in items["paris"]
+    assert all(i["wordtag_label"] for i in out["items"])
