@@ -66,11 +66,20 @@ class GenerationMixin:
         if gen.max_length:
             gen.max_new_tokens = max(1, gen.max_length - input_ids.shape[1])
 
+        if gen.num_return_sequences > 1 and gen.num_beams == 1 \
+                and not gen.do_sample:
+            raise ValueError(
+                "num_return_sequences > 1 needs do_sample or beam search")
         if gen.num_beams > 1 and gen.num_beam_groups > 1:
             return self.group_beam_search(input_ids, gen)
         if gen.num_beams > 1:
             return self.beam_search(input_ids, gen)
         if gen.do_sample:
+            if gen.num_return_sequences > 1:
+                # reference semantics: expand the batch n times, sample
+                # independently, return [B*n, L]
+                input_ids = input_ids.repeat_interleave(
+                    gen.num_return_sequences, dim=0)
             return self.sample(input_ids, gen)
         return self.greedy_search(input_ids, gen)
 
@@ -185,21 +194,26 @@ class GenerationMixin:
             if all(len(f) >= K for f in finished):
                 break
 
+        n_ret = max(1, min(gen.num_return_sequences, K))
         results = []
         for b in range(B):
-            cands = finished[b]
-            if not cands:
-                # no EOS reached: use live beams
+            cands = sorted(finished[b], key=lambda x: -x[0])
+            if len(cands) < n_ret:
+                # top up with live beams (finished candidates stay preferred)
+                live = []
                 for j in range(K):
-                    lp = (ids.shape[1] - prompt_len) ** gen.length_penalty
-                    cands.append((beam_scores[b * K + j].item() / lp, ids[b * K + j]))
-            cands.sort(key=lambda x: -x[0])
-            results.append(cands[0][1][prompt_len:])
+                    lp = max(1, ids.shape[1] - prompt_len) ** gen.length_penalty
+                    live.append((beam_scores[b * K + j].item() / lp,
+                                 ids[b * K + j]))
+                cands.extend(sorted(live, key=lambda x: -x[0]))
+            for r in range(n_ret):
+                results.append(cands[r][1][prompt_len:])
         maxlen = max(r.shape[0] for r in results)
         pad_id = gen.pad_token_id or (eos_ids[0] if eos_ids else 0)
-        out = torch.full((B, maxlen), pad_id, dtype=torch.long, device=device)
-        for b, r in enumerate(results):
-            out[b, : r.shape[0]] = r
+        out = torch.full((B * n_ret, maxlen), pad_id, dtype=torch.long,
+                         device=device)
+        for i, r in enumerate(results):
+            out[i, : r.shape[0]] = r
         return out, None
 
 

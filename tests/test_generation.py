@@ -198,3 +198,35 @@ def test_stopping_criteria(model):
     assert not t(torch.zeros(1, 1), None)
     t2 = MaxTimeCriteria(max_time=0.0, initial_timestamp=time.time() - 1)
     assert t2(torch.zeros(1, 1), None)
+
+
+def test_num_return_sequences(model):
+    ids = torch.randint(3, 128, (2, 5))
+    # beam: top-2 sequences per batch item, flattened [B*n, L]
+    gen = GenerationConfig(max_new_tokens=5, num_beams=4,
+                           num_return_sequences=2, eos_token_id=2,
+                           pad_token_id=0)
+    out, _ = model.generate(ids, gen)
+    assert out.shape[0] == 4
+    # the first sequence per item is the best (same as n=1 beam search)
+    gen1 = GenerationConfig(max_new_tokens=5, num_beams=4, eos_token_id=2,
+                            pad_token_id=0)
+    best, _ = model.generate(ids, gen1)
+    L = min(best.shape[1], out.shape[1])
+    assert torch.equal(out[0, :L], best[0, :L])
+    assert torch.equal(out[2, :L], best[1, :L])
+
+    # sampling: independent expansions
+    torch.manual_seed(0)
+    gen_s = GenerationConfig(max_new_tokens=4, do_sample=True,
+                             num_return_sequences=3, eos_token_id=2,
+                             pad_token_id=0)
+    out_s, _ = model.generate(ids, gen_s)
+    assert out_s.shape[0] == 6
+
+    # greedy with n>1 is rejected (reference semantics)
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        model.generate(ids, GenerationConfig(max_new_tokens=2,
+                                             num_return_sequences=2))
