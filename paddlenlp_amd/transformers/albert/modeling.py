@@ -7,7 +7,6 @@ num_hidden_layers applications.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

@@ -5,7 +5,6 @@ y1 in a 0..max_2d grid, plus width/height), over the shared encoder core.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
