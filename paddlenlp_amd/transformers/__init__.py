@@ -97,3 +97,4 @@ from .ernie_layout import (  # noqa: F401
     ErnieLayoutForTokenClassification,
     ErnieLayoutModel,
 )
+from .jamba import JambaConfig, JambaForCausalLM, JambaModel  # noqa: F401

@@ -39,6 +39,8 @@ MODEL_REGISTRY = {
               "causal_lm": "MambaForCausalLM", "base": "MambaModel"},
     "gptj": {"module": "gptj", "config": "GPTJConfig",
              "causal_lm": "GPTJForCausalLM", "base": "GPTJModel"},
+    "jamba": {"module": "jamba", "config": "JambaConfig",
+              "causal_lm": "JambaForCausalLM", "base": "JambaModel"},
     "t5": {"module": "t5", "config": "T5Config", "base": "T5Model",
            "seq2seq_lm": "T5ForConditionalGeneration"},
     "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",
