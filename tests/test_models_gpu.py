@@ -96,3 +96,39 @@ def test_qlora_nf4_gpu(device):
                if p.requires_grad and "lora_B" in n]
     assert b_grads and all(g is not None and torch.isfinite(g).all()
                            for g in b_grads)
+
+
+def test_deepseek_v2_gpu(device):
+    from paddlenlp_amd.transformers import DeepseekV2Config, DeepseekV2ForCausalLM
+
+    torch.manual_seed(0)
+    cfg = DeepseekV2Config(
+        vocab_size=512, hidden_size=128, intermediate_size=256,
+        moe_intermediate_size=64, num_hidden_layers=2, num_attention_heads=4,
+        q_lora_rank=32, kv_lora_rank=16, qk_nope_head_dim=16,
+        qk_rope_head_dim=8, v_head_dim=16, n_routed_experts=4,
+        n_shared_experts=1, num_experts_per_tok=2, n_group=2, topk_group=1,
+        first_k_dense_replace=1, max_position_embeddings=128)
+    m = DeepseekV2ForCausalLM(cfg).to(device).to(torch.bfloat16)
+    ids = torch.randint(3, 512, (2, 32), device=device)
+    loss, _ = m(input_ids=ids, labels=ids)
+    loss.backward()
+    assert torch.isfinite(loss)
+
+
+def test_jamba_gpu(device):
+    from paddlenlp_amd.transformers import JambaConfig, JambaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = JambaConfig(vocab_size=512, hidden_size=128, intermediate_size=256,
+                      num_hidden_layers=4, num_attention_heads=4,
+                      num_key_value_heads=2, attn_layer_period=4,
+                      attn_layer_offset=1, expert_layer_period=2,
+                      expert_layer_offset=0, num_experts=4,
+                      num_experts_per_tok=2, mamba_d_state=8,
+                      max_position_embeddings=128)
+    m = JambaForCausalLM(cfg).to(device).to(torch.bfloat16)
+    ids = torch.randint(3, 512, (2, 32), device=device)
+    loss, _ = m(input_ids=ids, labels=ids)
+    loss.backward()
+    assert torch.isfinite(loss)
