@@ -96,6 +96,8 @@ def test_swiglu_fwd_bwd(C):
     (1, 128, 128, 2, 2, 64, True),      # head dim 64
     (2, 128, 128, 4, 4, 128, False),    # non-causal
     (1, 64, 192, 2, 2, 128, True),      # Skv > Sq (cached decode pattern)
+    (1, 128, 64, 2, 2, 128, False),     # Sq > Skv non-causal (zigzag CP block)
+    (1, 64, 128, 2, 2, 128, False),     # Skv > Sq non-causal (zigzag CP block)
 ])
 def test_flash_attention_fwd(C, B, Sq, Skv, Hq, Hk, D, causal):
     from paddlenlp_amd import ops
