@@ -98,3 +98,14 @@ from .ernie_layout import (  # noqa: F401
     ErnieLayoutModel,
 )
 from .jamba import JambaConfig, JambaForCausalLM, JambaModel  # noqa: F401
+from .qwen2 import Qwen2Config, Qwen2ForCausalLM, Qwen2Model  # noqa: F401
+from .mistral import (  # noqa: F401
+    MistralConfig,
+    MistralForCausalLM,
+    MistralModel,
+)
+from .mixtral import (  # noqa: F401
+    MixtralConfig,
+    MixtralForCausalLM,
+    MixtralModel,
+)

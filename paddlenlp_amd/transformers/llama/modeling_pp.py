@@ -16,7 +16,7 @@ from .configuration import LlamaConfig
 from .modeling import LlamaDecoderLayer, LlamaPretrainingCriterion, LlamaRMSNorm, _Linear
 from ...parallel.tensor_parallel import ColumnParallelLinear, VocabParallelEmbedding
 
-__all__ = ["LlamaForCausalLMPipe"]
+__all__ = ["LlamaForCausalLMPipe", "Qwen2ForCausalLMPipe", "MistralForCausalLMPipe"]
 
 
 class EmbeddingPipe(nn.Module):
@@ -62,9 +62,14 @@ class LlamaForCausalLMPipe(PipelineModule):
     Weight names intentionally match LlamaForCausalLM once remapped by
     `pp_param_name_map` (PipelinePretrainedModel behavior in the reference
     model_utils.py: maps pipe-layer param names back to base names).
+
+    Subclasses override `decoder_layer_cls` / `base_prefix` to give any
+    llama-shaped family a pipe variant (Qwen2/Mistral below).
     """
 
     config_class = LlamaConfig
+    decoder_layer_cls = LlamaDecoderLayer
+    base_prefix = "llama"
 
     @classmethod
     def _get_tensor_parallel_mappings(cls, config, is_split=True):
@@ -76,7 +81,8 @@ class LlamaForCausalLMPipe(PipelineModule):
         criterion = LlamaPretrainingCriterion(config)
         descs = [LayerDesc(EmbeddingPipe, config, name="embedding")]
         for i in range(config.num_hidden_layers):
-            descs.append(LayerDesc(LlamaDecoderLayer, config, i, name=f"layer_{i}"))
+            descs.append(LayerDesc(self.decoder_layer_cls, config, i,
+                                   name=f"layer_{i}"))
         descs.append(LayerDesc(RMSNormPipe, config, name="final_norm"))
         descs.append(LayerDesc(LMHeadPipe, config, name="lm_head"))
 
@@ -86,7 +92,7 @@ class LlamaForCausalLMPipe(PipelineModule):
         super().__init__(
             descs,
             loss_fn=loss_fn,
-            seg_method="layer:LlamaDecoderLayer",
+            seg_method=f"layer:{self.decoder_layer_cls.__name__}",
             topology=get_topology(),
             num_virtual_stages=num_virtual_stages,
         )
@@ -100,12 +106,13 @@ class LlamaForCausalLMPipe(PipelineModule):
             for pname, _ in layer.named_parameters():
                 local = f"local_layers.{idx}.{pname}"
                 if desc.name == "embedding":
-                    base = pname.replace("embed_tokens", "llama.embed_tokens")
+                    base = pname.replace(
+                        "embed_tokens", f"{self.base_prefix}.embed_tokens")
                 elif desc.name.startswith("layer_"):
                     li = desc.name.split("_")[1]
-                    base = f"llama.layers.{li}.{pname}"
+                    base = f"{self.base_prefix}.layers.{li}.{pname}"
                 elif desc.name == "final_norm":
-                    base = f"llama.norm.{pname}"
+                    base = f"{self.base_prefix}.norm.{pname}"
                 else:  # lm_head
                     base = pname if pname.startswith("lm_head") else f"lm_head.{pname}"
                 mapping[local] = base
@@ -125,3 +132,34 @@ class LlamaForCausalLMPipe(PipelineModule):
         missing, unexpected = self.load_state_dict(local_sd, strict=False)
         real_missing = [m for m in missing if not m.startswith("local_layers")] if strict else []
         return missing, unexpected
+
+
+def _make_pipe_variant(name, config_cls, layer_cls, prefix, doc):
+    return type(name, (LlamaForCausalLMPipe,), {
+        "config_class": config_cls,
+        "decoder_layer_cls": layer_cls,
+        "base_prefix": prefix,
+        "__doc__": doc,
+    })
+
+
+def _qwen2_pipe():
+    from ..qwen2.configuration import Qwen2Config
+    from ..qwen2.modeling import Qwen2DecoderLayer
+
+    return _make_pipe_variant(
+        "Qwen2ForCausalLMPipe", Qwen2Config, Qwen2DecoderLayer, "qwen2",
+        "Qwen2 pipeline variant (biased-QKV llama-shape layers).")
+
+
+def _mistral_pipe():
+    from ..mistral.configuration import MistralConfig
+
+    # mistral reuses the llama decoder layer
+    return _make_pipe_variant(
+        "MistralForCausalLMPipe", MistralConfig, LlamaDecoderLayer,
+        "mistral", "Mistral pipeline variant.")
+
+
+Qwen2ForCausalLMPipe = _qwen2_pipe()
+MistralForCausalLMPipe = _mistral_pipe()

@@ -230,3 +230,46 @@ def test_vpp_interleaved_parity():
 
 def test_vpp_v3_min_microbatches():
     _run_workers(_w_vpp_llama, extra=(3, 2))
+
+
+def _w_qwen2_pipe(rank, world):
+    """Qwen2 pipe variant: world-2 1F1B parity with the single-process model."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.pipeline import PipelineEngine
+    from paddlenlp_amd.transformers import Qwen2Config, Qwen2ForCausalLM
+    from paddlenlp_amd.transformers.llama.modeling_pp import Qwen2ForCausalLMPipe
+
+    topo = init_parallel_env(pp_degree=world, backend="gloo")
+    cfg = Qwen2Config(
+        vocab_size=96, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, dtype="float32")
+    torch.manual_seed(5)
+    full = Qwen2ForCausalLM.from_config(cfg)
+    pipe = Qwen2ForCausalLMPipe(cfg)
+    pipe.load_base_state_dict(full.state_dict())
+    engine = PipelineEngine(
+        pipe, hidden_shape_fn=lambda mb: (*mb["input_ids"].shape, cfg.hidden_size),
+        dtype=torch.float32, device=torch.device("cpu"))
+
+    g = torch.Generator().manual_seed(9)
+    mbs = [{"input_ids": torch.randint(0, 96, (2, 8), generator=g),
+            "labels": torch.randint(0, 96, (2, 8), generator=g)}
+           for _ in range(2)]
+    loss = engine.forward_backward(mbs, input_fn=lambda mb: mb["input_ids"])
+    ref_losses = []
+    for mb in mbs:
+        l, _ = full(input_ids=mb["input_ids"], labels=mb["labels"])
+        (l / len(mbs)).backward()
+        ref_losses.append(l.detach())
+    if pipe.is_last_stage:
+        assert torch.allclose(loss, torch.stack(ref_losses).mean(), atol=1e-5)
+    name_map = pipe.pp_param_name_map()
+    ref_params = dict(full.named_parameters())
+    for local_name, p in pipe.named_parameters():
+        assert torch.allclose(p.grad, ref_params[name_map[local_name]].grad,
+                              atol=1e-5), local_name
+
+
+def test_qwen2_pipe_parity():
+    _run_workers(_w_qwen2_pipe)
