@@ -118,10 +118,19 @@ def main():
         logger.warning(f"No tokenizer found at {tok_path}; continuing without one")
 
     if training_args.pipeline_parallel_degree > 1:
-        from paddlenlp_amd.transformers.llama.modeling_pp import LlamaForCausalLMPipe
+        from paddlenlp_amd.transformers.llama.modeling_pp import (
+            LlamaForCausalLMPipe,
+            MistralForCausalLMPipe,
+            Qwen2ForCausalLMPipe,
+        )
 
-        assert config.model_type == "llama", "pp entry currently wires the llama pipe model"
-        model = LlamaForCausalLMPipe(config)
+        pipe_registry = {"llama": LlamaForCausalLMPipe,
+                         "qwen2": Qwen2ForCausalLMPipe,
+                         "mistral": MistralForCausalLMPipe}
+        assert config.model_type in pipe_registry, \
+            f"no pipeline variant for model_type {config.model_type!r}"
+        model = pipe_registry[config.model_type](
+            config, num_virtual_stages=training_args.virtual_pp_degree)
         if training_args.bf16:
             model = model.to(torch.bfloat16)
     else:

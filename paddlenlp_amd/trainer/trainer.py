@@ -195,7 +195,11 @@ class Trainer:
 
         topo = self.topology
         if topo.pp_degree > 1:
-            from ..parallel.pipeline import PipelineEngine, PipelineModule
+            from ..parallel.pipeline import (
+                InterleavedPipelineEngine,
+                PipelineEngine,
+                PipelineModule,
+            )
 
             if not isinstance(unwrap_model(model), PipelineModule):
                 raise ValueError(
@@ -205,7 +209,9 @@ class Trainer:
             pipe = unwrap_model(model)
             hidden = pipe.config.hidden_size
 
-            self._pipe_engine = PipelineEngine(
+            engine_cls = (InterleavedPipelineEngine
+                          if pipe.num_virtual_stages > 1 else PipelineEngine)
+            self._pipe_engine = engine_cls(
                 pipe,
                 hidden_shape_fn=lambda mb: (*mb["input_ids"].shape, hidden),
                 dtype=args.compute_dtype,

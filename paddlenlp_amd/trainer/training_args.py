@@ -72,6 +72,8 @@ class TrainingArguments:
     sharding_comm_buffer_size_MB: int = 256
     # overlap the sharding-group gradient reduction with backward compute
     sharding_overlap_comm: bool = True
+    # interleaved virtual pipeline stages per rank (VPP; 1 = plain 1F1B)
+    virtual_pp_degree: int = 1
     # [[start, end], ...] 1-based global-step intervals whose data is skipped
     # (corrupted-range replay jump, reference trainer.py should_skip_data)
     skip_data_intervals: Optional[List[List[int]]] = None
