@@ -96,7 +96,13 @@ def apply_shift(model: nn.Module, dataloader, num_batches: int = 8):
     ahead of each norm->linear pair and compensate in the linear bias
     (reference apply_shift :119).  Requires the norm to expose a bias or the
     linears to have biases; linears without bias get one."""
+    from paddlenlp_amd.transformers.llama.modeling import LlamaRMSNorm
+
     pairs = _norm_linear_pairs(model)
+    # only norms whose forward APPLIES shift_bias may carry the shift —
+    # an inert buffer would silently corrupt outputs (bias compensation
+    # without the matching subtraction)
+    pairs = [(n, l) for n, l in pairs if isinstance(n, LlamaRMSNorm)]
     names = [n for _, lins in pairs for n, _ in lins]
     stats = collect_activation_stats(model, dataloader, num_batches, names)
     shifted = 0

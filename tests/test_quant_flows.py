@@ -143,3 +143,24 @@ def test_ptq_and_pack():
     with torch.no_grad():
         out = m2(input_ids=ids)
     assert torch.isfinite(out).all()
+
+
+def test_shift_skips_norms_without_shift_support():
+    """A norm class that does not apply shift_bias must be left alone
+    (silent output corruption guard)."""
+    from paddlenlp_amd.transformers import GemmaConfig, GemmaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = GemmaConfig(vocab_size=96, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=4,
+                      num_key_value_heads=2, head_dim=8,
+                      max_position_embeddings=64)
+    m = GemmaForCausalLM(cfg).eval()
+    ids = torch.randint(3, 96, (1, 8))
+    with torch.no_grad():
+        ref = m(input_ids=ids)
+    apply_shift(m, iter([{"input_ids": ids}]), 1)
+    assert not hasattr(m.gemma.layers[0].input_layernorm, "shift_bias")
+    with torch.no_grad():
+        out = m(input_ids=ids)
+    torch.testing.assert_close(out, ref)
