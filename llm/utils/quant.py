@@ -134,7 +134,12 @@ def apply_smooth(model: nn.Module, dataloader, alpha: float = 0.5,
     s (folded into the preceding RMSNorm weight), weights multiply by s.
     With do_awq=True, alpha is grid-searched per pair to minimize int8
     weight-quant output error (AWQ-style)."""
+    from paddlenlp_amd.transformers.llama.modeling import LlamaRMSNorm
+
     pairs = _norm_linear_pairs(model)
+    # scale folds into norms whose output is LINEAR in the weight
+    # (out = xhat * w); (1+w)-style norms (Gemma) cannot absorb a division
+    pairs = [(n, l) for n, l in pairs if isinstance(n, LlamaRMSNorm)]
     names = [n for _, lins in pairs for n, _ in lins]
     stats = collect_activation_stats(model, dataloader, num_batches, names)
     for norm, lins in pairs:

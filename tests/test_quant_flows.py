@@ -164,3 +164,21 @@ def test_shift_skips_norms_without_shift_support():
     with torch.no_grad():
         out = m(input_ids=ids)
     torch.testing.assert_close(out, ref)
+
+
+def test_smooth_skips_non_linear_norms():
+    from paddlenlp_amd.transformers import GemmaConfig, GemmaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = GemmaConfig(vocab_size=96, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=4,
+                      num_key_value_heads=2, head_dim=8,
+                      max_position_embeddings=64)
+    m = GemmaForCausalLM(cfg).eval()
+    ids = torch.randint(3, 96, (1, 8))
+    with torch.no_grad():
+        ref = m(input_ids=ids)
+    apply_smooth(m, iter([{"input_ids": ids}]), num_batches=1)
+    with torch.no_grad():
+        out = m(input_ids=ids)
+    torch.testing.assert_close(out, ref)  # untouched
