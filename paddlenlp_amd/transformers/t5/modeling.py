@@ -323,6 +323,16 @@ class T5ForConditionalGeneration(T5PretrainedModel):
         for k, v in kwargs.items():
             if hasattr(gen, k):
                 setattr(gen, k, v)
+        if gen.num_beams > 1:
+            from ...generation.seq2seq_utils import seq2seq_beam_search
+
+            eos = (gen.eos_ids() or [self.config.eos_token_id])[0]
+            pad = gen.pad_token_id if gen.pad_token_id is not None \
+                else self.config.pad_token_id
+            return seq2seq_beam_search(
+                self, input_ids, gen,
+                start_token_id=self.config.decoder_start_token_id,
+                eos_token_id=eos, pad_token_id=pad)
         B = input_ids.shape[0]
         device = input_ids.device
         enc = self.t5.encoder(input_ids)

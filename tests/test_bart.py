@@ -75,3 +75,10 @@ def test_bart_generate_and_save_load(tmp_path):
         a = m(input_ids=src, labels=labels)[1]
         b = m2(input_ids=src, labels=labels)[1]
     torch.testing.assert_close(a, b)
+
+
+def test_bart_beam_search():
+    m = BartForConditionalGeneration(tiny_bart()).eval()
+    src = torch.randint(3, 120, (2, 8))
+    out, _ = m.generate(src, max_new_tokens=5, num_beams=3)
+    assert out.shape[0] == 2 and out.shape[1] <= 5

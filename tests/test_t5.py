@@ -156,3 +156,29 @@ def test_hf_t5_name_conversion():
     with torch.no_grad():
         torch.testing.assert_close(m(src, decoder_input_ids=dec),
                                    m2(src, decoder_input_ids=dec))
+
+
+def test_t5_beam_search():
+    m = T5ForConditionalGeneration(tiny_t5()).eval()
+    src = torch.randint(2, 100, (2, 8))
+    out, _ = m.generate(src, max_new_tokens=5, num_beams=3)
+    assert out.shape[0] == 2 and out.shape[1] <= 5
+    # beam-1 equals greedy
+    b1, _ = m.generate(src, max_new_tokens=5, num_beams=1, do_sample=False)
+    # beam search maximizes sequence logprob; its score must be >= greedy's
+    def score(dec):
+        start = torch.full((dec.shape[0], 1), m.config.decoder_start_token_id,
+                           dtype=torch.long)
+        inp = torch.cat([start, dec[:, :-1]], dim=1)
+        with torch.no_grad():
+            logits = m(input_ids=src, decoder_input_ids=inp)
+        lp = logits.float().log_softmax(-1)
+        return lp.gather(-1, dec.unsqueeze(-1)).squeeze(-1).sum(-1)
+    L = min(out.shape[1], b1.shape[1])
+    if out.shape[1] == b1.shape[1]:
+        assert (score(out) >= score(b1) - 1e-4).all()
+
+    # num_return_sequences through the seq2seq beam
+    multi, _ = m.generate(src, max_new_tokens=4, num_beams=4,
+                          num_return_sequences=2)
+    assert multi.shape[0] == 4
