@@ -173,3 +173,176 @@ class ChunkEvaluator:
         r = self.n_correct / max(1, self.n_gold)
         return {"precision": p, "recall": r,
                 "f1": 2 * p * r / max(1e-12, p + r)}
+
+
+class RougeN:
+    """N-gram ROUGE recall (reference metrics/rouge.py RougeN)."""
+
+    def __init__(self, n: int = 2):
+        self.n = n
+        self.reset()
+
+    def reset(self):
+        self.overlap = 0
+        self.total = 0
+
+    def update(self, candidate: Sequence, references):
+        from collections import Counter
+
+        cand = Counter(_ngrams(list(candidate), self.n))
+        for ref in references:
+            refc = Counter(_ngrams(list(ref), self.n))
+            self.overlap += sum((cand & refc).values())
+            self.total += max(1, sum(refc.values()))
+
+    def accumulate(self) -> float:
+        return self.overlap / max(1, self.total)
+
+
+class Rouge2(RougeN):
+    def __init__(self):
+        super().__init__(n=2)
+
+
+class Distinct:
+    """distinct-n diversity (reference metrics/distinct.py): unique n-grams
+    over total n-grams across all updates."""
+
+    def __init__(self, n: int = 2):
+        self.n = n
+        self.reset()
+
+    def reset(self):
+        self.seen = set()
+        self.total = 0
+
+    def update(self, tokens: Sequence):
+        grams = _ngrams(list(tokens), self.n)
+        self.seen.update(grams)
+        self.total += len(grams)
+
+    def accumulate(self) -> float:
+        return len(self.seen) / max(1, self.total)
+
+
+class Mcc:
+    """Matthews correlation coefficient (reference metrics/glue.py Mcc)."""
+
+    def reset(self):
+        self.tp = self.fp = self.tn = self.fn = 0
+
+    def __init__(self):
+        self.reset()
+
+    def update(self, preds, labels):
+        for p, l in zip(_aslist(preds), _aslist(labels)):
+            if p == 1 and l == 1:
+                self.tp += 1
+            elif p == 1:
+                self.fp += 1
+            elif l == 1:
+                self.fn += 1
+            else:
+                self.tn += 1
+
+    def accumulate(self) -> float:
+        import math
+
+        num = self.tp * self.tn - self.fp * self.fn
+        den = math.sqrt((self.tp + self.fp) * (self.tp + self.fn)
+                        * (self.tn + self.fp) * (self.tn + self.fn))
+        return num / den if den else 0.0
+
+
+class PearsonAndSpearman:
+    """Pearson + Spearman correlations (reference metrics/glue.py)."""
+
+    def __init__(self):
+        self.reset()
+
+    def reset(self):
+        self.preds = []
+        self.labels = []
+
+    def update(self, preds, labels):
+        self.preds.extend(float(x) for x in _aslist(preds))
+        self.labels.extend(float(x) for x in _aslist(labels))
+
+    @staticmethod
+    def _pearson(a, b):
+        import math
+
+        n = len(a)
+        ma = sum(a) / n
+        mb = sum(b) / n
+        cov = sum((x - ma) * (y - mb) for x, y in zip(a, b))
+        va = math.sqrt(sum((x - ma) ** 2 for x in a))
+        vb = math.sqrt(sum((y - mb) ** 2 for y in b))
+        return cov / (va * vb) if va and vb else 0.0
+
+    @staticmethod
+    def _ranks(v):
+        order = sorted(range(len(v)), key=lambda i: v[i])
+        ranks = [0.0] * len(v)
+        for r, i in enumerate(order):
+            ranks[i] = float(r)
+        return ranks
+
+    def accumulate(self):
+        p = self._pearson(self.preds, self.labels)
+        s = self._pearson(self._ranks(self.preds), self._ranks(self.labels))
+        return {"pearson": p, "spearman": s,
+                "corr": (p + s) / 2}
+
+
+class MRR:
+    """Mean reciprocal rank (reference metrics/mrr.py)."""
+
+    def __init__(self):
+        self.reset()
+
+    def reset(self):
+        self.total = 0.0
+        self.count = 0
+
+    def update(self, rank: int):
+        """rank: 1-based position of the first relevant item (0 = none)."""
+        self.total += 1.0 / rank if rank > 0 else 0.0
+        self.count += 1
+
+    def accumulate(self) -> float:
+        return self.total / max(1, self.count)
+
+
+class SpanEvaluator:
+    """Span-level precision/recall/F1 over (start, end) pairs (reference
+    metrics/span.py, the UIE evaluation metric)."""
+
+    def __init__(self):
+        self.reset()
+
+    def reset(self):
+        self.num_correct = 0
+        self.num_infer = 0
+        self.num_label = 0
+
+    def update(self, pred_spans, gold_spans):
+        pred = set(map(tuple, pred_spans))
+        gold = set(map(tuple, gold_spans))
+        self.num_correct += len(pred & gold)
+        self.num_infer += len(pred)
+        self.num_label += len(gold)
+
+    def accumulate(self):
+        p = self.num_correct / self.num_infer if self.num_infer else 0.0
+        r = self.num_correct / self.num_label if self.num_label else 0.0
+        f1 = 2 * p * r / (p + r) if p + r else 0.0
+        return p, r, f1
+
+
+def _aslist(x):
+    if hasattr(x, "tolist"):
+        x = x.tolist()
+    if isinstance(x, (int, float)):
+        return [x]
+    return list(x)

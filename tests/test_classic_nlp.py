@@ -123,3 +123,46 @@ def test_char_augment():
         out = cls(aug_n=1, seed=3).augment(s)[0]
         assert out != s or cls is CharSwap  # swap can no-op on equal chars
         assert len(out.split()) == 2
+
+
+def test_extended_metrics():
+    from paddlenlp_amd.metrics import (
+        MRR,
+        Distinct,
+        Mcc,
+        PearsonAndSpearman,
+        Rouge2,
+        SpanEvaluator,
+    )
+
+    r2 = Rouge2()
+    r2.update("the cat sat on".split(), ["the cat sat down".split()])
+    assert 0 < r2.accumulate() < 1  # "the cat", "cat sat" match of 3
+
+    d = Distinct(2)
+    d.update(["a", "b", "a", "b"])
+    assert d.accumulate() == 2 / 3  # (a,b) (b,a) (a,b) -> 2 unique of 3
+
+    mcc = Mcc()
+    mcc.update([1, 1, 0, 0], [1, 1, 0, 0])
+    assert abs(mcc.accumulate() - 1.0) < 1e-9
+    mcc.reset()
+    mcc.update([1, 0], [0, 1])
+    assert mcc.accumulate() <= 0
+
+    ps = PearsonAndSpearman()
+    ps.update([1.0, 2.0, 3.0], [2.0, 4.0, 6.0])
+    out = ps.accumulate()
+    assert abs(out["pearson"] - 1.0) < 1e-9
+    assert abs(out["spearman"] - 1.0) < 1e-9
+
+    mrr = MRR()
+    mrr.update(1)
+    mrr.update(2)
+    mrr.update(0)
+    assert abs(mrr.accumulate() - (1 + 0.5 + 0) / 3) < 1e-9
+
+    se = SpanEvaluator()
+    se.update([(0, 2), (5, 7)], [(0, 2), (3, 4)])
+    p, r, f1 = se.accumulate()
+    assert p == 0.5 and r == 0.5 and abs(f1 - 0.5) < 1e-9
