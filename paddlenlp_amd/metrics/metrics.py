@@ -346,3 +346,43 @@ def _aslist(x):
     if isinstance(x, (int, float)):
         return [x]
     return list(x)
+
+
+class DetectionF1:
+    """Sentence-level error-detection F1 for text correction (reference
+    metrics/sighan.py): a prediction is correct when it flags exactly the
+    gold error positions."""
+
+    def __init__(self):
+        self.reset()
+
+    def reset(self):
+        self.tp = self.fp = self.fn = 0
+
+    def update(self, pred_positions, gold_positions):
+        pred = set(pred_positions)
+        gold = set(gold_positions)
+        if not gold:
+            if pred:
+                self.fp += 1
+            return
+        if pred == gold:
+            self.tp += 1
+        else:
+            if pred:
+                self.fp += 1
+            self.fn += 1
+
+    def accumulate(self):
+        p = self.tp / (self.tp + self.fp) if self.tp + self.fp else 0.0
+        r = self.tp / (self.tp + self.fn) if self.tp + self.fn else 0.0
+        f1 = 2 * p * r / (p + r) if p + r else 0.0
+        return p, r, f1
+
+
+class CorrectionF1(DetectionF1):
+    """Correction F1: positions AND corrected tokens must match."""
+
+    def update(self, pred_corrections, gold_corrections):
+        super().update(set(map(tuple, pred_corrections)),
+                       set(map(tuple, gold_corrections)))

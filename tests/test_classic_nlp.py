@@ -166,3 +166,20 @@ def test_extended_metrics():
     se.update([(0, 2), (5, 7)], [(0, 2), (3, 4)])
     p, r, f1 = se.accumulate()
     assert p == 0.5 and r == 0.5 and abs(f1 - 0.5) < 1e-9
+
+
+def test_sighan_detection_correction_f1():
+    from paddlenlp_amd.metrics import CorrectionF1, DetectionF1
+
+    det = DetectionF1()
+    det.update([2, 5], [2, 5])   # exact -> tp
+    det.update([1], [1, 3])      # partial -> fp + fn
+    det.update([], [])           # clean sentence, clean prediction
+    p, r, f1 = det.accumulate()
+    assert p == 0.5 and r == 0.5
+
+    cor = CorrectionF1()
+    cor.update([(2, "cat")], [(2, "cat")])
+    cor.update([(2, "dog")], [(2, "cat")])  # right position, wrong token
+    p, r, f1 = cor.accumulate()
+    assert p == 0.5 and r == 0.5
