@@ -109,3 +109,9 @@ from .mixtral import (  # noqa: F401
     MixtralForCausalLM,
     MixtralModel,
 )
+from .ernie_m import (  # noqa: F401
+    ErnieMConfig,
+    ErnieMForSequenceClassification,
+    ErnieMForTokenClassification,
+    ErnieMModel,
+)

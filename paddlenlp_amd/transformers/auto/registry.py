@@ -50,6 +50,10 @@ MODEL_REGISTRY = {
              "token_classification": "BertForTokenClassification",
              "question_answering": "BertForQuestionAnswering",
              "masked_lm": "BertForMaskedLM"},
+    "ernie_m": {"module": "ernie_m", "config": "ErnieMConfig",
+                "base": "ErnieMModel",
+                "sequence_classification": "ErnieMForSequenceClassification",
+                "token_classification": "ErnieMForTokenClassification"},
     "ernie": {"module": "ernie", "config": "ErnieConfig", "base": "ErnieModel",
               "sequence_classification": "ErnieForSequenceClassification",
               "token_classification": "ErnieForTokenClassification",
