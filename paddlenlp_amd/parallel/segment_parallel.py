@@ -17,10 +17,14 @@ import torch.distributed as dist
 from .topology import get_topology
 
 
-def split_inputs_sequence_dim(inputs: dict, group=None) -> dict:
-    """Slice every [B, S, ...] tensor to this rank's seq chunk.
+def split_inputs_sequence_dim(inputs: dict, group=None,
+                              balanced: bool = False) -> dict:
+    """Slice every [B, S, ...] tensor to this rank's seq chunk — contiguous,
+    or zigzag (chunks r and 2w-1-r) when `balanced` so causal ring-attention
+    work is equal across ranks.
 
-    Reference: trainer.py:972-973 split_inputs_sequence_dim."""
+    Reference: trainer.py:972-973 split_inputs_sequence_dim +
+    context_parallel_utils.py:43-50 load-balanced chunking."""
     group = group or get_topology().sep_parallel_group
     if group is None:
         return inputs
@@ -28,8 +32,14 @@ def split_inputs_sequence_dim(inputs: dict, group=None) -> dict:
     world = dist.get_world_size(group)
     out = {}
     for key, v in inputs.items():
-        if isinstance(v, torch.Tensor) and v.dim() >= 2 and v.shape[1] % world == 0:
-            out[key] = v.chunk(world, dim=1)[rank].contiguous()
+        if isinstance(v, torch.Tensor) and v.dim() >= 2 and \
+                v.shape[1] % (2 * world if balanced else world) == 0:
+            if balanced:
+                from .ring_attention import zigzag_split
+
+                out[key] = zigzag_split(v, world, rank, dim=1)
+            else:
+                out[key] = v.chunk(world, dim=1)[rank].contiguous()
         else:
             out[key] = v
     return out

@@ -432,7 +432,9 @@ class Trainer:
             # (reference split_inputs_sequence_dim, trainer.py:972-975)
             from ..parallel.segment_parallel import split_inputs_sequence_dim
 
-            inputs = split_inputs_sequence_dim(inputs, self.topology.sep_parallel_group)
+            inputs = split_inputs_sequence_dim(
+                inputs, self.topology.sep_parallel_group,
+                balanced=self.args.context_parallel_balanced)
         loss = self.compute_loss(model, inputs)
         if self.args.gradient_accumulation_steps > 1:
             loss = loss / self.args.gradient_accumulation_steps

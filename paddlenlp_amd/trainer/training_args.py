@@ -68,6 +68,8 @@ class TrainingArguments:
     sharding_parallel_degree: int = -1
     sep_parallel_degree: int = 1
     context_parallel_degree: int = 1
+    # zigzag load-balanced CP shards (rank r owns chunks r and 2w-1-r)
+    context_parallel_balanced: bool = False
     sharding: str = ""  # "stage1" | "stage2" | "stage3" | "" (space-separated options)
     sharding_comm_buffer_size_MB: int = 256
     # overlap the sharding-group gradient reduction with backward compute

@@ -40,6 +40,7 @@ class PretrainedConfig:
         self.sequence_parallel = kwargs.pop("sequence_parallel", False)
         self.sep_parallel_degree = kwargs.pop("sep_parallel_degree", 1)
         self.context_parallel_degree = kwargs.pop("context_parallel_degree", 1)
+        self.context_parallel_balanced = kwargs.pop("context_parallel_balanced", False)
         self.pipeline_parallel_degree = kwargs.pop("pipeline_parallel_degree", 1)
         self.recompute = kwargs.pop("recompute", False)
         self.recompute_granularity = kwargs.pop("recompute_granularity", "full")
@@ -142,6 +143,7 @@ class LlmMetaConfig:
         "sequence_parallel",
         "sep_parallel_degree",
         "context_parallel_degree",
+        "context_parallel_balanced",
         "pipeline_parallel_degree",
         # recompute
         "recompute",

@@ -183,11 +183,24 @@ class LlamaAttention(nn.Module):
         v = v.view(B, S, kl, d)
 
         # sep/cp: this rank holds a seq chunk; rope positions are offset
-        if self.sep_degree > 1 or self.cp_degree > 1:
+        balanced_cp = (self.cp_degree > 1 and
+                       getattr(self.config, "context_parallel_balanced", False))
+        if balanced_cp:
+            # zigzag shard: local seq = [chunk r | chunk 2w-1-r] of 2w chunks
             sep_rank = get_topology().get_rank_in("sep")
-            position_offset = position_offset + sep_rank * S
-
-        cos, sin = self.rotary_emb(S, hidden_states.device, position_offset)
+            w = self.cp_degree
+            C = S // 2
+            dev = hidden_states.device
+            cos1, sin1 = self.rotary_emb(C, dev, position_offset + sep_rank * C)
+            cos2, sin2 = self.rotary_emb(
+                C, dev, position_offset + (2 * w - 1 - sep_rank) * C)
+            cos = torch.cat([cos1, cos2], dim=0)
+            sin = torch.cat([sin1, sin2], dim=0)
+        else:
+            if self.sep_degree > 1 or self.cp_degree > 1:
+                sep_rank = get_topology().get_rank_in("sep")
+                position_offset = position_offset + sep_rank * S
+            cos, sin = self.rotary_emb(S, hidden_states.device, position_offset)
         if self.config.use_fused_rope:
             q, k = ops.fused_rope(q, k, cos, sin)
         else:
@@ -201,7 +214,9 @@ class LlamaAttention(nn.Module):
         if self.cp_degree > 1:
             from ...parallel.ring_attention import ring_flash_attention
 
-            attn_out = ring_flash_attention(q, k, v, causal=True)
+            attn_out = ring_flash_attention(
+                q, k, v, causal=True,
+                balanced=getattr(self.config, "context_parallel_balanced", False))
         elif self.sep_degree > 1:
             # Ulysses: [B, S/sep, H, D] -> [B, S, H/sep, D] around the core
             q = self.reshard.seq_to_head(q)

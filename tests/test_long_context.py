@@ -312,3 +312,35 @@ def test_zigzag_split_gather_roundtrip():
 
 def test_ring_attention_balanced_zigzag_world4():
     _run_workers(_w_ring_balanced, world_size=4)
+
+
+def _w_cp_balanced_llama(rank, world):
+    """Model-level balanced (zigzag) CP: logits parity with single process."""
+    from paddlenlp_amd.parallel.ring_attention import zigzag_split
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    init_parallel_env(sep_degree=world, backend="gloo")
+    base_cfg = dict(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, dtype="float32",
+    )
+    torch.manual_seed(41)
+    full = LlamaForCausalLM.from_config(LlamaConfig(**base_cfg))
+    cfg = LlamaConfig(**{**base_cfg, "context_parallel_degree": world,
+                         "context_parallel_balanced": True})
+    torch.manual_seed(41)
+    cp_model = LlamaForCausalLM.from_config(cfg)
+    ids = torch.randint(0, 128, (2, 16),
+                        generator=torch.Generator().manual_seed(42))
+    with torch.no_grad():
+        ref = full(input_ids=ids)
+        local = cp_model(input_ids=zigzag_split(ids, world, rank))
+    expect = zigzag_split(ref, world, rank)
+    assert torch.allclose(local, expect, atol=1e-4), \
+        (local - expect).abs().max()
+
+
+def test_cp_balanced_llama_parity():
+    _run_workers(_w_cp_balanced_llama)
