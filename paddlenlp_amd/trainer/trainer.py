@@ -434,7 +434,10 @@ class Trainer:
 
             inputs = split_inputs_sequence_dim(
                 inputs, self.topology.sep_parallel_group,
-                balanced=self.args.context_parallel_balanced)
+                # zigzag only applies to ring/context parallel; Ulysses
+                # all-to-all needs contiguous chunks
+                balanced=(self.args.context_parallel_balanced
+                          and self.args.context_parallel_degree > 1))
         loss = self.compute_loss(model, inputs)
         if self.args.gradient_accumulation_steps > 1:
             loss = loss / self.args.gradient_accumulation_steps
