@@ -7,10 +7,12 @@ moments/master-weights in optimizer.safetensors / master_weights.safetensors
 with their own indexes; load path re-splits TP on the fly and dispatches
 tensors to whichever rank needs them.
 
-v1 scope (this milestone): correct save/resume round-trip for DP / ZeRO
-sharding under the SAME parallel config, with per-rank optimizer shards and
-a global index.  TP merge-on-save / split-on-load and fully dynamic
-cross-config resharding land with the TP/PP milestone.
+Implemented here: TP merge-on-save (gather + merge actions from
+_get_tensor_parallel_mappings) and split-on-load, per-PP-stage model
+shards with a gathered index, per-parameter-name optimizer/master-weight
+shards, and name-addressed mmap loading that makes resume work across
+world-size and ZeRO-stage changes (tests/test_unified_checkpoint.py
+covers the stage-switch and w2->w1 matrix).
 """
 from __future__ import annotations
 
