@@ -141,7 +141,9 @@ def main():
     if use_sharding:
         zero = ZeroShardedEngine(model, optimizer, stage=2,
                                  group=topo.sharding_parallel_group, bucket_mb=256)
-        zero.enable_overlap_comm()
+        # backward-overlapped grad reduce (PNLP_NO_OVERLAP=1 disables)
+        if os.environ.get("PNLP_NO_OVERLAP", "0") != "1":
+            zero.enable_overlap_comm()
 
     # synthetic data of the benchmark shape (no network: random tokens)
     g = torch.Generator(device="cpu").manual_seed(1234 + rank)
@@ -157,7 +159,8 @@ def main():
         optimizer.zero_grad(set_to_none=True)
         for a in range(args.accum):
             batch = batches[(step_idx * args.accum + a) % len(batches)]
-            if zero is not None and a == args.accum - 1:
+            if zero is not None and getattr(zero, "_overlap", False) \
+                    and a == args.accum - 1:
                 zero.overlap_active = True  # reduce buckets during backward
             loss, _ = model(**batch)
             (loss / args.accum).backward()
