@@ -115,3 +115,9 @@ from .ernie_m import (  # noqa: F401
     ErnieMForTokenClassification,
     ErnieMModel,
 )
+from .blip import (  # noqa: F401
+    BlipConfig,
+    BlipForConditionalGeneration,
+    BlipForImageTextRetrieval,
+    BlipModel,
+)

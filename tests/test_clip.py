@@ -60,3 +60,54 @@ def test_clip_vision_patches():
     x, pooled = m.vision_model(torch.randn(2, 3, 32, 32))
     assert x.shape == (2, 1 + 16, 32)  # class token + (32/8)^2 patches
     assert pooled.shape == (2, 32)
+
+
+def tiny_blip():
+    from paddlenlp_amd.transformers import BlipConfig
+
+    return BlipConfig(
+        text_config=dict(vocab_size=100, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64,
+                         max_position_embeddings=32, bos_token_id=1,
+                         eos_token_id=2),
+        vision_config=dict(hidden_size=32, num_hidden_layers=2,
+                           num_attention_heads=4, intermediate_size=64,
+                           image_size=32, patch_size=8),
+        projection_dim=16)
+
+
+def test_blip_itc_itm_caption():
+    from paddlenlp_amd.transformers import (
+        BlipForConditionalGeneration,
+        BlipForImageTextRetrieval,
+        BlipModel,
+    )
+
+    torch.manual_seed(0)
+    cfg = tiny_blip()
+    ids = torch.randint(3, 100, (3, 8))
+    pix = torch.randn(3, 3, 32, 32)
+
+    itc = BlipModel(cfg)
+    loss, li, lt = itc(ids, pix, return_loss=True)
+    assert li.shape == (3, 3)
+    loss.backward()
+
+    itm = BlipForImageTextRetrieval(cfg)
+    loss, logits = itm(ids, pix, labels=torch.tensor([1, 0, 1]))
+    assert logits.shape == (3, 2)
+    loss.backward()
+    # cross-attention is live: different images change the ITM logits
+    with torch.no_grad():
+        a = itm(ids, pix)
+        b = itm(ids, torch.randn(3, 3, 32, 32))
+    assert not torch.allclose(a, b)
+
+    cap = BlipForConditionalGeneration(cfg)
+    labels = ids.clone()
+    labels[:, :2] = -100
+    loss, logits = cap(pix, ids, labels=labels)
+    assert logits.shape == (3, 8, 100)
+    loss.backward()
+    out = cap.generate(pix, max_new_tokens=4)
+    assert out.shape[0] == 3 and out.shape[1] <= 4
