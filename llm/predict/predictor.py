@@ -48,6 +48,9 @@ class PredictorArgument:
     quant_type: str = field(default="")  # "" | "fp8" | "weight_only_int8"
     use_hipgraph: bool = field(default=False)  # capture decode in a hipGraph
     cachekv_int8: bool = field(default=False)  # int8 paged KV (2x capacity)
+    # speculative decoding: local path of a small draft model (dygraph mode)
+    speculate_model: str = field(default="")
+    speculate_gamma: int = field(default=4)
 
 
 class BasePredictor:
@@ -62,12 +65,24 @@ class BasePredictor:
 
 
 class DygraphPredictor(BasePredictor):
-    """Vanilla model.generate path (reference DygraphPredictor :232)."""
+    """Vanilla model.generate path (reference DygraphPredictor :232);
+    with `speculate_model` set, per-sequence speculative decoding (a small
+    draft proposes speculate_gamma tokens, the target verifies in one
+    forward — greedy output is identical to target-only decoding)."""
 
-    def __init__(self, config, model=None, tokenizer=None):
+    def __init__(self, config, model=None, tokenizer=None, draft_model=None):
         super().__init__(config, tokenizer)
         self.model = model
         self.model.eval()
+        self.draft_model = draft_model
+        if draft_model is None and getattr(config, "speculate_model", ""):
+            from paddlenlp_amd.transformers import AutoModelForCausalLM
+
+            self.draft_model = AutoModelForCausalLM.from_pretrained(
+                config.speculate_model)
+        if self.draft_model is not None:
+            self.draft_model = self.draft_model.to(
+                next(self.model.parameters()).device).eval()
 
     @torch.no_grad()
     def predict(self, texts: List[str]) -> List[str]:
@@ -83,6 +98,24 @@ class DygraphPredictor(BasePredictor):
             eos_token_id=self.tokenizer.eos_token_id,
             pad_token_id=self.tokenizer.pad_token_id or 0,
         )
+        if self.draft_model is not None:
+            from paddlenlp_amd.generation.speculative import (
+                speculative_generate,
+            )
+
+            outs = []
+            mask = inputs.get("attention_mask")
+            for b in range(ids.shape[0]):
+                row = ids[b:b + 1]
+                if mask is not None:  # strip left padding per sequence
+                    keep = mask[b].bool()
+                    row = row[:, keep.to(row.device)]
+                out, _ = speculative_generate(
+                    self.model, self.draft_model, row, gen,
+                    gamma=self.config.speculate_gamma)
+                outs.append(out[0])
+            return [self.tokenizer.decode(o, skip_special_tokens=True)
+                    for o in outs]
         out, _ = self.model.generate(ids, gen)
         return self.tokenizer.batch_decode(out, skip_special_tokens=True)
 

@@ -264,3 +264,40 @@ def test_predictor_int8_cachekv():
     out = pred.predict(["hello world", "the cat"])
     assert len(out) == 2 and all(isinstance(o, str) for o in out)
     assert pred.engine.k_caches[0].dtype == torch.int8
+
+
+def test_dygraph_predictor_speculative():
+    """DygraphPredictor with a draft model: greedy speculative output equals
+    plain greedy output."""
+    import sys
+
+    sys.path.insert(0, "llm")
+    import importlib
+
+    import predict.predictor as predictor_mod
+
+    importlib.reload(predictor_mod)
+    import torch
+
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=128,
+                      eos_token_id=2)
+    model = LlamaForCausalLM.from_config(cfg).eval()
+    draft_cfg = LlamaConfig(vocab_size=128, hidden_size=32,
+                            intermediate_size=64, num_hidden_layers=1,
+                            num_attention_heads=2, num_key_value_heads=2,
+                            max_position_embeddings=128, eos_token_id=2)
+    draft = LlamaForCausalLM.from_config(draft_cfg).eval()
+    tok = _make_tiny_tokenizer()
+    args = predictor_mod.PredictorArgument(
+        src_length=32, max_length=8, decode_strategy="greedy",
+        dtype="float32", speculate_gamma=3)
+    plain = predictor_mod.DygraphPredictor(args, model=model, tokenizer=tok)
+    spec = predictor_mod.DygraphPredictor(args, model=model, tokenizer=tok,
+                                          draft_model=draft)
+    texts = ["hello world", "the cat sat"]
+    assert spec.predict(texts) == plain.predict(texts)
