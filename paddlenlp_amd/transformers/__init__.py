@@ -121,3 +121,15 @@ from .blip import (  # noqa: F401
     BlipForImageTextRetrieval,
     BlipModel,
 )
+from .distilbert import (  # noqa: F401
+    DistilBertConfig,
+    DistilBertForMaskedLM,
+    DistilBertForSequenceClassification,
+    DistilBertModel,
+)
+from .roformer import (  # noqa: F401
+    RoFormerConfig,
+    RoFormerForMaskedLM,
+    RoFormerForSequenceClassification,
+    RoFormerModel,
+)

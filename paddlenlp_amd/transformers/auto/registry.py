@@ -67,6 +67,14 @@ MODEL_REGISTRY = {
     "albert": {"module": "albert", "config": "AlbertConfig", "base": "AlbertModel",
                "sequence_classification": "AlbertForSequenceClassification",
                "masked_lm": "AlbertForMaskedLM"},
+    "distilbert": {"module": "distilbert", "config": "DistilBertConfig",
+                   "base": "DistilBertModel",
+                   "sequence_classification": "DistilBertForSequenceClassification",
+                   "masked_lm": "DistilBertForMaskedLM"},
+    "roformer": {"module": "roformer", "config": "RoFormerConfig",
+                 "base": "RoFormerModel",
+                 "sequence_classification": "RoFormerForSequenceClassification",
+                 "masked_lm": "RoFormerForMaskedLM"},
     "electra": {"module": "electra", "config": "ElectraConfig", "base": "ElectraModel",
                 "sequence_classification": "ElectraForSequenceClassification",
                 "token_classification": "ElectraForTokenClassification"},

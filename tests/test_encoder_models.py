@@ -314,3 +314,47 @@ def test_ernie_m_no_token_type():
         ids, labels=torch.tensor([0, 2]))
     loss.backward()
     assert logits.shape == (2, 3)
+
+
+def test_distilbert_and_roformer():
+    from paddlenlp_amd.transformers import (
+        DistilBertConfig,
+        DistilBertForSequenceClassification,
+        RoFormerConfig,
+        RoFormerForMaskedLM,
+        RoFormerModel,
+    )
+
+    dcfg = DistilBertConfig(vocab_size=120, hidden_size=32,
+                            num_hidden_layers=2, num_attention_heads=4,
+                            intermediate_size=64, max_position_embeddings=64,
+                            hidden_dropout_prob=0.0,
+                            attention_probs_dropout_prob=0.0, num_labels=3)
+    ids = torch.randint(0, 120, (2, 10))
+    m = DistilBertForSequenceClassification(dcfg)
+    # distilled: no token-type embeddings, no pooler
+    assert m.distilbert.embeddings.token_type_embeddings is None
+    assert not hasattr(m.distilbert, "pooler")
+    loss, logits = m(ids, labels=torch.tensor([0, 2]))
+    loss.backward()
+    assert logits.shape == (2, 3)
+
+    rcfg = RoFormerConfig(vocab_size=120, hidden_size=32, num_hidden_layers=2,
+                          num_attention_heads=4, intermediate_size=64,
+                          max_position_embeddings=64, hidden_dropout_prob=0.0,
+                          attention_probs_dropout_prob=0.0)
+    rm = RoFormerModel(rcfg).eval()
+    # rotary encoder: NO absolute position table anywhere
+    assert not any("position" in n for n, _ in rm.named_parameters())
+    # position sensitivity comes from the rotation: swapped tokens differ
+    with torch.no_grad():
+        a, _ = rm(ids)
+        b, _ = rm(ids.flip(dims=[1]))
+    assert not torch.allclose(a[:, 0], b[:, -1], atol=1e-4)
+
+    mlm = RoFormerForMaskedLM(rcfg)
+    labels = ids.clone()
+    labels[:, :5] = -100
+    loss, logits = mlm(ids, labels=labels)
+    loss.backward()
+    assert logits.shape == (2, 10, 120)
