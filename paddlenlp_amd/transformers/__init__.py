@@ -133,3 +133,9 @@ from .roformer import (  # noqa: F401
     RoFormerForSequenceClassification,
     RoFormerModel,
 )
+from .deberta import (  # noqa: F401
+    DebertaConfig,
+    DebertaForMaskedLM,
+    DebertaForSequenceClassification,
+    DebertaModel,
+)

@@ -67,6 +67,10 @@ MODEL_REGISTRY = {
     "albert": {"module": "albert", "config": "AlbertConfig", "base": "AlbertModel",
                "sequence_classification": "AlbertForSequenceClassification",
                "masked_lm": "AlbertForMaskedLM"},
+    "deberta": {"module": "deberta", "config": "DebertaConfig",
+                "base": "DebertaModel",
+                "sequence_classification": "DebertaForSequenceClassification",
+                "masked_lm": "DebertaForMaskedLM"},
     "distilbert": {"module": "distilbert", "config": "DistilBertConfig",
                    "base": "DistilBertModel",
                    "sequence_classification": "DistilBertForSequenceClassification",

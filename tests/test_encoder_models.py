@@ -358,3 +358,35 @@ def test_distilbert_and_roformer():
     loss, logits = mlm(ids, labels=labels)
     loss.backward()
     assert logits.shape == (2, 10, 120)
+
+
+def test_deberta_disentangled_attention():
+    from paddlenlp_amd.transformers import (
+        DebertaConfig,
+        DebertaForSequenceClassification,
+        DebertaModel,
+    )
+
+    cfg = DebertaConfig(vocab_size=120, hidden_size=32, num_hidden_layers=2,
+                        num_attention_heads=4, intermediate_size=64,
+                        max_relative_positions=8, hidden_dropout_prob=0.0,
+                        attention_probs_dropout_prob=0.0, num_labels=2)
+    m = DebertaModel(cfg).eval()
+    # no absolute position table; one shared relative table of 2k rows
+    assert m.rel_embeddings.num_embeddings == 16
+    assert not any("position" in n and "rel" not in n
+                   for n, _ in m.named_parameters())
+    ids = torch.randint(0, 120, (2, 10))
+    seq = m(ids)
+    assert seq.shape == (2, 10, 32)
+    # position-aware: reversing the sequence changes per-token outputs
+    with torch.no_grad():
+        a = m(ids)
+        b = m(ids.flip(dims=[1]))
+    assert not torch.allclose(a[:, 0], b[:, -1], atol=1e-4)
+
+    cls_model = DebertaForSequenceClassification(cfg)
+    loss, logits = cls_model(ids, labels=torch.tensor([0, 1]))
+    loss.backward()
+    assert logits.shape == (2, 2)
+    assert m.rel_embeddings.weight.shape == (16, 32)
