@@ -139,3 +139,9 @@ from .deberta import (  # noqa: F401
     DebertaForSequenceClassification,
     DebertaModel,
 )
+from .mt5 import (  # noqa: F401
+    MT5Config,
+    MT5EncoderModel,
+    MT5ForConditionalGeneration,
+    MT5Model,
+)

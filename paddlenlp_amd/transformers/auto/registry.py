@@ -45,6 +45,8 @@ MODEL_REGISTRY = {
            "seq2seq_lm": "T5ForConditionalGeneration"},
     "bart": {"module": "bart", "config": "BartConfig", "base": "BartModel",
              "seq2seq_lm": "BartForConditionalGeneration"},
+    "mt5": {"module": "mt5", "config": "MT5Config", "base": "MT5Model",
+            "seq2seq_lm": "MT5ForConditionalGeneration"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",

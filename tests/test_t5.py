@@ -182,3 +182,19 @@ def test_t5_beam_search():
     multi, _ = m.generate(src, max_new_tokens=4, num_beams=4,
                           num_return_sequences=2)
     assert multi.shape[0] == 4
+
+
+def test_mt5_defaults():
+    from paddlenlp_amd.transformers import MT5Config, MT5ForConditionalGeneration
+
+    cfg = MT5Config(vocab_size=100, d_model=32, d_kv=8, d_ff=64,
+                    num_layers=2, num_heads=4, dropout_rate=0.0)
+    assert cfg.is_gated_act and not cfg.tie_word_embeddings
+    m = MT5ForConditionalGeneration(cfg)
+    assert hasattr(m.t5.encoder.blocks[0].ff, "wi_0")  # gated
+    assert m.lm_head.weight.data_ptr() != m.t5.shared.weight.data_ptr()
+    src = torch.randint(0, 100, (2, 6))
+    labels = torch.randint(1, 100, (2, 4))
+    loss, logits = m(input_ids=src, labels=labels)
+    loss.backward()
+    assert logits.shape == (2, 4, 100)
