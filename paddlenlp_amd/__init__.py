@@ -6,7 +6,7 @@ PyTorch-ROCm with hand-written CDNA4 (gfx950) HIP kernels for the hot
 path and RCCL collectives over xGMI for 4D parallelism.
 
 Public surface mirrors the reference layer map:
-  - paddlenlp_amd.transformers: model zoo (31 families) + Auto* registries
+  - paddlenlp_amd.transformers: model zoo (32 families) + Auto* registries
   - paddlenlp_amd.trainer:      Trainer / TrainingArguments / callbacks /
                                 unified checkpoint / compression
   - paddlenlp_amd.data:         collators, Megatron-style causal datasets,
