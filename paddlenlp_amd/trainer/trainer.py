@@ -45,6 +45,7 @@ from .integrations import get_reporting_integration_callbacks
 from .plugins.timer import get_timers
 from ..utils.profiler import add_profiler_step
 from .trainer_utils import (
+    DynamicLossScaler,
     TrainerMemoryTracker,
     TrainOutput,
     caculate_llm_flops,
@@ -234,6 +235,7 @@ class Trainer:
         self.is_in_train = True
         mem_tracker = TrainerMemoryTracker(skip=args.skip_memory_metrics)
         mem_tracker.start("train")
+        self._loss_scaler = DynamicLossScaler() if args.fp16 else None
 
         train_dataloader = self.get_train_dataloader()
         steps_per_epoch = max(1, len(train_dataloader) // args.gradient_accumulation_steps)
@@ -441,7 +443,10 @@ class Trainer:
         loss = self.compute_loss(model, inputs)
         if self.args.gradient_accumulation_steps > 1:
             loss = loss / self.args.gradient_accumulation_steps
-        loss.backward()
+        if getattr(self, "_loss_scaler", None) is not None:
+            self._loss_scaler.scale_loss(loss).backward()
+        else:
+            loss.backward()
         return loss.detach()
 
     def optimizer_step(self, model: nn.Module):
@@ -465,6 +470,18 @@ class Trainer:
                     dist.all_reduce(p.grad, group=topo.model_parallel_group)
 
         timers("all-reduce").stop()
+        if getattr(self, "_loss_scaler", None) is not None:
+            finite = self._loss_scaler.unscale_and_check(
+                [p for p in model.parameters() if p.grad is not None])
+            self._loss_scaler.update(found_inf=not finite)
+            if not finite:
+                # overflow: skip this step entirely (reference AMP behavior)
+                logger.warning(
+                    f"fp16 overflow: skipping step, loss scale -> "
+                    f"{self._loss_scaler.scale:g}")
+                self.optimizer.zero_grad(set_to_none=True)
+                self.lr_scheduler.step()
+                return
         if args.max_grad_norm and args.max_grad_norm > 0:
             if self._zero is not None and self._zero.stage == 3:
                 self._zero.clip_grads(args.max_grad_norm)

@@ -256,3 +256,53 @@ def enable_determinism(seed: int = 42):
     os.environ.setdefault("MIOPEN_FIND_MODE", "1")
     os.environ.setdefault("CUBLAS_WORKSPACE_CONFIG", ":4096:8")
     torch.use_deterministic_algorithms(True, warn_only=True)
+
+
+class DynamicLossScaler:
+    """Dynamic fp16 loss scaling (reference paddle AMP O2 scaler semantics):
+    scale the loss up before backward, unscale grads before clipping/step,
+    skip the step and halve the scale on inf/nan, grow after
+    `growth_interval` consecutive good steps."""
+
+    def __init__(self, init_scale: float = 2.0 ** 16, growth_factor: float = 2.0,
+                 backoff_factor: float = 0.5, growth_interval: int = 2000,
+                 min_scale: float = 1.0):
+        self.scale = float(init_scale)
+        self.growth_factor = growth_factor
+        self.backoff_factor = backoff_factor
+        self.growth_interval = growth_interval
+        self.min_scale = min_scale
+        self._good_steps = 0
+
+    def scale_loss(self, loss):
+        return loss * self.scale
+
+    def unscale_and_check(self, params) -> bool:
+        """Divide grads by the scale in place; returns True when all grads
+        are finite (step may proceed)."""
+        inv = 1.0 / self.scale
+        finite = True
+        for p in params:
+            if p.grad is None:
+                continue
+            p.grad.mul_(inv)
+            if finite and not torch.isfinite(p.grad).all():
+                finite = False
+        return finite
+
+    def update(self, found_inf: bool):
+        if found_inf:
+            self.scale = max(self.min_scale, self.scale * self.backoff_factor)
+            self._good_steps = 0
+        else:
+            self._good_steps += 1
+            if self._good_steps >= self.growth_interval:
+                self.scale *= self.growth_factor
+                self._good_steps = 0
+
+    def state_dict(self):
+        return {"scale": self.scale, "good_steps": self._good_steps}
+
+    def load_state_dict(self, sd):
+        self.scale = sd["scale"]
+        self._good_steps = sd["good_steps"]

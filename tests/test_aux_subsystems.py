@@ -155,3 +155,77 @@ def test_import_utils_and_tools():
     assert compare_version("2.10.0", "2.9.1") == 1
     assert compare_version("1.0", "1.0.0") == 0
     assert compare_version("1.2", "1.10") == -1
+
+
+def test_dynamic_loss_scaler():
+    from paddlenlp_amd.trainer.trainer_utils import DynamicLossScaler
+
+    s = DynamicLossScaler(init_scale=4.0, growth_interval=2)
+    p = torch.nn.Parameter(torch.ones(3))
+    loss = (p * 2).sum()
+    s.scale_loss(loss).backward()
+    torch.testing.assert_close(p.grad, torch.full((3,), 8.0))  # 2 * scale 4
+    assert s.unscale_and_check([p])
+    torch.testing.assert_close(p.grad, torch.full((3,), 2.0))
+    s.update(found_inf=False)
+    s.update(found_inf=False)
+    assert s.scale == 8.0  # grew after 2 good steps
+
+    p.grad[0] = float("inf")
+    assert not s.unscale_and_check([p])
+    s.update(found_inf=True)
+    assert s.scale == 4.0  # backed off
+
+    sd = s.state_dict()
+    s2 = DynamicLossScaler()
+    s2.load_state_dict(sd)
+    assert s2.scale == s.scale
+
+
+def test_fp16_trainer_skips_overflow_steps(tmp_path):
+    """An inf grad must skip the optimizer step and shrink the scale,
+    leaving weights untouched."""
+    from paddlenlp_amd.trainer import Trainer
+    from paddlenlp_amd.trainer.trainer_utils import DynamicLossScaler
+    from paddlenlp_amd.trainer.training_args import TrainingArguments
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=2,
+                      num_key_value_heads=2, max_position_embeddings=32,
+                      dtype="float32")
+    model = LlamaForCausalLM.from_config(cfg)
+    data = [{"input_ids": torch.randint(0, 64, (8,)),
+             "labels": torch.randint(0, 64, (8,))} for _ in range(2)]
+    args = TrainingArguments(output_dir=str(tmp_path), do_train=True,
+                             max_steps=1, per_device_train_batch_size=1,
+                             logging_steps=1000, save_steps=1 << 30,
+                             report_to=[])
+    trainer = Trainer(model=model, args=args, train_dataset=data)
+    trainer.train()  # builds optimizer/scheduler (clean run)
+
+    trainer._loss_scaler = DynamicLossScaler(init_scale=8.0)
+    model.zero_grad(set_to_none=True)
+    b = {k: v[None] for k, v in data[0].items()}
+    # scaled backward, then poison one grad with inf
+    trainer.training_step(model, b)
+    next(p for p in model.parameters() if p.grad is not None).grad[...] = \
+        float("inf")
+    w_before = {n: p.detach().clone() for n, p in model.named_parameters()}
+    trainer.optimizer_step(model)
+    assert trainer._loss_scaler.scale == 4.0  # backed off
+    for n, p in model.named_parameters():
+        torch.testing.assert_close(p, w_before[n])  # step skipped
+
+    # clean grads step normally and grow the good-step counter
+    # (the 1-step linear schedule has decayed lr to 0: restore it)
+    for group in trainer.optimizer.param_groups:
+        group["lr"] = 1e-3
+    trainer.lr_scheduler.step = lambda *a, **k: None  # freeze for the check
+    model.zero_grad(set_to_none=True)
+    trainer.training_step(model, b)
+    trainer.optimizer_step(model)
+    changed = any(not torch.equal(p, w_before[n])
+                  for n, p in model.named_parameters())
+    assert changed
