@@ -612,6 +612,9 @@ class Trainer:
 
         if args.process_index == 0:
             torch.save(self.lr_scheduler.state_dict(), os.path.join(ckpt_dir, SCHEDULER_NAME))
+            if getattr(self, "_loss_scaler", None) is not None:
+                torch.save(self._loss_scaler.state_dict(),
+                           os.path.join(ckpt_dir, "scaler.pt"))
             self.state.save_to_json(os.path.join(ckpt_dir, TRAINER_STATE_NAME))
             self._save_rng_state(ckpt_dir)
             self._rotate_checkpoints()
@@ -682,6 +685,10 @@ class Trainer:
             # construction-time LR until the next scheduler.step() otherwise
             for group, lr in zip(self.optimizer.param_groups, self.lr_scheduler.get_last_lr()):
                 group["lr"] = lr
+        scaler_path = os.path.join(ckpt_dir, "scaler.pt")
+        if os.path.isfile(scaler_path) and getattr(self, "_loss_scaler", None) is not None:
+            self._loss_scaler.load_state_dict(
+                torch.load(scaler_path, weights_only=False))
         state_path = os.path.join(ckpt_dir, TRAINER_STATE_NAME)
         if os.path.isfile(state_path):
             self.state = TrainerState.load_from_json(state_path)
