@@ -1,7 +1,14 @@
 from .configuration_utils import LlmMetaConfig, PretrainedConfig  # noqa: F401
 from .model_utils import PretrainedModel, unwrap_model  # noqa: F401
 from .tokenizer_utils import PretrainedTokenizer  # noqa: F401
-from .auto import AutoConfig, AutoModel, AutoModelForCausalLM, AutoTokenizer  # noqa: F401
+from .auto import (  # noqa: F401
+    AutoConfig,
+    AutoModel,
+    AutoModelForCausalLM,
+    AutoModelForConditionalGeneration,
+    AutoModelForSeq2SeqLM,
+    AutoTokenizer,
+)
 from .llama import (  # noqa: F401
     LlamaConfig,
     LlamaForCausalLM,
@@ -67,6 +74,11 @@ from .bart import (  # noqa: F401
     BartConfig,
     BartForConditionalGeneration,
     BartModel,
+)
+from .pegasus import (  # noqa: F401
+    PegasusConfig,
+    PegasusForConditionalGeneration,
+    PegasusModel,
 )
 from .gemma import GemmaConfig, GemmaForCausalLM, GemmaModel  # noqa: F401
 from .opt import OPTConfig, OPTForCausalLM, OPTModel  # noqa: F401

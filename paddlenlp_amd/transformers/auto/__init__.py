@@ -2,7 +2,9 @@ from .configuration import AutoConfig  # noqa: F401
 from .modeling import (  # noqa: F401
     AutoModel,
     AutoModelForCausalLM,
+    AutoModelForConditionalGeneration,
     AutoModelForMaskedLM,
+    AutoModelForSeq2SeqLM,
     AutoModelForQuestionAnswering,
     AutoModelForSequenceClassification,
     AutoModelForTokenClassification,

@@ -25,6 +25,14 @@ class AutoModelForCausalLM(_AutoBase):
     _kind = "causal_lm"
 
 
+class AutoModelForSeq2SeqLM(_AutoBase):
+    _kind = "seq2seq_lm"
+
+
+class AutoModelForConditionalGeneration(AutoModelForSeq2SeqLM):
+    """Reference's name for the seq2seq auto class (auto/modeling.py)."""
+
+
 class AutoModelForSequenceClassification(_AutoBase):
     _kind = "sequence_classification"
 

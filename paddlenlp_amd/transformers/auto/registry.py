@@ -47,6 +47,9 @@ MODEL_REGISTRY = {
              "seq2seq_lm": "BartForConditionalGeneration"},
     "mt5": {"module": "mt5", "config": "MT5Config", "base": "MT5Model",
             "seq2seq_lm": "MT5ForConditionalGeneration"},
+    "pegasus": {"module": "pegasus", "config": "PegasusConfig",
+                "base": "PegasusModel",
+                "seq2seq_lm": "PegasusForConditionalGeneration"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",
