@@ -80,6 +80,22 @@ from .pegasus import (  # noqa: F401
     PegasusForConditionalGeneration,
     PegasusModel,
 )
+from .mbart import (  # noqa: F401
+    MBartConfig,
+    MBartForConditionalGeneration,
+    MBartModel,
+)
+from .qwen import (  # noqa: F401
+    QWenConfig,
+    QWenForCausalLM,
+    QWenLMHeadModel,
+    QWenModel,
+)
+from .codegen import (  # noqa: F401
+    CodeGenConfig,
+    CodeGenForCausalLM,
+    CodeGenModel,
+)
 from .gemma import GemmaConfig, GemmaForCausalLM, GemmaModel  # noqa: F401
 from .opt import OPTConfig, OPTForCausalLM, OPTModel  # noqa: F401
 from .bloom import BloomConfig, BloomForCausalLM, BloomModel  # noqa: F401

@@ -50,6 +50,13 @@ MODEL_REGISTRY = {
     "pegasus": {"module": "pegasus", "config": "PegasusConfig",
                 "base": "PegasusModel",
                 "seq2seq_lm": "PegasusForConditionalGeneration"},
+    "mbart": {"module": "mbart", "config": "MBartConfig",
+              "base": "MBartModel",
+              "seq2seq_lm": "MBartForConditionalGeneration"},
+    "qwen": {"module": "qwen", "config": "QWenConfig",
+             "causal_lm": "QWenForCausalLM", "base": "QWenModel"},
+    "codegen": {"module": "codegen", "config": "CodeGenConfig",
+                "causal_lm": "CodeGenForCausalLM", "base": "CodeGenModel"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",

@@ -60,6 +60,11 @@ TINY = dict(
                     max_position_embeddings=64),
     mamba=dict(vocab_size=96, hidden_size=32, num_hidden_layers=2,
                state_size=8),
+    qwen=dict(vocab_size=96, hidden_size=32, intermediate_size=128,
+              num_hidden_layers=2, num_attention_heads=4, seq_length=32,
+              max_position_embeddings=64),
+    codegen=dict(vocab_size=96, n_embd=32, n_layer=2, n_head=4,
+                 rotary_dim=4, max_position_embeddings=64),
     jamba=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
                num_hidden_layers=4, num_attention_heads=4,
                num_key_value_heads=2, attn_layer_period=4,
