@@ -96,6 +96,31 @@ from .codegen import (  # noqa: F401
     CodeGenForCausalLM,
     CodeGenModel,
 )
+from .tinybert import (  # noqa: F401
+    TinyBertConfig,
+    TinyBertForPretraining,
+    TinyBertForQuestionAnswering,
+    TinyBertForSequenceClassification,
+    TinyBertModel,
+)
+from .ppminilm import (  # noqa: F401
+    PPMiniLMConfig,
+    PPMiniLMForQuestionAnswering,
+    PPMiniLMForSequenceClassification,
+    PPMiniLMModel,
+)
+from .skep import (  # noqa: F401
+    SkepConfig,
+    SkepCrfForTokenClassification,
+    SkepForSequenceClassification,
+    SkepForTokenClassification,
+    SkepModel,
+)
+from .distill_utils import (  # noqa: F401
+    calc_minilm_loss,
+    calc_multi_relation_loss,
+    to_distill,
+)
 from .gemma import GemmaConfig, GemmaForCausalLM, GemmaModel  # noqa: F401
 from .opt import OPTConfig, OPTForCausalLM, OPTModel  # noqa: F401
 from .bloom import BloomConfig, BloomForCausalLM, BloomModel  # noqa: F401

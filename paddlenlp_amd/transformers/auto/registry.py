@@ -57,6 +57,17 @@ MODEL_REGISTRY = {
              "causal_lm": "QWenForCausalLM", "base": "QWenModel"},
     "codegen": {"module": "codegen", "config": "CodeGenConfig",
                 "causal_lm": "CodeGenForCausalLM", "base": "CodeGenModel"},
+    "tinybert": {"module": "tinybert", "config": "TinyBertConfig",
+                 "base": "TinyBertModel",
+                 "sequence_classification": "TinyBertForSequenceClassification",
+                 "question_answering": "TinyBertForQuestionAnswering"},
+    "ppminilm": {"module": "ppminilm", "config": "PPMiniLMConfig",
+                 "base": "PPMiniLMModel",
+                 "sequence_classification": "PPMiniLMForSequenceClassification",
+                 "question_answering": "PPMiniLMForQuestionAnswering"},
+    "skep": {"module": "skep", "config": "SkepConfig", "base": "SkepModel",
+             "sequence_classification": "SkepForSequenceClassification",
+             "token_classification": "SkepForTokenClassification"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",
