@@ -116,6 +116,7 @@ from .skep import (  # noqa: F401
     SkepForTokenClassification,
     SkepModel,
 )
+from .yuan import YuanConfig, YuanForCausalLM, YuanModel  # noqa: F401
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

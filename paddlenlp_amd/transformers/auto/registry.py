@@ -68,6 +68,8 @@ MODEL_REGISTRY = {
     "skep": {"module": "skep", "config": "SkepConfig", "base": "SkepModel",
              "sequence_classification": "SkepForSequenceClassification",
              "token_classification": "SkepForTokenClassification"},
+    "yuan": {"module": "yuan", "config": "YuanConfig",
+             "causal_lm": "YuanForCausalLM", "base": "YuanModel"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",

@@ -78,3 +78,48 @@ def test_codegen_partial_rotary():
     out = _gptj_rope(x, cos, sin, 4)
     torch.testing.assert_close(out[..., 4:], x[..., 4:])
     assert not torch.allclose(out[..., :4], x[..., :4])
+
+
+def test_yuan_lfa_memory_exactness():
+    """Token-by-token decode must match full forward — this exercises the
+    LFA 2-token raw-hidden memory carried in the cache's third slot."""
+    from paddlenlp_amd.transformers import YuanConfig, YuanForCausalLM
+
+    m = YuanForCausalLM(YuanConfig(
+        vocab_size=96, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64)).eval()
+    ids = torch.randint(3, 96, (2, 12))
+    with torch.no_grad():
+        full = m(input_ids=ids)
+        past, outs = None, []
+        for t in range(ids.shape[1]):
+            lg, past = m(input_ids=ids[:, t:t + 1], use_cache=True,
+                         past_key_values=past)
+            outs.append(lg[:, 0])
+    torch.testing.assert_close(torch.stack(outs, 1), full, rtol=2e-4, atol=2e-4)
+    # cache carries (k, v, memory[B,2,H])
+    assert len(past[0]) == 3 and past[0][2].shape == (2, 2, 32)
+
+
+def test_yuan_v_reads_raw_hidden():
+    """V must come from raw hidden states, Q/K from the filtered path."""
+    from paddlenlp_amd.transformers import YuanConfig, YuanForCausalLM
+
+    m = YuanForCausalLM(YuanConfig(
+        vocab_size=96, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=1, num_attention_heads=4, num_key_value_heads=4,
+        max_position_embeddings=64)).eval()
+    attn = m.yuan.layers[0].self_attn
+    x = torch.randn(1, 5, 32)
+    captured = {}
+    orig = attn.v_proj.forward
+
+    def spy(inp):
+        captured["v_in"] = inp
+        return orig(inp)
+
+    attn.v_proj.forward = spy
+    with torch.no_grad():
+        attn(x)
+    torch.testing.assert_close(captured["v_in"], x)

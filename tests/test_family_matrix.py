@@ -65,6 +65,9 @@ TINY = dict(
               max_position_embeddings=64),
     codegen=dict(vocab_size=96, n_embd=32, n_layer=2, n_head=4,
                  rotary_dim=4, max_position_embeddings=64),
+    yuan=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
+              num_hidden_layers=2, num_attention_heads=4,
+              num_key_value_heads=2, max_position_embeddings=64),
     jamba=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
                num_hidden_layers=4, num_attention_heads=4,
                num_key_value_heads=2, attn_layer_period=4,
