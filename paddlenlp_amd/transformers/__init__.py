@@ -117,6 +117,16 @@ from .skep import (  # noqa: F401
     SkepModel,
 )
 from .yuan import YuanConfig, YuanForCausalLM, YuanModel  # noqa: F401
+from .blenderbot import (  # noqa: F401
+    BlenderbotConfig,
+    BlenderbotForConditionalGeneration,
+    BlenderbotModel,
+)
+from .blenderbot_small import (  # noqa: F401
+    BlenderbotSmallConfig,
+    BlenderbotSmallForConditionalGeneration,
+    BlenderbotSmallModel,
+)
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

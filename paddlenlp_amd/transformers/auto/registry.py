@@ -70,6 +70,13 @@ MODEL_REGISTRY = {
              "token_classification": "SkepForTokenClassification"},
     "yuan": {"module": "yuan", "config": "YuanConfig",
              "causal_lm": "YuanForCausalLM", "base": "YuanModel"},
+    "blenderbot": {"module": "blenderbot", "config": "BlenderbotConfig",
+                   "base": "BlenderbotModel",
+                   "seq2seq_lm": "BlenderbotForConditionalGeneration"},
+    "blenderbot_small": {
+        "module": "blenderbot_small", "config": "BlenderbotSmallConfig",
+        "base": "BlenderbotSmallModel",
+        "seq2seq_lm": "BlenderbotSmallForConditionalGeneration"},
     "bert": {"module": "bert", "config": "BertConfig", "base": "BertModel",
              "sequence_classification": "BertForSequenceClassification",
              "token_classification": "BertForTokenClassification",
