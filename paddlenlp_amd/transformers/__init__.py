@@ -127,6 +127,19 @@ from .blenderbot_small import (  # noqa: F401
     BlenderbotSmallForConditionalGeneration,
     BlenderbotSmallModel,
 )
+from .nezha import (  # noqa: F401
+    NeZhaConfig,
+    NeZhaForQuestionAnswering,
+    NeZhaForSequenceClassification,
+    NeZhaForTokenClassification,
+    NeZhaModel,
+)
+from .mpnet import (  # noqa: F401
+    MPNetConfig,
+    MPNetForMaskedLM,
+    MPNetForSequenceClassification,
+    MPNetModel,
+)
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

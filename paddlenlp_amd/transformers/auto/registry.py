@@ -114,6 +114,13 @@ MODEL_REGISTRY = {
     "electra": {"module": "electra", "config": "ElectraConfig", "base": "ElectraModel",
                 "sequence_classification": "ElectraForSequenceClassification",
                 "token_classification": "ElectraForTokenClassification"},
+    "nezha": {"module": "nezha", "config": "NeZhaConfig", "base": "NeZhaModel",
+              "sequence_classification": "NeZhaForSequenceClassification",
+              "token_classification": "NeZhaForTokenClassification",
+              "question_answering": "NeZhaForQuestionAnswering"},
+    "mpnet": {"module": "mpnet", "config": "MPNetConfig", "base": "MPNetModel",
+              "sequence_classification": "MPNetForSequenceClassification",
+              "masked_lm": "MPNetForMaskedLM"},
 }
 
 
