@@ -52,6 +52,37 @@ class _GenerationTask:
         return results[0] if single else results
 
 
+class _Seq2SeqTask:
+    """Seq2seq-backed pipeline (reference: taskflow/text_summarization.py,
+    which runs Pegasus).  Any AutoModelForSeq2SeqLM checkpoint works."""
+
+    def __init__(self, task: str, model_path: str, max_new_tokens: int = 64,
+                 num_beams: int = 4, **kwargs):
+        from ..transformers import AutoModelForSeq2SeqLM
+
+        self.task = task
+        self.tokenizer = AutoTokenizer.from_pretrained(model_path)
+        self.model = AutoModelForSeq2SeqLM.from_pretrained(model_path)
+        if torch.cuda.is_available():
+            self.model = self.model.to("cuda:0")
+        self.model.eval()
+        self.max_new_tokens = max_new_tokens
+        self.num_beams = num_beams
+
+    def __call__(self, inputs):
+        single = isinstance(inputs, str)
+        texts = [inputs] if single else list(inputs)
+        enc = self.tokenizer(texts, padding=True, return_tensors="pt")
+        device = next(self.model.parameters()).device
+        out, _ = self.model.generate(
+            enc["input_ids"].to(device),
+            max_new_tokens=self.max_new_tokens, num_beams=self.num_beams)
+        decoded = self.tokenizer.batch_decode(out, skip_special_tokens=True)
+        return decoded[0] if single else decoded
+
+
+SEQ2SEQ_TASKS = {"text_summarization"}
+
 GENERATION_TASKS = set(_GenerationTask.TEMPLATES)
 
 # encoder-backed pipelines (taskflow/tasks.py): task name -> class
@@ -92,21 +123,24 @@ ENCODER_TASKS = {
 # remaining pipelines need task models that don't exist offline.
 PENDING_TASKS = set()  # every registered pipeline is live
 
-TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | PENDING_TASKS)
+TASKS = sorted(GENERATION_TASKS | set(ENCODER_TASKS) | SEQ2SEQ_TASKS
+               | PENDING_TASKS)
 
 
 class Taskflow:
     def __init__(self, task: str, model: Optional[str] = None, **kwargs):
-        if task in GENERATION_TASKS or task in ENCODER_TASKS:
+        if task in GENERATION_TASKS or task in ENCODER_TASKS \
+                or task in SEQ2SEQ_TASKS:
             if model is None:
                 raise ValueError(
                     f"Taskflow('{task}') needs a local model path via model= "
                     "(no network access in this environment)")
-            cls = ENCODER_TASKS.get(task, _GenerationTask)
-            if cls is _GenerationTask:
-                self.task_instance = _GenerationTask(task, model, **kwargs)
+            if task in SEQ2SEQ_TASKS:
+                self.task_instance = _Seq2SeqTask(task, model, **kwargs)
+            elif task in ENCODER_TASKS:
+                self.task_instance = ENCODER_TASKS[task](model, **kwargs)
             else:
-                self.task_instance = cls(model, **kwargs)
+                self.task_instance = _GenerationTask(task, model, **kwargs)
         elif task in PENDING_TASKS:
             raise NotImplementedError(
                 f"Task '{task}' is registered but its task model is not ported "

@@ -270,3 +270,25 @@ def test_knowledge_mining(tmp_path):
     assert items["alice"]["termid"] == "person_alice_001"
     assert "termid" not in items["paris"]
     assert all(i["wordtag_label"] for i in out["items"])
+
+
+def test_text_summarization_task(tmp_path):
+    from paddlenlp_amd.transformers import (
+        PegasusConfig,
+        PegasusForConditionalGeneration,
+    )
+
+    m = PegasusForConditionalGeneration(PegasusConfig(
+        vocab_size=len(VOCAB), d_model=32, encoder_layers=2,
+        decoder_layers=2, encoder_attention_heads=4,
+        decoder_attention_heads=4, encoder_ffn_dim=64, decoder_ffn_dim=64,
+        max_position_embeddings=64, eos_token_id=1, pad_token_id=0,
+        decoder_start_token_id=0))
+    m.save_pretrained(str(tmp_path))
+    _save_tokenizer(tmp_path)
+    flow = Taskflow("text_summarization", model=str(tmp_path),
+                    max_new_tokens=6, num_beams=2)
+    out = flow("the cat sat on the mat")
+    assert isinstance(out, str)
+    outs = flow(["the cat sat", "the dog ran fast"])
+    assert isinstance(outs, list) and len(outs) == 2
