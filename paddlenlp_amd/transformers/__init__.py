@@ -140,6 +140,31 @@ from .mpnet import (  # noqa: F401
     MPNetForSequenceClassification,
     MPNetModel,
 )
+from .fnet import (  # noqa: F401
+    FNetConfig,
+    FNetForMaskedLM,
+    FNetForSequenceClassification,
+    FNetModel,
+)
+from .ernie_gram import (  # noqa: F401
+    ErnieGramConfig,
+    ErnieGramForQuestionAnswering,
+    ErnieGramForSequenceClassification,
+    ErnieGramForTokenClassification,
+    ErnieGramModel,
+)
+from .rembert import (  # noqa: F401
+    RemBertConfig,
+    RemBertForMaskedLM,
+    RemBertForSequenceClassification,
+    RemBertModel,
+)
+from .xlm import (  # noqa: F401
+    XLMConfig,
+    XLMForSequenceClassification,
+    XLMModel,
+    XLMWithLMHeadModel,
+)
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

@@ -121,6 +121,21 @@ MODEL_REGISTRY = {
     "mpnet": {"module": "mpnet", "config": "MPNetConfig", "base": "MPNetModel",
               "sequence_classification": "MPNetForSequenceClassification",
               "masked_lm": "MPNetForMaskedLM"},
+    "fnet": {"module": "fnet", "config": "FNetConfig", "base": "FNetModel",
+             "sequence_classification": "FNetForSequenceClassification",
+             "masked_lm": "FNetForMaskedLM"},
+    "ernie_gram": {"module": "ernie_gram", "config": "ErnieGramConfig",
+                   "base": "ErnieGramModel",
+                   "sequence_classification": "ErnieGramForSequenceClassification",
+                   "token_classification": "ErnieGramForTokenClassification",
+                   "question_answering": "ErnieGramForQuestionAnswering"},
+    "rembert": {"module": "rembert", "config": "RemBertConfig",
+                "base": "RemBertModel",
+                "sequence_classification": "RemBertForSequenceClassification",
+                "masked_lm": "RemBertForMaskedLM"},
+    "xlm": {"module": "xlm", "config": "XLMConfig", "base": "XLMModel",
+            "sequence_classification": "XLMForSequenceClassification",
+            "masked_lm": "XLMWithLMHeadModel"},
 }
 
 
