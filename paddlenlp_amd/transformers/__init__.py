@@ -165,6 +165,12 @@ from .xlm import (  # noqa: F401
     XLMModel,
     XLMWithLMHeadModel,
 )
+from .unified_transformer import (  # noqa: F401
+    UnifiedTransformerConfig,
+    UnifiedTransformerLMHeadModel,
+    UnifiedTransformerModel,
+)
+from .unimo import UNIMOConfig, UNIMOLMHeadModel, UNIMOModel  # noqa: F401
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

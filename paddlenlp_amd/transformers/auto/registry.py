@@ -136,6 +136,12 @@ MODEL_REGISTRY = {
     "xlm": {"module": "xlm", "config": "XLMConfig", "base": "XLMModel",
             "sequence_classification": "XLMForSequenceClassification",
             "masked_lm": "XLMWithLMHeadModel"},
+    "unified_transformer": {
+        "module": "unified_transformer", "config": "UnifiedTransformerConfig",
+        "base": "UnifiedTransformerModel",
+        "causal_lm": "UnifiedTransformerLMHeadModel"},
+    "unimo": {"module": "unimo", "config": "UNIMOConfig",
+              "base": "UNIMOModel", "causal_lm": "UNIMOLMHeadModel"},
 }
 
 

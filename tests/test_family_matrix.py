@@ -68,6 +68,13 @@ TINY = dict(
     yuan=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
               num_hidden_layers=2, num_attention_heads=4,
               num_key_value_heads=2, max_position_embeddings=64),
+    unified_transformer=dict(vocab_size=96, hidden_size=32,
+                             intermediate_size=64, num_hidden_layers=2,
+                             num_attention_heads=4,
+                             max_position_embeddings=64),
+    unimo=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
+               num_hidden_layers=2, num_attention_heads=4,
+               max_position_embeddings=64),
     jamba=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
                num_hidden_layers=4, num_attention_heads=4,
                num_key_value_heads=2, attn_layer_period=4,
