@@ -171,6 +171,11 @@ from .unified_transformer import (  # noqa: F401
     UnifiedTransformerModel,
 )
 from .unimo import UNIMOConfig, UNIMOLMHeadModel, UNIMOModel  # noqa: F401
+from .chatglm import (  # noqa: F401
+    ChatGLMConfig,
+    ChatGLMForCausalLM,
+    ChatGLMModel,
+)
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

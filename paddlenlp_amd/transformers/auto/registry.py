@@ -142,6 +142,8 @@ MODEL_REGISTRY = {
         "causal_lm": "UnifiedTransformerLMHeadModel"},
     "unimo": {"module": "unimo", "config": "UNIMOConfig",
               "base": "UNIMOModel", "causal_lm": "UNIMOLMHeadModel"},
+    "chatglm": {"module": "chatglm", "config": "ChatGLMConfig",
+                "causal_lm": "ChatGLMForCausalLM", "base": "ChatGLMModel"},
 }
 
 

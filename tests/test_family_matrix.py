@@ -75,6 +75,9 @@ TINY = dict(
     unimo=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
                num_hidden_layers=2, num_attention_heads=4,
                max_position_embeddings=64),
+    chatglm=dict(vocab_size=96, hidden_size=32, num_hidden_layers=2,
+                 num_attention_heads=4, inner_hidden_size=64,
+                 max_sequence_length=64),
     jamba=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
                num_hidden_layers=4, num_attention_heads=4,
                num_key_value_heads=2, attn_layer_period=4,
