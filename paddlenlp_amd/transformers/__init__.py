@@ -176,6 +176,18 @@ from .chatglm import (  # noqa: F401
     ChatGLMForCausalLM,
     ChatGLMModel,
 )
+from .megatronbert import (  # noqa: F401
+    MegatronBertConfig,
+    MegatronBertForQuestionAnswering,
+    MegatronBertForSequenceClassification,
+    MegatronBertModel,
+)
+from .layoutlm import (  # noqa: F401
+    LayoutLMConfig,
+    LayoutLMForSequenceClassification,
+    LayoutLMForTokenClassification,
+    LayoutLMModel,
+)
 from .distill_utils import (  # noqa: F401
     calc_minilm_loss,
     calc_multi_relation_loss,

@@ -144,6 +144,15 @@ MODEL_REGISTRY = {
               "base": "UNIMOModel", "causal_lm": "UNIMOLMHeadModel"},
     "chatglm": {"module": "chatglm", "config": "ChatGLMConfig",
                 "causal_lm": "ChatGLMForCausalLM", "base": "ChatGLMModel"},
+    "megatronbert": {
+        "module": "megatronbert", "config": "MegatronBertConfig",
+        "base": "MegatronBertModel",
+        "sequence_classification": "MegatronBertForSequenceClassification",
+        "question_answering": "MegatronBertForQuestionAnswering"},
+    "layoutlm": {"module": "layoutlm", "config": "LayoutLMConfig",
+                 "base": "LayoutLMModel",
+                 "sequence_classification": "LayoutLMForSequenceClassification",
+                 "token_classification": "LayoutLMForTokenClassification"},
 }
 
 

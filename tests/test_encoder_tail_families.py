@@ -113,3 +113,56 @@ def test_xlm_sinusoidal_positions():
     assert logits.shape == (2, 2)
     # sinusoid table never serialized
     assert all("position_table" not in k for k in m.state_dict())
+
+
+def test_megatronbert_pre_ln():
+    from paddlenlp_amd.transformers import (
+        MegatronBertConfig,
+        MegatronBertForQuestionAnswering,
+        MegatronBertForSequenceClassification,
+        MegatronBertModel,
+    )
+
+    cfg = MegatronBertConfig(**TINY)
+    m = MegatronBertModel(cfg)
+    # pre-LN tower: extra final LayerNorm after the stack
+    assert isinstance(m.ln, torch.nn.LayerNorm)
+    ids = torch.randint(2, 100, (2, 10))
+    seq, pooled = m(ids)
+    assert seq.shape == (2, 10, 32)
+    loss, _ = MegatronBertForSequenceClassification(cfg)(
+        ids, labels=torch.tensor([0, 1]))
+    loss.backward()
+    start, end = MegatronBertForQuestionAnswering(cfg)(ids)
+    assert start.shape == (2, 10)
+
+
+def test_layoutlm_bbox_embeddings():
+    from paddlenlp_amd.transformers import (
+        LayoutLMConfig,
+        LayoutLMForTokenClassification,
+        LayoutLMModel,
+    )
+
+    cfg = LayoutLMConfig(max_2d_position_embeddings=128,
+                         hidden_dropout_prob=0.0,
+                         attention_probs_dropout_prob=0.0, **TINY)
+    m = LayoutLMModel(cfg).eval()
+    ids = torch.randint(2, 100, (2, 10))
+    bbox = torch.randint(0, 100, (2, 10, 4))
+    bbox[..., 2:] = torch.minimum(
+        bbox[..., :2] + torch.randint(1, 20, (2, 10, 2)),
+        torch.tensor(127))
+    with torch.no_grad():
+        a, _ = m(ids, bbox=bbox)
+        b, _ = m(ids)  # no layout
+        bbox2 = bbox.clone()
+        # move one box's right edge (stays >= x0 and < 128)
+        bbox2[0, 0, 2] = 127 if bbox[0, 0, 2] != 127 else 126
+        c, _ = m(ids, bbox=bbox2)
+    assert not torch.allclose(a, b)  # layout matters
+    assert not torch.allclose(a[0], c[0])  # a single bbox matters
+    loss, logits = LayoutLMForTokenClassification(cfg)(
+        ids, bbox=bbox, labels=torch.randint(0, 2, (2, 10)))
+    loss.backward()
+    assert logits.shape == (2, 10, 2)
