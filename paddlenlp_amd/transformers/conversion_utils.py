@@ -109,11 +109,41 @@ def _convert_t5(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
     return out
 
 
+def _unblock_codegen_qkv(w: torch.Tensor, mp_num: int = 4) -> torch.Tensor:
+    """Re-block a CodeGen fused qkv weight into plain [q|k|v] rows.
+
+    HF/reference CodeGen stores qkv_proj as mp_num blocks, each laid out
+    [q_i; v_i; k_i] (codegen/modeling.py:160-164 splits in q, v, k order).
+    Returns the [3h, h] weight ordered [q; k; v] for this framework's
+    straightforward chunk(3).
+    """
+    three_h, h = w.shape
+    part = three_h // mp_num          # rows per block
+    piece = part // 3                 # rows per tensor within a block
+    blocks = w.view(mp_num, 3, piece, h)
+    q = blocks[:, 0].reshape(-1, h)
+    v = blocks[:, 1].reshape(-1, h)
+    k = blocks[:, 2].reshape(-1, h)
+    return torch.cat([q, k, v], dim=0)
+
+
+def _convert_codegen(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
+    out = {}
+    for k, v in sd.items():
+        nk = k
+        if nk.endswith("attn.qkv_proj.weight"):
+            v = _unblock_codegen_qkv(v)
+        out[nk] = v
+    return out
+
+
 def convert_hf_state_dict(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
     """HF names (model.layers.N...) -> framework names + fused projections."""
     model_type = config.model_type
     if model_type == "bert":
         return _convert_bert(sd, config)
+    if model_type == "codegen":
+        return _convert_codegen(sd, config)
     if model_type == "t5":
         converted = _convert_t5(sd, config)
         # T5ForConditionalGeneration nests the stacks under "t5."

@@ -123,3 +123,35 @@ def test_yuan_v_reads_raw_hidden():
     with torch.no_grad():
         attn(x)
     torch.testing.assert_close(captured["v_in"], x)
+
+
+def test_codegen_qkv_conversion_roundtrip():
+    """Blocked (mp_num=4, [q_i; v_i; k_i] per block) -> plain [q; k; v]."""
+    from paddlenlp_amd.transformers.conversion_utils import (
+        _unblock_codegen_qkv,
+        convert_hf_state_dict,
+    )
+
+    h, mp = 32, 4
+    q = torch.randn(h, h)
+    k = torch.randn(h, h)
+    v = torch.randn(h, h)
+    piece = h // mp
+    blocks = []
+    for i in range(mp):
+        blocks += [q[i * piece:(i + 1) * piece],
+                   v[i * piece:(i + 1) * piece],
+                   k[i * piece:(i + 1) * piece]]
+    blocked = torch.cat(blocks, dim=0)
+    torch.testing.assert_close(_unblock_codegen_qkv(blocked),
+                               torch.cat([q, k, v], dim=0))
+
+    cfg = CodeGenConfig(vocab_size=96, n_embd=h, n_layer=1, n_head=4,
+                        rotary_dim=4, max_position_embeddings=64)
+    sd = {"transformer.h.0.attn.qkv_proj.weight": blocked,
+          "transformer.wte.weight": torch.randn(96, h)}
+    out = convert_hf_state_dict(sd, cfg)
+    torch.testing.assert_close(out["transformer.h.0.attn.qkv_proj.weight"],
+                               torch.cat([q, k, v], dim=0))
+    torch.testing.assert_close(out["transformer.wte.weight"],
+                               sd["transformer.wte.weight"])
