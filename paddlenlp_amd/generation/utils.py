@@ -209,7 +209,7 @@ class GenerationMixin:
             for r in range(n_ret):
                 results.append(cands[r][1][prompt_len:])
         maxlen = max(r.shape[0] for r in results)
-        pad_id = gen.pad_token_id or (eos_ids[0] if eos_ids else 0)
+        pad_id = gen.pad_token_id if gen.pad_token_id is not None else (eos_ids[0] if eos_ids else 0)
         out = torch.full((B * n_ret, maxlen), pad_id, dtype=torch.long,
                          device=device)
         for i, r in enumerate(results):
@@ -303,7 +303,7 @@ class GenerationMixin:
             cands.sort(key=lambda x: -x[0])
             results.append(cands[0][1][prompt_len:])
         maxlen = max(r.shape[0] for r in results)
-        pad_id = gen.pad_token_id or (eos_ids[0] if eos_ids else 0)
+        pad_id = gen.pad_token_id if gen.pad_token_id is not None else (eos_ids[0] if eos_ids else 0)
         out = torch.full((B, maxlen), pad_id, dtype=torch.long, device=device)
         for b, r in enumerate(results):
             out[b, : r.shape[0]] = r

@@ -2,214 +2,250 @@
 
 Reference behavior (SURVEY.md §2.2 row "Sharding stage1/2/3"): stage1 shards
 optimizer states (grads all-reduced like DP); stage2 additionally shards
-gradients (reduce to the owning rank only); both broadcast updated params
-from the owner after the step (reference: fleet group_sharded / DygraphShardingOptimizer
-wired at trainer.py:2016-2095).
+gradients; both make updated params visible on every rank after the step
+(reference: fleet group_sharded / DygraphShardingOptimizer "split_param"
+with allgather/broadcast overlap, trainer.py:2016-2095, comm-buffer fusion
+training_args.py:564).
 
-Design here: parameters are greedily partitioned across the sharding group by
-size.  Non-owned params get `grad=None` before optimizer.step() so the
-optimizer only touches the local shard (optimizer states therefore only
-materialize for owned params = the ZeRO memory saving).  After the step the
-owner broadcasts updated param data.
+MI355X design — flat shard-aligned buckets:
 
-xGMI note: gradient reduction uses per-owner flat buckets
-(sharding_comm_buffer_size_MB) so each RCCL call is large enough to hit
-per-link bandwidth; reduce-scatter decomposition of all-reduce is what the
-ring would do internally anyway, and owner-bucketing gives it to us with
-stage-2 memory savings for free.
+  * Parameters are packed (in REVERSE registration order, i.e. backward
+    production order) into dtype-homogeneous buckets of ~bucket_mb.  Within
+    a bucket every param is assigned whole to one rank (greedy size
+    balance); each rank's segment is padded to a common ``shard_elems`` so
+    the bucket buffer is exactly ``[world * shard_elems]``.
+  * ``p.data`` and ``p.grad`` become views into the bucket's persistent
+    param/grad buffers.  Backward accumulates straight into the flat grad
+    buffer — zero-copy.
+  * Gradient exchange is ONE ``dist.reduce_scatter_tensor`` per bucket
+    (in-place: output = the local shard view).  That is the
+    bandwidth-optimal schedule for 7-link xGMI rings: n-1 per-link hops of
+    1/n of the data, half the bytes of the old per-owner reduce+broadcast
+    and no serialized latency chain.
+  * After optimizer.step() (which touched only locally-owned params, whose
+    data views live inside the local shard) params are republished with ONE
+    in-place ``dist.all_gather_into_tensor`` per bucket, launched async
+    across buckets.
+  * With ``enable_overlap_comm()`` each bucket's reduce-scatter launches
+    asynchronously the moment its last gradient accumulates, overlapping
+    the remaining backward compute (mirrors the reference's
+    stage1-v2 comm/compute overlap switches, trainer.py:2083-2095).
+
+Stage 1 vs stage 2 differ only in bookkeeping here: both use
+reduce-scatter (stage 1 gains the same bandwidth win; its "unsharded
+grads" property is not observable through the Trainer, which clips with a
+sharding-group reduction either way).
 """
 from __future__ import annotations
 
-from typing import Dict
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist
 
 from ..utils.log import logger
 
+_ALIGN = 64  # element alignment of each rank shard inside a bucket
+
+
+class _Bucket:
+    __slots__ = ("params", "owner", "shard_elems", "param_buf", "grad_buf",
+                 "offsets", "fired", "work")
+
+    def __init__(self, params, owner, shard_elems, param_buf, grad_buf, offsets):
+        self.params = params            # all params in this bucket
+        self.owner = owner              # param -> group rank
+        self.shard_elems = shard_elems
+        self.param_buf = param_buf      # [world * shard_elems]
+        self.grad_buf = grad_buf        # [world * shard_elems]
+        self.offsets = offsets          # param -> start offset in buf
+        self.fired = 0
+        self.work = None
+
 
 class ZeroShardedEngine:
     def __init__(self, model, optimizer, stage: int, group, bucket_mb: int = 256):
-        assert stage in (1, 2), "stage3 lands in a later milestone"
+        assert stage in (1, 2), "stage3 lives in parallel.zero3"
         self.model = model
         self.optimizer = optimizer
         self.stage = stage
         self.group = group
         self.world = dist.get_world_size(group) if group is not None else 1
         self.rank = dist.get_rank(group) if group is not None else 0
-        self.group_ranks = dist.get_process_group_ranks(group) if group is not None else [0]
-        self.bucket_bytes = bucket_mb * 1024 * 1024
+        self.bucket_bytes = max(bucket_mb, 1) * 1024 * 1024
 
-        # greedy size-balanced partition: owner[param] = rank in group
+        params, seen = [], set()
+        for g in optimizer.param_groups:
+            for p in g["params"]:
+                if p.requires_grad and id(p) not in seen:
+                    seen.add(id(p))
+                    params.append(p)
+        self._param_list = params
+
+        self.buckets: List[_Bucket] = []
         self.owner: Dict[torch.nn.Parameter, int] = {}
-        sizes = [0] * self.world
-        params = [p for g in optimizer.param_groups for p in g["params"] if p.requires_grad]
-        for p in sorted(params, key=lambda p: -p.numel()):
-            r = sizes.index(min(sizes))
-            self.owner[p] = r
-            sizes[r] += p.numel()
+        self._build_buckets(params)
         self.owned_params = [p for p in params if self.owner[p] == self.rank]
+        self._attach_grad_views()
+
+        total = sum(p.numel() for p in params)
+        pad = sum(b.param_buf.numel() for b in self.buckets) - total
         logger.info(
             f"ZeRO stage{stage}: rank {self.rank}/{self.world} owns "
-            f"{sum(p.numel() for p in self.owned_params):,} / {sum(sizes):,} params"
-        )
+            f"{sum(p.numel() for p in self.owned_params):,} / {total:,} params "
+            f"in {len(self.buckets)} buckets (pad {pad:,} elems)")
 
-        # ---- optional comm/compute overlap (enable_overlap_comm) ----
-        # Buckets in REVERSE parameter order (the backward pass produces
-        # grads last-layer-first) so each bucket's reduce launches as soon
-        # as its grads exist, overlapping with the rest of backward.
         self._overlap = False
         self.overlap_active = False
-        self._param_list = params
-        self._buckets = []          # dicts: params, owner_rank, fired
         self._param_bucket = {}
-        self._pending = []          # (work, flat, grads, owner_rank)
         self._hooks = []
+
+    # ------------------------------------------------------------------
+    def _build_buckets(self, params):
+        # reverse order => backward produces a bucket's grads consecutively
+        by_dtype: Dict[torch.dtype, list] = {}
+        for p in reversed(params):
+            by_dtype.setdefault(p.dtype, []).append(p)
+        for dtype, plist in by_dtype.items():
+            cur, size = [], 0
+            for p in plist:
+                cur.append(p)
+                size += p.numel() * p.element_size()
+                if size >= self.bucket_bytes:
+                    self._make_bucket(cur, dtype)
+                    cur, size = [], 0
+            if cur:
+                self._make_bucket(cur, dtype)
+
+    def _make_bucket(self, plist, dtype):
+        # greedy size-balanced whole-param assignment inside the bucket
+        loads = [0] * self.world
+        per_rank = [[] for _ in range(self.world)]
+        owner = {}
+        for p in sorted(plist, key=lambda p: -p.numel()):
+            r = loads.index(min(loads))
+            owner[p] = r
+            per_rank[r].append(p)
+            loads[r] += p.numel()
+        shard = max(loads)
+        shard = (shard + _ALIGN - 1) // _ALIGN * _ALIGN
+        if shard == 0:
+            return
+        device = plist[0].device
+        param_buf = torch.zeros(self.world * shard, dtype=dtype, device=device)
+        grad_buf = torch.zeros(self.world * shard, dtype=dtype, device=device)
+        offsets = {}
+        for r in range(self.world):
+            off = r * shard
+            for p in per_rank[r]:
+                n = p.numel()
+                param_buf[off:off + n].copy_(p.data.reshape(-1))
+                p.data = param_buf[off:off + n].view(p.shape)
+                offsets[p] = off
+                off += n
+        b = _Bucket(plist, owner, shard, param_buf, grad_buf, offsets)
+        self.buckets.append(b)
+        self.owner.update(owner)
+
+    def _grad_view(self, b: _Bucket, p):
+        off = b.offsets[p]
+        return b.grad_buf[off:off + p.numel()].view(p.shape)
+
+    def _attach_grad_views(self, owned_only: bool = False):
+        for b in self.buckets:
+            for p in b.params:
+                if owned_only and b.owner[p] != self.rank:
+                    p.grad = None
+                else:
+                    p.grad = self._grad_view(b, p)
+
+    def zero_grad(self, set_to_none: bool = True):  # signature-compatible
+        """Zero the flat grad buffers and re-attach every grad view.
+
+        Replaces ``optimizer.zero_grad``: with views attached, backward
+        accumulates straight into the bucket buffers (zero extra copies).
+        """
+        for b in self.buckets:
+            b.grad_buf.zero_()
+            b.fired = 0
+            b.work = None
+        self._attach_grad_views()
 
     # ------------------------------------------------------------------
     # comm overlap
     # ------------------------------------------------------------------
     def enable_overlap_comm(self):
-        """Register per-param hooks that reduce each owner-bucket
-        asynchronously the moment its last gradient accumulates (only while
-        `overlap_active`, i.e. during the final micro-batch's backward)."""
+        """Per-param hooks launch each bucket's reduce-scatter the moment
+        its last gradient accumulates (only while ``overlap_active``, i.e.
+        during the final micro-batch's backward)."""
         if self.group is None or self._overlap:
             return
         self._overlap = True
-        per_owner = {r: [] for r in range(self.world)}
-        for p in reversed(self._param_list):
-            per_owner[self.owner[p]].append(p)
-        for owner_rank, plist in per_owner.items():
-            bucket, size = [], 0
-            for p in plist:
-                bucket.append(p)
-                size += p.numel() * p.element_size()
-                if size >= self.bucket_bytes:
-                    self._buckets.append(
-                        {"params": bucket, "owner_rank": owner_rank, "fired": 0})
-                    bucket, size = [], 0
-            if bucket:
-                self._buckets.append(
-                    {"params": bucket, "owner_rank": owner_rank, "fired": 0})
-        for bi, b in enumerate(self._buckets):
-            for p in b["params"]:
+        for bi, b in enumerate(self.buckets):
+            for p in b.params:
                 self._param_bucket[p] = bi
-                self._hooks.append(p.register_post_accumulate_grad_hook(
-                    self._on_grad_ready))
-        logger.info(f"ZeRO: overlap enabled over {len(self._buckets)} buckets")
+                self._hooks.append(
+                    p.register_post_accumulate_grad_hook(self._on_grad_ready))
+        logger.info(f"ZeRO: overlap enabled over {len(self.buckets)} buckets")
 
     def _on_grad_ready(self, p):
         if not self.overlap_active:
             return
-        b = self._buckets[self._param_bucket[p]]
-        b["fired"] += 1
-        if b["fired"] == len(b["params"]):
+        b = self.buckets[self._param_bucket[p]]
+        b.fired += 1
+        if b.fired == len(b.params):
             self._launch_bucket(b)
 
-    def _launch_bucket(self, b):
-        grads = [p.grad for p in b["params"] if p.grad is not None]
-        if not grads:
-            b["fired"] = -1  # mark done
-            return
-        flat = torch._utils._flatten_dense_tensors(grads)
-        if self.stage == 1:
-            work = dist.all_reduce(flat, group=self.group, async_op=True)
-        else:
-            work = dist.reduce(flat, dst=self.group_ranks[b["owner_rank"]],
-                               group=self.group, async_op=True)
-        self._pending.append((work, flat, grads, b["owner_rank"]))
-        b["fired"] = -1
+    def _sync_bucket_grads(self, b: _Bucket):
+        # tolerate externally re-allocated grads (tests call
+        # optimizer.zero_grad(set_to_none=True) directly): copy any grad
+        # that is not our view back into the flat buffer
+        for p in b.params:
+            g = p.grad
+            if g is None:
+                continue
+            off = b.offsets[p]
+            view = b.grad_buf[off:off + p.numel()]
+            if g.data_ptr() != view.data_ptr():
+                view.copy_(g.reshape(-1))
+                p.grad = view.view(p.shape)
 
-    def _finish_overlap(self):
-        # launch any bucket that never completed (params without grads)
-        for b in self._buckets:
-            if b["fired"] >= 0:
-                self._launch_bucket(b)
-        for work, flat, grads, owner_rank in self._pending:
-            work.wait()
-            if self.stage == 1 or owner_rank == self.rank:
-                flat.div_(self.world)
-                for g, out in zip(grads,
-                                  torch._utils._unflatten_dense_tensors(flat, grads)):
-                    g.copy_(out)
-        self._pending.clear()
-        for b in self._buckets:
-            b["fired"] = 0
-        self.overlap_active = False
+    def _launch_bucket(self, b: _Bucket, async_op: bool = True):
+        self._sync_bucket_grads(b)
+        shard = b.grad_buf[self.rank * b.shard_elems:(self.rank + 1) * b.shard_elems]
+        b.work = dist.reduce_scatter_tensor(
+            shard, b.grad_buf, group=self.group, async_op=async_op)
+        b.fired = -1
 
     # ------------------------------------------------------------------
     def reduce_gradients_and_step_pre(self):
-        """Reduce grads over the sharding group, then drop non-owned grads."""
+        """One reduce-scatter per bucket; then only locally-owned grads
+        remain attached (optimizer states materialize only for the local
+        shard = the ZeRO memory saving)."""
         if self.group is None:
             return
-        if self._overlap and (self.overlap_active or self._pending):
-            self._finish_overlap()
-            for p in self.owner:
-                if self.owner[p] != self.rank:
-                    p.grad = None
-            return
-        params = [p for p in self.owner if p.grad is not None]
-        if self.stage == 1:
-            # grads replicated: flat all-reduce (average)
-            from .data_parallel import fused_allreduce_gradients
-
-            fused_allreduce_gradients(params, self.group, bucket_bytes=self.bucket_bytes)
-        else:
-            # stage2: reduce each owner's shard to the owner only
-            for owner_rank in range(self.world):
-                bucket, size = [], 0
-                owner_global = self.group_ranks[owner_rank]
-
-                def flush():
-                    nonlocal bucket, size
-                    if not bucket:
-                        return
-                    flat = torch._utils._flatten_dense_tensors(bucket)
-                    dist.reduce(flat, dst=owner_global, group=self.group)
-                    if owner_rank == self.rank:
-                        flat.div_(self.world)
-                        for g, out in zip(bucket, torch._utils._unflatten_dense_tensors(flat, bucket)):
-                            g.copy_(out)
-                    bucket, size = [], 0
-
-                for p in params:
-                    if self.owner[p] != owner_rank:
-                        continue
-                    bucket.append(p.grad)
-                    size += p.grad.numel() * p.grad.element_size()
-                    if size >= self.bucket_bytes:
-                        flush()
-                flush()
-        # drop non-owned grads so the optimizer (and its states) only touch
-        # the local shard
-        for p in self.owner:
-            if self.owner[p] != self.rank:
-                p.grad = None
+        for b in self.buckets:
+            if b.fired >= 0:            # not yet launched by overlap hooks
+                self._launch_bucket(b)
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+                b.work = None
+            shard = b.grad_buf[self.rank * b.shard_elems:(self.rank + 1) * b.shard_elems]
+            shard.div_(self.world)
+            b.fired = 0
+        self._attach_grad_views(owned_only=True)
+        self.overlap_active = False
 
     def step_post(self):
-        """Broadcast updated params from their owners (flat-bucketed)."""
+        """Republish updated params: one async in-place all-gather per
+        bucket (param views make the result visible with zero copies)."""
         if self.group is None:
             return
-        for owner_rank in range(self.world):
-            owner_global = self.group_ranks[owner_rank]
-            bucket, size = [], 0
-
-            def flush():
-                nonlocal bucket, size
-                if not bucket:
-                    return
-                flat = torch._utils._flatten_dense_tensors(bucket)
-                dist.broadcast(flat, src=owner_global, group=self.group)
-                if owner_rank != self.rank:
-                    for p, out in zip(bucket, torch._utils._unflatten_dense_tensors(flat, bucket)):
-                        p.copy_(out)
-                bucket, size = [], 0
-
-            for p in self.owner:
-                if self.owner[p] != owner_rank:
-                    continue
-                bucket.append(p.data)
-                size += p.numel() * p.element_size()
-                if size >= self.bucket_bytes:
-                    flush()
-            flush()
+        works = []
+        for b in self.buckets:
+            shard = b.param_buf[self.rank * b.shard_elems:(self.rank + 1) * b.shard_elems]
+            works.append(dist.all_gather_into_tensor(
+                b.param_buf, shard, group=self.group, async_op=True))
+        for w in works:
+            w.wait()

@@ -371,6 +371,22 @@ class Trainer:
                     done = True
                     break
             epoch += 1
+            # epoch end is a step boundary (reference semantics): flush a
+            # partially filled gradient-accumulation window instead of
+            # silently carrying grads into the next epoch / dropping the
+            # final micro-batches
+            if accum_count > 0 and not done:
+                if is_pipeline and pipe_buffer:
+                    loss = self.training_pipeline_step(pipe_buffer)
+                    pipe_buffer = []
+                    tr_loss += loss.detach()
+                self.optimizer_step(model)
+                self.state.global_step += 1
+                accum_count = 0
+                self.control = self.callback_handler.on_step_end(args, self.state, self.control)
+                self._maybe_log_save_evaluate(tr_loss, model, start_time)
+                if self.control.should_training_stop or self.state.global_step >= max_steps:
+                    done = True
             # the consumed-samples resume skip applies only to the first
             # (resumed) epoch; later epochs start from sample 0
             sampler = getattr(train_dataloader, "batch_sampler", None)
@@ -473,14 +489,24 @@ class Trainer:
         if getattr(self, "_loss_scaler", None) is not None:
             finite = self._loss_scaler.unscale_and_check(
                 [p for p in model.parameters() if p.grad is not None])
+            # Under TP/PP/sharding ranks hold different gradient shards, so
+            # the overflow verdict must be agreed globally or ranks diverge
+            # into mismatched collectives (reference AMP all-reduces
+            # found_inf across the hybrid groups).
+            if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+                flag = torch.tensor(
+                    0.0 if finite else 1.0,
+                    device=next(model.parameters()).device)
+                dist.all_reduce(flag, op=dist.ReduceOp.MAX)
+                finite = flag.item() == 0.0
             self._loss_scaler.update(found_inf=not finite)
             if not finite:
-                # overflow: skip this step entirely (reference AMP behavior)
+                # overflow: skip this step entirely (reference AMP behavior;
+                # the LR schedule only advances when the optimizer ran)
                 logger.warning(
                     f"fp16 overflow: skipping step, loss scale -> "
                     f"{self._loss_scaler.scale:g}")
-                self.optimizer.zero_grad(set_to_none=True)
-                self.lr_scheduler.step()
+                self._zero_grads()
                 return
         if args.max_grad_norm and args.max_grad_norm > 0:
             if self._zero is not None and self._zero.stage == 3:
@@ -494,7 +520,15 @@ class Trainer:
             self._zero.step_post()
         timers("optimizer-step").stop()
         self.lr_scheduler.step()
-        self.optimizer.zero_grad(set_to_none=True)
+        self._zero_grads()
+
+    def _zero_grads(self):
+        if self._zero is not None and hasattr(self._zero, "zero_grad"):
+            # flat-bucket engine: zero the persistent buffers and re-attach
+            # grad views so backward accumulates in place (zero-copy)
+            self._zero.zero_grad()
+        else:
+            self.optimizer.zero_grad(set_to_none=True)
 
     def _clip_grad_norm(self, model):
         """Global grad-norm clip, TP/PP-aware: local sum-of-squares, reduced
@@ -519,7 +553,9 @@ class Trainer:
         total += replicated_sq
         if topo.pp_degree > 1 and topo.pipe_parallel_group is not None:
             dist.all_reduce(total, group=topo.pipe_parallel_group)
-        if self._zero is not None and self._zero.stage >= 2:
+        if self._zero is not None and self._zero.stage >= 1:
+            # grads are reduce-scattered for stage 1 and 2 alike: each rank
+            # holds only its owned shard at clip time
             dist.all_reduce(total, group=topo.sharding_parallel_group)
         norm = total.sqrt()
         clip = (args.max_grad_norm / (norm + 1e-6)).clamp(max=1.0)

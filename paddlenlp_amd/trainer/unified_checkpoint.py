@@ -42,10 +42,15 @@ def _param_names(model) -> Dict[int, str]:
 
 def save_unified_model(model, output_dir: str, topology) -> None:
     """Write model weights.  tp=1: dp/sharding rank 0 writes everything.
-    tp>1: every mp rank gathers-and-merges its TP-split params to mp rank 0
-    (full merge; per-rank slice writing is a later optimization).
-    pp>1: each stage writes a model-XXXXX-of-N.safetensors shard with base
-    names + a gathered index (reference unified_checkpoint_into_shards :928)."""
+    tp>1: params are TP-merged over the mp group but each merged tensor is
+    assigned round-robin to ONE mp-rank writer, which streams it into its
+    own model-XXXXX-of-N.safetensors shard — no rank ever materializes the
+    full state dict (reference unified_checkpoint_into_shards :928 writes
+    per-rank slices for the same reason; a 70B full merge would be a
+    ~140 GB host allocation on one rank).
+    pp>1: each stage writes its own shard(s) with base names; the index is
+    gathered over the world with EVERY rank participating (non-writers
+    contribute an empty map) so collectives always match."""
     os.makedirs(output_dir, exist_ok=True)
     from ..transformers.model_utils import PretrainedModel
     from ..parallel.pipeline import PipelineModule
@@ -68,37 +73,55 @@ def save_unified_model(model, output_dir: str, topology) -> None:
                 save_file(sd, os.path.join(output_dir, "model.safetensors"), metadata={"format": "pt"})
         return
 
-    # TP merge: gather each sharded param over the mp group
+    from safetensors.torch import save_file
+    from ..utils.env import SAFE_WEIGHTS_INDEX_NAME, SAFE_WEIGHTS_NAME
+
     mp_group = topology.model_parallel_group
     mp_rank = topology.get_rank_in("mp")
-    merged = {}
     actions = {}
     if isinstance(model, PretrainedModel):
         actions = type(model)._get_tensor_parallel_mappings(model.config, is_split=False)
-    for name, p in model.named_parameters():
+
+    mine = {}
+    for idx, (name, p) in enumerate(model.named_parameters()):
+        writer = idx % tp
         if name in actions:
+            # all mp ranks (incl. non-dp0 — they own shard data the writer
+            # needs only when dp0; but keeping every rank in the gather
+            # keeps the collective schedule identical on all ranks)
             shards = [torch.empty_like(p.data) for _ in range(tp)]
             dist.all_gather(shards, p.data.contiguous(), group=mp_group)
-            if mp_rank == 0 and is_dp0:
-                merged[name] = actions[name]([s.cpu() for s in shards])
-        else:
-            if mp_rank == 0 and is_dp0:
-                merged[name] = p.data.cpu()
-    if mp_rank == 0 and is_dp0 and topology.coords.get("pp", 0) == 0 and topology.pp_degree == 1:
-        if isinstance(model, PretrainedModel):
-            model.save_pretrained(output_dir, state_dict=merged)
-        else:
-            from safetensors.torch import save_file
+            if mp_rank == writer and is_dp0:
+                mine[name] = actions[name]([s.cpu() for s in shards]).contiguous()
+        elif mp_rank == writer and is_dp0:
+            mine[name] = p.data.detach().cpu().contiguous()
 
-            save_file(
-                {k: v.contiguous() for k, v in merged.items()},
-                os.path.join(output_dir, "model.safetensors"),
-                metadata={"format": "pt"},
-            )
+    fname = SAFE_WEIGHTS_NAME.replace(
+        ".safetensors", f"-{mp_rank + 1:05d}-of-{tp:05d}.safetensors")
+    local_map = {}
+    if is_dp0:
+        save_file(mine, os.path.join(output_dir, fname), metadata={"format": "pt"})
+        local_map = {k: fname for k in mine}
+        if mp_rank == 0 and isinstance(model, PretrainedModel):
+            model.config.save_pretrained(output_dir)
+
+    if dist.is_initialized():
+        maps = [None] * topology.world_size
+        dist.all_gather_object(maps, local_map)
+    else:
+        maps = [local_map]
+    if topology.rank == 0:
+        weight_map = {}
+        for m in maps:
+            weight_map.update(m or {})
+        with open(os.path.join(output_dir, SAFE_WEIGHTS_INDEX_NAME), "w") as f:
+            json.dump({"metadata": {"total_size": 0}, "weight_map": weight_map}, f, indent=2)
 
 
 def _save_unified_pipe_model(model, output_dir, topology, is_dp0):
-    """Per-pp-stage shards with base param names; TP merged within the stage."""
+    """Per-(pp-stage, mp-rank) shards with base param names; TP merged
+    within the stage but streamed to round-robin writers so no rank holds a
+    whole stage's merged state."""
     from safetensors.torch import save_file
     from ..utils.env import SAFE_WEIGHTS_INDEX_NAME, SAFE_WEIGHTS_NAME
 
@@ -111,24 +134,27 @@ def _save_unified_pipe_model(model, output_dir, topology, is_dp0):
     actions = {}
     if tp > 1 and hasattr(type(model), "_get_tensor_parallel_mappings"):
         actions = type(model)._get_tensor_parallel_mappings(model.config, is_split=False)
-    merged = {}
+    mine = {}
     mp_group = topology.model_parallel_group
-    for name, t in sd.items():
+    for idx, (name, t) in enumerate(sd.items()):
+        writer = idx % tp
         if name in actions and tp > 1:
             shards = [torch.empty_like(t) for _ in range(tp)]
             dist.all_gather(shards, t.contiguous(), group=mp_group)
-            if mp_rank == 0 and is_dp0:
-                merged[name] = actions[name]([x.cpu() for x in shards]).contiguous()
-        elif mp_rank == 0 and is_dp0:
-            merged[name] = t.detach().cpu().contiguous()
+            if mp_rank == writer and is_dp0:
+                mine[name] = actions[name]([x.cpu() for x in shards]).contiguous()
+        elif mp_rank == writer and is_dp0:
+            mine[name] = t.detach().cpu().contiguous()
 
+    n_files = pp_deg * tp
     fname = SAFE_WEIGHTS_NAME.replace(
-        ".safetensors", f"-{pp_rank + 1:05d}-of-{pp_deg:05d}.safetensors")
+        ".safetensors",
+        f"-{pp_rank * tp + mp_rank + 1:05d}-of-{n_files:05d}.safetensors")
     local_map = {}
-    if mp_rank == 0 and is_dp0:
-        save_file(merged, os.path.join(output_dir, fname), metadata={"format": "pt"})
-        local_map = {k: fname for k in merged}
-        if hasattr(model, "config"):
+    if is_dp0:
+        save_file(mine, os.path.join(output_dir, fname), metadata={"format": "pt"})
+        local_map = {k: fname for k in mine}
+        if pp_rank == 0 and mp_rank == 0 and hasattr(model, "config"):
             model.config.save_pretrained(output_dir)
 
     if dist.is_initialized():
@@ -154,12 +180,10 @@ def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=Non
 
     names = _param_names(model)
     # which ranks hold distinct optimizer state: one dp-replica suffices,
-    # but under ZeRO every sharding rank holds a distinct shard.
+    # but under ZeRO every sharding rank holds a distinct shard.  Non-dp0
+    # ranks still take part in the index all_gather_object below (with an
+    # empty index) so the collective schedule matches on every rank.
     is_dp0 = topology.coords.get("dp", 0) == 0
-    if not is_dp0:
-        if dist.is_initialized():
-            dist.barrier()
-        return
 
     shard_coords = (
         topology.coords.get("sharding", 0),
@@ -174,34 +198,36 @@ def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=Non
     )
 
     opt_tensors, master_tensors, meta = {}, {}, {}
-    for group in optimizer.param_groups:
-        for p in group["params"]:
-            state = optimizer.state.get(p)
-            if not state or "exp_avg" not in state:
-                continue
-            name = names.get(id(p))
-            if name is None:
-                continue
-            opt_tensors[f"{name}/{MOMENT1}"] = state["exp_avg"].cpu()
-            opt_tensors[f"{name}/{MOMENT2}"] = state["exp_avg_sq"].cpu()
-            meta[name] = {"step": state.get("step", 0)}
-            if state.get("master") is not None:
-                master_tensors[name] = state["master"].cpu()
+    if is_dp0:
+        for group in optimizer.param_groups:
+            for p in group["params"]:
+                state = optimizer.state.get(p)
+                if not state or "exp_avg" not in state:
+                    continue
+                name = names.get(id(p))
+                if name is None:
+                    continue
+                opt_tensors[f"{name}/{MOMENT1}"] = state["exp_avg"].cpu()
+                opt_tensors[f"{name}/{MOMENT2}"] = state["exp_avg_sq"].cpu()
+                meta[name] = {"step": state.get("step", 0)}
+                if state.get("master") is not None:
+                    master_tensors[name] = state["master"].cpu()
 
-    fname = SAFE_OPTIMIZER_NAME.replace(
-        ".safetensors", f"-{writer_idx + 1:05d}-of-{n_writers:05d}.safetensors"
-    )
-    save_file(opt_tensors, os.path.join(output_dir, fname), metadata={"format": "pt"})
-    mname = None
-    if master_tensors:
-        mname = SAFE_MASTER_WEIGHTS_NAME.replace(
+    fname = mname = None
+    if is_dp0:
+        fname = SAFE_OPTIMIZER_NAME.replace(
             ".safetensors", f"-{writer_idx + 1:05d}-of-{n_writers:05d}.safetensors"
         )
-        save_file(master_tensors, os.path.join(output_dir, mname), metadata={"format": "pt"})
+        save_file(opt_tensors, os.path.join(output_dir, fname), metadata={"format": "pt"})
+        if master_tensors:
+            mname = SAFE_MASTER_WEIGHTS_NAME.replace(
+                ".safetensors", f"-{writer_idx + 1:05d}-of-{n_writers:05d}.safetensors"
+            )
+            save_file(master_tensors, os.path.join(output_dir, mname), metadata={"format": "pt"})
 
-    # gather the global index on rank 0
+    # gather the global index on rank 0 (every rank participates)
     local_index = {
-        "weight_map": {k: fname for k in opt_tensors},
+        "weight_map": {k: fname for k in opt_tensors} if fname else {},
         "master_weight_map": {k: mname for k in master_tensors} if mname else {},
         "steps": meta,
     }
