@@ -161,6 +161,25 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
     return C;
 }
 
+void launch_mfma32_probe(const void*, const void*, float*, hipStream_t);
+void launch_permlane_probe(int*, int*, hipStream_t);
+
+torch::Tensor mfma32_probe(torch::Tensor A, torch::Tensor B) {
+    CHECK_GPU(A); CHECK_BF16(A);
+    auto C = torch::empty({32, 32}, A.options().dtype(torch::kFloat32));
+    launch_mfma32_probe(A.contiguous().data_ptr(), B.contiguous().data_ptr(),
+                        C.data_ptr<float>(), cur_stream());
+    return C;
+}
+
+std::vector<torch::Tensor> permlane_probe() {
+    auto opt = torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA);
+    auto o0 = torch::empty({64}, opt);
+    auto o1 = torch::empty({64}, opt);
+    launch_permlane_probe(o0.data_ptr<int>(), o1.data_ptr<int>(), cur_stream());
+    return {o0, o1};
+}
+
 std::vector<torch::Tensor> flash_attn_fwd_ex(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                              bool causal, int64_t variant) {
     int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
@@ -397,6 +416,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("cross_entropy_fwd", &cross_entropy_fwd);
     m.def("cross_entropy_bwd", &cross_entropy_bwd);
     m.def("mfma_probe", &mfma_probe);
+    m.def("mfma32_probe", &mfma32_probe);
+    m.def("permlane_probe", &permlane_probe);
     m.def("flash_attn_fwd", &flash_attn_fwd);
     m.def("flash_attn_fwd_ex", &flash_attn_fwd_ex);
     m.def("flash_attn_bwd", &flash_attn_bwd);

@@ -998,12 +998,27 @@ void launch_flash_fwd_mask(const void* q, const void* k, const void* v, void* o,
     else if (D == 32) flash_fwd_t<32, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
 }
 
-// variant dispatch for A/B benchmarking (variant: 0=MF1+swz [default],
-// 1=MF2+swz, 2=MF1 linear, 3=MF2 linear)
+// v2 kernel (flash_attn_v2.hip): 8-wave 32x32 swapped-QK^T structure
+bool launch_flash_fwd2(const void* q, const void* k, const void* v, void* o,
+                       float* lse, int B, int Sq, int Skv, int Hq, int Hk,
+                       int D, float scale, bool causal, hipStream_t stream);
+
+// variant dispatch for A/B benchmarking (variant: 0=default [v2 when D=128],
+// 1=MF2+swz, 2=MF1 linear, 3=MF2 linear, 4=GLL, 5=16x16 pipe, 6=v2 forced,
+// 7=v1 MF1 linear explicitly)
 void launch_flash_fwd_variant(const void* q, const void* k, const void* v, void* o,
                               float* lse, int B, int Sq, int Skv, int Hq, int Hk,
                               int D, float scale, bool causal, int variant,
                               hipStream_t stream) {
+    if (variant == 6) {
+        if (launch_flash_fwd2(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, D, scale, causal, stream))
+            return;
+        variant = 0;
+    }
+    if (variant == 0 &&
+        launch_flash_fwd2(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, D, scale, causal, stream))
+        return;
+    if (variant == 7) variant = 2;
     if (D != 128) variant = 0;
     switch (variant) {
         case 5: {

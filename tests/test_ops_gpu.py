@@ -33,6 +33,45 @@ def test_mfma_layout_probe(C):
     assert torch.allclose(out, ref, atol=2e-2, rtol=2e-2), (out - ref).abs().max()
 
 
+def test_mfma32_layout_probe(C):
+    """32x32x16 fragment layouts (the v2 flash kernel builds on these)."""
+    torch.manual_seed(1)
+    A = _bf16(torch.randn(32, 16, device="cuda"))
+    B = _bf16(torch.randn(16, 32, device="cuda"))
+    out = C.mfma32_probe(A, B)
+    ref = A.float() @ B.float()
+    assert torch.allclose(out, ref, atol=2e-2, rtol=2e-2), (out - ref).abs().max()
+
+
+def test_permlane32_swap_semantics(C):
+    """v_permlane32_swap_b32: new_a = {a.lo32, b.lo32}, new_b = {a.hi32, b.hi32}
+    viewed per-lane: l<32: (a[l], a[l+32]); l>=32: (b[l-32], b[l])."""
+    o0, o1 = C.permlane_probe()
+    o0, o1 = o0.cpu(), o1.cpu()
+    for l in range(64):
+        if l < 32:
+            assert o0[l] == l, (l, o0[l])
+            assert o1[l] == l + 32, (l, o1[l])
+        else:
+            assert o0[l] == 1000 + l - 32, (l, o0[l])
+            assert o1[l] == 1000 + l, (l, o1[l])
+
+
+def test_flash_attention_fwd_bench_shape(C):
+    """Forward numerics at the training bench shape (S=4096, GQA 32:8)."""
+    from paddlenlp_amd import ops
+
+    torch.manual_seed(0)
+    B, S, Hq, Hk, D = 1, 4096, 32, 8, 128
+    q = _bf16(torch.randn(B, S, Hq, D, device="cuda"))
+    k = _bf16(torch.randn(B, S, Hk, D, device="cuda"))
+    v = _bf16(torch.randn(B, S, Hk, D, device="cuda"))
+    out = ops.flash_attention(q, k, v, causal=True)
+    ref = ops.reference.flash_attention(q.float(), k.float(), v.float(), causal=True)
+    assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), \
+        (out.float() - ref).abs().max()
+
+
 def test_rms_norm_fwd_bwd(C):
     from paddlenlp_amd import ops
 

@@ -14,7 +14,7 @@ import torch
 
 from paddlenlp_amd.ops.functional import _load_extension
 
-VARIANTS = {0: "default", 2: "MF1+linear", 4: "MF1+gll", 5: "MF1+pipe"}
+VARIANTS = {6: "v2-32x32", 7: "v1-linear", 5: "MF1+pipe"}
 
 
 def main():
@@ -38,11 +38,12 @@ def main():
     flops = 0.5 * 2 * 2 * args.B * args.Hq * args.S * args.S * args.D
 
     # correctness cross-check between variants first
-    ref_o, ref_lse = C.flash_attn_fwd_ex(q, k, v, True, 0)
-    for var in (1, 2, 3):
+    ref_o, ref_lse = C.flash_attn_fwd_ex(q, k, v, True, 7)
+    for var in VARIANTS:
         o, lse = C.flash_attn_fwd_ex(q, k, v, True, var)
         err = (o.float() - ref_o.float()).abs().max().item()
-        assert err < 1e-2, (var, err)
+        print(f"  variant {var} vs v1 max err: {err:.4f}")
+        assert err < 3e-2, (var, err)
 
     results = {v: [] for v in VARIANTS}
     for rep in range(args.reps):
