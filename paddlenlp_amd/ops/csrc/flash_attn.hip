@@ -1091,11 +1091,27 @@ void launch_flash_bwd_mask(const void* dout, const void* q, const void* k, const
     else if (D == 32) flash_bwd_t<32, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
 }
 
+// v2 (flash_attn_bwd_v2.hip): 8-wave 32x32 swapped-operand structure
+bool launch_flash_bwd2(const void* dout, const void* q, const void* k, const void* v,
+                       const void* o, const float* lse, float* delta,
+                       void* dq, void* dk, void* dv,
+                       int B, int Sq, int Skv, int Hq, int Hk, int D,
+                       float scale, bool causal, hipStream_t stream);
+
 void launch_flash_bwd_variant(const void* dout, const void* q, const void* k, const void* v,
                               const void* o, const float* lse, float* delta,
                               void* dq, void* dk, void* dv,
                               int B, int Sq, int Skv, int Hq, int Hk, int D,
                               float scale, bool causal, int variant, hipStream_t stream) {
+    // variant 0 = default (v2 when D=128), 2 = v2 forced, 3 = v1 linear,
+    // 1 = v1 swizzled
+    if (variant == 2 || variant == 0) {
+        if (launch_flash_bwd2(dout, q, k, v, o, lse, delta, dq, dk, dv,
+                              B, Sq, Skv, Hq, Hk, D, scale, causal, stream))
+            return;
+        variant = 0;
+    }
+    if (variant == 3) variant = 0;
     // within-probe A/B: linear 202 TF > swizzled 172 for the bwd pair
     // (their extra LDS tiles make the swizzle's address VALU the bottleneck)
     if (variant == 0 && D == 128) {

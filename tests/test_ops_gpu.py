@@ -156,6 +156,8 @@ def test_flash_attention_fwd(C, B, Sq, Skv, Hq, Hk, D, causal):
     (1, 128, 4, 4, 128),
     (1, 128, 4, 2, 128),
     (1, 100, 2, 2, 128),
+    (1, 300, 8, 2, 128),     # multi-block both grids, ragged
+    (1, 2048, 32, 8, 128),   # bench-shape GQA ratio at depth
 ])
 def test_flash_attention_bwd(C, B, S, Hq, Hk, D):
     from paddlenlp_amd import ops

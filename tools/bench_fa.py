@@ -60,22 +60,27 @@ def main():
         mean = sum(ts) / len(ts)
         print(f"  fwd variant {var} ({name:<11}): {mean*1e3:8.2f} ms  {flops/mean/1e12:7.1f} TF/s")
 
-    # backward variants A/B (0=linear [default], 1=swizzled)
+    # backward variants A/B (2=v2, 3=v1 linear)
     o, lse = C.flash_attn_fwd_ex(q, k, v, True, 0)
     do = torch.randn_like(o)
-    bres = {0: [], 1: []}
+    ref_dq, ref_dk, ref_dv = C.flash_attn_bwd_ex(do, q, k, v, o, lse, True, 3)
+    dq2, dk2, dv2 = C.flash_attn_bwd_ex(do, q, k, v, o, lse, True, 2)
+    for nm, a, b in (("dq", ref_dq, dq2), ("dk", ref_dk, dk2), ("dv", ref_dv, dv2)):
+        err = (a.float() - b.float()).abs().max().item()
+        print(f"  bwd v2 vs v1 {nm} max err: {err:.4f}")
+    bres = {2: [], 3: []}
     for rep in range(args.reps):
-        for var in (0, 1):
+        for var in (2, 3):
             torch.cuda.synchronize()
             t0 = time.perf_counter()
             C.flash_attn_bwd_ex(do, q, k, v, o, lse, True, var)
             torch.cuda.synchronize()
             bres[var].append(time.perf_counter() - t0)
     bwd_flops = flops * 3.5  # 7 matmuls vs fwd's 2
-    for var, name in ((0, "linear"), (1, "swz")):
+    for var, name in ((2, "v2-32x32"), (3, "v1-linear")):
         ts = sorted(bres[var])[1:-1] or bres[var]
         mean = sum(ts) / len(ts)
-        print(f"  bwd variant {var} ({name:<6}): {mean*1e3:8.2f} ms  {bwd_flops/mean/1e12:7.1f} TF/s")
+        print(f"  bwd variant {var} ({name:<9}): {mean*1e3:8.2f} ms  {bwd_flops/mean/1e12:7.1f} TF/s")
 
 
 if __name__ == "__main__":
