@@ -408,12 +408,31 @@ int paged_decode_nsplit(int B, int Hk) {
     return nsplit;
 }
 
+// v2 (paged_attn_v2.hip): MFMA-tiled cooperative-staging decode kernel
+bool launch_paged_decode_attn2(const void* q, const void* k_cache, const void* v_cache,
+                               const float* k_scale, const float* v_scale,
+                               const int* block_table, const int* seq_lens, void* out,
+                               float* partials, int nsplit,
+                               int B, int Hq, int Hk, int D, int block_size,
+                               int max_blocks, float scale, hipStream_t stream);
+
 void launch_paged_decode_attn(const void* q, const void* k_cache, const void* v_cache,
                               const float* k_scale, const float* v_scale,
                               const int* block_table, const int* seq_lens, void* out,
                               float* partials, int nsplit,
                               int B, int Hq, int Hk, int D, int block_size,
                               int max_blocks, float scale, hipStream_t stream) {
+    if (launch_paged_decode_attn2(q, k_cache, v_cache, k_scale, v_scale,
+                                  block_table, seq_lens, out, partials, nsplit,
+                                  B, Hq, Hk, D, block_size, max_blocks, scale, stream)) {
+        if (nsplit > 1) {
+            if (D == 128)
+                hipLaunchKernelGGL(paged_decode_merge_kernel<128>, dim3(B, Hk, Hq / Hk),
+                                   dim3(64), 0, stream, partials, (ushort_t*)out,
+                                   seq_lens, B, Hq, Hk, nsplit);
+        }
+        return;
+    }
     dim3 grid(B, Hk, nsplit);
     const bool c8 = (k_scale != nullptr);
 #define PA_LAUNCH(DD, CC)                                                          \
