@@ -145,8 +145,37 @@ class BlockInferencePredictor(BasePredictor):
         else:
             self.decode_fn = self.engine.decode_step
 
-    def _sample(self, logits, prev_ids=None):
+    def _sample(self, logits, prev_ids=None, prev_lens=None):
         cfg = self.config
+        # fused HIP path (reference token_penalty_multi_scores_v2 +
+        # top_p_sampling_reject): penalties scatter + one histogram-threshold
+        # sampling kernel instead of a 128k-vocab sort + multinomial
+        if logits.is_cuda:
+            try:
+                from paddlenlp_amd.ops.functional import _load_extension
+
+                C = _load_extension()
+                B = logits.size(0)
+                dev = logits.device
+                logits = logits.contiguous().to(torch.bfloat16)
+                if cfg.repetition_penalty != 1.0 and prev_ids is not None and prev_ids.numel():
+                    if prev_lens is None:
+                        prev_lens = torch.full((B,), prev_ids.size(1),
+                                               dtype=torch.int32, device=dev)
+                    C.apply_repetition_penalty(
+                        logits, prev_ids.to(torch.int64),
+                        prev_lens.to(torch.int32),
+                        torch.full((B,), cfg.repetition_penalty,
+                                   dtype=torch.float32, device=dev))
+                greedy = cfg.decode_strategy == "greedy"
+                temp = torch.full((B,), max(cfg.temperature, 1e-6),
+                                  dtype=torch.float32, device=dev)
+                top_p = torch.full((B,), 0.0 if greedy else min(cfg.top_p, 1.0),
+                                   dtype=torch.float32, device=dev)
+                u = torch.rand(B, dtype=torch.float32, device=dev)
+                return C.topp_sample(logits, temp, top_p, u)
+            except Exception:
+                pass
         if cfg.repetition_penalty != 1.0 and prev_ids is not None and prev_ids.numel():
             score = torch.gather(logits, 1, prev_ids)
             score = torch.where(score < 0, score * cfg.repetition_penalty,

@@ -163,6 +163,75 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
 
 void launch_mfma32_probe(const void*, const void*, float*, hipStream_t);
 void launch_permlane_probe(int*, int*, hipStream_t);
+void launch_repetition_penalty(void*, const long long*, const int*, const float*,
+                               int, long long, int, hipStream_t);
+void launch_topp_sample(const void*, const float*, const float*, const float*,
+                        const long long*, int, const int*, const int*,
+                        long long*, int, long long, hipStream_t);
+void launch_decode_update(long long*, signed char*, const signed char*, long long*,
+                          int*, int*, const long long*, int, int*, const int*,
+                          long long, int, int, hipStream_t);
+void launch_block_step(int*, int*, signed char*, signed char*, int*, int*,
+                       signed char*, int, int, int, hipStream_t);
+
+void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pre_ids,
+                              torch::Tensor pre_lens, torch::Tensor rep_pen) {
+    CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+    int B = logits.size(0);
+    long long V = logits.size(1);
+    launch_repetition_penalty(logits.data_ptr(),
+                              reinterpret_cast<const long long*>(pre_ids.data_ptr<int64_t>()),
+                              pre_lens.data_ptr<int>(), rep_pen.data_ptr<float>(),
+                              B, V, pre_ids.size(1), cur_stream());
+}
+
+torch::Tensor topp_sample(torch::Tensor logits, torch::Tensor temperature,
+                          torch::Tensor top_p, torch::Tensor uniform,
+                          c10::optional<torch::Tensor> ban_eos,
+                          c10::optional<torch::Tensor> cur_lens,
+                          c10::optional<torch::Tensor> min_lens) {
+    CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+    int B = logits.size(0);
+    long long V = logits.size(1);
+    auto out = torch::empty({B}, logits.options().dtype(torch::kInt64));
+    launch_topp_sample(
+        logits.data_ptr(), temperature.data_ptr<float>(), top_p.data_ptr<float>(),
+        uniform.data_ptr<float>(),
+        ban_eos ? reinterpret_cast<const long long*>(ban_eos->data_ptr<int64_t>()) : nullptr,
+        ban_eos ? (int)ban_eos->numel() : 0,
+        cur_lens ? cur_lens->data_ptr<int>() : nullptr,
+        min_lens ? min_lens->data_ptr<int>() : nullptr,
+        reinterpret_cast<long long*>(out.data_ptr<int64_t>()), B, V, cur_stream());
+    return out;
+}
+
+void decode_update(torch::Tensor tokens, torch::Tensor stop_flags, torch::Tensor active,
+                   torch::Tensor pre_ids, torch::Tensor pre_lens, torch::Tensor seq_lens,
+                   torch::Tensor eos_ids, torch::Tensor not_need_stop,
+                   c10::optional<torch::Tensor> max_new, int64_t pad_id) {
+    int B = tokens.size(0);
+    launch_decode_update(
+        reinterpret_cast<long long*>(tokens.data_ptr<int64_t>()),
+        stop_flags.data_ptr<signed char>(), active.data_ptr<signed char>(),
+        reinterpret_cast<long long*>(pre_ids.data_ptr<int64_t>()),
+        pre_lens.data_ptr<int>(), seq_lens.data_ptr<int>(),
+        reinterpret_cast<const long long*>(eos_ids.data_ptr<int64_t>()),
+        (int)eos_ids.numel(), not_need_stop.data_ptr<int>(),
+        max_new ? max_new->data_ptr<int>() : nullptr,
+        (long long)pad_id, B, (int)pre_ids.size(1), cur_stream());
+}
+
+void block_step(torch::Tensor block_table, torch::Tensor seq_lens,
+                torch::Tensor stop_flags, torch::Tensor active,
+                torch::Tensor free_list, torch::Tensor free_count,
+                torch::Tensor is_block_step, int64_t block_size) {
+    int B = block_table.size(0);
+    launch_block_step(block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                      stop_flags.data_ptr<signed char>(), active.data_ptr<signed char>(),
+                      free_list.data_ptr<int>(), free_count.data_ptr<int>(),
+                      is_block_step.data_ptr<signed char>(), B, (int)block_size,
+                      (int)block_table.size(1), cur_stream());
+}
 
 torch::Tensor mfma32_probe(torch::Tensor A, torch::Tensor B) {
     CHECK_GPU(A); CHECK_BF16(A);
@@ -418,6 +487,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_probe", &mfma_probe);
     m.def("mfma32_probe", &mfma32_probe);
     m.def("permlane_probe", &permlane_probe);
+    m.def("apply_repetition_penalty", &apply_repetition_penalty);
+    m.def("topp_sample", &topp_sample,
+          py::arg("logits"), py::arg("temperature"), py::arg("top_p"),
+          py::arg("uniform"), py::arg("ban_eos") = py::none(),
+          py::arg("cur_lens") = py::none(), py::arg("min_lens") = py::none());
+    m.def("decode_update", &decode_update,
+          py::arg("tokens"), py::arg("stop_flags"), py::arg("active"),
+          py::arg("pre_ids"), py::arg("pre_lens"), py::arg("seq_lens"),
+          py::arg("eos_ids"), py::arg("not_need_stop"),
+          py::arg("max_new") = py::none(), py::arg("pad_id") = 0);
+    m.def("block_step", &block_step);
     m.def("flash_attn_fwd", &flash_attn_fwd);
     m.def("flash_attn_fwd_ex", &flash_attn_fwd_ex);
     m.def("flash_attn_bwd", &flash_attn_bwd);

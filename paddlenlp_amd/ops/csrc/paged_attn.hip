@@ -337,6 +337,7 @@ __global__ void rope_cache_append_kernel(
         } else {
             int hk = head - Hq;
             int blk = block_table[(long long)b * max_blocks + pos / block_size];
+            if (blk < 0) return;  // inactive slot (device scheduler): no cache write
             const long long tok =
                 ((long long)blk * block_size + pos % block_size) * Hk + hk;
             if (C8) {
@@ -368,6 +369,7 @@ __global__ void rope_cache_append_kernel(
         // V: copy (or quantize) into the cache
         int hv = head - Hq - Hk;
         int blk = block_table[(long long)b * max_blocks + pos / block_size];
+        if (blk < 0) return;  // inactive slot (device scheduler): no cache write
         const long long tok =
             ((long long)blk * block_size + pos % block_size) * Hk + hv;
         constexpr int VE = D / 64;

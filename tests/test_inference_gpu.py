@@ -274,3 +274,161 @@ def test_int8_kv_cache_kernels(C):
     rel = (out.cpu().float() - out_ref.float()).abs().max() \
         / out_ref.float().abs().max()
     assert rel < 0.03, rel
+
+
+# ---------------------------------------------------------------------------
+# fused sampling + device block scheduler (sampling.hip)
+# ---------------------------------------------------------------------------
+def test_topp_sample_greedy_matches_argmax(C):
+    torch.manual_seed(0)
+    B, V = 16, 1000
+    logits = torch.randn(B, V, device="cuda", dtype=torch.bfloat16) * 4
+    temp = torch.ones(B, device="cuda")
+    top_p = torch.zeros(B, device="cuda")     # greedy
+    u = torch.rand(B, device="cuda")
+    out = C.topp_sample(logits, temp, top_p, u)
+    assert torch.equal(out, logits.float().argmax(-1))
+
+
+def test_topp_sample_membership(C):
+    """Sampled tokens always land inside the torch-computed top-p set
+    (allowing the documented 1/64-of-a-bin boundary slack)."""
+    torch.manual_seed(1)
+    B, V = 32, 2000
+    logits = torch.randn(B, V, device="cuda", dtype=torch.bfloat16) * 3
+    temp = torch.full((B,), 0.9, device="cuda")
+    tp = 0.7
+    top_p = torch.full((B,), tp, device="cuda")
+    # torch reference top-p mask (slightly widened for boundary slack)
+    lf = logits.float() / 0.9
+    sorted_l, sorted_i = torch.sort(lf, descending=True)
+    probs = sorted_l.softmax(-1)
+    cum = probs.cumsum(-1)
+    keep_sorted = (cum - probs) <= tp + 0.02
+    keep = torch.zeros_like(keep_sorted).scatter(1, sorted_i, keep_sorted)
+    for trial in range(8):
+        u = torch.rand(B, device="cuda")
+        out = C.topp_sample(logits, temp, top_p, u)
+        ok = keep.gather(1, out.unsqueeze(1)).squeeze(1)
+        assert ok.all(), (trial, out[~ok])
+
+
+def test_topp_sample_min_length_bans_eos(C):
+    torch.manual_seed(2)
+    B, V = 4, 100
+    logits = torch.full((B, V), -5.0, device="cuda", dtype=torch.bfloat16)
+    logits[:, 7] = 10.0          # eos would win
+    logits[:, 3] = 5.0
+    temp = torch.ones(B, device="cuda")
+    top_p = torch.zeros(B, device="cuda")
+    u = torch.rand(B, device="cuda")
+    eos = torch.tensor([7], dtype=torch.int64, device="cuda")
+    cur = torch.zeros(B, dtype=torch.int32, device="cuda")
+    minl = torch.full((B,), 5, dtype=torch.int32, device="cuda")
+    out = C.topp_sample(logits, temp, top_p, u, eos, cur, minl)
+    assert (out == 3).all(), out
+
+
+def test_repetition_penalty_kernel(C):
+    torch.manual_seed(3)
+    B, V, L = 4, 500, 10
+    logits = torch.randn(B, V, device="cuda", dtype=torch.bfloat16)
+    ref = logits.float().clone()
+    pre = torch.randint(0, V, (B, L), device="cuda", dtype=torch.int64)
+    lens = torch.full((B,), L, dtype=torch.int32, device="cuda")
+    rp = torch.full((B,), 1.3, device="cuda")
+    C.apply_repetition_penalty(logits, pre, lens, rp)
+    for b in range(B):
+        for t in set(pre[b].tolist()):
+            l = ref[b, t]
+            ref[b, t] = l * 1.3 if l < 0 else l / 1.3
+    assert torch.allclose(logits.float(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_block_step_kernel_free_alloc_preempt(C):
+    B, bs, mb, nblocks = 4, 16, 8, 10
+    dev = "cuda"
+    bt = torch.full((B, mb), -1, dtype=torch.int32, device=dev)
+    # seq0: stopped with 2 blocks; seq1: active at a block boundary (needs
+    # one); seq2: active mid-block; seq3: inactive
+    bt[0, :2] = torch.tensor([5, 6], dtype=torch.int32)
+    bt[1, :2] = torch.tensor([0, 1], dtype=torch.int32)
+    bt[2, 0] = 2
+    seq_lens = torch.tensor([20, 32, 5, 0], dtype=torch.int32, device=dev)
+    stop = torch.tensor([1, 0, 0, 0], dtype=torch.int8, device=dev)
+    active = torch.tensor([1, 1, 1, 0], dtype=torch.int8, device=dev)
+    free_list = torch.tensor([9, 8, 7], dtype=torch.int32, device=dev)
+    free_list = torch.cat([free_list, torch.zeros(nblocks - 3, dtype=torch.int32, device=dev)])
+    free_count = torch.tensor([3], dtype=torch.int32, device=dev)
+    is_bs = torch.zeros(B, dtype=torch.int8, device=dev)
+    C.block_step(bt, seq_lens, stop, active, free_list, free_count, is_bs, bs)
+    torch.cuda.synchronize()
+    # seq0 freed (2 blocks back), inactive now
+    assert int(active[0]) == 0 and int(seq_lens[0]) == 0
+    assert (bt[0] == -1).all()
+    # seq1 got a new block at position 2
+    assert int(bt[1, 2]) >= 0
+    # free count: 3 + 2 freed - 1 allocated = 4
+    assert int(free_count[0]) == 4
+    assert int(is_bs.sum()) == 0
+
+
+def test_device_decode_loop_matches_host_loop(C):
+    """Greedy device-resident loop == host-managed loop, token for token."""
+    from paddlenlp_amd.experimental import BlockManager, FusedMultiTransformer
+    from paddlenlp_amd.experimental.device_scheduler import DeviceDecodeLoop
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(5)
+    cfg = LlamaConfig(
+        vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=256,
+    )
+    model = LlamaForCausalLM.from_config(cfg, dtype=torch.bfloat16, device="cuda")
+    model.eval()
+    eng = FusedMultiTransformer.from_llama(model, block_size=16, max_seq_len=256).to("cuda")
+    eng.allocate_caches(num_blocks=64, device="cuda")
+    B, T, STEPS = 2, 24, 10
+    ids = torch.randint(3, 512, (B, T), device="cuda")
+    lens = torch.tensor([T] * B, dtype=torch.int32, device="cuda")
+
+    # ---- host loop (bench_infer pattern) ----
+    mgr = BlockManager(64, 16, 16, B)
+    slots = [mgr.allocate_slot(T) for _ in range(B)]
+    bt = torch.stack([mgr.block_table[s] for s in slots]).to("cuda", torch.int32)
+    logits = eng.prefill(ids, bt, lens)
+    # cast to bf16 before argmax on both sides: the device loop samples on
+    # bf16 logits, and fp32-vs-bf16 argmax can flip on sub-eps ties
+    tok = logits.to(torch.bfloat16).float().argmax(-1, keepdim=True)
+    host_tokens = [tok.squeeze(1).cpu().clone()]
+    for _ in range(STEPS - 1):
+        lens_before = torch.tensor([int(mgr.seq_lens[s]) for s in slots],
+                                   dtype=torch.int32, device="cuda")
+        for s in slots:
+            assert mgr.extend(s, 1)
+        bt = torch.stack([mgr.block_table[s] for s in slots]).to("cuda", torch.int32)
+        lg = eng.decode_step(tok, bt, lens_before)
+        tok = lg.to(torch.bfloat16).float().argmax(-1, keepdim=True)
+        host_tokens.append(tok.squeeze(1).cpu().clone())
+
+    # ---- device loop (fresh caches to avoid stale KV) ----
+    eng.allocate_caches(num_blocks=64, device="cuda")
+    loop = DeviceDecodeLoop(eng, max_batch=B, num_blocks=64,
+                            max_blocks_per_seq=16, device="cuda",
+                            eos_ids=[], max_gen_len=64)
+    blocks = [loop.allocate_for_prefill(T) for _ in range(B)]
+    bt2 = torch.stack([torch.cat([b, torch.full((16 - b.numel(),), -1,
+                                                dtype=torch.int32, device="cuda")])
+                       for b in blocks])
+    logits2 = eng.prefill(ids, bt2, lens)
+    first = logits2.to(torch.bfloat16).float().argmax(-1)
+    for i in range(B):
+        loop.add_request(i, T, int(first[i]), top_p=0.0, blocks=blocks[i])
+    loop.decode_steps(STEPS - 1)
+    torch.cuda.synchronize()
+
+    for i in range(B):
+        dev_hist = [int(first[i])] + loop.pre_ids[i, :STEPS - 1].cpu().tolist()
+        host_hist = [int(host_tokens[s][i]) for s in range(STEPS)]
+        assert dev_hist == host_hist, (i, dev_hist, host_hist)

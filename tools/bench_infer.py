@@ -87,6 +87,35 @@ def bench_decode(eng, B, prompt_len, steps, warmup, device, use_graph=False):
     return B * steps / elapsed, elapsed / steps * 1000
 
 
+def bench_device_loop(eng, B, prompt_len, steps, warmup, device):
+    """Zero-host-sync decode: fused sampling + in-kernel block scheduler."""
+    from paddlenlp_amd.experimental.device_scheduler import DeviceDecodeLoop
+
+    c = eng.config
+    nblocks = eng.k_caches[0].shape[0]
+    mbs = (prompt_len + steps + warmup) // c.block_size + 4
+    loop = DeviceDecodeLoop(eng, max_batch=B, num_blocks=nblocks,
+                            max_blocks_per_seq=mbs, device=device,
+                            eos_ids=[], max_gen_len=steps + warmup + 8)
+    ids = torch.randint(3, c.vocab_size, (B, prompt_len), device=device)
+    lens = torch.full((B,), prompt_len, dtype=torch.int32, device=device)
+    blocks = [loop.allocate_for_prefill(prompt_len) for _ in range(B)]
+    bt = torch.stack([torch.cat([b, torch.full((mbs - b.numel(),), -1,
+                                               dtype=torch.int32, device=device)])
+                      for b in blocks])
+    logits = eng.prefill(ids, bt, lens)
+    first = logits.argmax(-1)
+    for i in range(B):
+        loop.add_request(i, prompt_len, int(first[i]), top_p=0.0, blocks=blocks[i])
+    loop.decode_steps(warmup)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    loop.decode_steps(steps)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    return B * steps / elapsed, elapsed / steps * 1000
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="llama3-8b")
@@ -96,6 +125,7 @@ def main():
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--quant", default="")          # "" | fp8 | weight_only_int8
     p.add_argument("--graph", action="store_true")
+    p.add_argument("--device-sched", action="store_true")
     p.add_argument("--cachekv", default="bf16", choices=["bf16", "int8"])
     args = p.parse_args()
 
@@ -117,13 +147,18 @@ def main():
 
     for b in args.batches.split(","):
         B = int(b)
-        tps, ms = bench_decode(eng, B, args.prompt_len, args.steps, args.warmup,
-                               device, use_graph=args.graph)
+        if args.device_sched:
+            tps, ms = bench_device_loop(eng, B, args.prompt_len, args.steps,
+                                        args.warmup, device)
+        else:
+            tps, ms = bench_decode(eng, B, args.prompt_len, args.steps, args.warmup,
+                                   device, use_graph=args.graph)
         print(json.dumps({
             "metric": "decode_tokens_per_sec", "model": args.model, "batch": B,
             "prompt_len": args.prompt_len, "value": round(tps, 1),
             "ms_per_step": round(ms, 2),
             "quant": args.quant or "bf16", "hipgraph": bool(args.graph),
+            "device_sched": bool(args.device_sched),
             "cachekv": args.cachekv,
         }))
 
