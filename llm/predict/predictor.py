@@ -207,7 +207,9 @@ class BlockInferencePredictor(BasePredictor):
         mgr = self.manager
         prev_tokens = {}
 
-        def try_insert():
+        def prepare_inserts():
+            """Tokenize + allocate slots for as many pending prompts as fit.
+            Returns (inserted, batch, lens, bt) or None — no GPU compute."""
             inserted = []
             while pending:
                 req_id, text = pending[0]
@@ -220,32 +222,49 @@ class BlockInferencePredictor(BasePredictor):
                 generated[slot] = 0
                 prev_tokens[slot] = list(ids)
                 inserted.append((slot, ids))
-            if inserted:
-                T = max(len(ids) for _, ids in inserted)
-                batch = torch.zeros(len(inserted), T, dtype=torch.long, device=self.device)
-                lens = torch.zeros(len(inserted), dtype=torch.int32, device=self.device)
-                bt = torch.empty(len(inserted), mgr.max_blocks_per_seq, dtype=torch.int32,
-                                 device=self.device)
-                for i, (slot, ids) in enumerate(inserted):
-                    batch[i, :len(ids)] = torch.tensor(ids, device=self.device)
-                    lens[i] = len(ids)
-                    bt[i] = mgr.block_table[slot].to(self.device)
-                logits = self.engine.prefill(batch, bt, lens)
-                prev = torch.tensor([prev_tokens[s] + [0] * (T - len(prev_tokens[s]))
-                                     for s, _ in inserted], device=self.device, dtype=torch.long)
-                tokens = self._sample(logits, prev)
-                for i, (slot, ids) in enumerate(inserted):
-                    t = int(tokens[i])
-                    self._commit_token(slot, t, results, slot_to_req, generated, prev_tokens, eos)
+            if not inserted:
+                return None
+            T = max(len(ids) for _, ids in inserted)
+            batch = torch.zeros(len(inserted), T, dtype=torch.long, device=self.device)
+            lens = torch.zeros(len(inserted), dtype=torch.int32, device=self.device)
+            bt = torch.empty(len(inserted), mgr.max_blocks_per_seq, dtype=torch.int32,
+                             device=self.device)
+            for i, (slot, ids) in enumerate(inserted):
+                batch[i, :len(ids)] = torch.tensor(ids, device=self.device)
+                lens[i] = len(ids)
+                bt[i] = mgr.block_table[slot].to(self.device)
+            return inserted, batch, lens, bt
+
+        def finish_inserts(ins, logits):
+            inserted, batch, lens, bt = ins
+            T = batch.shape[1]
+            prev = torch.tensor([prev_tokens[s] + [0] * (T - len(prev_tokens[s]))
+                                 for s, _ in inserted], device=self.device, dtype=torch.long)
+            plens = torch.tensor([len(prev_tokens[s]) for s, _ in inserted],
+                                 dtype=torch.int32, device=self.device)
+            tokens = self._sample(logits, prev, plens)
+            for i, (slot, ids) in enumerate(inserted):
+                self._commit_token(slot, int(tokens[i]), results, slot_to_req,
+                                   generated, prev_tokens, eos)
 
         def _active_slots():
             return sorted(slot_to_req.keys())
 
-        try_insert()
         while slot_to_req or pending:
+            ins = prepare_inserts()
             slots = _active_slots()
+            # note: slots admitted by `ins` this round are in slot_to_req but
+            # have no KV yet — exclude them from this round's decode batch
+            if ins:
+                new_slots = {s for s, _ in ins[0]}
+                slots = [s for s in slots if s not in new_slots]
             if not slots:
-                try_insert()
+                if ins:
+                    logits = self.engine.prefill(ins[1], ins[3], ins[2])
+                    finish_inserts(ins, logits)
+                elif pending:
+                    raise RuntimeError(
+                        "cannot admit pending prompt: prompt larger than KV capacity")
                 continue
             B = len(slots)
             input_ids = torch.tensor(
@@ -257,20 +276,27 @@ class BlockInferencePredictor(BasePredictor):
                 if not mgr.extend(s, 1):
                     mgr.preempt_longest()  # (v1: preempted request is dropped back to pending)
             bt = torch.stack([mgr.block_table[s] for s in slots]).to(self.device, torch.int32)
-            logits = self.decode_fn(input_ids, bt, lens_before)
+            if ins:
+                # admit concurrently: the prefill overlaps this decode step
+                # on a side stream (engine.mixed_step) instead of stalling it
+                logits, p_logits = self.engine.mixed_step(
+                    input_ids, bt, lens_before, ins[1], ins[3], ins[2])
+            else:
+                logits = self.decode_fn(input_ids, bt, lens_before)
+                p_logits = None
             maxlen = max(len(prev_tokens[s]) for s in slots)
             prev = torch.zeros(B, maxlen, dtype=torch.long, device=self.device)
+            plens = torch.zeros(B, dtype=torch.int32, device=self.device)
             for i, s in enumerate(slots):
                 prev[i, :len(prev_tokens[s])] = torch.tensor(
                     prev_tokens[s], device=self.device)
-            tokens = self._sample(logits, prev)
-            finished_any = False
+                plens[i] = len(prev_tokens[s])
+            tokens = self._sample(logits, prev, plens)
             for i, s in enumerate(slots):
-                t = int(tokens[i])
-                done = self._commit_token(s, t, results, slot_to_req, generated, prev_tokens, eos)
-                finished_any = finished_any or done
-            if finished_any:
-                try_insert()
+                self._commit_token(s, int(tokens[i]), results, slot_to_req,
+                                   generated, prev_tokens, eos)
+            if p_logits is not None:
+                finish_inserts(ins, p_logits)
 
         return [tok.decode(results[i], skip_special_tokens=True) for i in range(len(texts))]
 

@@ -379,6 +379,36 @@ class FusedMultiTransformer(nn.Module):
         return (last @ self.lm_head.t()).float()
 
 
+    @torch.no_grad()
+    def mixed_step(self, decode_ids, decode_bt, decode_lens,
+                   prefill_ids=None, prefill_bt=None, prefill_lens=None):
+        """One engine call serving a mixed batch: the running decode batch
+        plus a freshly admitted prompt batch.  The reference runs encoder
+        and decoder work on dual streams with event hand-off
+        (csrc/gpu/append_attention.cu:120-160); here the prefill runs on a
+        side HIP stream overlapping the decode pass — weights are
+        read-only and the cache blocks are disjoint, so no ordering is
+        needed beyond the join.  Continuous batching therefore admits new
+        prompts without stalling the decode batch.
+
+        Returns (decode_logits, prefill_logits_or_None)."""
+        if prefill_ids is None:
+            return self.decode_step(decode_ids, decode_bt, decode_lens), None
+        if not decode_ids.is_cuda:
+            d = self.decode_step(decode_ids, decode_bt, decode_lens)
+            p = self.prefill(prefill_ids, prefill_bt, prefill_lens)
+            return d, p
+        if not hasattr(self, "_side_stream") or self._side_stream is None:
+            self._side_stream = torch.cuda.Stream()
+        cur = torch.cuda.current_stream()
+        self._side_stream.wait_stream(cur)
+        with torch.cuda.stream(self._side_stream):
+            p_logits = self.prefill(prefill_ids, prefill_bt, prefill_lens)
+        d_logits = self.decode_step(decode_ids, decode_bt, decode_lens)
+        cur.wait_stream(self._side_stream)
+        return d_logits, p_logits
+
+
 class GraphDecodeRunner:
     """hipGraph-captured decode_step.
 

@@ -432,3 +432,54 @@ def test_device_decode_loop_matches_host_loop(C):
         dev_hist = [int(first[i])] + loop.pre_ids[i, :STEPS - 1].cpu().tolist()
         host_hist = [int(host_tokens[s][i]) for s in range(STEPS)]
         assert dev_hist == host_hist, (i, dev_hist, host_hist)
+
+
+def test_mixed_prefill_decode_step(C):
+    """mixed_step (prefill on a side stream overlapping decode) must produce
+    exactly the same logits as running the two calls sequentially."""
+    from paddlenlp_amd.experimental import BlockManager, FusedMultiTransformer
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(7)
+    cfg = LlamaConfig(
+        vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=256,
+    )
+    model = LlamaForCausalLM.from_config(cfg, dtype=torch.bfloat16, device="cuda")
+    model.eval()
+    eng = FusedMultiTransformer.from_llama(model, block_size=16, max_seq_len=256).to("cuda")
+    eng.allocate_caches(num_blocks=96, device="cuda")
+    mgr = BlockManager(96, 16, 16, 4)
+
+    # running decode batch of 2
+    T = 24
+    ids = torch.randint(3, 512, (2, T), device="cuda")
+    lens = torch.tensor([T, T], dtype=torch.int32, device="cuda")
+    slots = [mgr.allocate_slot(T) for _ in range(2)]
+    bt = torch.stack([mgr.block_table[s] for s in slots]).to("cuda", torch.int32)
+    logits = eng.prefill(ids, bt, lens)
+    tok = logits.argmax(-1, keepdim=True)
+    lens_before = torch.tensor([int(mgr.seq_lens[s]) for s in slots],
+                               dtype=torch.int32, device="cuda")
+    for s in slots:
+        mgr.extend(s, 1)
+    bt = torch.stack([mgr.block_table[s] for s in slots]).to("cuda", torch.int32)
+
+    # incoming prompt batch of 2
+    p_ids = torch.randint(3, 512, (2, 20), device="cuda")
+    p_lens = torch.tensor([20, 15], dtype=torch.int32, device="cuda")
+    p_slots = [mgr.allocate_slot(20) for _ in range(2)]
+    p_bt = torch.stack([mgr.block_table[s] for s in p_slots]).to("cuda", torch.int32)
+
+    d_mix, p_mix = eng.mixed_step(tok, bt, lens_before, p_ids, p_bt, p_lens)
+    torch.cuda.synchronize()
+
+    # sequential reference on fresh caches (same block ids, rebuilt state)
+    eng.allocate_caches(num_blocks=96, device="cuda")
+    eng.prefill(ids, torch.stack([mgr.block_table[s] for s in slots]).to("cuda", torch.int32)[:, :],
+                lens)
+    d_seq = eng.decode_step(tok, bt, lens_before)
+    p_seq = eng.prefill(p_ids, p_bt, p_lens)
+    assert torch.equal(d_mix, d_seq)
+    assert torch.equal(p_mix, p_seq)
