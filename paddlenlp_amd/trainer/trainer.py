@@ -117,6 +117,24 @@ class Trainer:
             seed=self.args.seed,
             consumed_samples=self.state.consumed_samples,
         )
+        if self.args.distributed_dataloader and (
+            self.topology.mp_degree > 1 or self.topology.pp_degree > 1
+            or self.topology.sep_degree > 1
+        ):
+            # only dp-source ranks read; mp/pp/sep peers receive broadcasts
+            # (reference DistDataLoader, data/dist_dataloader.py:41)
+            from ..data.dist_dataloader import DistDataLoader
+
+            loader = DistDataLoader(
+                self.train_dataset,
+                batch_sampler=sampler,
+                collate_fn=self.data_collator,
+                num_workers=self.args.dataloader_num_workers,
+                pin_memory=torch.cuda.is_available(),
+                topology=self.topology,
+            )
+            loader.batch_sampler = sampler  # consumed-samples resume hook
+            return loader
         return DataLoader(
             self.train_dataset,
             batch_sampler=sampler,
@@ -790,6 +808,18 @@ class Trainer:
         model.train()
         logits = torch.cat(all_logits) if all_logits else None
         labels = torch.cat(all_labels) if all_labels else None
+        # merge across data-parallel ranks (reference _nested_gather /
+        # distributed_concat, trainer.py:3302): each dp rank saw a disjoint
+        # dataset shard; every rank returns the full concatenation
+        dp_group = self.topology.data_parallel_group
+        if dist.is_initialized() and dp_group is not None and \
+                dist.get_world_size(dp_group) > 1:
+            gathered = [None] * dist.get_world_size(dp_group)
+            dist.all_gather_object(gathered, (logits, labels), group=dp_group)
+            lg = [g[0] for g in gathered if g[0] is not None]
+            lb = [g[1] for g in gathered if g[1] is not None]
+            logits = torch.cat(lg) if lg else logits
+            labels = torch.cat(lb) if lb else labels
         metrics = {}
         if self.compute_metrics is not None and logits is not None:
             metrics = self.compute_metrics((logits, labels))

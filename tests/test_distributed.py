@@ -560,3 +560,96 @@ def _w_tp2_sharding2(rank, world):
 
 def test_tp2_sharding2_parity():
     _run_workers(_w_tp2_sharding2, world_size=4)
+
+
+def _w_dist_dataloader_trainer(rank, world):
+    """mp=2 with distributed_dataloader: only mp rank 0 reads; training is
+    identical to the plain loader (same data via broadcast)."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.trainer import Trainer, TrainingArguments
+
+    topo = init_parallel_env(mp_degree=world, backend="gloo")
+
+    class DS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            g = torch.Generator().manual_seed(100 + i)
+            return {"x": torch.randn(4, generator=g), "labels": torch.randn(1, generator=g)}
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(7)
+            self.lin = torch.nn.Linear(4, 1)
+
+        def forward(self, x, labels=None):
+            out = self.lin(x)
+            return ((out - labels) ** 2).mean()
+
+    import tempfile
+
+    def run(flag):
+        with tempfile.TemporaryDirectory() as td:
+            args = TrainingArguments(
+                output_dir=td, per_device_train_batch_size=2, max_steps=4,
+                learning_rate=1e-2, distributed_dataloader=flag,
+                logging_steps=100, save_steps=0)
+            model = M()
+            tr = Trainer(model=model, args=args, train_dataset=DS())
+            tr.train()
+            return {n: p.detach().clone() for n, p in model.named_parameters()}
+
+    a = run(True)
+    b = run(False)
+    for n in a:
+        assert torch.allclose(a[n], b[n], atol=1e-7), (n,)
+
+
+def test_dist_dataloader_trainer():
+    _run_workers(_w_dist_dataloader_trainer)
+
+
+def _w_predict_dp_gather(rank, world):
+    """Trainer.predict with dp=2 returns the FULL concatenation on every
+    rank (reference distributed_concat, trainer.py:3302)."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.trainer import Trainer, TrainingArguments
+
+    topo = init_parallel_env(dp_degree=world, backend="gloo")
+
+    class DS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            g = torch.Generator().manual_seed(i)
+            return {"x": torch.randn(4, generator=g), "labels": torch.randn(1, generator=g)}
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(3)
+            self.lin = torch.nn.Linear(4, 1)
+
+        def forward(self, x, labels=None):
+            out = self.lin(x)
+            loss = ((out - labels) ** 2).mean()
+            return loss, out
+
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as td:
+        args = TrainingArguments(
+            output_dir=td, per_device_eval_batch_size=2, max_steps=1,
+            logging_steps=100, save_steps=0)
+        tr = Trainer(model=M(), args=args, train_dataset=DS())
+        logits, labels, _ = tr.predict(DS())
+    # all 8 samples present on every rank
+    assert logits.shape[0] == 8, logits.shape
+    assert labels.shape[0] == 8
+
+
+def test_predict_dp_gather():
+    _run_workers(_w_predict_dp_gather)
