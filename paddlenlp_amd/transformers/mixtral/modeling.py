@@ -18,7 +18,11 @@ from torch.utils.checkpoint import checkpoint
 
 from ... import ops
 from ...generation import GenerationConfig, GenerationMixin
-from ...parallel.expert_parallel import dispatch_and_combine, mark_moe_params_no_sync
+from ...parallel.expert_parallel import (
+    GroupedExperts,
+    dispatch_and_combine,
+    mark_moe_params_no_sync,
+)
 from ...parallel.topology import get_topology
 from ..model_utils import PretrainedModel
 from ..llama.modeling import LlamaAttention, LlamaRMSNorm
@@ -72,12 +76,13 @@ class MixtralSparseMoeBlock(nn.Module):
             self.experts_per_rank = self.num_experts // self.ep_degree
         else:
             self.experts_per_rank = self.num_experts
-        self.experts = nn.ModuleList(
-            [MixtralExpertMLP(config) for _ in range(self.experts_per_rank)]
-        )
+        # stacked weights -> one batched GEMM per projection
+        # (parallel.expert_parallel.GroupedExperts; per-expert state-dict
+        # keys are preserved)
+        self.experts = GroupedExperts(
+            self.experts_per_rank, config.hidden_size, config.intermediate_size)
         if self.ep_degree > 1:
-            for e in self.experts:
-                mark_moe_params_no_sync(e)
+            mark_moe_params_no_sync(self.experts)
 
     def forward(self, hidden):  # [B, S, H]
         B, S, H = hidden.shape
@@ -94,7 +99,7 @@ class MixtralSparseMoeBlock(nn.Module):
         flat_e = topk_e.reshape(-1)                               # [T*k]
         out_flat = dispatch_and_combine(
             flat_x, flat_e, self.num_experts,
-            expert_fn=lambda le, toks: self.experts[le](toks),
+            grouped_fn=self.experts.forward_grouped,
             group=self.ep_group,
         )
         out = (out_flat.reshape(T, self.top_k, H) * topk_w[..., None]).sum(1)

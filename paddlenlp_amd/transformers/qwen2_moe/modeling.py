@@ -15,7 +15,11 @@ from torch.utils.checkpoint import checkpoint
 
 from ... import ops
 from ...generation import GenerationConfig, GenerationMixin
-from ...parallel.expert_parallel import dispatch_and_combine, mark_moe_params_no_sync
+from ...parallel.expert_parallel import (
+    GroupedExperts,
+    dispatch_and_combine,
+    mark_moe_params_no_sync,
+)
 from ...parallel.topology import get_topology
 from ..llama.modeling import LlamaRMSNorm
 from ..mixtral.modeling import load_balancing_loss_func
@@ -63,12 +67,12 @@ class Qwen2MoeSparseMoeBlock(nn.Module):
             self.experts_per_rank = self.num_experts // self.ep_degree
         else:
             self.experts_per_rank = self.num_experts
-        self.experts = nn.ModuleList(
-            [Qwen2MoeMLP(config, config.moe_intermediate_size)
-             for _ in range(self.experts_per_rank)])
+        self.experts = GroupedExperts(
+            self.experts_per_rank, config.hidden_size,
+            config.moe_intermediate_size,
+            names=("gate_proj", "up_proj", "down_proj"))
         if self.ep_degree > 1:
-            for e in self.experts:
-                mark_moe_params_no_sync(e)
+            mark_moe_params_no_sync(self.experts)
 
     def forward(self, hidden):  # [B, S, H]
         B, S, H = hidden.shape
@@ -83,13 +87,20 @@ class Qwen2MoeSparseMoeBlock(nn.Module):
         T = x.shape[0]
         flat_x = x.repeat_interleave(self.top_k, dim=0)
         flat_e = topk_e.reshape(-1)
+        # the shared expert computes while the dispatch all-to-all is in
+        # flight (reference overlaps fused_moe with shared-expert compute)
+        shared_out = {}
+
+        def _shared():
+            shared_out["y"] = self.shared_expert(x) * torch.sigmoid(
+                self.shared_expert_gate(x))
+
         out_flat = dispatch_and_combine(
             flat_x, flat_e, self.num_experts,
-            expert_fn=lambda le, toks: self.experts[le](toks),
-            group=self.ep_group)
+            grouped_fn=self.experts.forward_grouped,
+            group=self.ep_group, overlap_fn=_shared)
         routed = (out_flat.reshape(T, self.top_k, H) * topk_w[..., None]).sum(1)
-        shared = self.shared_expert(x) * torch.sigmoid(self.shared_expert_gate(x))
-        out = routed + shared
+        out = routed + shared_out["y"]
         return out.reshape(B, S, H), router_logits
 
 

@@ -98,8 +98,8 @@ def _w_ep_parity(rank, world):
         for le in range(ep_block.experts_per_rank):
             ge = rank * ep_block.experts_per_rank + le
             for name in ("w1", "w2", "w3"):
-                getattr(ep_block.experts[le], name).weight.copy_(
-                    getattr(full_block.experts[ge], name).weight)
+                getattr(ep_block.experts, name)[le].copy_(
+                    getattr(full_block.experts, name)[ge])
 
     x = torch.randn(2, 8, 64, generator=torch.Generator().manual_seed(7), requires_grad=True)
     out, _ = ep_block(x)
@@ -109,7 +109,7 @@ def _w_ep_parity(rank, world):
     out.sum().backward()
     assert x.grad is not None and torch.isfinite(x.grad).all()
     # expert params are flagged no_sync (dp grad-allreduce exclusion)
-    assert all(getattr(p, "no_sync", False) for p in ep_block.experts[0].parameters())
+    assert all(getattr(p, "no_sync", False) for p in ep_block.experts.parameters())
 
 
 def test_expert_parallel_parity():
@@ -172,3 +172,29 @@ def test_qwen2_moe_cached_decode_parity():
         logits, past = m(input_ids=ids[:, :-1], use_cache=True)
         step, _ = m(input_ids=ids[:, -1:], use_cache=True, past_key_values=past)
     torch.testing.assert_close(step[:, 0], full[:, -1], rtol=1e-3, atol=1e-3)
+
+
+def test_grouped_experts_state_dict_roundtrip():
+    """GroupedExperts keeps per-expert ModuleList-style keys so existing
+    checkpoints and conversion mappings still apply."""
+    from paddlenlp_amd.parallel.expert_parallel import GroupedExperts
+
+    torch.manual_seed(11)
+    ge = GroupedExperts(4, 16, 32)
+    sd = ge.state_dict()
+    assert "0.w1.weight" in sd and "3.w2.weight" in sd
+    assert sd["0.w1.weight"].shape == (32, 16)   # nn.Linear [out, in] layout
+    ge2 = GroupedExperts(4, 16, 32)
+    ge2.load_state_dict(sd)
+    for n in ("w1", "w2", "w3"):
+        assert torch.equal(getattr(ge, n), getattr(ge2, n))
+    # grouped forward == per-expert loop
+    x = torch.randn(10, 16)
+    counts = torch.tensor([3, 0, 5, 2])
+    out = ge.forward_grouped(x, counts)
+    start = 0
+    for e, n in enumerate(counts.tolist()):
+        if n:
+            ref = ge[e](x[start:start + n])
+            assert torch.allclose(out[start:start + n], ref, atol=1e-5)
+        start += n
