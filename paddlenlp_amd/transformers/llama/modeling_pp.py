@@ -33,7 +33,16 @@ class EmbeddingPipe(nn.Module):
         return self.embed_tokens.weight
 
     def forward(self, input_ids):
-        return self.embed_tokens(input_ids)
+        h = self.embed_tokens(input_ids)
+        if getattr(self.config, "sequence_parallel", False) and \
+                self.config.tensor_parallel_degree > 1:
+            # shard activations on the sequence dim for the whole stack
+            # (reference modeling_pp first-stage ScatterOp; the shape-
+            # negotiated p2p carries the [B, s/mp, H] layout across stages)
+            from ...parallel.sequence_parallel import ScatterOp
+
+            h = ScatterOp(h)
+        return h
 
 
 class RMSNormPipe(LlamaRMSNorm):
@@ -53,6 +62,13 @@ class LMHeadPipe(nn.Module):
             self.lm_head = _Linear(config.hidden_size, config.vocab_size, bias=False)
 
     def forward(self, hidden):
+        if getattr(self.config, "sequence_parallel", False) and \
+                self.config.tensor_parallel_degree > 1:
+            # re-assemble the full sequence before the vocab projection
+            # (reference llama/modeling.py:1896 GatherOp)
+            from ...parallel.sequence_parallel import GatherOp
+
+            hidden = GatherOp(hidden)
         return self.lm_head(hidden)
 
 
