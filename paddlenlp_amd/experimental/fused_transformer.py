@@ -41,6 +41,9 @@ class FusedMultiTransformerConfig:
     dtype: torch.dtype = torch.bfloat16
     # explicit head_dim for TP shards (local num_heads != hidden/head_dim)
     head_dim_override: int = 0
+    qkv_bias: bool = False          # qwen2-style attention bias
+    moe_num_experts: int = 0        # >0: FFN is a routed MoE (mixtral)
+    moe_top_k: int = 2
 
     @property
     def head_dim(self):
@@ -140,8 +143,19 @@ class FusedMultiTransformer(nn.Module):
         self.qkv_weights = nn.ParameterList([mk(qkv_out, h) for _ in range(L)])
         self.out_proj_weights = nn.ParameterList([mk(h, c.num_heads * hd) for _ in range(L)])
         self.ffn_ln_scales = nn.ParameterList([mk(h) for _ in range(L)])
-        self.gate_up_weights = nn.ParameterList([mk(2 * c.intermediate_size, h) for _ in range(L)])
-        self.down_weights = nn.ParameterList([mk(h, c.intermediate_size) for _ in range(L)])
+        if c.qkv_bias:
+            self.qkv_biases = nn.ParameterList([mk(qkv_out) for _ in range(L)])
+        if c.moe_num_experts > 0:
+            E, I = c.moe_num_experts, c.intermediate_size
+            self.moe_gates = nn.ParameterList([mk(E, h) for _ in range(L)])
+            self.moe_w1 = nn.ParameterList([mk(E, h, I) for _ in range(L)])
+            self.moe_w3 = nn.ParameterList([mk(E, h, I) for _ in range(L)])
+            self.moe_w2 = nn.ParameterList([mk(E, I, h) for _ in range(L)])
+            self.gate_up_weights = nn.ParameterList([])
+            self.down_weights = nn.ParameterList([])
+        else:
+            self.gate_up_weights = nn.ParameterList([mk(2 * c.intermediate_size, h) for _ in range(L)])
+            self.down_weights = nn.ParameterList([mk(h, c.intermediate_size) for _ in range(L)])
         self.embed_tokens = mk(c.vocab_size, h)
         self.final_norm = mk(h)
         self.lm_head = mk(c.vocab_size, h)
@@ -216,6 +230,105 @@ class FusedMultiTransformer(nn.Module):
         eng.lm_head.copy_(model.lm_head.weight)
         return eng
 
+    @classmethod
+    @torch.no_grad()
+    def from_qwen2(cls, model, block_size: int = 64, max_seq_len: int = 8192):
+        """Import a Qwen2ForCausalLM: llama layout + attention bias
+        (reference experimental/transformers/qwen2/modeling.py)."""
+        mc = model.config
+        Hq, Hk = mc.num_attention_heads, mc.num_key_value_heads
+        I = mc.intermediate_size
+        hd = mc.hidden_size // Hq
+        has_bias = getattr(mc, "attention_bias", True)
+        cfg = FusedMultiTransformerConfig(
+            hidden_size=mc.hidden_size, num_heads=Hq, num_kv_heads=Hk,
+            intermediate_size=I, num_layers=mc.num_hidden_layers,
+            vocab_size=mc.vocab_size, rms_norm_eps=mc.rms_norm_eps,
+            rope_theta=mc.rope_theta, block_size=block_size,
+            max_seq_len=max_seq_len, dtype=next(model.parameters()).dtype,
+            head_dim_override=hd, qkv_bias=has_bias,
+        )
+        eng = cls(cfg)
+        base = model.qwen2
+        for i, layer in enumerate(base.layers):
+            eng.ln_scales[i].copy_(layer.input_layernorm.weight)
+            a = layer.self_attn
+            if hasattr(a, "qkv_proj"):
+                w = a.qkv_proj.weight
+                qw, kw, vw = w[:Hq * hd], w[Hq * hd:(Hq + Hk) * hd], w[(Hq + Hk) * hd:]
+                if has_bias:
+                    b = a.qkv_proj.bias
+                    eng.qkv_biases[i].copy_(b)
+            else:
+                qw, kw, vw = a.q_proj.weight, a.k_proj.weight, a.v_proj.weight
+                if has_bias:
+                    eng.qkv_biases[i].copy_(torch.cat(
+                        [a.q_proj.bias, a.k_proj.bias, a.v_proj.bias]))
+            eng.qkv_weights[i].copy_(torch.cat([qw, kw, vw], dim=0))
+            eng.out_proj_weights[i].copy_(a.o_proj.weight)
+            eng.ffn_ln_scales[i].copy_(layer.post_attention_layernorm.weight)
+            m = layer.mlp
+            if hasattr(m, "gate_up_fused_proj"):
+                eng.gate_up_weights[i].copy_(m.gate_up_fused_proj.weight)
+            else:
+                eng.gate_up_weights[i].copy_(torch.cat(
+                    [m.gate_proj.weight, m.up_proj.weight], dim=0))
+            eng.down_weights[i].copy_(m.down_proj.weight)
+        eng.embed_tokens.copy_(base.embed_tokens.weight)
+        eng.final_norm.copy_(base.norm.weight)
+        eng.lm_head.copy_(model.lm_head.weight)
+        return eng
+
+    @classmethod
+    @torch.no_grad()
+    def from_mixtral(cls, model, block_size: int = 64, max_seq_len: int = 8192):
+        """Import a MixtralForCausalLM: llama attention + routed-MoE FFN
+        (reference experimental/transformers/mixtral/modeling.py)."""
+        mc = model.config
+        Hq, Hk = mc.num_attention_heads, mc.num_key_value_heads
+        hd = mc.hidden_size // Hq
+        cfg = FusedMultiTransformerConfig(
+            hidden_size=mc.hidden_size, num_heads=Hq, num_kv_heads=Hk,
+            intermediate_size=mc.intermediate_size,
+            num_layers=mc.num_hidden_layers, vocab_size=mc.vocab_size,
+            rms_norm_eps=mc.rms_norm_eps, rope_theta=mc.rope_theta,
+            block_size=block_size, max_seq_len=max_seq_len,
+            dtype=next(model.parameters()).dtype, head_dim_override=hd,
+            moe_num_experts=mc.num_local_experts,
+            moe_top_k=mc.num_experts_per_tok,
+        )
+        eng = cls(cfg)
+        base = model.mixtral
+        for i, layer in enumerate(base.layers):
+            eng.ln_scales[i].copy_(layer.input_layernorm.weight)
+            a = layer.self_attn
+            if hasattr(a, "qkv_proj"):
+                eng.qkv_weights[i].copy_(a.qkv_proj.weight)
+            else:
+                eng.qkv_weights[i].copy_(torch.cat(
+                    [a.q_proj.weight, a.k_proj.weight, a.v_proj.weight], dim=0))
+            eng.out_proj_weights[i].copy_(a.o_proj.weight)
+            eng.ffn_ln_scales[i].copy_(layer.post_attention_layernorm.weight)
+            moe = layer.block_sparse_moe
+            eng.moe_gates[i].copy_(moe.gate.weight)
+            eng.moe_w1[i].copy_(moe.experts.w1)
+            eng.moe_w3[i].copy_(moe.experts.w3)
+            eng.moe_w2[i].copy_(moe.experts.w2)
+        eng.embed_tokens.copy_(base.embed_tokens.weight)
+        eng.final_norm.copy_(base.norm.weight)
+        eng.lm_head.copy_(model.lm_head.weight)
+        return eng
+
+    @classmethod
+    def from_model(cls, model, **kw):
+        """Generic importer: dispatch on the model family."""
+        name = type(model).__name__
+        if "Qwen2" in name:
+            return cls.from_qwen2(model, **kw)
+        if "Mixtral" in name:
+            return cls.from_mixtral(model, **kw)
+        return cls.from_llama(model, **kw)
+
     def _tp_reduce(self, t):
         if self.tp_group is not None:
             import torch.distributed as dist
@@ -251,6 +364,45 @@ class FusedMultiTransformer(nn.Module):
         q, sc = self._qw[name][i]
         algo = self.quant_algo if self.quant_algo == "fp8" else "weight_only_int8"
         return weight_only_linear(x, q, sc, None, algo)
+
+    def _moe_ffn(self, h, i):
+        """Routed MoE FFN (mixtral): top-k router + capacity-padded batched
+        GEMMs over the stacked expert weights (same grouped-GEMM scheme as
+        parallel.expert_parallel.GroupedExperts; reference fused_moe,
+        fused_transformer_layers.py:951-1008)."""
+        c = self.config
+        shp = h.shape
+        x = h.reshape(-1, shp[-1])
+        T = x.shape[0]
+        E, K = c.moe_num_experts, c.moe_top_k
+        router = x @ self.moe_gates[i].t()
+        probs = router.float().softmax(-1)
+        topw, tope = probs.topk(K, dim=-1)
+        topw = (topw / topw.sum(-1, keepdim=True)).to(x.dtype)
+        flat_x = x.repeat_interleave(K, dim=0)
+        flat_e = tope.reshape(-1)
+        sort_idx = torch.argsort(flat_e, stable=True)
+        sorted_x = flat_x[sort_idx]
+        counts = torch.bincount(flat_e, minlength=E)
+        C = int(counts.max().item())
+        out_sorted = torch.zeros_like(sorted_x)
+        if C > 0:
+            dev = x.device
+            offs = torch.cumsum(counts, 0) - counts
+            tok_e = torch.repeat_interleave(torch.arange(E, device=dev), counts)
+            pos = torch.arange(sorted_x.shape[0], device=dev) - offs[tok_e]
+            idx = tok_e * C + pos
+            xp = sorted_x.new_zeros(E * C, shp[-1]).index_copy(0, idx, sorted_x)
+            xp = xp.view(E, C, shp[-1])
+            gu = torch.cat([torch.bmm(xp, self.moe_w1[i]),
+                            torch.bmm(xp, self.moe_w3[i])], dim=-1)
+            act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
+            yp = torch.bmm(act, self.moe_w2[i])
+            out_sorted = yp.reshape(E * C, shp[-1]).index_select(0, idx)
+        out_flat = torch.empty_like(out_sorted)
+        out_flat[sort_idx] = out_sorted
+        out = (out_flat.reshape(T, K, shp[-1]) * topw[..., None]).sum(1)
+        return out.reshape(shp)
 
     def allocate_caches(self, num_blocks: int, device, cachekv_dtype: str = "bf16"):
         """cachekv_dtype="int8" halves KV memory: caches store int8 with one
@@ -330,13 +482,18 @@ class FusedMultiTransformer(nn.Module):
         for i in range(c.num_layers):
             h = self._rms(x, self.ln_scales[i])
             qkv = self._mm(h, "qkv_weights", i)          # [B, 1, qkv_out]
+            if c.qkv_bias:
+                qkv = qkv + self.qkv_biases[i]
             q = self._rope_append(i, qkv, block_table, seq_lens_before)
             attn = self._paged_attn(i, q[:, 0], block_table, seq_lens_after)
             x = x + self._tp_reduce(self._mm(attn.reshape(B, 1, -1), "out_proj_weights", i))
             h = self._rms(x, self.ffn_ln_scales[i])
-            gu = self._mm(h, "gate_up_weights", i)
-            act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
-            x = x + self._tp_reduce(self._mm(act, "down_weights", i))
+            if c.moe_num_experts > 0:
+                x = x + self._tp_reduce(self._moe_ffn(h, i))
+            else:
+                gu = self._mm(h, "gate_up_weights", i)
+                act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
+                x = x + self._tp_reduce(self._mm(act, "down_weights", i))
         x = self._rms(x, self.final_norm)
         logits = x[:, 0] @ self.lm_head.t()
         return logits.float()
@@ -355,6 +512,8 @@ class FusedMultiTransformer(nn.Module):
         for i in range(c.num_layers):
             h = self._rms(x, self.ln_scales[i])
             qkv = self._mm(h, "qkv_weights", i)          # [B, T, *]
+            if c.qkv_bias:
+                qkv = qkv + self.qkv_biases[i]
             q = self._rope_append(i, qkv, block_table, zeros, token_counts=prompt_lens)
             # contiguous roped K + raw V for the flash kernel
             qkv_v = qkv.view(B, T, c.num_heads + 2 * c.num_kv_heads, c.head_dim)
@@ -370,9 +529,12 @@ class FusedMultiTransformer(nn.Module):
                 attn = reference.flash_attention(q, k_roped, v, causal=True)
             x = x + self._tp_reduce(self._mm(attn.reshape(B, T, -1), "out_proj_weights", i))
             h = self._rms(x, self.ffn_ln_scales[i])
-            gu = self._mm(h, "gate_up_weights", i)
-            act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
-            x = x + self._tp_reduce(self._mm(act, "down_weights", i))
+            if c.moe_num_experts > 0:
+                x = x + self._tp_reduce(self._moe_ffn(h, i))
+            else:
+                gu = self._mm(h, "gate_up_weights", i)
+                act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
+                x = x + self._tp_reduce(self._mm(act, "down_weights", i))
         x = self._rms(x, self.final_norm)
         idx = (prompt_lens.long() - 1).clamp(min=0)
         last = x[torch.arange(B, device=x.device), idx]
