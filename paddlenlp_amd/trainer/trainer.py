@@ -268,6 +268,32 @@ class Trainer:
             resume_from_checkpoint is None and args.resume_from_checkpoint
         ):
             resume_from_checkpoint = args.resume_from_checkpoint or get_last_checkpoint(args.output_dir)
+            # cross-rank completeness negotiation (reference all-reduces the
+            # resume flag, trainer.py:711-721): with non-shared filesystems
+            # ranks can see different checkpoints — every rank must resume
+            # from the SAME one, or none at all
+            if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+                mine = resume_from_checkpoint if isinstance(resume_from_checkpoint, str) else None
+                views = [None] * dist.get_world_size()
+                dist.all_gather_object(views, mine)
+                common = set(views)
+                if len(common) > 1:
+                    agreed = None
+                    if None not in common:
+                        # all ranks found something but disagree: take the
+                        # oldest (the one every rank is guaranteed to have
+                        # completed past)
+                        def _step_of(p):
+                            import re as _re
+
+                            m = _re.search(r"(\d+)$", p.rstrip("/"))
+                            return int(m.group(1)) if m else -1
+
+                        agreed = min(common, key=_step_of)
+                    logger.warning(
+                        f"checkpoint negotiation: ranks saw {sorted(x or '<none>' for x in common)}; "
+                        f"agreeing on {agreed or '<fresh start>'}")
+                    resume_from_checkpoint = agreed
 
         model = self._wrap_model(self.model)
         self._model_wrapped = model
@@ -660,7 +686,15 @@ class Trainer:
         if args.unified_checkpoint:
             from .unified_checkpoint import save_unified_optimizer
 
-            save_unified_optimizer(self.optimizer, unwrap_model(model), ckpt_dir, self.topology, zero=self._zero)
+            saver = None
+            if args.async_save:
+                if not hasattr(self, "_shm_saver"):
+                    from .utils.shm_save import ShmAsyncSaver
+
+                    self._shm_saver = ShmAsyncSaver()
+                saver = self._shm_saver
+            save_unified_optimizer(self.optimizer, unwrap_model(model), ckpt_dir,
+                                   self.topology, zero=self._zero, saver=saver)
         elif args.process_index == 0 or (self._zero is not None):
             torch.save(self.optimizer.state_dict(), os.path.join(ckpt_dir, OPTIMIZER_STATE_NAME))
 
@@ -672,8 +706,19 @@ class Trainer:
             self.state.save_to_json(os.path.join(ckpt_dir, TRAINER_STATE_NAME))
             self._save_rng_state(ckpt_dir)
             self._rotate_checkpoints()
+        if getattr(self, "_shm_saver", None) is not None:
+            # the marker below asserts completeness: drain the writer first
+            self._shm_saver.wait_all()
         if dist.is_initialized():
             dist.barrier()
+        # integrity marker AFTER every rank finished writing (reference PDC
+        # .checkpoint_done, trainer.py:2520-2524); get_last_checkpoint only
+        # resumes from marked checkpoints
+        if args.process_index == 0:
+            from .trainer_utils import CHECKPOINT_DONE_MARKER
+
+            with open(os.path.join(ckpt_dir, CHECKPOINT_DONE_MARKER), "w") as f:
+                f.write(str(self.state.global_step))
 
     def _save_rng_state(self, ckpt_dir):
         rng = {

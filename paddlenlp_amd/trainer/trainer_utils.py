@@ -161,8 +161,15 @@ def get_scheduler(
     return torch.optim.lr_scheduler.LambdaLR(optimizer, fns[name], last_epoch=last_epoch)
 
 
+CHECKPOINT_DONE_MARKER = ".checkpoint_done"
+
+
 def get_last_checkpoint(folder: str) -> Optional[str]:
-    """Reference: trainer_utils.py get_last_checkpoint."""
+    """Latest COMPLETE checkpoint: only directories carrying the
+    `.checkpoint_done` integrity marker qualify (reference PDC marker,
+    trainer.py:2520-2524), so a save interrupted mid-write is never
+    resumed from.  Falls back to the newest unmarked checkpoint only when
+    no marked one exists (pre-marker checkpoints)."""
     if not os.path.isdir(folder):
         return None
     pat = re.compile(rf"^{PREFIX_CHECKPOINT_DIR}-(\d+)$")
@@ -172,7 +179,10 @@ def get_last_checkpoint(folder: str) -> Optional[str]:
     ]
     if not ckpts:
         return None
-    return os.path.join(folder, max(ckpts, key=lambda d: int(pat.match(d).group(1))))
+    done = [d for d in ckpts
+            if os.path.isfile(os.path.join(folder, d, CHECKPOINT_DONE_MARKER))]
+    pool = done or ckpts
+    return os.path.join(folder, max(pool, key=lambda d: int(pat.match(d).group(1))))
 
 
 class TrainOutput:

@@ -104,6 +104,9 @@ class TrainingArguments:
 
     # checkpoint
     unified_checkpoint: bool = True
+    # optimizer shards written by a shared-memory writer process
+    # (reference unified_checkpoint async_save, :159-299)
+    async_save: bool = False
     resume_from_checkpoint: Optional[str] = None
     save_on_each_node: bool = False
     ignore_data_skip: bool = False

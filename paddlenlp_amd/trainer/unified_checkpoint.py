@@ -171,7 +171,8 @@ def _save_unified_pipe_model(model, output_dir, topology, is_dp0):
             json.dump({"metadata": {"total_size": total}, "weight_map": weight_map}, f, indent=2)
 
 
-def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=None) -> None:
+def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=None,
+                           saver=None) -> None:
     """Per-rank optimizer shards: optimizer-XXXXX-of-N.safetensors + index.
 
     Keys are '<param_name>/moment1' etc. so the file is self-describing and
@@ -213,17 +214,23 @@ def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=Non
                 if state.get("master") is not None:
                     master_tensors[name] = state["master"].cpu()
 
+    def _write(tensors, path):
+        if saver is not None:
+            saver.save_safetensors(tensors, path)   # shm + writer process
+        else:
+            save_file(tensors, path, metadata={"format": "pt"})
+
     fname = mname = None
     if is_dp0:
         fname = SAFE_OPTIMIZER_NAME.replace(
             ".safetensors", f"-{writer_idx + 1:05d}-of-{n_writers:05d}.safetensors"
         )
-        save_file(opt_tensors, os.path.join(output_dir, fname), metadata={"format": "pt"})
+        _write(opt_tensors, os.path.join(output_dir, fname))
         if master_tensors:
             mname = SAFE_MASTER_WEIGHTS_NAME.replace(
                 ".safetensors", f"-{writer_idx + 1:05d}-of-{n_writers:05d}.safetensors"
             )
-            save_file(master_tensors, os.path.join(output_dir, mname), metadata={"format": "pt"})
+            _write(master_tensors, os.path.join(output_dir, mname))
 
     # gather the global index on rank 0 (every rank participates)
     local_index = {

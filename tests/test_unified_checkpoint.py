@@ -149,3 +149,67 @@ if int(os.environ.get("RANK", "0")) == 0:
     # weights have been trained further, but shapes/keys must match
     model_sd = tr2.model.state_dict()
     assert set(saved.keys()) == set(model_sd.keys())
+
+
+def test_shm_async_saver_roundtrip(tmp_path):
+    """shared-memory writer process: bytes on disk == the snapshot."""
+    from paddlenlp_amd.trainer.utils.shm_save import ShmAsyncSaver
+    from safetensors.torch import load_file
+
+    sd = {
+        "a": torch.randn(4, 6),
+        "b": torch.randn(3, dtype=torch.float32).to(torch.bfloat16),
+        "c": torch.arange(5, dtype=torch.int64),
+    }
+    saver = ShmAsyncSaver()
+    path = str(tmp_path / "opt.safetensors")
+    saver.save_safetensors(sd, path)
+    assert saver.wait_all(timeout=60)
+    back = load_file(path)
+    for k in sd:
+        assert back[k].dtype == sd[k].dtype
+        assert torch.equal(back[k], sd[k]), k
+    saver.shutdown()
+
+
+def test_trainer_async_save_resume(tmp_path):
+    """async_save=True checkpoints resume identically to sync ones."""
+    from paddlenlp_amd.trainer import Trainer, TrainingArguments
+
+    class DS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            g = torch.Generator().manual_seed(i)
+            return {"x": torch.randn(4, generator=g), "labels": torch.randn(1, generator=g)}
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(0)
+            self.lin = torch.nn.Linear(4, 1)
+
+        def forward(self, x, labels=None):
+            return ((self.lin(x) - labels) ** 2).mean()
+
+    def run(async_save, out):
+        args = TrainingArguments(
+            output_dir=str(out), per_device_train_batch_size=2, max_steps=4,
+            save_steps=2, logging_steps=100, async_save=async_save)
+        m = M()
+        tr = Trainer(model=m, args=args, train_dataset=DS())
+        tr.train()
+        # resume from the step-2 checkpoint and continue
+        args2 = TrainingArguments(
+            output_dir=str(out), per_device_train_batch_size=2, max_steps=4,
+            save_steps=100, logging_steps=100, async_save=async_save)
+        m2 = M()
+        tr2 = Trainer(model=m2, args=args2, train_dataset=DS())
+        tr2.train(resume_from_checkpoint=str(out / "checkpoint-2"))
+        return {n: p.detach().clone() for n, p in m2.named_parameters()}
+
+    a = run(True, tmp_path / "a")
+    b = run(False, tmp_path / "b")
+    for n in a:
+        assert torch.allclose(a[n], b[n], atol=1e-7), n
