@@ -55,9 +55,13 @@ class PretrainedConfig:
         for key, value in kwargs.items():
             try:
                 setattr(self, key, value)
-            except AttributeError as err:
-                logger.error(f"Can't set {key} with value {value} for {self}")
-                raise err
+            except AttributeError:
+                # read-only property (e.g. a derived `head_dim`) colliding
+                # with a serialized field from a foreign (HF) config.json:
+                # keep the derived value
+                logger.warning(
+                    f"Ignoring config field '{key}'={value}: read-only on "
+                    f"{type(self).__name__}")
 
     # ---- attribute_map support (legacy-name aliasing) ----
     def __setattr__(self, key, value):
@@ -103,6 +107,16 @@ class PretrainedConfig:
     def from_dict(cls, config_dict: Dict[str, Any], **kwargs) -> "PretrainedConfig":
         config_dict = dict(config_dict)
         config_dict.pop("model_type", None)
+        # newer HF config.json nests rope under "rope_parameters"; promote
+        # the fields so they beat subclass defaults
+        rp = config_dict.get("rope_parameters")
+        if isinstance(rp, dict):
+            if "rope_theta" in rp and "rope_theta" not in config_dict:
+                config_dict["rope_theta"] = rp["rope_theta"]
+            if rp.get("rope_type", "default") != "default" and \
+                    "rope_scaling" not in config_dict:
+                config_dict["rope_scaling"] = {
+                    k: v for k, v in rp.items() if k != "rope_theta"}
         config_dict.update(kwargs)
         return cls(**config_dict)
 
