@@ -61,12 +61,24 @@ def paged_decode_attn_ref(q, k_cache, v_cache, block_table, seq_lens,
     bs = k_cache.shape[1]
     Hk = k_cache.shape[2]
     out = torch.empty_like(q)
+
+    def _unpack4(t):
+        # packed uint8 nibbles (offset-binary q+8) -> float
+        hi = (t >> 4).to(torch.int8) - 8
+        lo = (t & 0xF).to(torch.int8) - 8
+        return torch.stack([hi, lo], dim=-1).reshape(*t.shape[:-1], D).float()
+
+    is_int4 = k_cache.dtype == torch.uint8
     for b in range(B):
         L = int(seq_lens[b])
         nb = (L + bs - 1) // bs
         blocks = block_table[b, :nb].long()
-        k = k_cache[blocks].reshape(-1, Hk, D)[:L].float()  # [L, Hk, D]
-        v = v_cache[blocks].reshape(-1, Hk, D)[:L].float()
+        if is_int4:
+            k = _unpack4(k_cache[blocks]).reshape(-1, Hk, D)[:L]
+            v = _unpack4(v_cache[blocks]).reshape(-1, Hk, D)[:L]
+        else:
+            k = k_cache[blocks].reshape(-1, Hk, D)[:L].float()  # [L, Hk, D]
+            v = v_cache[blocks].reshape(-1, Hk, D)[:L].float()
         if k_scale is not None:
             ks = k_scale[blocks].reshape(-1, Hk)[:L].float()
             vs = v_scale[blocks].reshape(-1, Hk)[:L].float()
@@ -87,7 +99,7 @@ def rope_cache_append_ref(qkv, k_cache, v_cache, block_table, seq_lens_before,
                           k_scale=None, v_scale=None):
     """CPU mirror of the fused kernel; returns roped q [B, T, Hq, D]."""
     B, T, _ = qkv.shape
-    D = k_cache.shape[3]
+    D = k_cache.shape[3] * (2 if k_cache.dtype == torch.uint8 else 1)
     bs = k_cache.shape[1]
     qkv = qkv.view(B, T, Hq + 2 * Hk, D)
     q = qkv[:, :, :Hq]
@@ -106,7 +118,23 @@ def rope_cache_append_ref(qkv, k_cache, v_cache, block_table, seq_lens_before,
                 cos[None], sin[None])
             q_out[b, t] = qr[0, 0].to(q.dtype)
             blk = int(block_table[b, pos // bs])
-            if k_scale is not None:  # int8 cache: per-(token, head) absmax
+            if k_scale is not None and k_cache.dtype == torch.uint8:
+                # int4: per-(token, head) absmax/7, offset-binary nibbles
+                kv32 = kr[0, 0].float()
+                vv32 = v[b, t].float()
+                ks = kv32.abs().amax(-1).clamp(min=1e-8) / 7.0
+                vs = vv32.abs().amax(-1).clamp(min=1e-8) / 7.0
+
+                def _pack4(x, sc):
+                    qq = torch.round(x / sc[:, None]).clamp(-7, 7).to(torch.int8) + 8
+                    u = qq.to(torch.uint8)
+                    return (u[:, 0::2] << 4) | u[:, 1::2]
+
+                k_cache[blk, pos % bs] = _pack4(kv32, ks)
+                v_cache[blk, pos % bs] = _pack4(vv32, vs)
+                k_scale[blk, pos % bs] = ks
+                v_scale[blk, pos % bs] = vs
+            elif k_scale is not None:  # int8 cache: per-(token, head) absmax
                 kv32 = kr[0, 0].float()
                 vv32 = v[b, t].float()
                 ks = kv32.abs().amax(-1).clamp(min=1e-8) / 127.0
@@ -412,7 +440,21 @@ class FusedMultiTransformer(nn.Module):
         c = self.config
         shape = (num_blocks, c.block_size, c.num_kv_heads, c.head_dim)
         self.cachekv_dtype = cachekv_dtype
-        if cachekv_dtype == "int8":
+        if cachekv_dtype == "int4":
+            # packed nibbles: 4x capacity vs bf16 (reference
+            # append_attention_c4_impl); per-(token, head) absmax scale
+            assert c.head_dim % 2 == 0
+            p4 = (num_blocks, c.block_size, c.num_kv_heads, c.head_dim // 2)
+            self.k_caches = [torch.zeros(p4, dtype=torch.uint8, device=device)
+                             for _ in range(c.num_layers)]
+            self.v_caches = [torch.zeros(p4, dtype=torch.uint8, device=device)
+                             for _ in range(c.num_layers)]
+            sshape = (num_blocks, c.block_size, c.num_kv_heads)
+            self.k_scales = [torch.zeros(sshape, dtype=torch.float32, device=device)
+                             for _ in range(c.num_layers)]
+            self.v_scales = [torch.zeros(sshape, dtype=torch.float32, device=device)
+                             for _ in range(c.num_layers)]
+        elif cachekv_dtype == "int8":
             self.k_caches = [torch.zeros(shape, dtype=torch.int8, device=device)
                              for _ in range(c.num_layers)]
             self.v_caches = [torch.zeros(shape, dtype=torch.int8, device=device)

@@ -31,10 +31,10 @@ struct AdamWChunk {
     int is_bf16;
 };
 void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, float, float, hipStream_t);
-void launch_paged_decode_attn(const void*, const void*, const void*, const float*, const float*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, hipStream_t);
+void launch_paged_decode_attn(const void*, const void*, const void*, const float*, const float*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, int, hipStream_t);
 int paged_decode_nsplit(int, int);
 void launch_wint8_gemv(const void*, const void*, const float*, void*, int, int, int, hipStream_t);
-void launch_rope_cache_append(const void*, void*, void*, void*, float*, float*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, hipStream_t);
+void launch_rope_cache_append(const void*, void*, void*, void*, float*, float*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, int, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -405,9 +405,13 @@ torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor k_cache, torch::T
     int B = q.size(0), Hq = q.size(1), D = q.size(2);
     int block_size = k_cache.size(1), Hk = k_cache.size(2);
     int max_blocks = block_table.size(1);
-    const bool c8 = (k_cache.scalar_type() == torch::kChar);
-    TORCH_CHECK(!c8 || (k_scale.has_value() && v_scale.has_value()),
-                "int8 KV cache needs k_scale/v_scale");
+    int cache_mode = 0;
+    if (k_cache.scalar_type() == torch::kChar) cache_mode = 1;
+    else if (k_cache.scalar_type() == torch::kByte) cache_mode = 2;
+    TORCH_CHECK(cache_mode == 0 || (k_scale.has_value() && v_scale.has_value()),
+                "quantized KV cache needs k_scale/v_scale");
+    TORCH_CHECK(cache_mode != 2 || (D == 128 && Hq / Hk <= 16),
+                "int4 KV cache requires head dim 128 and GQA group <= 16");
     auto out = torch::empty_like(q);
     float scale = 1.0f / std::sqrt((float)D);
     int G = Hq / Hk;
@@ -420,11 +424,11 @@ torch::Tensor paged_decode_attn(torch::Tensor q, torch::Tensor k_cache, torch::T
         pptr = partials.data_ptr<float>();
     }
     launch_paged_decode_attn(q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-                             c8 ? k_scale->data_ptr<float>() : nullptr,
-                             c8 ? v_scale->data_ptr<float>() : nullptr,
+                             cache_mode ? k_scale->data_ptr<float>() : nullptr,
+                             cache_mode ? v_scale->data_ptr<float>() : nullptr,
                              block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                              out.data_ptr(), pptr, nsplit, B, Hq, Hk, D, block_size,
-                             max_blocks, scale, cur_stream());
+                             max_blocks, scale, cache_mode, cur_stream());
     return out;
 }
 
@@ -438,7 +442,10 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     CHECK_GPU(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
     TORCH_CHECK(qkv.dim() == 3, "qkv must be [B, T, (Hq+2Hk)*D]");
     int B = qkv.size(0), T = qkv.size(1);
-    int D = k_cache.size(3);
+    int cache_mode = 0;
+    if (k_cache.scalar_type() == torch::kChar) cache_mode = 1;
+    else if (k_cache.scalar_type() == torch::kByte) cache_mode = 2;
+    int D = (int)k_cache.size(3) * (cache_mode == 2 ? 2 : 1);
     int block_size = k_cache.size(1);
     int max_blocks = block_table.size(1);
     // zeros (not empty): tokens beyond token_counts[b] are skipped by the
@@ -448,17 +455,17 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     auto sf = sin_t.to(torch::kFloat32).contiguous();
     const int* tc = nullptr;
     if (token_counts.has_value()) tc = token_counts->data_ptr<int>();
-    const bool c8 = (k_cache.scalar_type() == torch::kChar);
-    TORCH_CHECK(!c8 || (k_scale.has_value() && v_scale.has_value()),
-                "int8 KV cache needs k_scale/v_scale");
+    TORCH_CHECK(cache_mode == 0 || (k_scale.has_value() && v_scale.has_value()),
+                "quantized KV cache needs k_scale/v_scale");
+    TORCH_CHECK(cache_mode != 2 || D == 128, "int4 KV cache requires head dim 128");
     launch_rope_cache_append(qkv.data_ptr(), q_out.data_ptr(),
                              k_cache.data_ptr(), v_cache.data_ptr(),
-                             c8 ? k_scale->data_ptr<float>() : nullptr,
-                             c8 ? v_scale->data_ptr<float>() : nullptr,
+                             cache_mode ? k_scale->data_ptr<float>() : nullptr,
+                             cache_mode ? v_scale->data_ptr<float>() : nullptr,
                              block_table.data_ptr<int>(), seq_lens_before.data_ptr<int>(),
                              cf.data_ptr<float>(), sf.data_ptr<float>(),
                              B, T, (int)Hq, (int)Hk, D, block_size, max_blocks, tc,
-                             cur_stream());
+                             cache_mode, cur_stream());
     return q_out;
 }
 

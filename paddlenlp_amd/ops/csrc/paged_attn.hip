@@ -285,7 +285,9 @@ __global__ void paged_decode_merge_kernel(
 // positions p = seq_lens_before[b] + t; writes roped q to q_out [B, T, Hq, D]
 // and roped k / raw v into the paged cache.
 // ---------------------------------------------------------------------------
-template <int D, bool C8>
+// MODE: 0 = bf16, 1 = int8 (absmax/127), 2 = int4 packed nibbles
+// (offset-binary q+8, absmax/7) — reference append_attention_c8/_c4_impl
+template <int D, int MODE>
 __global__ void rope_cache_append_kernel(
     const ushort_t* __restrict__ qkv,
     ushort_t* __restrict__ q_out,
@@ -340,7 +342,7 @@ __global__ void rope_cache_append_kernel(
             if (blk < 0) return;  // inactive slot (device scheduler): no cache write
             const long long tok =
                 ((long long)blk * block_size + pos % block_size) * Hk + hk;
-            if (C8) {
+            if (MODE == 1) {
                 float amax = 0.f;
 #pragma unroll
                 for (int e = 0; e < EPL; e++)
@@ -353,6 +355,29 @@ __global__ void rope_cache_append_kernel(
                     int i = lane * EPL + e;
                     dst[i] = (signed char)lrintf(fminf(fmaxf(o1[e] / sc, -127.f), 127.f));
                     dst[i + half] = (signed char)lrintf(fminf(fmaxf(o2[e] / sc, -127.f), 127.f));
+                }
+                if (lane == 0) k_scale[tok] = sc;
+            } else if (MODE == 2) {
+                // int4: EPL(=2 at D=128) adjacent elements per lane pack
+                // into one byte; nibbles are offset-binary (q+8)
+                static_assert(MODE != 2 || D % 128 == 0 || D == 64, "");
+                float amax = 0.f;
+#pragma unroll
+                for (int e = 0; e < EPL; e++)
+                    amax = fmaxf(amax, fmaxf(fabsf(o1[e]), fabsf(o2[e])));
+                amax = wave_reduce_max(amax);
+                const float sc = fmaxf(amax, 1e-8f) / 7.f;
+                unsigned char* dst = (unsigned char*)k_cache + tok * (D / 2);
+                auto nib = [&](float v) {
+                    int q = (int)lrintf(fminf(fmaxf(v / sc, -7.f), 7.f)) + 8;
+                    return (unsigned)q;
+                };
+#pragma unroll
+                for (int e = 0; e < EPL; e += 2) {
+                    int i = lane * EPL + e;
+                    dst[i / 2] = (unsigned char)((nib(o1[e]) << 4) | nib(o1[e + 1]));
+                    dst[(i + half) / 2] =
+                        (unsigned char)((nib(o2[e]) << 4) | nib(o2[e + 1]));
                 }
                 if (lane == 0) k_scale[tok] = sc;
             } else {
@@ -373,7 +398,7 @@ __global__ void rope_cache_append_kernel(
         const long long tok =
             ((long long)blk * block_size + pos % block_size) * Hk + hv;
         constexpr int VE = D / 64;
-        if (C8) {
+        if (MODE == 1) {
             float vals[VE];
             float amax = 0.f;
 #pragma unroll
@@ -388,6 +413,24 @@ __global__ void rope_cache_append_kernel(
             for (int e = 0; e < VE; e++)
                 dst[lane * VE + e] =
                     (signed char)lrintf(fminf(fmaxf(vals[e] / sc, -127.f), 127.f));
+            if (lane == 0) v_scale[tok] = sc;
+        } else if (MODE == 2) {
+            float vals[VE];
+            float amax = 0.f;
+#pragma unroll
+            for (int e = 0; e < VE; e++) {
+                vals[e] = bf16_to_f32(src[lane * VE + e]);
+                amax = fmaxf(amax, fabsf(vals[e]));
+            }
+            amax = wave_reduce_max(amax);
+            const float sc = fmaxf(amax, 1e-8f) / 7.f;
+            unsigned char* dst = (unsigned char*)v_cache + tok * (D / 2);
+#pragma unroll
+            for (int e = 0; e < VE; e += 2) {
+                int q0 = (int)lrintf(fminf(fmaxf(vals[e] / sc, -7.f), 7.f)) + 8;
+                int q1 = (int)lrintf(fminf(fmaxf(vals[e + 1] / sc, -7.f), 7.f)) + 8;
+                dst[(lane * VE + e) / 2] = (unsigned char)((q0 << 4) | q1);
+            }
             if (lane == 0) v_scale[tok] = sc;
         } else {
             ushort_t* dst = (ushort_t*)v_cache + tok * D;
@@ -416,17 +459,20 @@ bool launch_paged_decode_attn2(const void* q, const void* k_cache, const void* v
                                const int* block_table, const int* seq_lens, void* out,
                                float* partials, int nsplit,
                                int B, int Hq, int Hk, int D, int block_size,
-                               int max_blocks, float scale, hipStream_t stream);
+                               int max_blocks, float scale, int cache_mode,
+                               hipStream_t stream);
 
 void launch_paged_decode_attn(const void* q, const void* k_cache, const void* v_cache,
                               const float* k_scale, const float* v_scale,
                               const int* block_table, const int* seq_lens, void* out,
                               float* partials, int nsplit,
                               int B, int Hq, int Hk, int D, int block_size,
-                              int max_blocks, float scale, hipStream_t stream) {
+                              int max_blocks, float scale, int cache_mode,
+                              hipStream_t stream) {
     if (launch_paged_decode_attn2(q, k_cache, v_cache, k_scale, v_scale,
                                   block_table, seq_lens, out, partials, nsplit,
-                                  B, Hq, Hk, D, block_size, max_blocks, scale, stream)) {
+                                  B, Hq, Hk, D, block_size, max_blocks, scale,
+                                  cache_mode, stream)) {
         if (nsplit > 1) {
             if (D == 128)
                 hipLaunchKernelGGL(paged_decode_merge_kernel<128>, dim3(B, Hk, Hq / Hk),
@@ -461,19 +507,21 @@ void launch_rope_cache_append(const void* qkv, void* q_out, void* k_cache, void*
                               const int* block_table, const int* seq_lens_before,
                               const float* cos_t, const float* sin_t,
                               int B, int T, int Hq, int Hk, int D, int block_size,
-                              int max_blocks, const int* token_counts, hipStream_t stream) {
+                              int max_blocks, const int* token_counts,
+                              int cache_mode, hipStream_t stream) {
     dim3 grid(B * T, Hq + 2 * Hk);
-    const bool c8 = (k_scale != nullptr);
-#define RC_LAUNCH(DD, CC)                                                         \
-    hipLaunchKernelGGL((rope_cache_append_kernel<DD, CC>), grid, dim3(64), 0,     \
+#define RC_LAUNCH(DD, MM)                                                         \
+    hipLaunchKernelGGL((rope_cache_append_kernel<DD, MM>), grid, dim3(64), 0,     \
                        stream, (const ushort_t*)qkv, (ushort_t*)q_out,            \
                        k_cache, v_cache, k_scale, v_scale,                        \
                        block_table, seq_lens_before, cos_t, sin_t,                \
                        B, T, Hq, Hk, block_size, max_blocks, token_counts)
     if (D == 128) {
-        if (c8) RC_LAUNCH(128, true); else RC_LAUNCH(128, false);
+        if (cache_mode == 2) RC_LAUNCH(128, 2);
+        else if (cache_mode == 1) RC_LAUNCH(128, 1);
+        else RC_LAUNCH(128, 0);
     } else if (D == 64) {
-        if (c8) RC_LAUNCH(64, true); else RC_LAUNCH(64, false);
+        if (cache_mode == 1) RC_LAUNCH(64, 1); else RC_LAUNCH(64, 0);
     }
 #undef RC_LAUNCH
 }

@@ -45,7 +45,8 @@ static __device__ __forceinline__ char* swzd(ushort_t* base, int row, int col_el
            (((row * LD + col_elem) * 2) ^ ((row & 7) << 4));
 }
 
-template <int D, bool C8>
+// MODE: 0 = bf16 cache, 1 = int8, 2 = int4 packed nibbles (q+8 offset)
+template <int D, int MODE>
 __global__ __launch_bounds__(PD2_BLOCK) void paged_decode_attn2_kernel(
     const ushort_t* __restrict__ q,        // [B, Hq, D]
     const void* __restrict__ k_cache,      // [nblocks, bs, Hk, D] bf16|int8
@@ -130,13 +131,27 @@ __global__ __launch_bounds__(PD2_BLOCK) void paged_decode_attn2_kernel(
                 int blk = bt[tok / block_size];
                 long long rec = ((long long)blk * block_size + (tok % block_size)) * Hk + hk;
                 long long base = rec * D + s_c;
-                if (C8) {
+                if (MODE == 1) {
                     float vs = v_scale[rec];
 #pragma unroll
                     for (int cc = 0; cc < 4; cc++)
 #pragma unroll
                         for (int j = 0; j < 8; j++)
                             vr[rr][cc][j] = (short)f32_to_bf16((float)v8[base + cc * 8 + j] * vs);
+                } else if (MODE == 2) {
+                    float vs = v_scale[rec];
+                    const unsigned char* v4 =
+                        (const unsigned char*)v_cache + (rec * D + s_c) / 2;
+#pragma unroll
+                    for (int cc = 0; cc < 4; cc++)
+#pragma unroll
+                        for (int j = 0; j < 4; j++) {
+                            unsigned byte = v4[cc * 4 + j];
+                            vr[rr][cc][j * 2] = (short)f32_to_bf16(
+                                ((int)(byte >> 4) - 8) * vs);
+                            vr[rr][cc][j * 2 + 1] = (short)f32_to_bf16(
+                                ((int)(byte & 0xF) - 8) * vs);
+                        }
                 } else {
 #pragma unroll
                     for (int cc = 0; cc < 4; cc++)
@@ -189,7 +204,7 @@ __global__ __launch_bounds__(PD2_BLOCK) void paged_decode_attn2_kernel(
                 int blk = bt[tokc / block_size];
                 long long rec = ((long long)blk * block_size + (tokc % block_size)) * Hk + hk;
                 const long long kbase = rec * D;
-                if (C8) {
+                if (MODE == 1) {
                     float ks = k_scale[rec];
                     frag8 ak[DSTEPS];
 #pragma unroll
@@ -198,6 +213,25 @@ __global__ __launch_bounds__(PD2_BLOCK) void paged_decode_attn2_kernel(
                         for (int j = 0; j < 8; j++)
                             ak[kk][j] = (short)f32_to_bf16(
                                 (float)k8[kbase + kk * 16 + hi * 8 + j] * ks);
+                    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+                    for (int kk = 0; kk < DSTEPS; kk++)
+                        st = mfma32d(ak[kk], qT[kk], st);
+                    __builtin_amdgcn_s_setprio(0);
+                } else if (MODE == 2) {
+                    float ks = k_scale[rec];
+                    const unsigned char* k4 = (const unsigned char*)k_cache + kbase / 2;
+                    frag8 ak[DSTEPS];
+#pragma unroll
+                    for (int kk = 0; kk < DSTEPS; kk++)
+#pragma unroll
+                        for (int j = 0; j < 4; j++) {
+                            unsigned byte = k4[(kk * 16 + hi * 8) / 2 + j];
+                            ak[kk][j * 2] = (short)f32_to_bf16(
+                                ((int)(byte >> 4) - 8) * ks);
+                            ak[kk][j * 2 + 1] = (short)f32_to_bf16(
+                                ((int)(byte & 0xF) - 8) * ks);
+                        }
                     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
                     for (int kk = 0; kk < DSTEPS; kk++)
@@ -365,19 +399,18 @@ bool launch_paged_decode_attn2(const void* q, const void* k_cache, const void* v
                                const int* block_table, const int* seq_lens, void* out,
                                float* partials, int nsplit,
                                int B, int Hq, int Hk, int D, int block_size,
-                               int max_blocks, float scale, hipStream_t stream) {
+                               int max_blocks, float scale, int cache_mode,
+                               hipStream_t stream) {
     if (D != 128 || (Hq % Hk) != 0 || Hq / Hk > PD2_MAXG) return false;
     dim3 grid(B, Hk, nsplit);
-    const bool c8 = (k_scale != nullptr);
-    if (c8)
-        hipLaunchKernelGGL((paged_decode_attn2_kernel<128, true>), grid, dim3(PD2_BLOCK),
-                           0, stream, (const ushort_t*)q, k_cache, v_cache, k_scale,
-                           v_scale, block_table, seq_lens, (ushort_t*)out, partials,
-                           B, Hq, Hk, block_size, max_blocks, scale, nsplit);
-    else
-        hipLaunchKernelGGL((paged_decode_attn2_kernel<128, false>), grid, dim3(PD2_BLOCK),
-                           0, stream, (const ushort_t*)q, k_cache, v_cache, k_scale,
-                           v_scale, block_table, seq_lens, (ushort_t*)out, partials,
-                           B, Hq, Hk, block_size, max_blocks, scale, nsplit);
+#define PD2_LAUNCH(MM)                                                             \
+    hipLaunchKernelGGL((paged_decode_attn2_kernel<128, MM>), grid, dim3(PD2_BLOCK),\
+                       0, stream, (const ushort_t*)q, k_cache, v_cache, k_scale,   \
+                       v_scale, block_table, seq_lens, (ushort_t*)out, partials,   \
+                       B, Hq, Hk, block_size, max_blocks, scale, nsplit)
+    if (cache_mode == 2) PD2_LAUNCH(2);
+    else if (cache_mode == 1) PD2_LAUNCH(1);
+    else PD2_LAUNCH(0);
+#undef PD2_LAUNCH
     return true;
 }

@@ -375,3 +375,45 @@ def test_engine_from_mixtral_matches_dygraph():
             ref = (out[0] if isinstance(out, tuple) else out)[:, -1]
         assert torch.allclose(logits, ref.float(), atol=1e-3), \
             (step, (logits - ref.float()).abs().max())
+
+
+def test_int4_kv_cache_cpu():
+    """int4 paged KV cache (packed nibbles, absmax/7 per token-head): CPU
+    reference path stays close to the bf16-cache engine."""
+    import torch
+
+    from paddlenlp_amd.experimental import BlockManager, FusedMultiTransformer
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64)
+    model = LlamaForCausalLM.from_config(cfg).eval()
+
+    outs = {}
+    for dtype in ("bf16", "int4"):
+        eng = FusedMultiTransformer.from_llama(model, block_size=8, max_seq_len=64)
+        eng.allocate_caches(16, torch.device("cpu"), cachekv_dtype=dtype)
+        mgr = BlockManager(16, 8, 8, 2)
+        ids = torch.randint(3, 128, (2, 10), generator=torch.Generator().manual_seed(1))
+        lens = torch.tensor([10, 10], dtype=torch.int32)
+        slots = [mgr.allocate_slot(10) for _ in range(2)]
+        bt = torch.stack([mgr.block_table[s] for s in slots]).to(torch.int32)
+        logits = eng.prefill(ids, bt, lens)
+        tok = logits.argmax(-1, keepdim=True)
+        lens_before = torch.tensor([10, 10], dtype=torch.int32)
+        for s in slots:
+            assert mgr.extend(s, 1)
+        bt = torch.stack([mgr.block_table[s] for s in slots]).to(torch.int32)
+        step = eng.decode_step(tok, bt, lens_before)
+        outs[dtype] = (logits.float(), step.float())
+        if dtype == "int4":
+            assert eng.k_caches[0].dtype == torch.uint8
+            assert eng.k_caches[0].shape[-1] == 8   # head_dim 16 packed /2
+            assert float(eng.k_scales[0].abs().sum()) > 0
+
+    for a, b in zip(outs["bf16"], outs["int4"]):
+        rel = (a - b).abs().max() / a.abs().max().clamp(min=1e-6)
+        assert rel < 0.25, rel  # 4-bit KV: coarser but bounded

@@ -483,3 +483,58 @@ def test_mixed_prefill_decode_step(C):
     p_seq = eng.prefill(p_ids, p_bt, p_lens)
     assert torch.equal(d_mix, d_seq)
     assert torch.equal(p_mix, p_seq)
+
+
+def test_int4_kv_cache_gpu(C):
+    """int4 cache kernels (quantized rope-append + dequantizing MFMA decode)
+    match the CPU reference mirror."""
+    from paddlenlp_amd.experimental.fused_transformer import (
+        paged_decode_attn_ref, rope_cache_append_ref)
+
+    torch.manual_seed(4)
+    B, T, Hq, Hk, D = 2, 12, 8, 2, 128
+    bs, nblocks, mb = 16, 16, 4
+    qkv = torch.randn(B, T, (Hq + 2 * Hk) * D, device="cuda",
+                      dtype=torch.bfloat16)
+    from paddlenlp_amd import ops
+    cos, sin = ops.build_rope_cache(64, D, device="cuda")
+    bt = torch.arange(B * mb, dtype=torch.int32, device="cuda").reshape(B, mb)
+    lens0 = torch.zeros(B, dtype=torch.int32, device="cuda")
+    counts = torch.tensor([T, T - 3], dtype=torch.int32, device="cuda")
+
+    kc = torch.zeros(nblocks, bs, Hk, D // 2, dtype=torch.uint8, device="cuda")
+    vc = torch.zeros_like(kc)
+    ks = torch.zeros(nblocks, bs, Hk, dtype=torch.float32, device="cuda")
+    vs = torch.zeros_like(ks)
+    q = C.rope_cache_append(qkv, kc, vc, bt, lens0, cos, sin, Hq, Hk,
+                            counts, ks, vs)
+
+    kc_ref = torch.zeros_like(kc, device="cpu")
+    vc_ref = torch.zeros_like(vc, device="cpu")
+    ks_ref = torch.zeros_like(ks, device="cpu")
+    vs_ref = torch.zeros_like(vs, device="cpu")
+    q_ref = rope_cache_append_ref(qkv.cpu(), kc_ref, vc_ref, bt.cpu(),
+                                  lens0.cpu(), cos.cpu(), sin.cpu(), Hq, Hk,
+                                  token_counts=counts.cpu(),
+                                  k_scale=ks_ref, v_scale=vs_ref)
+    assert torch.allclose(q.cpu().float(), q_ref.reshape(B, T, Hq, D).float(),
+                          atol=2e-2, rtol=2e-2)
+    # packed nibbles agree up to rounding ties (each nibble within 1 step);
+    # scales must match closely
+    assert torch.allclose(ks.cpu(), ks_ref, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(vs.cpu(), vs_ref, atol=1e-4, rtol=1e-3)
+    kd = (kc.cpu().int() - kc_ref.int()).abs()
+    vd = (vc.cpu().int() - vc_ref.int()).abs()
+    assert int(kd.max()) <= 17 and int(vd.max()) <= 17, (kd.max(), vd.max())
+
+    # decode against the (GPU-written) int4 cache; the CPU mirror reads the
+    # SAME cache — comparing decodes over independently rounded caches
+    # amplifies nibble ties through the softmax
+    seq_lens = counts.clone()
+    qd = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    out = C.paged_decode_attn(qd, kc, vc, bt, seq_lens, ks, vs)
+    out_ref = paged_decode_attn_ref(qd.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                    seq_lens.cpu(), ks.cpu(), vs.cpu())
+    assert torch.allclose(out.cpu().float(), out_ref.float(),
+                          atol=5e-2, rtol=5e-2), \
+        (out.cpu().float() - out_ref.float()).abs().max()

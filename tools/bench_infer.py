@@ -126,7 +126,7 @@ def main():
     p.add_argument("--quant", default="")          # "" | fp8 | weight_only_int8
     p.add_argument("--graph", action="store_true")
     p.add_argument("--device-sched", action="store_true")
-    p.add_argument("--cachekv", default="bf16", choices=["bf16", "int8"])
+    p.add_argument("--cachekv", default="bf16", choices=["bf16", "int8", "int4"])
     args = p.parse_args()
 
     device = "cuda:0"
