@@ -80,3 +80,68 @@ def test_save_load_and_auto(tmp_path, tok):
     assert loaded.encode("hello world") == tok.encode("hello world")
     assert loaded.pad_token_id == tok.pad_token_id
     assert loaded.eos_token == "</s>"
+
+
+def _word_tok():
+    from tokenizers import Tokenizer, models, pre_tokenizers
+    from paddlenlp_amd.transformers.tokenizer_utils import PretrainedTokenizer
+
+    vocab = {"<unk>": 0, "<s>": 1, "</s>": 2, "[CLS]": 3, "[SEP]": 4, "[MASK]": 5}
+    for i, w in enumerate("the quick brown fox jumps over lazy dog a and cat sat mat".split()):
+        vocab[w] = 6 + i
+    tok = Tokenizer(models.WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    return PretrainedTokenizer(tokenizer=tok, bos_token="<s>", eos_token="</s>",
+                               pad_token="</s>", unk_token="<unk>",
+                               cls_token="[CLS]", sep_token="[SEP]",
+                               mask_token="[MASK]")
+
+
+def test_text_pair_token_type_ids_and_truncation():
+    tok = _word_tok()
+    out = tok("the quick brown fox", text_pair="lazy dog", truncation="longest_first",
+              max_length=5)
+    assert len(out["input_ids"]) == 5
+    assert out["token_type_ids"] == [0, 0, 0, 1, 1]
+    out2 = tok("the quick brown fox", text_pair="lazy dog",
+               truncation="only_first", max_length=4)
+    # only_first trims the first segment, keeps the pair
+    assert out2["token_type_ids"] == [0, 0, 1, 1]
+    out3 = tok("the quick", text_pair="lazy dog cat sat",
+               truncation="only_second", max_length=4)
+    assert out3["token_type_ids"] == [0, 0, 1, 1]
+
+
+def test_offsets_mapping_and_padding_sides():
+    tok = _word_tok()
+    out = tok("the quick fox", return_offsets_mapping=True)
+    offs = out["offset_mapping"]
+    assert offs[0] == (0, 3) and offs[1] == (4, 9)
+    # right padding (encoder style) + pad_to_multiple_of
+    out = tok(["the quick fox", "dog"], padding=True, padding_side="right",
+              pad_to_multiple_of=4)
+    assert all(len(i) == 4 for i in out["input_ids"])
+    assert out["attention_mask"][1] == [1, 0, 0, 0]
+    # left padding (decoder default)
+    out = tok(["the quick fox", "dog"], padding=True)
+    assert out["attention_mask"][1] == [0, 0, 1]
+
+
+def test_special_token_registration_and_roundtrip(tmp_path):
+    from paddlenlp_amd.transformers.tokenizer_utils import PretrainedTokenizer
+
+    tok = _word_tok()
+    n = tok.add_special_tokens({"additional_special_tokens": ["<|im_start|>", "<|im_end|>"]})
+    assert n == 2
+    assert "<|im_start|>" in tok.all_special_tokens
+    sid = tok.convert_tokens_to_ids("<|im_start|>")
+    assert sid is not None and sid in tok.all_special_ids
+    # special tokens survive decode-skip
+    ids = tok.encode("the quick") + [sid]
+    assert "<|im_start|>" not in tok.decode(ids, skip_special_tokens=True)
+    # save / reload keeps everything
+    tok.save_pretrained(str(tmp_path))
+    back = PretrainedTokenizer.from_pretrained(str(tmp_path))
+    assert back.cls_token == "[CLS]" and back.mask_token == "[MASK]"
+    assert "<|im_end|>" in back.additional_special_tokens
+    assert back.convert_tokens_to_ids("<|im_start|>") == sid
