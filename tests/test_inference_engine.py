@@ -417,3 +417,22 @@ def test_int4_kv_cache_cpu():
     for a, b in zip(outs["bf16"], outs["int4"]):
         rel = (a - b).abs().max() / a.abs().max().clamp(min=1e-6)
         assert rel < 0.25, rel  # 4-bit KV: coarser but bounded
+
+
+def test_static_export_roundtrip(tmp_path):
+    """torch.export static program == eager forward, across dynamic shapes
+    (reference llm/predict/export_model.py dy2static equivalent)."""
+    import sys
+    sys.path.insert(0, ".")
+    from llm.predict.export_model import export_model, load_exported
+
+    model = tiny_llama(seed=9)
+    export_model(model, str(tmp_path))
+    fn = load_exported(str(tmp_path))
+    for B, S in ((1, 8), (2, 5), (3, 12)):
+        ids = torch.randint(3, 128, (B, S), generator=torch.Generator().manual_seed(B))
+        with torch.no_grad():
+            ref = model(input_ids=ids)
+            ref = ref[0] if isinstance(ref, tuple) else ref
+            got = fn(ids)
+        assert torch.allclose(got, ref, atol=1e-5), (B, S, (got - ref).abs().max())
