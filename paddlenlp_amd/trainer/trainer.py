@@ -253,7 +253,8 @@ class Trainer:
         self.is_in_train = True
         mem_tracker = TrainerMemoryTracker(skip=args.skip_memory_metrics)
         mem_tracker.start("train")
-        self._loss_scaler = DynamicLossScaler() if args.fp16 else None
+        self._loss_scaler = (DynamicLossScaler(init_scale=args.scale_loss)
+                             if args.fp16 else None)
 
         train_dataloader = self.get_train_dataloader()
         steps_per_epoch = max(1, len(train_dataloader) // args.gradient_accumulation_steps)

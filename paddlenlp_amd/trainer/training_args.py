@@ -119,12 +119,61 @@ class TrainingArguments:
     max_evaluate_steps: int = -1
     skip_profile_timer: bool = True
 
+    # ---- reference-compatible config strings (training_args.py:655-690):
+    # space-separated option switches parsed in __post_init__ and mapped
+    # onto the native knobs
+    tensor_parallel_config: str = ""
+    pipeline_parallel_config: str = ""
+    sharding_parallel_config: str = ""
+    hybrid_parallel_topo_order: str = "dp_first"   # "dp_first" | "sharding_first"
+    # fp16 initial loss scale (reference scale_loss)
+    scale_loss: float = 2.0 ** 15
+    fp16_opt_level: str = "O2"
+    amp_custom_black_list: Optional[List[str]] = None
+    amp_custom_white_list: Optional[List[str]] = None
+    bf16_full_eval: bool = False
+    eval_accumulation_steps: Optional[int] = None
+    prediction_loss_only: bool = False
+    metric_for_best_model: Optional[str] = None
+    greater_is_better: Optional[bool] = None
+    load_best_model_at_end: bool = False
+
     def __post_init__(self):
         self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
         self.local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
         if self.fp16 and self.bf16:
             raise ValueError("Pick one of fp16/bf16")
+
+        # reference-style option strings -> native switches
+        self.tp_options = set(self.tensor_parallel_config.split())
+        self.pp_options = set(self.pipeline_parallel_config.split())
+        self.sd_options = set(self.sharding_parallel_config.split())
+        known_tp = {"enable_mp_async_allreduce", "enable_mp_skip_c_identity",
+                    "enable_mp_fused_linear_param_grad_add", "sync_param",
+                    "sync_grad", "sync_moment"}
+        known_pp = {"disable_p2p_cache_shape", "disable_partial_send_recv",
+                    "enable_delay_scale_loss", "enable_dp_comm_overlap",
+                    "enable_sharding_comm_overlap", "enable_release_grads",
+                    "enable_timer", "enable_overlap_p2p_comm"}
+        known_sd = {"enable_stage1_tensor_fusion", "enable_stage1_overlap",
+                    "enable_stage2_overlap", "split_param",
+                    "enable_stage1_broadcast_overlap", "disable_stage1_reduce_avg",
+                    "enable_stage1_allgather_overlap"}
+        for opts, known, name in ((self.tp_options, known_tp, "tensor"),
+                                  (self.pp_options, known_pp, "pipeline"),
+                                  (self.sd_options, known_sd, "sharding")):
+            unknown = opts - known
+            if unknown:
+                logger.warning(f"ignoring unknown {name}_parallel_config "
+                               f"options: {sorted(unknown)}")
+        if {"enable_stage1_overlap", "enable_stage2_overlap",
+            "enable_stage1_allgather_overlap"} & self.sd_options:
+            self.sharding_overlap_comm = True
+        if "enable_timer" in self.pp_options:
+            self.skip_profile_timer = False
+        # flat shard-aligned comm buffers are the native default, so
+        # enable_stage1_tensor_fusion / split_param are already in effect
 
         # sharding options
         self.sharding_options = set()

@@ -146,3 +146,33 @@ def test_save_total_limit():
         tr.train()
         ckpts = sorted(x for x in os.listdir(d) if x.startswith("checkpoint"))
         assert ckpts == ["checkpoint-4", "checkpoint-6"]
+
+
+def test_reference_config_strings_parse():
+    """Reference-style recipe JSON (tp/pp/sharding config strings) parses
+    and maps onto the native switches (reference training_args.py:655-690)."""
+    import json
+    import tempfile
+
+    from paddlenlp_amd.trainer import TrainingArguments
+    from paddlenlp_amd.trainer.argparser import PdArgumentParser
+
+    recipe = {
+        "output_dir": "out",
+        "per_device_train_batch_size": 1,
+        "tensor_parallel_config": "enable_mp_async_allreduce enable_mp_skip_c_identity",
+        "pipeline_parallel_config": "enable_timer disable_p2p_cache_shape",
+        "sharding_parallel_config": "enable_stage1_tensor_fusion enable_stage1_overlap split_param",
+        "scale_loss": 1024.0,
+        "fp16_opt_level": "O2",
+        "hybrid_parallel_topo_order": "sharding_first",
+    }
+    with tempfile.NamedTemporaryFile("w", suffix=".json", delete=False) as f:
+        json.dump(recipe, f)
+        path = f.name
+    parser = PdArgumentParser([TrainingArguments])
+    (args,) = parser.parse_json_file(path)
+    assert "enable_mp_async_allreduce" in args.tp_options
+    assert args.skip_profile_timer is False          # enable_timer
+    assert args.sharding_overlap_comm is True        # enable_stage1_overlap
+    assert args.scale_loss == 1024.0
