@@ -4,6 +4,7 @@ from .modeling import (  # noqa: F401
     AutoModelForCausalLM,
     AutoModelForConditionalGeneration,
     AutoModelForMaskedLM,
+    AutoModelForMultipleChoice,
     AutoModelForSeq2SeqLM,
     AutoModelForQuestionAnswering,
     AutoModelForSequenceClassification,

@@ -45,5 +45,9 @@ class AutoModelForQuestionAnswering(_AutoBase):
     _kind = "question_answering"
 
 
+class AutoModelForMultipleChoice(_AutoBase):
+    _kind = "multiple_choice"
+
+
 class AutoModelForMaskedLM(_AutoBase):
     _kind = "masked_lm"

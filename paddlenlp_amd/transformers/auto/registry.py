@@ -171,15 +171,30 @@ def resolve_model_type(path: str) -> str:
     return model_type
 
 
+_TASK_HEAD_KINDS = {"sequence_classification", "token_classification",
+                    "question_answering", "multiple_choice"}
+
+
 def get_class(model_type: str, kind: str):
     """kind in {config, base, causal_lm, sequence_classification,
-    token_classification, question_answering, masked_lm}."""
+    token_classification, question_answering, multiple_choice, masked_lm}.
+
+    Task-head kinds missing from a family's registry entry are synthesized
+    over its base model (transformers.task_heads) — reference parity:
+    nearly every family exposes these heads."""
     if model_type not in MODEL_REGISTRY:
         raise ValueError(
             f"Unknown model_type '{model_type}'. Registered: {sorted(MODEL_REGISTRY)}"
         )
     entry = MODEL_REGISTRY[model_type]
     if kind not in entry:
+        if kind in _TASK_HEAD_KINDS and "base" in entry:
+            from ..task_heads import synthesize_head
+
+            module = importlib.import_module(
+                f"paddlenlp_amd.transformers.{entry['module']}")
+            base_cls = getattr(module, entry["base"])
+            return synthesize_head(base_cls, kind)
         raise ValueError(f"model_type '{model_type}' has no {kind} head")
     module = importlib.import_module(
         f"paddlenlp_amd.transformers.{entry['module']}")
