@@ -117,9 +117,10 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     constexpr int NDT = D / 32;
     constexpr int BLKM = FB2_WAVES * 32;
 
-    __shared__ ushort_t k_lds[BLKN * D];    // row-major [kv][d]
-    __shared__ ushort_t v_lds[BLKN * D];    // row-major [kv][d]
-    __shared__ ushort_t kt_lds[D * BLKN];   // transposed [d][kv]
+    // double-buffered (one barrier per KV tile; see flash_attn_v2.hip)
+    __shared__ ushort_t k_lds[2][BLKN * D];    // row-major [kv][d]
+    __shared__ ushort_t v_lds[2][BLKN * D];    // row-major [kv][d]
+    __shared__ ushort_t kt_lds[2][D * BLKN];   // transposed [d][kv]
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -191,25 +192,26 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
             sv1 = *reinterpret_cast<const short8v*>(v_ptr + (long long)(g0 + 1) * kv_row_stride + s_col);
         }
     };
-    auto write_tile = [&]() {
-        *reinterpret_cast<short8v*>(swzb2<D>(k_lds, s_row0, s_col)) = sk0;
-        *reinterpret_cast<short8v*>(swzb2<D>(k_lds, s_row0 + 1, s_col)) = sk1;
-        *reinterpret_cast<short8v*>(swzb2<D>(v_lds, s_row0, s_col)) = sv0;
-        *reinterpret_cast<short8v*>(swzb2<D>(v_lds, s_row0 + 1, s_col)) = sv1;
+    auto write_tile = [&](int buf) {
+        *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0, s_col)) = sk0;
+        *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
+        *reinterpret_cast<short8v*>(swzb2<D>(v_lds[buf], s_row0, s_col)) = sv0;
+        *reinterpret_cast<short8v*>(swzb2<D>(v_lds[buf], s_row0 + 1, s_col)) = sv1;
 #pragma unroll
         for (int j = 0; j < 8; j++) {
             unsigned p32 = ((unsigned)(unsigned short)sk0[j]) |
                            (((unsigned)(unsigned short)sk1[j]) << 16);
-            *reinterpret_cast<unsigned*>(swzb2<BLKN>(kt_lds, s_col + j, s_row0)) = p32;
+            *reinterpret_cast<unsigned*>(swzb2<BLKN>(kt_lds[buf], s_col + j, s_row0)) = p32;
         }
     };
 
     load_tile(0);
-    write_tile();
+    write_tile(0);
     __syncthreads();
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
         const int kv_base = kvt * BLKN;
+        const int cur = kvt & 1;
         const bool wave_skip =
             causal && (kv_base > qw + 31 + causal_off);
 
@@ -223,9 +225,9 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
 #pragma unroll
             for (int kk = 0; kk < DSTEPS; kk++) {
                 frag8 ak = *reinterpret_cast<const frag8*>(
-                    swzb2<D>(k_lds, sub * 32 + l32, kk * 16 + hi * 8));
+                    swzb2<D>(k_lds[cur], sub * 32 + l32, kk * 16 + hi * 8));
                 frag8 av = *reinterpret_cast<const frag8*>(
-                    swzb2<D>(v_lds, sub * 32 + l32, kk * 16 + hi * 8));
+                    swzb2<D>(v_lds[cur], sub * 32 + l32, kk * 16 + hi * 8));
                 st = mfma32b(ak, aq[kk], st);
                 dp = mfma32b(av, ado[kk], dp);
             }
@@ -252,7 +254,7 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
 #pragma unroll
                 for (int n = 0; n < NDT; n++) {
                     frag8 bkf = *reinterpret_cast<const frag8*>(
-                        swzb2<BLKN>(kt_lds, n * 32 + l32, ks * 16 + hi * 8));
+                        swzb2<BLKN>(kt_lds[cur], n * 32 + l32, ks * 16 + hi * 8));
                     acc_dq[n] = mfma32b(ads, bkf, acc_dq[n]);
                 }
             }
@@ -263,11 +265,8 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
         if (kvt + 1 < n_kv_tiles) load_tile(kv_base + BLKN);
         if (!wave_skip) process_sub(1);
 
+        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1);
         __syncthreads();
-        if (kvt + 1 < n_kv_tiles) {
-            write_tile();
-            __syncthreads();
-        }
     }
 
     // epilogue: acc_dq C-layout [q][d] (col = d, row = q)
@@ -299,12 +298,13 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
     constexpr int NDT = D / 32;
     constexpr int BLKKV = FB2_WAVES * 32;  // 256 kv rows per block
 
-    __shared__ ushort_t q_lds[BLKQ * D];     // row-major [q][d]
-    __shared__ ushort_t do_lds[BLKQ * D];    // row-major [q][d]
-    __shared__ ushort_t qt_lds[D * BLKQ];    // transposed [d][q]
-    __shared__ ushort_t dot_lds[D * BLKQ];   // transposed [d][q]
-    __shared__ float lse_lds[BLKQ];
-    __shared__ float dl_lds[BLKQ];
+    // double-buffered q-side tiles (one barrier per q tile)
+    __shared__ ushort_t q_lds[2][BLKQ * D];     // row-major [q][d]
+    __shared__ ushort_t do_lds[2][BLKQ * D];    // row-major [q][d]
+    __shared__ ushort_t qt_lds[2][D * BLKQ];    // transposed [d][q]
+    __shared__ ushort_t dot_lds[2][D * BLKQ];   // transposed [d][q]
+    __shared__ float lse_lds[2][BLKQ];
+    __shared__ float dl_lds[2][BLKQ];
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -376,33 +376,34 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
                 sd1 = *reinterpret_cast<const short8v*>(do_ptr + (long long)(g0 + 1) * q_row_stride + s_col);
             }
         };
-        auto write_qtile = [&](int q_tb) {
-            *reinterpret_cast<short8v*>(swzb2<D>(q_lds, s_row0, s_col)) = sq0;
-            *reinterpret_cast<short8v*>(swzb2<D>(q_lds, s_row0 + 1, s_col)) = sq1;
-            *reinterpret_cast<short8v*>(swzb2<D>(do_lds, s_row0, s_col)) = sd0;
-            *reinterpret_cast<short8v*>(swzb2<D>(do_lds, s_row0 + 1, s_col)) = sd1;
+        auto write_qtile = [&](int q_tb, int buf) {
+            *reinterpret_cast<short8v*>(swzb2<D>(q_lds[buf], s_row0, s_col)) = sq0;
+            *reinterpret_cast<short8v*>(swzb2<D>(q_lds[buf], s_row0 + 1, s_col)) = sq1;
+            *reinterpret_cast<short8v*>(swzb2<D>(do_lds[buf], s_row0, s_col)) = sd0;
+            *reinterpret_cast<short8v*>(swzb2<D>(do_lds[buf], s_row0 + 1, s_col)) = sd1;
 #pragma unroll
             for (int j = 0; j < 8; j++) {
                 unsigned pq = ((unsigned)(unsigned short)sq0[j]) |
                               (((unsigned)(unsigned short)sq1[j]) << 16);
                 unsigned pd = ((unsigned)(unsigned short)sd0[j]) |
                               (((unsigned)(unsigned short)sd1[j]) << 16);
-                *reinterpret_cast<unsigned*>(swzb2<BLKQ>(qt_lds, s_col + j, s_row0)) = pq;
-                *reinterpret_cast<unsigned*>(swzb2<BLKQ>(dot_lds, s_col + j, s_row0)) = pd;
+                *reinterpret_cast<unsigned*>(swzb2<BLKQ>(qt_lds[buf], s_col + j, s_row0)) = pq;
+                *reinterpret_cast<unsigned*>(swzb2<BLKQ>(dot_lds[buf], s_col + j, s_row0)) = pd;
             }
             if (tid < BLKQ) {
                 int qgl = q_tb + tid;
-                lse_lds[tid] = (qgl < Sq) ? lse_row[qgl] : INFINITY;
-                dl_lds[tid] = (qgl < Sq) ? dl_row[qgl] : 0.f;
+                lse_lds[buf][tid] = (qgl < Sq) ? lse_row[qgl] : INFINITY;
+                dl_lds[buf][tid] = (qgl < Sq) ? dl_row[qgl] : 0.f;
             }
         };
 
         load_qtile(qt0 * BLKQ);
-        write_qtile(qt0 * BLKQ);
+        write_qtile(qt0 * BLKQ, qt0 & 1);
         __syncthreads();
 
         for (int qt = qt0; qt < n_q_tiles; qt++) {
             const int q_tb = qt * BLKQ;
+            const int cur = qt & 1;
             // skip q tiles fully below this wave's causal diagonal
             const bool wave_skip =
                 causal && (q_tb + BLKQ - 1 + causal_off < kvw);
@@ -420,9 +421,9 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
 #pragma unroll
                     for (int kk = 0; kk < DSTEPS; kk++) {
                         frag8 aqf = *reinterpret_cast<const frag8*>(
-                            swzb2<D>(q_lds, sub * 32 + l32, kk * 16 + hi * 8));
+                            swzb2<D>(q_lds[cur], sub * 32 + l32, kk * 16 + hi * 8));
                         frag8 adf = *reinterpret_cast<const frag8*>(
-                            swzb2<D>(do_lds, sub * 32 + l32, kk * 16 + hi * 8));
+                            swzb2<D>(do_lds[cur], sub * 32 + l32, kk * 16 + hi * 8));
                         st = mfma32b(aqf, bk[kk], st);
                         dp = mfma32b(adf, bv[kk], dp);
                     }
@@ -435,8 +436,8 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
                     for (int r = 0; r < 16; r++) {
                         int qrow = sub * 32 + crow32b(r, hi);
                         int qgl = q_tb + qrow;
-                        float ls = lse_lds[qrow];
-                        float dl = dl_lds[qrow];
+                        float ls = lse_lds[cur][qrow];
+                        float dl = dl_lds[cur][qrow];
                         bool vis = (qgl < Sq) && (kvg_lane < Skv) &&
                                    (!causal || kvg_lane <= qgl + causal_off);
                         float p = (vis && ls != INFINITY)
@@ -456,9 +457,9 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
 #pragma unroll
                         for (int n = 0; n < NDT; n++) {
                             frag8 bdo = *reinterpret_cast<const frag8*>(
-                                swzb2<BLKQ>(dot_lds, n * 32 + l32, ks * 16 + hi * 8));
+                                swzb2<BLKQ>(dot_lds[cur], n * 32 + l32, ks * 16 + hi * 8));
                             frag8 bq = *reinterpret_cast<const frag8*>(
-                                swzb2<BLKQ>(qt_lds, n * 32 + l32, ks * 16 + hi * 8));
+                                swzb2<BLKQ>(qt_lds[cur], n * 32 + l32, ks * 16 + hi * 8));
                             acc_dv[n] = mfma32b(apt, bdo, acc_dv[n]);
                             acc_dk[n] = mfma32b(ads, bq, acc_dk[n]);
                         }
@@ -467,15 +468,14 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
                 }
             }
 
-            __syncthreads();
             if (qt + 1 < n_q_tiles) {
-                // no register prefetch here (unlike fwd/dq): the dkv
-                // register file is already at capacity with two
-                // accumulator pairs + stationary K/V fragments
+                // no register prefetch (register file at capacity); the
+                // load+write go straight into the OTHER buffer, so only
+                // one barrier per tile
                 load_qtile(q_tb + BLKQ);
-                write_qtile(q_tb + BLKQ);
-                __syncthreads();
+                write_qtile(q_tb + BLKQ, cur ^ 1);
             }
+            __syncthreads();
         }
         __syncthreads();   // before the next g reuses the LDS tiles
     }

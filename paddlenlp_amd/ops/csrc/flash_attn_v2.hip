@@ -107,8 +107,11 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     constexpr int NDT = D / 32;       // 32-col d-tiles of O
     constexpr int KVT = FA2_BLKN / 32;  // kv sub-tiles (2)
 
-    __shared__ ushort_t k_lds[FA2_BLKN * D];   // row-major [kv][d], swizzled
-    __shared__ ushort_t vt_lds[D * FA2_BLKN];  // transposed [d][kv], swizzled
+    // double-buffered: compute tile t from buf[t&1] while staging t+1 into
+    // buf[(t+1)&1] -> ONE barrier per KV tile instead of two (the barrier
+    // drain before s_barrier is the dominant structural stall, guide §5)
+    __shared__ ushort_t k_lds[2][FA2_BLKN * D];
+    __shared__ ushort_t vt_lds[2][D * FA2_BLKN];
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -175,23 +178,24 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             sv1 = *reinterpret_cast<const short8v*>(v_ptr + (long long)(g0 + 1) * kv_row_stride + s_col);
         }
     };
-    auto write_tile = [&]() {
-        *reinterpret_cast<short8v*>(swz2<D>(k_lds, s_row0, s_col)) = sk0;
-        *reinterpret_cast<short8v*>(swz2<D>(k_lds, s_row0 + 1, s_col)) = sk1;
+    auto write_tile = [&](int buf) {
+        *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0, s_col)) = sk0;
+        *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
 #pragma unroll
         for (int j = 0; j < 8; j++) {
             unsigned p32 = ((unsigned)(unsigned short)sv0[j]) |
                            (((unsigned)(unsigned short)sv1[j]) << 16);
-            *reinterpret_cast<unsigned*>(swz2<FA2_BLKN>(vt_lds, s_col + j, s_row0)) = p32;
+            *reinterpret_cast<unsigned*>(swz2<FA2_BLKN>(vt_lds[buf], s_col + j, s_row0)) = p32;
         }
     };
 
     load_tile(0);
-    write_tile();
+    write_tile(0);
     __syncthreads();
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
         const int kv_base = kvt * FA2_BLKN;
+        const int cur = kvt & 1;
         // a wave whose q rows all precede this kv tile contributes nothing:
         // skip its compute but keep it in the staging barriers
         const bool wave_skip =
@@ -209,7 +213,7 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
 #pragma unroll
                 for (int nt = 0; nt < KVT; nt++) {
                     frag8 ak = *reinterpret_cast<const frag8*>(
-                        swz2<D>(k_lds, nt * 32 + l32, kk * 16 + hi * 8));
+                        swz2<D>(k_lds[cur], nt * 32 + l32, kk * 16 + hi * 8));
                     st[nt] = mfma32(ak, aq[kk], st[nt]);
                 }
             }
@@ -328,18 +332,19 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
 #pragma unroll
                 for (int n = 0; n < NDT; n++) {
                     frag8 bv = *reinterpret_cast<const frag8*>(
-                        swz2<FA2_BLKN>(vt_lds, n * 32 + l32, ks * 16 + hi * 8));
+                        swz2<FA2_BLKN>(vt_lds[cur], n * 32 + l32, ks * 16 + hi * 8));
                     acc_o[n] = mfma32(pa, bv, acc_o[n]);
                 }
             }
             __builtin_amdgcn_s_setprio(0);
         }
 
-        __syncthreads();                  // everyone done with k_lds/vt_lds
-        if (kvt + 1 < n_kv_tiles) {
-            write_tile();
-            __syncthreads();
-        }
+        // stage t+1 into the OTHER buffer while this tile's compute may
+        // still be in flight on other waves, then one barrier: it both
+        // publishes buf[cur^1] and guarantees every wave has finished
+        // reading buf[cur] before iteration t+1 overwrites it
+        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1);
+        __syncthreads();
     }
 
     // epilogue: O / l, LSE.  inv_l lives in the lane that owns q; O rows
