@@ -260,13 +260,101 @@ def save_unified_optimizer(optimizer, model, output_dir: str, topology, zero=Non
         dist.barrier()
 
 
+def _bcast_device():
+    import torch as _t
+
+    return _t.device("cuda") if (_t.cuda.is_available() and
+                                 dist.get_backend() == "nccl") else _t.device("cpu")
+
+
+def _plan_file_dispatch(ckpt_dir: str, files) -> Dict[str, int]:
+    """Cross-rank file-availability negotiation (reference dynamic load,
+    unified_checkpoint.py:1583 create_dispatch_table): returns
+    {fname: owner_rank} for every file MISSING on at least one rank
+    (owner = lowest rank that has it).  Files present everywhere load
+    locally; a file present nowhere raises on every rank coherently."""
+    if not (dist.is_available() and dist.is_initialized() and
+            dist.get_world_size() > 1):
+        return {}
+    local = {f for f in files if os.path.isfile(os.path.join(ckpt_dir, f))}
+    views = [None] * dist.get_world_size()
+    dist.all_gather_object(views, local)
+    dispatch = {}
+    for f in files:
+        have = [r for r, v in enumerate(views) if f in v]
+        if not have:
+            raise FileNotFoundError(f"checkpoint shard {f} missing on every rank")
+        if len(have) < dist.get_world_size():
+            dispatch[f] = have[0]
+    return dispatch
+
+
+def _dispatch_file(ckpt_dir: str, fname: str, owner: int, consume) -> None:
+    """Owner reads every tensor in `fname` and broadcasts; every rank's
+    `consume(key, tensor)` decides what to keep (reference
+    distributed_send_recv :1943 — broadcast used here since resume
+    bandwidth is not the constraint and it keeps the schedule uniform)."""
+    import torch as _t
+    from safetensors import safe_open
+
+    rank = dist.get_rank()
+    dev = _bcast_device()
+    if rank == owner:
+        with safe_open(os.path.join(ckpt_dir, fname), framework="pt", device="cpu") as f:
+            keys = list(f.keys())
+            metas = [(k, tuple(f.get_slice(k).get_shape()),
+                      str(f.get_slice(k).get_dtype())) for k in keys]
+            obj = [metas]
+            dist.broadcast_object_list(obj, src=owner)
+            for k in keys:
+                t = f.get_tensor(k).to(dev)
+                dist.broadcast(t, src=owner)
+                consume(k, t.cpu())
+    else:
+        obj = [None]
+        dist.broadcast_object_list(obj, src=owner)
+        _SAFE_DT = {"F32": _t.float32, "F16": _t.float16, "BF16": _t.bfloat16,
+                    "I64": _t.int64, "I32": _t.int32, "I8": _t.int8,
+                    "U8": _t.uint8, "BOOL": _t.bool,
+                    "torch.float32": _t.float32, "torch.float16": _t.float16,
+                    "torch.bfloat16": _t.bfloat16, "torch.int64": _t.int64}
+        for k, shape, dt in obj[0]:
+            t = _t.empty(shape, dtype=_SAFE_DT.get(dt, _t.float32), device=dev)
+            dist.broadcast(t, src=owner)
+            consume(k, t.cpu())
+
+
+def _load_json_negotiated(ckpt_dir: str, fname: str):
+    """Read a JSON sidecar, receiving it from another rank when the local
+    filesystem lacks it."""
+    path = os.path.join(ckpt_dir, fname)
+    have = os.path.isfile(path)
+    if not (dist.is_available() and dist.is_initialized() and
+            dist.get_world_size() > 1):
+        if not have:
+            return None
+        with open(path) as f:
+            return json.load(f)
+    views = [None] * dist.get_world_size()
+    dist.all_gather_object(views, have)
+    owners = [r for r, v in enumerate(views) if v]
+    if not owners:
+        return None
+    obj = [None]
+    if dist.get_rank() == owners[0]:
+        with open(path) as f:
+            obj = [json.load(f)]
+    dist.broadcast_object_list(obj, src=owners[0])
+    return obj[0]
+
+
 def load_unified_checkpoint(model, optimizer, ckpt_dir: str, topology, zero=None) -> None:
     """Load model weights + per-name optimizer states from a unified ckpt.
 
-    Each rank reads exactly the tensors it needs by name (mmap safe_open),
-    so the load works regardless of which rank wrote which shard — the
-    same-name-lookup is what makes same-config resume and (later)
-    cross-config resharding share one code path."""
+    Each rank reads exactly the tensors it needs by name (mmap safe_open);
+    when a shard file is absent from a rank's local filesystem the
+    lowest rank holding it reads and broadcasts (dynamic dispatch), so
+    resume works on non-shared storage too."""
     from safetensors import safe_open
 
     # ---- model weights ----
@@ -280,16 +368,18 @@ def load_unified_checkpoint(model, optimizer, ckpt_dir: str, topology, zero=None
         if isinstance(model, PretrainedModel):
             actions = type(model)._get_tensor_parallel_mappings(model.config, is_split=True)
 
-    index_file = os.path.join(ckpt_dir, SAFE_WEIGHTS_INDEX_NAME)
-    if os.path.isfile(index_file):
-        with open(index_file) as f:
-            weight_map = json.load(f)["weight_map"]
-    else:
-        weight_map = None
+    index = _load_json_negotiated(ckpt_dir, SAFE_WEIGHTS_INDEX_NAME)
+    weight_map = index["weight_map"] if index else None
 
     params = dict(model.named_parameters())
     buffers = dict(model.named_buffers())
     targets = {**params, **buffers}
+
+    def consume(key, t):
+        if key in targets:
+            if key in actions:
+                t = actions[key](t)
+            targets[key].data.copy_(t.to(targets[key].dtype))
 
     def load_tensor(f, key):
         t = f.get_tensor(key)
@@ -302,13 +392,28 @@ def load_unified_checkpoint(model, optimizer, ckpt_dir: str, topology, zero=None
         for key, fname in weight_map.items():
             if key in targets:
                 by_file.setdefault(fname, []).append(key)
-        for fname, keys in by_file.items():
-            with safe_open(os.path.join(ckpt_dir, fname), framework="pt", device="cpu") as f:
-                for key in keys:
-                    targets[key].data.copy_(load_tensor(f, key).to(targets[key].dtype))
+        # the dispatch plan must be identical on every rank: plan over the
+        # UNION of files any rank needs (PP stages need different shards)
+        union = sorted(by_file)
+        if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+            views = [None] * dist.get_world_size()
+            dist.all_gather_object(views, set(by_file))
+            union = sorted(set().union(*views))
+        dispatch = _plan_file_dispatch(ckpt_dir, union)
+        for fname in union:
+            if fname in dispatch:
+                _dispatch_file(ckpt_dir, fname, dispatch[fname], consume)
+            elif fname in by_file:
+                with safe_open(os.path.join(ckpt_dir, fname), framework="pt", device="cpu") as f:
+                    for key in by_file[fname]:
+                        targets[key].data.copy_(load_tensor(f, key).to(targets[key].dtype))
     else:
         single = os.path.join(ckpt_dir, SAFE_WEIGHTS_NAME)
-        if os.path.isfile(single):
+        dispatch = _plan_file_dispatch(ckpt_dir, [SAFE_WEIGHTS_NAME]) \
+            if not os.path.isfile(single) or dist.is_initialized() else {}
+        if SAFE_WEIGHTS_NAME in dispatch:
+            _dispatch_file(ckpt_dir, SAFE_WEIGHTS_NAME, dispatch[SAFE_WEIGHTS_NAME], consume)
+        elif os.path.isfile(single):
             with safe_open(single, framework="pt", device="cpu") as f:
                 for key in f.keys():
                     if key in targets:
@@ -319,19 +424,14 @@ def load_unified_checkpoint(model, optimizer, ckpt_dir: str, topology, zero=None
     # ---- optimizer states ----
     if optimizer is None:
         return
-    opt_index_file = os.path.join(ckpt_dir, SAFE_OPTIMIZER_INDEX_NAME)
-    if not os.path.isfile(opt_index_file):
+    opt_index = _load_json_negotiated(ckpt_dir, SAFE_OPTIMIZER_INDEX_NAME)
+    if opt_index is None:
         logger.warning(f"No optimizer index in {ckpt_dir}; optimizer starts fresh")
         return
-    with open(opt_index_file) as f:
-        opt_index = json.load(f)
     opt_map = opt_index["weight_map"]
     steps = opt_index.get("steps", {})
-    master_map = {}
-    m_index_file = os.path.join(ckpt_dir, SAFE_MASTER_WEIGHTS_INDEX_NAME)
-    if os.path.isfile(m_index_file):
-        with open(m_index_file) as f:
-            master_map = json.load(f)["weight_map"]
+    m_index = _load_json_negotiated(ckpt_dir, SAFE_MASTER_WEIGHTS_INDEX_NAME)
+    master_map = m_index["weight_map"] if m_index else {}
 
     names = _param_names(model)
     # which params does this rank's optimizer step? (ZeRO: owned only)
@@ -344,12 +444,45 @@ def load_unified_checkpoint(model, optimizer, ckpt_dir: str, topology, zero=None
             if name is not None:
                 wanted[name] = p
 
+    # which optimizer shard files does ANY rank need, and which of those
+    # are missing somewhere (dynamic dispatch set — must be identical on
+    # every rank so the broadcast schedule matches)
+    needed_files = set()
+    for name in wanted:
+        k1 = f"{name}/{MOMENT1}"
+        if k1 in opt_map:
+            needed_files.add(opt_map[k1])
+            needed_files.add(opt_map[f"{name}/{MOMENT2}"])
+        if name in master_map:
+            needed_files.add(master_map[name])
+    if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+        views = [None] * dist.get_world_size()
+        dist.all_gather_object(views, needed_files)
+        union_files = sorted(set().union(*views))
+    else:
+        union_files = sorted(needed_files)
+    dispatch = _plan_file_dispatch(ckpt_dir, union_files)
+
+    received: Dict[str, Dict[str, torch.Tensor]] = {}
+    wanted_keys = set()
+    for name in wanted:
+        wanted_keys.update((f"{name}/{MOMENT1}", f"{name}/{MOMENT2}", name))
+    for fname in sorted(dispatch):
+        bucket = received.setdefault(fname, {})
+
+        def consume(key, t, bucket=bucket):
+            if key in wanted_keys:
+                bucket[key] = t
+        _dispatch_file(ckpt_dir, fname, dispatch[fname], consume)
+
     open_files = {}
 
-    def get(fname):
+    def get_tensor(fname, key):
+        if fname in received:
+            return received[fname][key]
         if fname not in open_files:
             open_files[fname] = safe_open(os.path.join(ckpt_dir, fname), framework="pt", device="cpu")
-        return open_files[fname]
+        return open_files[fname].get_tensor(key)
 
     try:
         for name, p in wanted.items():
@@ -357,11 +490,11 @@ def load_unified_checkpoint(model, optimizer, ckpt_dir: str, topology, zero=None
             if key1 not in opt_map:
                 continue
             state = optimizer.state[p]
-            state["exp_avg"] = get(opt_map[key1]).get_tensor(key1).to(p.device)
-            state["exp_avg_sq"] = get(opt_map[key2]).get_tensor(key2).to(p.device)
+            state["exp_avg"] = get_tensor(opt_map[key1], key1).to(p.device)
+            state["exp_avg_sq"] = get_tensor(opt_map[key2], key2).to(p.device)
             state["step"] = steps.get(name, {}).get("step", 0)
             if name in master_map:
-                state["master"] = get(master_map[name]).get_tensor(name).to(p.device)
+                state["master"] = get_tensor(master_map[name], name).to(p.device)
             elif p.dtype in (torch.bfloat16, torch.float16) and getattr(optimizer, "master_weights", False):
                 state["master"] = p.detach().float().clone()
             else:

@@ -653,3 +653,60 @@ def _w_predict_dp_gather(rank, world):
 
 def test_predict_dp_gather():
     _run_workers(_w_predict_dp_gather)
+
+
+def _w_dynamic_ckpt_dispatch(rank, world):
+    """Dynamic checkpoint dispatch: rank 1's local dir has NO shard files
+    (non-shared filesystem); it must receive every tensor from rank 0
+    (reference load_unified_checkpoint_dynamically + distributed_send_recv)."""
+    import os
+    import shutil
+    import tempfile
+
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.trainer.optimizer import FusedAdamW
+    from paddlenlp_amd.trainer.unified_checkpoint import (
+        load_unified_checkpoint, save_unified_model, save_unified_optimizer)
+
+    topo = init_parallel_env(dp_degree=world, backend="gloo")
+
+    torch.manual_seed(77)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+    opt = FusedAdamW(model.parameters(), lr=1e-2, master_weights=False)
+    # one step so optimizer states exist
+    loss = model(torch.randn(4, 8)).sum()
+    loss.backward()
+    opt.step()
+
+    shared = "/tmp/pdnlp_dyn_ckpt_test"
+    if rank == 0:
+        shutil.rmtree(shared, ignore_errors=True)
+        os.makedirs(shared, exist_ok=True)
+    dist.barrier()
+    save_unified_model(model, shared, topo)
+    save_unified_optimizer(opt, model, shared, topo)
+    dist.barrier()
+
+    # rank 1 loads from an EMPTY directory (simulated lost local disk)
+    my_dir = shared if rank == 0 else tempfile.mkdtemp()
+
+    torch.manual_seed(99 + rank)
+    m2 = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+    o2 = FusedAdamW(m2.parameters(), lr=1e-2, master_weights=False)
+    load_unified_checkpoint(m2, o2, my_dir, topo)
+
+    for (n1, p1), (n2, p2) in zip(model.named_parameters(), m2.named_parameters()):
+        assert torch.allclose(p1, p2, atol=0), (rank, n1)
+    # optimizer moments arrived too
+    for p1, p2 in zip(model.parameters(), m2.parameters()):
+        s1, s2 = opt.state.get(p1, {}), o2.state.get(p2, {})
+        if "exp_avg" in s1:
+            assert "exp_avg" in s2, rank
+            assert torch.allclose(s1["exp_avg"], s2["exp_avg"]), rank
+    dist.barrier()
+    if rank == 0:
+        shutil.rmtree(shared, ignore_errors=True)
+
+
+def test_dynamic_checkpoint_dispatch():
+    _run_workers(_w_dynamic_ckpt_dispatch)
