@@ -156,7 +156,12 @@ def main():
         })
 
     def one_step(step_idx: int):
-        optimizer.zero_grad(set_to_none=True)
+        if zero is not None:
+            # flat-bucket engine: zero the persistent grad buffers and
+            # re-attach views so backward accumulates in place
+            zero.zero_grad()
+        else:
+            optimizer.zero_grad(set_to_none=True)
         for a in range(args.accum):
             batch = batches[(step_idx * args.accum + a) % len(batches)]
             if zero is not None and getattr(zero, "_overlap", False) \
