@@ -34,6 +34,7 @@ void launch_adamw(const AdamWChunk*, int, float, float, float, float, float, flo
 void launch_paged_decode_attn(const void*, const void*, const void*, const float*, const float*, const int*, const int*, void*, float*, int, int, int, int, int, int, int, float, int, hipStream_t);
 int paged_decode_nsplit(int, int);
 void launch_wint8_gemv(const void*, const void*, const float*, void*, int, int, int, hipStream_t);
+void launch_fp8_rowwise_quant(const void*, void*, float*, long long, int, hipStream_t);
 void launch_rope_cache_append(const void*, void*, void*, void*, float*, float*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, int, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -469,6 +470,19 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     return q_out;
 }
 
+std::vector<torch::Tensor> fp8_rowwise_quant(torch::Tensor x) {
+    CHECK_GPU(x); CHECK_BF16(x);
+    auto x2 = x.contiguous().view({-1, x.size(-1)});
+    long long R = x2.size(0);
+    int K = x2.size(1);
+    TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+    auto y = torch::empty({R, (long)K}, x.options().dtype(torch::kFloat8_e4m3fn));
+    auto scales = torch::empty({R}, x.options().dtype(torch::kFloat32));
+    launch_fp8_rowwise_quant(x2.data_ptr(), y.data_ptr(), scales.data_ptr<float>(),
+                             R, K, cur_stream());
+    return {y, scales};
+}
+
 torch::Tensor wint8_gemv(torch::Tensor x, torch::Tensor wq, torch::Tensor scale) {
     CHECK_GPU(x); CHECK_BF16(x);
     auto x2 = x.contiguous().view({-1, x.size(-1)});
@@ -523,4 +537,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("token_counts") = c10::nullopt,
           py::arg("k_scale") = c10::nullopt, py::arg("v_scale") = c10::nullopt);
     m.def("wint8_gemv", &wint8_gemv);
+    m.def("fp8_rowwise_quant", &fp8_rowwise_quant);
 }

@@ -9,6 +9,7 @@
 // dequantize in-register and dot against an x chunk staged in LDS as fp32
 // (same-address broadcast reads are conflict-free).
 #include "common.h"
+#include <hip/hip_fp8.h>
 
 #define WG_BLOCK 256
 #define ROWS_PER_WG 32   // output rows (N) per workgroup, 8 lanes per row
@@ -99,4 +100,57 @@ void launch_wint8_gemv(const void* x, const void* wq, const float* scale, void* 
         hipLaunchKernelGGL(wint8_gemv_kernel<16>, grid, dim3(WG_BLOCK), lds, stream,
                            (const ushort_t*)x, (const signed char*)wq, scale,
                            (ushort_t*)y, M, N, K, KC);
+}
+
+// ---------------------------------------------------------------------------
+// fused rowwise fp8 activation quant: one wave per row computes the row
+// absmax and casts to OCP e4m3fn in a single launch (the torch-op chain
+// costs ~6 kernel launches ≈ 40 us at decode shapes; this is one).
+// Feeds torch._scaled_mm rowwise-scaled fp8 GEMMs (reference fp8 A8W8
+// path, fused_transformer_layers.py FP8 variant).
+// ---------------------------------------------------------------------------
+#define FP8_E4M3_MAX 448.0f
+
+__global__ __launch_bounds__(256) void fp8_rowwise_quant_kernel(
+    const ushort_t* __restrict__ x,   // [R, K] bf16
+    unsigned char* __restrict__ y,    // [R, K] fp8 e4m3fn bit patterns
+    float* __restrict__ scales,       // [R]
+    long long R, int K) {
+    const int wave_in_block = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const long long row = (long long)blockIdx.x * 4 + wave_in_block;
+    if (row >= R) return;
+    const ushort_t* xr = x + row * K;
+    float amax = 0.f;
+    for (int k = lane * 8; k < K; k += 64 * 8) {
+        short8v v = *reinterpret_cast<const short8v*>(xr + k);
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+            amax = fmaxf(amax, fabsf(bf16_to_f32((ushort_t)v[j])));
+    }
+    amax = wave_reduce_max(amax);
+    const float sc = fmaxf(amax, 1e-8f) / FP8_E4M3_MAX;
+    const float inv = 1.0f / sc;
+    unsigned char* yr = y + row * K;
+    for (int k = lane * 8; k < K; k += 64 * 8) {
+        short8v v = *reinterpret_cast<const short8v*>(xr + k);
+        unsigned char out8[8];
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+            float f = bf16_to_f32((ushort_t)v[j]) * inv;
+            f = fminf(fmaxf(f, -FP8_E4M3_MAX), FP8_E4M3_MAX);
+            __hip_fp8_e4m3 h(f);
+            out8[j] = h.__x;
+        }
+        *reinterpret_cast<unsigned long long*>(yr + k) =
+            *reinterpret_cast<unsigned long long*>(out8);
+    }
+    if (lane == 0) scales[row] = sc;
+}
+
+void launch_fp8_rowwise_quant(const void* x, void* y, float* scales,
+                              long long R, int K, hipStream_t stream) {
+    long long blocks = (R + 3) / 4;
+    hipLaunchKernelGGL(fp8_rowwise_quant_kernel, dim3((unsigned)blocks), dim3(256),
+                       0, stream, (const ushort_t*)x, (unsigned char*)y, scales, R, K);
 }

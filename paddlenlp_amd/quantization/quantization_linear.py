@@ -48,9 +48,11 @@ def dequantize_int4(packed: torch.Tensor, scale: torch.Tensor,
 
 
 def quantize_fp8(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Per-tensor fp8 e4m3fn: w -> (fp8 w, fp32 scale scalar)."""
-    scale = (w.abs().amax().clamp(min=1e-8).float() / FP8_E4M3_MAX)
-    q = (w.float() / scale).clamp(-FP8_E4M3_MAX, FP8_E4M3_MAX).to(torch.float8_e4m3fn)
+    """Per-output-channel fp8 e4m3fn: w [out, in] -> (fp8 w, fp32 scale
+    [out]).  Channelwise scales feed rowwise-scaled _scaled_mm (hipBLASLt
+    fp8 MFMA) and quantize better than a single tensor scale."""
+    scale = (w.abs().amax(dim=1).clamp(min=1e-8).float() / FP8_E4M3_MAX)
+    q = (w.float() / scale[:, None]).clamp(-FP8_E4M3_MAX, FP8_E4M3_MAX).to(torch.float8_e4m3fn)
     return q, scale
 
 
@@ -62,15 +64,23 @@ def weight_only_linear(x: torch.Tensor, qweight: torch.Tensor, scale: torch.Tens
         orig_shape = x.shape
         x2 = x.reshape(-1, orig_shape[-1])
         if x2.is_cuda:
-            # dynamic per-tensor activation scale -> fp8 x fp8 scaled GEMM
-            x_scale = (x2.abs().amax().clamp(min=1e-8).float() / FP8_E4M3_MAX)
-            x8 = (x2.float() / x_scale).clamp(-FP8_E4M3_MAX, FP8_E4M3_MAX).to(torch.float8_e4m3fn)
+            # fused one-kernel rowwise activation quant (per-token scale) +
+            # rowwise/channelwise-scaled fp8 GEMM on the fp8 MFMA pipe
+            from ..ops.functional import _load_extension
+
+            C = _load_extension()
+            x8, x_scale = C.fp8_rowwise_quant(x2.contiguous())
+            if scale.dim() == 0:      # legacy per-tensor weight scale
+                w_scale = scale.reshape(1, 1).expand(1, qweight.shape[0]).contiguous()
+            else:
+                w_scale = scale.reshape(1, -1).contiguous()
             y = torch._scaled_mm(
-                x8, qweight.t(), scale_a=x_scale, scale_b=scale,
-                bias=None, out_dtype=x.dtype,
+                x8, qweight.t(), scale_a=x_scale.unsqueeze(1),
+                scale_b=w_scale, bias=None, out_dtype=x.dtype,
             )
         else:
-            y = x2 @ (qweight.float() * scale).t().to(x.dtype)
+            ws = scale if scale.dim() else scale.reshape(1)
+            y = x2 @ (qweight.float() * (ws[:, None] if ws.dim() else ws)).t().to(x.dtype)
         y = y.reshape(*orig_shape[:-1], -1)
     elif algo == "weight_only_int4":
         w = dequantize_int4(qweight, scale, x.dtype)
@@ -106,7 +116,8 @@ class QuantizationLinear(nn.Module):
         if quant_algo == "fp8":
             self.register_buffer("quant_weight",
                                  torch.zeros(out_features, in_features, dtype=torch.float8_e4m3fn))
-            self.register_buffer("quant_scale", torch.ones((), dtype=torch.float32))
+            # per-output-channel scales (rowwise-scaled fp8 GEMM)
+            self.register_buffer("quant_scale", torch.ones(out_features, dtype=torch.float32))
         elif quant_algo == "weight_only_int4":
             self.register_buffer("quant_weight",
                                  torch.zeros(out_features, in_features // 2, dtype=torch.uint8))
