@@ -10,6 +10,7 @@
 //
 // Cache layout: [num_blocks, block_size, Hk, D] bf16 per layer.
 #include "common.h"
+#include <cstdlib>
 
 #define PA_WAVES 4
 #define PA_BLOCK (PA_WAVES * 64)
@@ -453,7 +454,14 @@ int paged_decode_nsplit(int B, int Hk) {
     return nsplit;
 }
 
-// v2 (paged_attn_v2.hip): MFMA-tiled cooperative-staging decode kernel
+// v2/v3 (paged_attn_v2.hip): MFMA-tiled cooperative-staging decode kernels
+bool launch_paged_decode_attn3(const void* q, const void* k_cache, const void* v_cache,
+                               const float* k_scale, const float* v_scale,
+                               const int* block_table, const int* seq_lens, void* out,
+                               float* partials, int nsplit,
+                               int B, int Hq, int Hk, int D, int block_size,
+                               int max_blocks, float scale, int cache_mode,
+                               hipStream_t stream);
 bool launch_paged_decode_attn2(const void* q, const void* k_cache, const void* v_cache,
                                const float* k_scale, const float* v_scale,
                                const int* block_table, const int* seq_lens, void* out,
@@ -469,7 +477,14 @@ void launch_paged_decode_attn(const void* q, const void* k_cache, const void* v_
                               int B, int Hq, int Hk, int D, int block_size,
                               int max_blocks, float scale, int cache_mode,
                               hipStream_t stream) {
-    if (launch_paged_decode_attn2(q, k_cache, v_cache, k_scale, v_scale,
+    static const char* kver = getenv("PNLP_DECODE_KERNEL");  // "2" forces v2
+    const bool want_v2 = (kver != nullptr && kver[0] == '2');
+    if ((!want_v2 &&
+         launch_paged_decode_attn3(q, k_cache, v_cache, k_scale, v_scale,
+                                   block_table, seq_lens, out, partials, nsplit,
+                                   B, Hq, Hk, D, block_size, max_blocks, scale,
+                                   cache_mode, stream)) ||
+        launch_paged_decode_attn2(q, k_cache, v_cache, k_scale, v_scale,
                                   block_table, seq_lens, out, partials, nsplit,
                                   B, Hq, Hk, D, block_size, max_blocks, scale,
                                   cache_mode, stream)) {
