@@ -365,15 +365,28 @@ class FusedMultiTransformer(nn.Module):
         return t
 
     @torch.no_grad()
-    def quantize(self, algo: str = "fp8"):
+    def quantize(self, algo: str = "fp8", names=None):
         """Weight-only quantization of the projection weights (fp8 e4m3fn via
-        the gfx950 fp8 MFMA path, or per-channel int8).  Norm scales,
-        embeddings and the LM head stay in the compute dtype."""
+        the gfx950 fp8 MFMA path, or per-channel int8).  Norm scales and
+        embeddings stay in the compute dtype.
+
+        `names` restricts quantization, e.g. the bandwidth-bound subset
+        ("gate_up_weights", "down_weights", "lm_head") — at decode batch
+        sizes the skinny qkv/o GEMMs are latency-bound and fp8 does not
+        pay there (see profiles/r02_decode_throughput.md)."""
         from ..quantization import quantize_fp8, quantize_int8
 
         qfn = quantize_fp8 if algo == "fp8" else quantize_int8
         self._qw = {}
-        for name in ("qkv_weights", "out_proj_weights", "gate_up_weights", "down_weights"):
+        all_names = ("qkv_weights", "out_proj_weights", "gate_up_weights",
+                     "down_weights")
+        names = tuple(names) if names is not None else all_names
+        for name in names:
+            if name == "lm_head":
+                q, sc = qfn(self.lm_head.data)
+                self._qw["lm_head"] = [(q, sc)]
+                self.lm_head.data = self.lm_head.data.new_zeros(1)
+                continue
             plist = getattr(self, name)
             qs = []
             for w in plist:
@@ -385,7 +398,7 @@ class FusedMultiTransformer(nn.Module):
         return self
 
     def _mm(self, x, name: str, i: int):
-        if self.quant_algo is None:
+        if self.quant_algo is None or name not in self._qw:
             return x @ getattr(self, name)[i].t()
         from ..quantization import weight_only_linear
 
@@ -537,7 +550,13 @@ class FusedMultiTransformer(nn.Module):
                 act = ops.swiglu(gu) if gu.is_cuda else reference.swiglu(gu)
                 x = x + self._tp_reduce(self._mm(act, "down_weights", i))
         x = self._rms(x, self.final_norm)
-        logits = x[:, 0] @ self.lm_head.t()
+        if "lm_head" in self._qw:
+            from ..quantization import weight_only_linear
+
+            q, sc = self._qw["lm_head"][0]
+            logits = weight_only_linear(x[:, 0], q, sc, None, self.quant_algo)
+        else:
+            logits = x[:, 0] @ self.lm_head.t()
         return logits.float()
 
     @torch.no_grad()
@@ -580,6 +599,11 @@ class FusedMultiTransformer(nn.Module):
         x = self._rms(x, self.final_norm)
         idx = (prompt_lens.long() - 1).clamp(min=0)
         last = x[torch.arange(B, device=x.device), idx]
+        if "lm_head" in self._qw:
+            from ..quantization import weight_only_linear
+
+            q, sc = self._qw["lm_head"][0]
+            return weight_only_linear(last, q, sc, None, self.quant_algo).float()
         return (last @ self.lm_head.t()).float()
 
 

@@ -123,7 +123,7 @@ def main():
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--steps", type=int, default=64)
     p.add_argument("--warmup", type=int, default=8)
-    p.add_argument("--quant", default="")          # "" | fp8 | weight_only_int8
+    p.add_argument("--quant", default="")          # "" | fp8 | fp8_ffn | weight_only_int8
     p.add_argument("--graph", action="store_true")
     p.add_argument("--device-sched", action="store_true")
     p.add_argument("--cachekv", default="bf16", choices=["bf16", "int8", "int4"])
@@ -137,7 +137,11 @@ def main():
     eng = FusedMultiTransformer.from_llama(model, block_size=64, max_seq_len=8192).to(device)
     del model
     torch.cuda.empty_cache()
-    if args.quant:
+    if args.quant == "fp8_ffn":
+        # bandwidth-bound subset only: big FFN projections + lm head
+        eng.quantize("fp8", names=("gate_up_weights", "down_weights", "lm_head"))
+        torch.cuda.empty_cache()
+    elif args.quant:
         eng.quantize(args.quant)
         torch.cuda.empty_cache()
 
