@@ -35,6 +35,8 @@ void launch_paged_decode_attn(const void*, const void*, const void*, const float
 int paged_decode_nsplit(int, int);
 void launch_wint8_gemv(const void*, const void*, const float*, void*, int, int, int, hipStream_t);
 void launch_fp8_rowwise_quant(const void*, void*, float*, long long, int, hipStream_t);
+void launch_skinny_gemm(const void*, const void*, float*, void*, int, int, int, int, hipStream_t);
+int skinny_gemm_ksplit(int, int, int);
 void launch_rope_cache_append(const void*, void*, void*, void*, float*, float*, const int*, const int*, const float*, const float*, int, int, int, int, int, int, int, const int*, int, hipStream_t);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -470,6 +472,24 @@ torch::Tensor rope_cache_append(torch::Tensor qkv, torch::Tensor k_cache, torch:
     return q_out;
 }
 
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w, int64_t ksplit) {
+    CHECK_GPU(x); CHECK_BF16(x); CHECK_BF16(w);
+    auto x2 = x.contiguous().view({-1, x.size(-1)});
+    int M = x2.size(0), K = x2.size(1), N = w.size(0);
+    TORCH_CHECK(M <= 64, "skinny_gemm is for decode batches (M <= 64)");
+    TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+    int ks = skinny_gemm_ksplit(N, K, (int)ksplit);
+    auto partial = torch::empty({(long)ks, (long)M, (long)N},
+                                x.options().dtype(torch::kFloat32));
+    auto sizes = x.sizes().vec();
+    sizes.back() = N;
+    auto y = torch::empty(sizes, x.options());
+    launch_skinny_gemm(x2.data_ptr(), w.contiguous().data_ptr(),
+                       partial.data_ptr<float>(), y.data_ptr(),
+                       M, N, K, (int)ksplit, cur_stream());
+    return y;
+}
+
 std::vector<torch::Tensor> fp8_rowwise_quant(torch::Tensor x) {
     CHECK_GPU(x); CHECK_BF16(x);
     auto x2 = x.contiguous().view({-1, x.size(-1)});
@@ -538,4 +558,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("k_scale") = c10::nullopt, py::arg("v_scale") = c10::nullopt);
     m.def("wint8_gemv", &wint8_gemv);
     m.def("fp8_rowwise_quant", &fp8_rowwise_quant);
+    m.def("skinny_gemm", &skinny_gemm, py::arg("x"), py::arg("w"),
+          py::arg("ksplit") = 8);
 }

@@ -294,3 +294,14 @@ def test_fused_head_and_loss_fn_gpu(C):
     ref.backward()
     assert torch.allclose(hidden.grad.float(), hr.grad, atol=2e-2, rtol=5e-2)
     assert torch.allclose(weight.grad.float(), wr.grad, atol=2e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("M,N,K,ks", [(64, 512, 1024, 4), (17, 384, 512, 2)])
+def test_skinny_gemm(C, M, N, K, ks):
+    torch.manual_seed(0)
+    x = _bf16(torch.randn(M, K, device="cuda"))
+    w = _bf16(torch.randn(N, K, device="cuda"))
+    y = C.skinny_gemm(x, w, ks)
+    ref = x.float() @ w.float().t()
+    assert torch.allclose(y.float(), ref, atol=0.5, rtol=2e-2), \
+        (y.float() - ref).abs().max()
