@@ -442,6 +442,10 @@ class Trainer:
             if epoch >= num_train_epochs:
                 done = True
 
+        if (args.load_best_model_at_end and self.state.best_model_checkpoint
+                and os.path.isdir(self.state.best_model_checkpoint)):
+            self._load_best_model()
+
         self.control = self.callback_handler.on_train_end(args, self.state, self.control)
         metrics = speed_metrics(
             "train", start_time,
@@ -660,12 +664,34 @@ class Trainer:
                 if timer_msg:
                     logs["timers"] = timer_msg
             self.log(logs)
+        metrics = None
         if control.should_evaluate and self.eval_dataset is not None:
             metrics = self.evaluate()
             self.control = self.callback_handler.on_evaluate(args, state, control, metrics=metrics)
         if control.should_save:
             self._save_checkpoint(model)
+            self._update_best_checkpoint(metrics)
             self.control = self.callback_handler.on_save(args, state, self.control)
+
+    def _update_best_checkpoint(self, metrics):
+        """Track best checkpoint for load_best_model_at_end (reference
+        trainer.py:2464-2477)."""
+        args = self.args
+        if not args.metric_for_best_model or metrics is None:
+            return
+        key = args.metric_for_best_model
+        if not key.startswith("eval_"):
+            key = f"eval_{key}"
+        if key not in metrics:
+            return
+        value = metrics[key]
+        greater = args.greater_is_better
+        if greater is None:
+            greater = not key.endswith("loss")
+        best = self.state.best_metric
+        if best is None or (value > best if greater else value < best):
+            self.state.best_metric = value
+            self.state.best_model_checkpoint = self._checkpoint_dir()
 
     def log(self, logs: Dict[str, float]):
         logs["epoch"] = round(self.state.epoch, 4)
@@ -745,6 +771,12 @@ class Trainer:
             ),
             key=lambda d: int(pat.match(d).group(1)),
         )
+        # never rotate away the best checkpoint (load_best_model_at_end)
+        best = self.state.best_model_checkpoint
+        if best is not None:
+            best = os.path.basename(os.path.normpath(best))
+            if best in ckpts:
+                ckpts.remove(best)
         for d in ckpts[:-limit]:
             shutil.rmtree(os.path.join(self.args.output_dir, d), ignore_errors=True)
 
@@ -762,6 +794,29 @@ class Trainer:
                 torch.save(model.state_dict(), os.path.join(output_dir, "pytorch_model.bin"))
         if self.tokenizer is not None and self.args.process_index == 0:
             self.tokenizer.save_pretrained(output_dir)
+
+    def _load_best_model(self):
+        """Reload the best checkpoint's model weights at train end
+        (reference _load_best_model, trainer.py:2490-2516)."""
+        ckpt_dir = self.state.best_model_checkpoint
+        logger.info(
+            f"Loading best model from {ckpt_dir} "
+            f"({self.args.metric_for_best_model}={self.state.best_metric})")
+        model = unwrap_model(self.model)
+        if self.args.unified_checkpoint:
+            from .unified_checkpoint import load_unified_checkpoint
+
+            load_unified_checkpoint(model, None, ckpt_dir, self.topology)
+        else:
+            from safetensors.torch import load_file
+
+            path = os.path.join(ckpt_dir, "model.safetensors")
+            if os.path.isfile(path):
+                model.load_state_dict(load_file(path), strict=False)
+            else:
+                path = os.path.join(ckpt_dir, "pytorch_model.bin")
+                model.load_state_dict(
+                    torch.load(path, weights_only=True), strict=False)
 
     def _load_from_checkpoint(self, ckpt_dir: str):
         logger.info(f"Resuming from checkpoint {ckpt_dir}")

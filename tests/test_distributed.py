@@ -710,3 +710,54 @@ def _w_dynamic_ckpt_dispatch(rank, world):
 
 def test_dynamic_checkpoint_dispatch():
     _run_workers(_w_dynamic_ckpt_dispatch)
+
+
+def _w_zero2_engine_zero_grad(rank, world):
+    """The bench.py hot path: engine.zero_grad() (persistent grad views) +
+    overlap + stage2 must match optimizer.zero_grad(set_to_none=True)."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.zero import ZeroShardedEngine
+    from paddlenlp_amd.parallel.data_parallel import broadcast_parameters
+    from paddlenlp_amd.trainer.optimizer import FusedAdamW
+
+    topo = init_parallel_env(sharding_degree=world, backend="gloo")
+
+    def build():
+        torch.manual_seed(55)
+        return torch.nn.Sequential(
+            torch.nn.Linear(16, 32), torch.nn.Tanh(), torch.nn.Linear(32, 4))
+
+    torch.manual_seed(66)
+    xs = [torch.randn(world * 4, 16) for _ in range(4)]
+    ys = [torch.randn(world * 4, 4) for _ in range(4)]
+
+    def run(use_engine_zero):
+        model = build()
+        broadcast_parameters(model, topo.sharding_parallel_group)
+        opt = FusedAdamW(model.parameters(), lr=1e-2, master_weights=False)
+        zero = ZeroShardedEngine(model, opt, stage=2,
+                                 group=topo.sharding_parallel_group, bucket_mb=0)
+        zero.enable_overlap_comm()
+        for x, y in zip(xs, ys):
+            if use_engine_zero:
+                zero.zero_grad()
+            else:
+                opt.zero_grad(set_to_none=True)
+            zero.overlap_active = True
+            xl = x[rank * 4:(rank + 1) * 4]
+            yl = y[rank * 4:(rank + 1) * 4]
+            loss = ((model(xl) - yl) ** 2).mean()
+            loss.backward()
+            zero.reduce_gradients_and_step_pre()
+            opt.step()
+            zero.step_post()
+        return {n: p.detach().clone() for n, p in model.named_parameters()}
+
+    a = run(True)
+    b = run(False)
+    for n in a:
+        assert torch.allclose(a[n], b[n], atol=1e-6), (n,)
+
+
+def test_zero2_engine_zero_grad_parity():
+    _run_workers(_w_zero2_engine_zero_grad)

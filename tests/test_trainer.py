@@ -176,3 +176,31 @@ def test_reference_config_strings_parse():
     assert args.skip_profile_timer is False          # enable_timer
     assert args.sharding_overlap_comm is True        # enable_stage1_overlap
     assert args.scale_loss == 1024.0
+
+
+def test_load_best_model_at_end():
+    """Best checkpoint (by eval_loss) is tracked, survives rotation, and is
+    reloaded at train end (reference trainer.py:2464-2516)."""
+    with tempfile.TemporaryDirectory() as d:
+        ds = RandDS(n=32, s=16)
+        model = tiny_model()
+        args = make_args(
+            d, max_steps=12, save_steps=4, eval_steps=4,
+            evaluation_strategy="steps", load_best_model_at_end=True,
+            metric_for_best_model="eval_loss", save_total_limit=1,
+            learning_rate=5e-3,
+        )
+        tr = Trainer(model=model, args=args, train_dataset=ds,
+                     eval_dataset=RandDS(n=16, s=16, seed=1))
+        tr.train()
+        best = tr.state.best_model_checkpoint
+        assert best is not None and os.path.isdir(best), best
+        assert tr.state.best_metric is not None
+        # model weights must equal the best checkpoint's saved weights
+        from safetensors.torch import load_file
+
+        path = os.path.join(best, "model.safetensors")
+        saved = load_file(path)
+        live = {k: v for k, v in model.state_dict().items()}
+        for k, v in saved.items():
+            assert torch.equal(v, live[k].to(v.dtype)), k
