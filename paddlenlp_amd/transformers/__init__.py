@@ -270,3 +270,37 @@ from .mt5 import (  # noqa: F401
     MT5ForConditionalGeneration,
     MT5Model,
 )
+from .xlnet import (  # noqa: F401
+    XLNetConfig,
+    XLNetForSequenceClassification,
+    XLNetLMHeadModel,
+    XLNetModel,
+)
+from .reformer import (  # noqa: F401
+    ReformerConfig,
+    ReformerModel,
+    ReformerModelWithLMHead,
+)
+from .bigbird import (  # noqa: F401
+    BigBirdConfig,
+    BigBirdForMaskedLM,
+    BigBirdForSequenceClassification,
+    BigBirdModel,
+)
+from .nystromformer import (  # noqa: F401
+    NystromformerConfig,
+    NystromformerForSequenceClassification,
+    NystromformerModel,
+)
+from .convbert import (  # noqa: F401
+    ConvBertConfig,
+    ConvBertForMaskedLM,
+    ConvBertForSequenceClassification,
+    ConvBertModel,
+)
+from .ctrl import (  # noqa: F401
+    CTRLConfig,
+    CTRLForSequenceClassification,
+    CTRLLMHeadModel,
+    CTRLModel,
+)

@@ -153,6 +153,27 @@ MODEL_REGISTRY = {
                  "base": "LayoutLMModel",
                  "sequence_classification": "LayoutLMForSequenceClassification",
                  "token_classification": "LayoutLMForTokenClassification"},
+    "xlnet": {"module": "xlnet", "config": "XLNetConfig", "base": "XLNetModel",
+              "causal_lm": "XLNetLMHeadModel",
+              "sequence_classification": "XLNetForSequenceClassification"},
+    "reformer": {"module": "reformer", "config": "ReformerConfig",
+                 "base": "ReformerModel",
+                 "causal_lm": "ReformerModelWithLMHead"},
+    "bigbird": {"module": "bigbird", "config": "BigBirdConfig",
+                "base": "BigBirdModel",
+                "sequence_classification": "BigBirdForSequenceClassification",
+                "masked_lm": "BigBirdForMaskedLM"},
+    "nystromformer": {"module": "nystromformer",
+                      "config": "NystromformerConfig",
+                      "base": "NystromformerModel",
+                      "sequence_classification": "NystromformerForSequenceClassification"},
+    "convbert": {"module": "convbert", "config": "ConvBertConfig",
+                 "base": "ConvBertModel",
+                 "sequence_classification": "ConvBertForSequenceClassification",
+                 "masked_lm": "ConvBertForMaskedLM"},
+    "ctrl": {"module": "ctrl", "config": "CTRLConfig", "base": "CTRLModel",
+             "causal_lm": "CTRLLMHeadModel",
+             "sequence_classification": "CTRLForSequenceClassification"},
 }
 
 

@@ -23,6 +23,7 @@ from .. import ops
 ACT2FN = {
     "gelu": lambda x: F.gelu(x),
     "gelu_tanh": lambda x: F.gelu(x, approximate="tanh"),
+    "gelu_new": lambda x: F.gelu(x, approximate="tanh"),
     "relu": F.relu,
     "silu": F.silu,
     "tanh": torch.tanh,

@@ -1,0 +1,344 @@
+"""XLNet / Reformer / BigBird / Nystromformer / ConvBERT / CTRL families.
+
+Each test exercises the family's distinctive mechanism, not just shapes:
+XLNet relative-shift + mems, Reformer LSH bucketing vs dense, BigBird
+block-sparse vs dense agreement on short inputs, Nystromformer landmark
+approximation quality, ConvBERT dynamic-conv locality, CTRL causality.
+Reference behavior: paddlenlp/transformers/{xlnet,reformer,bigbird,
+nystromformer,convbert,ctrl}/modeling.py.
+"""
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from paddlenlp_amd.transformers import (
+    BigBirdConfig,
+    BigBirdForMaskedLM,
+    BigBirdModel,
+    ConvBertConfig,
+    ConvBertForMaskedLM,
+    ConvBertModel,
+    CTRLConfig,
+    CTRLLMHeadModel,
+    NystromformerConfig,
+    NystromformerForSequenceClassification,
+    NystromformerModel,
+    ReformerConfig,
+    ReformerModel,
+    ReformerModelWithLMHead,
+    XLNetConfig,
+    XLNetForSequenceClassification,
+    XLNetLMHeadModel,
+    XLNetModel,
+)
+
+V = 120
+
+
+# ---------------------------------------------------------------- xlnet
+def xlnet_cfg(**kw):
+    d = dict(vocab_size=V, hidden_size=48, num_hidden_layers=2,
+             num_attention_heads=4, intermediate_size=96, dropout=0.0,
+             mem_len=16)
+    d.update(kw)
+    return XLNetConfig(**d)
+
+
+def test_xlnet_forward_and_loss():
+    torch.manual_seed(0)
+    m = XLNetLMHeadModel(xlnet_cfg())
+    ids = torch.randint(0, V, (2, 12))
+    loss, logits = m(ids, labels=ids)
+    assert logits.shape == (2, 12, V)
+    loss.backward()
+    assert m.transformer.layers[0].rel_attn.r_r_bias.grad is not None
+
+
+def test_xlnet_rel_shift():
+    """rel_shift_bnij must pick diagonal-shifted entries: row i of the
+    output at column j equals input[i, klen + i - j] (relative index)."""
+    from paddlenlp_amd.transformers.xlnet.modeling import XLNetRelativeAttention
+
+    klen, qlen = 5, 3
+    # x[b,n,i,j] with j indexing positions klen..-qlen (length klen+qlen)
+    x = torch.arange(klen + qlen).float().view(1, 1, 1, -1).repeat(1, 1, qlen, 1)
+    for i in range(qlen):
+        x[0, 0, i] += 100 * i
+    out = XLNetRelativeAttention.rel_shift_bnij(x, klen)
+    assert out.shape == (1, 1, qlen, klen)
+    # relative distance of q_i to k_j is i + (klen - qlen) - j + qlen... the
+    # shift property we rely on: along a row, consecutive columns step by 1
+    # and rows are offset by one relative position
+    row_step = out[0, 0, 0, 1] - out[0, 0, 0, 0]
+    assert row_step == 1
+    assert (out[0, 0, 1, 0] - 100) == out[0, 0, 0, 0] - 1
+
+
+def test_xlnet_mems_recurrence():
+    """Two-segment forward with mems must differ from memless forward and
+    must match attending over the concatenation implicitly (mems grow)."""
+    torch.manual_seed(1)
+    m = XLNetModel(xlnet_cfg()).eval()
+    a = torch.randint(0, V, (1, 8))
+    b = torch.randint(0, V, (1, 8))
+    with torch.no_grad():
+        _, mems = m(a, use_mems=True)
+        assert len(mems) == 2 and mems[0].shape[1] == 8
+        out_with, _ = m(b, mems=mems, use_mems=True)
+        out_without, _ = m(b)
+    assert not torch.allclose(out_with, out_without, atol=1e-4)
+
+
+def test_xlnet_segment_encoding_sensitivity():
+    torch.manual_seed(2)
+    m = XLNetModel(xlnet_cfg()).eval()
+    ids = torch.randint(0, V, (1, 10))
+    seg_a = torch.zeros(1, 10, dtype=torch.long)
+    seg_b = torch.cat([torch.zeros(1, 5, dtype=torch.long),
+                       torch.ones(1, 5, dtype=torch.long)], dim=1)
+    with torch.no_grad():
+        oa, _ = m(ids, token_type_ids=seg_a)
+        ob, _ = m(ids, token_type_ids=seg_b)
+    # seg_embed init is small (std 0.02) so the shift is modest, but the
+    # same/diff segment one-hot must reach the scores
+    assert (oa - ob).abs().max() > 1e-6
+
+
+def test_xlnet_classifier():
+    m = XLNetForSequenceClassification(xlnet_cfg(num_labels=3))
+    ids = torch.randint(0, V, (2, 9))
+    mask = torch.ones(2, 9)
+    mask[1, 6:] = 0
+    loss, logits = m(ids, attention_mask=mask, labels=torch.tensor([0, 2]))
+    assert logits.shape == (2, 3)
+    loss.backward()
+
+
+# -------------------------------------------------------------- reformer
+def ref_cfg(**kw):
+    d = dict(vocab_size=V, hidden_size=32, num_attention_heads=2,
+             attention_head_size=16, feed_forward_size=64,
+             attn_layers=("local", "lsh"), lsh_attn_chunk_length=8,
+             local_attn_chunk_length=8, num_hashes=2, num_buckets=4,
+             axial_pos_shape=(4, 8), hidden_dropout_prob=0.0)
+    d.update(kw)
+    return ReformerConfig(**d)
+
+
+def test_reformer_lm_forward_backward():
+    torch.manual_seed(0)
+    m = ReformerModelWithLMHead(ref_cfg())
+    ids = torch.randint(0, V, (2, 24))     # padded to 32 internally
+    loss, logits = m(ids, labels=ids)
+    assert logits.shape == (2, 24, V)
+    loss.backward()
+
+
+def test_reformer_causality():
+    """Causal (is_decoder) reformer with LOCAL layers: output at position t
+    must not change when future tokens change.  (LSH layers are checked via
+    the bucket test below — their chunk composition legitimately depends on
+    the global bucket sort, so strict prefix-invariance only holds for the
+    local flavor, as in the original.)"""
+    torch.manual_seed(3)
+    m = ReformerModel(ref_cfg(attn_layers=("local", "local"))).eval()
+    a = torch.randint(0, V, (1, 32))
+    b = a.clone()
+    b[0, 24:] = (b[0, 24:] + 7) % V
+    with torch.no_grad():
+        oa = m(a)
+        ob = m(b)
+    assert torch.allclose(oa[0, :16], ob[0, :16], atol=1e-5)
+    assert not torch.allclose(oa[0, 24:], ob[0, 24:], atol=1e-4)
+
+
+def test_reformer_lsh_groups_similar_vectors():
+    """Identical hidden states must land in the same LSH bucket."""
+    from paddlenlp_amd.transformers.reformer.modeling import LSHSelfAttention
+
+    cfg = ref_cfg()
+    attn = LSHSelfAttention(cfg)
+    qk = torch.randn(1, cfg.num_attention_heads, 16, cfg.attention_head_size)
+    qk[0, :, 5] = qk[0, :, 11]            # duplicate vector
+    g = torch.Generator().manual_seed(0)
+    buckets = attn._hash(F.normalize(qk, dim=-1), g)
+    assert (buckets[0, :, :, 5] == buckets[0, :, :, 11]).all()
+
+
+def test_reformer_axial_positions_cover_sequence():
+    from paddlenlp_amd.transformers.reformer.modeling import (
+        AxialPositionEmbeddings,
+    )
+
+    cfg = ref_cfg()
+    ape = AxialPositionEmbeddings(cfg)
+    emb = ape(30, torch.float32)
+    assert emb.shape == (30, cfg.hidden_size)
+    assert not torch.allclose(emb[0], emb[9])
+
+
+# --------------------------------------------------------------- bigbird
+def bb_cfg(**kw):
+    d = dict(vocab_size=V, hidden_size=32, num_hidden_layers=2,
+             num_attention_heads=2, intermediate_size=64, block_size=4,
+             num_global_blocks=1, num_random_blocks=1, num_sliding_blocks=3,
+             max_position_embeddings=128, hidden_dropout_prob=0.0)
+    d.update(kw)
+    return BigBirdConfig(**d)
+
+
+def test_bigbird_mlm_forward_backward():
+    torch.manual_seed(0)
+    m = BigBirdForMaskedLM(bb_cfg())
+    ids = torch.randint(0, V, (2, 40))
+    loss, logits = m(ids, labels=ids)
+    assert logits.shape == (2, 40, V)
+    loss.backward()
+
+
+def test_bigbird_sparse_matches_dense_when_pattern_covers_all():
+    """With few blocks the sparse pattern degenerates to dense attention —
+    the two paths must agree."""
+    torch.manual_seed(4)
+    from paddlenlp_amd.transformers.bigbird.modeling import (
+        BigBirdSparseAttention,
+    )
+
+    cfg = bb_cfg()
+    attn = BigBirdSparseAttention(cfg).eval()
+    x = torch.randn(1, 16, 32)             # 4 blocks <= g+w+r=5 -> dense
+    with torch.no_grad():
+        out = attn(x)
+        # dense reference
+        q = attn.query(x).view(1, 16, 2, 16).transpose(1, 2)
+        k = attn.key(x).view(1, 16, 2, 16).transpose(1, 2)
+        v = attn.value(x).view(1, 16, 2, 16).transpose(1, 2)
+        ref = F.softmax(q @ k.transpose(-1, -2) / math.sqrt(16), -1) @ v
+        ref = attn.out(ref.transpose(1, 2).reshape(1, 16, 32))
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_bigbird_global_block_sees_everything():
+    """Perturbing the LAST block must change the GLOBAL (first) block's
+    output but not a distant non-adjacent middle block's output when the
+    last block is outside its window+random set."""
+    torch.manual_seed(5)
+    cfg = bb_cfg(num_random_blocks=0, num_hidden_layers=1)
+    m = BigBirdModel(cfg).eval()
+    a = torch.randint(0, V, (1, 64))       # 16 blocks: sparse path
+    b = a.clone()
+    b[0, -4:] = (b[0, -4:] + 3) % V
+    with torch.no_grad():
+        oa, _ = m(a)
+        ob, _ = m(b)
+    # global block (positions 0-3) attends densely -> must change
+    assert not torch.allclose(oa[0, :4], ob[0, :4], atol=1e-5)
+    # middle block (positions 24-27) has window 5,6,7 + global 0: unchanged
+    assert torch.allclose(oa[0, 24:28], ob[0, 24:28], atol=1e-5)
+
+
+# --------------------------------------------------------- nystromformer
+def test_nystromformer_forward_and_approximation():
+    torch.manual_seed(0)
+    cfg = NystromformerConfig(
+        vocab_size=V, hidden_size=32, num_hidden_layers=1,
+        num_attention_heads=2, intermediate_size=64, num_landmarks=8,
+        conv_kernel_size=0, max_position_embeddings=128,
+        hidden_dropout_prob=0.0)
+    m = NystromformerModel(cfg).eval()
+    ids = torch.randint(0, V, (1, 64))
+    seq, pooled = m(ids)
+    assert seq.shape == (1, 64, 32)
+
+    # landmark approximation should be close to dense attention for the
+    # attention submodule on smooth inputs
+    from paddlenlp_amd.transformers.nystromformer.modeling import (
+        NystromAttention,
+    )
+
+    attn = NystromAttention(cfg).eval()
+    x = torch.randn(1, 64, 32) * 0.1
+    with torch.no_grad():
+        approx = attn(x)
+        scale = 1 / math.sqrt(math.sqrt(16))
+        q = attn.query(x).view(1, 64, 2, 16).transpose(1, 2) * scale
+        k = attn.key(x).view(1, 64, 2, 16).transpose(1, 2) * scale
+        v = attn.value(x).view(1, 64, 2, 16).transpose(1, 2)
+        dense = F.softmax(q @ k.transpose(-1, -2), -1) @ v
+        dense = attn.out(dense.transpose(1, 2).reshape(1, 64, 32))
+    rel = (approx - dense).norm() / dense.norm()
+    assert rel < 0.05, rel.item()
+
+
+def test_nystromformer_classifier_backward():
+    cfg = NystromformerConfig(
+        vocab_size=V, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=2, intermediate_size=64, num_landmarks=8,
+        max_position_embeddings=128, num_labels=2)
+    m = NystromformerForSequenceClassification(cfg)
+    ids = torch.randint(0, V, (2, 32))
+    loss, logits = m(ids, labels=torch.tensor([0, 1]))
+    loss.backward()
+    assert logits.shape == (2, 2)
+
+
+# -------------------------------------------------------------- convbert
+def cb_cfg(**kw):
+    d = dict(vocab_size=V, hidden_size=32, num_hidden_layers=2,
+             num_attention_heads=4, intermediate_size=64, head_ratio=2,
+             conv_kernel_size=5, max_position_embeddings=64,
+             hidden_dropout_prob=0.0)
+    d.update(kw)
+    return ConvBertConfig(**d)
+
+
+def test_convbert_forward_backward():
+    torch.manual_seed(0)
+    m = ConvBertForMaskedLM(cb_cfg())
+    ids = torch.randint(0, V, (2, 20))
+    loss, logits = m(ids, labels=ids)
+    assert logits.shape == (2, 20, V)
+    loss.backward()
+
+
+def test_convbert_head_split():
+    """head_ratio=2 halves the self-attention heads; the mixed layer's
+    dense input is 2x the reduced head width."""
+    from paddlenlp_amd.transformers.convbert.modeling import (
+        ConvBertMixedAttention,
+    )
+
+    attn = ConvBertMixedAttention(cb_cfg())
+    assert attn.num_heads == 2             # 4 heads / ratio 2
+    assert attn.dense.in_features == 2 * attn.all_head
+
+
+# ------------------------------------------------------------------ ctrl
+def test_ctrl_lm_and_causality():
+    torch.manual_seed(0)
+    cfg = CTRLConfig(vocab_size=V, hidden_size=32, num_hidden_layers=2,
+                     num_attention_heads=4, intermediate_size=64,
+                     resid_pdrop=0.0, embd_pdrop=0.0)
+    m = CTRLLMHeadModel(cfg).eval()
+    a = torch.randint(0, V, (1, 16))
+    b = a.clone()
+    b[0, 12:] = (b[0, 12:] + 5) % V
+    with torch.no_grad():
+        la = m(a)
+        lb = m(b)
+    assert torch.allclose(la[0, :12], lb[0, :12], atol=1e-5)
+    loss, _ = m(a, labels=a)
+    loss.backward()
+    # tied head: embedding grad flows through lm_head
+    assert m.transformer.w.weight.grad is not None
+
+
+# ------------------------------------------------------------- registry
+def test_new_families_in_auto_registry():
+    from paddlenlp_amd.transformers.auto.registry import MODEL_REGISTRY
+
+    for fam in ("xlnet", "reformer", "bigbird", "nystromformer",
+                "convbert", "ctrl"):
+        assert fam in MODEL_REGISTRY, fam
