@@ -153,8 +153,9 @@ MODEL_REGISTRY = {
                  "base": "LayoutLMModel",
                  "sequence_classification": "LayoutLMForSequenceClassification",
                  "token_classification": "LayoutLMForTokenClassification"},
+    # XLNet's LM head is the permutation-LM objective (bidirectional
+    # content stream), not a standard causal LM: no causal_lm registration
     "xlnet": {"module": "xlnet", "config": "XLNetConfig", "base": "XLNetModel",
-              "causal_lm": "XLNetLMHeadModel",
               "sequence_classification": "XLNetForSequenceClassification"},
     "reformer": {"module": "reformer", "config": "ReformerConfig",
                  "base": "ReformerModel",

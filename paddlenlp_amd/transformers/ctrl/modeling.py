@@ -13,9 +13,10 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ... import ops
+from ...generation import GenerationMixin
 from ..configuration_utils import PretrainedConfig
 from ..encoder import init_encoder_weights
+from ..gpt.modeling import GPTAttention
 from ..model_utils import PretrainedModel
 
 __all__ = ["CTRLConfig", "CTRLModel", "CTRLLMHeadModel",
@@ -60,23 +61,22 @@ class CTRLBlock(nn.Module):
         h = config.hidden_size
         self.ln1 = nn.LayerNorm(h, eps=config.layer_norm_epsilon)
         self.ln2 = nn.LayerNorm(h, eps=config.layer_norm_epsilon)
-        self.num_heads = config.num_attention_heads
-        self.head_dim = config.head_dim
-        self.qkv = nn.Linear(h, 3 * h)
-        self.proj = nn.Linear(h, h)
+        self.attn = GPTAttention(config)
         self.fc_in = nn.Linear(h, config.intermediate_size)
         self.fc_out = nn.Linear(config.intermediate_size, h)
         self.dropout = nn.Dropout(config.resid_pdrop)
 
-    def forward(self, x):
-        B, S, H = x.shape
-        q, k, v = self.qkv(self.ln1(x)).chunk(3, dim=-1)
-        shape = (B, S, self.num_heads, self.head_dim)
-        out = ops.flash_attention(q.view(shape), k.view(shape), v.view(shape),
-                                  causal=True)
-        x = x + self.dropout(self.proj(out.reshape(B, S, H)))
+    def forward(self, x, past_key_value=None, use_cache=False):
+        out = self.attn(self.ln1(x), past_key_value, use_cache)
+        present = None
+        if use_cache:
+            out, present = out
+        x = x + self.dropout(out)
         y = self.fc_out(F.relu(self.fc_in(self.ln2(x))))
-        return x + self.dropout(y)
+        x = x + self.dropout(y)
+        if use_cache:
+            return x, present
+        return x
 
 
 class CTRLPretrainedModel(PretrainedModel):
@@ -96,6 +96,7 @@ class CTRLModel(CTRLPretrainedModel):
             [CTRLBlock(config) for _ in range(config.num_hidden_layers)])
         self.layernorm = nn.LayerNorm(config.hidden_size,
                                       eps=config.layer_norm_epsilon)
+        self.init_weights()
 
     def get_input_embeddings(self):
         return self.w
@@ -110,18 +111,31 @@ class CTRLModel(CTRLPretrainedModel):
         pe[:, 1::2] = ang.cos()
         return pe.to(dtype)
 
-    def forward(self, input_ids, attention_mask=None):
+    def forward(self, input_ids, attention_mask=None, past_key_values=None,
+                use_cache=False):
         B, S = input_ids.shape
+        past_len = (past_key_values[0][0].shape[1]
+                    if past_key_values is not None else 0)
         # reference scales token embeddings by sqrt(d) before adding PE
         x = self.w(input_ids) * math.sqrt(self.config.hidden_size)
-        x = x + self._sinusoid(S, x.device, x.dtype)
-        x = self.dropout(x)
-        for block in self.h:
-            x = block(x)
-        return self.layernorm(x)
+        pe = self._sinusoid(past_len + S, x.device, x.dtype)[past_len:]
+        x = self.dropout(x + pe)
+        presents = [] if use_cache else None
+        for i, block in enumerate(self.h):
+            past = past_key_values[i] if past_key_values is not None else None
+            out = block(x, past, use_cache)
+            if use_cache:
+                x, present = out
+                presents.append(present)
+            else:
+                x = out
+        x = self.layernorm(x)
+        if use_cache:
+            return x, presents
+        return x
 
 
-class CTRLLMHeadModel(CTRLPretrainedModel):
+class CTRLLMHeadModel(CTRLPretrainedModel, GenerationMixin):
     _tied_weights_keys = ["lm_head.weight"]
 
     def __init__(self, config: CTRLConfig):
@@ -129,19 +143,24 @@ class CTRLLMHeadModel(CTRLPretrainedModel):
         self.transformer = CTRLModel(config)
         self.lm_head = nn.Linear(config.hidden_size, config.vocab_size,
                                  bias=True)
+        self.init_weights()
         self.lm_head.weight = self.transformer.w.weight
 
-    def forward(self, input_ids, attention_mask=None, labels=None):
-        logits = self.lm_head(self.transformer(input_ids, attention_mask))
+    def forward(self, input_ids, attention_mask=None, labels=None,
+                past_key_values=None, use_cache=False, **kwargs):
+        out = self.transformer(input_ids, attention_mask,
+                               past_key_values, use_cache)
+        presents = None
+        if use_cache:
+            out, presents = out
+        logits = self.lm_head(out)
         if labels is not None:
+            # labels are pre-shifted by the caller (framework convention)
             loss = F.cross_entropy(
-                logits[:, :-1].reshape(-1, self.config.vocab_size),
-                labels[:, 1:].reshape(-1), ignore_index=-100)
-            return loss, logits
-        return logits
-
-    def prepare_inputs_for_generation(self, input_ids, **kwargs):
-        return {"input_ids": input_ids}
+                logits.reshape(-1, self.config.vocab_size),
+                labels.reshape(-1), ignore_index=-100)
+            return (loss, logits) if not use_cache else (loss, logits, presents)
+        return logits if not use_cache else (logits, presents)
 
 
 class CTRLForSequenceClassification(CTRLPretrainedModel):
@@ -150,6 +169,7 @@ class CTRLForSequenceClassification(CTRLPretrainedModel):
         self.transformer = CTRLModel(config)
         self.classifier = nn.Linear(config.hidden_size, config.num_labels,
                                     bias=False)
+        self.init_weights()
 
     def forward(self, input_ids, attention_mask=None, labels=None):
         seq = self.transformer(input_ids, attention_mask)

@@ -21,6 +21,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ...generation import GenerationMixin
 from ..configuration_utils import PretrainedConfig
 from ..encoder import ACT2FN, init_encoder_weights
 from ..model_utils import PretrainedModel
@@ -285,17 +286,31 @@ class ReformerModel(ReformerPretrainedModel):
         return out[:, :S0]
 
 
-class ReformerModelWithLMHead(ReformerPretrainedModel):
+class ReformerModelWithLMHead(ReformerPretrainedModel, GenerationMixin):
+    """Causal LM head.  `use_cache` keeps the token prefix (a recompute
+    cache): LSH bucketing and chunk composition depend on the whole
+    sequence, so there is no valid token-level KV cache — each decode
+    step re-runs the prefix, exactly as the reference does for its LSH
+    layers."""
+
     def __init__(self, config: ReformerConfig):
         super().__init__(config)
         self.reformer = ReformerModel(config)
         self.lm_head = nn.Linear(2 * config.hidden_size, config.vocab_size)
 
-    def forward(self, input_ids, attention_mask=None, labels=None):
+    def forward(self, input_ids, attention_mask=None, labels=None,
+                past_key_values=None, use_cache=False, **kwargs):
+        S_new = input_ids.shape[1]
+        if past_key_values is not None:
+            input_ids = torch.cat([past_key_values, input_ids], dim=1)
         logits = self.lm_head(self.reformer(input_ids, attention_mask))
+        if past_key_values is not None:
+            logits = logits[:, -S_new:]
+        present = input_ids if use_cache else None
         if labels is not None:
+            # labels are pre-shifted by the caller (framework convention)
             loss = F.cross_entropy(
-                logits[:, :-1].reshape(-1, self.config.vocab_size),
-                labels[:, 1:].reshape(-1), ignore_index=-100)
-            return loss, logits
-        return logits
+                logits.reshape(-1, self.config.vocab_size),
+                labels.reshape(-1), ignore_index=-100)
+            return (loss, logits) if not use_cache else (loss, logits, present)
+        return logits if not use_cache else (logits, present)

@@ -220,6 +220,7 @@ class XLNetLMHeadModel(XLNetPretrainedModel):
                                          attention_mask, mems, use_mems)
         logits = self.lm_loss(seq)
         if labels is not None:
+            # labels pre-aligned by the caller (perm-LM targets / shifted ids)
             loss = F.cross_entropy(logits.view(-1, self.config.vocab_size),
                                    labels.view(-1), ignore_index=-100)
             return loss, logits

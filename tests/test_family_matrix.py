@@ -84,6 +84,14 @@ TINY = dict(
                attn_layer_offset=1, expert_layer_period=2,
                expert_layer_offset=0, num_experts=4, num_experts_per_tok=2,
                mamba_d_state=8, max_position_embeddings=64),
+    ctrl=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
+              num_hidden_layers=2, num_attention_heads=4,
+              resid_pdrop=0.0, embd_pdrop=0.0),
+    reformer=dict(vocab_size=96, hidden_size=32, num_attention_heads=2,
+                  attention_head_size=16, feed_forward_size=64,
+                  attn_layers=("local", "lsh"), lsh_attn_chunk_length=8,
+                  local_attn_chunk_length=8, num_hashes=2, num_buckets=4,
+                  axial_pos_shape=(4, 8), hidden_dropout_prob=0.0),
 )
 # gpt2 aliases gpt; skip the alias row
 FAMILIES = sorted(mt for mt, entry in MODEL_REGISTRY.items()
