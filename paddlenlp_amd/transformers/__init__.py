@@ -304,3 +304,23 @@ from .ctrl import (  # noqa: F401
     CTRLLMHeadModel,
     CTRLModel,
 )
+from .glm import (  # noqa: F401
+    GLMConfig,
+    GLMForConditionalGeneration,
+    GLMModel,
+)
+from .mobilebert import (  # noqa: F401
+    MobileBertConfig,
+    MobileBertForSequenceClassification,
+    MobileBertModel,
+)
+from .squeezebert import (  # noqa: F401
+    SqueezeBertConfig,
+    SqueezeBertForSequenceClassification,
+    SqueezeBertModel,
+)
+from .gau_alpha import (  # noqa: F401
+    GAUAlphaConfig,
+    GAUAlphaForSequenceClassification,
+    GAUAlphaModel,
+)

@@ -175,6 +175,17 @@ MODEL_REGISTRY = {
     "ctrl": {"module": "ctrl", "config": "CTRLConfig", "base": "CTRLModel",
              "causal_lm": "CTRLLMHeadModel",
              "sequence_classification": "CTRLForSequenceClassification"},
+    "glm": {"module": "glm", "config": "GLMConfig", "base": "GLMModel",
+            "causal_lm": "GLMForConditionalGeneration"},
+    "mobilebert": {"module": "mobilebert", "config": "MobileBertConfig",
+                   "base": "MobileBertModel",
+                   "sequence_classification": "MobileBertForSequenceClassification"},
+    "squeezebert": {"module": "squeezebert", "config": "SqueezeBertConfig",
+                    "base": "SqueezeBertModel",
+                    "sequence_classification": "SqueezeBertForSequenceClassification"},
+    "gau_alpha": {"module": "gau_alpha", "config": "GAUAlphaConfig",
+                  "base": "GAUAlphaModel",
+                  "sequence_classification": "GAUAlphaForSequenceClassification"},
 }
 
 

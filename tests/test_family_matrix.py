@@ -92,6 +92,9 @@ TINY = dict(
                   attn_layers=("local", "lsh"), lsh_attn_chunk_length=8,
                   local_attn_chunk_length=8, num_hashes=2, num_buckets=4,
                   axial_pos_shape=(4, 8), hidden_dropout_prob=0.0),
+    glm=dict(vocab_size=96, hidden_size=32, num_hidden_layers=2,
+             num_attention_heads=4, max_position_embeddings=64,
+             hidden_dropout_prob=0.0),
 )
 # gpt2 aliases gpt; skip the alias row
 FAMILIES = sorted(mt for mt, entry in MODEL_REGISTRY.items()
