@@ -324,3 +324,20 @@ from .gau_alpha import (  # noqa: F401
     GAUAlphaForSequenceClassification,
     GAUAlphaModel,
 )
+from .deberta_v2 import (  # noqa: F401
+    DebertaV2Config,
+    DebertaV2ForMaskedLM,
+    DebertaV2ForSequenceClassification,
+    DebertaV2Model,
+)
+from .chinesebert import (  # noqa: F401
+    ChineseBertConfig,
+    ChineseBertForSequenceClassification,
+    ChineseBertModel,
+)
+from .funnel import (  # noqa: F401
+    FunnelConfig,
+    FunnelForSequenceClassification,
+    FunnelForTokenClassification,
+    FunnelModel,
+)

@@ -186,6 +186,17 @@ MODEL_REGISTRY = {
     "gau_alpha": {"module": "gau_alpha", "config": "GAUAlphaConfig",
                   "base": "GAUAlphaModel",
                   "sequence_classification": "GAUAlphaForSequenceClassification"},
+    "deberta-v2": {"module": "deberta_v2", "config": "DebertaV2Config",
+                   "base": "DebertaV2Model",
+                   "sequence_classification": "DebertaV2ForSequenceClassification",
+                   "masked_lm": "DebertaV2ForMaskedLM"},
+    "chinesebert": {"module": "chinesebert", "config": "ChineseBertConfig",
+                    "base": "ChineseBertModel",
+                    "sequence_classification": "ChineseBertForSequenceClassification"},
+    "funnel": {"module": "funnel", "config": "FunnelConfig",
+               "base": "FunnelModel",
+               "sequence_classification": "FunnelForSequenceClassification",
+               "token_classification": "FunnelForTokenClassification"},
 }
 
 

@@ -1,0 +1,159 @@
+"""DeBERTa-v2 / ChineseBERT / Funnel families.
+
+Distinctive mechanisms: v2 log-bucket relative positions, ChineseBERT's
+three-view (word/pinyin/glyph) fusion embedding, Funnel's pool-query
+blocks + upsampling decoder.  Reference behavior:
+paddlenlp/transformers/{deberta_v2,chinesebert,funnel}/modeling.py.
+"""
+import torch
+
+from paddlenlp_amd.transformers import (
+    ChineseBertConfig,
+    ChineseBertForSequenceClassification,
+    ChineseBertModel,
+    DebertaV2Config,
+    DebertaV2ForMaskedLM,
+    DebertaV2Model,
+    FunnelConfig,
+    FunnelForSequenceClassification,
+    FunnelForTokenClassification,
+    FunnelModel,
+)
+
+V = 96
+
+
+# ------------------------------------------------------------ deberta_v2
+def test_log_bucket_positions():
+    from paddlenlp_amd.transformers.deberta_v2.modeling import (
+        make_log_bucket_position,
+    )
+
+    rel = torch.arange(-63, 64)
+    b = make_log_bucket_position(rel, 16, 64)
+    # near field exact
+    assert (b[63 - 7:63 + 8] == rel[63 - 7:63 + 8]).all()
+    # far field compressed: strictly fewer distinct buckets than positions
+    far = b[63 + 8:]
+    assert far.max() <= 15 and len(far.unique()) < len(far)
+    # antisymmetric
+    assert (b + b.flip(0) == 0).all()
+
+
+def test_deberta_v2_forward_backward_and_conv():
+    torch.manual_seed(0)
+    cfg = DebertaV2Config(
+        vocab_size=V, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        max_relative_positions=32, position_buckets=8,
+        conv_kernel_size=3, hidden_dropout_prob=0.0)
+    m = DebertaV2ForMaskedLM(cfg)
+    ids = torch.randint(0, V, (2, 20))
+    loss, logits = m(ids, labels=ids)
+    assert logits.shape == (2, 20, V)
+    loss.backward()
+    assert m.deberta.conv is not None
+    assert m.deberta.rel_embeddings.weight.shape == (16, 32)  # 2*buckets
+
+
+def test_deberta_v2_relative_attention_is_translation_invariant():
+    """With no absolute positions, a content pattern placed at two
+    offsets must produce identical interior representations."""
+    torch.manual_seed(1)
+    cfg = DebertaV2Config(
+        vocab_size=V, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        max_relative_positions=32, position_buckets=8,
+        hidden_dropout_prob=0.0)
+    m = DebertaV2Model(cfg).eval()
+    pat = torch.randint(0, V, (1, 6))
+    a = torch.cat([pat, pat], dim=1)       # pattern at offsets 0 and 6
+    with torch.no_grad():
+        out = m(a)
+    # relative-only: token 2 inside first copy ~ token 8 inside second
+    # (identical local neighbourhood, same relative geometry to the copy)
+    assert (out[0, 2] - out[0, 8]).abs().max() < \
+        0.5 * out[0, 2].abs().max()
+
+
+# ----------------------------------------------------------- chinesebert
+def test_chinesebert_fusion_embedding():
+    torch.manual_seed(0)
+    cfg = ChineseBertConfig(
+        vocab_size=V, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        pinyin_embedding_size=16, glyph_embedding_dim=24,
+        hidden_dropout_prob=0.0)
+    m = ChineseBertModel(cfg).eval()
+    ids = torch.randint(0, V, (1, 10))
+    py_a = torch.zeros(1, 10, 8, dtype=torch.long)
+    py_b = torch.randint(0, cfg.pinyin_map_size, (1, 10, 8))
+    with torch.no_grad():
+        oa, _ = m(ids, pinyin_ids=py_a)
+        ob, _ = m(ids, pinyin_ids=py_b)
+    # pinyin view reaches the output
+    assert not torch.allclose(oa, ob, atol=1e-4)
+
+
+def test_chinesebert_classifier_backward():
+    cfg = ChineseBertConfig(
+        vocab_size=V, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        pinyin_embedding_size=16, glyph_embedding_dim=24)
+    m = ChineseBertForSequenceClassification(cfg)
+    ids = torch.randint(0, V, (2, 10))
+    loss, logits = m(ids, labels=torch.tensor([0, 1]))
+    loss.backward()
+    assert m.chinesebert.embeddings.glyph_embeddings.weight.grad is not None
+    assert m.chinesebert.embeddings.pinyin_embeddings.embedding.weight.grad \
+        is not None
+
+
+# ---------------------------------------------------------------- funnel
+def fun_cfg(**kw):
+    d = dict(vocab_size=V, hidden_size=32, num_attention_heads=4,
+             intermediate_size=64, block_sizes=(2, 2, 2),
+             num_decoder_layers=1, hidden_dropout_prob=0.0,
+             max_position_embeddings=64)
+    d.update(kw)
+    return FunnelConfig(**d)
+
+
+def test_funnel_pooling_halves_each_block():
+    torch.manual_seed(0)
+    m = FunnelModel(fun_cfg()).eval()
+    ids = torch.randint(0, V, (1, 16))
+    with torch.no_grad():
+        coarse, first, _ = m.encode(ids)
+    assert first.shape[1] == 16            # block 0 keeps full length
+    assert coarse.shape[1] == 4            # 16 -> 8 -> 4 over 3 blocks
+
+
+def test_funnel_decoder_restores_full_length():
+    torch.manual_seed(1)
+    m = FunnelModel(fun_cfg()).eval()
+    ids = torch.randint(0, V, (2, 16))
+    with torch.no_grad():
+        seq, coarse = m(ids)
+    assert seq.shape == (2, 16, 32)
+    assert coarse.shape == (2, 4, 32)
+
+
+def test_funnel_heads_backward():
+    cfg = fun_cfg()
+    ids = torch.randint(0, V, (2, 16))
+    m1 = FunnelForSequenceClassification(cfg)
+    loss, logits = m1(ids, labels=torch.tensor([0, 1]))
+    loss.backward()
+    assert logits.shape == (2, 2)
+    m2 = FunnelForTokenClassification(cfg)
+    loss, logits = m2(ids, labels=torch.randint(0, 2, (2, 16)))
+    loss.backward()
+    assert logits.shape == (2, 16, 2)
+
+
+def test_families_registered():
+    from paddlenlp_amd.transformers.auto.registry import MODEL_REGISTRY
+
+    for fam in ("deberta-v2", "chinesebert", "funnel"):
+        assert fam in MODEL_REGISTRY, fam
