@@ -341,3 +341,13 @@ from .funnel import (  # noqa: F401
     FunnelForTokenClassification,
     FunnelModel,
 )
+from .prophetnet import (  # noqa: F401
+    ProphetNetConfig,
+    ProphetNetForConditionalGeneration,
+    ProphetNetModel,
+)
+from .luke import (  # noqa: F401
+    LukeConfig,
+    LukeForEntityClassification,
+    LukeModel,
+)

@@ -197,6 +197,11 @@ MODEL_REGISTRY = {
                "base": "FunnelModel",
                "sequence_classification": "FunnelForSequenceClassification",
                "token_classification": "FunnelForTokenClassification"},
+    "prophetnet": {"module": "prophetnet", "config": "ProphetNetConfig",
+                   "base": "ProphetNetModel",
+                   "seq2seq_lm": "ProphetNetForConditionalGeneration"},
+    "luke": {"module": "luke", "config": "LukeConfig", "base": "LukeModel",
+             "entity_classification": "LukeForEntityClassification"},
 }
 
 

@@ -157,3 +157,81 @@ def test_families_registered():
 
     for fam in ("deberta-v2", "chinesebert", "funnel"):
         assert fam in MODEL_REGISTRY, fam
+
+
+# ------------------------------------------------------------ prophetnet
+def test_prophetnet_ngram_streams():
+    """Predict streams must carry the future-token signal: the n-gram
+    loss terms exist, predict-stream outputs differ per stream, and the
+    stream mask keeps the main stream causal."""
+    from paddlenlp_amd.transformers import (
+        ProphetNetConfig,
+        ProphetNetForConditionalGeneration,
+        ProphetNetModel,
+    )
+
+    torch.manual_seed(0)
+    cfg = ProphetNetConfig(
+        vocab_size=V, hidden_size=32, num_encoder_layers=2,
+        num_decoder_layers=2, num_attention_heads=4, intermediate_size=64,
+        ngram=2, dropout=0.0, max_position_embeddings=64,
+        pad_token_id=0, decoder_start_token_id=1)
+    m = ProphetNetModel(cfg).eval()
+    src = torch.randint(2, V, (1, 8))
+    tgt = torch.randint(2, V, (1, 6))
+    with torch.no_grad():
+        main, predict = m(src, tgt)
+    assert main.shape == (1, 6, 32)
+    assert predict.shape == (2, 1, 6, 32)
+    assert not torch.allclose(predict[0], predict[1], atol=1e-4)
+
+    # causality of the main stream
+    tgt2 = tgt.clone()
+    tgt2[0, 4] = (tgt2[0, 4] + 1) % (V - 2) + 2
+    with torch.no_grad():
+        main2, _ = m(src, tgt2)
+    assert torch.allclose(main[0, :4], main2[0, :4], atol=1e-5)
+
+    lm = ProphetNetForConditionalGeneration(cfg)
+    labels = torch.randint(2, V, (2, 6))
+    loss, logits = lm(torch.randint(2, V, (2, 8)), labels=labels)
+    assert logits.shape == (2, 6, V)
+    loss.backward()
+
+
+# ------------------------------------------------------------------ luke
+def test_luke_entity_aware_attention():
+    from paddlenlp_amd.transformers import (
+        LukeConfig,
+        LukeForEntityClassification,
+        LukeModel,
+    )
+
+    torch.manual_seed(0)
+    cfg = LukeConfig(
+        vocab_size=V, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        entity_vocab_size=50, entity_emb_size=16,
+        max_position_embeddings=64, hidden_dropout_prob=0.0)
+    m = LukeModel(cfg).eval()
+    ids = torch.randint(2, V, (1, 10))
+    ent = torch.tensor([[3, 7]])
+    # entity 0 spans tokens 2-4, entity 1 spans token 8
+    ent_pos = torch.tensor([[[2, 3, 4], [8, -1, -1]]])
+    with torch.no_grad():
+        w, e = m(ids, ent, ent_pos)
+    assert w.shape == (1, 10, 32) and e.shape == (1, 2, 32)
+    # the four query matrices are distinct parameters
+    layer = m.layers[0].attn
+    assert layer.w2e_query.weight.data_ptr() != layer.query.weight.data_ptr()
+
+    # entity stream affects word stream (joint attention)
+    with torch.no_grad():
+        w2, _ = m(ids, torch.tensor([[5, 7]]), ent_pos)
+    assert not torch.allclose(w, w2, atol=1e-4)
+
+    clf = LukeForEntityClassification(cfg)
+    loss, logits = clf(ids.repeat(2, 1), ent.repeat(2, 1),
+                       ent_pos.repeat(2, 1, 1), labels=torch.tensor([0, 1]))
+    loss.backward()
+    assert logits.shape == (2, 2)
