@@ -351,3 +351,22 @@ from .luke import (  # noqa: F401
     LukeForEntityClassification,
     LukeModel,
 )
+from .roformerv2 import (  # noqa: F401
+    RoFormerv2Config,
+    RoFormerv2ForSequenceClassification,
+    RoFormerv2Model,
+)
+from .ernie_ctm import (  # noqa: F401
+    ErnieCtmConfig,
+    ErnieCtmModel,
+    ErnieCtmWordtagModel,
+)
+from .ernie_doc import (  # noqa: F401
+    ErnieDocConfig,
+    ErnieDocForSequenceClassification,
+    ErnieDocModel,
+)
+from .transformer import (  # noqa: F401
+    TransformerConfig,
+    TransformerModel,
+)

@@ -202,6 +202,16 @@ MODEL_REGISTRY = {
                    "seq2seq_lm": "ProphetNetForConditionalGeneration"},
     "luke": {"module": "luke", "config": "LukeConfig", "base": "LukeModel",
              "entity_classification": "LukeForEntityClassification"},
+    "roformerv2": {"module": "roformerv2", "config": "RoFormerv2Config",
+                   "base": "RoFormerv2Model",
+                   "sequence_classification": "RoFormerv2ForSequenceClassification"},
+    "ernie_ctm": {"module": "ernie_ctm", "config": "ErnieCtmConfig",
+                  "base": "ErnieCtmModel"},
+    "ernie_doc": {"module": "ernie_doc", "config": "ErnieDocConfig",
+                  "base": "ErnieDocModel",
+                  "sequence_classification": "ErnieDocForSequenceClassification"},
+    "transformer": {"module": "transformer", "config": "TransformerConfig",
+                    "base": "TransformerModel"},
 }
 
 

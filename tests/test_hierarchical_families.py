@@ -235,3 +235,80 @@ def test_luke_entity_aware_attention():
                        ent_pos.repeat(2, 1, 1), labels=torch.tensor([0, 1]))
     loss.backward()
     assert logits.shape == (2, 2)
+
+
+# ------------------------------------------- roformerv2 / ernie_* / nmt
+def test_roformerv2_paramfree_norm_and_no_bias():
+    from paddlenlp_amd.transformers import (
+        RoFormerv2Config,
+        RoFormerv2ForSequenceClassification,
+        RoFormerv2Model,
+    )
+    from paddlenlp_amd.transformers.roformerv2.modeling import Norm
+
+    torch.manual_seed(0)
+    cfg = RoFormerv2Config(vocab_size=V, hidden_size=32,
+                           num_hidden_layers=2, num_attention_heads=4,
+                           intermediate_size=64, hidden_dropout_prob=0.0)
+    m = RoFormerv2Model(cfg)
+    assert m.layers[0].qkv_proj.bias is None       # bias-free
+    assert len(list(Norm().parameters())) == 0     # param-free norm
+    seq = m(torch.randint(0, V, (2, 10)))
+    assert seq.shape == (2, 10, 32)
+    clf = RoFormerv2ForSequenceClassification(cfg)
+    loss, _ = clf(torch.randint(0, V, (2, 10)), labels=torch.tensor([0, 1]))
+    loss.backward()
+
+
+def test_ernie_ctm_multi_cls_and_wordtag():
+    from paddlenlp_amd.transformers import ErnieCtmConfig, ErnieCtmWordtagModel
+
+    torch.manual_seed(0)
+    cfg = ErnieCtmConfig(vocab_size=V, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64,
+                         cls_num=2, num_tag=7)
+    m = ErnieCtmWordtagModel(cfg)
+    ids = torch.randint(0, V, (2, 12))
+    tags = torch.randint(0, 7, (2, 10))            # content region only
+    loss, tag_logits, sent_logits = m(ids, tag_labels=tags)
+    assert tag_logits.shape == (2, 10, 7)          # cls slots excluded
+    loss.backward()
+
+
+def test_ernie_doc_memory_recurrence():
+    from paddlenlp_amd.transformers import ErnieDocConfig, ErnieDocModel
+
+    torch.manual_seed(0)
+    cfg = ErnieDocConfig(vocab_size=V, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64,
+                         memory_len=8, hidden_dropout_prob=0.0,
+                         max_position_embeddings=64)
+    m = ErnieDocModel(cfg).eval()
+    seg1 = torch.randint(0, V, (1, 8))
+    seg2 = torch.randint(0, V, (1, 8))
+    with torch.no_grad():
+        _, mems = m(seg1)
+        assert len(mems) == 2 and mems[0].shape == (1, 8, 32)
+        with_mem, _ = m(seg2, memories=mems)
+        without, _ = m(seg2)
+    # the previous segment's memory must reach the current segment
+    assert not torch.allclose(with_mem, without, atol=1e-4)
+
+
+def test_classic_transformer_seq2seq():
+    from paddlenlp_amd.transformers import TransformerConfig, TransformerModel
+
+    torch.manual_seed(0)
+    cfg = TransformerConfig(src_vocab_size=V, trg_vocab_size=V,
+                            hidden_size=32, num_encoder_layers=2,
+                            num_decoder_layers=2, num_attention_heads=4,
+                            intermediate_size=64, dropout=0.0)
+    m = TransformerModel(cfg)
+    src = torch.randint(2, V, (2, 8))
+    labels = torch.randint(2, V, (2, 6))
+    loss, logits = m(src, labels=labels)
+    assert logits.shape == (2, 6, V)
+    loss.backward()
+    # tied projection (weight_sharing)
+    assert m.project_out.weight.data_ptr() == \
+        m.trg_embedding.weight.data_ptr()
