@@ -128,8 +128,18 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     const int l32 = lane & 31;
     const int hi = lane >> 5;
 
-    const int qt = blockIdx.x;
-    const int bh = blockIdx.y;
+    // XCD-aware remap (see flash_attn_v2.hip): keep one (b,h)'s tiles
+    // on one XCD so the shared K/V (dq) tiles re-hit that XCD's L2
+    int qt, bh;
+    if ((gridDim.y & 7) == 0) {
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        qt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        qt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
     const int q_base = qt * BLKM;
@@ -312,8 +322,17 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
     const int l32 = lane & 31;
     const int hi = lane >> 5;
 
-    const int kvt = blockIdx.x;
-    const int bh = blockIdx.y;
+    // XCD-aware remap: one (b,h)'s kv-tiles per XCD (shared Q/dO reuse)
+    int kvt, bh;
+    if ((gridDim.y & 7) == 0) {
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        kvt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        kvt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hk, hk = bh % Hk;
     const int G = Hq / Hk;
     const int kv_base = kvt * BLKKV;

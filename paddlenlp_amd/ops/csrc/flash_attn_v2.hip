@@ -119,8 +119,21 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     const int l32 = lane & 31;
     const int hi = lane >> 5;
 
-    const int qt = blockIdx.x;
-    const int bh = blockIdx.y;
+    // XCD-aware remap: the dispatcher round-robins flat workgroup ids
+    // over the 8 XCDs, so with the natural (qt fastest) order every XCD
+    // streams every head's KV and no L2 ever re-hits it.  Remap so one
+    // (b,h)'s q-tiles run on ONE XCD back-to-back: its 16 workgroups
+    // then consume each KV tile out of that XCD's L2 together.
+    int qt, bh;
+    if ((gridDim.y & 7) == 0) {
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        qt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        qt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
     const int q_base = qt * FA2_BLKM;

@@ -370,3 +370,19 @@ from .transformer import (  # noqa: F401
     TransformerConfig,
     TransformerModel,
 )
+from .chineseclip import (  # noqa: F401
+    ChineseCLIPConfig,
+    ChineseCLIPModel,
+)
+from .bit import (  # noqa: F401
+    BitConfig,
+    BitForImageClassification,
+    BitModel,
+)
+from .ernie_code import (  # noqa: F401
+    ErnieCodeConfig,
+    ErnieCodeEncoderModel,
+    ErnieCodeForConditionalGeneration,
+    ErnieCodeModel,
+)
+from .bert_japanese import BertJapaneseTokenizer  # noqa: F401

@@ -212,6 +212,13 @@ MODEL_REGISTRY = {
                   "sequence_classification": "ErnieDocForSequenceClassification"},
     "transformer": {"module": "transformer", "config": "TransformerConfig",
                     "base": "TransformerModel"},
+    "chineseclip": {"module": "chineseclip", "config": "ChineseCLIPConfig",
+                    "base": "ChineseCLIPModel"},
+    "bit": {"module": "bit", "config": "BitConfig", "base": "BitModel",
+            "image_classification": "BitForImageClassification"},
+    "ernie_code": {"module": "ernie_code", "config": "ErnieCodeConfig",
+                   "base": "ErnieCodeModel",
+                   "seq2seq_lm": "ErnieCodeForConditionalGeneration"},
 }
 
 
