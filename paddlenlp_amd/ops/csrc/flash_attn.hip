@@ -118,8 +118,17 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
     const int l16 = lane & 15;
     const int lk8 = (lane >> 4) * 8;
 
-    const int qt = blockIdx.x;
-    const int bh = blockIdx.y;
+    int qt, bh;
+    if ((gridDim.y & 7) == 0) {
+        // XCD-aware remap (see flash_attn_v2.hip): one (b,h) per XCD
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        qt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        qt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
     const int q_base = qt * FWD_BLKM;
@@ -368,8 +377,17 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_pipe_kernel(
     const int l16 = lane & 15;
     const int lk8 = (lane >> 4) * 8;
 
-    const int qt = blockIdx.x;
-    const int bh = blockIdx.y;
+    int qt, bh;
+    if ((gridDim.y & 7) == 0) {
+        // XCD-aware remap (see flash_attn_v2.hip): one (b,h) per XCD
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        qt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        qt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
     const int q_base = qt * BLKM;
@@ -611,8 +629,17 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dq_kernel(
     const int l16 = lane & 15;
     const int lk8 = (lane >> 4) * 8;
 
-    const int qt = blockIdx.x;
-    const int bh = blockIdx.y;
+    int qt, bh;
+    if ((gridDim.y & 7) == 0) {
+        // XCD-aware remap (see flash_attn_v2.hip): one (b,h) per XCD
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        qt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        qt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hq, hq = bh % Hq;
     const int hk = hq / (Hq / Hk);
     const int q_base = qt * BLK_M;
@@ -788,8 +815,17 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_bwd_dkv_kernel(
     const int l16 = lane & 15;
     const int lk8 = (lane >> 4) * 8;
 
-    const int kvt = blockIdx.x;
-    const int bh = blockIdx.y;           // b * Hk + hk (one block per kv head)
+    int kvt, bh;
+    if ((gridDim.y & 7) == 0) {
+        // XCD-aware remap (see flash_attn_v2.hip): one (b,h) per XCD
+        const int flat = blockIdx.x + gridDim.x * blockIdx.y;
+        const int idx = flat >> 3;
+        kvt = idx % gridDim.x;
+        bh = (flat & 7) * (gridDim.y >> 3) + idx / gridDim.x;
+    } else {
+        kvt = blockIdx.x;
+        bh = blockIdx.y;
+    }
     const int b = bh / Hk, hk = bh % Hk;
     const int G = Hq / Hk;               // q heads accumulated in-kernel
     const int kv_base = kvt * BLK_N;

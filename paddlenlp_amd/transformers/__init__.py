@@ -386,3 +386,13 @@ from .ernie_code import (  # noqa: F401
     ErnieCodeModel,
 )
 from .bert_japanese import BertJapaneseTokenizer  # noqa: F401
+from .speecht5 import (  # noqa: F401
+    SpeechT5Config,
+    SpeechT5ForSpeechToText,
+    SpeechT5ForTextToSpeech,
+    SpeechT5Model,
+)
+from .clap import (  # noqa: F401
+    ClapConfig,
+    ClapModel,
+)

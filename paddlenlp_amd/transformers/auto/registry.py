@@ -219,6 +219,9 @@ MODEL_REGISTRY = {
     "ernie_code": {"module": "ernie_code", "config": "ErnieCodeConfig",
                    "base": "ErnieCodeModel",
                    "seq2seq_lm": "ErnieCodeForConditionalGeneration"},
+    "speecht5": {"module": "speecht5", "config": "SpeechT5Config",
+                 "base": "SpeechT5Model"},
+    "clap": {"module": "clap", "config": "ClapConfig", "base": "ClapModel"},
 }
 
 

@@ -103,3 +103,76 @@ def test_bert_japanese_tokenizer(tmp_path):
         BertJapaneseTokenizer(vocab_file=str(vf),
                               word_tokenizer_type="mecab")
         assert any("MeCab" in str(x.message) for x in w)
+
+
+# -------------------------------------------------------- speech / audio
+def test_speecht5_asr_speech_to_text():
+    from paddlenlp_amd.transformers import SpeechT5Config, SpeechT5ForSpeechToText
+
+    torch.manual_seed(0)
+    cfg = SpeechT5Config(
+        vocab_size=40, hidden_size=32, encoder_layers=2, decoder_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        conv_dim=(16, 16), conv_stride=(5, 2), conv_kernel=(10, 3),
+        positional_conv_kernel=8, positional_conv_groups=4)
+    m = SpeechT5ForSpeechToText(cfg)
+    wav = torch.randn(2, 400)                      # raw audio
+    labels = torch.randint(3, 40, (2, 6))
+    loss, logits = m(wav, labels=labels)
+    assert logits.shape == (2, 6, 40)
+    loss.backward()
+    # the conv feature encoder downsamples the waveform
+    frames = m.speecht5.speech_encoder_prenet.feature_encoder(wav).shape[1]
+    assert 1 < frames < 400
+
+
+def test_speecht5_tts_text_to_mel():
+    from paddlenlp_amd.transformers import SpeechT5Config, SpeechT5ForTextToSpeech
+
+    torch.manual_seed(0)
+    cfg = SpeechT5Config(
+        vocab_size=40, hidden_size=32, encoder_layers=2, decoder_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        conv_dim=(16,), conv_stride=(5,), conv_kernel=(10,),
+        num_mel_bins=20, reduction_factor=2,
+        speech_decoder_prenet_units=16, speech_decoder_postnet_units=16,
+        positional_conv_kernel=8, positional_conv_groups=4)
+    m = SpeechT5ForTextToSpeech(cfg)
+    ids = torch.randint(3, 40, (2, 8))
+    mel = torch.randn(2, 12, 20)                   # target mel frames
+    loss, after = m(ids, labels=mel)
+    assert after.shape[0] == 2 and after.shape[2] == 20
+    loss.backward()
+    # the speech-decoder prenet keeps dropout ON in eval (TTS diversity)
+    m.eval()
+    with torch.no_grad():
+        a1 = m(ids, decoder_mel=mel[:, ::2])
+        a2 = m(ids, decoder_mel=mel[:, ::2])
+    assert not torch.allclose(a1[1], a2[1])
+
+
+def test_clap_contrastive_audio_text():
+    from paddlenlp_amd.transformers import ClapConfig, ClapModel
+
+    torch.manual_seed(0)
+    cfg = ClapConfig(
+        audio_config=dict(num_mel_bins=32, max_frames=64, patch_size=8,
+                          hidden_size=32, num_hidden_layers=2,
+                          num_attention_heads=4, intermediate_size=64),
+        text_config=dict(vocab_size=96, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64),
+        projection_dim=16)
+    m = ClapModel(cfg).eval()
+    ids = torch.randint(0, 96, (3, 10))
+    mel = torch.randn(2, 32, 64)
+    with torch.no_grad():
+        la, lt = m(ids, mel)
+    assert la.shape == (2, 3) and lt.shape == (3, 2)
+    # symmetric InfoNCE is trainable end to end
+    m.train()
+    la, lt = m(ids[:2], mel)
+    target = torch.arange(2)
+    loss = 0.5 * (torch.nn.functional.cross_entropy(la, target) +
+                  torch.nn.functional.cross_entropy(lt, target))
+    loss.backward()
+    assert m.audio_model.patch_embed.weight.grad is not None
