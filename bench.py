@@ -107,6 +107,9 @@ def main():
         use_fused_rms_norm=True,
         use_fused_rope=True,
         use_fused_swiglu=True,
+        # chunked head+CE (no [tokens, vocab] logits tensor); A/B via env
+        use_fused_linear_cross_entropy=(
+            os.environ.get("PNLP_FUSED_CE", "0") == "1"),
     )
     if rank == 0:
         n_params = (

@@ -396,3 +396,13 @@ from .clap import (  # noqa: F401
     ClapConfig,
     ClapModel,
 )
+from .blip_2 import (  # noqa: F401
+    Blip2Config,
+    Blip2Model,
+    Blip2QFormerModel,
+)
+from .dpt import (  # noqa: F401
+    DPTConfig,
+    DPTForDepthEstimation,
+    DPTModel,
+)

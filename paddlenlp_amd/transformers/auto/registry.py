@@ -222,6 +222,10 @@ MODEL_REGISTRY = {
     "speecht5": {"module": "speecht5", "config": "SpeechT5Config",
                  "base": "SpeechT5Model"},
     "clap": {"module": "clap", "config": "ClapConfig", "base": "ClapModel"},
+    "blip-2": {"module": "blip_2", "config": "Blip2Config",
+               "base": "Blip2Model"},
+    "dpt": {"module": "dpt", "config": "DPTConfig", "base": "DPTModel",
+            "depth_estimation": "DPTForDepthEstimation"},
 }
 
 
