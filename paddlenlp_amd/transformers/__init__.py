@@ -406,3 +406,20 @@ from .dpt import (  # noqa: F401
     DPTForDepthEstimation,
     DPTModel,
 )
+from .minigpt4 import (  # noqa: F401
+    MiniGPT4Config,
+    MiniGPT4ForConditionalGeneration,
+)
+from .visualglm import (  # noqa: F401
+    VisualGLMConfig,
+    VisualGLMForConditionalGeneration,
+)
+from .ernie_vil import (  # noqa: F401
+    ErnieViLConfig,
+    ErnieViLModel,
+)
+from .dallebart import (  # noqa: F401
+    DalleBartConfig,
+    DalleBartForConditionalGeneration,
+    DalleBartModel,
+)

@@ -226,6 +226,15 @@ MODEL_REGISTRY = {
                "base": "Blip2Model"},
     "dpt": {"module": "dpt", "config": "DPTConfig", "base": "DPTModel",
             "depth_estimation": "DPTForDepthEstimation"},
+    "minigpt4": {"module": "minigpt4", "config": "MiniGPT4Config",
+                 "base": "MiniGPT4ForConditionalGeneration"},
+    "visualglm": {"module": "visualglm", "config": "VisualGLMConfig",
+                  "base": "VisualGLMForConditionalGeneration"},
+    "ernie_vil": {"module": "ernie_vil", "config": "ErnieViLConfig",
+                  "base": "ErnieViLModel"},
+    "dallebart": {"module": "dallebart", "config": "DalleBartConfig",
+                  "base": "DalleBartModel",
+                  "seq2seq_lm": "DalleBartForConditionalGeneration"},
 }
 
 

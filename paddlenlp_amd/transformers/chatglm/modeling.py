@@ -184,19 +184,21 @@ class ChatGLMModel(ChatGLMPretrainedModel):
     def get_input_embeddings(self):
         return self.word_embeddings
 
-    def forward(self, input_ids, prefix_len=None, past_key_values=None,
-                use_cache=False):
-        B, S = input_ids.shape
+    def forward(self, input_ids=None, prefix_len=None, past_key_values=None,
+                use_cache=False, inputs_embeds=None):
+        x = (self.word_embeddings(input_ids) if inputs_embeds is None
+             else inputs_embeds)
+        B, S = x.shape[:2]
+        device = x.device
         past_len = 0
         if past_key_values is not None and past_key_values[0] is not None:
             past_len = past_key_values[0][0].shape[2]
         if prefix_len is None:
             prefix_len = torch.full((B,), past_len + S if past_len == 0 else 0,
-                                    dtype=torch.long, device=input_ids.device)
+                                    dtype=torch.long, device=device)
         pos_ids, block_ids = glm_2d_positions(prefix_len, past_len, S,
-                                              input_ids.device)
-        mask = prefix_lm_mask(S, prefix_len, past_len, input_ids.device)
-        x = self.word_embeddings(input_ids)
+                                              device)
+        mask = prefix_lm_mask(S, prefix_len, past_len, device)
         presents = [] if use_cache else None
         for i, layer in enumerate(self.layers):
             past = past_key_values[i] if past_key_values is not None else None
@@ -223,9 +225,11 @@ class ChatGLMForCausalLM(ChatGLMPretrainedModel):
     def get_input_embeddings(self):
         return self.chatglm.word_embeddings
 
-    def forward(self, input_ids, prefix_len=None, labels=None,
-                past_key_values=None, use_cache=False, **kwargs):
-        out = self.chatglm(input_ids, prefix_len, past_key_values, use_cache)
+    def forward(self, input_ids=None, prefix_len=None, labels=None,
+                past_key_values=None, use_cache=False, inputs_embeds=None,
+                **kwargs):
+        out = self.chatglm(input_ids, prefix_len, past_key_values, use_cache,
+                           inputs_embeds=inputs_embeds)
         if use_cache:
             hidden, presents = out
         else:

@@ -1,0 +1,60 @@
+"""VisualGLM (reference: paddlenlp/transformers/visualglm/modeling.py).
+
+BLIP-2-style bridge into ChatGLM: ViT + Q-Former (the reference
+duplicates the BLIP-2 stack :139-700; here it IS the shared Blip2
+bridge) projected to the ChatGLM hidden size and prepended as soft
+prompts to the ChatGLM embedding stream.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..blip_2.modeling import Blip2Config, Blip2Model
+from ..chatglm import ChatGLMConfig, ChatGLMForCausalLM
+from ..configuration_utils import PretrainedConfig
+from ..model_utils import PretrainedModel
+
+__all__ = ["VisualGLMConfig", "VisualGLMForConditionalGeneration"]
+
+
+class VisualGLMConfig(PretrainedConfig):
+    model_type = "visualglm"
+
+    def __init__(self, vision_config=None, qformer_config=None,
+                 text_config=None, num_query_tokens=32, **kwargs):
+        super().__init__(**kwargs)
+        self.text_config = ChatGLMConfig(**(text_config or {}))
+        self.bridge_config = Blip2Config(
+            vision_config=vision_config, qformer_config=qformer_config,
+            num_query_tokens=num_query_tokens,
+            lm_hidden_size=self.text_config.hidden_size)
+        self.num_query_tokens = num_query_tokens
+        self.initializer_range = 0.02
+
+
+class VisualGLMForConditionalGeneration(PretrainedModel):
+    config_class = VisualGLMConfig
+    base_model_prefix = "visualglm"
+
+    def _init_weights(self, module):
+        pass  # sub-models initialize themselves
+
+    def __init__(self, config: VisualGLMConfig):
+        super().__init__(config)
+        self.bridge = Blip2Model(config.bridge_config)
+        self.language_model = ChatGLMForCausalLM(config.text_config)
+
+    def forward(self, pixel_values, input_ids, labels=None):
+        prompt = self.bridge(pixel_values)
+        embed = self.language_model.get_input_embeddings()(input_ids)
+        inputs_embeds = torch.cat([prompt, embed], dim=1)
+        logits = self.language_model(inputs_embeds=inputs_embeds)
+        if labels is not None:
+            text_logits = logits[:, prompt.shape[1]:]
+            loss = F.cross_entropy(
+                text_logits[:, :-1].reshape(-1, logits.shape[-1]),
+                labels[:, 1:].reshape(-1), ignore_index=-100)
+            return loss, logits
+        return logits
