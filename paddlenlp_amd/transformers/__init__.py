@@ -423,3 +423,31 @@ from .dallebart import (  # noqa: F401
     DalleBartForConditionalGeneration,
     DalleBartModel,
 )
+from .artist import (  # noqa: F401
+    ArtistConfig,
+    ArtistForConditionalGeneration,
+    ArtistModel,
+)
+from .ernie_gen import (  # noqa: F401
+    ErnieGenConfig,
+    ErnieGenForGeneration,
+    ErnieGenModel,
+)
+from .clipseg import (  # noqa: F401
+    CLIPSegConfig,
+    CLIPSegForImageSegmentation,
+)
+from .semantic_search import (  # noqa: F401
+    ErnieCrossEncoder,
+    ErnieDualEncoder,
+)
+from .layoutlmv2 import (  # noqa: F401
+    LayoutLMv2Config,
+    LayoutLMv2ForTokenClassification,
+    LayoutLMv2Model,
+)
+from .layoutxlm import (  # noqa: F401
+    LayoutXLMConfig,
+    LayoutXLMForTokenClassification,
+    LayoutXLMModel,
+)

@@ -235,6 +235,19 @@ MODEL_REGISTRY = {
     "dallebart": {"module": "dallebart", "config": "DalleBartConfig",
                   "base": "DalleBartModel",
                   "seq2seq_lm": "DalleBartForConditionalGeneration"},
+    "artist": {"module": "artist", "config": "ArtistConfig",
+               "base": "ArtistModel",
+               "causal_lm": "ArtistForConditionalGeneration"},
+    "ernie_gen": {"module": "ernie_gen", "config": "ErnieGenConfig",
+                  "base": "ErnieGenModel"},
+    "clipseg": {"module": "clipseg", "config": "CLIPSegConfig",
+                "base": "CLIPSegForImageSegmentation"},
+    "layoutlmv2": {"module": "layoutlmv2", "config": "LayoutLMv2Config",
+                   "base": "LayoutLMv2Model",
+                   "token_classification": "LayoutLMv2ForTokenClassification"},
+    "layoutxlm": {"module": "layoutxlm", "config": "LayoutXLMConfig",
+                  "base": "LayoutXLMModel",
+                  "token_classification": "LayoutXLMForTokenClassification"},
 }
 
 

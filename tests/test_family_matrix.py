@@ -95,6 +95,9 @@ TINY = dict(
     glm=dict(vocab_size=96, hidden_size=32, num_hidden_layers=2,
              num_attention_heads=4, max_position_embeddings=64,
              hidden_dropout_prob=0.0),
+    artist=dict(vocab_size=96, hidden_size=32, intermediate_size=64,
+                num_hidden_layers=2, num_attention_heads=4,
+                max_position_embeddings=64),
 )
 # gpt2 aliases gpt; skip the alias row
 FAMILIES = sorted(mt for mt, entry in MODEL_REGISTRY.items()

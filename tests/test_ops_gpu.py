@@ -305,3 +305,36 @@ def test_skinny_gemm(C, M, N, K, ks):
     ref = x.float() @ w.float().t()
     assert torch.allclose(y.float(), ref, atol=0.5, rtol=2e-2), \
         (y.float() - ref).abs().max()
+
+
+def test_fused_head_and_loss_gpu_parity(C):
+    """The chunked head+CE autograd.Function on the GPU kernel path must
+    match the unfused lm_head @ + cross_entropy composition (loss, dhidden
+    and dweight)."""
+    from paddlenlp_amd.transformers.tensor_parallel_utils import (
+        fused_head_and_loss_fn,
+    )
+
+    torch.manual_seed(0)
+    N, H, V = 300, 64, 2048
+    hidden = _bf16(torch.randn(N, H, device="cuda")).requires_grad_()
+    weight = _bf16(torch.randn(V, H, device="cuda") * 0.02).requires_grad_()
+    labels = torch.randint(0, V, (N,), device="cuda")
+    labels[7] = -100
+
+    loss = fused_head_and_loss_fn(hidden, weight, labels, chunk_tokens=128)
+    loss.backward()
+
+    h2 = hidden.detach().clone().requires_grad_()
+    w2 = weight.detach().clone().requires_grad_()
+    logits = h2 @ w2.t()
+    ref = torch.nn.functional.cross_entropy(logits.float(), labels,
+                                            ignore_index=-100)
+    ref.backward()
+
+    assert torch.allclose(loss.float(), ref.float(), atol=2e-2, rtol=2e-2), \
+        (loss.item(), ref.item())
+    assert torch.allclose(hidden.grad.float(), h2.grad.float(),
+                          atol=2e-2, rtol=5e-2)
+    assert torch.allclose(weight.grad.float(), w2.grad.float(),
+                          atol=2e-2, rtol=5e-2)

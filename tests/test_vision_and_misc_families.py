@@ -316,3 +316,116 @@ def test_dallebart_text_to_image_tokens():
     assert isinstance(m.dallebart.encoder[0].glu, GLUFeedForward)
     grid = m.generate_image_tokens(text, top_k=8)
     assert grid.shape == (2, 9) and int(grid.max()) < 64
+
+
+# ------------------------------------ last reference-inventory families
+def test_ernie_gen_infill_bias():
+    from paddlenlp_amd.transformers import ErnieGenConfig, ErnieGenForGeneration
+    from paddlenlp_amd.transformers.ernie_gen.modeling import build_infill_bias
+
+    bias = build_infill_bias(3, 3, torch.device("cpu"), torch.float32)[0, 0]
+    neg = torch.finfo(torch.float32).min
+    assert bias[0, 2] == 0 and bias[0, 4] == neg   # src sees src, not tgt
+    assert bias[4, 1] == 0 and bias[4, 3] == 0      # tgt sees src + past tgt
+    assert bias[3, 4] == neg                        # tgt is causal
+
+    torch.manual_seed(0)
+    cfg = ErnieGenConfig(vocab_size=96, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64,
+                         hidden_dropout_prob=0.0)
+    m = ErnieGenForGeneration(cfg)
+    ids = torch.randint(0, 96, (2, 10))
+    labels = ids[:, 6:]
+    loss, logits = m(ids, src_len=6, labels=labels)
+    assert logits.shape == (2, 4, 96)
+    loss.backward()
+
+
+def test_clipseg_film_segmentation():
+    from paddlenlp_amd.transformers import CLIPSegConfig, CLIPSegForImageSegmentation
+
+    torch.manual_seed(0)
+    cfg = CLIPSegConfig(
+        text_config=dict(vocab_size=96, hidden_size=32, num_hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64),
+        vision_config=dict(hidden_size=32, num_hidden_layers=3,
+                           num_attention_heads=4, intermediate_size=64,
+                           image_size=32, patch_size=8),
+        extract_layers=(0, 1, 2), reduce_dim=16, projection_dim=16,
+        decoder_intermediate_size=32)
+    m = CLIPSegForImageSegmentation(cfg).eval()
+    ids = torch.randint(0, 96, (2, 8))
+    px = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        mask_a = m(ids, px)
+        # different prompt -> different mask (FiLM conditioning works)
+        mask_b = m((ids + 3) % 96, px)
+    assert mask_a.dim() == 3
+    assert not torch.allclose(mask_a, mask_b, atol=1e-5)
+    loss, _ = m(ids, px, labels=torch.randint(0, 2, mask_a.shape).float())
+    loss.backward()
+
+
+def test_semantic_search_dual_and_cross():
+    from paddlenlp_amd.transformers import ErnieCrossEncoder, ErnieDualEncoder
+    from paddlenlp_amd.transformers.ernie import ErnieConfig
+
+    torch.manual_seed(0)
+    cfg = ErnieConfig(vocab_size=96, hidden_size=32, num_hidden_layers=2,
+                      num_attention_heads=4, intermediate_size=64)
+    dual = ErnieDualEncoder(cfg, output_emb_size=16)
+    q = torch.randint(0, 96, (4, 8))
+    t = torch.randint(0, 96, (4, 8))
+    loss, logits = dual(q, t)
+    assert logits.shape == (4, 4)
+    loss.backward()
+    sim = dual.cosine_sim(q, t)
+    assert sim.shape == (4,) and sim.abs().max() <= 1.0 + 1e-5
+    # shared towers by default
+    assert dual.query_ernie is dual.title_ernie
+
+    cross = ErnieCrossEncoder(cfg)
+    loss, logits = cross(q, labels=torch.tensor([0, 1, 0, 1]))
+    loss.backward()
+
+
+def test_layoutlmv2_three_modalities():
+    from paddlenlp_amd.transformers import (
+        LayoutLMv2Config,
+        LayoutLMv2ForTokenClassification,
+        LayoutXLMModel,
+        LayoutXLMConfig,
+    )
+
+    torch.manual_seed(0)
+    cfg = LayoutLMv2Config(
+        vocab_size=96, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        image_feature_pool_shape=(4, 4, 32), rel_pos_bins=8,
+        max_rel_pos=32, rel_2d_pos_bins=8, max_rel_2d_pos=64,
+        num_labels=5, hidden_dropout_prob=0.0)
+    m = LayoutLMv2ForTokenClassification(cfg)
+    ids = torch.randint(0, 96, (2, 10))
+    bbox = torch.randint(0, 1000, (2, 10, 4))
+    bbox[:, :, 2:] = bbox[:, :, :2] + 10
+    img = torch.randn(2, 3, 32, 32)
+    loss, logits = m(ids, bbox, img, labels=torch.randint(0, 5, (2, 10)))
+    assert logits.shape == (2, 10, 5)
+    loss.backward()
+
+    # bbox reaches the output (spatial embeddings + 2d bias)
+    base = m.layoutlmv2
+    with torch.no_grad():
+        a, _ = base(ids, bbox, img)
+        b2 = bbox.clone()
+        b2[:, 5] += 200
+        b, _ = base(ids, b2, img)
+    assert not torch.allclose(a, b, atol=1e-5)
+
+    xlm = LayoutXLMModel(LayoutXLMConfig(
+        vocab_size=128, hidden_size=32, num_hidden_layers=1,
+        num_attention_heads=4, intermediate_size=64,
+        image_feature_pool_shape=(4, 4, 32), rel_pos_bins=8,
+        max_rel_pos=32, rel_2d_pos_bins=8, max_rel_2d_pos=64))
+    text, vis = xlm(torch.randint(0, 128, (1, 6)))
+    assert text.shape == (1, 6, 32) and vis.shape == (1, 16, 32)
