@@ -160,6 +160,7 @@ class LayoutLMv2Model(LayoutLMv2PretrainedModel):
         return self.embeddings
 
     def _spatial(self, bbox):
+        bbox = bbox.clamp(0, self.config.max_2d_position_embeddings - 1)
         x0 = self.x_embeddings(bbox[:, :, 0])
         y0 = self.y_embeddings(bbox[:, :, 1])
         x1 = self.x_embeddings(bbox[:, :, 2])
