@@ -23,6 +23,7 @@ class _GenerationTask:
         "dialogue": "{text}",
         "poetry_generation": "以「{text}」为题作诗：",
         "code_generation": "{text}",
+        "question_generation": "根据文章生成问题：{text}",
     }
 
     def __init__(self, task: str, model_path: str, max_new_tokens: int = 64,
@@ -81,7 +82,7 @@ class _Seq2SeqTask:
         return decoded[0] if single else decoded
 
 
-SEQ2SEQ_TASKS = {"text_summarization"}
+SEQ2SEQ_TASKS = {"text_summarization", "text2text_generation"}
 
 GENERATION_TASKS = set(_GenerationTask.TEMPLATES)
 
