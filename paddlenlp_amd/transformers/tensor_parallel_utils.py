@@ -77,7 +77,10 @@ class FusedHeadAndCrossEntropy(torch.autograd.Function):
                 valid = (lab != ignore_index).float()[:, None]
                 dlogits = ((p - onehot) * valid * scale).to(logits.dtype)
             dhidden[s:e] = dlogits @ weight
-            dweight += (dlogits.t().float() @ h.float())
+            # bf16 GEMM (MFMA accumulates fp32 internally), summed into the
+            # fp32 buffer — a full-fp32 GEMM here runs at a fraction of the
+            # bf16 MFMA rate and dominated the fused path's cost
+            dweight += (dlogits.t() @ h).float()
         return dhidden, dweight.to(weight.dtype), None, None, None
 
 
