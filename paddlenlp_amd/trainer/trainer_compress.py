@@ -28,22 +28,38 @@ from .compression_args import CompressionArguments
 # PTQ: static activation-scale calibration + simulated a8w8 linears
 # ---------------------------------------------------------------------------
 class _ActObserver:
-    """Collects an activation scale for one linear input."""
+    """Collects an activation scale for one linear input.
 
-    def __init__(self, algo: str):
+    Scalar algos: abs_max / avg / mse.  Vector algos (reference
+    llm/experimental/observer/{channel_wise,abs_max_headwise}.py):
+    channel_wise keeps one scale per input channel; abs_max_headwise
+    keeps one per attention head (head count via `heads`)."""
+
+    def __init__(self, algo: str, heads: int = 0):
         self.algo = algo
+        self.heads = heads
         self.max_seen = 0.0
         self.maxes: List[float] = []
         self.samples: List[torch.Tensor] = []
+        self.vec_max: Optional[torch.Tensor] = None
 
     def update(self, x: torch.Tensor):
+        if self.algo in ("channel_wise", "abs_max_headwise"):
+            v = x.detach().abs().reshape(-1, x.shape[-1]).amax(dim=0).float()
+            if self.algo == "abs_max_headwise":
+                v = v.reshape(self.heads, -1).amax(dim=-1)
+            self.vec_max = v if self.vec_max is None else                 torch.maximum(self.vec_max, v.to(self.vec_max.device))
+            return
         m = float(x.detach().abs().amax())
         self.max_seen = max(self.max_seen, m)
         self.maxes.append(m)
         if self.algo == "mse" and len(self.samples) < 8:
             self.samples.append(x.detach().flatten()[:4096].float().cpu())
 
-    def scale(self) -> float:
+    def scale(self):
+        if self.algo in ("channel_wise", "abs_max_headwise"):
+            v = self.vec_max if self.vec_max is not None else torch.ones(1)
+            return v.clamp(min=1e-8)
         if self.algo == "abs_max":
             s = self.max_seen
         elif self.algo == "avg":
