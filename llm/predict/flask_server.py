@@ -11,8 +11,11 @@ batch (reference broadcast_msg :107).
 """
 import argparse
 import json
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, StreamingResponse
 
 __all__ = ["ServerArgument", "PredictorServer", "build_app"]
 
@@ -53,13 +56,6 @@ class PredictorServer:
 
 
 def build_app(server: "PredictorServer"):
-    # module-scope annotation resolution: FastAPI resolves the Request
-    # annotation through typing.get_type_hints, so it must be importable
-    # at call time
-    from fastapi import FastAPI, Request
-    from fastapi.responses import JSONResponse, StreamingResponse
-    globals()["Request"] = Request
-
     app = FastAPI(title=server.args.title)
 
     @app.get("/health")
