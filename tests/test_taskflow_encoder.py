@@ -292,3 +292,20 @@ def test_text_summarization_task(tmp_path):
     assert isinstance(out, str)
     outs = flow(["the cat sat", "the dog ran fast"])
     assert isinstance(outs, list) and len(outs) == 2
+
+
+def test_task_registry_covers_reference_inventory():
+    """Every reference TASKS pipeline name resolves here (taskflow.py:48)."""
+    import paddlenlp_amd.taskflow.taskflow as tf
+
+    expected = {
+        "code_generation", "dependency_parsing", "dialogue",
+        "document_intelligence", "feature_extraction", "fill_mask",
+        "information_extraction", "knowledge_mining", "lexical_analysis",
+        "ner", "poetry_generation", "pos_tagging", "question_answering",
+        "question_generation", "sentiment_analysis", "text2text_generation",
+        "text_classification", "text_correction", "text_generation",
+        "text_similarity", "text_summarization", "word_segmentation",
+        "zero_shot_text_classification",
+    }
+    assert expected <= set(tf.TASKS), sorted(expected - set(tf.TASKS))
