@@ -8,7 +8,6 @@ heads; both halves concatenate back to the hidden size.
 """
 from __future__ import annotations
 
-import math
 
 import torch
 import torch.nn as nn

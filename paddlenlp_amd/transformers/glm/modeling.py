@@ -14,7 +14,7 @@ import torch.nn.functional as F
 
 from ...generation import GenerationMixin
 from ..configuration_utils import PretrainedConfig
-from ..encoder import ACT2FN, init_encoder_weights
+from ..encoder import init_encoder_weights
 from ..model_utils import PretrainedModel
 
 __all__ = ["GLMConfig", "GLMModel", "GLMForConditionalGeneration"]

@@ -11,7 +11,6 @@ future-n-gram objective).
 """
 from __future__ import annotations
 
-import math
 
 import torch
 import torch.nn as nn

@@ -15,7 +15,6 @@ inverse-pass (torch recompute hooks cover the memory story on MI355X).
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.nn as nn

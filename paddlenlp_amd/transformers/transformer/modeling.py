@@ -16,7 +16,7 @@ import torch.nn.functional as F
 
 from ...generation import GenerationMixin
 from ..configuration_utils import PretrainedConfig
-from ..encoder import expand_padding_mask, init_encoder_weights
+from ..encoder import init_encoder_weights
 from ..model_utils import PretrainedModel
 
 __all__ = ["TransformerConfig", "TransformerModel"]

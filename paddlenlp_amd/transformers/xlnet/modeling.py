@@ -11,7 +11,6 @@ what this implementation provides.
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.nn as nn
