@@ -67,10 +67,17 @@ class BigBirdConfig(PretrainedConfig):
         return self.hidden_size // self.num_attention_heads
 
 
+_RAND_MAP_CACHE = {}
+
+
 def _random_block_map(nblk: int, g: int, r: int, seed: int) -> torch.Tensor:
     """[nblk, r] random source blocks per query block (excluding self,
     globals and immediate window), fixed by seed (reference uses a static
-    rand pattern per sequence length)."""
+    rand pattern per sequence length).  Cached per shape: the pattern is
+    deterministic, so the Python loop runs once, not per forward."""
+    key = (nblk, g, r, seed)
+    if key in _RAND_MAP_CACHE:
+        return _RAND_MAP_CACHE[key]
     gen = torch.Generator().manual_seed(seed)
     rows = []
     for i in range(nblk):
@@ -79,7 +86,9 @@ def _random_block_map(nblk: int, g: int, r: int, seed: int) -> torch.Tensor:
             cand = [i]
         idx = torch.randint(0, len(cand), (r,), generator=gen)
         rows.append(torch.tensor([cand[j] for j in idx]))
-    return torch.stack(rows)                               # [nblk, r]
+    out = torch.stack(rows)                                # [nblk, r]
+    _RAND_MAP_CACHE[key] = out
+    return out
 
 
 class BigBirdSparseAttention(nn.Module):

@@ -193,10 +193,11 @@ class XLNetModel(XLNetPretrainedModel):
             mask = (1.0 - m[:, None, None, :].to(x.dtype)) * torch.finfo(x.dtype).min
 
         new_mems = []
+        mem_len = self.config.mem_len
         for i, layer in enumerate(self.layers):
-            if use_mems:
+            if use_mems and mem_len > 0:
                 cur = x if mems is None else torch.cat([mems[i], x], dim=1)
-                new_mems.append(cur[:, -self.config.mem_len:].detach())
+                new_mems.append(cur[:, -mem_len:].detach())
             x = layer(x, r, seg_mat, mask, mems[i] if mems else None)
         x = self.dropout(x)
         if use_mems:
