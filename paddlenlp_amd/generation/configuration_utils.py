@@ -26,6 +26,12 @@ class GenerationConfig:
     early_stopping: bool = False
     num_return_sequences: int = 1
     use_cache: bool = True
+    no_repeat_ngram_size: int = 0
+    forced_bos_token_id: Optional[int] = None
+    forced_eos_token_id: Optional[int] = None
+    bad_words_ids: Optional[List[List[int]]] = None
+    sequence_bias: Optional[dict] = None
+    prefix_allowed_tokens_fn: Optional[object] = None
     bos_token_id: Optional[int] = None
     eos_token_id: Optional[Union[int, List[int]]] = None
     pad_token_id: Optional[int] = None
@@ -39,8 +45,10 @@ class GenerationConfig:
 
     def save_pretrained(self, save_directory: str):
         os.makedirs(save_directory, exist_ok=True)
+        payload = {k: v for k, v in asdict(self).items()
+                   if not callable(v) and k != "prefix_allowed_tokens_fn"}
         with open(os.path.join(save_directory, GENERATION_CONFIG_NAME), "w") as f:
-            json.dump(asdict(self), f, indent=2)
+            json.dump(payload, f, indent=2)
 
     @classmethod
     def from_pretrained(cls, path: str):

@@ -12,9 +12,15 @@ import torch
 
 from .configuration_utils import GenerationConfig
 from .logits_process import (
+    ForcedBOSTokenLogitsProcessor,
+    ForcedEOSTokenLogitsProcessor,
     LogitsProcessorList,
     MinNewTokensLengthLogitsProcessor,
+    NoBadWordsLogitsProcessor,
+    NoRepeatNGramLogitsProcessor,
+    PrefixConstrainedLogitsProcessor,
     RepetitionPenaltyLogitsProcessor,
+    SequenceBiasLogitsProcessor,
     TemperatureLogitsWarper,
     TopKLogitsWarper,
     TopPLogitsWarper,
@@ -39,6 +45,24 @@ class GenerationMixin:
         if gen_config.min_new_tokens > 0 and gen_config.eos_ids():
             procs.append(MinNewTokensLengthLogitsProcessor(
                 prompt_len, gen_config.min_new_tokens, gen_config.eos_ids()))
+        if getattr(gen_config, "no_repeat_ngram_size", 0):
+            procs.append(NoRepeatNGramLogitsProcessor(
+                gen_config.no_repeat_ngram_size))
+        if getattr(gen_config, "forced_bos_token_id", None) is not None:
+            procs.append(ForcedBOSTokenLogitsProcessor(
+                prompt_len, gen_config.forced_bos_token_id))
+        if getattr(gen_config, "forced_eos_token_id", None) is not None \
+                and gen_config.max_new_tokens:
+            procs.append(ForcedEOSTokenLogitsProcessor(
+                prompt_len + gen_config.max_new_tokens,
+                gen_config.forced_eos_token_id))
+        if getattr(gen_config, "bad_words_ids", None):
+            procs.append(NoBadWordsLogitsProcessor(gen_config.bad_words_ids))
+        if getattr(gen_config, "sequence_bias", None):
+            procs.append(SequenceBiasLogitsProcessor(gen_config.sequence_bias))
+        if getattr(gen_config, "prefix_allowed_tokens_fn", None) is not None:
+            procs.append(PrefixConstrainedLogitsProcessor(
+                gen_config.prefix_allowed_tokens_fn, gen_config.num_beams))
         return procs
 
     def _get_logits_warpers(self, gen_config: GenerationConfig):

@@ -230,3 +230,71 @@ def test_num_return_sequences(model):
     with _pytest.raises(ValueError):
         model.generate(ids, GenerationConfig(max_new_tokens=2,
                                              num_return_sequences=2))
+
+
+def test_no_repeat_ngram_and_forced_tokens():
+    from paddlenlp_amd.generation.logits_process import (
+        ForcedBOSTokenLogitsProcessor,
+        ForcedEOSTokenLogitsProcessor,
+        NoRepeatNGramLogitsProcessor,
+    )
+
+    ids = torch.tensor([[1, 2, 3, 1, 2]])
+    logits = torch.zeros(1, 10)
+    out = NoRepeatNGramLogitsProcessor(3)(ids, logits.clone())
+    # prefix (1,2) previously continued with 3 -> 3 banned
+    assert out[0, 3] == -float("inf") and out[0, 4] == 0
+
+    bos = ForcedBOSTokenLogitsProcessor(prompt_len=5, bos_token_id=7)
+    out = bos(ids, torch.zeros(1, 10))
+    assert out[0, 7] == 0 and out[0, 0] == -float("inf")
+
+    eos = ForcedEOSTokenLogitsProcessor(max_total_len=6, eos_token_id=9)
+    out = eos(ids, torch.zeros(1, 10))
+    assert out[0, 9] == 0 and out[0, 1] == -float("inf")
+
+
+def test_bad_words_and_sequence_bias():
+    from paddlenlp_amd.generation.logits_process import (
+        NoBadWordsLogitsProcessor,
+        SequenceBiasLogitsProcessor,
+    )
+
+    ids = torch.tensor([[4, 5], [5, 6]])
+    logits = torch.zeros(2, 10)
+    # ban token 2 outright; ban "6 then 7" as a sequence
+    out = NoBadWordsLogitsProcessor([[2], [6, 7]])(ids, logits.clone())
+    assert (out[:, 2] == -float("inf")).all()
+    assert out[0, 7] == 0                    # row 0 tail is 5, allowed
+    assert out[1, 7] == -float("inf")        # row 1 tail is 6, banned
+
+    out = SequenceBiasLogitsProcessor({(3,): 2.5})(ids, torch.zeros(2, 10))
+    assert (out[:, 3] == 2.5).all()
+
+
+def test_prefix_constrained_generation():
+    from paddlenlp_amd.generation.logits_process import (
+        PrefixConstrainedLogitsProcessor,
+    )
+
+    proc = PrefixConstrainedLogitsProcessor(lambda b, ids: [1, 2], 1)
+    out = proc(torch.tensor([[0]]), torch.zeros(1, 5))
+    assert out[0, 1] == 0 and out[0, 3] == -float("inf")
+
+
+def test_generate_with_no_repeat_ngram(model):
+    torch.manual_seed(0)
+    ids = torch.randint(0, 96, (1, 6))
+    out, _ = model.generate(
+        ids, GenerationConfig(max_new_tokens=12, do_sample=False,
+                              no_repeat_ngram_size=2, pad_token_id=0))
+    toks = out[0].tolist()
+    seen = set()
+    full = ids[0].tolist() + toks
+    ok = True
+    for i in range(len(full) - 1):
+        pair = (full[i], full[i + 1])
+        if pair in seen:
+            ok = False
+        seen.add(pair)
+    assert ok, full
