@@ -1,6 +1,8 @@
 from .data_collator import (  # noqa: F401
     DataCollatorForLanguageModeling,
     DataCollatorForSeq2Seq,
+    DataCollatorForTokenClassification,
+    DataCollatorForWholeWordMask,
     DataCollatorWithPadding,
     default_data_collator,
 )

@@ -164,3 +164,66 @@ def test_vocab_and_greedy_tokenizer():
     assert tok.cut("深度学习") == ["深度", "学习"]  # longest match wins
     assert tok.cut("深度x") == ["深度", "x"]
     assert tok.encode("深度学习") == [0, 1]
+
+
+def test_mlm_collator_masks_and_labels():
+    import torch
+
+    from paddlenlp_amd.data import DataCollatorForLanguageModeling
+
+    class Tok:
+        pad_token_id = 0
+        cls_token_id = 1
+        sep_token_id = 2
+        mask_token_id = 3
+
+        def __len__(self):
+            return 50
+
+    torch.manual_seed(0)
+    feats = [{"input_ids": [1] + list(range(10, 28)) + [2]}
+             for _ in range(4)]
+    coll = DataCollatorForLanguageModeling(tokenizer=Tok(), mlm=True,
+                                           mlm_probability=0.5)
+    batch = coll(feats)
+    ids, labels = batch["input_ids"], batch["labels"]
+    # specials never masked
+    assert (labels[:, 0] == -100).all() and (labels[:, -1] == -100).all()
+    assert (ids[:, 0] == 1).all() and (ids[:, -1] == 2).all()
+    # some positions masked, labels hold the originals there
+    sel = labels != -100
+    assert sel.any()
+    orig = torch.tensor([f["input_ids"] for f in feats])
+    assert (labels[sel] == orig[sel]).all()
+    # masked positions mostly [MASK]
+    assert (ids[sel] == 3).float().mean() > 0.5
+
+
+def test_whole_word_mask_collator():
+    import torch
+
+    from paddlenlp_amd.data import DataCollatorForWholeWordMask
+
+    class Tok:
+        pad_token_id = 0
+        cls_token_id = None
+        sep_token_id = None
+        mask_token_id = 3
+
+        VOCAB = {10: "play", 11: "##ing", 12: "ball", 13: "##s", 14: "go"}
+
+        def __len__(self):
+            return 20
+
+        def convert_ids_to_tokens(self, ids):
+            return [self.VOCAB.get(i, f"tok{i}") for i in ids]
+
+    torch.manual_seed(1)
+    feats = [{"input_ids": [10, 11, 12, 13, 14]}]
+    coll = DataCollatorForWholeWordMask(tokenizer=Tok(), mlm_probability=0.99)
+    batch = coll(feats)
+    labels = batch["labels"][0]
+    # whole words masked together: 10/11 share fate, 12/13 share fate
+    assert (labels[0] == -100) == (labels[1] == -100)
+    assert (labels[2] == -100) == (labels[3] == -100)
+    assert (labels != -100).any()
