@@ -56,6 +56,9 @@ class ModelArgument:
 class DataArgument:
     dataset_name_or_path: str = field(default=None)
     max_length: int = field(default=2048)
+    # max prompt (source) length; the target gets the remaining budget
+    # (reference run_finetune DataArgument src_length)
+    src_length: int = field(default=1024)
     zero_padding: bool = field(default=False)
 
 
@@ -102,7 +105,9 @@ def main():
     except (FileNotFoundError, TypeError):
         pass
 
-    trans_fn = partial(convert_example, tokenizer=tokenizer, max_length=data_args.max_length)
+    trans_fn = partial(convert_example, tokenizer=tokenizer,
+                       max_length=data_args.max_length,
+                       src_length=data_args.src_length)
     train_ds = train_ds.map(trans_fn)
     if dev_ds is not None:
         dev_ds = dev_ds.map(trans_fn)

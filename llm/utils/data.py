@@ -9,7 +9,7 @@ from typing import Dict, List
 
 
 def convert_example(example: Dict, tokenizer, max_length: int = 2048,
-                    eos_token_id=None) -> Dict:
+                    src_length: int = None, eos_token_id=None) -> Dict:
     if "messages" in example:
         prompt_text = tokenizer.apply_chat_template(
             example["messages"][:-1], tokenize=False, add_generation_prompt=True)
@@ -26,6 +26,12 @@ def convert_example(example: Dict, tokenizer, max_length: int = 2048,
 
     prompt_ids = tokenizer.encode(prompt_text)
     full_ids = tokenizer.encode(full_text)
+    if src_length is not None and len(prompt_ids) > src_length:
+        # truncate the prompt head, keep its tail + the full target
+        # (reference src_length budgeting)
+        drop = len(prompt_ids) - src_length
+        prompt_ids = prompt_ids[drop:]
+        full_ids = full_ids[drop:]
     eos = eos_token_id if eos_token_id is not None else tokenizer.eos_token_id
     if eos is not None:
         full_ids = full_ids + [eos]
