@@ -29,8 +29,11 @@ def _run_workers(fn, world_size=WORLD, extra=()):
         p = ctx.Process(target=_worker_main, args=(fn.__module__, fn.__name__, rank, world_size, port, err_q, extra))
         p.start()
         procs.append(p)
+    # large worlds spawn world_size python processes that share the CPU
+    # with the rest of the suite: scale the join budget with the world
+    join_s = 180 + 60 * max(0, world_size - 2)
     for p in procs:
-        p.join(180)
+        p.join(join_s)
     errs = []
     while not err_q.empty():
         errs.append(err_q.get())
