@@ -115,7 +115,11 @@ def _chunked_attend(q, k, v, chunk: int, nbefore: int, causal: bool,
     if exclude_self:
         score = torch.where(diff == 0, torch.full_like(score, -1e5), score)
     lse = torch.logsumexp(score, dim=-1, keepdim=True)
-    out = (score - lse).exp() @ vc
+    probs = (score - lse).exp()
+    # renormalize: at the -1e5 self-penalty magnitude, fp32 lse rounding
+    # leaves probs summing to ~0.998 on self-only rows
+    probs = probs / probs.sum(dim=-1, keepdim=True).clamp_min(1e-20)
+    out = probs @ vc
     return out.reshape(B, H, S, D), lse.reshape(B, H, S)
 
 
