@@ -693,6 +693,41 @@ class Trainer:
             self.state.best_metric = value
             self.state.best_model_checkpoint = self._checkpoint_dir()
 
+    def log_metrics(self, split: str, metrics: Dict[str, float]):
+        """Pretty-print a metrics dict (reference log_metrics helper)."""
+        if not self.args.process_index == 0:
+            return
+        logger.info(f"***** {split} metrics *****")
+        for k in sorted(metrics):
+            logger.info(f"  {k} = {metrics[k]}")
+
+    def save_metrics(self, split: str, metrics: Dict[str, float]):
+        """Write metrics to <output_dir>/<split>_results.json and append to
+        all_results.json (reference save_metrics helper)."""
+        if not self.args.process_index == 0:
+            return
+        import json as _json
+
+        path = os.path.join(self.args.output_dir, f"{split}_results.json")
+        os.makedirs(self.args.output_dir, exist_ok=True)
+        with open(path, "w") as f:
+            _json.dump(metrics, f, indent=2, sort_keys=True)
+        all_path = os.path.join(self.args.output_dir, "all_results.json")
+        merged = {}
+        if os.path.isfile(all_path):
+            with open(all_path) as f:
+                merged = _json.load(f)
+        merged.update(metrics)
+        with open(all_path, "w") as f:
+            _json.dump(merged, f, indent=2, sort_keys=True)
+
+    def save_state(self):
+        """Persist trainer_state.json to output_dir (reference helper)."""
+        if self.args.process_index == 0:
+            os.makedirs(self.args.output_dir, exist_ok=True)
+            self.state.save_to_json(
+                os.path.join(self.args.output_dir, TRAINER_STATE_NAME))
+
     def log(self, logs: Dict[str, float]):
         logs["epoch"] = round(self.state.epoch, 4)
         self.state.log_history.append(dict(logs))

@@ -204,3 +204,20 @@ def test_load_best_model_at_end():
         live = {k: v for k, v in model.state_dict().items()}
         for k, v in saved.items():
             assert torch.equal(v, live[k].to(v.dtype)), k
+
+
+def test_log_save_metrics_and_state():
+    with tempfile.TemporaryDirectory() as d:
+        tr = Trainer(model=tiny_model(), args=make_args(d),
+                     train_dataset=RandDS(n=8))
+        m = {"eval_loss": 1.5, "eval_accuracy": 0.75}
+        tr.log_metrics("eval", m)
+        tr.save_metrics("eval", m)
+        tr.save_metrics("train", {"train_loss": 2.0})
+        with open(os.path.join(d, "eval_results.json")) as f:
+            assert json.load(f) == m
+        with open(os.path.join(d, "all_results.json")) as f:
+            merged = json.load(f)
+        assert merged["eval_loss"] == 1.5 and merged["train_loss"] == 2.0
+        tr.save_state()
+        assert os.path.isfile(os.path.join(d, "trainer_state.json"))
