@@ -58,7 +58,17 @@ class BasePredictor:
         self.config = config
         self.tokenizer = tokenizer
 
+    def _apply_chat_template(self, item):
+        """Accept raw strings or chat `messages` lists; message lists run
+        through the tokenizer's chat template (reference predictor applies
+        chat_template when the input is a conversation)."""
+        if isinstance(item, str):
+            return item
+        return self.tokenizer.apply_chat_template(
+            list(item), tokenize=False, add_generation_prompt=True)
+
     def _preprocess(self, texts: List[str]):
+        texts = [self._apply_chat_template(t) for t in texts]
         out = self.tokenizer(list(texts), padding=True, return_tensors="pt",
                              truncation=True, max_length=self.config.src_length)
         return out
@@ -213,6 +223,7 @@ class BlockInferencePredictor(BasePredictor):
             inserted = []
             while pending:
                 req_id, text = pending[0]
+                text = self._apply_chat_template(text)
                 ids = tok.encode(text)[-cfg.src_length:]
                 slot = mgr.allocate_slot(len(ids))
                 if slot is None:

@@ -436,3 +436,38 @@ def test_static_export_roundtrip(tmp_path):
             ref = ref[0] if isinstance(ref, tuple) else ref
             got = fn(ids)
         assert torch.allclose(got, ref, atol=1e-5), (B, S, (got - ref).abs().max())
+
+
+def test_predictor_accepts_chat_messages():
+    """A `messages` list runs through the tokenizer chat template before
+    tokenization (reference predictor chat_template handling)."""
+    from llm.predict.predictor import BasePredictor
+
+    class Tok:
+        chat_template = True
+
+        def apply_chat_template(self, conv, tokenize=False,
+                                add_generation_prompt=True):
+            return "|".join(m["content"] for m in conv) + "|ASSISTANT:"
+
+        def __call__(self, texts, **kw):
+            self.last = list(texts)
+            import torch
+
+            return {"input_ids": torch.zeros(len(texts), 3,
+                                             dtype=torch.long)}
+
+    class Cfg:
+        src_length = 64
+
+    p = BasePredictor.__new__(BasePredictor)
+    p.tokenizer = Tok()
+    p.config = Cfg()
+    out = p._preprocess([
+        [{"role": "user", "content": "hi"},
+         {"role": "assistant", "content": "yo"},
+         {"role": "user", "content": "q"}],
+        "plain text",
+    ])
+    assert p.tokenizer.last[0] == "hi|yo|q|ASSISTANT:"
+    assert p.tokenizer.last[1] == "plain text"
