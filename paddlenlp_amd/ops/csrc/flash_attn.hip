@@ -1025,10 +1025,25 @@ static void flash_fwd_t(const void* q, const void* k, const void* v, void* o,
                        causal ? 1 : 0);
 }
 
+bool launch_flash_fwd2_mask(const void* q, const void* k, const void* v,
+                            void* o, float* lse, const int* startend,
+                            int B, int Sq, int Skv, int Hq, int Hk,
+                            int D, float scale, hipStream_t stream);
+
 void launch_flash_fwd_mask(const void* q, const void* k, const void* v, void* o,
                            float* lse, const int* startend,
                            int B, int Sq, int Skv, int Hq, int Hk, int D,
                            float scale, hipStream_t stream) {
+    // v2 masked kernel on the D=128 hot path (PNLP_FLASHMASK_KERNEL=1
+    // forces the v1 fallback for A/B)
+    static const bool force_v1 = [] {
+        const char* e = getenv("PNLP_FLASHMASK_KERNEL");
+        return e && e[0] == '1';
+    }();
+    if (!force_v1 &&
+        launch_flash_fwd2_mask(q, k, v, o, lse, startend,
+                               B, Sq, Skv, Hq, Hk, D, scale, stream))
+        return;
     if (D == 128) flash_fwd_t<128, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
     else if (D == 64) flash_fwd_t<64, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
     else if (D == 32) flash_fwd_t<32, 1, false>(q, k, v, o, lse, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
@@ -1117,11 +1132,26 @@ static void flash_bwd_t(const void* dout, const void* q, const void* k, const vo
                        startend, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
 }
 
+bool launch_flash_bwd2_mask(const void* dout, const void* q, const void* k,
+                            const void* v, const void* o, const float* lse,
+                            float* delta, void* dq, void* dk, void* dv,
+                            const int* startend,
+                            int B, int Sq, int Skv, int Hq, int Hk, int D,
+                            float scale, hipStream_t stream);
+
 void launch_flash_bwd_mask(const void* dout, const void* q, const void* k, const void* v,
                            const void* o, const float* lse, float* delta,
                            void* dq, void* dk, void* dv, const int* startend,
                            int B, int Sq, int Skv, int Hq, int Hk, int D,
                            float scale, hipStream_t stream) {
+    static const bool force_v1 = [] {
+        const char* e = getenv("PNLP_FLASHMASK_KERNEL");
+        return e && e[0] == '1';
+    }();
+    if (!force_v1 &&
+        launch_flash_bwd2_mask(dout, q, k, v, o, lse, delta, dq, dk, dv,
+                               startend, B, Sq, Skv, Hq, Hk, D, scale, stream))
+        return;
     if (D == 128) flash_bwd_t<128, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
     else if (D == 64) flash_bwd_t<64, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);
     else if (D == 32) flash_bwd_t<32, false>(dout, q, k, v, o, lse, delta, dq, dk, dv, B, Sq, Skv, Hq, Hk, scale, true, stream, startend);

@@ -105,12 +105,12 @@ __global__ void flash_bwd2_delta_kernel(
 // delta are per-lane scalars (q = lane&31) and the T12 swap hands dS to the
 // dQ MFMA directly.
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool MASKED = false>
 __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    ushort_t* __restrict__ dq,
+    ushort_t* __restrict__ dq, const int* __restrict__ startend,
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int BLKN = 64;
     constexpr int DSTEPS = D / 16;
@@ -121,6 +121,7 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     __shared__ ushort_t k_lds[2][BLKN * D];    // row-major [kv][d]
     __shared__ ushort_t v_lds[2][BLKN * D];    // row-major [kv][d]
     __shared__ ushort_t kt_lds[2][D * BLKN];   // transposed [d][kv]
+    __shared__ int se_lds[2][MASKED ? BLKN : 1];  // FlashMask bounds
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -202,7 +203,12 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
             sv1 = *reinterpret_cast<const short8v*>(v_ptr + (long long)(g0 + 1) * kv_row_stride + s_col);
         }
     };
-    auto write_tile = [&](int buf) {
+    auto write_tile = [&](int buf, int kv_base) {
+        if (MASKED && tid < BLKN) {
+            int g = kv_base + tid;
+            se_lds[buf][tid] =
+                (g < Skv) ? startend[(long long)b * Skv + g] : 0;
+        }
         *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
         *reinterpret_cast<short8v*>(swzb2<D>(v_lds[buf], s_row0, s_col)) = sv0;
@@ -216,7 +222,7 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     };
 
     load_tile(0);
-    write_tile(0);
+    write_tile(0, 0);
     __syncthreads();
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
@@ -247,9 +253,11 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
             // scale (all per-lane: q = l32)
 #pragma unroll
             for (int r = 0; r < 16; r++) {
-                int kvg = kv_base + sub * 32 + crow32b(r, hi);
+                int row = sub * 32 + crow32b(r, hi);
+                int kvg = kv_base + row;
                 bool vis = (kvg < Skv) && (qg < Sq) &&
                            (!causal || kvg <= qg + causal_off);
+                if (MASKED) vis = vis && (qg < se_lds[cur][row]);
                 float p = vis ? __expf(st[r] * scale - lse_q) : 0.f;
                 st[r] = p * (dp[r] - dl_q) * scale;  // = dS^T
             }
@@ -275,7 +283,7 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
         if (kvt + 1 < n_kv_tiles) load_tile(kv_base + BLKN);
         if (!wave_skip) process_sub(1);
 
-        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1);
+        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1, kv_base + BLKN);
         __syncthreads();
     }
 
@@ -296,12 +304,13 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
 // in tiles of 64 across the whole GQA group.  Orientation: S and dP in
 // C-layout [q][kv] (col = kv) so T12 feeds dV += P^T·dO and dK += dS^T·Q.
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool MASKED = false>
 __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
     const ushort_t* __restrict__ q, const ushort_t* __restrict__ k,
     const ushort_t* __restrict__ v, const ushort_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     ushort_t* __restrict__ dk, ushort_t* __restrict__ dv,
+    const int* __restrict__ startend,
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int BLKQ = 64;               // q tile
     constexpr int DSTEPS = D / 16;
@@ -339,6 +348,11 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
     const int kvw = kv_base + wave * 32;     // this wave's first kv row
     const int kvg_lane = kvw + l32;          // lane's kv column
     const int causal_off = Skv - Sq;
+    // FlashMask: lane owns one kv column -> one bound scalar
+    int kv_end_lane = 0x7fffffff;
+    if (MASKED)
+        kv_end_lane = (kvg_lane < Skv)
+            ? startend[(long long)b * Skv + kvg_lane] : 0;
 
     const long long q_row_stride = (long long)Hq * D;
     const long long kv_row_stride = (long long)Hk * D;
@@ -459,6 +473,7 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
                         float dl = dl_lds[cur][qrow];
                         bool vis = (qgl < Sq) && (kvg_lane < Skv) &&
                                    (!causal || kvg_lane <= qgl + causal_off);
+                        if (MASKED) vis = vis && (qgl < kv_end_lane);
                         float p = (vis && ls != INFINITY)
                                       ? __expf(st[r] * scale - ls) : 0.f;
                         st[r] = p;
@@ -529,14 +544,41 @@ bool launch_flash_bwd2(const void* dout, const void* q, const void* k, const voi
     hipLaunchKernelGGL(flash_bwd2_delta_kernel<128>, dim3(dgrid), dim3(FB2_BLOCK), 0, stream,
                        (const ushort_t*)dout, (const ushort_t*)o, delta, B, Sq, Hq);
     dim3 gq((Sq + 255) / 256, B * Hq);
-    hipLaunchKernelGGL((flash_bwd2_dq_kernel<128>), gq, dim3(FB2_BLOCK), 0, stream,
+    hipLaunchKernelGGL((flash_bwd2_dq_kernel<128, false>), gq, dim3(FB2_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dq,
+                       (const int*)nullptr,
                        B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
     dim3 gkv((Skv + 255) / 256, B * Hk);
-    hipLaunchKernelGGL((flash_bwd2_dkv_kernel<128>), gkv, dim3(FB2_BLOCK), 0, stream,
+    hipLaunchKernelGGL((flash_bwd2_dkv_kernel<128, false>), gkv, dim3(FB2_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
                        (const ushort_t*)dout, lse, delta, (ushort_t*)dk, (ushort_t*)dv,
+                       (const int*)nullptr,
                        B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+    return true;
+}
+
+bool launch_flash_bwd2_mask(const void* dout, const void* q, const void* k,
+                            const void* v, const void* o, const float* lse,
+                            float* delta, void* dq, void* dk, void* dv,
+                            const int* startend,
+                            int B, int Sq, int Skv, int Hq, int Hk, int D,
+                            float scale, hipStream_t stream) {
+    if (D != 128 || (Hq % Hk) != 0) return false;
+    long long rows = (long long)B * Sq * Hq;
+    int waves_per_block = FB2_BLOCK / 64;
+    int dgrid = (int)((rows + waves_per_block - 1) / waves_per_block);
+    hipLaunchKernelGGL(flash_bwd2_delta_kernel<128>, dim3(dgrid), dim3(FB2_BLOCK), 0, stream,
+                       (const ushort_t*)dout, (const ushort_t*)o, delta, B, Sq, Hq);
+    dim3 gq((Sq + 255) / 256, B * Hq);
+    hipLaunchKernelGGL((flash_bwd2_dq_kernel<128, true>), gq, dim3(FB2_BLOCK), 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dout, lse, delta, (ushort_t*)dq, startend,
+                       B, Sq, Skv, Hq, Hk, scale, 1);
+    dim3 gkv((Skv + 255) / 256, B * Hk);
+    hipLaunchKernelGGL((flash_bwd2_dkv_kernel<128, true>), gkv, dim3(FB2_BLOCK), 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dout, lse, delta, (ushort_t*)dk, (ushort_t*)dv,
+                       startend, B, Sq, Skv, Hq, Hk, scale, 1);
     return true;
 }

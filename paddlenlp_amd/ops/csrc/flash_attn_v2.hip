@@ -95,13 +95,14 @@ __global__ void permlane_probe(int* out0, int* out1) {
 // ---------------------------------------------------------------------------
 // forward v2
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool MASKED = false>
 __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     const ushort_t* __restrict__ q,   // [B, Sq, Hq, D]
     const ushort_t* __restrict__ k,   // [B, Skv, Hk, D]
     const ushort_t* __restrict__ v,   // [B, Skv, Hk, D]
     ushort_t* __restrict__ o,         // [B, Sq, Hq, D]
     float* __restrict__ lse,          // [B, Hq, Sq]
+    const int* __restrict__ startend, // FlashMask [B, Skv] (MASKED only)
     int B, int Sq, int Skv, int Hq, int Hk, float scale, int causal) {
     constexpr int DSTEPS = D / 16;    // K-steps of the S^T MFMAs
     constexpr int NDT = D / 32;       // 32-col d-tiles of O
@@ -112,6 +113,8 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     // drain before s_barrier is the dominant structural stall, guide §5)
     __shared__ ushort_t k_lds[2][FA2_BLKN * D];
     __shared__ ushort_t vt_lds[2][D * FA2_BLKN];
+    // per-tile FlashMask column bounds (staged with the K/V tile)
+    __shared__ int se_lds[2][MASKED ? FA2_BLKN : 1];
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -191,7 +194,12 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             sv1 = *reinterpret_cast<const short8v*>(v_ptr + (long long)(g0 + 1) * kv_row_stride + s_col);
         }
     };
-    auto write_tile = [&](int buf) {
+    auto write_tile = [&](int buf, int kv_base) {
+        if (MASKED && tid < FA2_BLKN) {
+            int g = kv_base + tid;
+            se_lds[buf][tid] =
+                (g < Skv) ? startend[(long long)b * Skv + g] : 0;
+        }
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
 #pragma unroll
@@ -203,7 +211,7 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     };
 
     load_tile(0);
-    write_tile(0);
+    write_tile(0, 0);
     __syncthreads();
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
@@ -241,7 +249,7 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             // mask + scale.  A tile is "full" when every (q, kv) pair in this
             // wave's sub-block is visible — the common case away from the
             // diagonal; masking math is skipped entirely.
-            const bool full_tile =
+            const bool full_tile = !MASKED &&
                 (kv_base + FA2_BLKN <= Skv) &&
                 (!causal || (kv_base + FA2_BLKN - 1 <= qw + causal_off));
             if (full_tile) {
@@ -254,9 +262,11 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
                 for (int nt = 0; nt < KVT; nt++)
 #pragma unroll
                     for (int r = 0; r < 16; r++) {
-                        int kvg = kv_base + nt * 32 + crow32(r, hi);
+                        int row = nt * 32 + crow32(r, hi);
+                        int kvg = kv_base + row;
                         bool vis = (kvg < Skv) &&
                                    (!causal || kvg <= qg + causal_off);
+                        if (MASKED) vis = vis && (qg < se_lds[cur][row]);
                         st[nt][r] = vis ? st[nt][r] * scale : -INFINITY;
                     }
             }
@@ -356,7 +366,7 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
         // still be in flight on other waves, then one barrier: it both
         // publishes buf[cur^1] and guarantees every wave has finished
         // reading buf[cur] before iteration t+1 overwrites it
-        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1);
+        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1, kv_base + FA2_BLKN);
         __syncthreads();
     }
 
@@ -396,8 +406,22 @@ bool launch_flash_fwd2(const void* q, const void* k, const void* v, void* o,
                        int D, float scale, bool causal, hipStream_t stream) {
     if (D != 128 || (Hq % Hk) != 0) return false;
     dim3 grid((Sq + FA2_BLKM - 1) / FA2_BLKM, B * Hq);
-    hipLaunchKernelGGL((flash_fwd2_kernel<128>), grid, dim3(FA2_BLOCK), 0, stream,
+    hipLaunchKernelGGL((flash_fwd2_kernel<128, false>), grid, dim3(FA2_BLOCK), 0, stream,
                        (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
-                       (ushort_t*)o, lse, B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+                       (ushort_t*)o, lse, (const int*)nullptr,
+                       B, Sq, Skv, Hq, Hk, scale, causal ? 1 : 0);
+    return true;
+}
+
+bool launch_flash_fwd2_mask(const void* q, const void* k, const void* v,
+                            void* o, float* lse, const int* startend,
+                            int B, int Sq, int Skv, int Hq, int Hk,
+                            int D, float scale, hipStream_t stream) {
+    if (D != 128 || (Hq % Hk) != 0) return false;
+    dim3 grid((Sq + FA2_BLKM - 1) / FA2_BLKM, B * Hq);
+    hipLaunchKernelGGL((flash_fwd2_kernel<128, true>), grid, dim3(FA2_BLOCK), 0,
+                       stream, (const ushort_t*)q, (const ushort_t*)k,
+                       (const ushort_t*)v, (ushort_t*)o, lse, startend,
+                       B, Sq, Skv, Hq, Hk, scale, 1);
     return true;
 }
