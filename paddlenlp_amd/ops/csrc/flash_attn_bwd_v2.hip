@@ -122,6 +122,7 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     __shared__ ushort_t v_lds[2][BLKN * D];    // row-major [kv][d]
     __shared__ ushort_t kt_lds[2][D * BLKN];   // transposed [d][kv]
     __shared__ int se_lds[2][MASKED ? BLKN : 1];  // FlashMask bounds
+    __shared__ int se_max[2];                     // tile max (skip bound)
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -205,9 +206,14 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     };
     auto write_tile = [&](int buf, int kv_base) {
         if (MASKED && tid < BLKN) {
-            int g = kv_base + tid;
-            se_lds[buf][tid] =
-                (g < Skv) ? startend[(long long)b * Skv + g] : 0;
+            int val = (kv_base + tid < Skv)
+                ? startend[(long long)b * Skv + kv_base + tid] : 0;
+            se_lds[buf][tid] = val;
+            int mx = val;
+#pragma unroll
+            for (int off = 32; off; off >>= 1)
+                mx = max(mx, __shfl_xor(mx, off, 64));
+            if (tid == 0) se_max[buf] = mx;
         }
         *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
@@ -228,8 +234,11 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
         const int kv_base = kvt * BLKN;
         const int cur = kvt & 1;
-        const bool wave_skip =
+        bool wave_skip =
             causal && (kv_base > qw + 31 + causal_off);
+        // FlashMask whole-tile skip (see fwd): no bound exceeds this
+        // wave's first q row -> nothing visible
+        if (MASKED && qw >= se_max[cur]) wave_skip = true;
 
         // two 32-kv sub-iterations: one (S^T, dP^T) register pair live at
         // a time (register-pressure; see dkv kernel note)
@@ -348,11 +357,19 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
     const int kvw = kv_base + wave * 32;     // this wave's first kv row
     const int kvg_lane = kvw + l32;          // lane's kv column
     const int causal_off = Skv - Sq;
-    // FlashMask: lane owns one kv column -> one bound scalar
+    // FlashMask: lane owns one kv column -> one bound scalar; the
+    // wave-max bound drives whole-q-tile skipping
     int kv_end_lane = 0x7fffffff;
-    if (MASKED)
+    int wave_max_end = 0x7fffffff;
+    if (MASKED) {
         kv_end_lane = (kvg_lane < Skv)
             ? startend[(long long)b * Skv + kvg_lane] : 0;
+        wave_max_end = kv_end_lane;
+#pragma unroll
+        for (int off = 32; off; off >>= 1)
+            wave_max_end = max(wave_max_end,
+                               __shfl_xor(wave_max_end, off, 64));
+    }
 
     const long long q_row_stride = (long long)Hq * D;
     const long long kv_row_stride = (long long)Hk * D;
@@ -437,9 +454,11 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
         for (int qt = qt0; qt < n_q_tiles; qt++) {
             const int q_tb = qt * BLKQ;
             const int cur = qt & 1;
-            // skip q tiles fully below this wave's causal diagonal
-            const bool wave_skip =
+            // skip q tiles fully below this wave's causal diagonal, or
+            // (FlashMask) entirely past every bound this wave holds
+            bool wave_skip =
                 causal && (q_tb + BLKQ - 1 + causal_off < kvw);
+            if (MASKED && q_tb >= wave_max_end) wave_skip = true;
 
             // process the 64-q tile as two 32-q sub-iterations: only one
             // (S, dP) register pair is live at a time, which keeps the

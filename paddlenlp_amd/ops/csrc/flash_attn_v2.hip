@@ -113,8 +113,11 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     // drain before s_barrier is the dominant structural stall, guide §5)
     __shared__ ushort_t k_lds[2][FA2_BLKN * D];
     __shared__ ushort_t vt_lds[2][D * FA2_BLKN];
-    // per-tile FlashMask column bounds (staged with the K/V tile)
+    // per-tile FlashMask column bounds (staged with the K/V tile) and
+    // the tile's max bound (wave 0 reduces it during staging) for
+    // whole-tile skipping: a wave with qw >= max end sees nothing
     __shared__ int se_lds[2][MASKED ? FA2_BLKN : 1];
+    __shared__ int se_max[2];
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -196,9 +199,14 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     };
     auto write_tile = [&](int buf, int kv_base) {
         if (MASKED && tid < FA2_BLKN) {
-            int g = kv_base + tid;
-            se_lds[buf][tid] =
-                (g < Skv) ? startend[(long long)b * Skv + g] : 0;
+            int val = (kv_base + tid < Skv)
+                ? startend[(long long)b * Skv + kv_base + tid] : 0;
+            se_lds[buf][tid] = val;
+            int mx = val;
+#pragma unroll
+            for (int off = 32; off; off >>= 1)
+                mx = max(mx, __shfl_xor(mx, off, 64));
+            if (tid == 0) se_max[buf] = mx;
         }
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
@@ -219,8 +227,11 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
         const int cur = kvt & 1;
         // a wave whose q rows all precede this kv tile contributes nothing:
         // skip its compute but keep it in the staging barriers
-        const bool wave_skip =
+        bool wave_skip =
             causal && (kv_base > qw + FA2_QW - 1 + causal_off);
+        // FlashMask whole-tile skip: every kv bound in this tile is at or
+        // below the wave's first q row -> no visible pair
+        if (MASKED && qw >= se_max[cur]) wave_skip = true;
 
         f32x16 st[KVT];
         if (!wave_skip) {
