@@ -364,3 +364,38 @@ def test_flash_attention_long_context_8k(C):
                         (v.grad, vr.grad, "dv")):
         err = (g.float() - gr).abs().max().item()
         assert err < 8e-2, (name, err)
+
+
+def test_flashmask_irregular_packing_s1024(C):
+    """FlashMask v2 with tile-skipping: randomized, tile-unaligned sample
+    boundaries across 16 kv tiles must match the fp32 reference (fwd+bwd)."""
+    from paddlenlp_amd import ops
+
+    torch.manual_seed(3)
+    B, S, Hq, Hk, D = 2, 1024, 8, 2, 128
+    q = _bf16(torch.randn(B, S, Hq, D, device="cuda")).requires_grad_()
+    k = _bf16(torch.randn(B, S, Hk, D, device="cuda")).requires_grad_()
+    v = _bf16(torch.randn(B, S, Hk, D, device="cuda")).requires_grad_()
+    # irregular boundaries, different per batch row
+    se = torch.empty(B, 1, S, 1, dtype=torch.int32, device="cuda")
+    for bi, bounds in enumerate(([37, 191, 500, 501, 777, 1024],
+                                 [64, 640, 1024])):
+        lo = 0
+        for hi in bounds:
+            se[bi, 0, lo:hi, 0] = hi
+            lo = hi
+    out = ops.flash_attention(q, k, v, causal=True, startend_row_indices=se)
+    qr = q.detach().float().requires_grad_()
+    kr = k.detach().float().requires_grad_()
+    vr = v.detach().float().requires_grad_()
+    ref = ops.reference.flash_attention(qr, kr, vr, causal=True,
+                                        startend_row_indices=se)
+    assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), \
+        (out.float() - ref).abs().max()
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    ref.backward(dy.float())
+    for g, gr, n in ((q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                     (v.grad, vr.grad, "dv")):
+        err = (g.float() - gr).abs().max().item()
+        assert err < 8e-2, (n, err)
