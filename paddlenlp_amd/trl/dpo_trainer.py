@@ -57,10 +57,75 @@ class DPOTrainer(Trainer):
                 self.reference_model = self.reference_model.to(torch.bfloat16)
         return model
 
+    # -- packed (FlashMask) forward: chosen and rejected share one row --
+    def _pack_pair(self, inputs):
+        """[chosen ; rejected] per row with FlashMask startend bounds so
+        the two segments cannot attend to each other (the reference's
+        run_dpo packing).  Returns ids, labels, bounds, per-row split."""
+        ch_ids, rj_ids = inputs["chosen_input_ids"], inputs["rejected_input_ids"]
+        ch_lab, rj_lab = inputs["chosen_labels"], inputs["rejected_labels"]
+        B = ch_ids.shape[0]
+        Lc, Lr = ch_ids.shape[1], rj_ids.shape[1]
+        S = Lc + Lr
+        ids = torch.cat([ch_ids, rj_ids], dim=1)
+        labels = torch.cat([ch_lab, rj_lab], dim=1)
+        se = torch.empty(B, 1, S, 1, dtype=torch.int32, device=ids.device)
+        se[:, 0, :Lc, 0] = Lc
+        se[:, 0, Lc:, 0] = S
+        return ids, labels, se, Lc
+
     def compute_loss(self, model, inputs, return_outputs=False):
-        """inputs: chosen_input_ids/chosen_labels/rejected_input_ids/rejected_labels
-        — concatenated into one forward (reference concatenated_forward)."""
+        """inputs: chosen_input_ids/chosen_labels/rejected_input_ids/rejected_labels.
+        Packed into ONE row per pair with FlashMask segment bounds
+        (reference run_dpo packing; set dpo_packing=False on the trainer
+        to fall back to batch-concat rows)."""
         B = inputs["chosen_input_ids"].shape[0]
+        if getattr(self, "dpo_packing", None) is None:
+            # packing needs a model that explicitly implements the
+            # FlashMask kwarg (a **kwargs sink would silently DROP the
+            # segment mask and leak cross-pair attention)
+            import inspect
+
+            from ..transformers.model_utils import unwrap_model
+
+            fwd = unwrap_model(self.model).forward
+            self.dpo_packing = ("attn_mask_startend_row_indices"
+                                in inspect.signature(fwd).parameters)
+        if self.dpo_packing:
+            ids, labels, se, Lc = self._pack_pair(inputs)
+            logits = model(input_ids=ids,
+                           attn_mask_startend_row_indices=se)
+            if isinstance(logits, tuple):
+                logits = logits[0]
+            lab_c = labels.clone(); lab_c[:, Lc:] = -100
+            lab_r = labels.clone(); lab_r[:, :Lc] = -100
+            policy_chosen = sequence_logprob(logits, lab_c,
+                                             average=self.average_logps)
+            policy_rejected = sequence_logprob(logits, lab_r,
+                                               average=self.average_logps)
+            ref_chosen = ref_rejected = None
+            if self.reference_model is not None:
+                with torch.no_grad():
+                    rl = self.reference_model(
+                        input_ids=ids, attn_mask_startend_row_indices=se)
+                    if isinstance(rl, tuple):
+                        rl = rl[0]
+                    ref_chosen = sequence_logprob(rl, lab_c,
+                                                  average=self.average_logps)
+                    ref_rejected = sequence_logprob(rl, lab_r,
+                                                    average=self.average_logps)
+            loss, chosen_r, rejected_r = self.dpo_criterion(
+                policy_chosen, policy_rejected, ref_chosen, ref_rejected)
+            if self.loss_type == "orpo":
+                nll = -sequence_logprob(logits, lab_c, average=True).mean()
+                loss = loss + nll
+            if return_outputs:
+                return loss, {
+                    "rewards/chosen": chosen_r.mean(),
+                    "rewards/rejected": rejected_r.mean(),
+                    "rewards/accuracy": (chosen_r > rejected_r).float().mean(),
+                }
+            return loss
         maxlen = max(inputs["chosen_input_ids"].shape[1], inputs["rejected_input_ids"].shape[1])
 
         def pad(t, fill):

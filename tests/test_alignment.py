@@ -135,3 +135,49 @@ def test_ppo_step():
     # reference stayed frozen
     for p in reference.parameters():
         assert not p.requires_grad
+
+
+def test_dpo_packed_equals_rowwise():
+    """Packed (FlashMask one-row) DPO forward must produce the same loss
+    as the batch-concat row form: cross-segment attention is masked and
+    RoPE is relative, so the math is identical."""
+    import copy
+
+    import torch
+
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+    from paddlenlp_amd.trl.dpo_trainer import DPOTrainer
+    from paddlenlp_amd.trainer import TrainingArguments
+    import tempfile
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=96, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=64,
+                      use_flash_attention=True)
+    model = LlamaForCausalLM.from_config(cfg)
+
+    ch = torch.randint(2, 96, (2, 10))
+    rj = torch.randint(2, 96, (2, 12))
+    inputs = {
+        "chosen_input_ids": ch,
+        "chosen_labels": torch.cat([torch.full((2, 4), -100), ch[:, 4:]],
+                                   dim=1),
+        "rejected_input_ids": rj,
+        "rejected_labels": torch.cat([torch.full((2, 4), -100), rj[:, 4:]],
+                                     dim=1),
+    }
+    with tempfile.TemporaryDirectory() as d:
+        args = TrainingArguments(output_dir=d, max_steps=1,
+                                 per_device_train_batch_size=2)
+        tr = DPOTrainer(model=model, args=args, beta=0.1,
+                        loss_type="sigmoid",
+                        reference_model=copy.deepcopy(model))
+        tr.dpo_packing = True
+        packed = tr.compute_loss(model, {k: v.clone()
+                                         for k, v in inputs.items()})
+        tr.dpo_packing = False
+        rowwise = tr.compute_loss(model, {k: v.clone()
+                                          for k, v in inputs.items()})
+    assert torch.allclose(packed, rowwise, atol=1e-5), \
+        (float(packed), float(rowwise))
