@@ -80,7 +80,7 @@ class DPOTrainer(Trainer):
         (reference run_dpo packing; set dpo_packing=False on the trainer
         to fall back to batch-concat rows)."""
         B = inputs["chosen_input_ids"].shape[0]
-        if getattr(self, "dpo_packing", None) is None:
+        if getattr(self, "_can_pack", None) is None:
             # packing needs a model that explicitly implements the
             # FlashMask kwarg (a **kwargs sink would silently DROP the
             # segment mask and leak cross-pair attention)
@@ -89,9 +89,18 @@ class DPOTrainer(Trainer):
             from ..transformers.model_utils import unwrap_model
 
             fwd = unwrap_model(self.model).forward
-            self.dpo_packing = ("attn_mask_startend_row_indices"
-                                in inspect.signature(fwd).parameters)
-        if self.dpo_packing:
+            self._can_pack = ("attn_mask_startend_row_indices"
+                              in inspect.signature(fwd).parameters)
+        packing = getattr(self, "dpo_packing", None)
+        if packing is None:
+            # adaptive default: the packed row costs ~(Lc+Lr) per pair,
+            # the row-concat form 2*max(Lc, Lr) (both sides padded to the
+            # longer) — pack when padding waste exceeds ~10% (measured:
+            # at equal lengths the row form is ~10% faster on MI355X)
+            Lc = inputs["chosen_input_ids"].shape[1]
+            Lr = inputs["rejected_input_ids"].shape[1]
+            packing = (Lc + Lr) < 2 * max(Lc, Lr) * 0.9
+        if packing and self._can_pack:
             ids, labels, se, Lc = self._pack_pair(inputs)
             logits = model(input_ids=ids,
                            attn_mask_startend_row_indices=se)
