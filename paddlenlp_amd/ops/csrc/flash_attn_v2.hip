@@ -117,7 +117,8 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     // the tile's max bound (wave 0 reduces it during staging) for
     // whole-tile skipping: a wave with qw >= max end sees nothing
     __shared__ int se_lds[2][MASKED ? FA2_BLKN : 1];
-    __shared__ int se_max[2];
+    __shared__ int se_max[2];   // tile max bound: whole-tile skip
+    __shared__ int se_min[2];   // tile min bound: full-tile fast path
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -202,11 +203,13 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             int val = (kv_base + tid < Skv)
                 ? startend[(long long)b * Skv + kv_base + tid] : 0;
             se_lds[buf][tid] = val;
-            int mx = val;
+            int mx = val, mn = val;
 #pragma unroll
-            for (int off = 32; off; off >>= 1)
+            for (int off = 32; off; off >>= 1) {
                 mx = max(mx, __shfl_xor(mx, off, 64));
-            if (tid == 0) se_max[buf] = mx;
+                mn = min(mn, __shfl_xor(mn, off, 64));
+            }
+            if (tid == 0) { se_max[buf] = mx; se_min[buf] = mn; }
         }
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
@@ -260,9 +263,10 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             // mask + scale.  A tile is "full" when every (q, kv) pair in this
             // wave's sub-block is visible — the common case away from the
             // diagonal; masking math is skipped entirely.
-            const bool full_tile = !MASKED &&
+            const bool full_tile =
                 (kv_base + FA2_BLKN <= Skv) &&
-                (!causal || (kv_base + FA2_BLKN - 1 <= qw + causal_off));
+                (!causal || (kv_base + FA2_BLKN - 1 <= qw + causal_off)) &&
+                (!MASKED || (qw + FA2_QW - 1 < se_min[cur]));
             if (full_tile) {
 #pragma unroll
                 for (int nt = 0; nt < KVT; nt++)
