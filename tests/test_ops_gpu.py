@@ -399,3 +399,23 @@ def test_flashmask_irregular_packing_s1024(C):
                      (v.grad, vr.grad, "dv")):
         err = (g.float() - gr).abs().max().item()
         assert err < 8e-2, (n, err)
+
+
+def test_flashmask_bench_shape_gqa(C):
+    """Masked v2 at the training bench geometry (S=4096, GQA 32:8, 8
+    packed samples) — fwd vs fp32 reference."""
+    from paddlenlp_amd import ops
+
+    torch.manual_seed(5)
+    B, S, Hq, Hk, D = 1, 4096, 32, 8, 128
+    q = _bf16(torch.randn(B, S, Hq, D, device="cuda"))
+    k = _bf16(torch.randn(B, S, Hk, D, device="cuda"))
+    v = _bf16(torch.randn(B, S, Hk, D, device="cuda"))
+    se = torch.empty(B, 1, S, 1, dtype=torch.int32, device="cuda")
+    for i in range(8):
+        se[:, 0, i * 512:(i + 1) * 512, 0] = (i + 1) * 512
+    out = ops.flash_attention(q, k, v, causal=True, startend_row_indices=se)
+    ref = ops.reference.flash_attention(q.float(), k.float(), v.float(),
+                                        causal=True, startend_row_indices=se)
+    assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), \
+        (out.float() - ref).abs().max()
