@@ -227,3 +227,13 @@ def test_whole_word_mask_collator():
     assert (labels[0] == -100) == (labels[1] == -100)
     assert (labels[2] == -100) == (labels[3] == -100)
     assert (labels != -100).any()
+
+
+def test_token_classification_collator_pads_labels():
+    from paddlenlp_amd.data import DataCollatorForTokenClassification
+
+    feats = [{"input_ids": [5, 6, 7], "labels": [1, 2, 3]},
+             {"input_ids": [8], "labels": [0]}]
+    batch = DataCollatorForTokenClassification()(feats)
+    assert batch["input_ids"].shape == (2, 3)
+    assert batch["labels"][1].tolist() == [0, -100, -100]

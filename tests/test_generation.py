@@ -298,3 +298,28 @@ def test_generate_with_no_repeat_ngram(model):
             ok = False
         seen.add(pair)
     assert ok, full
+
+
+def test_forced_bos_in_generate(model):
+    """forced_bos_token_id pins the first generated token (the mBART
+    translation convention)."""
+    torch.manual_seed(0)
+    ids = torch.randint(0, 96, (2, 5))
+    out, _ = model.generate(
+        ids, GenerationConfig(max_new_tokens=4, do_sample=False,
+                              forced_bos_token_id=7, pad_token_id=0))
+    assert (out[:, 0] == 7).all()
+
+
+def test_bad_words_never_generated(model):
+    torch.manual_seed(0)
+    ids = torch.randint(0, 96, (1, 5))
+    # ban whatever greedy would pick first
+    base, _ = model.generate(ids, GenerationConfig(max_new_tokens=1,
+                                                   do_sample=False,
+                                                   pad_token_id=0))
+    banned = int(base[0, 0])
+    out, _ = model.generate(
+        ids, GenerationConfig(max_new_tokens=6, do_sample=False,
+                              bad_words_ids=[[banned]], pad_token_id=0))
+    assert banned not in out[0].tolist()
