@@ -122,7 +122,6 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     __shared__ ushort_t v_lds[2][BLKN * D];    // row-major [kv][d]
     __shared__ ushort_t kt_lds[2][D * BLKN];   // transposed [d][kv]
     __shared__ int se_lds[2][MASKED ? BLKN : 1];  // FlashMask bounds
-    __shared__ int se_max[2];                     // tile max (skip bound)
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -206,14 +205,9 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
     };
     auto write_tile = [&](int buf, int kv_base) {
         if (MASKED && tid < BLKN) {
-            int val = (kv_base + tid < Skv)
-                ? startend[(long long)b * Skv + kv_base + tid] : 0;
-            se_lds[buf][tid] = val;
-            int mx = val;
-#pragma unroll
-            for (int off = 32; off; off >>= 1)
-                mx = max(mx, __shfl_xor(mx, off, 64));
-            if (tid == 0) se_max[buf] = mx;
+            int g = kv_base + tid;
+            se_lds[buf][tid] =
+                (g < Skv) ? startend[(long long)b * Skv + g] : 0;
         }
         *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swzb2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
@@ -227,8 +221,19 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
         }
     };
 
+    // per-wave bound reduce (see flash_attn_v2.hip tile_bounds)
+    auto tile_se_max = [&](int kv_base) -> int {
+        int g = kv_base + lane;
+        int v = (g < Skv) ? startend[(long long)b * Skv + g] : 0;
+#pragma unroll
+        for (int off = 32; off; off >>= 1)
+            v = max(v, __shfl_xor(v, off, 64));
+        return v;
+    };
+
     load_tile(0);
     write_tile(0, 0);
+    int se_max_cur = MASKED ? tile_se_max(0) : 0x7fffffff;
     __syncthreads();
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
@@ -238,7 +243,7 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
             causal && (kv_base > qw + 31 + causal_off);
         // FlashMask whole-tile skip (see fwd): no bound exceeds this
         // wave's first q row -> nothing visible
-        if (MASKED && qw >= se_max[cur]) wave_skip = true;
+        if (MASKED && qw >= se_max_cur) wave_skip = true;
 
         // two 32-kv sub-iterations: one (S^T, dP^T) register pair live at
         // a time (register-pressure; see dkv kernel note)
@@ -289,10 +294,21 @@ __global__ __launch_bounds__(FB2_BLOCK) void flash_bwd2_dq_kernel(
         };
 
         if (!wave_skip) process_sub(0);
-        if (kvt + 1 < n_kv_tiles) load_tile(kv_base + BLKN);
+        int se_max_next = 0x7fffffff;
+        bool stage_skip = false;
+        if (kvt + 1 < n_kv_tiles) {
+            if (MASKED) {
+                se_max_next = tile_se_max(kv_base + BLKN);
+                // whole block past every bound: skip the K/V staging too
+                stage_skip = (q_base >= se_max_next);
+            }
+            if (!stage_skip) load_tile(kv_base + BLKN);
+        }
         if (!wave_skip) process_sub(1);
 
-        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1, kv_base + BLKN);
+        if (kvt + 1 < n_kv_tiles && !stage_skip)
+            write_tile(cur ^ 1, kv_base + BLKN);
+        if (MASKED) se_max_cur = se_max_next;
         __syncthreads();
     }
 
@@ -358,9 +374,13 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
     const int kvg_lane = kvw + l32;          // lane's kv column
     const int causal_off = Skv - Sq;
     // FlashMask: lane owns one kv column -> one bound scalar; the
-    // wave-max bound drives whole-q-tile skipping
+    // wave-max bound drives whole-q-tile skipping and the BLOCK max
+    // (published through LDS before the first barrier) lets the whole
+    // workgroup skip staging q tiles no kv row can see
+    __shared__ int blk_max_lds[MASKED ? FB2_WAVES : 1];
     int kv_end_lane = 0x7fffffff;
     int wave_max_end = 0x7fffffff;
+    int block_max_end = 0x7fffffff;
     if (MASKED) {
         kv_end_lane = (kvg_lane < Skv)
             ? startend[(long long)b * Skv + kvg_lane] : 0;
@@ -369,6 +389,7 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
         for (int off = 32; off; off >>= 1)
             wave_max_end = max(wave_max_end,
                                __shfl_xor(wave_max_end, off, 64));
+        if (lane == 0) blk_max_lds[wave] = wave_max_end;
     }
 
     const long long q_row_stride = (long long)Hq * D;
@@ -450,6 +471,14 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
         load_qtile(qt0 * BLKQ);
         write_qtile(qt0 * BLKQ, qt0 & 1);
         __syncthreads();
+        if (MASKED) {
+            // block max of the per-wave bounds (published above): q tiles
+            // wholly past it are invisible to EVERY kv row in the block
+            block_max_end = blk_max_lds[0];
+#pragma unroll
+            for (int w = 1; w < FB2_WAVES; w++)
+                block_max_end = max(block_max_end, blk_max_lds[w]);
+        }
 
         for (int qt = qt0; qt < n_q_tiles; qt++) {
             const int q_tb = qt * BLKQ;
@@ -521,10 +550,12 @@ __global__ __launch_bounds__(FB2_BLOCK, 1) void flash_bwd2_dkv_kernel(
                 }
             }
 
-            if (qt + 1 < n_q_tiles) {
+            if (qt + 1 < n_q_tiles &&
+                !(MASKED && q_tb + BLKQ >= block_max_end)) {
                 // no register prefetch (register file at capacity); the
                 // load+write go straight into the OTHER buffer, so only
-                // one barrier per tile
+                // one barrier per tile.  FlashMask: q tiles past the
+                // block's max bound skip the staging entirely
                 load_qtile(q_tb + BLKQ);
                 write_qtile(q_tb + BLKQ, cur ^ 1);
             }

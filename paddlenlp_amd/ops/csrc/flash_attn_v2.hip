@@ -116,9 +116,12 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     // per-tile FlashMask column bounds (staged with the K/V tile) and
     // the tile's max bound (wave 0 reduces it during staging) for
     // whole-tile skipping: a wave with qw >= max end sees nothing
+    // per-tile FlashMask column bounds (staged with the K/V tile);
+    // the tile min/max bounds travel in registers: every wave reduces
+    // the 64-int column itself (256 B from L2, 6 shfl_xor) one tile
+    // ahead, so whole-tile skips AND whole-block staging skips need no
+    // extra barrier
     __shared__ int se_lds[2][MASKED ? FA2_BLKN : 1];
-    __shared__ int se_max[2];   // tile max bound: whole-tile skip
-    __shared__ int se_min[2];   // tile min bound: full-tile fast path
 
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -200,16 +203,9 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
     };
     auto write_tile = [&](int buf, int kv_base) {
         if (MASKED && tid < FA2_BLKN) {
-            int val = (kv_base + tid < Skv)
-                ? startend[(long long)b * Skv + kv_base + tid] : 0;
-            se_lds[buf][tid] = val;
-            int mx = val, mn = val;
-#pragma unroll
-            for (int off = 32; off; off >>= 1) {
-                mx = max(mx, __shfl_xor(mx, off, 64));
-                mn = min(mn, __shfl_xor(mn, off, 64));
-            }
-            if (tid == 0) { se_max[buf] = mx; se_min[buf] = mn; }
+            int g = kv_base + tid;
+            se_lds[buf][tid] =
+                (g < Skv) ? startend[(long long)b * Skv + g] : 0;
         }
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0, s_col)) = sk0;
         *reinterpret_cast<short8v*>(swz2<D>(k_lds[buf], s_row0 + 1, s_col)) = sk1;
@@ -221,8 +217,25 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
         }
     };
 
+    // every wave reduces a tile's 64 bounds itself: returns (min, max)
+    // uniform across the wave, no shared-memory round trip
+    auto tile_bounds = [&](int kv_base, int& mn_out) -> int {
+        int g = kv_base + lane;
+        int v = (g < Skv) ? startend[(long long)b * Skv + g] : 0;
+        int mx = v, mn = (g < Skv) ? v : 0x7fffffff;
+#pragma unroll
+        for (int off = 32; off; off >>= 1) {
+            mx = max(mx, __shfl_xor(mx, off, 64));
+            mn = min(mn, __shfl_xor(mn, off, 64));
+        }
+        mn_out = mn;
+        return mx;
+    };
+
     load_tile(0);
     write_tile(0, 0);
+    int se_min_cur = 0, se_max_cur = 0x7fffffff;
+    if (MASKED) se_max_cur = tile_bounds(0, se_min_cur);
     __syncthreads();
 
     for (int kvt = 0; kvt < n_kv_tiles; kvt++) {
@@ -234,7 +247,7 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             causal && (kv_base > qw + FA2_QW - 1 + causal_off);
         // FlashMask whole-tile skip: every kv bound in this tile is at or
         // below the wave's first q row -> no visible pair
-        if (MASKED && qw >= se_max[cur]) wave_skip = true;
+        if (MASKED && qw >= se_max_cur) wave_skip = true;
 
         f32x16 st[KVT];
         if (!wave_skip) {
@@ -256,8 +269,18 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
         }
 
         // issue next tile's global loads now (T14): latency hides under
-        // softmax + PV
-        if (kvt + 1 < n_kv_tiles) load_tile(kv_base + FA2_BLKN);
+        // softmax + PV.  FlashMask: when the whole BLOCK's first q row
+        // clears the next tile's max bound, nothing will read it — skip
+        // the 32 KB K/V stage entirely (the bounds cost 256 B from L2)
+        int se_min_next = 0, se_max_next = 0x7fffffff;
+        bool stage_skip = false;
+        if (kvt + 1 < n_kv_tiles) {
+            if (MASKED) {
+                se_max_next = tile_bounds(kv_base + FA2_BLKN, se_min_next);
+                stage_skip = (q_base >= se_max_next);
+            }
+            if (!stage_skip) load_tile(kv_base + FA2_BLKN);
+        }
 
         if (!wave_skip) {
             // mask + scale.  A tile is "full" when every (q, kv) pair in this
@@ -266,7 +289,7 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
             const bool full_tile =
                 (kv_base + FA2_BLKN <= Skv) &&
                 (!causal || (kv_base + FA2_BLKN - 1 <= qw + causal_off)) &&
-                (!MASKED || (qw + FA2_QW - 1 < se_min[cur]));
+                (!MASKED || (qw + FA2_QW - 1 < se_min_cur));
             if (full_tile) {
 #pragma unroll
                 for (int nt = 0; nt < KVT; nt++)
@@ -381,7 +404,9 @@ __global__ __launch_bounds__(FA2_BLOCK) void flash_fwd2_kernel(
         // still be in flight on other waves, then one barrier: it both
         // publishes buf[cur^1] and guarantees every wave has finished
         // reading buf[cur] before iteration t+1 overwrites it
-        if (kvt + 1 < n_kv_tiles) write_tile(cur ^ 1, kv_base + FA2_BLKN);
+        if (kvt + 1 < n_kv_tiles && !stage_skip)
+            write_tile(cur ^ 1, kv_base + FA2_BLKN);
+        if (MASKED) { se_max_cur = se_max_next; se_min_cur = se_min_next; }
         __syncthreads();
     }
 
