@@ -181,3 +181,48 @@ def test_dpo_packed_equals_rowwise():
                                           for k, v in inputs.items()})
     assert torch.allclose(packed, rowwise, atol=1e-5), \
         (float(packed), float(rowwise))
+
+
+def test_dpo_packing_adaptive_default():
+    """Default packing decision follows the measured crossover: pack only
+    when the padded row form wastes >10%."""
+    import tempfile
+
+    import torch
+
+    from paddlenlp_amd.trainer import TrainingArguments
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+    from paddlenlp_amd.trl.dpo_trainer import DPOTrainer
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=96, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=128,
+                      use_flash_attention=True)
+    model = LlamaForCausalLM.from_config(cfg)
+
+    def mk(lc, lr):
+        ch = torch.randint(2, 96, (1, lc))
+        rj = torch.randint(2, 96, (1, lr))
+        return {"chosen_input_ids": ch, "chosen_labels": ch.clone(),
+                "rejected_input_ids": rj, "rejected_labels": rj.clone()}
+
+    calls = []
+    orig_forward = model.forward
+
+    def spy(*a, **kw):
+        calls.append("attn_mask_startend_row_indices" in kw
+                     and kw["attn_mask_startend_row_indices"] is not None)
+        return orig_forward(*a, **kw)
+
+    model.forward = spy
+    with tempfile.TemporaryDirectory() as d:
+        args = TrainingArguments(output_dir=d, max_steps=1,
+                                 per_device_train_batch_size=1)
+        tr = DPOTrainer(model=model, args=args, beta=0.1,
+                        loss_type="simpo")   # reference-free: one forward
+        # equal lengths: rowwise (no mask kwarg)
+        tr.compute_loss(model, mk(16, 16))
+        # heavy mismatch: packed (mask kwarg present)
+        tr.compute_loss(model, mk(8, 40))
+    assert calls[0] is False and calls[1] is True, calls
