@@ -823,3 +823,43 @@ def _w_zero2_world4_matches_single(rank, world):
 
 def test_zero2_world4_matches_single_process():
     _run_workers(_w_zero2_world4_matches_single, world_size=4)
+
+
+def _w_zero2_world8_smoke(rank, world):
+    """The N=8 SCALE-run topology (pure sharding, flat buckets, overlap)
+    at world 8: weights stay identical across ranks after 2 steps."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.zero import ZeroShardedEngine
+    from paddlenlp_amd.parallel.data_parallel import broadcast_parameters
+    from paddlenlp_amd.trainer.optimizer import FusedAdamW
+
+    topo = init_parallel_env(sharding_degree=world, backend="gloo")
+    torch.manual_seed(123)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.GELU(),
+                                torch.nn.Linear(16, 4))
+    broadcast_parameters(model, topo.sharding_parallel_group)
+    opt = FusedAdamW(model.parameters(), lr=1e-2, master_weights=True)
+    zero = ZeroShardedEngine(model, opt, stage=2,
+                             group=topo.sharding_parallel_group, bucket_mb=0)
+    zero.enable_overlap_comm()
+    g = torch.Generator().manual_seed(5 + rank)
+    for _ in range(2):
+        zero.zero_grad()
+        zero.overlap_active = True
+        x = torch.randn(2, 8, generator=g)
+        loss = model(x).pow(2).mean()
+        loss.backward()
+        zero.reduce_gradients_and_step_pre()
+        opt.step()
+        zero.step_post()
+    # all ranks must republish identical full weights
+    import torch.distributed as dist
+
+    for p in model.parameters():
+        ref = p.detach().clone()
+        dist.broadcast(ref, src=0, group=topo.sharding_parallel_group)
+        assert torch.allclose(p, ref, atol=1e-6)
+
+
+def test_zero2_world8_smoke():
+    _run_workers(_w_zero2_world8_smoke, world_size=8)
