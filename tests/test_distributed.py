@@ -863,3 +863,56 @@ def _w_zero2_world8_smoke(rank, world):
 
 def test_zero2_world8_smoke():
     _run_workers(_w_zero2_world8_smoke, world_size=8)
+
+
+def _w_zero2_odd_sizes(rank, world):
+    """Parameter sizes that don't divide by world*align exercise the flat
+    buckets' padding path; parity vs single-process must still hold."""
+    from paddlenlp_amd.parallel.topology import init_parallel_env
+    from paddlenlp_amd.parallel.zero import ZeroShardedEngine
+    from paddlenlp_amd.parallel.data_parallel import broadcast_parameters
+    from paddlenlp_amd.trainer.optimizer import FusedAdamW
+
+    topo = init_parallel_env(sharding_degree=world, backend="gloo")
+
+    def build():
+        torch.manual_seed(77)
+        # deliberately prime-ish shapes: 7, 13, 3 — none align to 64*world
+        return torch.nn.Sequential(
+            torch.nn.Linear(7, 13), torch.nn.Tanh(), torch.nn.Linear(13, 3))
+
+    torch.manual_seed(88)
+    xs = [torch.randn(world * 2, 7) for _ in range(3)]
+    ys = [torch.randn(world * 2, 3) for _ in range(3)]
+
+    model = build()
+    broadcast_parameters(model, topo.sharding_parallel_group)
+    opt = FusedAdamW(model.parameters(), lr=1e-2, master_weights=True)
+    zero = ZeroShardedEngine(model, opt, stage=2,
+                             group=topo.sharding_parallel_group, bucket_mb=0)
+    zero.enable_overlap_comm()
+    for x, y in zip(xs, ys):
+        zero.zero_grad()
+        zero.overlap_active = True
+        xl = x[rank * 2:(rank + 1) * 2]
+        yl = y[rank * 2:(rank + 1) * 2]
+        ((model(xl) - yl) ** 2).mean().backward()
+        zero.reduce_gradients_and_step_pre()
+        opt.step()
+        zero.step_post()
+
+    if rank == 0:
+        ref = build()
+        ref_opt = FusedAdamW(ref.parameters(), lr=1e-2, master_weights=True)
+        for x, y in zip(xs, ys):
+            ref_opt.zero_grad(set_to_none=True)
+            ((ref(x) - y) ** 2).mean().backward()
+            ref_opt.step()
+        for (n, p), (_, rp) in zip(model.named_parameters(),
+                                   ref.named_parameters()):
+            assert torch.allclose(p, rp, atol=1e-5), \
+                (n, (p - rp).abs().max().item())
+
+
+def test_zero2_odd_parameter_sizes():
+    _run_workers(_w_zero2_odd_sizes, world_size=3)
