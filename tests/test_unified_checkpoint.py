@@ -275,3 +275,24 @@ def test_tp2_checkpoint_loads_at_tp1(tmp_path):
         la = a(input_ids=ids)
         lb = b(input_ids=ids)
     assert torch.allclose(la, lb, atol=1e-6)
+
+
+def test_checkpoint_done_marker_gates_resume(tmp_path):
+    """get_last_checkpoint must ignore checkpoints without the integrity
+    marker (a crash mid-save leaves no .checkpoint_done)."""
+    import os
+
+    from paddlenlp_amd.trainer.trainer_utils import (
+        CHECKPOINT_DONE_MARKER,
+        get_last_checkpoint,
+    )
+
+    d = str(tmp_path)
+    for step, done in ((100, True), (200, False)):
+        ck = os.path.join(d, f"checkpoint-{step}")
+        os.makedirs(ck)
+        if done:
+            with open(os.path.join(ck, CHECKPOINT_DONE_MARKER), "w") as f:
+                f.write(str(step))
+    last = get_last_checkpoint(d)
+    assert last is not None and last.endswith("checkpoint-100"), last
