@@ -237,3 +237,24 @@ def test_token_classification_collator_pads_labels():
     batch = DataCollatorForTokenClassification()(feats)
     assert batch["input_ids"].shape == (2, 3)
     assert batch["labels"][1].tolist() == [0, -100, -100]
+
+
+def test_distributed_batch_sampler_consumed_samples_resume():
+    """Resume skips exactly the consumed samples on the first epoch
+    (reference consumed_samples replay, trainer.py:916-923)."""
+    from paddlenlp_amd.data.sampler import DistributedBatchSampler
+
+    ds = list(range(20))
+    fresh = DistributedBatchSampler(ds, batch_size=2, num_replicas=2,
+                                    rank=0, shuffle=False, drop_last=False,
+                                    seed=0)
+    all_batches = list(iter(fresh))
+    resumed = DistributedBatchSampler(ds, batch_size=2, num_replicas=2,
+                                      rank=0, shuffle=False, drop_last=False,
+                                      seed=0, consumed_samples=8)
+    rest = list(iter(resumed))
+    # 8 consumed over 2 replicas = 4 per rank = 2 batches of 2 skipped
+    assert rest == all_batches[2:], (rest, all_batches)
+    # second epoch: reset -> full pass again
+    resumed.consumed_samples = 0
+    assert list(iter(resumed)) == all_batches
