@@ -145,3 +145,22 @@ def test_special_token_registration_and_roundtrip(tmp_path):
     assert back.cls_token == "[CLS]" and back.mask_token == "[MASK]"
     assert "<|im_end|>" in back.additional_special_tokens
     assert back.convert_tokens_to_ids("<|im_start|>") == sid
+
+
+def test_chat_template_jinja_loop():
+    """A jinja template with a message loop renders through the jinja
+    backend (reference chat_template handling)."""
+    from paddlenlp_amd.transformers.tokenizer_utils import PretrainedTokenizer
+
+    tok = PretrainedTokenizer(
+        tokenizer=None,
+        chat_template=(
+            "{% for m in messages %}<|{{ m['role'] }}|>{{ m['content'] }}\n"
+            "{% endfor %}{% if add_generation_prompt %}<|assistant|>{% endif %}"
+        ))
+    text = tok.apply_chat_template(
+        [{"role": "user", "content": "hi"},
+         {"role": "assistant", "content": "hello"},
+         {"role": "user", "content": "bye?"}],
+        tokenize=False, add_generation_prompt=True)
+    assert text == "<|user|>hi\n<|assistant|>hello\n<|user|>bye?\n<|assistant|>"
