@@ -323,3 +323,20 @@ def test_bad_words_never_generated(model):
         ids, GenerationConfig(max_new_tokens=6, do_sample=False,
                               bad_words_ids=[[banned]], pad_token_id=0))
     assert banned not in out[0].tolist()
+
+
+def test_generation_config_roundtrip(tmp_path):
+    """save_pretrained drops non-serializable fields and round-trips the
+    rest."""
+    from paddlenlp_amd.generation import GenerationConfig
+
+    g = GenerationConfig(max_new_tokens=9, no_repeat_ngram_size=3,
+                         bad_words_ids=[[1, 2]], forced_eos_token_id=5,
+                         prefix_allowed_tokens_fn=lambda b, ids: [0])
+    g.save_pretrained(str(tmp_path))
+    loaded = GenerationConfig.from_pretrained(str(tmp_path))
+    assert loaded.max_new_tokens == 9
+    assert loaded.no_repeat_ngram_size == 3
+    assert loaded.bad_words_ids == [[1, 2]]
+    assert loaded.forced_eos_token_id == 5
+    assert loaded.prefix_allowed_tokens_fn is None
