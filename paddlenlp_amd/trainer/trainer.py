@@ -439,6 +439,9 @@ class Trainer:
                 sampler.consumed_samples = 0
                 sampler.set_epoch(int(epoch))
             self.control = self.callback_handler.on_epoch_end(args, self.state, self.control)
+            # epoch-strategy eval/save fire from on_epoch_end's control
+            # flags (reference runs the same hook after each epoch)
+            self._maybe_log_save_evaluate(tr_loss, model, start_time)
             if epoch >= num_train_epochs:
                 done = True
 

@@ -221,3 +221,25 @@ def test_log_save_metrics_and_state():
         assert merged["eval_loss"] == 1.5 and merged["train_loss"] == 2.0
         tr.save_state()
         assert os.path.isfile(os.path.join(d, "trainer_state.json"))
+
+
+def test_epoch_strategies_with_load_best():
+    """The SFT-preset pattern: evaluation_strategy=save_strategy=epoch +
+    load_best_model_at_end works end to end."""
+    with tempfile.TemporaryDirectory() as d:
+        ds = RandDS(n=24, s=16)
+        model = tiny_model()
+        args = make_args(
+            d, max_steps=-1, num_train_epochs=3, save_steps=0,
+            evaluation_strategy="epoch", save_strategy="epoch",
+            load_best_model_at_end=True, metric_for_best_model="eval_loss",
+            save_total_limit=2, learning_rate=3e-3,
+            per_device_train_batch_size=4)
+        tr = Trainer(model=model, args=args, train_dataset=ds,
+                     eval_dataset=RandDS(n=8, s=16, seed=3))
+        tr.train()
+        # 3 epochs -> 3 evals recorded; best checkpoint exists and loaded
+        evals = [l for l in tr.state.log_history if "eval_loss" in l]
+        assert len(evals) == 3, tr.state.log_history
+        assert tr.state.best_model_checkpoint is not None
+        assert os.path.isdir(tr.state.best_model_checkpoint)
