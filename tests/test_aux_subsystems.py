@@ -229,3 +229,39 @@ def test_fp16_trainer_skips_overflow_steps(tmp_path):
     changed = any(not torch.equal(p, w_before[n])
                   for n, p in model.named_parameters())
     assert changed
+
+
+def test_save_strategy_no_never_saves():
+    """save_strategy='no' writes no checkpoints even with save_steps set
+    — preset-compat guard."""
+    import tempfile
+
+    import torch
+
+    from paddlenlp_amd.trainer import Trainer, TrainingArguments
+    from paddlenlp_amd.transformers import LlamaConfig, LlamaForCausalLM
+    from torch.utils.data import Dataset
+
+    class DS(Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            ids = torch.randint(0, 64, (9,),
+                                generator=torch.Generator().manual_seed(i))
+            return {"input_ids": ids[:-1], "labels": ids[1:]}
+
+    cfg = LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=32)
+    with tempfile.TemporaryDirectory() as d:
+        args = TrainingArguments(output_dir=d, max_steps=4, save_steps=1,
+                                 save_strategy="no",
+                                 per_device_train_batch_size=2,
+                                 logging_steps=100)
+        tr = Trainer(model=LlamaForCausalLM.from_config(cfg), args=args,
+                     train_dataset=DS())
+        tr.train()
+        import os
+
+        assert not any(x.startswith("checkpoint-") for x in os.listdir(d))
