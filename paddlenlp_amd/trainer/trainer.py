@@ -906,6 +906,8 @@ class Trainer:
         dataloader = self.get_eval_dataloader(eval_dataset)
         model.eval()
         losses = []
+        collect = self.compute_metrics is not None
+        all_logits, all_labels = [], []
         start = time.time()
         n_samples = 0
         with torch.no_grad():
@@ -913,7 +915,16 @@ class Trainer:
                 if 0 < self.args.max_evaluate_steps <= i:
                     break
                 inputs = self._prepare_inputs(inputs)
-                loss = self.compute_loss(model, inputs)
+                if collect:
+                    loss, outputs = self.compute_loss(model, inputs,
+                                                      return_outputs=True)
+                    logits = outputs[1] if isinstance(outputs, tuple) else outputs
+                    if isinstance(logits, torch.Tensor):
+                        all_logits.append(logits.detach().float().cpu())
+                    if "labels" in inputs:
+                        all_labels.append(inputs["labels"].detach().cpu())
+                else:
+                    loss = self.compute_loss(model, inputs)
                 losses.append(loss.float())
                 n_samples += next(iter(inputs.values())).shape[0]
         model.train()
@@ -922,6 +933,15 @@ class Trainer:
             dist.all_reduce(mean_loss, group=self.topology.data_parallel_group)
             mean_loss /= dist.get_world_size(self.topology.data_parallel_group)
         metrics = {f"{metric_key_prefix}_loss": mean_loss.item()}
+        if collect and all_logits:
+            # reference evaluation_loop hands (predictions, labels) to
+            # compute_metrics and prefixes the result
+            extra = self.compute_metrics(
+                (torch.cat(all_logits),
+                 torch.cat(all_labels) if all_labels else None))
+            for k, v in (extra or {}).items():
+                key = k if k.startswith(metric_key_prefix) else                     f"{metric_key_prefix}_{k}"
+                metrics[key] = v
         try:
             metrics[f"{metric_key_prefix}_ppl"] = math.exp(mean_loss.item())
         except OverflowError:

@@ -243,3 +243,25 @@ def test_epoch_strategies_with_load_best():
         assert len(evals) == 3, tr.state.log_history
         assert tr.state.best_model_checkpoint is not None
         assert os.path.isdir(tr.state.best_model_checkpoint)
+
+
+def test_load_best_by_custom_metric():
+    """metric_for_best_model='accuracy' (greater_is_better inferred True)
+    tracks the best checkpoint via compute_metrics output."""
+    with tempfile.TemporaryDirectory() as d:
+        ds = RandDS(n=16, s=16)
+        vals = iter([0.2, 0.8, 0.5])
+
+        def metrics_fn(pack):
+            return {"accuracy": next(vals)}
+
+        args = make_args(d, max_steps=6, save_steps=2, eval_steps=2,
+                         evaluation_strategy="steps",
+                         load_best_model_at_end=True,
+                         metric_for_best_model="accuracy")
+        tr = Trainer(model=tiny_model(), args=args, train_dataset=ds,
+                     eval_dataset=RandDS(n=8, s=16, seed=2),
+                     compute_metrics=metrics_fn)
+        tr.train()
+        assert tr.state.best_metric == 0.8
+        assert tr.state.best_model_checkpoint.endswith("checkpoint-4")
