@@ -175,7 +175,9 @@ class DefaultFlowCallback(TrainerCallback):
     """Decides when to log / evaluate / save (reference :432)."""
 
     def on_step_end(self, args, state, control, **kwargs):
-        if args.logging_steps > 0 and state.global_step % args.logging_steps == 0:
+        if (getattr(args, "logging_strategy", "steps") == "steps"
+                and args.logging_steps > 0
+                and state.global_step % args.logging_steps == 0):
             control.should_log = True
         if (
             args.evaluation_strategy == "steps"
@@ -194,6 +196,8 @@ class DefaultFlowCallback(TrainerCallback):
         return control
 
     def on_epoch_end(self, args, state, control, **kwargs):
+        if getattr(args, "logging_strategy", "steps") == "epoch":
+            control.should_log = True
         if args.evaluation_strategy == "epoch":
             control.should_evaluate = True
         if args.save_strategy == "epoch":

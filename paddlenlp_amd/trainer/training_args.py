@@ -50,6 +50,7 @@ class TrainingArguments:
     logging_steps: int = 10
     logging_dir: Optional[str] = None
     save_strategy: str = "steps"
+    logging_strategy: str = "steps"
     save_steps: int = 500
     save_total_limit: Optional[int] = None
     evaluation_strategy: str = "no"
