@@ -33,6 +33,35 @@ from utils.data import convert_example
 
 @dataclass
 class FinetuneArguments(TrainingArguments):
+    # reference-preset compat: generation-based eval (BLEU/Rouge over
+    # generate()) — loss/accuracy eval runs when False (the default in
+    # most presets); True falls back to loss-eval with a warning
+    eval_with_do_generation: bool = False
+    # benchmark-mode flags (reference uses them to strip callbacks for
+    # throughput runs; accepted, benchmarking here is tools/bench_sft.py)
+    benchmark: bool = False
+    autotuner_benchmark: bool = False
+    # PTQ flow flags (reference QuantArgument; wired to
+    # trainer_compress.post_training_quantization)
+    do_ptq: bool = False
+    ptq_step: int = 32
+    do_gptq: bool = False
+    gptq_step: int = 8
+    smooth: bool = False
+    smooth_step: int = 16
+    smooth_all_linears: bool = False
+    smooth_piecewise_search: bool = False
+    smooth_k_piece: int = 3
+    smooth_search_piece: bool = False
+    auto_clip: bool = False
+    autoclip_step: int = 8
+    quant_type: str = "a8w8"
+    weight_quant_method: str = "abs_max_channel_wise"
+    act_quant_method: str = "avg"
+    cachekv_quant_method: str = "abs_max_headwise"
+    do_awq: bool = False
+    # resume base-model training from an SFT checkpoint (reference flag)
+    continue_training: bool = False
     pass
 
 
@@ -41,6 +70,11 @@ class ModelArgument:
     model_name_or_path: str = field(default=None)
     lora: bool = field(default=False)
     lora_rank: int = field(default=8)
+    # reference-preset compat (rsLoRA+ scales the B learning rate)
+    rslora_plus: bool = field(default=False)
+    pissa: bool = field(default=False)
+    vera: bool = field(default=False)
+    vera_rank: int = field(default=8)
     lora_alpha: float = field(default=16.0)
     lora_dropout: float = field(default=0.0)
     rslora: bool = field(default=False)
@@ -60,6 +94,10 @@ class DataArgument:
     # (reference run_finetune DataArgument src_length)
     src_length: int = field(default=1024)
     zero_padding: bool = field(default=False)
+    # reference-preset compat: lazy dataset loading is the default here
+    # (map-style loading is strict), the flag is accepted for parity
+    lazy: bool = field(default=False)
+    pad_to_multiple_of: int = field(default=0)
 
 
 def main():

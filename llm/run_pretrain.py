@@ -42,6 +42,8 @@ from paddlenlp_amd.utils.log import logger
 @dataclass
 class PreTrainingArguments(TrainingArguments):
     min_learning_rate: float = field(default=1e-5)
+    # start from a released checkpoint instead of random init (reference)
+    continue_training: bool = field(default=False)
     decay_steps: int = field(default=0)
 
     def __post_init__(self):
