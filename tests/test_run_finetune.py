@@ -88,3 +88,12 @@ def test_run_finetune_qlora(sft_setup):
     out = _run_finetune(sft_setup, ("--lora", "true", "--lora_rank", "4",
                                     "--weight_quantize_algo", "nf4"))
     assert (out / "lora_model_state.safetensors").is_file()
+
+
+def test_run_finetune_ptq_flow(sft_setup):
+    """do_ptq calibrates + swaps linears + saves a quantized state dict
+    (reference apply_ptq flow)."""
+    out = _run_finetune(sft_setup,
+                        extra=["--do_ptq", "1", "--ptq_step", "2",
+                               "--act_quant_method", "avg"])
+    assert (out / "ptq" / "quantized_model.pt").is_file()
